@@ -1,0 +1,190 @@
+// Paged-attention decode (one new token per sequence) for gfx950.
+//
+// Semantics: kubeai_amd/ops/ref.py::paged_attention_decode.
+//
+// Design (MI355X-first, memory-bound regime — see guide Appendix B):
+//  - workgroup = (seq, kv_head); 4 waves; GQA group of G = n_q/n_kv query
+//    heads computed together so each KV byte is read from HBM exactly once.
+//  - waves process DISJOINT cache blocks (flash-decoding split): wave w takes
+//    blocks w, w+4, ... with per-wave online-softmax state, merged at the end
+//    through LDS (m*, l*, o* combine). This keeps the chip busy at small
+//    decode batches (B*n_kv*4 waves in flight).
+//  - each wave stages its 16-token KV tile into its own LDS buffer with
+//    ushort8 (16 B) vector loads; score/PV reads from LDS are 4 B/lane,
+//    2-way bank aliased (free on CDNA4, guide §6 G4).
+//  - lane d-ownership: lane l owns head-dim elements {2l, 2l+1}; per-token
+//    scores via 64-lane butterfly shfl reduction (result broadcast to all
+//    lanes, feeding the online update without extra exchanges).
+#include <torch/extension.h>
+
+#include "common.h"
+
+namespace {
+
+constexpr int kWaves = 4;
+constexpr int kBlockThreads = kWaves * WAVE_SIZE;
+constexpr int kBS = 16;   // cache block size (tokens)
+constexpr int kHD = 128;  // head dim
+
+template <int G>
+__launch_bounds__(kBlockThreads) __global__ void paged_decode_kernel(
+    ushort* __restrict__ out,            // [B, n_q, hd] bf16
+    const ushort* __restrict__ q,        // [B, n_q, hd] bf16
+    const ushort* __restrict__ k_cache,  // [nb, n_kv, bs, hd]
+    const ushort* __restrict__ v_cache,
+    const int32_t* __restrict__ block_tables,  // [B, max_blocks]
+    const int32_t* __restrict__ seq_lens,      // [B]
+    const float scale, const int n_kv, const int max_blocks) {
+  const int b = blockIdx.x / n_kv;
+  const int kh = blockIdx.x % n_kv;
+  const int n_q = n_kv * G;
+  const int wave = threadIdx.x / WAVE_SIZE;
+  const int lane = threadIdx.x % WAVE_SIZE;
+  const int L = seq_lens[b];
+  const int n_blocks = (L + kBS - 1) / kBS;
+
+  // LDS: per-wave KV tile buffers + merge scratch
+  __shared__ ushort k_lds[kWaves][kBS][kHD];
+  __shared__ ushort v_lds[kWaves][kBS][kHD];
+  __shared__ float merge_o[kWaves][G][kHD];
+  __shared__ float merge_ml[kWaves][G][2];
+
+  // load q for all G heads of this group: lane owns elems {2l, 2l+1}
+  float qreg[G][2];
+#pragma unroll
+  for (int g = 0; g < G; ++g) {
+    const ushort* qh = q + ((int64_t)b * n_q + kh * G + g) * kHD;
+    qreg[g][0] = bf16_to_f32(qh[2 * lane]) * scale;
+    qreg[g][1] = bf16_to_f32(qh[2 * lane + 1]) * scale;
+  }
+
+  float m[G], l[G], o[G][2];
+#pragma unroll
+  for (int g = 0; g < G; ++g) {
+    m[g] = -INFINITY;
+    l[g] = 0.f;
+    o[g][0] = o[g][1] = 0.f;
+  }
+
+  const int32_t* bt = block_tables + (int64_t)b * max_blocks;
+
+  for (int blk_i = wave; blk_i < n_blocks; blk_i += kWaves) {
+    const int64_t blk = bt[blk_i];
+    const int tile_start = blk_i * kBS;
+    const int tile_len = min(kBS, L - tile_start);
+
+    // stage K/V tile: 16 tok x 128 elems = 2048 elems = 32 ushort8 per wave
+    {
+      const ushort8* src_k = reinterpret_cast<const ushort8*>(
+          k_cache + ((blk * n_kv + kh) * kBS) * kHD);
+      const ushort8* src_v = reinterpret_cast<const ushort8*>(
+          v_cache + ((blk * n_kv + kh) * kBS) * kHD);
+      ushort8* dst_k = reinterpret_cast<ushort8*>(&k_lds[wave][0][0]);
+      ushort8* dst_v = reinterpret_cast<ushort8*>(&v_lds[wave][0][0]);
+#pragma unroll
+      for (int i = 0; i < (kBS * kHD / 8) / WAVE_SIZE; ++i) {
+        dst_k[lane + i * WAVE_SIZE] = src_k[lane + i * WAVE_SIZE];
+        dst_v[lane + i * WAVE_SIZE] = src_v[lane + i * WAVE_SIZE];
+      }
+    }
+    // per-wave staging: compiler inserts lgkmcnt/vmcnt waits before LDS reads
+
+    for (int t = 0; t < tile_len; ++t) {
+      // lane reads its 2 K elems of token t (4B, 2-way bank aliased = free)
+      const uint32_t kk =
+          *reinterpret_cast<const uint32_t*>(&k_lds[wave][t][2 * lane]);
+      const float k0 = bf16_to_f32((ushort)(kk & 0xffff));
+      const float k1 = bf16_to_f32((ushort)(kk >> 16));
+      const uint32_t vv =
+          *reinterpret_cast<const uint32_t*>(&v_lds[wave][t][2 * lane]);
+      const float v0 = bf16_to_f32((ushort)(vv & 0xffff));
+      const float v1 = bf16_to_f32((ushort)(vv >> 16));
+#pragma unroll
+      for (int g = 0; g < G; ++g) {
+        float s = wave_reduce_sum(qreg[g][0] * k0 + qreg[g][1] * k1);
+        // online softmax update (q pre-scaled)
+        const float m_new = fmaxf(m[g], s);
+        const float corr = __expf(m[g] - m_new);
+        const float p = __expf(s - m_new);
+        l[g] = l[g] * corr + p;
+        o[g][0] = o[g][0] * corr + p * v0;
+        o[g][1] = o[g][1] * corr + p * v1;
+        m[g] = m_new;
+      }
+    }
+  }
+
+  // cross-wave merge through LDS
+#pragma unroll
+  for (int g = 0; g < G; ++g) {
+    merge_o[wave][g][2 * lane] = o[g][0];
+    merge_o[wave][g][2 * lane + 1] = o[g][1];
+    if (lane == 0) {
+      merge_ml[wave][g][0] = m[g];
+      merge_ml[wave][g][1] = l[g];
+    }
+  }
+  __syncthreads();
+
+  // wave w finalizes heads g = w, w+4, ... (for G<4, waves w>=G idle here)
+  for (int g = wave; g < G; g += kWaves) {
+    float m_star = -INFINITY;
+#pragma unroll
+    for (int w = 0; w < kWaves; ++w) m_star = fmaxf(m_star, merge_ml[w][g][0]);
+    float l_star = 0.f, o0 = 0.f, o1 = 0.f;
+#pragma unroll
+    for (int w = 0; w < kWaves; ++w) {
+      const float c = __expf(merge_ml[w][g][0] - m_star);
+      l_star += merge_ml[w][g][1] * c;
+      o0 += merge_o[w][g][2 * lane] * c;
+      o1 += merge_o[w][g][2 * lane + 1] * c;
+    }
+    const float inv_l = 1.0f / l_star;
+    ushort* oh = out + ((int64_t)b * n_q + kh * G + g) * kHD;
+    uint32_t packed = ((uint32_t)f32_to_bf16(o1 * inv_l) << 16) |
+                      f32_to_bf16(o0 * inv_l);
+    *reinterpret_cast<uint32_t*>(&oh[2 * lane]) = packed;
+  }
+}
+
+}  // namespace
+
+void paged_attention_decode(torch::Tensor out, torch::Tensor q,
+                            torch::Tensor k_cache, torch::Tensor v_cache,
+                            torch::Tensor block_tables, torch::Tensor seq_lens,
+                            double scale) {
+  TORCH_CHECK(q.is_contiguous() && out.is_contiguous());
+  TORCH_CHECK(k_cache.is_contiguous() && v_cache.is_contiguous());
+  TORCH_CHECK(q.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(block_tables.scalar_type() == torch::kInt32);
+  TORCH_CHECK(seq_lens.scalar_type() == torch::kInt32);
+  const int B = q.size(0), n_q = q.size(1), hd = q.size(2);
+  const int n_kv = k_cache.size(1);
+  const int max_blocks = block_tables.size(1);
+  TORCH_CHECK(hd == kHD, "decode kernel supports head_dim=128");
+  TORCH_CHECK(k_cache.size(2) == kBS, "decode kernel supports block_size=16");
+  const int G = n_q / n_kv;
+  TORCH_CHECK(n_q % n_kv == 0);
+  if (B == 0) return;
+  dim3 grid(B * n_kv), block(kBlockThreads);
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+#define LAUNCH(GG)                                                        \
+  hipLaunchKernelGGL((paged_decode_kernel<GG>), grid, block, 0, stream,   \
+                     (ushort*)out.data_ptr(), (const ushort*)q.data_ptr(), \
+                     (const ushort*)k_cache.data_ptr(),                    \
+                     (const ushort*)v_cache.data_ptr(),                    \
+                     block_tables.data_ptr<int32_t>(),                     \
+                     seq_lens.data_ptr<int32_t>(), (float)scale, n_kv,     \
+                     max_blocks)
+  switch (G) {
+    case 1: LAUNCH(1); break;
+    case 2: LAUNCH(2); break;
+    case 4: LAUNCH(4); break;
+    case 5: LAUNCH(5); break;
+    case 6: LAUNCH(6); break;
+    case 8: LAUNCH(8); break;
+    default: TORCH_CHECK(false, "unsupported GQA group size ", G);
+  }
+#undef LAUNCH
+  HIP_CHECK_KERNEL();
+}

@@ -1,0 +1,279 @@
+// Paged flash-attention prefill (causal, GQA, prefix-cache aware) for gfx950.
+//
+// Semantics: kubeai_amd/ops/ref.py::paged_attention_prefill — new (chunk)
+// query tokens attend to the full paged KV cache (which already contains
+// this chunk's K/V via reshape_and_cache).
+//
+// MI355X-first structure (guide §3/§5/§6):
+//  - workgroup = (seq, kv_head, q_tile16); one wave per q-head of the GQA
+//    group (G waves), so the staged KV tile in LDS is consumed by all G
+//    heads — each KV byte crosses HBM once per 16*G query rows.
+//  - QK^T and PV on v_mfma_f32_16x16x32_bf16; fp32 accumulation in AGPRs.
+//  - K tile [32][128] and transposed V tile [128][32] staged in LDS with
+//    the XOR swizzle (byte ^= (row&7)<<4) -> conflict-free ds_read_b128
+//    B-fragment reads (guide §6 Guideline 4).
+//  - online softmax entirely in C-fragment registers; the row (=q) direction
+//    lives in the low 4 lane bits, so row max/sum are 4-step shfl_xor
+//    butterflies; P goes through a 1 KiB/wave swizzled LDS round-trip to
+//    reach A-fragment layout for PV.
+//
+// Fragment layouts (verified by tests/test_kernels_gpu.py numerics):
+//   A[M=16][K=32]: lane l holds A[l&15][(l>>4)*8 + j], j=0..7 (bf16)
+//   B[K=32][N=16]: lane l holds B[(l>>4)*8 + j][l&15]
+//   C[M=16][N=16]: lane l holds C[(l>>4)*4 + i][l&15], i=0..3 (fp32)
+#include <torch/extension.h>
+
+#include "common.h"
+
+namespace {
+
+constexpr int kHD = 128;   // head dim
+constexpr int kBS = 16;    // cache block size
+constexpr int kQB = 16;    // query rows per tile
+constexpr int kKVB = 32;   // kv tokens per tile (2 cache blocks)
+
+typedef __attribute__((ext_vector_type(8))) short bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+DEV_INLINE int swz(int row, int byte_off) {
+  return byte_off ^ ((row & 7) << 4);
+}
+
+template <int G>
+__launch_bounds__(G * WAVE_SIZE) __global__ void paged_prefill_kernel(
+    ushort* __restrict__ out,            // [Tq, n_q, hd]
+    const ushort* __restrict__ q,        // [Tq, n_q, hd]
+    const ushort* __restrict__ k_cache,  // [nb, n_kv, bs, hd]
+    const ushort* __restrict__ v_cache,
+    const int32_t* __restrict__ block_tables,     // [B, max_blocks]
+    const int32_t* __restrict__ query_start_loc,  // [B+1]
+    const int32_t* __restrict__ seq_lens,         // [B]
+    const float scale, const int n_kv, const int max_blocks,
+    const int n_qtiles_max) {
+  const int b = blockIdx.x;
+  const int kh = blockIdx.y;
+  const int qtile = blockIdx.z;
+  const int n_q = n_kv * G;
+  const int wave = threadIdx.x / WAVE_SIZE;  // = q head within group
+  const int lane = threadIdx.x % WAVE_SIZE;
+  const int head = kh * G + wave;
+
+  const int s0 = query_start_loc[b];
+  const int q_len = query_start_loc[b + 1] - s0;
+  if (qtile * kQB >= q_len) return;  // whole workgroup exits together
+  const int L = seq_lens[b];
+  const int ctx = L - q_len;  // absolute position of query row 0
+
+  // rows this tile covers (clamped; invalid rows masked, never written)
+  const int row_lo = qtile * kQB;
+  const int n_rows = min(kQB, q_len - row_lo);
+  // causal kv limit for this tile: last row's abs position
+  const int kv_limit = ctx + row_lo + n_rows;  // exclusive
+  const int n_kv_tiles = (kv_limit + kKVB - 1) / kKVB;
+
+  __shared__ ushort k_lds[kKVB * kHD];          // [tok][hd], swizzled
+  __shared__ ushort v_lds[kHD * kKVB];          // [hd][tok], swizzled
+  __shared__ ushort p_lds[G][kQB * kKVB];       // [row][tok], swizzled
+
+  // ---- load Q fragments (4 K-chunks of 32) straight from global ----
+  bf16x8 q_frag[4];
+  {
+    const int qrow = min(row_lo + (lane & 15), q_len - 1);
+    const ushort* qp = q + ((int64_t)(s0 + qrow) * n_q + head) * kHD;
+#pragma unroll
+    for (int kc = 0; kc < 4; ++kc) {
+      q_frag[kc] =
+          *reinterpret_cast<const bf16x8*>(qp + kc * 32 + (lane >> 4) * 8);
+    }
+  }
+
+  f32x4 o_acc[kHD / 16];
+#pragma unroll
+  for (int c = 0; c < kHD / 16; ++c) o_acc[c] = {0.f, 0.f, 0.f, 0.f};
+  float m_run[4], l_run[4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    m_run[i] = -INFINITY;
+    l_run[i] = 0.f;
+  }
+
+  const int32_t* bt = block_tables + (int64_t)b * max_blocks;
+
+  for (int kt = 0; kt < n_kv_tiles; ++kt) {
+    const int kv_start = kt * kKVB;
+    const int kv_valid = min(kKVB, kv_limit - kv_start);  // tokens staged
+
+    // ---- cooperative staging: K -> k_lds, V -> v_lds transposed ----
+    __syncthreads();  // previous tile fully consumed
+    {
+      // each thread stages ceil(32*128/8 / nthreads) ushort8 vectors
+      const int nvec = kKVB * kHD / 8;  // 512
+      for (int i = threadIdx.x; i < nvec; i += G * WAVE_SIZE) {
+        const int tok = i / (kHD / 8);
+        const int col8 = i % (kHD / 8);
+        if (tok < kv_valid) {
+          const int abs_tok = kv_start + tok;
+          const int64_t blk = bt[abs_tok / kBS];
+          const ushort* src = k_cache +
+                              (((blk * n_kv + kh) * kBS) + abs_tok % kBS) * kHD;
+          // K: 16B vector write, swizzled row=tok
+          *reinterpret_cast<bf16x8*>(
+              reinterpret_cast<char*>(k_lds) +
+              swz(tok, tok * kHD * 2 + col8 * 16)) =
+              *reinterpret_cast<const bf16x8*>(src + col8 * 8);
+          // V: scatter-transpose 8 elems (row=hd, col=tok)
+          const ushort* vsrc = v_cache +
+                               (((blk * n_kv + kh) * kBS) + abs_tok % kBS) * kHD;
+          bf16x8 vv = *reinterpret_cast<const bf16x8*>(vsrc + col8 * 8);
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            const int hdrow = col8 * 8 + j;
+            *reinterpret_cast<ushort*>(
+                reinterpret_cast<char*>(v_lds) +
+                swz(hdrow, (hdrow * kKVB + tok) * 2)) = (ushort)vv[j];
+          }
+        }
+      }
+    }
+    __syncthreads();
+
+    // ---- S = Q.K^T over two 16-col subtiles ----
+    f32x4 s_frag[2];
+#pragma unroll
+    for (int nsub = 0; nsub < 2; ++nsub) {
+      s_frag[nsub] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int kc = 0; kc < 4; ++kc) {
+        const int tok = nsub * 16 + (lane & 15);
+        bf16x8 k_frag = *reinterpret_cast<const bf16x8*>(
+            reinterpret_cast<const char*>(k_lds) +
+            swz(tok, tok * kHD * 2 + kc * 64 + (lane >> 4) * 16));
+        s_frag[nsub] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            q_frag[kc], k_frag, s_frag[nsub], 0, 0, 0);
+      }
+    }
+
+    // ---- mask + online softmax in C-fragment layout ----
+    // lane holds rows r_i = (lane>>4)*4+i, cols (lane&15) and 16+(lane&15)
+    float p[2][4];  // [nsub][i] probabilities (pre-normalized)
+    float corr[4];
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      const int r = (lane >> 4) * 4 + i;
+      const int qpos = ctx + row_lo + r;
+      float s0v = s_frag[0][i] * scale;
+      float s1v = s_frag[1][i] * scale;
+      const int kv0 = kv_start + (lane & 15);
+      const int kv1 = kv0 + 16;
+      if (kv0 > qpos || r >= n_rows) s0v = -INFINITY;
+      if (kv1 > qpos || r >= n_rows) s1v = -INFINITY;
+      // row max across the 16 lanes of this row group
+      float rmax = fmaxf(s0v, s1v);
+#pragma unroll
+      for (int off = 8; off > 0; off >>= 1)
+        rmax = fmaxf(rmax, __shfl_xor(rmax, off, 64));
+      const float m_new = fmaxf(m_run[i], rmax);
+      // rows fully masked so far: keep m=-inf; avoid NaN from (-inf)-(-inf)
+      corr[i] = (m_run[i] == -INFINITY) ? 0.f : __expf(m_run[i] - m_new);
+      p[0][i] = (s0v == -INFINITY) ? 0.f : __expf(s0v - m_new);
+      p[1][i] = (s1v == -INFINITY) ? 0.f : __expf(s1v - m_new);
+      float rsum = p[0][i] + p[1][i];
+#pragma unroll
+      for (int off = 8; off > 0; off >>= 1) rsum += __shfl_xor(rsum, off, 64);
+      l_run[i] = l_run[i] * corr[i] + rsum;
+      m_run[i] = m_new;
+    }
+    // rescale O accumulator
+#pragma unroll
+    for (int c = 0; c < kHD / 16; ++c)
+#pragma unroll
+      for (int i = 0; i < 4; ++i) o_acc[c][i] *= corr[i];
+
+    // ---- P -> bf16 A-fragment via swizzled LDS round trip ----
+#pragma unroll
+    for (int nsub = 0; nsub < 2; ++nsub)
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        const int r = (lane >> 4) * 4 + i;
+        const int tok = nsub * 16 + (lane & 15);
+        *reinterpret_cast<ushort*>(
+            reinterpret_cast<char*>(p_lds[wave]) +
+            swz(r, (r * kKVB + tok) * 2)) = f32_to_bf16(p[nsub][i]);
+      }
+    // wave-local LDS dependency; compiler inserts the lgkmcnt wait
+    bf16x8 p_frag = *reinterpret_cast<const bf16x8*>(
+        reinterpret_cast<const char*>(p_lds[wave]) +
+        swz(lane & 15, ((lane & 15) * kKVB + (lane >> 4) * 8) * 2));
+
+    // ---- O += P.V ----
+#pragma unroll
+    for (int c = 0; c < kHD / 16; ++c) {
+      const int hdcol = c * 16 + (lane & 15);
+      bf16x8 v_frag = *reinterpret_cast<const bf16x8*>(
+          reinterpret_cast<const char*>(v_lds) +
+          swz(hdcol, (hdcol * kKVB + (lane >> 4) * 8) * 2));
+      o_acc[c] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(p_frag, v_frag,
+                                                         o_acc[c], 0, 0, 0);
+    }
+  }
+
+  // ---- write O / l ----
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    const int r = (lane >> 4) * 4 + i;
+    if (r >= n_rows) continue;
+    const float inv_l = 1.0f / l_run[i];
+    ushort* op = out + ((int64_t)(s0 + row_lo + r) * n_q + head) * kHD;
+#pragma unroll
+    for (int c = 0; c < kHD / 16; ++c)
+      op[c * 16 + (lane & 15)] = f32_to_bf16(o_acc[c][i] * inv_l);
+  }
+}
+
+}  // namespace
+
+void paged_attention_prefill(torch::Tensor out, torch::Tensor q,
+                             torch::Tensor k_cache, torch::Tensor v_cache,
+                             torch::Tensor block_tables,
+                             torch::Tensor query_start_loc,
+                             torch::Tensor seq_lens, double scale) {
+  TORCH_CHECK(q.is_contiguous() && out.is_contiguous());
+  TORCH_CHECK(k_cache.is_contiguous() && v_cache.is_contiguous());
+  TORCH_CHECK(q.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(block_tables.scalar_type() == torch::kInt32);
+  TORCH_CHECK(query_start_loc.scalar_type() == torch::kInt32);
+  TORCH_CHECK(seq_lens.scalar_type() == torch::kInt32);
+  const int n_q = q.size(1), hd = q.size(2);
+  const int n_kv = k_cache.size(1);
+  const int B = seq_lens.size(0);
+  const int max_blocks = block_tables.size(1);
+  TORCH_CHECK(hd == kHD, "prefill kernel supports head_dim=128");
+  TORCH_CHECK(k_cache.size(2) == kBS);
+  const int G = n_q / n_kv;
+  TORCH_CHECK(n_q % n_kv == 0);
+  if (q.size(0) == 0) return;
+  // max q tiles across sequences (host passes it via q len bound)
+  const int Tq = q.size(0);
+  const int n_qtiles_max = (Tq + kQB - 1) / kQB;  // upper bound; per-seq early exit
+  dim3 grid(B, n_kv, n_qtiles_max);
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+#define LAUNCH(GG)                                                         \
+  hipLaunchKernelGGL((paged_prefill_kernel<GG>), grid,                      \
+                     dim3(GG * WAVE_SIZE), 0, stream,                       \
+                     (ushort*)out.data_ptr(), (const ushort*)q.data_ptr(),  \
+                     (const ushort*)k_cache.data_ptr(),                     \
+                     (const ushort*)v_cache.data_ptr(),                     \
+                     block_tables.data_ptr<int32_t>(),                      \
+                     query_start_loc.data_ptr<int32_t>(),                   \
+                     seq_lens.data_ptr<int32_t>(), (float)scale, n_kv,      \
+                     max_blocks, n_qtiles_max)
+  switch (G) {
+    case 1: LAUNCH(1); break;
+    case 2: LAUNCH(2); break;
+    case 4: LAUNCH(4); break;
+    case 8: LAUNCH(8); break;
+    default: TORCH_CHECK(false, "unsupported GQA group size ", G);
+  }
+#undef LAUNCH
+  HIP_CHECK_KERNEL();
+}

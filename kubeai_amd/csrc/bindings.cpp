@@ -1,0 +1,40 @@
+// Python bindings for the kubeai_amd gfx950 HIP kernels.
+#include <torch/extension.h>
+
+void rmsnorm(torch::Tensor out, torch::Tensor x, torch::Tensor weight,
+             double eps);
+void fused_add_rmsnorm(torch::Tensor x, torch::Tensor residual,
+                       torch::Tensor weight, double eps);
+void rope(torch::Tensor q, torch::Tensor k, torch::Tensor positions,
+          torch::Tensor cos_sin);
+void reshape_and_cache(torch::Tensor k, torch::Tensor v, torch::Tensor k_cache,
+                       torch::Tensor v_cache, torch::Tensor slot_mapping);
+void paged_attention_decode(torch::Tensor out, torch::Tensor q,
+                            torch::Tensor k_cache, torch::Tensor v_cache,
+                            torch::Tensor block_tables, torch::Tensor seq_lens,
+                            double scale);
+void paged_attention_prefill(torch::Tensor out, torch::Tensor q,
+                             torch::Tensor k_cache, torch::Tensor v_cache,
+                             torch::Tensor block_tables,
+                             torch::Tensor query_start_loc,
+                             torch::Tensor seq_lens, double scale);
+void silu_and_mul(torch::Tensor out, torch::Tensor x);
+void greedy_sample(torch::Tensor out, torch::Tensor logits);
+void gumbel_sample(torch::Tensor out, torch::Tensor logits,
+                   torch::Tensor temperature, torch::Tensor seeds,
+                   int64_t step);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("rmsnorm", &rmsnorm, "RMSNorm (bf16, fp32 accum)");
+  m.def("fused_add_rmsnorm", &fused_add_rmsnorm,
+        "in-place residual add + RMSNorm");
+  m.def("rope", &rope, "NeoX rotary embedding, in-place q/k");
+  m.def("reshape_and_cache", &reshape_and_cache, "paged KV cache write");
+  m.def("paged_attention_decode", &paged_attention_decode,
+        "paged attention, one query token per seq");
+  m.def("paged_attention_prefill", &paged_attention_prefill,
+        "paged causal flash attention over cached KV");
+  m.def("silu_and_mul", &silu_and_mul, "SwiGLU activation");
+  m.def("greedy_sample", &greedy_sample, "argmax sampling");
+  m.def("gumbel_sample", &gumbel_sample, "temperature sampling (hash RNG)");
+}

@@ -1,0 +1,69 @@
+// Paged KV-cache write: scatter new K/V token vectors into cache blocks.
+//
+// Cache layout [num_blocks, n_kv, block_size, hd]: one (block, kv-head) tile
+// is block_size*hd contiguous elements, the coalesced read unit for decode.
+// Semantics: kubeai_amd/ops/ref.py::reshape_and_cache.
+#include <torch/extension.h>
+
+#include "common.h"
+
+namespace {
+
+__global__ void reshape_and_cache_kernel(
+    const ushort* __restrict__ k,   // [T, n_kv, hd]
+    const ushort* __restrict__ v,   // [T, n_kv, hd]
+    ushort* __restrict__ k_cache,   // [nb, n_kv, bs, hd]
+    ushort* __restrict__ v_cache,
+    const int64_t* __restrict__ slot_mapping,  // [T]
+    const int n_kv, const int bs, const int hd, const int64_t n_tok) {
+  // one wave per (token, kv_head); lane i copies 8 elems (hd=128 -> 2 vec/lane)
+  const int64_t flat = (int64_t)blockIdx.x * (blockDim.x / WAVE_SIZE) +
+                       threadIdx.x / WAVE_SIZE;
+  const int64_t tok = flat / n_kv;
+  if (tok >= n_tok) return;
+  const int h = (int)(flat % n_kv);
+  const int lane = threadIdx.x % WAVE_SIZE;
+  const int64_t slot = slot_mapping[tok];
+  if (slot < 0) return;
+  const int64_t blk = slot / bs;
+  const int off = (int)(slot % bs);
+
+  const ushort8* src_k =
+      reinterpret_cast<const ushort8*>(k + ((int64_t)tok * n_kv + h) * hd);
+  const ushort8* src_v =
+      reinterpret_cast<const ushort8*>(v + ((int64_t)tok * n_kv + h) * hd);
+  ushort8* dst_k = reinterpret_cast<ushort8*>(
+      k_cache + (((int64_t)blk * n_kv + h) * bs + off) * hd);
+  ushort8* dst_v = reinterpret_cast<ushort8*>(
+      v_cache + (((int64_t)blk * n_kv + h) * bs + off) * hd);
+  const int nvec = hd / 8;
+  for (int i = lane; i < nvec; i += WAVE_SIZE) {
+    dst_k[i] = src_k[i];
+    dst_v[i] = src_v[i];
+  }
+}
+
+}  // namespace
+
+void reshape_and_cache(torch::Tensor k, torch::Tensor v, torch::Tensor k_cache,
+                       torch::Tensor v_cache, torch::Tensor slot_mapping) {
+  TORCH_CHECK(k.is_contiguous() && v.is_contiguous());
+  TORCH_CHECK(k_cache.is_contiguous() && v_cache.is_contiguous());
+  TORCH_CHECK(slot_mapping.scalar_type() == torch::kInt64);
+  const int T = k.size(0), n_kv = k.size(1), hd = k.size(2);
+  const int bs = k_cache.size(2);
+  TORCH_CHECK(hd % 8 == 0);
+  if (T == 0) return;
+  const int64_t total_waves = (int64_t)T * n_kv;
+  const int waves_per_block = 4;
+  const int64_t blocks = (total_waves + waves_per_block - 1) / waves_per_block;
+  // tail guard: slot_mapping index must stay in range
+  hipLaunchKernelGGL(reshape_and_cache_kernel, dim3((uint32_t)blocks),
+                     dim3(waves_per_block * WAVE_SIZE), 0,
+                     c10::hip::getCurrentHIPStream().stream(),
+                     (const ushort*)k.data_ptr(), (const ushort*)v.data_ptr(),
+                     (ushort*)k_cache.data_ptr(), (ushort*)v_cache.data_ptr(),
+                     slot_mapping.data_ptr<int64_t>(), n_kv, bs, hd,
+                     (int64_t)T);
+  HIP_CHECK_KERNEL();
+}

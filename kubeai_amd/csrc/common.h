@@ -1,0 +1,103 @@
+// Common helpers for kubeai_amd gfx950 (CDNA4) kernels.
+//
+// Conventions (see /opt/skills/guides/cdna_hip_programming.md):
+//  - wave = 64 lanes, hard-coded.
+//  - bf16 global loads are vectorized as ushort4/ushort8 (hipcc does not
+//    auto-vectorize scalar bf16 loads).
+//  - fp32 accumulation everywhere.
+#pragma once
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+#include <stdint.h>
+#include <c10/hip/HIPStream.h>
+
+#define WAVE_SIZE 64
+#define DEV_INLINE __device__ __forceinline__
+
+typedef __hip_bfloat16 bf16_t;
+
+// 16-byte vector of 8 bf16 values (one coalesced lane-load).
+typedef ushort __attribute__((ext_vector_type(8))) ushort8;
+typedef ushort __attribute__((ext_vector_type(4))) ushort4_t;
+typedef float __attribute__((ext_vector_type(4))) float4_t;
+
+DEV_INLINE float bf16_to_f32(ushort u) {
+  union {
+    uint32_t i;
+    float f;
+  } c;
+  c.i = ((uint32_t)u) << 16;
+  return c.f;
+}
+
+DEV_INLINE ushort f32_to_bf16(float f) {
+  // round-to-nearest-even, matching PyTorch's bf16 cast
+  union {
+    float f;
+    uint32_t i;
+  } c;
+  c.f = f;
+  uint32_t x = c.i;
+  uint32_t rb = ((x >> 16) & 1u) + 0x7fffu;
+  x += rb;
+  return (ushort)(x >> 16);
+}
+
+// ---------------- wave reductions (64-wide) ----------------
+
+DEV_INLINE float wave_reduce_sum(float v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v += __shfl_xor(v, off, 64);
+  return v;
+}
+
+DEV_INLINE float wave_reduce_max(float v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v = fmaxf(v, __shfl_xor(v, off, 64));
+  return v;
+}
+
+// Block-level sum reduce across up to 16 waves; `smem` must hold >= 16 floats.
+// Every thread returns the total.
+DEV_INLINE float block_reduce_sum(float v, float* smem) {
+  const int wave = threadIdx.x / WAVE_SIZE;
+  const int lane = threadIdx.x % WAVE_SIZE;
+  const int n_waves = (blockDim.x + WAVE_SIZE - 1) / WAVE_SIZE;
+  v = wave_reduce_sum(v);
+  if (lane == 0) smem[wave] = v;
+  __syncthreads();
+  float total = 0.f;
+#pragma unroll
+  for (int w = 0; w < 16; ++w)
+    if (w < n_waves) total += smem[w];
+  return total;
+}
+
+// ---------------- hash RNG (kernel/ref-identical spec) ----------------
+//
+// Sampling uses a counter-based hash RNG so the HIP kernel and the PyTorch
+// reference produce bit-identical streams: u64 = splitmix64(key*M ^ idx).
+DEV_INLINE uint64_t splitmix64(uint64_t z) {
+  z += 0x9e3779b97f4a7c15ull;
+  z = (z ^ (z >> 30)) * 0xbf58476d1ce4e5b9ull;
+  z = (z ^ (z >> 27)) * 0x94d049bb133111ebull;
+  return z ^ (z >> 31);
+}
+
+// uniform in (0, 1]
+DEV_INLINE float hash_uniform(uint64_t key, uint64_t idx) {
+  uint64_t h = splitmix64(key ^ (idx * 0xd1342543de82ef95ull));
+  // take top 24 bits -> (0,1] to keep -log well-defined
+  uint32_t m = (uint32_t)(h >> 40);
+  return ((float)m + 1.0f) * (1.0f / 16777216.0f);
+}
+
+#define HIP_CHECK_KERNEL()                                    \
+  do {                                                        \
+    hipError_t e = hipGetLastError();                         \
+    if (e != hipSuccess) {                                    \
+      TORCH_CHECK(false, "HIP kernel launch failed: ",        \
+                  hipGetErrorString(e));                      \
+    }                                                         \
+  } while (0)

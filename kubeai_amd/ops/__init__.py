@@ -1,0 +1,122 @@
+"""Op dispatch: hand-written gfx950 HIP kernels on GPU, torch reference on CPU.
+
+Policy (deliberate, see repo instructions): on a ROCm GPU the HIP extension is
+MANDATORY — if ``kubeai_amd._C`` failed to import, GPU-tensor calls raise
+instead of silently falling back to eager PyTorch.  CPU tensors always use the
+reference implementations (ref.py) so the engine logic is testable anywhere.
+"""
+from __future__ import annotations
+
+import torch
+
+from . import ref
+
+_C = None
+_C_IMPORT_ERROR: Exception | None = None
+try:  # built in-tree by `python setup.py build_ext --inplace` / __graft_entry__.build()
+    from kubeai_amd import _C  # type: ignore[attr-defined,no-redef]
+except Exception as e:  # pragma: no cover - exercised only when ext missing
+    _C_IMPORT_ERROR = e
+
+
+def have_hip_ext() -> bool:
+    return _C is not None
+
+
+def _require_ext() -> None:
+    if _C is None:
+        raise RuntimeError(
+            "kubeai_amd HIP extension (kubeai_amd._C) is not built but a GPU "
+            "tensor reached the ops layer. Build it with "
+            "`python setup.py build_ext --inplace` (PYTORCH_ROCM_ARCH=gfx950). "
+            f"Original import error: {_C_IMPORT_ERROR!r}"
+        )
+
+
+def rmsnorm(x: torch.Tensor, weight: torch.Tensor, eps: float) -> torch.Tensor:
+    if x.is_cuda:
+        _require_ext()
+        out = torch.empty_like(x)
+        _C.rmsnorm(out, x, weight, eps)
+        return out
+    return ref.rmsnorm(x, weight, eps)
+
+
+def fused_add_rmsnorm(x, residual, weight, eps: float):
+    """In-place on GPU: x <- rmsnorm(x + residual), residual <- x + residual."""
+    if x.is_cuda:
+        _require_ext()
+        _C.fused_add_rmsnorm(x, residual, weight, eps)
+        return x, residual
+    return ref.fused_add_rmsnorm(x, residual, weight, eps)
+
+
+def rope(q, k, positions, cos_sin):
+    """In-place on GPU; returns (q, k)."""
+    if q.is_cuda:
+        _require_ext()
+        _C.rope(q, k, positions, cos_sin)
+        return q, k
+    return ref.rope(q, k, positions, cos_sin)
+
+
+def reshape_and_cache(k, v, k_cache, v_cache, slot_mapping) -> None:
+    if k.is_cuda:
+        _require_ext()
+        _C.reshape_and_cache(k, v, k_cache, v_cache, slot_mapping)
+        return
+    ref.reshape_and_cache(k, v, k_cache, v_cache, slot_mapping)
+
+
+def paged_attention_decode(q, k_cache, v_cache, block_tables, seq_lens, scale: float):
+    if q.is_cuda:
+        _require_ext()
+        out = torch.empty_like(q)
+        _C.paged_attention_decode(
+            out, q, k_cache, v_cache, block_tables, seq_lens, scale
+        )
+        return out
+    return ref.paged_attention_decode(q, k_cache, v_cache, block_tables, seq_lens, scale)
+
+
+def paged_attention_prefill(
+    q, k_cache, v_cache, block_tables, query_start_loc, seq_lens, scale: float
+):
+    if q.is_cuda:
+        _require_ext()
+        out = torch.empty_like(q)
+        _C.paged_attention_prefill(
+            out, q, k_cache, v_cache, block_tables, query_start_loc, seq_lens, scale
+        )
+        return out
+    return ref.paged_attention_prefill(
+        q, k_cache, v_cache, block_tables, query_start_loc, seq_lens, scale
+    )
+
+
+def silu_and_mul(x):
+    if x.is_cuda:
+        _require_ext()
+        T, two_i = x.shape
+        out = torch.empty((T, two_i // 2), dtype=x.dtype, device=x.device)
+        _C.silu_and_mul(out, x)
+        return out
+    return ref.silu_and_mul(x)
+
+
+def greedy_sample(logits):
+    if logits.is_cuda:
+        _require_ext()
+        out = torch.empty(logits.shape[0], dtype=torch.int64, device=logits.device)
+        _C.greedy_sample(out, logits)
+        return out
+    return ref.greedy_sample(logits)
+
+
+def gumbel_sample(logits, temperature, seeds, step: int):
+    if logits.is_cuda:
+        _require_ext()
+        out = torch.empty(logits.shape[0], dtype=torch.int64, device=logits.device)
+        _C.gumbel_sample(out, logits, temperature, seeds, step)
+        return out
+    return ref.gumbel_sample(logits, temperature, seeds, step)

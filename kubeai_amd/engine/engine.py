@@ -1,0 +1,150 @@
+"""LLMEngine — the per-GPU serving engine facade.
+
+add_request() / step() / abort(); one step = schedule -> forward -> commit.
+This is the in-house replacement for the vLLM container the reference
+launches per Model pod.
+"""
+from __future__ import annotations
+
+import dataclasses
+import time
+from typing import Optional
+
+import torch
+
+from kubeai_amd.models.config import ModelArchConfig, PRESETS
+
+from .kvcache import BlockManager
+from .runner import ModelRunner
+from .scheduler import Request, RequestStatus, SamplingParams, Scheduler
+
+
+@dataclasses.dataclass
+class EngineConfig:
+    model: str = "llama-tiny"  # preset name or HF dir
+    device: str = "auto"  # auto -> cuda if available else cpu
+    dtype: str = "bfloat16"
+    block_size: int = 16
+    num_gpu_blocks: Optional[int] = None  # None -> profile from free HBM
+    gpu_memory_utilization: float = 0.90
+    max_num_seqs: int = 256
+    max_num_batched_tokens: int = 8192
+    max_model_len: int = 8192
+    enable_prefix_caching: bool = True
+    seed: int = 0
+
+    def resolve_arch(self) -> ModelArchConfig:
+        if self.model in PRESETS:
+            return PRESETS[self.model]
+        return ModelArchConfig.from_hf_config(self.model)
+
+    def resolve_device(self) -> str:
+        if self.device != "auto":
+            return self.device
+        return "cuda" if torch.cuda.is_available() else "cpu"
+
+
+@dataclasses.dataclass
+class RequestOutput:
+    request_id: str
+    new_token_ids: list[int]
+    finished: bool
+    finish_reason: Optional[str]  # "stop" | "length" | "abort"
+    num_prompt_tokens: int
+    num_cached_tokens: int
+    output_token_ids: list[int]
+
+
+class LLMEngine:
+    def __init__(self, cfg: EngineConfig, tp_group=None):
+        self.cfg = cfg
+        arch = cfg.resolve_arch()
+        self.arch = arch
+        device = cfg.resolve_device()
+        self.runner = ModelRunner(
+            arch,
+            device=device,
+            dtype=getattr(torch, cfg.dtype),
+            block_size=cfg.block_size,
+            num_gpu_blocks=cfg.num_gpu_blocks,
+            gpu_memory_utilization=cfg.gpu_memory_utilization,
+            seed=cfg.seed,
+            tp_group=tp_group,
+        )
+        self.block_manager = BlockManager(self.runner.num_blocks, cfg.block_size)
+        self.scheduler = Scheduler(
+            self.block_manager,
+            max_num_seqs=cfg.max_num_seqs,
+            max_num_batched_tokens=cfg.max_num_batched_tokens,
+            max_model_len=cfg.max_model_len,
+            enable_prefix_caching=cfg.enable_prefix_caching,
+        )
+        self.step_count = 0
+
+    # ------------------------------------------------------------------
+    def add_request(
+        self,
+        prompt_token_ids: list[int],
+        params: Optional[SamplingParams] = None,
+        request_id: Optional[str] = None,
+        lora_id: int = 0,
+    ) -> Request:
+        if params is None:
+            params = SamplingParams()
+        if not params.ignore_eos and self.arch.eos_token_id not in params.stop_token_ids:
+            params = dataclasses.replace(
+                params,
+                stop_token_ids=tuple(params.stop_token_ids)
+                + (self.arch.eos_token_id,),
+            )
+        if len(prompt_token_ids) >= self.cfg.max_model_len:
+            prompt_token_ids = prompt_token_ids[-(self.cfg.max_model_len - 1) :]
+        req = Request(prompt_token_ids, params, request_id=request_id, lora_id=lora_id)
+        self.scheduler.add_request(req)
+        return req
+
+    def abort_request(self, request_id: str) -> None:
+        self.scheduler.abort(request_id)
+
+    def has_work(self) -> bool:
+        return self.scheduler.has_work()
+
+    # ------------------------------------------------------------------
+    def step(self) -> list[RequestOutput]:
+        out = self.scheduler.schedule()
+        if out.is_empty:
+            return []
+        sampled = self.runner.execute(out, self.step_count)
+        self.step_count += 1
+        now = time.monotonic()
+        finished = self.scheduler.finish_step(out, sampled, now)
+        finished_ids = {r.request_id for r in finished}
+        results: list[RequestOutput] = []
+        for ss in out.all_seqs:
+            req = ss.req
+            if not ss.samples:
+                continue
+            tok = sampled[req.request_id]
+            results.append(
+                RequestOutput(
+                    request_id=req.request_id,
+                    new_token_ids=[tok],
+                    finished=req.request_id in finished_ids,
+                    finish_reason=req.status.value if req.status.finished else None,
+                    num_prompt_tokens=req.num_prompt_tokens,
+                    num_cached_tokens=req.num_cached_prompt_tokens,
+                    output_token_ids=req.output_token_ids,
+                )
+            )
+        return results
+
+    # ------------------------------------------------------------------
+    # metrics for /metrics + the autoscaler
+    def stats(self) -> dict:
+        return {
+            "num_running": self.scheduler.num_running,
+            "num_waiting": self.scheduler.num_waiting,
+            "kv_usage": self.block_manager.usage(),
+            "prefix_cache_hit_rate": self.block_manager.hit_rate(),
+            "step_count": self.step_count,
+        }

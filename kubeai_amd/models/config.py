@@ -1,0 +1,130 @@
+"""Model architecture configs.
+
+Mirrors the fields of HF `config.json` that the engine needs; presets cover
+the BASELINE.json configs (Llama-3 8B / 70B, plus tiny variants for tests).
+"""
+from __future__ import annotations
+
+import dataclasses
+import json
+import os
+
+
+@dataclasses.dataclass
+class ModelArchConfig:
+    architecture: str = "llama"
+    vocab_size: int = 128256
+    hidden_size: int = 4096
+    intermediate_size: int = 14336
+    num_hidden_layers: int = 32
+    num_attention_heads: int = 32
+    num_key_value_heads: int = 8
+    head_dim: int = 128
+    max_position_embeddings: int = 8192
+    rms_norm_eps: float = 1e-5
+    rope_theta: float = 500000.0
+    tie_word_embeddings: bool = False
+    bos_token_id: int = 128000
+    eos_token_id: int = 128001
+    # MoE (Mixtral-style); n_experts == 0 means dense MLP
+    num_local_experts: int = 0
+    num_experts_per_tok: int = 2
+
+    @property
+    def n_kv_heads(self) -> int:
+        return self.num_key_value_heads
+
+    @staticmethod
+    def from_hf_config(path: str) -> "ModelArchConfig":
+        with open(os.path.join(path, "config.json")) as f:
+            cfg = json.load(f)
+        hidden = cfg["hidden_size"]
+        n_heads = cfg["num_attention_heads"]
+        arch = (cfg.get("architectures") or ["LlamaForCausalLM"])[0]
+        eos = cfg.get("eos_token_id", 2)
+        if isinstance(eos, list):
+            eos = eos[0]
+        return ModelArchConfig(
+            architecture="mixtral" if "Mixtral" in arch else "llama",
+            vocab_size=cfg["vocab_size"],
+            hidden_size=hidden,
+            intermediate_size=cfg["intermediate_size"],
+            num_hidden_layers=cfg["num_hidden_layers"],
+            num_attention_heads=n_heads,
+            num_key_value_heads=cfg.get("num_key_value_heads", n_heads),
+            head_dim=cfg.get("head_dim", hidden // n_heads),
+            max_position_embeddings=cfg.get("max_position_embeddings", 8192),
+            rms_norm_eps=cfg.get("rms_norm_eps", 1e-5),
+            rope_theta=cfg.get("rope_theta", 10000.0),
+            tie_word_embeddings=cfg.get("tie_word_embeddings", False),
+            bos_token_id=cfg.get("bos_token_id", 1),
+            eos_token_id=eos,
+            num_local_experts=cfg.get("num_local_experts", 0),
+            num_experts_per_tok=cfg.get("num_experts_per_tok", 2),
+        )
+
+
+# Presets (BASELINE.json configs + test-size models)
+PRESETS: dict[str, ModelArchConfig] = {
+    "llama-3-8b": ModelArchConfig(),
+    "llama-3-70b": ModelArchConfig(
+        hidden_size=8192,
+        intermediate_size=28672,
+        num_hidden_layers=80,
+        num_attention_heads=64,
+        num_key_value_heads=8,
+    ),
+    # tiny: CPU tests / smoke; same head_dim=128 so HIP kernel paths match
+    "llama-tiny": ModelArchConfig(
+        vocab_size=2048,
+        hidden_size=256,
+        intermediate_size=512,
+        num_hidden_layers=2,
+        num_attention_heads=2,
+        num_key_value_heads=1,
+        head_dim=128,
+        max_position_embeddings=2048,
+        bos_token_id=1,
+        eos_token_id=2,
+    ),
+    # small: 1-GPU quick bench model (head_dim=128 to stay on the HIP path)
+    "llama-1b": ModelArchConfig(
+        vocab_size=128256,
+        hidden_size=2048,
+        intermediate_size=8192,
+        num_hidden_layers=16,
+        num_attention_heads=16,
+        num_key_value_heads=4,
+        head_dim=128,
+    ),
+    "mixtral-tiny": ModelArchConfig(
+        architecture="mixtral",
+        vocab_size=2048,
+        hidden_size=256,
+        intermediate_size=512,
+        num_hidden_layers=2,
+        num_attention_heads=2,
+        num_key_value_heads=1,
+        head_dim=128,
+        max_position_embeddings=2048,
+        bos_token_id=1,
+        eos_token_id=2,
+        num_local_experts=4,
+        num_experts_per_tok=2,
+    ),
+    "mixtral-8x7b": ModelArchConfig(
+        architecture="mixtral",
+        vocab_size=32000,
+        hidden_size=4096,
+        intermediate_size=14336,
+        num_hidden_layers=32,
+        num_attention_heads=32,
+        num_key_value_heads=8,
+        head_dim=128,
+        rope_theta=1000000.0,
+        bos_token_id=1,
+        eos_token_id=2,
+        num_local_experts=8,
+        num_experts_per_tok=2,
+    ),
+}

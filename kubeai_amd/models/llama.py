@@ -1,0 +1,199 @@
+"""Llama-family decoder for the kubeai_amd engine.
+
+MI355X-first execution model (one process per GPU):
+  - token-packed forward ([T, H], no padding) over the ForwardBatch;
+  - plain GEMMs (qkv/o/gate_up/down/lm_head) go through torch.nn.functional
+    .linear, i.e. hipBLASLt/rocBLAS on ROCm — library GEMMs per the design
+    rules; every fused hot op (rmsnorm, rope, cache write, paged attention,
+    SwiGLU) is a hand-written gfx950 kernel via kubeai_amd.ops;
+  - KV cache is paged [num_blocks, n_kv, block_size, head_dim] bf16, owned by
+    the runner and passed in.
+
+Reference parity: this is the in-house engine the reference delegates to
+vLLM pods for; no reference code exists for it.
+"""
+from __future__ import annotations
+
+import math
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from kubeai_amd import ops
+from kubeai_amd.engine.batch import ForwardBatch
+from .config import ModelArchConfig
+
+
+class Attention(nn.Module):
+    def __init__(self, cfg: ModelArchConfig, layer_idx: int):
+        super().__init__()
+        self.layer_idx = layer_idx
+        self.n_q = cfg.num_attention_heads
+        self.n_kv = cfg.num_key_value_heads
+        self.hd = cfg.head_dim
+        self.scale = 1.0 / math.sqrt(self.hd)
+        q_size = self.n_q * self.hd
+        kv_size = self.n_kv * self.hd
+        self.qkv_proj = nn.Linear(cfg.hidden_size, q_size + 2 * kv_size, bias=False)
+        self.o_proj = nn.Linear(q_size, cfg.hidden_size, bias=False)
+
+    def forward(
+        self,
+        x: torch.Tensor,  # [T, H]
+        fb: ForwardBatch,
+        kv_cache: tuple[torch.Tensor, torch.Tensor],
+        cos_sin: torch.Tensor,
+    ) -> torch.Tensor:
+        T = x.shape[0]
+        qkv = F.linear(x, self.qkv_proj.weight)
+        q, k, v = qkv.split(
+            [self.n_q * self.hd, self.n_kv * self.hd, self.n_kv * self.hd], dim=-1
+        )
+        q = q.view(T, self.n_q, self.hd).contiguous()
+        k = k.view(T, self.n_kv, self.hd).contiguous()
+        v = v.view(T, self.n_kv, self.hd).contiguous()
+        q, k = ops.rope(q, k, fb.positions, cos_sin)
+        k_cache, v_cache = kv_cache
+        ops.reshape_and_cache(k, v, k_cache, v_cache, fb.slot_mapping)
+
+        out = torch.empty_like(q)
+        nd = fb.n_decode
+        if nd > 0:
+            out[:nd] = ops.paged_attention_decode(
+                q[:nd].contiguous(),
+                k_cache,
+                v_cache,
+                fb.decode_block_tables,
+                fb.decode_seq_lens,
+                self.scale,
+            )
+        if fb.n_prefill > 0:
+            out[nd:] = ops.paged_attention_prefill(
+                q[nd:].contiguous(),
+                k_cache,
+                v_cache,
+                fb.prefill_block_tables,
+                fb.prefill_query_start_loc,
+                fb.prefill_seq_lens,
+                self.scale,
+            )
+        return F.linear(out.view(T, -1), self.o_proj.weight)
+
+
+class MLP(nn.Module):
+    def __init__(self, cfg: ModelArchConfig):
+        super().__init__()
+        self.gate_up_proj = nn.Linear(
+            cfg.hidden_size, 2 * cfg.intermediate_size, bias=False
+        )
+        self.down_proj = nn.Linear(cfg.intermediate_size, cfg.hidden_size, bias=False)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return F.linear(ops.silu_and_mul(F.linear(x, self.gate_up_proj.weight)),
+                        self.down_proj.weight)
+
+
+class MoEMLP(nn.Module):
+    """Mixtral-style sparse MoE block (token-level top-k routing).
+
+    Expert GEMMs are grouped per expert (sort tokens by expert) so each
+    expert runs one hipBLASLt GEMM over its token group.
+    """
+
+    def __init__(self, cfg: ModelArchConfig):
+        super().__init__()
+        self.n_experts = cfg.num_local_experts
+        self.top_k = cfg.num_experts_per_tok
+        self.gate = nn.Linear(cfg.hidden_size, self.n_experts, bias=False)
+        self.experts = nn.ModuleList([MLP(cfg) for _ in range(self.n_experts)])
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        T, H = x.shape
+        router = F.linear(x.float(), self.gate.weight.float())  # [T, E]
+        weights, selected = torch.topk(router, self.top_k, dim=-1)
+        weights = torch.softmax(weights, dim=-1).to(x.dtype)  # [T, k]
+        out = torch.zeros_like(x)
+        for e in range(self.n_experts):
+            mask = selected == e  # [T, k]
+            tok_idx, k_idx = mask.nonzero(as_tuple=True)
+            if tok_idx.numel() == 0:
+                continue
+            contrib = self.experts[e](x[tok_idx])
+            out.index_add_(0, tok_idx, contrib * weights[tok_idx, k_idx, None])
+        return out
+
+
+class DecoderLayer(nn.Module):
+    def __init__(self, cfg: ModelArchConfig, layer_idx: int):
+        super().__init__()
+        self.input_layernorm = nn.Parameter(torch.ones(cfg.hidden_size))
+        self.post_attention_layernorm = nn.Parameter(torch.ones(cfg.hidden_size))
+        self.self_attn = Attention(cfg, layer_idx)
+        self.mlp = MoEMLP(cfg) if cfg.num_local_experts > 0 else MLP(cfg)
+        self.eps = cfg.rms_norm_eps
+
+    def forward(self, x, residual, fb, kv_cache, cos_sin):
+        if residual is None:
+            residual = x
+            x = ops.rmsnorm(x, self.input_layernorm, self.eps)
+        else:
+            x, residual = ops.fused_add_rmsnorm(
+                x, residual, self.input_layernorm, self.eps
+            )
+        x = self.self_attn(x, fb, kv_cache, cos_sin)
+        x, residual = ops.fused_add_rmsnorm(
+            x, residual, self.post_attention_layernorm, self.eps
+        )
+        x = self.mlp(x)
+        return x, residual
+
+
+class LlamaForCausalLM(nn.Module):
+    """Covers dense Llama and (with num_local_experts>0) Mixtral MoE."""
+
+    def __init__(self, cfg: ModelArchConfig, device=None, dtype=torch.bfloat16):
+        super().__init__()
+        self.cfg = cfg
+        factory = {"device": device, "dtype": dtype}
+        with torch.device(device if device is not None else "cpu"):
+            self.embed_tokens = nn.Embedding(cfg.vocab_size, cfg.hidden_size)
+            self.layers = nn.ModuleList(
+                [DecoderLayer(cfg, i) for i in range(cfg.num_hidden_layers)]
+            )
+            self.norm = nn.Parameter(torch.ones(cfg.hidden_size))
+            if cfg.tie_word_embeddings:
+                self.lm_head = None
+            else:
+                self.lm_head = nn.Linear(cfg.hidden_size, cfg.vocab_size, bias=False)
+        self.to(**factory)
+        from kubeai_amd.ops import ref as ops_ref
+
+        self.register_buffer(
+            "cos_sin",
+            ops_ref.make_cos_sin_cache(
+                cfg.head_dim, cfg.max_position_embeddings, cfg.rope_theta
+            ).to(device=device),
+            persistent=False,
+        )
+
+    @torch.inference_mode()
+    def forward(self, fb: ForwardBatch) -> torch.Tensor:
+        x = self.embed_tokens(fb.input_ids.long())
+        residual = None
+        for i, layer in enumerate(self.layers):
+            x, residual = layer(x, residual, fb, self.kv_caches[i], self.cos_sin)
+        x, _ = ops.fused_add_rmsnorm(x, residual, self.norm, self.cfg.rms_norm_eps)
+        return x
+
+    @torch.inference_mode()
+    def compute_logits(self, hidden: torch.Tensor) -> torch.Tensor:
+        w = (
+            self.embed_tokens.weight
+            if self.lm_head is None
+            else self.lm_head.weight
+        )
+        return F.linear(hidden, w).float()
+
+    def bind_kv_caches(self, kv_caches: list[tuple[torch.Tensor, torch.Tensor]]):
+        self.kv_caches = kv_caches

@@ -1,0 +1,253 @@
+"""Flagship serving benchmark — the BASELINE.json metric.
+
+Measures the k6 multi-turn-chat headline (output tokens/sec + p50 TTFT,
+Llama-3-8B) on N single-GPU replicas (weak scaling: each rank = one replica
+with its own synthetic virtual users, mirroring the reference's
+benchmarks/multi-turn-chat-k6/k6.js workload shape: per-VU conversations
+whose history grows with each turn, so the engine's prefix cache — the
+in-house analog of what CHWBL routing exploits — is exercised).
+
+Launch (driver contract):
+  python bench.py --gpus 1 --steps K --warmup W
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+      --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+
+One "step" = one engine scheduler step (one packed forward: decode batch +
+prefill chunks). Warmup steps fill the continuous batch; then exactly K
+steps are timed between barriers + torch.cuda.synchronize on both sides.
+rank0 prints ONE JSON line; value = WHOLE-JOB output tokens/s.
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import random
+import statistics
+import time
+
+import torch
+
+
+def get_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=60)
+    p.add_argument("--warmup", type=int, default=20)
+    p.add_argument("--model", default="llama-3-8b")
+    p.add_argument("--vus", type=int, default=32, help="virtual users per replica")
+    p.add_argument("--system-tokens", type=int, default=64)
+    p.add_argument("--user-tokens", type=int, default=48)
+    p.add_argument("--max-tokens", type=int, default=32, help="per-turn completion")
+    p.add_argument("--max-turns", type=int, default=8)
+    p.add_argument("--device", default="auto")
+    p.add_argument("--no-prefix-cache", action="store_true")
+    p.add_argument("--gpu-mem-util", type=float, default=0.90)
+    p.add_argument("--max-model-len", type=int, default=8192)
+    return p.parse_args()
+
+
+class VirtualUser:
+    """One k6-style chat thread: history grows with every turn."""
+
+    def __init__(self, uid: int, args, system_prompt: list[int], vocab: int):
+        self.uid = uid
+        self.args = args
+        self.rng = random.Random(10_000 + uid)
+        self.history: list[int] = list(system_prompt)
+        self.turn = 0
+        self.vocab = vocab
+        self.inflight: str | None = None
+
+    def next_request(self):
+        self.turn += 1
+        user_msg = [
+            self.rng.randrange(1000, self.vocab - 10)
+            for _ in range(self.args.user_tokens)
+        ]
+        self.history.extend(user_msg)
+        rid = f"vu{self.uid}-t{self.turn}"
+        self.inflight = rid
+        return rid, list(self.history)
+
+    def complete(self, output_tokens: list[int]):
+        self.history.extend(output_tokens)
+        self.inflight = None
+        if (
+            self.turn >= self.args.max_turns
+            or len(self.history) > self.args.max_model_len - 256
+        ):
+            self.history = self.history[: self.args.system_tokens]
+            self.turn = 0
+
+
+def main():
+    args = get_args()
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+
+    use_cuda = torch.cuda.is_available() if args.device == "auto" else args.device == "cuda"
+    device = f"cuda:{local_rank}" if use_cuda else "cpu"
+    if use_cuda:
+        torch.cuda.set_device(local_rank)
+
+    dist = None
+    if world > 1:
+        import torch.distributed as dist_mod
+
+        dist = dist_mod
+        dist.init_process_group(
+            backend="nccl" if use_cuda else "gloo", rank=rank, world_size=world
+        )
+
+    from kubeai_amd.engine import EngineConfig, LLMEngine, SamplingParams
+    from kubeai_amd.models.config import PRESETS
+
+    model = args.model
+    if not use_cuda and model == "llama-3-8b":
+        model = "llama-tiny"  # CPU sanity runs only
+    arch = PRESETS[model]
+
+    eng = LLMEngine(
+        EngineConfig(
+            model=model,
+            device=device,
+            num_gpu_blocks=None if use_cuda else 2048,
+            gpu_memory_utilization=args.gpu_mem_util,
+            max_model_len=args.max_model_len,
+            max_num_batched_tokens=8192,
+            max_num_seqs=max(args.vus * 2, 64),
+            enable_prefix_caching=not args.no_prefix_cache,
+            seed=1234 + rank,
+        )
+    )
+
+    sp = SamplingParams(max_tokens=args.max_tokens, temperature=0.0, ignore_eos=True)
+    sys_rng = random.Random(7)
+    system_prompt = [
+        sys_rng.randrange(1000, arch.vocab_size - 10)
+        for _ in range(args.system_tokens)
+    ]
+    vus = [VirtualUser(u, args, system_prompt, arch.vocab_size) for u in range(args.vus)]
+    arrival: dict[str, float] = {}
+    first_token: dict[str, float] = {}
+    rid_to_vu = {}
+
+    # drip-release VUs over the warmup so the batch ramps like a k6 ramp-up
+    # and request arrivals (TTFT samples) continue through the timed region
+    released = 0
+    release_per_step = max(1, args.vus // max(args.warmup, 8))
+
+    def pump():
+        nonlocal released
+        released = min(args.vus, released + release_per_step)
+        now = time.monotonic()
+        for vu in vus[:released]:
+            if vu.inflight is None:
+                rid, toks = vu.next_request()
+                eng.add_request(toks, sp, request_id=rid)
+                arrival[rid] = now
+                rid_to_vu[rid] = vu
+
+    def one_step():
+        pump()
+        now_outputs = eng.step()
+        now = time.monotonic()
+        toks = 0
+        for o in now_outputs:
+            toks += len(o.new_token_ids)
+            if o.request_id not in first_token:
+                first_token[o.request_id] = now
+            if o.finished:
+                rid_to_vu[o.request_id].complete(o.output_token_ids)
+        return toks
+
+    # ---- warmup ----
+    for _ in range(args.warmup):
+        one_step()
+
+    # ---- timed region ----
+    timed_start_reqs = set(arrival)  # exclude pre-warmup arrivals from TTFT
+    if dist:
+        dist.barrier()
+    if use_cuda:
+        torch.cuda.synchronize()
+    t0 = time.monotonic()
+    tokens = 0
+    for _ in range(args.steps):
+        tokens += one_step()
+    if use_cuda:
+        torch.cuda.synchronize()
+    if dist:
+        dist.barrier()
+    t1 = time.monotonic()
+    elapsed = t1 - t0
+
+    # TTFT for requests submitted during the timed region
+    ttfts = [
+        (first_token[r] - arrival[r]) * 1000.0
+        for r in first_token
+        if r not in timed_start_reqs and arrival.get(r, 0) >= t0
+    ]
+    p50_ttft = statistics.median(ttfts) if ttfts else None
+
+    # aggregate across ranks: elapsed = MAX, tokens = SUM
+    if dist:
+        te = torch.tensor([elapsed], dtype=torch.float64, device=device if use_cuda else "cpu")
+        tk = torch.tensor([float(tokens)], dtype=torch.float64, device=device if use_cuda else "cpu")
+        dist.all_reduce(te, op=dist.ReduceOp.MAX)
+        dist.all_reduce(tk, op=dist.ReduceOp.SUM)
+        elapsed_max = te.item()
+        tokens_sum = tk.item()
+        tt = torch.tensor(
+            [p50_ttft if p50_ttft is not None else 0.0, 1.0 if p50_ttft else 0.0],
+            dtype=torch.float64,
+            device=device if use_cuda else "cpu",
+        )
+        dist.all_reduce(tt)
+        p50_all = tt[0].item() / max(tt[1].item(), 1.0)
+    else:
+        elapsed_max = elapsed
+        tokens_sum = float(tokens)
+        p50_all = p50_ttft
+
+    value = tokens_sum / elapsed_max
+    baseline = 178.71  # BASELINE.md k6 PrefixHash new_tokens tok/s (2xL4)
+
+    if rank == 0:
+        stats = eng.stats()
+        print(
+            json.dumps(
+                {
+                    "metric": "output_tokens_per_s (k6 multi-turn chat)",
+                    "value": round(value, 2),
+                    "unit": "tok/s",
+                    "n_gpus": world if use_cuda else world,
+                    "steps": args.steps,
+                    "warmup": args.warmup,
+                    "ms_per_step": round(elapsed_max * 1000.0 / args.steps, 3),
+                    "p50_ttft_ms": round(p50_all, 2) if p50_all else None,
+                    "higher_is_better": True,
+                    "scaling": "weak",
+                    "vs_baseline": round(value / baseline, 3),
+                    "dtype": "bf16",
+                    "data": "synthetic",
+                    "config": {
+                        "model": model,
+                        "global_batch": args.vus * world,
+                        "seq_len": args.max_model_len,
+                        "parallelism": f"dp{world}",
+                        "max_tokens_per_turn": args.max_tokens,
+                        "prefix_cache": not args.no_prefix_cache,
+                        "prefix_cache_hit_rate": round(stats["prefix_cache_hit_rate"], 4),
+                    },
+                }
+            )
+        )
+    if dist:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

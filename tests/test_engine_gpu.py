@@ -1,0 +1,100 @@
+"""Engine integration on the GPU (HIP kernel path end-to-end).
+
+Validates cross-kernel consistency (prefill vs decode vs prefix-cached
+paths must yield identical greedy generations) and CPU-reference logits
+agreement with shared weights.
+"""
+import pytest
+import torch
+
+from kubeai_amd.engine import EngineConfig, LLMEngine, SamplingParams
+
+pytestmark = pytest.mark.gpu
+
+
+def make_engine(device, **kw):
+    base = dict(
+        model="llama-tiny",
+        device=device,
+        max_model_len=512,
+        num_gpu_blocks=256,
+        seed=0,
+    )
+    base.update(kw)
+    return LLMEngine(EngineConfig(**base))
+
+
+def drain(eng, max_steps=300):
+    outs = {}
+    for _ in range(max_steps):
+        if not eng.has_work():
+            break
+        for o in eng.step():
+            if o.finished:
+                outs[o.request_id] = o
+    assert not eng.has_work()
+    return outs
+
+
+def test_greedy_deterministic_gpu():
+    prompt = list(range(10, 140))
+    a = []
+    for _ in range(2):
+        eng = make_engine("cuda")
+        eng.add_request(prompt, SamplingParams(max_tokens=12), request_id="r")
+        a.append(drain(eng)["r"].output_token_ids)
+    assert a[0] == a[1]
+
+
+def test_prefix_cache_equivalence_gpu():
+    prompt = list(range(5, 120))
+    eng = make_engine("cuda")
+    eng.add_request(prompt, SamplingParams(max_tokens=8), request_id="cold")
+    cold = drain(eng)["cold"]
+    eng.add_request(prompt, SamplingParams(max_tokens=8), request_id="warm")
+    warm = drain(eng)["warm"]
+    assert warm.num_cached_tokens >= 96
+    assert warm.output_token_ids == cold.output_token_ids
+
+
+def test_chunked_prefill_equivalence_gpu():
+    prompt = list(range(5, 260))
+    eng1 = make_engine("cuda", max_num_batched_tokens=4096)
+    eng1.add_request(prompt, SamplingParams(max_tokens=6), request_id="a")
+    full = drain(eng1)["a"].output_token_ids
+    eng2 = make_engine("cuda", max_num_batched_tokens=96)
+    eng2.add_request(prompt, SamplingParams(max_tokens=6), request_id="a")
+    chunked = drain(eng2)["a"].output_token_ids
+    assert full == chunked
+
+
+def test_gpu_logits_match_cpu_reference():
+    # identical weights on both devices; compare last-token logits of a
+    # prefill — full-stack numerics check of the HIP path vs torch ref
+    gpu = make_engine("cuda")
+    cpu = make_engine("cpu")
+    cpu.runner.model.load_state_dict(
+        {k: v.cpu() for k, v in gpu.runner.model.state_dict().items()}
+    )
+    prompt = list(range(20, 150))
+    for eng, rid in ((gpu, "g"), (cpu, "c")):
+        eng.add_request(prompt, SamplingParams(max_tokens=4), request_id=rid)
+    og = drain(gpu)["g"].output_token_ids
+    oc = drain(cpu)["c"].output_token_ids
+    # bf16 GEMM backends differ (hipBLASLt vs CPU); require the greedy
+    # argmax to agree on at least the first generated token
+    assert og[0] == oc[0], f"gpu={og} cpu={oc}"
+
+
+def test_batch_throughput_many_seqs_gpu():
+    eng = make_engine("cuda", max_num_seqs=64)
+    for i in range(32):
+        eng.add_request(
+            list(range(i + 3, i + 90)),
+            SamplingParams(max_tokens=8),
+            request_id=f"r{i}",
+        )
+    outs = drain(eng, max_steps=600)
+    assert len(outs) == 32
+    for o in outs.values():
+        assert len(o.output_token_ids) == 8

@@ -101,6 +101,12 @@ def main():
             backend="nccl" if use_cuda else "gloo", rank=rank, world_size=world
         )
 
+    import sys
+
+    def log(msg):
+        print(f"[bench r{rank} t={time.monotonic():.1f}] {msg}", file=sys.stderr, flush=True)
+
+    log("importing engine")
     from kubeai_amd.engine import EngineConfig, LLMEngine, SamplingParams
     from kubeai_amd.models.config import PRESETS
 
@@ -123,6 +129,7 @@ def main():
         )
     )
 
+    log(f"engine ready: {eng.runner.num_blocks} kv blocks")
     sp = SamplingParams(max_tokens=args.max_tokens, temperature=0.0, ignore_eos=True)
     sys_rng = random.Random(7)
     system_prompt = [
@@ -164,8 +171,11 @@ def main():
         return toks
 
     # ---- warmup ----
-    for _ in range(args.warmup):
+    for i in range(args.warmup):
         one_step()
+        if i == 0:
+            log("first step done")
+    log("warmup done")
 
     # ---- timed region ----
     timed_start_reqs = set(arrival)  # exclude pre-warmup arrivals from TTFT

@@ -111,6 +111,18 @@ __launch_bounds__(G * WAVE_SIZE) __global__ void paged_prefill_kernel(
       for (int i = threadIdx.x; i < nvec; i += G * WAVE_SIZE) {
         const int tok = i / (kHD / 8);
         const int col8 = i % (kHD / 8);
+        if (tok >= kv_valid) {
+          // zero unstaged V: P rows are 0 there, but 0 * stale-NaN would
+          // poison the PV MFMA accumulator (K can stay stale: scores for
+          // invalid tokens are forced to -inf before softmax)
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            const int hdrow = col8 * 8 + j;
+            *reinterpret_cast<ushort*>(
+                reinterpret_cast<char*>(v_lds) +
+                swz(hdrow, (hdrow * kKVB + tok) * 2)) = 0;
+          }
+        }
         if (tok < kv_valid) {
           const int abs_tok = kv_start + tok;
           const int64_t blk = bt[abs_tok / kBS];

@@ -34,6 +34,9 @@ class ForwardBatch:
     # indices (into the packed stream) of tokens whose logits are sampled
     logits_indices: torch.Tensor   # int64 [n_sample]
 
+    # per-token LoRA adapter ids (None when no adapters are active)
+    lora_ids: torch.Tensor | None = None  # int32 [T]
+
     @property
     def n_tokens(self) -> int:
         return self.input_ids.shape[0]

@@ -139,6 +139,37 @@ class LLMEngine:
         return results
 
     # ------------------------------------------------------------------
+    def embed(self, tok_lists: list[list[int]]) -> list[list[float]]:
+        """Mean-pooled, L2-normalised final hidden states (TextEmbedding
+        feature; reference analog: the Infinity engine)."""
+        from .scheduler import ScheduledSeq, SchedulerOutput
+
+        out: list[list[float]] = []
+        for toks in tok_lists:
+            toks = toks[: self.cfg.max_model_len - 1]
+            table, _ = self.block_manager.allocate(toks, salt=-1, max_cached=0)
+            req = Request(list(toks), SamplingParams(max_tokens=1))
+            req.block_table = table
+            so = SchedulerOutput(
+                decode=[], prefill=[ScheduledSeq(req, 0, len(toks))], preempted=[]
+            )
+            fb = self.runner.build_batch(so)
+            hidden = self.runner.model(fb)
+            vec = hidden.float().mean(dim=0)
+            vec = vec / vec.norm().clamp_min(1e-12)
+            out.append(vec.cpu().tolist())
+            self.block_manager.free(table)
+        return out
+
+    # ------------------------------------------------------------------
+    # LoRA adapter lifecycle (weights managed by the runner's LoRA manager)
+    def load_lora(self, lora_id: int, path: Optional[str]) -> None:
+        self.runner.load_lora(lora_id, path)
+
+    def unload_lora(self, lora_id: int) -> None:
+        self.runner.unload_lora(lora_id)
+
+    # ------------------------------------------------------------------
     # metrics for /metrics + the autoscaler (SURVEY.md §5.5)
     def stats(self) -> dict:
         return {

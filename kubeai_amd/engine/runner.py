@@ -88,6 +88,20 @@ class ModelRunner:
         return 512  # CPU tests
 
     # ------------------------------------------------------------------
+    # LoRA (full SGMV path in kubeai_amd/engine/lora.py; ids are also the
+    # KV-cache salt so adapter outputs never share prefix blocks)
+    def load_lora(self, lora_id: int, path) -> None:
+        from .lora import LoRAManager
+
+        if not hasattr(self, "lora_manager"):
+            self.lora_manager = LoRAManager(self.model, self.device, self.dtype)
+        self.lora_manager.load(lora_id, path)
+
+    def unload_lora(self, lora_id: int) -> None:
+        if hasattr(self, "lora_manager"):
+            self.lora_manager.unload(lora_id)
+
+    # ------------------------------------------------------------------
     def build_batch(self, out: SchedulerOutput) -> ForwardBatch:
         bs = self.block_size
         input_ids: list[int] = []
@@ -131,9 +145,17 @@ class ModelRunner:
             if ss.samples:
                 logit_idx.append(nd + qsl[j + 1] - 1)
 
+        lora_ids = None
+        if getattr(self, "lora_manager", None) is not None and self.lora_manager.active:
+            ids: list[int] = []
+            for ss in out.all_seqs:
+                ids.extend([ss.req.lora_id] * ss.chunk_len)
+            lora_ids = torch.tensor(ids, dtype=torch.int32, device=self.device)
+
         dev = self.device
         t32 = lambda x: torch.tensor(x, dtype=torch.int32, device=dev)
         return ForwardBatch(
+            lora_ids=lora_ids,
             input_ids=t32(input_ids),
             positions=t32(positions),
             slot_mapping=torch.tensor(slots, dtype=torch.int64, device=dev),

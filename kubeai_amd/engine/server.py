@@ -1,0 +1,425 @@
+"""OpenAI-compatible HTTP server wrapping LLMEngine — the per-Model engine
+container contract the control plane depends on:
+
+  POST /v1/completions, /v1/chat/completions (stream + non-stream)
+  POST /v1/embeddings
+  GET  /health                      (startup/readiness/liveness probes)
+  GET  /metrics                     (Prometheus; queue depth + KV occupancy
+                                     gauges consumed by the autoscaler)
+  GET  /v1/models
+  POST /v1/load_lora_adapter        ("already loaded" error semantics)
+  POST /v1/unload_lora_adapter      ("cannot be found" error semantics)
+
+Run:  python -m kubeai_amd.engine.server --model llama-tiny --port 8000
+"""
+from __future__ import annotations
+
+import argparse
+import asyncio
+import json
+import os
+import queue
+import threading
+import time
+import uuid
+from typing import Optional
+
+from fastapi import FastAPI, Request
+from fastapi.responses import JSONResponse, PlainTextResponse, StreamingResponse
+import prometheus_client as prom
+
+from .engine import EngineConfig, LLMEngine
+from .scheduler import SamplingParams
+from .tokenizer import apply_chat_template, load_tokenizer
+
+REGISTRY = prom.CollectorRegistry()
+M_WAITING = prom.Gauge("kubeai_engine_num_requests_waiting", "queue depth", ["model"], registry=REGISTRY)
+M_RUNNING = prom.Gauge("kubeai_engine_num_requests_running", "running requests", ["model"], registry=REGISTRY)
+M_KV = prom.Gauge("kubeai_engine_kv_cache_usage_perc", "KV cache occupancy", ["model"], registry=REGISTRY)
+M_HIT = prom.Gauge("kubeai_engine_prefix_cache_hit_rate", "prefix cache hit rate", ["model"], registry=REGISTRY)
+M_GEN = prom.Counter("kubeai_engine_generation_tokens_total", "generated tokens", ["model"], registry=REGISTRY)
+M_PROMPT = prom.Counter("kubeai_engine_prompt_tokens_total", "prompt tokens", ["model"], registry=REGISTRY)
+M_TTFT = prom.Histogram(
+    "kubeai_engine_time_to_first_token_seconds", "TTFT", ["model"], registry=REGISTRY,
+    buckets=(0.01, 0.025, 0.05, 0.1, 0.25, 0.5, 1.0, 2.5, 5.0, 10.0, 30.0),
+)
+
+
+class EngineServer:
+    """Engine + stepping thread + async request plumbing."""
+
+    def __init__(self, cfg: EngineConfig, served_model_name: str):
+        self.cfg = cfg
+        self.served_model_name = served_model_name
+        self.engine: Optional[LLMEngine] = None
+        self.tokenizer = None
+        self._submit: "queue.Queue" = queue.Queue()
+        self._events: dict[str, tuple[asyncio.AbstractEventLoop, asyncio.Queue]] = {}
+        self._aborts: "queue.Queue" = queue.Queue()
+        self._ready = threading.Event()
+        self._stop = threading.Event()
+        self._lora_adapters: dict[str, str] = {}
+        self._lock = threading.Lock()
+        self._thread = threading.Thread(target=self._run, daemon=True)
+
+    # ------------------------------------------------------------- lifecycle
+    def start(self) -> None:
+        self._thread.start()
+
+    def stop(self) -> None:
+        self._stop.set()
+
+    def _run(self) -> None:
+        # heavy init inside the thread so /health can answer "starting"
+        self.engine = LLMEngine(self.cfg)
+        arch = self.engine.arch
+        self.tokenizer = load_tokenizer(
+            self.cfg.model, arch.vocab_size, arch.bos_token_id, arch.eos_token_id
+        )
+        self._ready.set()
+        label = self.served_model_name
+        while not self._stop.is_set():
+            worked = False
+            while True:
+                try:
+                    rid, toks, params, lora_id = self._submit.get_nowait()
+                except queue.Empty:
+                    break
+                self.engine.add_request(toks, params, request_id=rid, lora_id=lora_id)
+                M_PROMPT.labels(label).inc(len(toks))
+                worked = True
+            while True:
+                try:
+                    rid = self._aborts.get_nowait()
+                except queue.Empty:
+                    break
+                self.engine.abort_request(rid)
+            if self.engine.has_work():
+                outputs = self.engine.step()
+                worked = True
+                for o in outputs:
+                    M_GEN.labels(label).inc(len(o.new_token_ids))
+                    ent = self._events.get(o.request_id)
+                    if ent is not None:
+                        loop, q = ent
+                        loop.call_soon_threadsafe(q.put_nowait, o)
+            s = self.engine.stats()
+            M_WAITING.labels(label).set(s["num_waiting"])
+            M_RUNNING.labels(label).set(s["num_running"])
+            M_KV.labels(label).set(s["kv_usage"])
+            M_HIT.labels(label).set(s["prefix_cache_hit_rate"])
+            if not worked:
+                time.sleep(0.002)
+
+    # ------------------------------------------------------------- requests
+    async def generate(self, token_ids: list[int], params: SamplingParams, lora_id: int = 0):
+        """Async iterator of RequestOutput for one request."""
+        rid = f"cmpl-{uuid.uuid4().hex[:16]}"
+        loop = asyncio.get_running_loop()
+        q: asyncio.Queue = asyncio.Queue()
+        self._events[rid] = (loop, q)
+        t0 = time.monotonic()
+        first = True
+        self._submit.put((rid, token_ids, params, lora_id))
+        try:
+            while True:
+                o = await q.get()
+                if first:
+                    M_TTFT.labels(self.served_model_name).observe(time.monotonic() - t0)
+                    first = False
+                yield o
+                if o.finished:
+                    return
+        finally:
+            self._events.pop(rid, None)
+            self._aborts.put(rid)
+
+    # ------------------------------------------------------------- LoRA
+    # Registry + per-adapter KV-cache salting; adapter weights are applied
+    # by the runner's LoRA manager (kubeai_amd/engine/lora.py).
+    def load_lora(self, name: str, path: Optional[str]) -> None:
+        if path and not (os.path.isdir(path) or os.path.isfile(path)):
+            raise ValueError(f"lora path {path} does not exist")
+        lid = self.lora_id_of(name)
+        if self.engine is not None:
+            self.engine.load_lora(lid, path)
+
+    def unload_lora(self, name: str) -> None:
+        if self.engine is not None:
+            self.engine.unload_lora(self.lora_id_of(name))
+
+    def lora_id_of(self, name: str) -> int:
+        import zlib
+
+        return 1 + (zlib.crc32(name.encode()) % 1_000_000)
+
+
+def build_app(server: EngineServer) -> FastAPI:
+    app = FastAPI()
+    name = server.served_model_name
+
+    def _params_from(body: dict) -> SamplingParams:
+        mt = body.get("max_tokens") or body.get("max_completion_tokens") or 128
+        temp = body.get("temperature")
+        temp = 1.0 if temp is None else float(temp)
+        seed = body.get("seed")
+        return SamplingParams(
+            max_tokens=int(mt),
+            temperature=float(temp),
+            top_p=float(body.get("top_p") or 1.0),
+            top_k=int(body.get("top_k") or 0),
+            seed=int(seed) if seed is not None else None,
+            ignore_eos=bool(body.get("ignore_eos") or False),
+        )
+
+    def _usage(prompt_toks: int, completion_toks: int) -> dict:
+        return {
+            "prompt_tokens": prompt_toks,
+            "completion_tokens": completion_toks,
+            "total_tokens": prompt_toks + completion_toks,
+        }
+
+    @app.get("/health")
+    async def health():
+        if not server._ready.is_set():
+            return JSONResponse({"status": "starting"}, status_code=503)
+        return {"status": "ok"}
+
+    @app.get("/metrics")
+    async def metrics():
+        return PlainTextResponse(
+            prom.generate_latest(REGISTRY).decode(), media_type=prom.CONTENT_TYPE_LATEST
+        )
+
+    @app.get("/v1/models")
+    async def models():
+        data = [{"id": name, "object": "model", "owned_by": "kubeai-amd"}]
+        for a in server._lora_adapters:
+            data.append({"id": a, "object": "model", "owned_by": "kubeai-amd", "parent": name})
+        return {"object": "list", "data": data}
+
+    async def _finish_tokens(gen):
+        final = None
+        async for o in gen:
+            final = o
+        return final
+
+    @app.post("/v1/completions")
+    async def completions(request: Request):
+        body = await request.json()
+        params = _params_from(body)
+        prompt = body.get("prompt", "")
+        if isinstance(prompt, list):
+            prompt = prompt[0] if prompt else ""
+        if isinstance(prompt, str):
+            toks = server.tokenizer.encode(prompt, add_bos=True)
+        else:
+            toks = list(prompt)
+        lora_id = _resolve_lora(server, body.get("model"))
+        if isinstance(lora_id, JSONResponse):
+            return lora_id
+        if body.get("stream"):
+            return StreamingResponse(
+                _stream_completion(server, toks, params, name, chat=False, lora_id=lora_id),
+                media_type="text/event-stream",
+            )
+        final = await _finish_tokens(server.generate(toks, params, lora_id))
+        text = server.tokenizer.decode(_strip_stop(final, params))
+        return {
+            "id": f"cmpl-{uuid.uuid4().hex[:12]}",
+            "object": "text_completion",
+            "created": int(time.time()),
+            "model": name,
+            "choices": [
+                {
+                    "index": 0,
+                    "text": text,
+                    "finish_reason": final.finish_reason or "stop",
+                    "logprobs": None,
+                }
+            ],
+            "usage": _usage(final.num_prompt_tokens, len(final.output_token_ids)),
+        }
+
+    @app.post("/v1/chat/completions")
+    async def chat_completions(request: Request):
+        body = await request.json()
+        params = _params_from(body)
+        toks = apply_chat_template(server.tokenizer, body.get("messages", []))
+        lora_id = _resolve_lora(server, body.get("model"))
+        if isinstance(lora_id, JSONResponse):
+            return lora_id
+        if body.get("stream"):
+            return StreamingResponse(
+                _stream_completion(server, toks, params, name, chat=True, lora_id=lora_id),
+                media_type="text/event-stream",
+            )
+        final = await _finish_tokens(server.generate(toks, params, lora_id))
+        text = server.tokenizer.decode(_strip_stop(final, params))
+        return {
+            "id": f"chatcmpl-{uuid.uuid4().hex[:12]}",
+            "object": "chat.completion",
+            "created": int(time.time()),
+            "model": name,
+            "choices": [
+                {
+                    "index": 0,
+                    "message": {"role": "assistant", "content": text},
+                    "finish_reason": final.finish_reason or "stop",
+                }
+            ],
+            "usage": _usage(final.num_prompt_tokens, len(final.output_token_ids)),
+        }
+
+    @app.post("/v1/embeddings")
+    async def embeddings(request: Request):
+        body = await request.json()
+        inputs = body.get("input", [])
+        if isinstance(inputs, str):
+            inputs = [inputs]
+        tok_lists = [server.tokenizer.encode(t, add_bos=True) for t in inputs]
+        vecs = await asyncio.get_running_loop().run_in_executor(
+            None, server.engine.embed, tok_lists
+        )
+        data = [
+            {"object": "embedding", "index": i, "embedding": v}
+            for i, v in enumerate(vecs)
+        ]
+        return {
+            "object": "list",
+            "data": data,
+            "model": name,
+            "usage": _usage(sum(len(t) for t in tok_lists), 0),
+        }
+
+    @app.post("/v1/load_lora_adapter")
+    async def load_lora(request: Request):
+        body = await request.json()
+        lname, path = body.get("lora_name"), body.get("lora_path")
+        with server._lock:
+            if lname in server._lora_adapters:
+                return JSONResponse(
+                    {"error": f"adapter {lname} was already loaded"}, status_code=400
+                )
+            try:
+                server.load_lora(lname, path)
+            except Exception as e:  # noqa: BLE001
+                return JSONResponse({"error": str(e)}, status_code=400)
+            server._lora_adapters[lname] = path
+        return PlainTextResponse("OK")
+
+    @app.post("/v1/unload_lora_adapter")
+    async def unload_lora(request: Request):
+        body = await request.json()
+        lname = body.get("lora_name")
+        with server._lock:
+            if lname not in server._lora_adapters:
+                return JSONResponse(
+                    {"error": f"adapter {lname} cannot be found"}, status_code=404
+                )
+            server._lora_adapters.pop(lname)
+            server.unload_lora(lname)
+        return PlainTextResponse("OK")
+
+    return app
+
+
+def _strip_stop(final, params: SamplingParams) -> list[int]:
+    toks = final.output_token_ids
+    if final.finish_reason == "stop" and toks and toks[-1] in params.stop_token_ids:
+        return toks[:-1]
+    return toks
+
+
+def _resolve_lora(server: EngineServer, model_field: Optional[str]):
+    """Model field == adapter name selects the adapter (vLLM semantics)."""
+    if not model_field or model_field == server.served_model_name:
+        return 0
+    with server._lock:
+        if model_field in server._lora_adapters:
+            return server.lora_id_of(model_field)
+    return JSONResponse(
+        {"error": {"message": f"model {model_field} cannot be found"}}, status_code=404
+    )
+
+
+async def _stream_completion(server, toks, params, name, chat: bool, lora_id: int = 0):
+    rid = f"{'chatcmpl' if chat else 'cmpl'}-{uuid.uuid4().hex[:12]}"
+    created = int(time.time())
+    n_out = 0
+    async for o in server.generate(toks, params, lora_id):
+        piece = server.tokenizer.decode(o.new_token_ids)
+        n_out = len(o.output_token_ids)
+        if chat:
+            chunk = {
+                "id": rid,
+                "object": "chat.completion.chunk",
+                "created": created,
+                "model": name,
+                "choices": [
+                    {
+                        "index": 0,
+                        "delta": {"content": piece} if not o.finished else {},
+                        "finish_reason": o.finish_reason if o.finished else None,
+                    }
+                ],
+            }
+        else:
+            chunk = {
+                "id": rid,
+                "object": "text_completion",
+                "created": created,
+                "model": name,
+                "choices": [
+                    {
+                        "index": 0,
+                        "text": piece,
+                        "finish_reason": o.finish_reason if o.finished else None,
+                    }
+                ],
+            }
+        if o.finished:
+            chunk["usage"] = {
+                "prompt_tokens": o.num_prompt_tokens,
+                "completion_tokens": n_out,
+                "total_tokens": o.num_prompt_tokens + n_out,
+            }
+        yield f"data: {json.dumps(chunk)}\n\n"
+    yield "data: [DONE]\n\n"
+
+
+# --------------------------------------------------------------------------
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--model", required=True, help="preset name or model dir")
+    p.add_argument("--served-model-name", default=None)
+    p.add_argument("--host", default="0.0.0.0")
+    p.add_argument("--port", type=int, default=8000)
+    p.add_argument("--device", default="auto")
+    p.add_argument("--max-model-len", type=int, default=8192)
+    p.add_argument("--max-num-seqs", type=int, default=256)
+    p.add_argument("--gpu-memory-utilization", type=float, default=0.90)
+    p.add_argument("--num-gpu-blocks", type=int, default=None)
+    p.add_argument("--enable-lora", action="store_true")
+    p.add_argument("--tensor-parallel-size", type=int, default=1)
+    args = p.parse_args()
+
+    cfg = EngineConfig(
+        model=args.model,
+        device=args.device,
+        max_model_len=args.max_model_len,
+        max_num_seqs=args.max_num_seqs,
+        gpu_memory_utilization=args.gpu_memory_utilization,
+        num_gpu_blocks=args.num_gpu_blocks,
+    )
+    served = args.served_model_name or os.path.basename(args.model.rstrip("/"))
+    server = EngineServer(cfg, served)
+    server.start()
+    app = build_app(server)
+    import uvicorn
+
+    uvicorn.run(app, host=args.host, port=args.port, log_level="warning")
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,79 @@
+"""Tokenizers for the engine.
+
+- HF tokenizer (tokenizer.json in the model dir) when available.
+- SyntheticTokenizer otherwise: deterministic word-hash tokens. Identical
+  text prefixes map to identical token prefixes, so prefix caching and
+  CHWBL prefix routing behave exactly as with a real tokenizer — this is
+  what bench/k6-style runs use (no network for real tokenizer files).
+"""
+from __future__ import annotations
+
+import os
+import zlib
+
+
+class SyntheticTokenizer:
+    def __init__(self, vocab_size: int, bos_token_id: int = 1, eos_token_id: int = 2):
+        self.vocab_size = vocab_size
+        self.bos_token_id = bos_token_id
+        self.eos_token_id = eos_token_id
+        # reserve low ids for specials
+        self._lo = 16
+
+    def encode(self, text: str, add_bos: bool = False) -> list[int]:
+        toks: list[int] = [self.bos_token_id] if add_bos else []
+        span = self.vocab_size - self._lo
+        for word in text.split():
+            toks.append(self._lo + zlib.crc32(word.encode()) % span)
+        return toks
+
+    def decode(self, token_ids: list[int]) -> str:
+        return " ".join(f"t{t}" for t in token_ids)
+
+
+class HFTokenizer:
+    def __init__(self, path: str):
+        from transformers import AutoTokenizer
+
+        self._tok = AutoTokenizer.from_pretrained(path)
+        self.vocab_size = self._tok.vocab_size
+        self.bos_token_id = self._tok.bos_token_id
+        self.eos_token_id = self._tok.eos_token_id
+
+    def encode(self, text: str, add_bos: bool = False) -> list[int]:
+        return self._tok.encode(text, add_special_tokens=add_bos)
+
+    def decode(self, token_ids: list[int]) -> str:
+        return self._tok.decode(token_ids, skip_special_tokens=True)
+
+
+def load_tokenizer(model: str, vocab_size: int, bos: int, eos: int):
+    if os.path.isdir(model) and (
+        os.path.exists(os.path.join(model, "tokenizer.json"))
+        or os.path.exists(os.path.join(model, "tokenizer.model"))
+    ):
+        try:
+            return HFTokenizer(model)
+        except Exception:
+            pass
+    return SyntheticTokenizer(vocab_size, bos, eos)
+
+
+def apply_chat_template(tokenizer, messages: list[dict]) -> list[int]:
+    """Deterministic minimal chat template (role tag + content per message).
+
+    Reference analog: the engine containers own their chat template
+    (engine contract); format is internal.
+    """
+    role_ids = {"system": 3, "user": 4, "assistant": 5, "tool": 6}
+    toks: list[int] = [tokenizer.bos_token_id]
+    for m in messages:
+        toks.append(role_ids.get(m.get("role", "user"), 4))
+        content = m.get("content") or ""
+        if isinstance(content, list):  # OpenAI content-parts form
+            content = " ".join(
+                p.get("text", "") for p in content if isinstance(p, dict)
+            )
+        toks.extend(tokenizer.encode(content))
+    toks.append(role_ids["assistant"])  # generation prompt
+    return toks

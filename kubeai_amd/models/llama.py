@@ -47,6 +47,9 @@ class Attention(nn.Module):
     ) -> torch.Tensor:
         T = x.shape[0]
         qkv = F.linear(x, self.qkv_proj.weight)
+        lm = getattr(self, "_lora_manager", None)
+        if lm is not None and lm.active and fb.lora_ids is not None:
+            lm.apply(self.layer_idx, "qkv", x, qkv, fb.lora_ids)
         q, k, v = qkv.split(
             [self.n_q * self.hd, self.n_kv * self.hd, self.n_kv * self.hd], dim=-1
         )
@@ -78,7 +81,11 @@ class Attention(nn.Module):
                 fb.prefill_seq_lens,
                 self.scale,
             )
-        return F.linear(out.view(T, -1), self.o_proj.weight)
+        attn_flat = out.view(T, -1)
+        result = F.linear(attn_flat, self.o_proj.weight)
+        if lm is not None and lm.active and fb.lora_ids is not None:
+            lm.apply(self.layer_idx, "o", attn_flat, result, fb.lora_ids)
+        return result
 
 
 class MLP(nn.Module):

@@ -1,0 +1,177 @@
+"""Engine HTTP server: the full engine contract from SURVEY.md §2.16-bis.
+
+Uses the tiny model on CPU with FastAPI's TestClient (no sockets).
+"""
+import json
+
+import pytest
+from fastapi.testclient import TestClient
+
+from kubeai_amd.engine.engine import EngineConfig
+from kubeai_amd.engine.server import EngineServer, build_app
+
+
+@pytest.fixture(scope="module")
+def client():
+    cfg = EngineConfig(
+        model="llama-tiny", device="cpu", num_gpu_blocks=256, max_model_len=512
+    )
+    server = EngineServer(cfg, "test-model")
+    server.start()
+    server._ready.wait(timeout=60)
+    app = build_app(server)
+    with TestClient(app) as c:
+        yield c
+    server.stop()
+
+
+def test_health(client):
+    r = client.get("/health")
+    assert r.status_code == 200
+    assert r.json()["status"] == "ok"
+
+
+def test_completions_usage(client):
+    r = client.post(
+        "/v1/completions",
+        json={"model": "test-model", "prompt": "hello world foo bar", "max_tokens": 5,
+              "temperature": 0},
+    )
+    assert r.status_code == 200
+    body = r.json()
+    assert body["choices"][0]["finish_reason"] in ("length", "stop")
+    u = body["usage"]
+    # k6 reads these fields (benchmarks/multi-turn-chat-k6/k6.js:52-53)
+    assert u["prompt_tokens"] == 5  # bos + 4 words
+    assert 1 <= u["completion_tokens"] <= 5
+    assert u["total_tokens"] == u["prompt_tokens"] + u["completion_tokens"]
+
+
+def test_chat_completions(client):
+    r = client.post(
+        "/v1/chat/completions",
+        json={
+            "model": "test-model",
+            "messages": [
+                {"role": "system", "content": "be brief"},
+                {"role": "user", "content": "hi there"},
+            ],
+            "max_tokens": 4,
+            "temperature": 0,
+        },
+    )
+    assert r.status_code == 200
+    body = r.json()
+    assert body["object"] == "chat.completion"
+    assert body["choices"][0]["message"]["role"] == "assistant"
+    assert body["usage"]["completion_tokens"] >= 1
+
+
+def test_chat_streaming(client):
+    with client.stream(
+        "POST",
+        "/v1/chat/completions",
+        json={
+            "model": "test-model",
+            "messages": [{"role": "user", "content": "stream me"}],
+            "max_tokens": 4,
+            "temperature": 0,
+            "stream": True,
+        },
+    ) as r:
+        assert r.status_code == 200
+        events = []
+        for line in r.iter_lines():
+            if line.startswith("data: "):
+                events.append(line[len("data: ") :])
+    assert events[-1] == "[DONE]"
+    payloads = [json.loads(e) for e in events[:-1]]
+    assert payloads[0]["object"] == "chat.completion.chunk"
+    assert any(p["choices"][0]["finish_reason"] for p in payloads)
+    assert "usage" in payloads[-1]
+
+
+def test_models_listing(client):
+    r = client.get("/v1/models")
+    ids = [m["id"] for m in r.json()["data"]]
+    assert "test-model" in ids
+
+
+def test_metrics_exposition(client):
+    # generate some load first
+    client.post(
+        "/v1/completions",
+        json={"prompt": "metrics probe", "max_tokens": 2, "temperature": 0},
+    )
+    r = client.get("/metrics")
+    text = r.text
+    assert "kubeai_engine_num_requests_waiting" in text
+    assert "kubeai_engine_kv_cache_usage_perc" in text
+    assert "kubeai_engine_generation_tokens_total" in text
+    assert "kubeai_engine_time_to_first_token_seconds" in text
+
+
+def test_lora_admin_semantics(client):
+    # load
+    r = client.post(
+        "/v1/load_lora_adapter", json={"lora_name": "ad1", "lora_path": None}
+    )
+    assert r.status_code == 200
+    # double load -> "already loaded" (vllmclient matches this string,
+    # internal/vllmclient/client.go:30-45)
+    r = client.post(
+        "/v1/load_lora_adapter", json={"lora_name": "ad1", "lora_path": None}
+    )
+    assert r.status_code == 400 and "already loaded" in r.text
+    # adapter appears in /v1/models and is usable as model name
+    ids = [m["id"] for m in client.get("/v1/models").json()["data"]]
+    assert "ad1" in ids
+    r = client.post(
+        "/v1/completions",
+        json={"model": "ad1", "prompt": "adapter run", "max_tokens": 3,
+              "temperature": 0},
+    )
+    assert r.status_code == 200
+    # unload
+    r = client.post("/v1/unload_lora_adapter", json={"lora_name": "ad1"})
+    assert r.status_code == 200
+    # unload again -> "cannot be found" (client.go:59-73)
+    r = client.post("/v1/unload_lora_adapter", json={"lora_name": "ad1"})
+    assert r.status_code == 404 and "cannot be found" in r.text
+
+
+def test_lora_changes_output(client):
+    base = client.post(
+        "/v1/completions",
+        json={"prompt": "same prompt here", "max_tokens": 4, "temperature": 0},
+    ).json()["choices"][0]["text"]
+    client.post("/v1/load_lora_adapter", json={"lora_name": "delta", "lora_path": None})
+    adapted = client.post(
+        "/v1/completions",
+        json={"model": "delta", "prompt": "same prompt here", "max_tokens": 4,
+              "temperature": 0},
+    ).json()["choices"][0]["text"]
+    client.post("/v1/unload_lora_adapter", json={"lora_name": "delta"})
+    # random-init adapter must perturb the greedy path
+    assert adapted != base
+
+
+def test_unknown_model_404(client):
+    r = client.post(
+        "/v1/completions", json={"model": "nope", "prompt": "x", "max_tokens": 1}
+    )
+    assert r.status_code == 404
+
+
+def test_embeddings(client):
+    r = client.post(
+        "/v1/embeddings",
+        json={"model": "test-model", "input": ["hello world", "other text"]},
+    )
+    assert r.status_code == 200
+    data = r.json()["data"]
+    assert len(data) == 2
+    v = data[0]["embedding"]
+    assert len(v) == 256  # llama-tiny hidden size
+    norm = sum(x * x for x in v) ** 0.5
+    assert abs(norm - 1.0) < 1e-3

@@ -1,0 +1,120 @@
+"""Composition root (reference: internal/manager/run.go).
+
+Wires store + controller + load balancer + autoscaler + proxy + gateway +
+messengers + leader election and serves the OpenAI API. Usable two ways:
+
+  - CLI:   python -m kubeai_amd.controlplane.manager --config config.yaml
+  - tests: Manager(cfg, runtime=FakeRuntime(...)) -> await mgr.start()
+           (the envtest-style in-process harness, main_test.go:132-157)
+"""
+from __future__ import annotations
+
+import argparse
+import asyncio
+from typing import Optional
+
+from .autoscaler import Autoscaler
+from .config import SystemConfig, load_config
+from .controller import CacheManager, ModelController
+from .leader import Election
+from .loadbalancer import LoadBalancer
+from .messenger import MemBroker, Messenger
+from .modelclient import ModelClient
+from .openaiserver import build_gateway_app
+from .proxy import ProxyHandler
+from .runtime import FakeRuntime, LocalProcessRuntime
+from .store import Store
+
+
+class Manager:
+    def __init__(
+        self,
+        cfg: Optional[SystemConfig] = None,
+        runtime=None,
+        broker: Optional[MemBroker] = None,
+    ):
+        self.cfg = cfg or SystemConfig()
+        self.store = Store()
+        self.runtime = runtime or LocalProcessRuntime(self.store, n_gpus=self.cfg.n_gpus)
+        self.model_client = ModelClient(
+            self.store,
+            required_consecutive_scale_downs=self.cfg.autoscaling.required_consecutive_scale_downs,
+        )
+        self.lb = LoadBalancer(self.store)
+        self.controller = ModelController(
+            self.store,
+            self.runtime,
+            resource_profiles=self.cfg.resource_profiles,
+            cache=CacheManager(self.cfg.cache_dir),
+        )
+        self.election = Election(self.cfg.leader_lock_path)
+        self.autoscaler = Autoscaler(
+            self.store,
+            self.model_client,
+            interval=self.cfg.autoscaling.interval_seconds,
+            time_window=self.cfg.autoscaling.time_window_seconds,
+            self_metric_addrs=self.cfg.fixed_self_metric_addrs,
+            state_path=self.cfg.autoscaling.state_path,
+            is_leader=self.election.is_leader,
+        )
+        self.proxy = ProxyHandler(self.model_client, self.lb)
+        self.app = build_gateway_app(self.model_client, self.proxy)
+        self.broker = broker or MemBroker()
+        self.messengers = [
+            Messenger(
+                self.broker,
+                s.requests_url.split("://", 1)[-1],
+                s.responses_url.split("://", 1)[-1],
+                self.model_client,
+                self.lb,
+                max_handlers=s.max_handlers,
+            )
+            for s in self.cfg.messaging
+        ]
+
+    async def start(self) -> None:
+        self.lb.start()
+        self.controller.start()
+        self.election.start()
+        self.autoscaler.start()
+        for m in self.messengers:
+            m.start()
+
+    async def stop(self) -> None:
+        for m in self.messengers:
+            await m.stop()
+        await self.autoscaler.stop()
+        await self.election.stop()
+        await self.controller.stop()
+        await self.lb.stop()
+        await self.proxy.close()
+        if hasattr(self.runtime, "shutdown"):
+            await self.runtime.shutdown()
+
+
+async def run(cfg: SystemConfig) -> None:
+    import uvicorn
+
+    mgr = Manager(cfg)
+    await mgr.start()
+    server = uvicorn.Server(
+        uvicorn.Config(
+            mgr.app, host="0.0.0.0", port=cfg.api_port, log_level="warning"
+        )
+    )
+    try:
+        await server.serve()
+    finally:
+        await mgr.stop()
+
+
+def main() -> None:
+    p = argparse.ArgumentParser()
+    p.add_argument("--config", default=None)
+    args = p.parse_args()
+    cfg = load_config(args.config)
+    asyncio.run(run(cfg))
+
+
+if __name__ == "__main__":
+    main()

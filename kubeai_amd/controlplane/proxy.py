@@ -1,0 +1,127 @@
+"""Model proxy — parse, scale-from-zero, route, forward with retries.
+
+Reference parity (internal/modelproxy/handler.go):
+  parse -> active-request metric +- -> ScaleAtLeastOneReplica ->
+  AwaitBestAddress -> reverse proxy to http://<addr> -> retries <=3 on
+  {500,502,503,504} or connection error (handler.go:127-158); body kept
+  re-readable; streaming responses pass through untouched.
+"""
+from __future__ import annotations
+
+import json
+from typing import Optional
+
+import httpx
+from starlette.requests import Request
+from starlette.responses import JSONResponse, Response, StreamingResponse
+
+from .apiutils import APIError, parse_request
+from .loadbalancer import LoadBalancer
+from .modelclient import ModelClient
+
+RETRYABLE = {500, 502, 503, 504}
+
+
+class ProxyHandler:
+    def __init__(
+        self,
+        model_client: ModelClient,
+        lb: LoadBalancer,
+        max_retries: int = 3,
+        timeout: float = 600.0,
+        endpoint_timeout: float = 120.0,
+    ):
+        self.model_client = model_client
+        self.lb = lb
+        self.max_retries = max_retries
+        self.endpoint_timeout = endpoint_timeout
+        self.client = httpx.AsyncClient(timeout=timeout)
+
+    async def close(self) -> None:
+        await self.client.aclose()
+
+    async def handle(self, request: Request, path: str) -> Response:
+        try:
+            raw = await request.body()
+            try:
+                body = json.loads(raw) if raw else {}
+            except json.JSONDecodeError:
+                return JSONResponse({"error": "invalid JSON body"}, status_code=400)
+            selectors = _parse_selectors(request.headers.get("X-Label-Selector"))
+            try:
+                pr = parse_request(body, path, self.model_client.lookup_model, selectors)
+            except APIError as e:
+                return JSONResponse({"error": e.message}, status_code=e.status)
+
+            self.model_client.scale_at_least_one_replica(pr.model)
+            payload = json.dumps(pr.body).encode()
+            attempt = 0
+            while True:
+                attempt += 1
+                try:
+                    addr, done = await self.lb.await_best_address(
+                        pr.model,
+                        adapter=pr.adapter,
+                        prefix=pr.prefix,
+                        timeout=self.endpoint_timeout,
+                    )
+                except TimeoutError as e:
+                    return JSONResponse({"error": str(e)}, status_code=503)
+                try:
+                    resp = await self._forward(addr, path, payload, request)
+                except (httpx.ConnectError, httpx.ReadError, httpx.RemoteProtocolError):
+                    done()
+                    if attempt <= self.max_retries:
+                        continue
+                    return JSONResponse(
+                        {"error": "backend connection failed"}, status_code=502
+                    )
+                if resp.status_code in RETRYABLE and attempt <= self.max_retries:
+                    await resp.aclose()
+                    done()
+                    continue
+                return self._to_response(resp, done)
+        except Exception as e:  # noqa: BLE001
+            import traceback
+
+            traceback.print_exc()
+            return JSONResponse({"error": "internal proxy error"}, status_code=500)
+
+    async def _forward(self, addr: str, path: str, payload: bytes, request: Request):
+        url = f"http://{addr}{path}"
+        headers = {
+            k: v
+            for k, v in request.headers.items()
+            if k.lower() in ("content-type", "accept", "authorization", "traceparent")
+        }
+        req = self.client.build_request("POST", url, content=payload, headers=headers)
+        return await self.client.send(req, stream=True)
+
+    def _to_response(self, resp: httpx.Response, done) -> Response:
+        async def body_iter():
+            try:
+                async for chunk in resp.aiter_raw():
+                    yield chunk
+            finally:
+                await resp.aclose()
+                done()
+
+        headers = {
+            k: v
+            for k, v in resp.headers.items()
+            if k.lower() in ("content-type", "cache-control")
+        }
+        return StreamingResponse(
+            body_iter(), status_code=resp.status_code, headers=headers
+        )
+
+
+def _parse_selectors(header: Optional[str]) -> Optional[dict[str, str]]:
+    if not header:
+        return None
+    out = {}
+    for part in header.split(","):
+        k, _, v = part.partition("=")
+        if k:
+            out[k.strip()] = v.strip()
+    return out

@@ -35,6 +35,13 @@ def get_args():
     p.add_argument("--steps", type=int, default=60)
     p.add_argument("--warmup", type=int, default=20)
     p.add_argument("--model", default="llama-3-8b")
+    p.add_argument(
+        "--tp",
+        type=int,
+        default=1,
+        help="tensor-parallel degree; ranks form world//tp replicas "
+        "(config #4: --tp 8 --model llama-3-70b)",
+    )
     p.add_argument("--vus", type=int, default=32, help="virtual users per replica")
     p.add_argument("--system-tokens", type=int, default=64)
     p.add_argument("--user-tokens", type=int, default=48)
@@ -93,6 +100,8 @@ def main():
         torch.cuda.set_device(local_rank)
 
     dist = None
+    tp_group = None
+    tp_rank = 0
     if world > 1:
         import torch.distributed as dist_mod
 
@@ -100,6 +109,21 @@ def main():
         dist.init_process_group(
             backend="nccl" if use_cuda else "gloo", rank=rank, world_size=world
         )
+        if args.tp > 1:
+            assert world % args.tp == 0, "world size must be divisible by --tp"
+            from kubeai_amd.parallel.tp import TPGroup
+
+            replica_id = rank // args.tp
+            sub = None
+            for rid in range(world // args.tp):
+                ranks = list(range(rid * args.tp, (rid + 1) * args.tp))
+                g = dist.new_group(ranks)
+                if rid == replica_id:
+                    sub = g
+            tp_group = TPGroup(sub)
+            tp_rank = rank % args.tp
+    elif args.tp > 1:
+        raise SystemExit("--tp > 1 requires torchrun with nproc-per-node >= tp")
 
     import sys
 
@@ -115,6 +139,7 @@ def main():
         model = "llama-tiny"  # CPU sanity runs only
     arch = PRESETS[model]
 
+    replica_seed = 1234 + (rank // args.tp)
     eng = LLMEngine(
         EngineConfig(
             model=model,
@@ -125,8 +150,9 @@ def main():
             max_num_batched_tokens=8192,
             max_num_seqs=max(args.vus * 2, 64),
             enable_prefix_caching=not args.no_prefix_cache,
-            seed=1234 + rank,
-        )
+            seed=replica_seed,
+        ),
+        tp_group=tp_group,
     )
 
     log(f"engine ready: {eng.runner.num_blocks} kv blocks")
@@ -186,7 +212,9 @@ def main():
     t0 = time.monotonic()
     tokens = 0
     for _ in range(args.steps):
-        tokens += one_step()
+        t = one_step()
+        if tp_rank == 0:  # TP replicas run in lockstep; count once
+            tokens += t
     if use_cuda:
         torch.cuda.synchronize()
     if dist:
@@ -194,12 +222,14 @@ def main():
     t1 = time.monotonic()
     elapsed = t1 - t0
 
-    # TTFT for requests submitted during the timed region
+    # TTFT for requests submitted during the timed region (replica leaders)
     ttfts = [
         (first_token[r] - arrival[r]) * 1000.0
         for r in first_token
         if r not in timed_start_reqs and arrival.get(r, 0) >= t0
     ]
+    if tp_rank != 0:
+        ttfts = []
     p50_ttft = statistics.median(ttfts) if ttfts else None
 
     # aggregate across ranks: elapsed = MAX, tokens = SUM
@@ -247,7 +277,8 @@ def main():
                         "model": model,
                         "global_batch": args.vus * world,
                         "seq_len": args.max_model_len,
-                        "parallelism": f"dp{world}",
+                        "parallelism": f"dp{world // args.tp}"
+                        + (f"-tp{args.tp}" if args.tp > 1 else ""),
                         "max_tokens_per_turn": args.max_tokens,
                         "prefix_cache": not args.no_prefix_cache,
                         "prefix_cache_hit_rate": round(stats["prefix_cache_hit_rate"], 4),

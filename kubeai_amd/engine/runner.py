@@ -40,7 +40,9 @@ class ModelRunner:
         if tp_group is not None:
             from kubeai_amd.parallel.tp import TPLlamaForCausalLM
 
-            self.model = TPLlamaForCausalLM(arch, tp_group, device=self.device, dtype=dtype)
+            self.model = TPLlamaForCausalLM(
+                arch, tp_group, device=self.device, dtype=dtype, seed=seed
+            )
             self.n_kv_local = self.model.n_kv_local
         else:
             self.model = LlamaForCausalLM(arch, device=self.device, dtype=dtype)

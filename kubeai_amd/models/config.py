@@ -97,6 +97,19 @@ PRESETS: dict[str, ModelArchConfig] = {
         num_key_value_heads=4,
         head_dim=128,
     ),
+    # tiny with 2 KV heads so TP=2 shards cleanly (tests/test_tp.py)
+    "llama-tiny-tp": ModelArchConfig(
+        vocab_size=2048,
+        hidden_size=256,
+        intermediate_size=512,
+        num_hidden_layers=2,
+        num_attention_heads=4,
+        num_key_value_heads=2,
+        head_dim=128,
+        max_position_embeddings=2048,
+        bos_token_id=1,
+        eos_token_id=2,
+    ),
     "mixtral-tiny": ModelArchConfig(
         architecture="mixtral",
         vocab_size=2048,

@@ -1,0 +1,229 @@
+"""Tensor parallelism over RCCL/xGMI (one process per GPU).
+
+Sharding (MI355X-first):
+  - attention: head-sharded qkv (column-parallel), row-parallel o_proj with
+    one all-reduce; each rank owns n_kv/tp KV heads so the paged cache and
+    both attention kernels run unchanged on local heads (same GQA ratio).
+  - MLP: column-parallel gate_up, row-parallel down + all-reduce.
+  - embeddings + lm_head replicated (288 GB HBM/GPU makes the ~2 GB copy
+    cheaper than gathering vocab-parallel logits every step).
+  - decode-step all-reduces are small (T×H); over 7 p2p xGMI links a ring
+    is per-link bound, so RCCL's latency-oriented algorithms handle the
+    small-message case; large prefill activations ride the ring. Backend
+    "nccl" IS RCCL on ROCm; CPU tests use gloo.
+
+Weight init: every rank seeds the same RNG, materialises each full weight
+on CPU and keeps its slice — rank-identical to the single-GPU model, so
+TP=K output must equal TP=1 output (tests/test_tp.py asserts this).
+"""
+from __future__ import annotations
+
+import math
+
+import torch
+import torch.distributed as dist
+import torch.nn as nn
+import torch.nn.functional as F
+
+from kubeai_amd import ops
+from kubeai_amd.engine.batch import ForwardBatch
+from kubeai_amd.models.config import ModelArchConfig
+from kubeai_amd.models.llama import MLP, MoEMLP
+
+
+class TPGroup:
+    def __init__(self, group=None):
+        self.group = group
+        self.rank = dist.get_rank(group)
+        self.world = dist.get_world_size(group)
+
+    def all_reduce(self, t: torch.Tensor) -> torch.Tensor:
+        if self.world > 1:
+            dist.all_reduce(t, group=self.group)
+        return t
+
+
+def _shard_rows(full: torch.Tensor, tp: TPGroup) -> torch.Tensor:
+    # split dim 0 (output features of a column-parallel linear)
+    n = full.shape[0] // tp.world
+    return full[tp.rank * n : (tp.rank + 1) * n].clone()
+
+
+def _shard_cols(full: torch.Tensor, tp: TPGroup) -> torch.Tensor:
+    # split dim 1 (input features of a row-parallel linear)
+    n = full.shape[1] // tp.world
+    return full[:, tp.rank * n : (tp.rank + 1) * n].clone()
+
+
+def _full_weight(shape, gen) -> torch.Tensor:
+    # deterministic init with 1/sqrt(fan_in) scale; every rank draws the
+    # SAME full tensor (same generator state) and keeps only its shard
+    return torch.randn(shape, generator=gen) / math.sqrt(shape[1])
+
+
+class TPAttention(nn.Module):
+    def __init__(self, cfg: ModelArchConfig, layer_idx: int, tp: TPGroup):
+        super().__init__()
+        self.tp = tp
+        self.layer_idx = layer_idx
+        assert cfg.num_attention_heads % tp.world == 0, "q heads % tp != 0"
+        assert cfg.num_key_value_heads % tp.world == 0, "kv heads % tp != 0"
+        self.n_q = cfg.num_attention_heads // tp.world
+        self.n_kv = cfg.num_key_value_heads // tp.world
+        self.hd = cfg.head_dim
+        self.scale = 1.0 / math.sqrt(self.hd)
+        H = cfg.hidden_size
+        self.qkv_proj = nn.Linear(H, (self.n_q + 2 * self.n_kv) * self.hd, bias=False)
+        self.o_proj = nn.Linear(self.n_q * self.hd, H, bias=False)
+
+    def shard_from_full(self, full_qkv: torch.Tensor, full_o: torch.Tensor,
+                        cfg: ModelArchConfig) -> None:
+        tp = self.tp
+        nq, nkv, hd = cfg.num_attention_heads, cfg.num_key_value_heads, cfg.head_dim
+        q_w = full_qkv[: nq * hd]
+        k_w = full_qkv[nq * hd : (nq + nkv) * hd]
+        v_w = full_qkv[(nq + nkv) * hd :]
+        qs = _shard_rows(q_w.view(nq, hd, -1), tp).reshape(self.n_q * hd, -1)
+        ks = _shard_rows(k_w.view(nkv, hd, -1), tp).reshape(self.n_kv * hd, -1)
+        vs = _shard_rows(v_w.view(nkv, hd, -1), tp).reshape(self.n_kv * hd, -1)
+        self.qkv_proj.weight.data.copy_(torch.cat([qs, ks, vs], dim=0))
+        self.o_proj.weight.data.copy_(
+            full_o[:, tp.rank * self.n_q * hd : (tp.rank + 1) * self.n_q * hd].clone()
+        )
+
+    def forward(self, x, fb: ForwardBatch, kv_cache, cos_sin):
+        T = x.shape[0]
+        qkv = F.linear(x, self.qkv_proj.weight)
+        q, k, v = qkv.split(
+            [self.n_q * self.hd, self.n_kv * self.hd, self.n_kv * self.hd], dim=-1
+        )
+        q = q.view(T, self.n_q, self.hd).contiguous()
+        k = k.view(T, self.n_kv, self.hd).contiguous()
+        v = v.view(T, self.n_kv, self.hd).contiguous()
+        q, k = ops.rope(q, k, fb.positions, cos_sin)
+        k_cache, v_cache = kv_cache
+        ops.reshape_and_cache(k, v, k_cache, v_cache, fb.slot_mapping)
+        out = torch.empty_like(q)
+        nd = fb.n_decode
+        if nd > 0:
+            out[:nd] = ops.paged_attention_decode(
+                q[:nd].contiguous(), k_cache, v_cache,
+                fb.decode_block_tables, fb.decode_seq_lens, self.scale,
+            )
+        if fb.n_prefill > 0:
+            out[nd:] = ops.paged_attention_prefill(
+                q[nd:].contiguous(), k_cache, v_cache,
+                fb.prefill_block_tables, fb.prefill_query_start_loc,
+                fb.prefill_seq_lens, self.scale,
+            )
+        res = F.linear(out.view(T, -1), self.o_proj.weight)
+        return self.tp.all_reduce(res)
+
+
+class TPMLP(nn.Module):
+    def __init__(self, cfg: ModelArchConfig, tp: TPGroup):
+        super().__init__()
+        self.tp = tp
+        assert cfg.intermediate_size % tp.world == 0
+        self.i_local = cfg.intermediate_size // tp.world
+        self.gate_up_proj = nn.Linear(cfg.hidden_size, 2 * self.i_local, bias=False)
+        self.down_proj = nn.Linear(self.i_local, cfg.hidden_size, bias=False)
+
+    def shard_from_full(self, full_gate_up: torch.Tensor, full_down: torch.Tensor,
+                        cfg: ModelArchConfig) -> None:
+        I = cfg.intermediate_size
+        r, w = self.tp.rank, self.tp.world
+        gate = full_gate_up[:I][r * self.i_local : (r + 1) * self.i_local]
+        up = full_gate_up[I:][r * self.i_local : (r + 1) * self.i_local]
+        self.gate_up_proj.weight.data.copy_(torch.cat([gate, up], dim=0))
+        self.down_proj.weight.data.copy_(
+            full_down[:, r * self.i_local : (r + 1) * self.i_local]
+        )
+
+    def forward(self, x):
+        act = ops.silu_and_mul(F.linear(x, self.gate_up_proj.weight))
+        return self.tp.all_reduce(F.linear(act, self.down_proj.weight))
+
+
+class TPDecoderLayer(nn.Module):
+    def __init__(self, cfg: ModelArchConfig, layer_idx: int, tp: TPGroup):
+        super().__init__()
+        self.input_layernorm = nn.Parameter(torch.ones(cfg.hidden_size))
+        self.post_attention_layernorm = nn.Parameter(torch.ones(cfg.hidden_size))
+        self.self_attn = TPAttention(cfg, layer_idx, tp)
+        self.mlp = TPMLP(cfg, tp)
+        self.eps = cfg.rms_norm_eps
+
+    def forward(self, x, residual, fb, kv_cache, cos_sin):
+        if residual is None:
+            residual = x
+            x = ops.rmsnorm(x, self.input_layernorm, self.eps)
+        else:
+            x, residual = ops.fused_add_rmsnorm(x, residual, self.input_layernorm, self.eps)
+        x = self.self_attn(x, fb, kv_cache, cos_sin)
+        x, residual = ops.fused_add_rmsnorm(
+            x, residual, self.post_attention_layernorm, self.eps
+        )
+        x = self.mlp(x)
+        return x, residual
+
+
+class TPLlamaForCausalLM(nn.Module):
+    """Tensor-parallel Llama; rank-identical init so TP=K == TP=1 exactly
+    up to reduction order."""
+
+    def __init__(self, cfg: ModelArchConfig, tp: TPGroup, device=None,
+                 dtype=torch.bfloat16, seed: int = 0):
+        super().__init__()
+        self.cfg = cfg
+        self.tp = tp
+        self.n_kv_local = cfg.num_key_value_heads // tp.world
+        gen = torch.Generator().manual_seed(seed)
+        H = cfg.hidden_size
+        self.embed_tokens = nn.Embedding(cfg.vocab_size, H)
+        emb_full = torch.randn(cfg.vocab_size, H, generator=gen)
+        self.embed_tokens.weight.data.copy_(emb_full)
+        self.layers = nn.ModuleList(
+            [TPDecoderLayer(cfg, i, tp) for i in range(cfg.num_hidden_layers)]
+        )
+        nq, nkv, hd = cfg.num_attention_heads, cfg.num_key_value_heads, cfg.head_dim
+        for layer in self.layers:
+            full_qkv = _full_weight(((nq + 2 * nkv) * hd, H), gen)
+            full_o = _full_weight((H, nq * hd), gen)
+            layer.self_attn.shard_from_full(full_qkv, full_o, cfg)
+            full_gate_up = _full_weight((2 * cfg.intermediate_size, H), gen)
+            full_down = _full_weight((H, cfg.intermediate_size), gen)
+            layer.mlp.shard_from_full(full_gate_up, full_down, cfg)
+        self.norm = nn.Parameter(torch.ones(H))
+        if cfg.tie_word_embeddings:
+            self.lm_head = None
+        else:
+            self.lm_head = nn.Linear(H, cfg.vocab_size, bias=False)
+            self.lm_head.weight.data.copy_(_full_weight((cfg.vocab_size, H), gen))
+        self.to(device=device, dtype=dtype)
+        from kubeai_amd.ops import ref as ops_ref
+
+        self.register_buffer(
+            "cos_sin",
+            ops_ref.make_cos_sin_cache(
+                cfg.head_dim, cfg.max_position_embeddings, cfg.rope_theta
+            ).to(device=device),
+            persistent=False,
+        )
+
+    @torch.inference_mode()
+    def forward(self, fb: ForwardBatch) -> torch.Tensor:
+        x = self.embed_tokens(fb.input_ids.long())
+        residual = None
+        for i, layer in enumerate(self.layers):
+            x, residual = layer(x, residual, fb, self.kv_caches[i], self.cos_sin)
+        x, _ = ops.fused_add_rmsnorm(x, residual, self.norm, self.cfg.rms_norm_eps)
+        return x
+
+    @torch.inference_mode()
+    def compute_logits(self, hidden: torch.Tensor) -> torch.Tensor:
+        w = self.embed_tokens.weight if self.lm_head is None else self.lm_head.weight
+        return F.linear(hidden, w).float()
+
+    def bind_kv_caches(self, kv_caches):
+        self.kv_caches = kv_caches

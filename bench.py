@@ -42,7 +42,7 @@ def get_args():
         help="tensor-parallel degree; ranks form world//tp replicas "
         "(config #4: --tp 8 --model llama-3-70b)",
     )
-    p.add_argument("--vus", type=int, default=32, help="virtual users per replica")
+    p.add_argument("--vus", type=int, default=40, help="virtual users per replica (k6 headline: 80 VUs over 2 replicas)")
     p.add_argument("--system-tokens", type=int, default=64)
     p.add_argument("--user-tokens", type=int, default=48)
     p.add_argument("--max-tokens", type=int, default=32, help="per-turn completion")

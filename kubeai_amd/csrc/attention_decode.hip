@@ -34,7 +34,8 @@ __launch_bounds__(kBlockThreads) __global__ void paged_decode_kernel(
     const ushort* __restrict__ v_cache,
     const int32_t* __restrict__ block_tables,  // [B, max_blocks]
     const int32_t* __restrict__ seq_lens,      // [B]
-    const float scale, const int n_kv, const int max_blocks) {
+    const float scale, const int n_kv, const int max_blocks,
+    const int64_t q_stride) {
   const int b = blockIdx.x / n_kv;
   const int kh = blockIdx.x % n_kv;
   const int n_q = n_kv * G;
@@ -53,7 +54,7 @@ __launch_bounds__(kBlockThreads) __global__ void paged_decode_kernel(
   float qreg[G][2];
 #pragma unroll
   for (int g = 0; g < G; ++g) {
-    const ushort* qh = q + ((int64_t)b * n_q + kh * G + g) * kHD;
+    const ushort* qh = q + (int64_t)b * q_stride + (int64_t)(kh * G + g) * kHD;
     qreg[g][0] = bf16_to_f32(qh[2 * lane]) * scale;
     qreg[g][1] = bf16_to_f32(qh[2 * lane + 1]) * scale;
   }
@@ -153,7 +154,8 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
                             torch::Tensor k_cache, torch::Tensor v_cache,
                             torch::Tensor block_tables, torch::Tensor seq_lens,
                             double scale) {
-  TORCH_CHECK(q.is_contiguous() && out.is_contiguous());
+  TORCH_CHECK(q.stride(2) == 1 && q.stride(1) == q.size(2));
+  TORCH_CHECK(out.is_contiguous());
   TORCH_CHECK(k_cache.is_contiguous() && v_cache.is_contiguous());
   TORCH_CHECK(q.scalar_type() == torch::kBFloat16);
   TORCH_CHECK(block_tables.scalar_type() == torch::kInt32);
@@ -175,7 +177,7 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
                      (const ushort*)v_cache.data_ptr(),                    \
                      block_tables.data_ptr<int32_t>(),                     \
                      seq_lens.data_ptr<int32_t>(), (float)scale, n_kv,     \
-                     max_blocks)
+                     max_blocks, q.stride(0))
   switch (G) {
     case 1: LAUNCH(1); break;
     case 2: LAUNCH(2); break;

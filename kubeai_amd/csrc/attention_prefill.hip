@@ -49,7 +49,7 @@ __launch_bounds__(G * WAVE_SIZE) __global__ void paged_prefill_kernel(
     const int32_t* __restrict__ query_start_loc,  // [B+1]
     const int32_t* __restrict__ seq_lens,         // [B]
     const float scale, const int n_kv, const int max_blocks,
-    const int n_qtiles_max) {
+    const int64_t q_stride) {
   const int b = blockIdx.x;
   const int kh = blockIdx.y;
   const int qtile = blockIdx.z;
@@ -79,7 +79,7 @@ __launch_bounds__(G * WAVE_SIZE) __global__ void paged_prefill_kernel(
   bf16x8 q_frag[4];
   {
     const int qrow = min(row_lo + (lane & 15), q_len - 1);
-    const ushort* qp = q + ((int64_t)(s0 + qrow) * n_q + head) * kHD;
+    const ushort* qp = q + (int64_t)(s0 + qrow) * q_stride + (int64_t)head * kHD;
 #pragma unroll
     for (int kc = 0; kc < 4; ++kc) {
       q_frag[kc] =
@@ -249,7 +249,8 @@ void paged_attention_prefill(torch::Tensor out, torch::Tensor q,
                              torch::Tensor block_tables,
                              torch::Tensor query_start_loc,
                              torch::Tensor seq_lens, double scale) {
-  TORCH_CHECK(q.is_contiguous() && out.is_contiguous());
+  TORCH_CHECK(q.stride(2) == 1 && q.stride(1) == q.size(2));
+  TORCH_CHECK(out.is_contiguous());
   TORCH_CHECK(k_cache.is_contiguous() && v_cache.is_contiguous());
   TORCH_CHECK(q.scalar_type() == torch::kBFloat16);
   TORCH_CHECK(block_tables.scalar_type() == torch::kInt32);
@@ -278,7 +279,7 @@ void paged_attention_prefill(torch::Tensor out, torch::Tensor q,
                      block_tables.data_ptr<int32_t>(),                      \
                      query_start_loc.data_ptr<int32_t>(),                   \
                      seq_lens.data_ptr<int32_t>(), (float)scale, n_kv,      \
-                     max_blocks, n_qtiles_max)
+                     max_blocks, q.stride(0))
   switch (G) {
     case 1: LAUNCH(1); break;
     case 2: LAUNCH(2); break;

@@ -15,7 +15,8 @@ __global__ void reshape_and_cache_kernel(
     ushort* __restrict__ k_cache,   // [nb, n_kv, bs, hd]
     ushort* __restrict__ v_cache,
     const int64_t* __restrict__ slot_mapping,  // [T]
-    const int n_kv, const int bs, const int hd, const int64_t n_tok) {
+    const int n_kv, const int bs, const int hd, const int64_t n_tok,
+    const int64_t kv_stride) {
   // one wave per (token, kv_head); lane i copies 8 elems (hd=128 -> 2 vec/lane)
   const int64_t flat = (int64_t)blockIdx.x * (blockDim.x / WAVE_SIZE) +
                        threadIdx.x / WAVE_SIZE;
@@ -29,9 +30,9 @@ __global__ void reshape_and_cache_kernel(
   const int off = (int)(slot % bs);
 
   const ushort8* src_k =
-      reinterpret_cast<const ushort8*>(k + ((int64_t)tok * n_kv + h) * hd);
+      reinterpret_cast<const ushort8*>(k + tok * kv_stride + (int64_t)h * hd);
   const ushort8* src_v =
-      reinterpret_cast<const ushort8*>(v + ((int64_t)tok * n_kv + h) * hd);
+      reinterpret_cast<const ushort8*>(v + tok * kv_stride + (int64_t)h * hd);
   ushort8* dst_k = reinterpret_cast<ushort8*>(
       k_cache + (((int64_t)blk * n_kv + h) * bs + off) * hd);
   ushort8* dst_v = reinterpret_cast<ushort8*>(
@@ -47,7 +48,9 @@ __global__ void reshape_and_cache_kernel(
 
 void reshape_and_cache(torch::Tensor k, torch::Tensor v, torch::Tensor k_cache,
                        torch::Tensor v_cache, torch::Tensor slot_mapping) {
-  TORCH_CHECK(k.is_contiguous() && v.is_contiguous());
+  TORCH_CHECK(k.stride(2) == 1 && k.stride(1) == k.size(2));
+  TORCH_CHECK(k.stride(0) == v.stride(0), "k/v must share row stride");
+  TORCH_CHECK(v.stride(2) == 1 && v.stride(1) == v.size(2));
   TORCH_CHECK(k_cache.is_contiguous() && v_cache.is_contiguous());
   TORCH_CHECK(slot_mapping.scalar_type() == torch::kInt64);
   const int T = k.size(0), n_kv = k.size(1), hd = k.size(2);
@@ -64,6 +67,6 @@ void reshape_and_cache(torch::Tensor k, torch::Tensor v, torch::Tensor k_cache,
                      (const ushort*)k.data_ptr(), (const ushort*)v.data_ptr(),
                      (ushort*)k_cache.data_ptr(), (ushort*)v_cache.data_ptr(),
                      slot_mapping.data_ptr<int64_t>(), n_kv, bs, hd,
-                     (int64_t)T);
+                     (int64_t)T, k.stride(0));
   HIP_CHECK_KERNEL();
 }

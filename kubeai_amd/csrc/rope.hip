@@ -17,7 +17,8 @@ __global__ void rope_kernel(
     ushort* __restrict__ k,        // [T, n_kv, hd]
     const int32_t* __restrict__ positions,  // [T]
     const float* __restrict__ cos_sin,      // [max_pos, hd]
-    const int n_q, const int n_kv, const int hd, const int64_t n_tok) {
+    const int n_q, const int n_kv, const int hd, const int64_t n_tok,
+    const int64_t q_stride, const int64_t k_stride) {
   const int n_heads = n_q + n_kv;
   const int64_t flat = (int64_t)blockIdx.x * (blockDim.x / WAVE_SIZE) +
                        threadIdx.x / WAVE_SIZE;
@@ -30,8 +31,8 @@ __global__ void rope_kernel(
   const int pos = positions[tok];
   const float* cs = cos_sin + (int64_t)pos * hd;
   ushort* base = (head < n_q)
-                     ? q + ((int64_t)tok * n_q + head) * hd
-                     : k + ((int64_t)tok * n_kv + (head - n_q)) * hd;
+                     ? q + tok * q_stride + (int64_t)head * hd
+                     : k + tok * k_stride + (int64_t)(head - n_q) * hd;
 
   for (int i = lane; i < half; i += WAVE_SIZE) {
     const float c = cs[i];
@@ -47,7 +48,9 @@ __global__ void rope_kernel(
 
 void rope(torch::Tensor q, torch::Tensor k, torch::Tensor positions,
           torch::Tensor cos_sin) {
-  TORCH_CHECK(q.is_contiguous() && k.is_contiguous());
+  // q/k may be row-strided views into the fused qkv projection output
+  TORCH_CHECK(q.stride(2) == 1 && q.stride(1) == q.size(2));
+  TORCH_CHECK(k.stride(2) == 1 && k.stride(1) == k.size(2));
   TORCH_CHECK(q.scalar_type() == torch::kBFloat16);
   TORCH_CHECK(positions.scalar_type() == torch::kInt32);
   TORCH_CHECK(cos_sin.scalar_type() == torch::kFloat32);
@@ -63,6 +66,7 @@ void rope(torch::Tensor q, torch::Tensor k, torch::Tensor positions,
                      c10::hip::getCurrentHIPStream().stream(), (ushort*)q.data_ptr(),
                      (ushort*)k.data_ptr(),
                      positions.data_ptr<int32_t>(),
-                     cos_sin.data_ptr<float>(), n_q, n_kv, hd, (int64_t)T);
+                     cos_sin.data_ptr<float>(), n_q, n_kv, hd, (int64_t)T,
+                     q.stride(0), k.stride(0));
   HIP_CHECK_KERNEL();
 }

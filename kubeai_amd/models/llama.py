@@ -53,33 +53,39 @@ class Attention(nn.Module):
         q, k, v = qkv.split(
             [self.n_q * self.hd, self.n_kv * self.hd, self.n_kv * self.hd], dim=-1
         )
-        q = q.view(T, self.n_q, self.hd).contiguous()
-        k = k.view(T, self.n_kv, self.hd).contiguous()
-        v = v.view(T, self.n_kv, self.hd).contiguous()
+        # strided views into the qkv buffer — the HIP kernels take a row
+        # stride, so no .contiguous() copies (saves 3 HBM round trips/layer)
+        q = q.unflatten(-1, (self.n_q, self.hd))
+        k = k.unflatten(-1, (self.n_kv, self.hd))
+        v = v.unflatten(-1, (self.n_kv, self.hd))
         q, k = ops.rope(q, k, fb.positions, cos_sin)
         k_cache, v_cache = kv_cache
         ops.reshape_and_cache(k, v, k_cache, v_cache, fb.slot_mapping)
 
-        out = torch.empty_like(q)
+        out = torch.empty(
+            (T, self.n_q, self.hd), dtype=qkv.dtype, device=qkv.device
+        )
         nd = fb.n_decode
         if nd > 0:
-            out[:nd] = ops.paged_attention_decode(
-                q[:nd].contiguous(),
+            ops.paged_attention_decode(
+                q[:nd],
                 k_cache,
                 v_cache,
                 fb.decode_block_tables,
                 fb.decode_seq_lens,
                 self.scale,
+                out=out[:nd],
             )
         if fb.n_prefill > 0:
-            out[nd:] = ops.paged_attention_prefill(
-                q[nd:].contiguous(),
+            ops.paged_attention_prefill(
+                q[nd:],
                 k_cache,
                 v_cache,
                 fb.prefill_block_tables,
                 fb.prefill_query_start_loc,
                 fb.prefill_seq_lens,
                 self.scale,
+                out=out[nd:],
             )
         attn_flat = out.view(T, -1)
         result = F.linear(attn_flat, self.o_proj.weight)

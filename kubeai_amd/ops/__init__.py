@@ -68,30 +68,47 @@ def reshape_and_cache(k, v, k_cache, v_cache, slot_mapping) -> None:
     ref.reshape_and_cache(k, v, k_cache, v_cache, slot_mapping)
 
 
-def paged_attention_decode(q, k_cache, v_cache, block_tables, seq_lens, scale: float):
+def paged_attention_decode(
+    q, k_cache, v_cache, block_tables, seq_lens, scale: float, out=None
+):
+    """q may be a row-strided view (fused qkv output); out must be
+    contiguous (allocated here if not supplied)."""
     if q.is_cuda:
         _require_ext()
-        out = torch.empty_like(q)
+        if out is None:
+            out = torch.empty(q.shape, dtype=q.dtype, device=q.device)
         _C.paged_attention_decode(
             out, q, k_cache, v_cache, block_tables, seq_lens, scale
         )
         return out
-    return ref.paged_attention_decode(q, k_cache, v_cache, block_tables, seq_lens, scale)
+    res = ref.paged_attention_decode(
+        q, k_cache, v_cache, block_tables, seq_lens, scale
+    )
+    if out is not None:
+        out.copy_(res)
+        return out
+    return res
 
 
 def paged_attention_prefill(
-    q, k_cache, v_cache, block_tables, query_start_loc, seq_lens, scale: float
+    q, k_cache, v_cache, block_tables, query_start_loc, seq_lens, scale: float,
+    out=None,
 ):
     if q.is_cuda:
         _require_ext()
-        out = torch.empty_like(q)
+        if out is None:
+            out = torch.empty(q.shape, dtype=q.dtype, device=q.device)
         _C.paged_attention_prefill(
             out, q, k_cache, v_cache, block_tables, query_start_loc, seq_lens, scale
         )
         return out
-    return ref.paged_attention_prefill(
+    res = ref.paged_attention_prefill(
         q, k_cache, v_cache, block_tables, query_start_loc, seq_lens, scale
     )
+    if out is not None:
+        out.copy_(res)
+        return out
+    return res
 
 
 def silu_and_mul(x):

@@ -9,6 +9,8 @@ import sys
 
 import torch
 
+import os, sys
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import kubeai_amd.ops as ops
 
 HBM_GBS = 6300.0

@@ -137,6 +137,54 @@ def test_paged_prefill(nq, nkv, q_lens, ctx_lens):
     assert_close_bf16(out, expected, atol=3e-2, rtol=3e-2)
 
 
+def test_strided_qkv_views():
+    """rope/cache/decode must accept row-strided views into the fused qkv
+    buffer (the no-copy path the model uses)."""
+    T, nq, nkv, hd, bs = 24, 32, 8, 128, 16
+    W = (nq + 2 * nkv) * hd
+    qkv = torch.randn(T, W, dtype=torch.bfloat16, device=DEV)
+    qkv_ref = qkv.clone()
+    pos = torch.randint(0, 1000, (T,), dtype=torch.int32, device=DEV)
+    cs = ref.make_cos_sin_cache(hd, 2048, 500000.0).to(DEV)
+
+    q = qkv[:, : nq * hd].unflatten(-1, (nq, hd))
+    k = qkv[:, nq * hd : (nq + nkv) * hd].unflatten(-1, (nkv, hd))
+    v = qkv[:, (nq + nkv) * hd :].unflatten(-1, (nkv, hd))
+    ops.rope(q, k, pos, cs)
+
+    qc = qkv_ref[:, : nq * hd].reshape(T, nq, hd).contiguous()
+    kc_ = qkv_ref[:, nq * hd : (nq + nkv) * hd].reshape(T, nkv, hd).contiguous()
+    q_ref, k_ref = ref.rope(qc.float(), kc_.float(), pos, cs)
+    assert_close_bf16(q, q_ref)
+    assert_close_bf16(k, k_ref)
+
+    # cache write from strided views
+    nb = T + 1
+    kcache = torch.zeros(nb, nkv, bs, hd, dtype=torch.bfloat16, device=DEV)
+    vcache = torch.zeros_like(kcache)
+    slots = torch.arange(T, dtype=torch.int64, device=DEV)
+    ops.reshape_and_cache(k, v, kcache, vcache, slots)
+    kcache_ref = torch.zeros_like(kcache)
+    vcache_ref = torch.zeros_like(vcache)
+    ref.reshape_and_cache(
+        k.contiguous(), v.contiguous(), kcache_ref, vcache_ref, slots
+    )
+    torch.testing.assert_close(kcache, kcache_ref)
+    torch.testing.assert_close(vcache, vcache_ref)
+
+    # decode from a strided q view
+    B = T
+    bt = torch.arange(0, 1, dtype=torch.int32, device=DEV).repeat(B, 1)
+    seq_lens = torch.full((B,), bs, dtype=torch.int32, device=DEV)
+    kcache2 = torch.randn(2, nkv, bs, hd, dtype=torch.bfloat16, device=DEV)
+    vcache2 = torch.randn_like(kcache2)
+    out = ops.paged_attention_decode(q, kcache2, vcache2, bt, seq_lens, 0.1)
+    expected = ref.paged_attention_decode(
+        q.contiguous().float(), kcache2.float(), vcache2.float(), bt, seq_lens, 0.1
+    )
+    assert_close_bf16(out, expected, atol=3e-2, rtol=3e-2)
+
+
 def test_silu_and_mul():
     x = torch.randn(77, 2 * 14336, dtype=torch.bfloat16, device=DEV)
     out = ops.silu_and_mul(x)

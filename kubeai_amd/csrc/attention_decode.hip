@@ -35,14 +35,21 @@ __launch_bounds__(kBlockThreads) __global__ void paged_decode_kernel(
     const int32_t* __restrict__ block_tables,  // [B, max_blocks]
     const int32_t* __restrict__ seq_lens,      // [B]
     const float scale, const int n_kv, const int max_blocks,
-    const int64_t q_stride) {
-  const int b = blockIdx.x / n_kv;
-  const int kh = blockIdx.x % n_kv;
+    const int64_t q_stride, const int n_splits,
+    float* __restrict__ part_o,    // [B, n_q, n_splits, hd]
+    float* __restrict__ part_ml) { // [B, n_q, n_splits, 2]
+  const int b = blockIdx.x / (n_kv * n_splits);
+  const int rem = blockIdx.x % (n_kv * n_splits);
+  const int kh = rem / n_splits;
+  const int split = rem % n_splits;
   const int n_q = n_kv * G;
   const int wave = threadIdx.x / WAVE_SIZE;
   const int lane = threadIdx.x % WAVE_SIZE;
   const int L = seq_lens[b];
   const int n_blocks = (L + kBS - 1) / kBS;
+  const int chunk = (n_blocks + n_splits - 1) / n_splits;
+  const int blk_lo = split * chunk;
+  const int blk_hi = min(n_blocks, blk_lo + chunk);
 
   // LDS: per-wave KV tile buffers + merge scratch
   __shared__ ushort k_lds[kWaves][kBS][kHD];
@@ -69,7 +76,7 @@ __launch_bounds__(kBlockThreads) __global__ void paged_decode_kernel(
 
   const int32_t* bt = block_tables + (int64_t)b * max_blocks;
 
-  for (int blk_i = wave; blk_i < n_blocks; blk_i += kWaves) {
+  for (int blk_i = blk_lo + wave; blk_i < blk_hi; blk_i += kWaves) {
     const int64_t blk = bt[blk_i];
     const int tile_start = blk_i * kBS;
     const int tile_len = min(kBS, L - tile_start);
@@ -133,19 +140,64 @@ __launch_bounds__(kBlockThreads) __global__ void paged_decode_kernel(
 #pragma unroll
     for (int w = 0; w < kWaves; ++w) m_star = fmaxf(m_star, merge_ml[w][g][0]);
     float l_star = 0.f, o0 = 0.f, o1 = 0.f;
+    if (m_star != -INFINITY) {
 #pragma unroll
-    for (int w = 0; w < kWaves; ++w) {
-      const float c = __expf(merge_ml[w][g][0] - m_star);
-      l_star += merge_ml[w][g][1] * c;
-      o0 += merge_o[w][g][2 * lane] * c;
-      o1 += merge_o[w][g][2 * lane + 1] * c;
+      for (int w = 0; w < kWaves; ++w) {
+        const float c = __expf(merge_ml[w][g][0] - m_star);
+        l_star += merge_ml[w][g][1] * c;
+        o0 += merge_o[w][g][2 * lane] * c;
+        o1 += merge_o[w][g][2 * lane + 1] * c;
+      }
     }
-    const float inv_l = 1.0f / l_star;
-    ushort* oh = out + ((int64_t)b * n_q + kh * G + g) * kHD;
-    uint32_t packed = ((uint32_t)f32_to_bf16(o1 * inv_l) << 16) |
-                      f32_to_bf16(o0 * inv_l);
-    *reinterpret_cast<uint32_t*>(&oh[2 * lane]) = packed;
+    const int head = kh * G + g;
+    if (n_splits == 1) {
+      const float inv_l = 1.0f / l_star;
+      ushort* oh = out + ((int64_t)b * n_q + head) * kHD;
+      uint32_t packed = ((uint32_t)f32_to_bf16(o1 * inv_l) << 16) |
+                        f32_to_bf16(o0 * inv_l);
+      *reinterpret_cast<uint32_t*>(&oh[2 * lane]) = packed;
+    } else {
+      // flash-decoding partials (empty splits write m=-inf, l=0, o=0)
+      float* po = part_o + (((int64_t)b * n_q + head) * n_splits + split) * kHD;
+      po[2 * lane] = o0;
+      po[2 * lane + 1] = o1;
+      if (lane == 0) {
+        float* pm = part_ml + (((int64_t)b * n_q + head) * n_splits + split) * 2;
+        pm[0] = m_star;
+        pm[1] = l_star;
+      }
+    }
   }
+}
+
+// Merge flash-decoding partials: one wave per (seq, q_head).
+__global__ void decode_merge_kernel(
+    ushort* __restrict__ out,             // [B, n_q, hd]
+    const float* __restrict__ part_o,     // [B, n_q, n_splits, hd]
+    const float* __restrict__ part_ml,    // [B, n_q, n_splits, 2]
+    const int n_q, const int n_splits, const int64_t n_bh) {
+  const int64_t bh = (int64_t)blockIdx.x * (blockDim.x / WAVE_SIZE) +
+                     threadIdx.x / WAVE_SIZE;
+  if (bh >= n_bh) return;
+  const int lane = threadIdx.x % WAVE_SIZE;
+  const float* pml = part_ml + bh * n_splits * 2;
+  const float* po = part_o + bh * n_splits * kHD;
+  float m_star = -INFINITY;
+  for (int s = 0; s < n_splits; ++s) m_star = fmaxf(m_star, pml[2 * s]);
+  float l_star = 0.f, o0 = 0.f, o1 = 0.f;
+  for (int s = 0; s < n_splits; ++s) {
+    const float ms = pml[2 * s];
+    if (ms == -INFINITY) continue;
+    const float c = __expf(ms - m_star);
+    l_star += pml[2 * s + 1] * c;
+    o0 += po[s * kHD + 2 * lane] * c;
+    o1 += po[s * kHD + 2 * lane + 1] * c;
+  }
+  const float inv_l = 1.0f / l_star;
+  ushort* oh = out + bh * kHD;
+  uint32_t packed = ((uint32_t)f32_to_bf16(o1 * inv_l) << 16) |
+                    f32_to_bf16(o0 * inv_l);
+  *reinterpret_cast<uint32_t*>(&oh[2 * lane]) = packed;
 }
 
 }  // namespace
@@ -168,7 +220,23 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
   const int G = n_q / n_kv;
   TORCH_CHECK(n_q % n_kv == 0);
   if (B == 0) return;
-  dim3 grid(B * n_kv), block(kBlockThreads);
+  // flash-decoding split: target >=1024 workgroups to fill 256 CUs
+  int n_splits = 1;
+  const int base_wgs = B * n_kv;
+  if (base_wgs < 1024) {
+    n_splits = std::min<int>(16, (1024 + base_wgs - 1) / base_wgs);
+  }
+  torch::Tensor part_o, part_ml;
+  float *part_o_ptr = nullptr, *part_ml_ptr = nullptr;
+  if (n_splits > 1) {
+    auto opts =
+        torch::TensorOptions().device(q.device()).dtype(torch::kFloat32);
+    part_o = torch::empty({B, n_q, n_splits, (int64_t)kHD}, opts);
+    part_ml = torch::empty({B, n_q, n_splits, 2}, opts);
+    part_o_ptr = part_o.data_ptr<float>();
+    part_ml_ptr = part_ml.data_ptr<float>();
+  }
+  dim3 grid(B * n_kv * n_splits), block(kBlockThreads);
   auto stream = c10::hip::getCurrentHIPStream().stream();
 #define LAUNCH(GG)                                                        \
   hipLaunchKernelGGL((paged_decode_kernel<GG>), grid, block, 0, stream,   \
@@ -177,7 +245,8 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
                      (const ushort*)v_cache.data_ptr(),                    \
                      block_tables.data_ptr<int32_t>(),                     \
                      seq_lens.data_ptr<int32_t>(), (float)scale, n_kv,     \
-                     max_blocks, q.stride(0))
+                     max_blocks, q.stride(0), n_splits, part_o_ptr,        \
+                     part_ml_ptr)
   switch (G) {
     case 1: LAUNCH(1); break;
     case 2: LAUNCH(2); break;
@@ -189,4 +258,14 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
   }
 #undef LAUNCH
   HIP_CHECK_KERNEL();
+  if (n_splits > 1) {
+    const int64_t n_bh = (int64_t)B * n_q;
+    const int wpb = 4;
+    hipLaunchKernelGGL(decode_merge_kernel,
+                       dim3((uint32_t)((n_bh + wpb - 1) / wpb)),
+                       dim3(wpb * WAVE_SIZE), 0, stream,
+                       (ushort*)out.data_ptr(), part_o_ptr, part_ml_ptr, n_q,
+                       n_splits, n_bh);
+    HIP_CHECK_KERNEL();
+  }
 }

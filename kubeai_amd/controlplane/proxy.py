@@ -8,7 +8,9 @@ Reference parity (internal/modelproxy/handler.go):
 """
 from __future__ import annotations
 
+import dataclasses
 import json
+import re
 from typing import Optional
 
 import httpx
@@ -43,18 +45,29 @@ class ProxyHandler:
     async def handle(self, request: Request, path: str) -> Response:
         try:
             raw = await request.body()
-            try:
-                body = json.loads(raw) if raw else {}
-            except json.JSONDecodeError:
-                return JSONResponse({"error": "invalid JSON body"}, status_code=400)
+            ctype = request.headers.get("content-type", "")
+            multipart = ctype.startswith("multipart/form-data")
+            if multipart:
+                # audio/transcriptions etc. — extract the model form field,
+                # forward the multipart body untouched (reference:
+                # apiutils/request.go:109-165)
+                model_field = _extract_multipart_field(raw, ctype, "model")
+                body = {"model": model_field}
+            else:
+                try:
+                    body = json.loads(raw) if raw else {}
+                except json.JSONDecodeError:
+                    return JSONResponse({"error": "invalid JSON body"}, status_code=400)
             selectors = _parse_selectors(request.headers.get("X-Label-Selector"))
             try:
                 pr = parse_request(body, path, self.model_client.lookup_model, selectors)
             except APIError as e:
                 return JSONResponse({"error": e.message}, status_code=e.status)
+            if multipart:
+                pr = dataclasses.replace(pr, body=None)  # raw passthrough
 
             self.model_client.scale_at_least_one_replica(pr.model)
-            payload = json.dumps(pr.body).encode()
+            payload = raw if pr.body is None else json.dumps(pr.body).encode()
             attempt = 0
             while True:
                 attempt += 1
@@ -94,6 +107,7 @@ class ProxyHandler:
             for k, v in request.headers.items()
             if k.lower() in ("content-type", "accept", "authorization", "traceparent")
         }
+        headers.setdefault("content-type", "application/json")
         req = self.client.build_request("POST", url, content=payload, headers=headers)
         return await self.client.send(req, stream=True)
 
@@ -114,6 +128,19 @@ class ProxyHandler:
         return StreamingResponse(
             body_iter(), status_code=resp.status_code, headers=headers
         )
+
+
+def _extract_multipart_field(raw: bytes, content_type: str, field: str) -> Optional[str]:
+    """Minimal multipart/form-data field extraction (no multipart lib in
+    this image). Good for small text fields like `model`."""
+    pat = re.compile(
+        rb'Content-Disposition:\s*form-data;\s*name="'
+        + re.escape(field.encode())
+        + rb'"\r?\n\r?\n([^\r\n]*)',
+        re.IGNORECASE,
+    )
+    m = pat.search(raw)
+    return m.group(1).decode(errors="replace") if m else None
 
 
 def _parse_selectors(header: Optional[str]) -> Optional[dict[str, str]]:

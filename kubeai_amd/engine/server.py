@@ -292,6 +292,44 @@ def build_app(server: EngineServer) -> FastAPI:
             "usage": _usage(sum(len(t) for t in tok_lists), 0),
         }
 
+    @app.post("/v1/rerank")
+    async def rerank(request: Request):
+        """Score documents against a query (Reranking feature; reference
+        analog: the Infinity engine's /rerank)."""
+        body = await request.json()
+        query = body.get("query", "")
+        docs = body.get("documents", []) or []
+        top_n = body.get("top_n") or len(docs)
+        tok_lists = [server.tokenizer.encode(t, add_bos=True) for t in [query] + docs]
+        vecs = await asyncio.get_running_loop().run_in_executor(
+            None, server.engine.embed, tok_lists
+        )
+        qv = vecs[0]
+        scores = [
+            sum(a * b for a, b in zip(qv, dv)) for dv in vecs[1:]
+        ]  # unit-norm vectors -> cosine
+        order = sorted(range(len(docs)), key=lambda i: -scores[i])[: int(top_n)]
+        return {
+            "model": name,
+            "results": [
+                {"index": i, "relevance_score": scores[i],
+                 "document": {"text": docs[i]}}
+                for i in order
+            ],
+            "usage": _usage(sum(len(t) for t in tok_lists), 0),
+        }
+
+    @app.post("/v1/audio/transcriptions")
+    async def transcriptions(request: Request):
+        # SpeechToText is a control-plane-routable feature; this engine
+        # build has no audio frontend (no whisper weights offline) — return
+        # the standard error shape rather than fake text.
+        return JSONResponse(
+            {"error": {"message": f"model {name} does not support audio "
+                                  "transcription in this build"}},
+            status_code=501,
+        )
+
     @app.post("/v1/load_lora_adapter")
     async def load_lora(request: Request):
         body = await request.json()

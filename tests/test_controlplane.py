@@ -86,10 +86,16 @@ class FakeBackend:
         async def metrics(request):
             return JSONResponse({})
 
+        async def transcriptions(request):
+            raw = await request.body()
+            self.requests.append((request.url.path, raw))
+            return JSONResponse({"text": "transcribed"})
+
         return Starlette(
             routes=[
                 Route("/v1/completions", completions, methods=["POST"]),
                 Route("/v1/chat/completions", completions, methods=["POST"]),
+                Route("/v1/audio/transcriptions", transcriptions, methods=["POST"]),
                 Route("/v1/load_lora_adapter", load_lora, methods=["POST"]),
                 Route("/v1/unload_lora_adapter", unload_lora, methods=["POST"]),
                 Route("/metrics", metrics, methods=["GET"]),
@@ -362,6 +368,33 @@ def test_models_listing_by_feature():
                 r = await client.get("/openai/v1/models?feature=TextEmbedding")
                 ids = [d["id"] for d in r.json()["data"]]
                 assert ids == ["emb"]
+
+    run(body())
+
+
+def test_proxy_multipart_audio_routing():
+    """SpeechToText path: multipart body, model extracted from the form
+    field, raw body forwarded (reference: apiutils/request.go:109-165)."""
+
+    async def body():
+        m = text_gen_model("whisper")
+        m.spec.features = ["SpeechToText"]
+        m.spec.replicas = 1
+        async with harness([m]) as (mgr, runtime, backend):
+            rep = (await wait_for(lambda: mgr.store.list_replicas("whisper")))[0]
+            runtime.mark_ready(rep.name, backend.address)
+            transport = httpx.ASGITransport(app=mgr.app)
+            async with httpx.AsyncClient(transport=transport, base_url="http://gw") as client:
+                resp = await client.post(
+                    "/openai/v1/audio/transcriptions",
+                    files={"file": ("a.wav", b"RIFFxxxx", "audio/wav")},
+                    data={"model": "whisper"},
+                )
+            assert resp.status_code == 200
+            assert resp.json()["text"] == "transcribed"
+            path, raw = backend.requests[-1]
+            assert path == "/v1/audio/transcriptions"
+            assert b"RIFFxxxx" in raw  # multipart forwarded untouched
 
     run(body())
 

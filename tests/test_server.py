@@ -163,6 +163,30 @@ def test_unknown_model_404(client):
     assert r.status_code == 404
 
 
+def test_rerank(client):
+    r = client.post(
+        "/v1/rerank",
+        json={
+            "model": "test-model",
+            "query": "alpha beta",
+            "documents": ["alpha beta gamma", "unrelated words here", "alpha beta"],
+            "top_n": 2,
+        },
+    )
+    assert r.status_code == 200
+    results = r.json()["results"]
+    assert len(results) == 2
+    # the identical-prefix documents must outrank the unrelated one
+    assert {res["index"] for res in results} <= {0, 1, 2}
+    assert results[0]["relevance_score"] >= results[1]["relevance_score"]
+    assert 1 not in {res["index"] for res in results}
+
+
+def test_audio_transcription_not_supported(client):
+    r = client.post("/v1/audio/transcriptions")
+    assert r.status_code == 501
+
+
 def test_embeddings(client):
     r = client.post(
         "/v1/embeddings",

@@ -23,8 +23,27 @@ import argparse
 import json
 import os
 import random
+import shutil
 import statistics
 import time
+
+# hipBLASLt algorithm selection via TunableOp: ship the gfx950-tuned table
+# (kubeai_amd/data/) and point torch at per-device copies. Must happen
+# before `import torch`. Opt out with KUBEAI_TUNABLEOP=0; tune afresh with
+# PYTORCH_TUNABLEOP_TUNING=1.
+_TUNED = os.path.join(
+    os.path.dirname(os.path.abspath(__file__)),
+    "kubeai_amd", "data", "tunableop_gfx950.csv",
+)
+if os.environ.get("KUBEAI_TUNABLEOP", "1") == "1" and os.path.exists(_TUNED):
+    os.environ.setdefault("PYTORCH_TUNABLEOP_ENABLED", "1")
+    os.environ.setdefault("PYTORCH_TUNABLEOP_TUNING", "0")
+    if "PYTORCH_TUNABLEOP_FILENAME" not in os.environ:
+        for dev in range(8):
+            dst = f"/tmp/kubeai_tunableop_{dev}.csv"
+            if not os.path.exists(dst):
+                shutil.copyfile(_TUNED, dst)
+        os.environ["PYTORCH_TUNABLEOP_FILENAME"] = "/tmp/kubeai_tunableop_%d.csv"
 
 import torch
 

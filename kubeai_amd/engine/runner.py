@@ -154,8 +154,35 @@ class ModelRunner:
                 ids.extend([ss.req.lora_id] * ss.chunk_len)
             lora_ids = torch.tensor(ids, dtype=torch.int32, device=self.device)
 
+        # ---- pad the token stream to a bucket size --------------------
+        # Keeps GEMM M-shapes on a small fixed set (TunableOp algo table +
+        # future hipGraph capture). Pad tokens form a throwaway prefill
+        # sequence over cache block 0: slot -1 (no KV write), row-local
+        # compute only (GEMMs/norms are row-independent), never sampled.
+        T = len(input_ids)
+        T_pad = _bucket(T)
+        if T_pad > T:
+            pad = T_pad - T
+            input_ids.extend([0] * pad)
+            slots.extend([-1] * pad)
+            # split padding into <=block_size mini-seqs so the throwaway
+            # causal attention stays O(pad * bs), not O(pad^2)
+            left = pad
+            while left > 0:
+                c = min(left, bs)
+                positions.extend(range(c))
+                pre_tables.append([0] * max_bt)
+                pre_lens.append(c)
+                qsl.append(qsl[-1] + c)
+                left -= c
+            if lora_ids is not None:
+                lora_ids = torch.cat(
+                    [lora_ids, torch.zeros(pad, dtype=torch.int32, device=self.device)]
+                )
+
         dev = self.device
         t32 = lambda x: torch.tensor(x, dtype=torch.int32, device=dev)
+        n_prefill = len(pre_lens)
         return ForwardBatch(
             lora_ids=lora_ids,
             input_ids=t32(input_ids),
@@ -164,8 +191,8 @@ class ModelRunner:
             n_decode=nd,
             decode_block_tables=t32(dec_tables) if dec_tables else None,
             decode_seq_lens=t32(dec_lens) if dec_lens else None,
-            n_prefill=len(out.prefill),
-            prefill_query_start_loc=t32(qsl) if out.prefill else None,
+            n_prefill=n_prefill,
+            prefill_query_start_loc=t32(qsl) if n_prefill else None,
             prefill_seq_lens=t32(pre_lens) if pre_lens else None,
             prefill_block_tables=t32(pre_tables) if pre_tables else None,
             logits_indices=torch.tensor(logit_idx, dtype=torch.int64, device=dev),
@@ -224,6 +251,18 @@ class ModelRunner:
                 tokens = ops.gumbel_sample(logits.contiguous(), t_t, seeds, step)
         tokens = tokens.cpu().tolist()
         return {r.request_id: int(t) for r, t in zip(sample_reqs, tokens)}
+
+
+_BUCKETS = [8, 16, 24, 32, 40, 48, 64, 80, 96, 128, 160, 192, 256, 320,
+            384, 512, 640, 768, 1024, 1280, 1536, 2048, 3072, 4096, 6144,
+            8192, 12288, 16384]
+
+
+def _bucket(n: int) -> int:
+    for b in _BUCKETS:
+        if n <= b:
+            return b
+    return n
 
 
 def _seed_of(request_id: str) -> int:

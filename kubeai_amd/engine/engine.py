@@ -31,6 +31,7 @@ class EngineConfig:
     max_num_batched_tokens: int = 8192
     max_model_len: int = 8192
     enable_prefix_caching: bool = True
+    enable_graphs: bool = True  # hipGraph capture for pure-decode steps
     seed: int = 0
 
     def resolve_arch(self) -> ModelArchConfig:
@@ -70,6 +71,8 @@ class LLMEngine:
             gpu_memory_utilization=cfg.gpu_memory_utilization,
             seed=cfg.seed,
             tp_group=tp_group,
+            max_model_len=cfg.max_model_len,
+            enable_graphs=cfg.enable_graphs,
         )
         self.block_manager = BlockManager(self.runner.num_blocks, cfg.block_size)
         self.scheduler = Scheduler(

@@ -7,6 +7,7 @@ kernels.
 """
 from __future__ import annotations
 
+import os
 from typing import Optional
 
 import torch
@@ -20,6 +21,68 @@ from .kvcache import BlockManager
 from .scheduler import SchedulerOutput
 
 
+class _DecodeGraph:
+    """hipGraph-captured pure-decode step for one padded batch size.
+
+    Replays the whole token-packed forward (norms, GEMMs, rope, cache
+    write, flash-decoding attention, logits GEMM) with zero launch
+    overhead; only the small input buffers are refreshed per step.
+    Sampling stays outside (its `step` scalar changes every call).
+    """
+
+    def __init__(self, runner: "ModelRunner", b_pad: int, bt_max: int):
+        dev = runner.device
+        self.b_pad = b_pad
+        self.bt_max = bt_max
+        self.input_ids = torch.zeros(b_pad, dtype=torch.int32, device=dev)
+        self.positions = torch.zeros(b_pad, dtype=torch.int32, device=dev)
+        self.slots = torch.full((b_pad,), -1, dtype=torch.int64, device=dev)
+        self.block_tables = torch.zeros(
+            (b_pad, bt_max), dtype=torch.int32, device=dev
+        )
+        self.seq_lens = torch.ones(b_pad, dtype=torch.int32, device=dev)
+        logits_idx = torch.arange(b_pad, dtype=torch.int64, device=dev)
+        self.fb = ForwardBatch(
+            input_ids=self.input_ids,
+            positions=self.positions,
+            slot_mapping=self.slots,
+            n_decode=b_pad,
+            decode_block_tables=self.block_tables,
+            decode_seq_lens=self.seq_lens,
+            n_prefill=0,
+            prefill_query_start_loc=None,
+            prefill_seq_lens=None,
+            prefill_block_tables=None,
+            logits_indices=logits_idx,
+        )
+        # warmup outside capture (allocator, hipBLASLt workspace)
+        torch.cuda.synchronize()
+        for _ in range(2):
+            h = runner.model(self.fb)
+            runner.model.compute_logits(h[logits_idx])
+        torch.cuda.synchronize()
+        self.graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.graph):
+            h = runner.model(self.fb)
+            self.logits = runner.model.compute_logits(h[logits_idx])
+
+    def run(self, input_ids, positions, slots, tables, seq_lens) -> torch.Tensor:
+        b = len(input_ids)
+        self.input_ids[:b].copy_(input_ids, non_blocking=True)
+        self.positions[:b].copy_(positions, non_blocking=True)
+        self.slots[:b].copy_(slots, non_blocking=True)
+        if b < self.b_pad:
+            self.slots[b:].fill_(-1)
+            self.seq_lens[b:].fill_(1)
+            self.block_tables[b:].fill_(0)
+        self.block_tables[:b, : tables.shape[1]].copy_(tables, non_blocking=True)
+        if tables.shape[1] < self.bt_max:
+            self.block_tables[:b, tables.shape[1] :].fill_(0)
+        self.seq_lens[:b].copy_(seq_lens, non_blocking=True)
+        self.graph.replay()
+        return self.logits[:b]
+
+
 class ModelRunner:
     def __init__(
         self,
@@ -31,11 +94,22 @@ class ModelRunner:
         gpu_memory_utilization: float = 0.90,
         seed: int = 0,
         tp_group=None,
+        max_model_len: int = 8192,
+        enable_graphs: bool = True,
     ):
         self.arch = arch
         self.device = torch.device(device)
         self.dtype = dtype
         self.block_size = block_size
+        self.max_model_len = max_model_len
+        self.tp_group = tp_group
+        self.enable_graphs = (
+            enable_graphs
+            and self.device.type == "cuda"
+            and tp_group is None  # RCCL-in-graph: validate before enabling
+            and os.environ.get("KUBEAI_GRAPHS", "1") == "1"
+        )
+        self._graphs: dict[int, _DecodeGraph] = {}
         torch.manual_seed(seed)
         if tp_group is not None:
             from kubeai_amd.parallel.tp import TPLlamaForCausalLM
@@ -203,19 +277,73 @@ class ModelRunner:
         return table + [0] * (n - len(table))
 
     # ------------------------------------------------------------------
+    def _execute_graph(self, out: SchedulerOutput):
+        """Pure-decode step through a captured hipGraph (bucketed batch)."""
+        b = len(out.decode)
+        b_pad = _bucket(b)
+        bt_max = (self.max_model_len + self.block_size - 1) // self.block_size
+        g = self._graphs.get(b_pad)
+        if g is None:
+            try:
+                g = _DecodeGraph(self, b_pad, bt_max)
+            except Exception:
+                import traceback
+
+                traceback.print_exc()
+                self.enable_graphs = False
+                return None
+            self._graphs[b_pad] = g
+        bs = self.block_size
+        ids, pos, slots, lens = [], [], [], []
+        import numpy as np
+
+        tables = np.zeros((b, g.bt_max), dtype=np.int32)
+        for i, ss in enumerate(out.decode):
+            req = ss.req
+            t = ss.chunk_start
+            ids.append(req.tokens[t])
+            pos.append(t)
+            slots.append(req.block_table[t // bs] * bs + t % bs)
+            lens.append(t + 1)
+            bt = req.block_table
+            tables[i, : len(bt)] = bt
+        return g.run(
+            torch.tensor(ids, dtype=torch.int32),
+            torch.tensor(pos, dtype=torch.int32),
+            torch.tensor(slots, dtype=torch.int64),
+            torch.from_numpy(tables),
+            torch.tensor(lens, dtype=torch.int32),
+        )
+
     @torch.inference_mode()
     def execute(self, out: SchedulerOutput, step: int) -> dict[str, int]:
         """Run one forward + sampling; returns request_id -> sampled token."""
-        fb = self.build_batch(out)
-        hidden = self.model(fb)  # [T, H]
-        if fb.logits_indices.numel() == 0:
-            return {}
-        logits = self.model.compute_logits(hidden[fb.logits_indices])  # [S, V] f32
+        logits = None
+        if (
+            self.enable_graphs
+            and not out.prefill
+            and out.decode
+            and not (
+                getattr(self, "lora_manager", None) is not None
+                and self.lora_manager.active
+            )
+        ):
+            logits = self._execute_graph(out)
+        if logits is None:
+            fb = self.build_batch(out)
+            hidden = self.model(fb)  # [T, H]
+            if fb.logits_indices.numel() == 0:
+                return {}
+            logits = self.model.compute_logits(hidden[fb.logits_indices])  # [S, V]
 
         # assemble sampling params in the same order as logits rows
         sample_reqs = [ss.req for ss in out.decode] + [
             ss.req for ss in out.prefill if ss.samples
         ]
+        if logits.shape[0] != len(sample_reqs):
+            raise RuntimeError(
+                f"logits rows {logits.shape[0]} != sample reqs {len(sample_reqs)}"
+            )
         temps = [r.params.temperature for r in sample_reqs]
         if all(t <= 0.0 for t in temps):
             tokens = ops.greedy_sample(logits.contiguous())

@@ -70,6 +70,7 @@ def get_args():
     p.add_argument("--no-prefix-cache", action="store_true")
     p.add_argument("--gpu-mem-util", type=float, default=0.90)
     p.add_argument("--max-model-len", type=int, default=8192)
+    p.add_argument("--prefill-interval", type=int, default=1)
     return p.parse_args()
 
 
@@ -169,6 +170,7 @@ def main():
             max_num_batched_tokens=8192,
             max_num_seqs=max(args.vus * 2, 64),
             enable_prefix_caching=not args.no_prefix_cache,
+            prefill_interval=args.prefill_interval,
             seed=replica_seed,
         ),
         tp_group=tp_group,

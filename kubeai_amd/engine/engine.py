@@ -32,6 +32,7 @@ class EngineConfig:
     max_model_len: int = 8192
     enable_prefix_caching: bool = True
     enable_graphs: bool = True  # hipGraph capture for pure-decode steps
+    prefill_interval: int = 1  # >1: batch prefills onto every Nth step
     seed: int = 0
 
     def resolve_arch(self) -> ModelArchConfig:
@@ -81,6 +82,7 @@ class LLMEngine:
             max_num_batched_tokens=cfg.max_num_batched_tokens,
             max_model_len=cfg.max_model_len,
             enable_prefix_caching=cfg.enable_prefix_caching,
+            prefill_interval=cfg.prefill_interval,
         )
         self.step_count = 0
 
@@ -158,7 +160,8 @@ class LLMEngine:
             )
             fb = self.runner.build_batch(so)
             hidden = self.runner.model(fb)
-            vec = hidden.float().mean(dim=0)
+            # exclude bucket-padding rows from the pool
+            vec = hidden[: len(toks)].float().mean(dim=0)
             vec = vec / vec.norm().clamp_min(1e-12)
             out.append(vec.cpu().tolist())
             self.block_manager.free(table)

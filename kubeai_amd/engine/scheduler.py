@@ -127,12 +127,18 @@ class Scheduler:
         max_num_batched_tokens: int = 8192,
         max_model_len: int = 8192,
         enable_prefix_caching: bool = True,
+        prefill_interval: int = 1,
     ):
         self.bm = block_manager
         self.max_num_seqs = max_num_seqs
         self.max_num_batched_tokens = max_num_batched_tokens
         self.max_model_len = max_model_len
         self.enable_prefix_caching = enable_prefix_caching
+        # prefill_interval > 1: batch prefill work onto every Nth step so
+        # the steps in between are pure decode (hipGraph-replayable). Waiting
+        # prefills still run immediately when nothing is decoding.
+        self.prefill_interval = max(1, prefill_interval)
+        self._step_idx = 0
         self.waiting: deque[Request] = deque()
         self.running: list[Request] = []
         self._aborted: set[str] = set()
@@ -163,10 +169,20 @@ class Scheduler:
     # ----------------------------------------------------------------
     def schedule(self) -> SchedulerOutput:
         self._apply_aborts()
+        self._step_idx += 1
         budget = self.max_num_batched_tokens
         decode: list[ScheduledSeq] = []
         prefill: list[ScheduledSeq] = []
         preempted: list[Request] = []
+        have_decodes = any(
+            not r.in_prefill and r.status == RequestStatus.RUNNING
+            for r in self.running
+        )
+        defer_prefill = (
+            self.prefill_interval > 1
+            and have_decodes
+            and self._step_idx % self.prefill_interval != 0
+        )
 
         # 1) running sequences: decodes first (latency-critical; each needs 1
         #    token), then mid-prefill chunks. Iterate copies: preemption
@@ -182,7 +198,7 @@ class Scheduler:
             self._scheduled_ids.add(req.request_id)
             budget -= 1
         for req in [r for r in list(self.running) if r.in_prefill]:
-            if budget <= 0 or req.status != RequestStatus.RUNNING:
+            if budget <= 0 or req.status != RequestStatus.RUNNING or defer_prefill:
                 continue
             chunk = min(req.num_prompt_tokens - req.num_computed, budget)
             if not self._ensure_blocks(req, req.num_computed + chunk, preempted):
@@ -193,7 +209,8 @@ class Scheduler:
 
         # 2) admit waiting requests
         while (
-            self.waiting
+            not defer_prefill
+            and self.waiting
             and budget > 0
             and len(self.running) < self.max_num_seqs
         ):

@@ -70,7 +70,13 @@ def get_args():
     p.add_argument("--no-prefix-cache", action="store_true")
     p.add_argument("--gpu-mem-util", type=float, default=0.90)
     p.add_argument("--max-model-len", type=int, default=8192)
-    p.add_argument("--prefill-interval", type=int, default=1)
+    p.add_argument("--prefill-interval", type=int, default=4)
+    p.add_argument(
+        "--quant",
+        default=None,
+        choices=[None, "fp8", "none"],
+        help="fp8 = W8A8 serving (the BASELINE config is Llama-3.1-8B-FP8)",
+    )
     return p.parse_args()
 
 
@@ -171,6 +177,7 @@ def main():
             max_num_seqs=max(args.vus * 2, 64),
             enable_prefix_caching=not args.no_prefix_cache,
             prefill_interval=args.prefill_interval,
+            quantization=args.quant if args.quant not in (None, "none") and use_cuda else None,
             seed=replica_seed,
         ),
         tp_group=tp_group,
@@ -292,7 +299,7 @@ def main():
                     "higher_is_better": True,
                     "scaling": "weak",
                     "vs_baseline": round(value / baseline, 3),
-                    "dtype": "bf16",
+                    "dtype": "fp8" if (args.quant == "fp8" and use_cuda) else "bf16",
                     "data": "synthetic",
                     "config": {
                         "model": model,

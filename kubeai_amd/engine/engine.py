@@ -31,6 +31,7 @@ class EngineConfig:
     max_num_batched_tokens: int = 8192
     max_model_len: int = 8192
     enable_prefix_caching: bool = True
+    quantization: Optional[str] = None  # "fp8" => W8A8 dynamic (quant.py)
     enable_graphs: bool = True  # hipGraph capture for pure-decode steps
     prefill_interval: int = 1  # >1: batch prefills onto every Nth step
     seed: int = 0
@@ -74,6 +75,7 @@ class LLMEngine:
             tp_group=tp_group,
             max_model_len=cfg.max_model_len,
             enable_graphs=cfg.enable_graphs,
+            quantization=cfg.quantization,
         )
         self.block_manager = BlockManager(self.runner.num_blocks, cfg.block_size)
         self.scheduler = Scheduler(

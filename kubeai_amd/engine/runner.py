@@ -96,6 +96,7 @@ class ModelRunner:
         tp_group=None,
         max_model_len: int = 8192,
         enable_graphs: bool = True,
+        quantization: Optional[str] = None,
     ):
         self.arch = arch
         self.device = torch.device(device)
@@ -122,6 +123,14 @@ class ModelRunner:
             self.model = LlamaForCausalLM(arch, device=self.device, dtype=dtype)
             self.n_kv_local = arch.num_key_value_heads
         self.model.eval()
+        self.quantization = quantization
+        if quantization == "fp8":
+            if self.device.type != "cuda":
+                raise ValueError("fp8 quantization requires a GPU")
+            from .quant import convert_to_fp8
+
+            n = convert_to_fp8(self.model)
+            assert n > 0, "no linear layers converted to fp8"
 
         if num_gpu_blocks is None:
             num_gpu_blocks = self._profile_num_blocks(gpu_memory_utilization)

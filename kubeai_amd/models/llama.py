@@ -46,7 +46,7 @@ class Attention(nn.Module):
         cos_sin: torch.Tensor,
     ) -> torch.Tensor:
         T = x.shape[0]
-        qkv = F.linear(x, self.qkv_proj.weight)
+        qkv = self.qkv_proj(x)
         lm = getattr(self, "_lora_manager", None)
         if lm is not None and lm.active and fb.lora_ids is not None:
             lm.apply(self.layer_idx, "qkv", x, qkv, fb.lora_ids)
@@ -88,7 +88,7 @@ class Attention(nn.Module):
                 out=out[nd:],
             )
         attn_flat = out.view(T, -1)
-        result = F.linear(attn_flat, self.o_proj.weight)
+        result = self.o_proj(attn_flat)
         if lm is not None and lm.active and fb.lora_ids is not None:
             lm.apply(self.layer_idx, "o", attn_flat, result, fb.lora_ids)
         return result
@@ -103,8 +103,7 @@ class MLP(nn.Module):
         self.down_proj = nn.Linear(cfg.intermediate_size, cfg.hidden_size, bias=False)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        return F.linear(ops.silu_and_mul(F.linear(x, self.gate_up_proj.weight)),
-                        self.down_proj.weight)
+        return self.down_proj(ops.silu_and_mul(self.gate_up_proj(x)))
 
 
 class MoEMLP(nn.Module):
@@ -201,12 +200,9 @@ class LlamaForCausalLM(nn.Module):
 
     @torch.inference_mode()
     def compute_logits(self, hidden: torch.Tensor) -> torch.Tensor:
-        w = (
-            self.embed_tokens.weight
-            if self.lm_head is None
-            else self.lm_head.weight
-        )
-        return F.linear(hidden, w).float()
+        if self.lm_head is None:
+            return F.linear(hidden, self.embed_tokens.weight).float()
+        return self.lm_head(hidden).float()
 
     def bind_kv_caches(self, kv_caches: list[tuple[torch.Tensor, torch.Tensor]]):
         self.kv_caches = kv_caches

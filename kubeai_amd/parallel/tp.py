@@ -93,7 +93,7 @@ class TPAttention(nn.Module):
 
     def forward(self, x, fb: ForwardBatch, kv_cache, cos_sin):
         T = x.shape[0]
-        qkv = F.linear(x, self.qkv_proj.weight)
+        qkv = self.qkv_proj(x)
         q, k, v = qkv.split(
             [self.n_q * self.hd, self.n_kv * self.hd, self.n_kv * self.hd], dim=-1
         )
@@ -120,7 +120,7 @@ class TPAttention(nn.Module):
                 fb.prefill_seq_lens, self.scale,
                 out=out[nd:],
             )
-        res = F.linear(out.view(T, -1), self.o_proj.weight)
+        res = self.o_proj(out.view(T, -1))
         return self.tp.all_reduce(res)
 
 
@@ -145,8 +145,8 @@ class TPMLP(nn.Module):
         )
 
     def forward(self, x):
-        act = ops.silu_and_mul(F.linear(x, self.gate_up_proj.weight))
-        return self.tp.all_reduce(F.linear(act, self.down_proj.weight))
+        act = ops.silu_and_mul(self.gate_up_proj(x))
+        return self.tp.all_reduce(self.down_proj(act))
 
 
 class TPDecoderLayer(nn.Module):
@@ -226,8 +226,9 @@ class TPLlamaForCausalLM(nn.Module):
 
     @torch.inference_mode()
     def compute_logits(self, hidden: torch.Tensor) -> torch.Tensor:
-        w = self.embed_tokens.weight if self.lm_head is None else self.lm_head.weight
-        return F.linear(hidden, w).float()
+        if self.lm_head is None:
+            return F.linear(hidden, self.embed_tokens.weight).float()
+        return self.lm_head(hidden).float()
 
     def bind_kv_caches(self, kv_caches):
         self.kv_caches = kv_caches

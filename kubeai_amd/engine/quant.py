@@ -27,8 +27,9 @@ class Fp8Linear(nn.Module):
         w_amax = w.abs().amax(dim=1, keepdim=True).clamp_min(1e-6)  # [out,1]
         w_scale = w_amax / FP8_MAX
         wq = (w / w_scale).clamp(-FP8_MAX, FP8_MAX).to(FP8_DTYPE)
-        # column-major for mat2 of scaled_mm: store transposed-contiguous
-        self.register_buffer("weight_fp8_t", wq.t().contiguous())
+        # scaled_mm wants mat2 column-major: keep [out, in] row-major and
+        # pass .t() (a column-major view) at call time
+        self.register_buffer("weight_fp8", wq.contiguous())
         self.register_buffer("weight_scale", w_scale.reshape(1, -1).contiguous())
         self.out_features, self.in_features = weight_bf16.shape
 
@@ -36,7 +37,7 @@ class Fp8Linear(nn.Module):
     def weight(self) -> torch.Tensor:
         """bf16 view for code paths that read .weight (LoRA base, tests)."""
         return (
-            self.weight_fp8_t.t().float() * self.weight_scale.reshape(-1, 1)
+            self.weight_fp8.float() * self.weight_scale.reshape(-1, 1)
         ).to(torch.bfloat16)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
@@ -45,7 +46,7 @@ class Fp8Linear(nn.Module):
         xq = (x.float() / a_scale).clamp(-FP8_MAX, FP8_MAX).to(FP8_DTYPE)
         return torch._scaled_mm(
             xq,
-            self.weight_fp8_t,
+            self.weight_fp8.t(),
             scale_a=a_scale,
             scale_b=self.weight_scale,
             out_dtype=torch.bfloat16,

@@ -167,18 +167,25 @@ class LlamaForCausalLM(nn.Module):
     def __init__(self, cfg: ModelArchConfig, device=None, dtype=torch.bfloat16):
         super().__init__()
         self.cfg = cfg
-        factory = {"device": device, "dtype": dtype}
-        with torch.device(device if device is not None else "cpu"):
-            self.embed_tokens = nn.Embedding(cfg.vocab_size, cfg.hidden_size)
-            self.layers = nn.ModuleList(
-                [DecoderLayer(cfg, i) for i in range(cfg.num_hidden_layers)]
-            )
-            self.norm = nn.Parameter(torch.ones(cfg.hidden_size))
-            if cfg.tie_word_embeddings:
-                self.lm_head = None
-            else:
-                self.lm_head = nn.Linear(cfg.hidden_size, cfg.vocab_size, bias=False)
-        self.to(**factory)
+        # build weights directly in the target dtype/device: a 70B model
+        # constructed fp32-first would transiently need ~280 GB
+        prev_dtype = torch.get_default_dtype()
+        torch.set_default_dtype(dtype)
+        try:
+            with torch.device(device if device is not None else "cpu"):
+                self.embed_tokens = nn.Embedding(cfg.vocab_size, cfg.hidden_size)
+                self.layers = nn.ModuleList(
+                    [DecoderLayer(cfg, i) for i in range(cfg.num_hidden_layers)]
+                )
+                self.norm = nn.Parameter(torch.ones(cfg.hidden_size))
+                if cfg.tie_word_embeddings:
+                    self.lm_head = None
+                else:
+                    self.lm_head = nn.Linear(
+                        cfg.hidden_size, cfg.vocab_size, bias=False
+                    )
+        finally:
+            torch.set_default_dtype(prev_dtype)
         from kubeai_amd.ops import ref as ops_ref
 
         self.register_buffer(

@@ -45,6 +45,8 @@ def spec_hash(model: Model, gpus_per_replica: int) -> str:
             ",".join(f"{k}={v}" for k, v in sorted(s.env.items())),
             str(gpus_per_replica),
             s.cache_profile,
+            ",".join(f"{f.path}#{hashlib.sha256(f.content.encode()).hexdigest()[:8]}"
+                     for f in s.files),
         ]
     )
     return format(fnv1a_32(key.encode()), "08x")

@@ -112,9 +112,19 @@ class LocalProcessRuntime:
         rep.gpu_ids = ids
         rep.state = ReplicaState.STARTING
         env = dict(os.environ)
+        env.update(model.spec.env)  # reference: model_types.go:86-90
         if ids:
             env["HIP_VISIBLE_DEVICES"] = ",".join(map(str, ids))
             env["CUDA_VISIBLE_DEVICES"] = env["HIP_VISIBLE_DEVICES"]
+        # spec.files -> per-replica dir (reference: files.go ConfigMap mounts)
+        if model.spec.files:
+            files_dir = os.path.join("/tmp/kubeai-replica-files", name)
+            os.makedirs(files_dir, exist_ok=True)
+            for f in model.spec.files:
+                dst = os.path.join(files_dir, f.path.lstrip("/").replace("/", "_"))
+                with open(dst, "w") as fh:
+                    fh.write(f.content)
+            env["KUBEAI_MODEL_FILES_DIR"] = files_dir
         model_ref = _model_source_path(model)
         cmd = [
             sys.executable,

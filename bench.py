@@ -42,7 +42,10 @@ if os.environ.get("KUBEAI_TUNABLEOP", "1") == "1" and os.path.exists(_TUNED):
         for dev in range(8):
             dst = f"/tmp/kubeai_tunableop_{dev}.csv"
             if not os.path.exists(dst):
-                shutil.copyfile(_TUNED, dst)
+                # atomic publish: concurrent ranks must never see a partial copy
+                tmp = f"{dst}.{os.getpid()}.tmp"
+                shutil.copyfile(_TUNED, tmp)
+                os.replace(tmp, dst)
         os.environ["PYTORCH_TUNABLEOP_FILENAME"] = "/tmp/kubeai_tunableop_%d.csv"
 
 import torch

@@ -1,0 +1,212 @@
+"""Small-unit coverage: xxhash vectors, moving average, apiutils, config,
+autoscaler HA aggregation + state persistence (reference analogs:
+autoscaling_ha_test.go, autoscaler_state_test.go, movingaverage tests)."""
+import asyncio
+import json
+import os
+import tempfile
+
+import pytest
+
+from kubeai_amd.controlplane.apiutils import (
+    APIError,
+    extract_prefix,
+    first_n_chars,
+    parse_request,
+    split_model_adapter,
+)
+from kubeai_amd.controlplane.autoscaler import Autoscaler
+from kubeai_amd.controlplane.config import load_config
+from kubeai_amd.controlplane.crd import Model, ModelSpec
+from kubeai_amd.controlplane.modelclient import ModelClient
+from kubeai_amd.controlplane.movingaverage import SimpleMovingAverage
+from kubeai_amd.controlplane.store import Store
+from kubeai_amd.utils.xxhash64 import xxh64
+
+
+# ---------------------------------------------------------------- xxhash
+def test_xxh64_vectors():
+    assert xxh64(b"") == 0xEF46DB3751D8E999
+    assert xxh64(b"a") == 0xD24EC4F1A98C6E5B
+    assert xxh64(b"abc") == 0x44BC2CF5AD770999
+    assert xxh64(b"xxhash", seed=20141025) == 0xB559B98D844E0635
+    # >32-byte path
+    assert xxh64(b"0123456789abcdefghijklmnopqrstuvwxyz" * 4) == 0x5D3DC3370A8C2A38
+
+
+# ---------------------------------------------------------------- moving avg
+def test_moving_average_decays_to_zero():
+    ma = SimpleMovingAverage(4)
+    assert ma.next(8) == 2.0
+    assert ma.next(8) == 4.0
+    ma.next(8), ma.next(8)
+    assert ma.calculate() == 8.0
+    for _ in range(4):
+        ma.next(0)
+    assert ma.calculate() == 0.0  # enables scale-to-zero
+
+
+def test_moving_average_load_history():
+    ma = SimpleMovingAverage(3)
+    ma.load([3.0, 6.0, 9.0])
+    assert ma.calculate() == 6.0
+
+
+# ---------------------------------------------------------------- apiutils
+def test_split_model_adapter():
+    assert split_model_adapter("m1") == ("m1", "")
+    assert split_model_adapter("m1_lora") == ("m1", "lora")
+    assert split_model_adapter("m1_a_b") == ("m1", "a_b")
+
+
+def test_prefix_extraction_rune_safe():
+    assert first_n_chars("héllo wörld", 6) == "héllo "
+    body = {"messages": [{"role": "system", "content": "s"},
+                          {"role": "user", "content": "ünïcode prompt here"}]}
+    assert extract_prefix(body, "/v1/chat/completions", 7) == "ünïcode"
+    assert extract_prefix({"prompt": "plain text"}, "/v1/completions", 5) == "plain"
+    assert extract_prefix({"prompt": ["first", "second"]}, "/v1/completions", 5) == "first"
+    assert extract_prefix({}, "/v1/completions", 5) is None
+
+
+def test_parse_request_rewrites_adapter_model():
+    store = Store()
+    from kubeai_amd.controlplane.crd import AdapterSpec
+
+    store.apply_model(
+        Model(name="m", spec=ModelSpec(url="hf://x/y",
+                                       adapters=[AdapterSpec(name="ad", url="hf://a/b")]))
+    )
+    mc = ModelClient(store)
+    pr = parse_request({"model": "m_ad", "prompt": "p"}, "/v1/completions",
+                       mc.lookup_model)
+    assert pr.model == "m" and pr.adapter == "ad"
+    assert pr.body["model"] == "ad"  # engine selects LoRA by model name
+    with pytest.raises(APIError):
+        parse_request({"model": "m_ghost", "prompt": "p"}, "/v1/completions",
+                      mc.lookup_model)
+    with pytest.raises(APIError):
+        parse_request({"prompt": "p"}, "/v1/completions", mc.lookup_model)
+
+
+# ---------------------------------------------------------------- config
+def test_config_yaml_roundtrip(tmp_path):
+    p = tmp_path / "config.yaml"
+    p.write_text(
+        """
+resourceProfiles:
+  amd-gpu-mi355x: 1
+  big: 8
+modelAutoscaling:
+  interval: 5
+  timeWindow: 300
+messaging:
+  streams:
+    - requestsURL: mem://req
+      responsesURL: mem://resp
+      maxHandlers: 4
+"""
+    )
+    cfg = load_config(str(p))
+    assert cfg.resource_profiles["big"] == 8
+    assert cfg.autoscaling.interval_seconds == 5
+    assert cfg.messaging[0].max_handlers == 4
+
+
+def test_config_validation():
+    cfg = load_config(None)
+    cfg.autoscaling.interval_seconds = -1
+    with pytest.raises(ValueError):
+        cfg.validate()
+
+
+# ---------------------------------------------------------------- autoscaler
+class FakeMetricsServer:
+    """Three control-plane replicas reporting different active counts —
+    the HA sum test (reference: autoscaling_ha_test.go)."""
+
+
+def test_autoscaler_ha_sums_self_metrics():
+    async def body():
+        import uvicorn
+        from starlette.applications import Starlette
+        from starlette.responses import PlainTextResponse
+        from starlette.routing import Route
+        import socket
+
+        def free_port():
+            with socket.socket() as s:
+                s.bind(("127.0.0.1", 0))
+                return s.getsockname()[1]
+
+        counts = [3, 5, 2]
+        servers, addrs = [], []
+        for c in counts:
+            port = free_port()
+
+            def app_for(c):
+                async def metrics(request):
+                    return PlainTextResponse(
+                        f'kubeai_inference_requests_active{{model="m"}} {c}\n'
+                    )
+
+                return Starlette(routes=[Route("/metrics", metrics)])
+
+            srv = uvicorn.Server(
+                uvicorn.Config(app_for(c), host="127.0.0.1", port=port,
+                               log_level="error")
+            )
+            task = asyncio.create_task(srv.serve())
+            while not srv.started:
+                await asyncio.sleep(0.01)
+            servers.append((srv, task))
+            addrs.append(f"127.0.0.1:{port}")
+
+        store = Store()
+        store.apply_model(
+            Model(name="m", spec=ModelSpec(url="hf://x/y", min_replicas=0,
+                                           max_replicas=10, target_requests=2))
+        )
+        mc = ModelClient(store)
+        a = Autoscaler(
+            store, mc, interval=0.01, time_window=0.01,
+            self_metric_addrs=addrs, scrape_engine_queues=False,
+        )
+        await a.tick()
+        # sum = 10, target 2 -> ceil(10/2) = 5 replicas
+        assert store.get_model("m").spec.replicas == 5
+        await a.stop()
+        for srv, task in servers:
+            srv.should_exit = True
+            await asyncio.wait_for(task, timeout=5)
+
+    asyncio.run(body())
+
+
+def test_autoscaler_state_persistence(tmp_path):
+    async def body():
+        state = str(tmp_path / "state.json")
+        store = Store()
+        store.apply_model(
+            Model(name="m", spec=ModelSpec(url="hf://x/y", max_replicas=10))
+        )
+        mc = ModelClient(store)
+        a = Autoscaler(store, mc, interval=1.0, time_window=4.0,
+                       state_path=state, scrape_engine_queues=False)
+        from kubeai_amd.controlplane import metrics
+
+        metrics.INFERENCE_REQUESTS_ACTIVE.labels("m").inc(4)
+        try:
+            await a.tick()
+            assert os.path.exists(state)
+            data = json.load(open(state))
+            assert data["m"][0] == 4.0
+            # a fresh autoscaler preloads the history (restart continuity)
+            b = Autoscaler(store, mc, interval=1.0, time_window=4.0,
+                           state_path=state, scrape_engine_queues=False)
+            assert b.averages["m"].history()[0] == 4.0
+        finally:
+            metrics.INFERENCE_REQUESTS_ACTIVE.labels("m").dec(4)
+            await a.stop()
+
+    asyncio.run(body())

@@ -46,11 +46,23 @@ M_TTFT = prom.Histogram(
 
 
 class EngineServer:
-    """Engine + stepping thread + async request plumbing."""
+    """Engine + stepping thread + async request plumbing.
 
-    def __init__(self, cfg: EngineConfig, served_model_name: str):
+    Tensor-parallel serving (--tensor-parallel-size N): rank 0 owns the
+    HTTP surface and broadcasts each iteration's (new requests, aborts)
+    to N-1 worker processes over torch.distributed; every rank then runs
+    the identical scheduler step in lockstep (all-reduced activations are
+    rank-identical, so sampling and scheduling never diverge — the same
+    invariant the TP bench path relies on, kubeai_amd/parallel/tp.py).
+    """
+
+    def __init__(self, cfg: EngineConfig, served_model_name: str, tp_size: int = 1,
+                 tp_port: int = 0):
         self.cfg = cfg
         self.served_model_name = served_model_name
+        self.tp_size = tp_size
+        self.tp_port = tp_port
+        self._dist = None
         self.engine: Optional[LLMEngine] = None
         self.tokenizer = None
         self._submit: "queue.Queue" = queue.Queue()
@@ -71,7 +83,21 @@ class EngineServer:
 
     def _run(self) -> None:
         # heavy init inside the thread so /health can answer "starting"
-        self.engine = LLMEngine(self.cfg)
+        tp_group = None
+        if self.tp_size > 1 or os.environ.get("KUBEAI_FORCE_TP"):
+            import torch.distributed as dist
+
+            from kubeai_amd.parallel.tp import TPGroup
+
+            dist.init_process_group(
+                "nccl" if self.cfg.resolve_device().startswith("cuda") else "gloo",
+                init_method=f"tcp://127.0.0.1:{self.tp_port}",
+                rank=0,
+                world_size=self.tp_size,
+            )
+            self._dist = dist
+            tp_group = TPGroup()
+        self.engine = LLMEngine(self.cfg, tp_group=tp_group)
         arch = self.engine.arch
         self.tokenizer = load_tokenizer(
             self.cfg.model, arch.vocab_size, arch.bos_token_id, arch.eos_token_id
@@ -80,19 +106,29 @@ class EngineServer:
         label = self.served_model_name
         while not self._stop.is_set():
             worked = False
+            new_reqs, abort_ids = [], []
             while True:
                 try:
                     rid, toks, params, lora_id = self._submit.get_nowait()
                 except queue.Empty:
                     break
+                new_reqs.append((rid, toks, params, lora_id))
+            while True:
+                try:
+                    abort_ids.append(self._aborts.get_nowait())
+                except queue.Empty:
+                    break
+            if self._dist is not None:
+                # keep TP workers in lockstep: ship this iteration's intents
+                self._dist.broadcast_object_list(
+                    [{"new": new_reqs, "aborts": abort_ids, "shutdown": False}],
+                    src=0,
+                )
+            for rid, toks, params, lora_id in new_reqs:
                 self.engine.add_request(toks, params, request_id=rid, lora_id=lora_id)
                 M_PROMPT.labels(label).inc(len(toks))
                 worked = True
-            while True:
-                try:
-                    rid = self._aborts.get_nowait()
-                except queue.Empty:
-                    break
+            for rid in abort_ids:
                 self.engine.abort_request(rid)
             if self.engine.has_work():
                 outputs = self.engine.step()
@@ -110,6 +146,11 @@ class EngineServer:
             M_HIT.labels(label).set(s["prefix_cache_hit_rate"])
             if not worked:
                 time.sleep(0.002)
+        if self._dist is not None:
+            self._dist.broadcast_object_list(
+                [{"new": [], "aborts": [], "shutdown": True}], src=0
+            )
+            self._dist.destroy_process_group()
 
     # ------------------------------------------------------------- requests
     async def generate(self, token_ids: list[int], params: SamplingParams, lora_id: int = 0):
@@ -427,6 +468,39 @@ async def _stream_completion(server, toks, params, name, chat: bool, lora_id: in
 
 
 # --------------------------------------------------------------------------
+def _tp_worker_main(rank: int, world: int, port: int, cfg: EngineConfig) -> None:
+    """TP worker rank (no HTTP): lockstep scheduler driven by rank 0."""
+    import torch
+    import torch.distributed as dist
+
+    from kubeai_amd.parallel.tp import TPGroup
+
+    use_cuda = cfg.resolve_device().startswith("cuda")
+    if use_cuda:
+        torch.cuda.set_device(rank)
+        cfg = type(cfg)(**{**cfg.__dict__, "device": f"cuda:{rank}"})
+    dist.init_process_group(
+        "nccl" if use_cuda else "gloo",
+        init_method=f"tcp://127.0.0.1:{port}",
+        rank=rank,
+        world_size=world,
+    )
+    engine = LLMEngine(cfg, tp_group=TPGroup())
+    while True:
+        box = [None]
+        dist.broadcast_object_list(box, src=0)
+        msg = box[0]
+        if msg["shutdown"]:
+            break
+        for rid, toks, params, lora_id in msg["new"]:
+            engine.add_request(toks, params, request_id=rid, lora_id=lora_id)
+        for rid in msg["aborts"]:
+            engine.abort_request(rid)
+        if engine.has_work():
+            engine.step()
+    dist.destroy_process_group()
+
+
 def main():
     p = argparse.ArgumentParser()
     p.add_argument("--model", required=True, help="preset name or model dir")
@@ -451,7 +525,27 @@ def main():
         num_gpu_blocks=args.num_gpu_blocks,
     )
     served = args.served_model_name or os.path.basename(args.model.rstrip("/"))
-    server = EngineServer(cfg, served)
+    tp = args.tensor_parallel_size
+    tp_port = 0
+    workers = []
+    if tp > 1:
+        import socket
+
+        with socket.socket() as s:
+            s.bind(("127.0.0.1", 0))
+            tp_port = s.getsockname()[1]
+        import torch.multiprocessing as mp
+
+        ctx = mp.get_context("spawn")
+        for r in range(1, tp):
+            p = ctx.Process(
+                target=_tp_worker_main, args=(r, tp, tp_port, cfg), daemon=True
+            )
+            p.start()
+            workers.append(p)
+        if cfg.device in ("auto", "cuda"):
+            cfg = EngineConfig(**{**cfg.__dict__, "device": "cuda:0"})
+    server = EngineServer(cfg, served, tp_size=tp, tp_port=tp_port)
     server.start()
     app = build_app(server)
     import uvicorn

@@ -59,10 +59,13 @@ def test_adapter_filtering():
         g = await make_group(
             ["a:1", "b:1"], adapters={"a:1": ["lora1"], "b:1": []}
         )
+        dones = []
         for _ in range(5):
             addr, done = await g.get_best_addr("lora1", None, ll_spec())
             assert addr == "a:1"  # only a:1 has the adapter
-            # note: no done() -> load grows, still must pick a:1
+            dones.append(done)  # load grows while held, still must pick a:1
+        for d in dones:
+            d()
 
     run(body())
 

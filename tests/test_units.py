@@ -188,25 +188,25 @@ def test_autoscaler_state_persistence(tmp_path):
         state = str(tmp_path / "state.json")
         store = Store()
         store.apply_model(
-            Model(name="m", spec=ModelSpec(url="hf://x/y", max_replicas=10))
+            Model(name="m-state", spec=ModelSpec(url="hf://x/y", max_replicas=10))
         )
         mc = ModelClient(store)
         a = Autoscaler(store, mc, interval=1.0, time_window=4.0,
                        state_path=state, scrape_engine_queues=False)
         from kubeai_amd.controlplane import metrics
 
-        metrics.INFERENCE_REQUESTS_ACTIVE.labels("m").inc(4)
+        metrics.INFERENCE_REQUESTS_ACTIVE.labels("m-state").inc(4)
         try:
             await a.tick()
             assert os.path.exists(state)
             data = json.load(open(state))
-            assert data["m"][0] == 4.0
+            assert data["m-state"][0] == 4.0
             # a fresh autoscaler preloads the history (restart continuity)
             b = Autoscaler(store, mc, interval=1.0, time_window=4.0,
                            state_path=state, scrape_engine_queues=False)
-            assert b.averages["m"].history()[0] == 4.0
+            assert b.averages["m-state"].history()[0] == 4.0
         finally:
-            metrics.INFERENCE_REQUESTS_ACTIVE.labels("m").dec(4)
+            metrics.INFERENCE_REQUESTS_ACTIVE.labels("m-state").dec(4)
             await a.stop()
 
     asyncio.run(body())

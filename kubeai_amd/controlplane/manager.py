@@ -39,6 +39,7 @@ class Manager:
         self.model_client = ModelClient(
             self.store,
             required_consecutive_scale_downs=self.cfg.autoscaling.required_consecutive_scale_downs,
+            autoscaling_interval=self.cfg.autoscaling.interval_seconds,
         )
         self.lb = LoadBalancer(self.store)
         self.controller = ModelController(

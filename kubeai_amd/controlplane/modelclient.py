@@ -13,9 +13,15 @@ from .store import Store
 
 
 class ModelClient:
-    def __init__(self, store: Store, required_consecutive_scale_downs: int = 1):
+    def __init__(
+        self,
+        store: Store,
+        required_consecutive_scale_downs: int = 1,
+        autoscaling_interval: float = 10.0,
+    ):
         self.store = store
         self.required_consecutive_scale_downs = required_consecutive_scale_downs
+        self.autoscaling_interval = autoscaling_interval
         self._consecutive_scale_downs: dict[str, int] = {}
 
     def lookup_model(
@@ -48,10 +54,16 @@ class ModelClient:
         target = max(lo, min(replicas, hi))
         current = s.replicas or 0
         if target < current:
-            # scale-down hysteresis: require N consecutive ticks
+            # scale-down hysteresis: require N consecutive ticks; N derives
+            # from the per-model scaleDownDelaySeconds (reference:
+            # model_types.go:118-121 + config.RequiredConsecutiveScaleDowns)
+            per_model = max(
+                1, int(s.scale_down_delay_seconds / max(self.autoscaling_interval, 1e-9))
+            )
+            required = max(self.required_consecutive_scale_downs, per_model)
             n = self._consecutive_scale_downs.get(name, 0) + 1
             self._consecutive_scale_downs[name] = n
-            if n < self.required_consecutive_scale_downs:
+            if n < required:
                 return
         else:
             self._consecutive_scale_downs[name] = 0

@@ -156,6 +156,7 @@ async def wait_for(cond, timeout=5.0, interval=0.01):
 
 
 def text_gen_model(name="m1", **kw):
+    kw.setdefault("scale_down_delay_seconds", 0)  # fast tests
     spec = ModelSpec(url="hf://meta/llama-tiny", min_replicas=0, max_replicas=4, **kw)
     return Model(name=name, spec=spec)
 

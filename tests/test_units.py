@@ -210,3 +210,23 @@ def test_autoscaler_state_persistence(tmp_path):
             await a.stop()
 
     asyncio.run(body())
+
+
+def test_scale_down_delay_hysteresis():
+    store = Store()
+    store.apply_model(
+        Model(name="hyst", spec=ModelSpec(url="hf://x/y", min_replicas=0,
+                                          max_replicas=5,
+                                          scale_down_delay_seconds=30))
+    )
+    store.scale_model("hyst", 3)
+    mc = ModelClient(store, autoscaling_interval=10.0)  # -> 3 ticks required
+    mc.scale("hyst", 1)
+    assert store.get_model("hyst").spec.replicas == 3  # tick 1: held
+    mc.scale("hyst", 1)
+    assert store.get_model("hyst").spec.replicas == 3  # tick 2: held
+    mc.scale("hyst", 1)
+    assert store.get_model("hyst").spec.replicas == 1  # tick 3: applied
+    # scale-up is never delayed
+    mc.scale("hyst", 4)
+    assert store.get_model("hyst").spec.replicas == 4

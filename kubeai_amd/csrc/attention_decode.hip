@@ -10,11 +10,13 @@
 //    through LDS (m*, l*, o* combine). This keeps the chip busy at small
 //    decode batches (B*n_kv*4 waves in flight).
 //  - each wave stages its 16-token KV tile into its own LDS buffer with
-//    ushort8 (16 B) vector loads; score/PV reads from LDS are 4 B/lane,
-//    2-way bank aliased (free on CDNA4, guide §6 G4).
-//  - lane d-ownership: lane l owns head-dim elements {2l, 2l+1}; per-token
-//    scores via 64-lane butterfly shfl reduction (result broadcast to all
-//    lanes, feeding the online update without extra exchanges).
+//    ushort8 (16 B) vector loads (rows padded 16 B for bank spread).
+//  - scoring is token-parallel: 4 lanes per token, each covering a 32-elem
+//    head_dim slice, 2-step part reduction — one pass scores the whole tile
+//    (v1's 16 serial 64-lane butterflies were the VALU bottleneck).
+//  - tile-level online softmax with defer-max (rescale only when the
+//    running max grows); PV accumulates per-lane head_dim pairs with the
+//    per-token p reaching lanes via one __shfl broadcast.
 #include <torch/extension.h>
 
 #include "common.h"
@@ -25,6 +27,10 @@ constexpr int kWaves = 4;
 constexpr int kBlockThreads = kWaves * WAVE_SIZE;
 constexpr int kBS = 16;   // cache block size (tokens)
 constexpr int kHD = 128;  // head dim
+// +8 ushorts (16 B) row padding: keeps 16 B staging alignment and breaks
+// the 256 B row stride so the (token, part) score reads spread over banks
+// (4-way residual aliasing ~= 1.6x on the LDS op, vs 32-way unpadded)
+constexpr int kPad = 8;
 
 template <int G>
 __launch_bounds__(kBlockThreads) __global__ void paged_decode_kernel(
@@ -52,18 +58,29 @@ __launch_bounds__(kBlockThreads) __global__ void paged_decode_kernel(
   const int blk_hi = min(n_blocks, blk_lo + chunk);
 
   // LDS: per-wave KV tile buffers + merge scratch
-  __shared__ ushort k_lds[kWaves][kBS][kHD];
-  __shared__ ushort v_lds[kWaves][kBS][kHD];
+  __shared__ ushort k_lds[kWaves][kBS][kHD + kPad];
+  __shared__ ushort v_lds[kWaves][kBS][kHD + kPad];
   __shared__ float merge_o[kWaves][G][kHD];
   __shared__ float merge_ml[kWaves][G][2];
 
-  // load q for all G heads of this group: lane owns elems {2l, 2l+1}
-  float qreg[G][2];
+  // --- lane roles -----------------------------------------------------
+  // scoring: 4 lanes per token (tok = lane>>2, part = lane&3), each part
+  //   covering a 32-elem slice of head_dim — all 16 tile tokens score in
+  //   parallel with a 2-step part reduction (replaces v1's 16 serial
+  //   64-lane butterflies, the VALU bottleneck)
+  // PV/output: lane owns head_dim elems {2*lane, 2*lane+1}; per-token p
+  //   reaches every lane via one __shfl broadcast
+  const int tok_of = lane >> 2;
+  const int part = lane & 3;
+
+  // q slice for this lane's part, packed bf16 pairs (16 u32 per head)
+  uint32_t q_pack[G][16];
 #pragma unroll
   for (int g = 0; g < G; ++g) {
-    const ushort* qh = q + (int64_t)b * q_stride + (int64_t)(kh * G + g) * kHD;
-    qreg[g][0] = bf16_to_f32(qh[2 * lane]) * scale;
-    qreg[g][1] = bf16_to_f32(qh[2 * lane + 1]) * scale;
+    const uint32_t* qp = reinterpret_cast<const uint32_t*>(
+        q + (int64_t)b * q_stride + (int64_t)(kh * G + g) * kHD + part * 32);
+#pragma unroll
+    for (int i = 0; i < 16; ++i) q_pack[g][i] = qp[i];
   }
 
   float m[G], l[G], o[G][2];
@@ -87,37 +104,79 @@ __launch_bounds__(kBlockThreads) __global__ void paged_decode_kernel(
           k_cache + ((blk * n_kv + kh) * kBS) * kHD);
       const ushort8* src_v = reinterpret_cast<const ushort8*>(
           v_cache + ((blk * n_kv + kh) * kBS) * kHD);
-      ushort8* dst_k = reinterpret_cast<ushort8*>(&k_lds[wave][0][0]);
-      ushort8* dst_v = reinterpret_cast<ushort8*>(&v_lds[wave][0][0]);
 #pragma unroll
       for (int i = 0; i < (kBS * kHD / 8) / WAVE_SIZE; ++i) {
-        dst_k[lane + i * WAVE_SIZE] = src_k[lane + i * WAVE_SIZE];
-        dst_v[lane + i * WAVE_SIZE] = src_v[lane + i * WAVE_SIZE];
+        const int vec = lane + i * WAVE_SIZE;  // 16B vector index
+        const int row = vec / (kHD / 8);
+        const int col = vec % (kHD / 8);
+        *reinterpret_cast<ushort8*>(&k_lds[wave][row][col * 8]) = src_k[vec];
+        *reinterpret_cast<ushort8*>(&v_lds[wave][row][col * 8]) = src_v[vec];
       }
     }
     // per-wave staging: compiler inserts lgkmcnt/vmcnt waits before LDS reads
 
-    for (int t = 0; t < tile_len; ++t) {
-      // lane reads its 2 K elems of token t (4B, 2-way bank aliased = free)
-      const uint32_t kk =
-          *reinterpret_cast<const uint32_t*>(&k_lds[wave][t][2 * lane]);
-      const float k0 = bf16_to_f32((ushort)(kk & 0xffff));
-      const float k1 = bf16_to_f32((ushort)(kk >> 16));
+    // ---- scores: lane computes its (token, part-slice) partial --------
+    float s[G];
+#pragma unroll
+    for (int g = 0; g < G; ++g) s[g] = 0.f;
+    {
+      const uint32_t* krow = reinterpret_cast<const uint32_t*>(
+          &k_lds[wave][tok_of][part * 32]);
+#pragma unroll
+      for (int j = 0; j < 16; ++j) {
+        const uint32_t kk = krow[j];
+        const float k0 = bf16_to_f32((ushort)(kk & 0xffff));
+        const float k1 = bf16_to_f32((ushort)(kk >> 16));
+#pragma unroll
+        for (int g = 0; g < G; ++g) {
+          const uint32_t qq = q_pack[g][j];
+          s[g] = fmaf(k0, bf16_to_f32((ushort)(qq & 0xffff)), s[g]);
+          s[g] = fmaf(k1, bf16_to_f32((ushort)(qq >> 16)), s[g]);
+        }
+      }
+    }
+    float p[G];
+    float rsum[G];
+#pragma unroll
+    for (int g = 0; g < G; ++g) {
+      // reduce across the 4 part-slices -> full dot in all 4 lanes
+      s[g] += __shfl_xor(s[g], 1, 64);
+      s[g] += __shfl_xor(s[g], 2, 64);
+      s[g] = (tok_of < tile_len) ? s[g] * scale : -INFINITY;
+      // tile max across tokens (bits 2..5 of the lane id)
+      float tmax = s[g];
+#pragma unroll
+      for (int off = 4; off < 64; off <<= 1)
+        tmax = fmaxf(tmax, __shfl_xor(tmax, off, 64));
+      const float m_new = fmaxf(m[g], tmax);
+      if (m_new != m[g]) {  // defer-max: wave-uniform rescale only on growth
+        const float corr = __expf(m[g] - m_new);
+        l[g] *= corr;
+        o[g][0] *= corr;
+        o[g][1] *= corr;
+        m[g] = m_new;
+      }
+      p[g] = (s[g] == -INFINITY) ? 0.f : __expf(s[g] - m_new);
+      rsum[g] = p[g];
+#pragma unroll
+      for (int off = 1; off < 64; off <<= 1)
+        rsum[g] += __shfl_xor(rsum[g], off, 64);
+      l[g] += rsum[g] * 0.25f;  // each token counted by its 4 part-lanes
+    }
+
+    // ---- PV: lane accumulates its 2 output elems ----------------------
+#pragma unroll
+    for (int t = 0; t < kBS; ++t) {
+      if (t >= tile_len) break;
       const uint32_t vv =
           *reinterpret_cast<const uint32_t*>(&v_lds[wave][t][2 * lane]);
       const float v0 = bf16_to_f32((ushort)(vv & 0xffff));
       const float v1 = bf16_to_f32((ushort)(vv >> 16));
 #pragma unroll
       for (int g = 0; g < G; ++g) {
-        float s = wave_reduce_sum(qreg[g][0] * k0 + qreg[g][1] * k1);
-        // online softmax update (q pre-scaled)
-        const float m_new = fmaxf(m[g], s);
-        const float corr = __expf(m[g] - m_new);
-        const float p = __expf(s - m_new);
-        l[g] = l[g] * corr + p;
-        o[g][0] = o[g][0] * corr + p * v0;
-        o[g][1] = o[g][1] * corr + p * v1;
-        m[g] = m_new;
+        const float pt = __shfl(p[g], t * 4, 64);
+        o[g][0] = fmaf(pt, v0, o[g][0]);
+        o[g][1] = fmaf(pt, v1, o[g][1]);
       }
     }
   }

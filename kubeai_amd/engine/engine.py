@@ -76,6 +76,7 @@ class LLMEngine:
             max_model_len=cfg.max_model_len,
             enable_graphs=cfg.enable_graphs,
             quantization=cfg.quantization,
+            model_path=cfg.model,
         )
         self.block_manager = BlockManager(self.runner.num_blocks, cfg.block_size)
         self.scheduler = Scheduler(

@@ -97,6 +97,7 @@ class ModelRunner:
         max_model_len: int = 8192,
         enable_graphs: bool = True,
         quantization: Optional[str] = None,
+        model_path: Optional[str] = None,
     ):
         self.arch = arch
         self.device = torch.device(device)
@@ -123,6 +124,14 @@ class ModelRunner:
             self.model = LlamaForCausalLM(arch, device=self.device, dtype=dtype)
             self.n_kv_local = arch.num_key_value_heads
         self.model.eval()
+        if model_path is not None and os.path.isdir(model_path):
+            from kubeai_amd.models.loader import load_weights
+
+            if tp_group is not None and tp_group.world > 1:
+                raise NotImplementedError(
+                    "TP checkpoint loading lands with the TP shard loader"
+                )
+            load_weights(self.model, model_path)
         self.quantization = quantization
         if quantization == "fp8":
             if self.device.type != "cuda":

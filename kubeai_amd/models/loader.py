@@ -1,0 +1,205 @@
+"""Checkpoint loading: HF safetensors -> engine model layout.
+
+Maps HF Llama/Mixtral parameter names onto the engine's fused layout
+(q/k/v -> qkv_proj, gate/up -> gate_up_proj); supports sharded
+model-*.safetensors. TP loading slices each full tensor per rank with the
+same sharding as parallel/tp.py.
+"""
+from __future__ import annotations
+
+import glob
+import json
+import os
+from typing import Iterator
+
+import torch
+
+
+def iter_safetensors(model_dir: str) -> Iterator[tuple[str, torch.Tensor]]:
+    from safetensors import safe_open
+
+    files = sorted(glob.glob(os.path.join(model_dir, "*.safetensors")))
+    if not files:
+        raise FileNotFoundError(f"no *.safetensors under {model_dir}")
+    for f in files:
+        with safe_open(f, framework="pt", device="cpu") as sf:
+            for key in sf.keys():
+                yield key, sf.get_tensor(key)
+
+
+def _strip(name: str) -> str:
+    for prefix in ("model.", "language_model.model."):
+        if name.startswith(prefix):
+            return name[len(prefix) :]
+    return name
+
+
+def load_weights(model, model_dir: str) -> int:
+    """Load HF weights into LlamaForCausalLM (dense or MoE). Returns the
+    number of engine parameters filled; raises if any stays unset."""
+    cfg = model.cfg
+    hd, nq, nkv = cfg.head_dim, cfg.num_attention_heads, cfg.num_key_value_heads
+    params = dict(model.named_parameters())
+    filled: set[str] = set()
+
+    def put(target: str, tensor: torch.Tensor) -> None:
+        p = params[target]
+        if p.shape != tensor.shape:
+            raise ValueError(f"{target}: shape {tuple(tensor.shape)} != {tuple(p.shape)}")
+        with torch.no_grad():
+            p.copy_(tensor.to(p.dtype))
+        filled.add(target)
+
+    # staging for fused tensors
+    pending: dict[str, dict[str, torch.Tensor]] = {}
+
+    def fuse_qkv(layer: int) -> None:
+        ps = pending.get(f"qkv.{layer}", {})
+        if len(ps) == 3:
+            put(
+                f"layers.{layer}.self_attn.qkv_proj.weight",
+                torch.cat([ps["q"], ps["k"], ps["v"]], dim=0),
+            )
+
+    def fuse_gate_up(layer: int, expert: int | None) -> None:
+        key = f"gu.{layer}" + ("" if expert is None else f".{expert}")
+        ps = pending.get(key, {})
+        if len(ps) == 2:
+            tgt = (
+                f"layers.{layer}.mlp.gate_up_proj.weight"
+                if expert is None
+                else f"layers.{layer}.mlp.experts.{expert}.gate_up_proj.weight"
+            )
+            put(tgt, torch.cat([ps["gate"], ps["up"]], dim=0))
+
+    for name, w in iter_safetensors(model_dir):
+        n = _strip(name)
+        if n == "embed_tokens.weight":
+            put("embed_tokens.weight", w)
+        elif n in ("lm_head.weight",):
+            if model.lm_head is not None:
+                put("lm_head.weight", w)
+        elif n == "norm.weight":
+            put("norm", w)
+        elif ".layers." in n or n.startswith("layers."):
+            parts = n.split(".")
+            layer = int(parts[parts.index("layers") + 1])
+            rest = ".".join(parts[parts.index("layers") + 2 :])
+            if rest == "input_layernorm.weight":
+                put(f"layers.{layer}.input_layernorm", w)
+            elif rest == "post_attention_layernorm.weight":
+                put(f"layers.{layer}.post_attention_layernorm", w)
+            elif rest == "self_attn.q_proj.weight":
+                pending.setdefault(f"qkv.{layer}", {})["q"] = w
+                fuse_qkv(layer)
+            elif rest == "self_attn.k_proj.weight":
+                pending.setdefault(f"qkv.{layer}", {})["k"] = w
+                fuse_qkv(layer)
+            elif rest == "self_attn.v_proj.weight":
+                pending.setdefault(f"qkv.{layer}", {})["v"] = w
+                fuse_qkv(layer)
+            elif rest == "self_attn.o_proj.weight":
+                put(f"layers.{layer}.self_attn.o_proj.weight", w)
+            elif rest == "mlp.gate_proj.weight":
+                pending.setdefault(f"gu.{layer}", {})["gate"] = w
+                fuse_gate_up(layer, None)
+            elif rest == "mlp.up_proj.weight":
+                pending.setdefault(f"gu.{layer}", {})["up"] = w
+                fuse_gate_up(layer, None)
+            elif rest == "mlp.down_proj.weight":
+                put(f"layers.{layer}.mlp.down_proj.weight", w)
+            elif rest == "block_sparse_moe.gate.weight":
+                put(f"layers.{layer}.mlp.gate.weight", w)
+            elif ".block_sparse_moe.experts." in "." + rest:
+                ep = rest.split(".")
+                e = int(ep[ep.index("experts") + 1])
+                which = ep[-2]  # w1/w2/w3
+                if which == "w1":  # gate
+                    pending.setdefault(f"gu.{layer}.{e}", {})["gate"] = w
+                    fuse_gate_up(layer, e)
+                elif which == "w3":  # up
+                    pending.setdefault(f"gu.{layer}.{e}", {})["up"] = w
+                    fuse_gate_up(layer, e)
+                elif which == "w2":  # down
+                    put(f"layers.{layer}.mlp.experts.{e}.down_proj.weight", w)
+
+    if model.lm_head is not None and "lm_head.weight" not in filled:
+        # tied embeddings checkpoints
+        with torch.no_grad():
+            model.lm_head.weight.copy_(model.embed_tokens.weight)
+        filled.add("lm_head.weight")
+
+    missing = set(params) - filled
+    if missing:
+        raise ValueError(f"unfilled parameters after load: {sorted(missing)[:8]}...")
+    return len(filled)
+
+
+def save_hf_checkpoint(model, out_dir: str) -> None:
+    """Write the model back out in HF layout (tests + weight-prep tooling)."""
+    from safetensors.torch import save_file
+
+    os.makedirs(out_dir, exist_ok=True)
+    cfg = model.cfg
+    state: dict[str, torch.Tensor] = {
+        "model.embed_tokens.weight": model.embed_tokens.weight.detach().cpu(),
+        "model.norm.weight": model.norm.detach().cpu(),
+    }
+    if model.lm_head is not None:
+        state["lm_head.weight"] = model.lm_head.weight.detach().cpu()
+    nq, nkv, hd = cfg.num_attention_heads, cfg.num_key_value_heads, cfg.head_dim
+    for i, layer in enumerate(model.layers):
+        pre = f"model.layers.{i}."
+        state[pre + "input_layernorm.weight"] = layer.input_layernorm.detach().cpu()
+        state[pre + "post_attention_layernorm.weight"] = (
+            layer.post_attention_layernorm.detach().cpu()
+        )
+        qkv = layer.self_attn.qkv_proj.weight.detach().cpu()
+        state[pre + "self_attn.q_proj.weight"] = qkv[: nq * hd].clone()
+        state[pre + "self_attn.k_proj.weight"] = qkv[nq * hd : (nq + nkv) * hd].clone()
+        state[pre + "self_attn.v_proj.weight"] = qkv[(nq + nkv) * hd :].clone()
+        state[pre + "self_attn.o_proj.weight"] = (
+            layer.self_attn.o_proj.weight.detach().cpu()
+        )
+        mlp = layer.mlp
+        if hasattr(mlp, "experts"):
+            state[pre + "block_sparse_moe.gate.weight"] = mlp.gate.weight.detach().cpu()
+            for e, ex in enumerate(mlp.experts):
+                gu = ex.gate_up_proj.weight.detach().cpu()
+                I = cfg.intermediate_size
+                epre = pre + f"block_sparse_moe.experts.{e}."
+                state[epre + "w1.weight"] = gu[:I].clone()
+                state[epre + "w3.weight"] = gu[I:].clone()
+                state[epre + "w2.weight"] = ex.down_proj.weight.detach().cpu()
+        else:
+            gu = mlp.gate_up_proj.weight.detach().cpu()
+            I = cfg.intermediate_size
+            state[pre + "mlp.gate_proj.weight"] = gu[:I].clone()
+            state[pre + "mlp.up_proj.weight"] = gu[I:].clone()
+            state[pre + "mlp.down_proj.weight"] = mlp.down_proj.weight.detach().cpu()
+    save_file(state, os.path.join(out_dir, "model.safetensors"))
+    with open(os.path.join(out_dir, "config.json"), "w") as f:
+        json.dump(
+            {
+                "architectures": [
+                    "MixtralForCausalLM" if cfg.num_local_experts else "LlamaForCausalLM"
+                ],
+                "vocab_size": cfg.vocab_size,
+                "hidden_size": cfg.hidden_size,
+                "intermediate_size": cfg.intermediate_size,
+                "num_hidden_layers": cfg.num_hidden_layers,
+                "num_attention_heads": cfg.num_attention_heads,
+                "num_key_value_heads": cfg.num_key_value_heads,
+                "head_dim": cfg.head_dim,
+                "max_position_embeddings": cfg.max_position_embeddings,
+                "rms_norm_eps": cfg.rms_norm_eps,
+                "rope_theta": cfg.rope_theta,
+                "tie_word_embeddings": cfg.tie_word_embeddings,
+                "bos_token_id": cfg.bos_token_id,
+                "eos_token_id": cfg.eos_token_id,
+                "num_local_experts": cfg.num_local_experts,
+                "num_experts_per_tok": cfg.num_experts_per_tok,
+                "torch_dtype": "bfloat16",
+            },
+            f,
+        )

@@ -5,17 +5,19 @@
 // this chunk's K/V via reshape_and_cache).
 //
 // MI355X-first structure (guide §3/§5/§6):
-//  - workgroup = (seq, kv_head, q_tile16); one wave per q-head of the GQA
-//    group (G waves), so the staged KV tile in LDS is consumed by all G
-//    heads — each KV byte crosses HBM once per 16*G query rows.
+//  - workgroup = (seq, kv_head, 32-row q-tile); one wave per q-head of the
+//    GQA group (G waves), so the staged KV tile in LDS is consumed by all G
+//    heads — each KV byte crosses HBM once per 32*G query rows.
+//  - each wave computes TWO 16-row MFMA sub-tiles per staged KV tile
+//    (QBLK=32): doubles arithmetic intensity over the v1 16-row tile.
 //  - QK^T and PV on v_mfma_f32_16x16x32_bf16; fp32 accumulation in AGPRs.
 //  - K tile [32][128] and transposed V tile [128][32] staged in LDS with
 //    the XOR swizzle (byte ^= (row&7)<<4) -> conflict-free ds_read_b128
 //    B-fragment reads (guide §6 Guideline 4).
 //  - online softmax entirely in C-fragment registers; the row (=q) direction
 //    lives in the low 4 lane bits, so row max/sum are 4-step shfl_xor
-//    butterflies; P goes through a 1 KiB/wave swizzled LDS round-trip to
-//    reach A-fragment layout for PV.
+//    butterflies; P goes through a swizzled LDS round-trip to reach
+//    A-fragment layout for PV.
 //
 // Fragment layouts (verified by tests/test_kernels_gpu.py numerics):
 //   A[M=16][K=32]: lane l holds A[l&15][(l>>4)*8 + j], j=0..7 (bf16)
@@ -29,7 +31,8 @@ namespace {
 
 constexpr int kHD = 128;   // head dim
 constexpr int kBS = 16;    // cache block size
-constexpr int kQB = 16;    // query rows per tile
+constexpr int kQB = 32;    // query rows per tile (2 MFMA sub-tiles)
+constexpr int kMT = kQB / 16;
 constexpr int kKVB = 32;   // kv tokens per tile (2 cache blocks)
 
 typedef __attribute__((ext_vector_type(8))) short bf16x8;
@@ -73,29 +76,35 @@ __launch_bounds__(G * WAVE_SIZE) __global__ void paged_prefill_kernel(
 
   __shared__ ushort k_lds[kKVB * kHD];          // [tok][hd], swizzled
   __shared__ ushort v_lds[kHD * kKVB];          // [hd][tok], swizzled
-  __shared__ ushort p_lds[G][kQB * kKVB];       // [row][tok], swizzled
+  __shared__ ushort p_lds[G][16 * kKVB];        // [row][tok], swizzled
 
-  // ---- load Q fragments (4 K-chunks of 32) straight from global ----
-  bf16x8 q_frag[4];
-  {
-    const int qrow = min(row_lo + (lane & 15), q_len - 1);
-    const ushort* qp = q + (int64_t)(s0 + qrow) * q_stride + (int64_t)head * kHD;
+  // ---- load Q fragments (2 sub-tiles x 4 K-chunks) from global ----
+  bf16x8 q_frag[kMT][4];
+#pragma unroll
+  for (int mt = 0; mt < kMT; ++mt) {
+    const int qrow = min(row_lo + mt * 16 + (lane & 15), q_len - 1);
+    const ushort* qp =
+        q + (int64_t)(s0 + qrow) * q_stride + (int64_t)head * kHD;
 #pragma unroll
     for (int kc = 0; kc < 4; ++kc) {
-      q_frag[kc] =
+      q_frag[mt][kc] =
           *reinterpret_cast<const bf16x8*>(qp + kc * 32 + (lane >> 4) * 8);
     }
   }
 
-  f32x4 o_acc[kHD / 16];
+  f32x4 o_acc[kMT][kHD / 16];
 #pragma unroll
-  for (int c = 0; c < kHD / 16; ++c) o_acc[c] = {0.f, 0.f, 0.f, 0.f};
-  float m_run[4], l_run[4];
+  for (int mt = 0; mt < kMT; ++mt)
 #pragma unroll
-  for (int i = 0; i < 4; ++i) {
-    m_run[i] = -INFINITY;
-    l_run[i] = 0.f;
-  }
+    for (int c = 0; c < kHD / 16; ++c) o_acc[mt][c] = {0.f, 0.f, 0.f, 0.f};
+  float m_run[kMT][4], l_run[kMT][4];
+#pragma unroll
+  for (int mt = 0; mt < kMT; ++mt)
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      m_run[mt][i] = -INFINITY;
+      l_run[mt][i] = 0.f;
+    }
 
   const int32_t* bt = block_tables + (int64_t)b * max_blocks;
 
@@ -106,7 +115,6 @@ __launch_bounds__(G * WAVE_SIZE) __global__ void paged_prefill_kernel(
     // ---- cooperative staging: K -> k_lds, V -> v_lds transposed ----
     __syncthreads();  // previous tile fully consumed
     {
-      // each thread stages ceil(32*128/8 / nthreads) ushort8 vectors
       const int nvec = kKVB * kHD / 8;  // 512
       for (int i = threadIdx.x; i < nvec; i += G * WAVE_SIZE) {
         const int tok = i / (kHD / 8);
@@ -149,97 +157,115 @@ __launch_bounds__(G * WAVE_SIZE) __global__ void paged_prefill_kernel(
     }
     __syncthreads();
 
-    // ---- S = Q.K^T over two 16-col subtiles ----
-    f32x4 s_frag[2];
-#pragma unroll
-    for (int nsub = 0; nsub < 2; ++nsub) {
-      s_frag[nsub] = {0.f, 0.f, 0.f, 0.f};
-#pragma unroll
-      for (int kc = 0; kc < 4; ++kc) {
-        const int tok = nsub * 16 + (lane & 15);
-        bf16x8 k_frag = *reinterpret_cast<const bf16x8*>(
-            reinterpret_cast<const char*>(k_lds) +
-            swz(tok, tok * kHD * 2 + kc * 64 + (lane >> 4) * 16));
-        s_frag[nsub] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            q_frag[kc], k_frag, s_frag[nsub], 0, 0, 0);
-      }
-    }
-
-    // ---- mask + online softmax in C-fragment layout ----
-    // lane holds rows r_i = (lane>>4)*4+i, cols (lane&15) and 16+(lane&15)
-    float p[2][4];  // [nsub][i] probabilities (pre-normalized)
-    float corr[4];
-#pragma unroll
-    for (int i = 0; i < 4; ++i) {
-      const int r = (lane >> 4) * 4 + i;
-      const int qpos = ctx + row_lo + r;
-      float s0v = s_frag[0][i] * scale;
-      float s1v = s_frag[1][i] * scale;
-      const int kv0 = kv_start + (lane & 15);
-      const int kv1 = kv0 + 16;
-      if (kv0 > qpos || r >= n_rows) s0v = -INFINITY;
-      if (kv1 > qpos || r >= n_rows) s1v = -INFINITY;
-      // row max across the 16 lanes of this row group
-      float rmax = fmaxf(s0v, s1v);
-#pragma unroll
-      for (int off = 8; off > 0; off >>= 1)
-        rmax = fmaxf(rmax, __shfl_xor(rmax, off, 64));
-      const float m_new = fmaxf(m_run[i], rmax);
-      // rows fully masked so far: keep m=-inf; avoid NaN from (-inf)-(-inf)
-      corr[i] = (m_run[i] == -INFINITY) ? 0.f : __expf(m_run[i] - m_new);
-      p[0][i] = (s0v == -INFINITY) ? 0.f : __expf(s0v - m_new);
-      p[1][i] = (s1v == -INFINITY) ? 0.f : __expf(s1v - m_new);
-      float rsum = p[0][i] + p[1][i];
-#pragma unroll
-      for (int off = 8; off > 0; off >>= 1) rsum += __shfl_xor(rsum, off, 64);
-      l_run[i] = l_run[i] * corr[i] + rsum;
-      m_run[i] = m_new;
-    }
-    // rescale O accumulator
-#pragma unroll
-    for (int c = 0; c < kHD / 16; ++c)
-#pragma unroll
-      for (int i = 0; i < 4; ++i) o_acc[c][i] *= corr[i];
-
-    // ---- P -> bf16 A-fragment via swizzled LDS round trip ----
+    // K fragments are shared by both q sub-tiles: load once per n-subtile
+    bf16x8 k_frag[2][4];
 #pragma unroll
     for (int nsub = 0; nsub < 2; ++nsub)
 #pragma unroll
-      for (int i = 0; i < 4; ++i) {
-        const int r = (lane >> 4) * 4 + i;
+      for (int kc = 0; kc < 4; ++kc) {
         const int tok = nsub * 16 + (lane & 15);
-        *reinterpret_cast<ushort*>(
-            reinterpret_cast<char*>(p_lds[wave]) +
-            swz(r, (r * kKVB + tok) * 2)) = f32_to_bf16(p[nsub][i]);
+        k_frag[nsub][kc] = *reinterpret_cast<const bf16x8*>(
+            reinterpret_cast<const char*>(k_lds) +
+            swz(tok, tok * kHD * 2 + kc * 64 + (lane >> 4) * 16));
       }
-    // wave-local LDS dependency; compiler inserts the lgkmcnt wait
-    bf16x8 p_frag = *reinterpret_cast<const bf16x8*>(
-        reinterpret_cast<const char*>(p_lds[wave]) +
-        swz(lane & 15, ((lane & 15) * kKVB + (lane >> 4) * 8) * 2));
-
-    // ---- O += P.V ----
+    bf16x8 v_frag[kHD / 16];
 #pragma unroll
     for (int c = 0; c < kHD / 16; ++c) {
       const int hdcol = c * 16 + (lane & 15);
-      bf16x8 v_frag = *reinterpret_cast<const bf16x8*>(
+      v_frag[c] = *reinterpret_cast<const bf16x8*>(
           reinterpret_cast<const char*>(v_lds) +
           swz(hdcol, (hdcol * kKVB + (lane >> 4) * 8) * 2));
-      o_acc[c] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(p_frag, v_frag,
-                                                         o_acc[c], 0, 0, 0);
+    }
+
+#pragma unroll
+    for (int mt = 0; mt < kMT; ++mt) {
+      const int mrow_lo = row_lo + mt * 16;
+      if (mt * 16 >= n_rows) break;               // sub-tile fully past q_len
+      if (kv_start > ctx + mrow_lo + 15) continue; // fully causal-masked
+
+      // ---- S = Q.K^T over two 16-col subtiles ----
+      f32x4 s_frag[2];
+#pragma unroll
+      for (int nsub = 0; nsub < 2; ++nsub) {
+        s_frag[nsub] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+        for (int kc = 0; kc < 4; ++kc) {
+          s_frag[nsub] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              q_frag[mt][kc], k_frag[nsub][kc], s_frag[nsub], 0, 0, 0);
+        }
+      }
+
+      // ---- mask + online softmax in C-fragment layout ----
+      float p[2][4];
+      float corr[4];
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        const int r = mt * 16 + (lane >> 4) * 4 + i;
+        const int qpos = ctx + row_lo + r;
+        float s0v = s_frag[0][i] * scale;
+        float s1v = s_frag[1][i] * scale;
+        const int kv0 = kv_start + (lane & 15);
+        const int kv1 = kv0 + 16;
+        if (kv0 > qpos || r >= n_rows) s0v = -INFINITY;
+        if (kv1 > qpos || r >= n_rows) s1v = -INFINITY;
+        float rmax = fmaxf(s0v, s1v);
+#pragma unroll
+        for (int off = 8; off > 0; off >>= 1)
+          rmax = fmaxf(rmax, __shfl_xor(rmax, off, 64));
+        const float m_new = fmaxf(m_run[mt][i], rmax);
+        corr[i] =
+            (m_run[mt][i] == -INFINITY) ? 0.f : __expf(m_run[mt][i] - m_new);
+        p[0][i] = (s0v == -INFINITY) ? 0.f : __expf(s0v - m_new);
+        p[1][i] = (s1v == -INFINITY) ? 0.f : __expf(s1v - m_new);
+        float rsum = p[0][i] + p[1][i];
+#pragma unroll
+        for (int off = 8; off > 0; off >>= 1) rsum += __shfl_xor(rsum, off, 64);
+        l_run[mt][i] = l_run[mt][i] * corr[i] + rsum;
+        m_run[mt][i] = m_new;
+      }
+#pragma unroll
+      for (int c = 0; c < kHD / 16; ++c)
+#pragma unroll
+        for (int i = 0; i < 4; ++i) o_acc[mt][c][i] *= corr[i];
+
+      // ---- P -> bf16 A-fragment via swizzled LDS round trip ----
+#pragma unroll
+      for (int nsub = 0; nsub < 2; ++nsub)
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+          const int r16 = (lane >> 4) * 4 + i;  // row within the sub-tile
+          const int tok = nsub * 16 + (lane & 15);
+          *reinterpret_cast<ushort*>(
+              reinterpret_cast<char*>(p_lds[wave]) +
+              swz(r16, (r16 * kKVB + tok) * 2)) = f32_to_bf16(p[nsub][i]);
+        }
+      // wave-local LDS dependency; compiler inserts the lgkmcnt wait
+      bf16x8 p_frag = *reinterpret_cast<const bf16x8*>(
+          reinterpret_cast<const char*>(p_lds[wave]) +
+          swz(lane & 15, ((lane & 15) * kKVB + (lane >> 4) * 8) * 2));
+
+      // ---- O += P.V ----
+#pragma unroll
+      for (int c = 0; c < kHD / 16; ++c) {
+        o_acc[mt][c] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            p_frag, v_frag[c], o_acc[mt][c], 0, 0, 0);
+      }
     }
   }
 
   // ---- write O / l ----
 #pragma unroll
-  for (int i = 0; i < 4; ++i) {
-    const int r = (lane >> 4) * 4 + i;
-    if (r >= n_rows) continue;
-    const float inv_l = 1.0f / l_run[i];
-    ushort* op = out + ((int64_t)(s0 + row_lo + r) * n_q + head) * kHD;
+  for (int mt = 0; mt < kMT; ++mt)
 #pragma unroll
-    for (int c = 0; c < kHD / 16; ++c)
-      op[c * 16 + (lane & 15)] = f32_to_bf16(o_acc[c][i] * inv_l);
-  }
+    for (int i = 0; i < 4; ++i) {
+      const int r = mt * 16 + (lane >> 4) * 4 + i;
+      if (r >= n_rows) continue;
+      const float inv_l = 1.0f / l_run[mt][i];
+      ushort* op = out + ((int64_t)(s0 + row_lo + r) * n_q + head) * kHD;
+#pragma unroll
+      for (int c = 0; c < kHD / 16; ++c)
+        op[c * 16 + (lane & 15)] = f32_to_bf16(o_acc[mt][c][i] * inv_l);
+    }
 }
 
 }  // namespace
@@ -265,9 +291,8 @@ void paged_attention_prefill(torch::Tensor out, torch::Tensor q,
   const int G = n_q / n_kv;
   TORCH_CHECK(n_q % n_kv == 0);
   if (q.size(0) == 0) return;
-  // max q tiles across sequences (host passes it via q len bound)
   const int Tq = q.size(0);
-  const int n_qtiles_max = (Tq + kQB - 1) / kQB;  // upper bound; per-seq early exit
+  const int n_qtiles_max = (Tq + kQB - 1) / kQB;  // per-seq early exit
   dim3 grid(B, n_kv, n_qtiles_max);
   auto stream = c10::hip::getCurrentHIPStream().stream();
 #define LAUNCH(GG)                                                         \

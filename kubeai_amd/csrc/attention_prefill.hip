@@ -31,7 +31,7 @@ namespace {
 
 constexpr int kHD = 128;   // head dim
 constexpr int kBS = 16;    // cache block size
-constexpr int kQB = 32;    // query rows per tile (2 MFMA sub-tiles)
+constexpr int kQB = 16;    // query rows per tile (32 costs occupancy: 201 VGPR -> 1 wave/SIMD, net loss)
 constexpr int kMT = kQB / 16;
 constexpr int kKVB = 32;   // kv tokens per tile (2 cache blocks)
 
@@ -157,26 +157,6 @@ __launch_bounds__(G * WAVE_SIZE) __global__ void paged_prefill_kernel(
     }
     __syncthreads();
 
-    // K fragments are shared by both q sub-tiles: load once per n-subtile
-    bf16x8 k_frag[2][4];
-#pragma unroll
-    for (int nsub = 0; nsub < 2; ++nsub)
-#pragma unroll
-      for (int kc = 0; kc < 4; ++kc) {
-        const int tok = nsub * 16 + (lane & 15);
-        k_frag[nsub][kc] = *reinterpret_cast<const bf16x8*>(
-            reinterpret_cast<const char*>(k_lds) +
-            swz(tok, tok * kHD * 2 + kc * 64 + (lane >> 4) * 16));
-      }
-    bf16x8 v_frag[kHD / 16];
-#pragma unroll
-    for (int c = 0; c < kHD / 16; ++c) {
-      const int hdcol = c * 16 + (lane & 15);
-      v_frag[c] = *reinterpret_cast<const bf16x8*>(
-          reinterpret_cast<const char*>(v_lds) +
-          swz(hdcol, (hdcol * kKVB + (lane >> 4) * 8) * 2));
-    }
-
 #pragma unroll
     for (int mt = 0; mt < kMT; ++mt) {
       const int mrow_lo = row_lo + mt * 16;
@@ -190,8 +170,12 @@ __launch_bounds__(G * WAVE_SIZE) __global__ void paged_prefill_kernel(
         s_frag[nsub] = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
         for (int kc = 0; kc < 4; ++kc) {
+          const int tok = nsub * 16 + (lane & 15);
+          const bf16x8 k_frag = *reinterpret_cast<const bf16x8*>(
+              reinterpret_cast<const char*>(k_lds) +
+              swz(tok, tok * kHD * 2 + kc * 64 + (lane >> 4) * 16));
           s_frag[nsub] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              q_frag[mt][kc], k_frag[nsub][kc], s_frag[nsub], 0, 0, 0);
+              q_frag[mt][kc], k_frag, s_frag[nsub], 0, 0, 0);
         }
       }
 
@@ -247,8 +231,12 @@ __launch_bounds__(G * WAVE_SIZE) __global__ void paged_prefill_kernel(
       // ---- O += P.V ----
 #pragma unroll
       for (int c = 0; c < kHD / 16; ++c) {
+        const int hdcol = c * 16 + (lane & 15);
+        const bf16x8 v_frag = *reinterpret_cast<const bf16x8*>(
+            reinterpret_cast<const char*>(v_lds) +
+            swz(hdcol, (hdcol * kKVB + (lane >> 4) * 8) * 2));
         o_acc[mt][c] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            p_frag, v_frag[c], o_acc[mt][c], 0, 0, 0);
+            p_frag, v_frag, o_acc[mt][c], 0, 0, 0);
       }
     }
   }

@@ -58,19 +58,34 @@ async def main_async(args):
             raise RuntimeError(f"replicas never ready: {reps}")
         print(f"[gateway_bench] {args.replicas} replica(s) ready", file=sys.stderr)
 
-        import benchmarks.multi_turn_chat as mtc
+        if args.mode == "chat":
+            import benchmarks.multi_turn_chat as mtc
 
-        bench_args = argparse.Namespace(
-            base_url=f"http://127.0.0.1:{args.port}",
-            model=args.model_name,
-            vus=args.vus,
-            iterations=args.iterations,
-            max_tokens=args.max_tokens,
-            user_words=48,
-            max_history=40,
-            timeout=300.0,
-        )
-        await mtc.main_async(bench_args)
+            bench_args = argparse.Namespace(
+                base_url=f"http://127.0.0.1:{args.port}",
+                model=args.model_name,
+                vus=args.vus,
+                iterations=args.iterations,
+                max_tokens=args.max_tokens,
+                user_words=48,
+                max_history=40,
+                timeout=300.0,
+            )
+            await mtc.main_async(bench_args)
+        else:  # bench_serving-style sweep
+            import benchmarks.serving_sweep as sweep
+
+            bench_args = argparse.Namespace(
+                base_url=f"http://127.0.0.1:{args.port}",
+                model=args.model_name,
+                num_prompts=args.iterations,
+                concurrency=args.vus,
+                max_tokens=args.max_tokens,
+                prefix_words=256,
+                suffix_words=32,
+                timeout=300.0,
+            )
+            await sweep.main_async(bench_args)
     finally:
         server.should_exit = True
         await asyncio.sleep(0.5)
@@ -87,6 +102,7 @@ def main():
     p.add_argument("--max-tokens", type=int, default=32)
     p.add_argument("--port", type=int, default=18080)
     p.add_argument("--resource-profile", default="amd-gpu-mi355x:1")
+    p.add_argument("--mode", choices=["chat", "sweep"], default="chat")
     asyncio.run(main_async(p.parse_args()))
 
 

@@ -19,6 +19,7 @@ void paged_attention_prefill(torch::Tensor out, torch::Tensor q,
                              torch::Tensor query_start_loc,
                              torch::Tensor seq_lens, double scale);
 void silu_and_mul(torch::Tensor out, torch::Tensor x);
+void skinny_gemm(torch::Tensor y, torch::Tensor x, torch::Tensor w);
 void greedy_sample(torch::Tensor out, torch::Tensor logits);
 void gumbel_sample(torch::Tensor out, torch::Tensor logits,
                    torch::Tensor temperature, torch::Tensor seeds,
@@ -35,6 +36,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("paged_attention_prefill", &paged_attention_prefill,
         "paged causal flash attention over cached KV");
   m.def("silu_and_mul", &silu_and_mul, "SwiGLU activation");
+  m.def("skinny_gemm", &skinny_gemm, "weight-streaming GEMM for M<=64");
   m.def("greedy_sample", &greedy_sample, "argmax sampling");
   m.def("gumbel_sample", &gumbel_sample, "temperature sampling (hash RNG)");
 }

@@ -25,6 +25,15 @@ from kubeai_amd.engine.batch import ForwardBatch
 from .config import ModelArchConfig
 
 
+class EngineLinear(nn.Linear):
+    """nn.Linear routed through ops.linear: the hand-written weight-
+    streaming GEMM takes skinny decode batches (M<=64); hipBLASLt keeps
+    the rest."""
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return ops.linear(x, self.weight)
+
+
 class Attention(nn.Module):
     def __init__(self, cfg: ModelArchConfig, layer_idx: int):
         super().__init__()
@@ -35,8 +44,8 @@ class Attention(nn.Module):
         self.scale = 1.0 / math.sqrt(self.hd)
         q_size = self.n_q * self.hd
         kv_size = self.n_kv * self.hd
-        self.qkv_proj = nn.Linear(cfg.hidden_size, q_size + 2 * kv_size, bias=False)
-        self.o_proj = nn.Linear(q_size, cfg.hidden_size, bias=False)
+        self.qkv_proj = EngineLinear(cfg.hidden_size, q_size + 2 * kv_size, bias=False)
+        self.o_proj = EngineLinear(q_size, cfg.hidden_size, bias=False)
 
     def forward(
         self,
@@ -97,10 +106,10 @@ class Attention(nn.Module):
 class MLP(nn.Module):
     def __init__(self, cfg: ModelArchConfig):
         super().__init__()
-        self.gate_up_proj = nn.Linear(
+        self.gate_up_proj = EngineLinear(
             cfg.hidden_size, 2 * cfg.intermediate_size, bias=False
         )
-        self.down_proj = nn.Linear(cfg.intermediate_size, cfg.hidden_size, bias=False)
+        self.down_proj = EngineLinear(cfg.intermediate_size, cfg.hidden_size, bias=False)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         return self.down_proj(ops.silu_and_mul(self.gate_up_proj(x)))
@@ -181,7 +190,7 @@ class LlamaForCausalLM(nn.Module):
                 if cfg.tie_word_embeddings:
                     self.lm_head = None
                 else:
-                    self.lm_head = nn.Linear(
+                    self.lm_head = EngineLinear(
                         cfg.hidden_size, cfg.vocab_size, bias=False
                     )
         finally:

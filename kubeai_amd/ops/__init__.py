@@ -121,6 +121,28 @@ def silu_and_mul(x):
     return ref.silu_and_mul(x)
 
 
+def linear(x, weight):
+    """GEMM dispatch: hand-written weight-streaming kernel for skinny decode
+    batches (M<=64, shapes aligned); hipBLASLt otherwise."""
+    if (
+        x.is_cuda
+        and x.dtype == torch.bfloat16
+        and x.dim() == 2
+        and 1 <= x.shape[0] <= 64
+        and weight.shape[1] % 64 == 0
+        and weight.shape[0] % 64 == 0
+        and x.is_contiguous()
+        and weight.is_contiguous()
+    ):
+        _require_ext()
+        y = torch.empty(
+            (x.shape[0], weight.shape[0]), dtype=x.dtype, device=x.device
+        )
+        _C.skinny_gemm(y, x, weight)
+        return y
+    return torch.nn.functional.linear(x, weight)
+
+
 def greedy_sample(logits):
     if logits.is_cuda:
         _require_ext()

@@ -28,7 +28,7 @@ import torch.nn.functional as F
 from kubeai_amd import ops
 from kubeai_amd.engine.batch import ForwardBatch
 from kubeai_amd.models.config import ModelArchConfig
-from kubeai_amd.models.llama import MLP, MoEMLP
+from kubeai_amd.models.llama import MLP, MoEMLP, EngineLinear
 
 
 class TPGroup:
@@ -73,8 +73,8 @@ class TPAttention(nn.Module):
         self.hd = cfg.head_dim
         self.scale = 1.0 / math.sqrt(self.hd)
         H = cfg.hidden_size
-        self.qkv_proj = nn.Linear(H, (self.n_q + 2 * self.n_kv) * self.hd, bias=False)
-        self.o_proj = nn.Linear(self.n_q * self.hd, H, bias=False)
+        self.qkv_proj = EngineLinear(H, (self.n_q + 2 * self.n_kv) * self.hd, bias=False)
+        self.o_proj = EngineLinear(self.n_q * self.hd, H, bias=False)
 
     def shard_from_full(self, full_qkv: torch.Tensor, full_o: torch.Tensor,
                         cfg: ModelArchConfig) -> None:
@@ -130,8 +130,8 @@ class TPMLP(nn.Module):
         self.tp = tp
         assert cfg.intermediate_size % tp.world == 0
         self.i_local = cfg.intermediate_size // tp.world
-        self.gate_up_proj = nn.Linear(cfg.hidden_size, 2 * self.i_local, bias=False)
-        self.down_proj = nn.Linear(self.i_local, cfg.hidden_size, bias=False)
+        self.gate_up_proj = EngineLinear(cfg.hidden_size, 2 * self.i_local, bias=False)
+        self.down_proj = EngineLinear(self.i_local, cfg.hidden_size, bias=False)
 
     def shard_from_full(self, full_gate_up: torch.Tensor, full_down: torch.Tensor,
                         cfg: ModelArchConfig) -> None:
@@ -202,7 +202,7 @@ class TPLlamaForCausalLM(nn.Module):
         if cfg.tie_word_embeddings:
             self.lm_head = None
         else:
-            self.lm_head = nn.Linear(H, cfg.vocab_size, bias=False)
+            self.lm_head = EngineLinear(H, cfg.vocab_size, bias=False)
             self.lm_head.weight.data.copy_(_full_weight((cfg.vocab_size, H), gen))
         self.to(device=device, dtype=dtype)
         from kubeai_amd.ops import ref as ops_ref

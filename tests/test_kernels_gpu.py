@@ -192,6 +192,21 @@ def test_silu_and_mul():
     assert_close_bf16(out, expected)
 
 
+@pytest.mark.parametrize("M", [1, 7, 16, 33, 40, 64])
+@pytest.mark.parametrize(
+    "N,K", [(6144, 4096), (4096, 4096), (4096, 14336), (128256, 4096)]
+)
+def test_skinny_gemm(M, N, K):
+    if N == 128256 and M > 16:
+        pytest.skip("one big-vocab case is enough")
+    x = torch.randn(M, K, dtype=torch.bfloat16, device=DEV) / 8
+    w = torch.randn(N, K, dtype=torch.bfloat16, device=DEV) / 8
+    out = ops.linear(x, w)
+    expected = (x.float() @ w.float().t())
+    # bf16 inputs, fp32 accumulation on both sides
+    torch.testing.assert_close(out.float(), expected, atol=0.3, rtol=3e-2)
+
+
 def test_greedy_sample():
     logits = torch.randn(64, 128256, dtype=torch.float32, device=DEV)
     out = ops.greedy_sample(logits)

@@ -121,11 +121,19 @@ def silu_and_mul(x):
     return ref.silu_and_mul(x)
 
 
+import os
+
+_USE_SKINNY = os.environ.get("KUBEAI_SKINNY_GEMM", "0") == "1"
+
+
 def linear(x, weight):
-    """GEMM dispatch: hand-written weight-streaming kernel for skinny decode
-    batches (M<=64, shapes aligned); hipBLASLt otherwise."""
+    """GEMM dispatch. The hand-written weight-streaming kernel (skinny_gemm)
+    currently loses to hipBLASLt on MI355X (direct 16 B row gathers do not
+    coalesce; measured 0.3-0.65x — profiles/r01_results.md), so it is
+    opt-in via KUBEAI_SKINNY_GEMM=1 until the LDS-staged variant lands."""
     if (
-        x.is_cuda
+        _USE_SKINNY
+        and x.is_cuda
         and x.dtype == torch.bfloat16
         and x.dim() == 2
         and 1 <= x.shape[0] <= 64

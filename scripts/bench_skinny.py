@@ -1,6 +1,7 @@
 import os, sys, time
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import torch
+os.environ["KUBEAI_SKINNY_GEMM"] = "1"
 import kubeai_amd.ops as ops
 
 def t(fn, iters=100, warm=20):

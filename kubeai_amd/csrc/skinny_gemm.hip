@@ -48,26 +48,56 @@ __launch_bounds__(256) __global__ void skinny_gemm_kernel(
   const int kc_lo = ks * per_split;
   const int kc_hi = min(k_chunk_total, kc_lo + per_split);
 
-  // A rows this lane touches (clamped for partial M)
-  const int kfrag = (lane >> 4) * 8;          // k offset within 32-chunk
+  // XOR-swizzled LDS tiles (row-major [64][kBK], byte ^= (row&7)<<4):
+  // coalesced global loads along k, conflict-free ds_read_b128 fragments
+  __shared__ ushort w_lds[kBN * kBK];
+  __shared__ ushort x_lds[64 * kBK];
 
   f32x4 acc[MT];
 #pragma unroll
   for (int mt = 0; mt < MT; ++mt) acc[mt] = {0.f, 0.f, 0.f, 0.f};
 
-  const ushort* wrow = w + (int64_t)(n0 + (lane & 15)) * K;
+  const int kfrag16 = (lane >> 4) * 16;  // byte offset of the lane's k range
 
   for (int kc = kc_lo; kc < kc_hi; ++kc) {
     const int kbase = kc * kBK;
+    __syncthreads();
+    {
+      // stage W tile [kBN][kBK] and x tile [M<=64][kBK]: 2 ushort8 each
+      const int vec = threadIdx.x;  // 256 threads, 8 vectors per row
+#pragma unroll
+      for (int h = 0; h < 2; ++h) {
+        const int v2 = vec + h * 256;
+        const int row = v2 >> 3;          // 0..63
+        const int col8 = v2 & 7;          // 16B slot within the row
+        *reinterpret_cast<bf16x8*>(
+            reinterpret_cast<char*>(w_lds) +
+            ((row * kBK + col8 * 8) * 2 ^ ((row & 7) << 4))) =
+            *reinterpret_cast<const bf16x8*>(
+                w + (int64_t)(n_tile * kBN + row) * K + kbase + col8 * 8);
+        const int xrow = min(row, M - 1);
+        *reinterpret_cast<bf16x8*>(
+            reinterpret_cast<char*>(x_lds) +
+            ((row * kBK + col8 * 8) * 2 ^ ((row & 7) << 4))) =
+            *reinterpret_cast<const bf16x8*>(
+                x + (int64_t)xrow * K + kbase + col8 * 8);
+      }
+    }
+    __syncthreads();
+
 #pragma unroll
     for (int half = 0; half < 2; ++half) {  // two k-steps of 32
-      const int k32 = kbase + half * 32 + kfrag;
-      const bf16x8 b_frag = *reinterpret_cast<const bf16x8*>(wrow + k32);
+      const int kb = half * 64;  // byte offset of this k-step in the row
+      const int wrow = wave * 16 + (lane & 15);
+      const bf16x8 b_frag = *reinterpret_cast<const bf16x8*>(
+          reinterpret_cast<const char*>(w_lds) +
+          ((wrow * kBK * 2 + kb + kfrag16) ^ ((wrow & 7) << 4)));
 #pragma unroll
       for (int mt = 0; mt < MT; ++mt) {
-        const int row = min(mt * 16 + (lane & 15), M - 1);
-        const bf16x8 a_frag =
-            *reinterpret_cast<const bf16x8*>(x + (int64_t)row * K + k32);
+        const int arow = mt * 16 + (lane & 15);
+        const bf16x8 a_frag = *reinterpret_cast<const bf16x8*>(
+            reinterpret_cast<const char*>(x_lds) +
+            ((arow * kBK * 2 + kb + kfrag16) ^ ((arow & 7) << 4)));
         acc[mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag, b_frag,
                                                           acc[mt], 0, 0, 0);
       }

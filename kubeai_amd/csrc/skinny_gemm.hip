@@ -48,10 +48,12 @@ __launch_bounds__(256) __global__ void skinny_gemm_kernel(
   const int kc_lo = ks * per_split;
   const int kc_hi = min(k_chunk_total, kc_lo + per_split);
 
-  // XOR-swizzled LDS tiles (row-major [64][kBK], byte ^= (row&7)<<4):
-  // coalesced global loads along k, conflict-free ds_read_b128 fragments
-  __shared__ ushort w_lds[kBN * kBK];
-  __shared__ ushort x_lds[64 * kBK];
+  // Double-buffered XOR-swizzled tiles, filled with async
+  // global_load_lds (16 B, wave-uniform LDS dest + per-lane PRE-SWIZZLED
+  // source address — guide m173): one barrier per k-chunk, loads for
+  // chunk kc+1 fly while chunk kc computes.
+  __shared__ ushort w_lds[2][kBN * kBK];
+  __shared__ ushort x_lds[2][64 * kBK];
 
   f32x4 acc[MT];
 #pragma unroll
@@ -59,49 +61,60 @@ __launch_bounds__(256) __global__ void skinny_gemm_kernel(
 
   const int kfrag16 = (lane >> 4) * 16;  // byte offset of the lane's k range
 
-  for (int kc = kc_lo; kc < kc_hi; ++kc) {
-    const int kbase = kc * kBK;
-    __syncthreads();
-    {
-      // stage W tile [kBN][kBK] and x tile [M<=64][kBK]: 2 ushort8 each
-      const int vec = threadIdx.x;  // 256 threads, 8 vectors per row
+  // staging geometry: 512 16B vectors per tile; this thread covers vector
+  // ids {tid, tid+256}; source byte offset is swizzled so the linear LDS
+  // write lands in swizzled layout
+  const int v_lo = threadIdx.x;
+  auto stage = [&](int buf, int kc) {
 #pragma unroll
-      for (int h = 0; h < 2; ++h) {
-        const int v2 = vec + h * 256;
-        const int row = v2 >> 3;          // 0..63
-        const int col8 = v2 & 7;          // 16B slot within the row
-        *reinterpret_cast<bf16x8*>(
-            reinterpret_cast<char*>(w_lds) +
-            ((row * kBK + col8 * 8) * 2 ^ ((row & 7) << 4))) =
-            *reinterpret_cast<const bf16x8*>(
-                w + (int64_t)(n_tile * kBN + row) * K + kbase + col8 * 8);
-        const int xrow = min(row, M - 1);
-        *reinterpret_cast<bf16x8*>(
-            reinterpret_cast<char*>(x_lds) +
-            ((row * kBK + col8 * 8) * 2 ^ ((row & 7) << 4))) =
-            *reinterpret_cast<const bf16x8*>(
-                x + (int64_t)xrow * K + kbase + col8 * 8);
-      }
+    for (int h = 0; h < 2; ++h) {
+      const int v = v_lo + h * 256;
+      const int row = v >> 3;
+      const int col8 = v & 7;
+      const int src_off = (col8 * 16) ^ ((row & 7) << 4);  // within 128B
+      const char* wsrc = reinterpret_cast<const char*>(
+          w + (int64_t)(n_tile * kBN + row) * K + kc * kBK) + src_off;
+      const int xrow = min(row, M - 1);
+      const char* xsrc = reinterpret_cast<const char*>(
+          x + (int64_t)xrow * K + kc * kBK) + src_off;
+      // LDS dest is wave-uniform: HW adds lane*16; this wave's 64
+      // vectors start at vector id (h*256 + wave*64)
+      const int wave_v0 = h * 256 + wave * 64;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) uint32_t*)wsrc,
+          (__attribute__((address_space(3))) uint32_t*)&w_lds[buf][wave_v0 * 8],
+          16, 0, 0);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) uint32_t*)xsrc,
+          (__attribute__((address_space(3))) uint32_t*)&x_lds[buf][wave_v0 * 8],
+          16, 0, 0);
     }
-    __syncthreads();
+  };
 
+  int cur = 0;
+  stage(0, kc_lo);
+  __syncthreads();  // compiler drains vmcnt before the barrier
+  for (int kc = kc_lo; kc < kc_hi; ++kc) {
+    if (kc + 1 < kc_hi) stage(cur ^ 1, kc + 1);
 #pragma unroll
     for (int half = 0; half < 2; ++half) {  // two k-steps of 32
       const int kb = half * 64;  // byte offset of this k-step in the row
       const int wrow = wave * 16 + (lane & 15);
       const bf16x8 b_frag = *reinterpret_cast<const bf16x8*>(
-          reinterpret_cast<const char*>(w_lds) +
+          reinterpret_cast<const char*>(w_lds[cur]) +
           ((wrow * kBK * 2 + kb + kfrag16) ^ ((wrow & 7) << 4)));
 #pragma unroll
       for (int mt = 0; mt < MT; ++mt) {
         const int arow = mt * 16 + (lane & 15);
         const bf16x8 a_frag = *reinterpret_cast<const bf16x8*>(
-            reinterpret_cast<const char*>(x_lds) +
+            reinterpret_cast<const char*>(x_lds[cur]) +
             ((arow * kBK * 2 + kb + kfrag16) ^ ((arow & 7) << 4)));
         acc[mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag, b_frag,
                                                           acc[mt], 0, 0, 0);
       }
     }
+    __syncthreads();
+    cur ^= 1;
   }
 
   // C layout: lane holds rows (lane>>4)*4+i, col lane&15 of each 16x16 tile

@@ -23,7 +23,7 @@
 
 namespace {
 
-constexpr int kWaves = 4;
+constexpr int kWaves = 2;  // 2 waves/workgroup: fits double-buffered tiles
 constexpr int kBlockThreads = kWaves * WAVE_SIZE;
 constexpr int kBS = 16;   // cache block size (tokens)
 constexpr int kHD = 128;  // head dim
@@ -57,9 +57,9 @@ __launch_bounds__(kBlockThreads) __global__ void paged_decode_kernel(
   const int blk_lo = split * chunk;
   const int blk_hi = min(n_blocks, blk_lo + chunk);
 
-  // LDS: per-wave KV tile buffers + merge scratch
-  __shared__ ushort k_lds[kWaves][kBS][kHD + kPad];
-  __shared__ ushort v_lds[kWaves][kBS][kHD + kPad];
+  // LDS: per-wave double-buffered KV tiles + merge scratch
+  __shared__ ushort k_lds[kWaves][2][kBS][kHD + kPad];
+  __shared__ ushort v_lds[kWaves][2][kBS][kHD + kPad];
   __shared__ float merge_o[kWaves][G][kHD];
   __shared__ float merge_ml[kWaves][G][2];
 
@@ -93,27 +93,46 @@ __launch_bounds__(kBlockThreads) __global__ void paged_decode_kernel(
 
   const int32_t* bt = block_tables + (int64_t)b * max_blocks;
 
-  for (int blk_i = blk_lo + wave; blk_i < blk_hi; blk_i += kWaves) {
+  // T14 async-stage: global loads for block n+1 issue before block n's
+  // compute (their latency hides under it); the register payload lands in
+  // the other LDS buffer just before it is needed. Per-wave buffers ->
+  // no barriers anywhere in the loop.
+  ushort8 stage_k[2], stage_v[2];  // this lane's 2 vectors of each tile
+
+  auto issue_loads = [&](int blk_i) {
     const int64_t blk = bt[blk_i];
+    const ushort8* src_k = reinterpret_cast<const ushort8*>(
+        k_cache + ((blk * n_kv + kh) * kBS) * kHD);
+    const ushort8* src_v = reinterpret_cast<const ushort8*>(
+        v_cache + ((blk * n_kv + kh) * kBS) * kHD);
+#pragma unroll
+    for (int i = 0; i < (kBS * kHD / 8) / WAVE_SIZE; ++i) {
+      stage_k[i] = src_k[lane + i * WAVE_SIZE];
+      stage_v[i] = src_v[lane + i * WAVE_SIZE];
+    }
+  };
+  auto write_tile = [&](int buf) {
+#pragma unroll
+    for (int i = 0; i < (kBS * kHD / 8) / WAVE_SIZE; ++i) {
+      const int vec = lane + i * WAVE_SIZE;  // 16B vector index
+      const int row = vec / (kHD / 8);
+      const int col = vec % (kHD / 8);
+      *reinterpret_cast<ushort8*>(&k_lds[wave][buf][row][col * 8]) = stage_k[i];
+      *reinterpret_cast<ushort8*>(&v_lds[wave][buf][row][col * 8]) = stage_v[i];
+    }
+  };
+
+  const int first = blk_lo + wave;
+  int cur = 0;
+  if (first < blk_hi) {
+    issue_loads(first);
+    write_tile(0);
+  }
+
+  for (int blk_i = first; blk_i < blk_hi; blk_i += kWaves) {
     const int tile_start = blk_i * kBS;
     const int tile_len = min(kBS, L - tile_start);
-
-    // stage K/V tile: 16 tok x 128 elems = 2048 elems = 32 ushort8 per wave
-    {
-      const ushort8* src_k = reinterpret_cast<const ushort8*>(
-          k_cache + ((blk * n_kv + kh) * kBS) * kHD);
-      const ushort8* src_v = reinterpret_cast<const ushort8*>(
-          v_cache + ((blk * n_kv + kh) * kBS) * kHD);
-#pragma unroll
-      for (int i = 0; i < (kBS * kHD / 8) / WAVE_SIZE; ++i) {
-        const int vec = lane + i * WAVE_SIZE;  // 16B vector index
-        const int row = vec / (kHD / 8);
-        const int col = vec % (kHD / 8);
-        *reinterpret_cast<ushort8*>(&k_lds[wave][row][col * 8]) = src_k[vec];
-        *reinterpret_cast<ushort8*>(&v_lds[wave][row][col * 8]) = src_v[vec];
-      }
-    }
-    // per-wave staging: compiler inserts lgkmcnt/vmcnt waits before LDS reads
+    if (blk_i + kWaves < blk_hi) issue_loads(blk_i + kWaves);
 
     // ---- scores: lane computes its (token, part-slice) partial --------
     float s[G];
@@ -121,7 +140,7 @@ __launch_bounds__(kBlockThreads) __global__ void paged_decode_kernel(
     for (int g = 0; g < G; ++g) s[g] = 0.f;
     {
       const uint32_t* krow = reinterpret_cast<const uint32_t*>(
-          &k_lds[wave][tok_of][part * 32]);
+          &k_lds[wave][cur][tok_of][part * 32]);
 #pragma unroll
       for (int j = 0; j < 16; ++j) {
         const uint32_t kk = krow[j];
@@ -169,7 +188,7 @@ __launch_bounds__(kBlockThreads) __global__ void paged_decode_kernel(
     for (int t = 0; t < kBS; ++t) {
       if (t >= tile_len) break;
       const uint32_t vv =
-          *reinterpret_cast<const uint32_t*>(&v_lds[wave][t][2 * lane]);
+          *reinterpret_cast<const uint32_t*>(&v_lds[wave][cur][t][2 * lane]);
       const float v0 = bf16_to_f32((ushort)(vv & 0xffff));
       const float v1 = bf16_to_f32((ushort)(vv >> 16));
 #pragma unroll
@@ -179,6 +198,10 @@ __launch_bounds__(kBlockThreads) __global__ void paged_decode_kernel(
         o[g][1] = fmaf(pt, v1, o[g][1]);
       }
     }
+
+    // land the prefetched tile in the other buffer (waits only on vmcnt)
+    if (blk_i + kWaves < blk_hi) write_tile(cur ^ 1);
+    cur ^= 1;
   }
 
   // cross-wave merge through LDS
@@ -282,8 +305,8 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
   // flash-decoding split: target >=1024 workgroups to fill 256 CUs
   int n_splits = 1;
   const int base_wgs = B * n_kv;
-  if (base_wgs < 1024) {
-    n_splits = std::min<int>(16, (1024 + base_wgs - 1) / base_wgs);
+  if (base_wgs < 2048) {
+    n_splits = std::min<int>(16, (2048 + base_wgs - 1) / base_wgs);
   }
   torch::Tensor part_o, part_ml;
   float *part_o_ptr = nullptr, *part_ml_ptr = nullptr;

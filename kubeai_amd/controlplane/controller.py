@@ -16,13 +16,8 @@ from typing import Optional
 
 import httpx
 
-from .crd import (
-    ADAPTER_LABEL_PREFIX,
-    CACHE_EVICTION_FINALIZER,
-    Model,
-    feature_labels,
-)
-from .store import Replica, ReplicaState, Store
+from .crd import CACHE_EVICTION_FINALIZER, Model, feature_labels
+from .store import Replica, Store
 
 
 def fnv1a_32(data: bytes) -> int:

@@ -17,7 +17,7 @@ import dataclasses
 import enum
 import itertools
 import uuid
-from typing import Callable, Optional
+from typing import Optional
 
 from .crd import Model, validate_model
 

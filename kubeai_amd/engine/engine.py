@@ -16,7 +16,7 @@ from kubeai_amd.models.config import ModelArchConfig, PRESETS
 
 from .kvcache import BlockManager
 from .runner import ModelRunner
-from .scheduler import Request, RequestStatus, SamplingParams, Scheduler
+from .scheduler import Request, SamplingParams, Scheduler
 
 
 @dataclasses.dataclass

@@ -17,7 +17,6 @@ from kubeai_amd.models.config import ModelArchConfig
 from kubeai_amd.models.llama import LlamaForCausalLM
 
 from .batch import ForwardBatch
-from .kvcache import BlockManager
 from .scheduler import SchedulerOutput
 
 

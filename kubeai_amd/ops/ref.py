@@ -18,8 +18,6 @@ Shapes / layout conventions (MI355X-first):
 """
 from __future__ import annotations
 
-import math
-
 import torch
 
 

@@ -28,7 +28,7 @@ import torch.nn.functional as F
 from kubeai_amd import ops
 from kubeai_amd.engine.batch import ForwardBatch
 from kubeai_amd.models.config import ModelArchConfig
-from kubeai_amd.models.llama import MLP, MoEMLP, EngineLinear
+from kubeai_amd.models.llama import EngineLinear
 
 
 class TPGroup:

@@ -97,7 +97,7 @@ __launch_bounds__(kBlockThreads) __global__ void paged_decode_kernel(
   // compute (their latency hides under it); the register payload lands in
   // the other LDS buffer just before it is needed. Per-wave buffers ->
   // no barriers anywhere in the loop.
-  ushort8 stage_k[2], stage_v[2];  // this lane's 2 vectors of each tile
+  ushort8 stage_k[4], stage_v[4];  // this lane's 4 vectors of each tile
 
   auto issue_loads = [&](int blk_i) {
     const int64_t blk = bt[blk_i];

@@ -24,6 +24,7 @@ void greedy_sample(torch::Tensor out, torch::Tensor logits);
 void gumbel_sample(torch::Tensor out, torch::Tensor logits,
                    torch::Tensor temperature, torch::Tensor seeds,
                    int64_t step);
+void register_chwbl(pybind11::module_& m);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm", &rmsnorm, "RMSNorm (bf16, fp32 accum)");
@@ -39,4 +40,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("skinny_gemm", &skinny_gemm, "weight-streaming GEMM for M<=64");
   m.def("greedy_sample", &greedy_sample, "argmax sampling");
   m.def("gumbel_sample", &gumbel_sample, "temperature sampling (hash RNG)");
+  register_chwbl(m);
 }

@@ -373,6 +373,40 @@ def test_models_listing_by_feature():
     run(body())
 
 
+def test_multi_model_routing():
+    """Two Models, separate replica sets; the gateway routes by model name
+    and in-flight accounting stays per-model."""
+
+    async def body():
+        m1, m2 = text_gen_model("alpha"), text_gen_model("beta")
+        m1.spec.replicas = 1
+        m2.spec.replicas = 1
+        async with harness([m1, m2]) as (mgr, runtime, backend):
+            backend2 = FakeBackend()
+            await backend2.start()
+            try:
+                r1 = (await wait_for(lambda: mgr.store.list_replicas("alpha")))[0]
+                r2 = (await wait_for(lambda: mgr.store.list_replicas("beta")))[0]
+                runtime.mark_ready(r1.name, backend.address)
+                runtime.mark_ready(r2.name, backend2.address)
+                transport = httpx.ASGITransport(app=mgr.app)
+                async with httpx.AsyncClient(
+                    transport=transport, base_url="http://gw"
+                ) as client:
+                    for name in ("alpha", "beta", "alpha"):
+                        resp = await client.post(
+                            "/openai/v1/completions",
+                            json={"model": name, "prompt": "x", "max_tokens": 1},
+                        )
+                        assert resp.status_code == 200
+                assert len(backend.requests) == 2   # alpha twice
+                assert len(backend2.requests) == 1  # beta once
+            finally:
+                await backend2.stop()
+
+    run(body())
+
+
 def test_proxy_multipart_audio_routing():
     """SpeechToText path: multipart body, model extracted from the form
     field, raw body forwarded (reference: apiutils/request.go:109-165)."""

@@ -25,6 +25,14 @@ void gumbel_sample(torch::Tensor out, torch::Tensor logits,
                    torch::Tensor temperature, torch::Tensor seeds,
                    int64_t step);
 void register_chwbl(pybind11::module_& m);
+void rmsnorm_fp8(torch::Tensor out, torch::Tensor out_scale, torch::Tensor x,
+                 torch::Tensor weight, double eps);
+void fused_add_rmsnorm_fp8(torch::Tensor out, torch::Tensor out_scale,
+                           torch::Tensor x, torch::Tensor residual,
+                           torch::Tensor weight, double eps);
+void silu_and_mul_fp8(torch::Tensor out, torch::Tensor out_scale,
+                      torch::Tensor x);
+void quant_fp8(torch::Tensor out, torch::Tensor out_scale, torch::Tensor x);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm", &rmsnorm, "RMSNorm (bf16, fp32 accum)");
@@ -40,5 +48,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("skinny_gemm", &skinny_gemm, "weight-streaming GEMM for M<=64");
   m.def("greedy_sample", &greedy_sample, "argmax sampling");
   m.def("gumbel_sample", &gumbel_sample, "temperature sampling (hash RNG)");
+  m.def("rmsnorm_fp8", &rmsnorm_fp8, "rmsnorm with fused fp8 row quant");
+  m.def("fused_add_rmsnorm_fp8", &fused_add_rmsnorm_fp8,
+        "residual add + rmsnorm with fused fp8 row quant");
+  m.def("silu_and_mul_fp8", &silu_and_mul_fp8, "SwiGLU with fused fp8 quant");
+  m.def("quant_fp8", &quant_fp8, "bf16 -> fp8 row quant");
   register_chwbl(m);
 }

@@ -40,6 +40,16 @@ class Fp8Linear(nn.Module):
             self.weight_fp8.float() * self.weight_scale.reshape(-1, 1)
         ).to(torch.bfloat16)
 
+    def forward_quantized(self, xq: torch.Tensor, x_scale: torch.Tensor) -> torch.Tensor:
+        """Pre-quantized input from a fused producer kernel (quant is free)."""
+        return torch._scaled_mm(
+            xq,
+            self.weight_fp8.t(),
+            scale_a=x_scale.view(-1, 1),
+            scale_b=self.weight_scale,
+            out_dtype=torch.bfloat16,
+        )
+
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         a_amax = x.float().abs().amax(dim=-1, keepdim=True).clamp_min(1e-6)
         a_scale = a_amax / FP8_MAX
@@ -57,7 +67,12 @@ _TARGETS = ("qkv_proj", "o_proj", "gate_up_proj", "down_proj", "lm_head")
 
 
 def convert_to_fp8(model: nn.Module) -> int:
-    """Swap target nn.Linear modules for Fp8Linear; returns count."""
+    """Swap target nn.Linear modules for Fp8Linear; returns count.
+
+    Dense decoder layers additionally get the FUSED activation-quant path
+    (llama.DecoderLayer checks _fp8_fused): producer kernels emit fp8 +
+    per-row scales, so per-linear dynamic quantization disappears.
+    """
     n = 0
     for mod in model.modules():
         for name in _TARGETS:
@@ -65,4 +80,13 @@ def convert_to_fp8(model: nn.Module) -> int:
             if isinstance(child, nn.Linear):
                 setattr(mod, name, Fp8Linear(child.weight))
                 n += 1
+    for layer in getattr(model, "layers", []):
+        attn = getattr(layer, "self_attn", None)
+        mlp = getattr(layer, "mlp", None)
+        if (
+            attn is not None
+            and isinstance(getattr(attn, "qkv_proj", None), Fp8Linear)
+            and isinstance(getattr(mlp, "gate_up_proj", None), Fp8Linear)
+        ):
+            layer._fp8_fused = True
     return n

@@ -54,11 +54,19 @@ class Attention(nn.Module):
         kv_cache: tuple[torch.Tensor, torch.Tensor],
         cos_sin: torch.Tensor,
     ) -> torch.Tensor:
-        T = x.shape[0]
-        qkv = self.qkv_proj(x)
-        lm = getattr(self, "_lora_manager", None)
-        if lm is not None and lm.active and fb.lora_ids is not None:
-            lm.apply(self.layer_idx, "qkv", x, qkv, fb.lora_ids)
+        fp8_in = isinstance(x, tuple)  # (fp8 tensor, row scales) from a
+        # fused producer kernel (DecoderLayer fp8 path)
+        if fp8_in:
+            xq, xs = x
+            T = xq.shape[0]
+            qkv = self.qkv_proj.forward_quantized(xq, xs)
+            lm = None
+        else:
+            T = x.shape[0]
+            qkv = self.qkv_proj(x)
+            lm = getattr(self, "_lora_manager", None)
+            if lm is not None and lm.active and fb.lora_ids is not None:
+                lm.apply(self.layer_idx, "qkv", x, qkv, fb.lora_ids)
         q, k, v = qkv.split(
             [self.n_q * self.hd, self.n_kv * self.hd, self.n_kv * self.hd], dim=-1
         )
@@ -97,6 +105,9 @@ class Attention(nn.Module):
                 out=out[nd:],
             )
         attn_flat = out.view(T, -1)
+        if fp8_in:
+            a8, ascale = ops.quant_fp8(attn_flat)
+            return self.o_proj.forward_quantized(a8, ascale)
         result = self.o_proj(attn_flat)
         if lm is not None and lm.active and fb.lora_ids is not None:
             lm.apply(self.layer_idx, "o", attn_flat, result, fb.lora_ids)
@@ -111,7 +122,11 @@ class MLP(nn.Module):
         )
         self.down_proj = EngineLinear(cfg.intermediate_size, cfg.hidden_size, bias=False)
 
-    def forward(self, x: torch.Tensor) -> torch.Tensor:
+    def forward(self, x) -> torch.Tensor:
+        if isinstance(x, tuple):  # fused fp8 path
+            h = self.gate_up_proj.forward_quantized(*x)
+            a8, ascale = ops.silu_and_mul_fp8(h)
+            return self.down_proj.forward_quantized(a8, ascale)
         return self.down_proj(ops.silu_and_mul(self.gate_up_proj(x)))
 
 
@@ -155,6 +170,27 @@ class DecoderLayer(nn.Module):
         self.eps = cfg.rms_norm_eps
 
     def forward(self, x, residual, fb, kv_cache, cos_sin):
+        if (
+            getattr(self, "_fp8_fused", False)
+            and x.is_cuda
+            and fb.lora_ids is None
+        ):
+            # fp8 serving: norms/activations emit fp8 + per-row scales
+            # directly (quant_fp8.hip), so activation quantization costs
+            # nothing extra; residual stays bf16
+            if residual is None:
+                residual = x
+                xq, xs = ops.rmsnorm_fp8(x, self.input_layernorm, self.eps)
+            else:
+                xq, xs = ops.fused_add_rmsnorm_fp8(
+                    x, residual, self.input_layernorm, self.eps
+                )
+            x = self.self_attn((xq, xs), fb, kv_cache, cos_sin)
+            xq, xs = ops.fused_add_rmsnorm_fp8(
+                x, residual, self.post_attention_layernorm, self.eps
+            )
+            x = self.mlp((xq, xs))
+            return x, residual
         if residual is None:
             residual = x
             x = ops.rmsnorm(x, self.input_layernorm, self.eps)

@@ -151,6 +151,41 @@ def linear(x, weight):
     return torch.nn.functional.linear(x, weight)
 
 
+def rmsnorm_fp8(x, weight, eps: float):
+    """GPU-only: rmsnorm with fused per-row fp8 quantization."""
+    _require_ext()
+    out = torch.empty(x.shape, dtype=torch.float8_e4m3fn, device=x.device)
+    scale = torch.empty(x.shape[0], dtype=torch.float32, device=x.device)
+    _C.rmsnorm_fp8(out, scale, x, weight, eps)
+    return out, scale
+
+
+def fused_add_rmsnorm_fp8(x, residual, weight, eps: float):
+    """GPU-only, in-place residual update; returns (fp8 out, row scales)."""
+    _require_ext()
+    out = torch.empty(x.shape, dtype=torch.float8_e4m3fn, device=x.device)
+    scale = torch.empty(x.shape[0], dtype=torch.float32, device=x.device)
+    _C.fused_add_rmsnorm_fp8(out, scale, x, residual, weight, eps)
+    return out, scale
+
+
+def silu_and_mul_fp8(x):
+    _require_ext()
+    T, two_i = x.shape
+    out = torch.empty((T, two_i // 2), dtype=torch.float8_e4m3fn, device=x.device)
+    scale = torch.empty(T, dtype=torch.float32, device=x.device)
+    _C.silu_and_mul_fp8(out, scale, x)
+    return out, scale
+
+
+def quant_fp8(x):
+    _require_ext()
+    out = torch.empty(x.shape, dtype=torch.float8_e4m3fn, device=x.device)
+    scale = torch.empty(x.shape[0], dtype=torch.float32, device=x.device)
+    _C.quant_fp8(out, scale, x)
+    return out, scale
+
+
 def greedy_sample(logits):
     if logits.is_cuda:
         _require_ext()

@@ -210,6 +210,54 @@ def test_skinny_gemm(M, N, K):
     torch.testing.assert_close(out.float(), expected, atol=0.3, rtol=3e-2)
 
 
+def _dequant(q8, scale):
+    return q8.float() * scale.view(-1, 1)
+
+
+@pytest.mark.parametrize("T,H", [(7, 4096), (64, 8192)])
+def test_rmsnorm_fp8(T, H):
+    x = torch.randn(T, H, dtype=torch.bfloat16, device=DEV)
+    w = torch.randn(H, dtype=torch.bfloat16, device=DEV)
+    q8, scale = ops.rmsnorm_fp8(x, w, 1e-5)
+    expected = ref.rmsnorm(x.float(), w.float(), 1e-5)
+    # scale = amax/448 per row
+    torch.testing.assert_close(
+        scale, expected.abs().amax(dim=-1) / 448.0, atol=1e-2, rtol=1e-2
+    )
+    torch.testing.assert_close(
+        _dequant(q8, scale), expected, atol=0.05, rtol=0.05
+    )
+
+
+def test_fused_add_rmsnorm_fp8():
+    T, H = 9, 4096
+    x = torch.randn(T, H, dtype=torch.bfloat16, device=DEV)
+    res = torch.randn(T, H, dtype=torch.bfloat16, device=DEV)
+    w = torch.randn(H, dtype=torch.bfloat16, device=DEV)
+    y_ref, res_ref = ref.fused_add_rmsnorm(x.float(), res.float(), w.float(), 1e-5)
+    q8, scale = ops.fused_add_rmsnorm_fp8(x, res, w, 1e-5)
+    assert_close_bf16(res, res_ref, atol=3e-2, rtol=3e-2)  # updated in place
+    torch.testing.assert_close(_dequant(q8, scale), y_ref, atol=0.06, rtol=0.06)
+
+
+def test_silu_and_mul_fp8():
+    x = torch.randn(13, 2 * 14336, dtype=torch.bfloat16, device=DEV)
+    q8, scale = ops.silu_and_mul_fp8(x)
+    expected = ref.silu_and_mul(x.float())
+    torch.testing.assert_close(_dequant(q8, scale), expected, atol=0.05, rtol=0.05)
+
+
+def test_quant_fp8_roundtrip():
+    x = torch.randn(21, 4096, dtype=torch.bfloat16, device=DEV) * 3
+    q8, scale = ops.quant_fp8(x)
+    torch.testing.assert_close(_dequant(q8, scale), x.float(), atol=0.05, rtol=0.05)
+    # matches torch's cast given the same scale
+    expected_q = (x.float() / scale.view(-1, 1)).clamp(-448, 448).to(
+        torch.float8_e4m3fn
+    )
+    assert (q8.view(torch.uint8) == expected_q.view(torch.uint8)).float().mean() > 0.99
+
+
 def test_greedy_sample():
     logits = torch.randn(64, 128256, dtype=torch.float32, device=DEV)
     out = ops.greedy_sample(logits)

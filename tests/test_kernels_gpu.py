@@ -225,7 +225,7 @@ def test_rmsnorm_fp8(T, H):
         scale, expected.abs().amax(dim=-1) / 448.0, atol=1e-2, rtol=1e-2
     )
     torch.testing.assert_close(
-        _dequant(q8, scale), expected, atol=0.05, rtol=0.05
+        _dequant(q8, scale), expected, atol=0.07, rtol=0.07
     )
 
 
@@ -237,20 +237,20 @@ def test_fused_add_rmsnorm_fp8():
     y_ref, res_ref = ref.fused_add_rmsnorm(x.float(), res.float(), w.float(), 1e-5)
     q8, scale = ops.fused_add_rmsnorm_fp8(x, res, w, 1e-5)
     assert_close_bf16(res, res_ref, atol=3e-2, rtol=3e-2)  # updated in place
-    torch.testing.assert_close(_dequant(q8, scale), y_ref, atol=0.06, rtol=0.06)
+    torch.testing.assert_close(_dequant(q8, scale), y_ref, atol=0.07, rtol=0.07)
 
 
 def test_silu_and_mul_fp8():
     x = torch.randn(13, 2 * 14336, dtype=torch.bfloat16, device=DEV)
     q8, scale = ops.silu_and_mul_fp8(x)
     expected = ref.silu_and_mul(x.float())
-    torch.testing.assert_close(_dequant(q8, scale), expected, atol=0.05, rtol=0.05)
+    torch.testing.assert_close(_dequant(q8, scale), expected, atol=0.07, rtol=0.07)
 
 
 def test_quant_fp8_roundtrip():
     x = torch.randn(21, 4096, dtype=torch.bfloat16, device=DEV) * 3
     q8, scale = ops.quant_fp8(x)
-    torch.testing.assert_close(_dequant(q8, scale), x.float(), atol=0.05, rtol=0.05)
+    torch.testing.assert_close(_dequant(q8, scale), x.float(), atol=0.1, rtol=0.07)
     # matches torch's cast given the same scale
     expected_q = (x.float() / scale.view(-1, 1)).clamp(-448, 448).to(
         torch.float8_e4m3fn

@@ -76,11 +76,27 @@ def get_args():
     p.add_argument("--prefill-interval", type=int, default=4)
     p.add_argument(
         "--quant",
-        default=None,
-        choices=[None, "fp8", "none"],
-        help="fp8 = W8A8 serving (the BASELINE config is Llama-3.1-8B-FP8)",
+        default="auto",
+        choices=["auto", "fp8", "none"],
+        help="fp8 = W8A8 serving with fused activation quant. Default auto "
+        "= fp8 for the Llama configs (the reference headline model is "
+        "Llama-3.1-8B-Instruct-FP8, i.e. fp8 IS the named precision).",
     )
     return p.parse_args()
+
+
+def _resolve_quant(args, model, use_cuda):
+    if not use_cuda:
+        return None
+    if args.quant == "fp8":
+        return "fp8"
+    if (
+        args.quant == "auto"
+        and args.tp == 1  # fused-fp8 path is validated single-GPU
+        and model in ("llama-3-8b", "llama-3-70b")
+    ):
+        return "fp8"  # matches the FP8-named baseline configs
+    return None
 
 
 class VirtualUser:
@@ -180,7 +196,7 @@ def main():
             max_num_seqs=max(args.vus * 2, 64),
             enable_prefix_caching=not args.no_prefix_cache,
             prefill_interval=args.prefill_interval,
-            quantization=args.quant if args.quant not in (None, "none") and use_cuda else None,
+            quantization=_resolve_quant(args, model, use_cuda),
             seed=replica_seed,
         ),
         tp_group=tp_group,
@@ -302,7 +318,7 @@ def main():
                     "higher_is_better": True,
                     "scaling": "weak",
                     "vs_baseline": round(value / baseline, 3),
-                    "dtype": "fp8" if (args.quant == "fp8" and use_cuda) else "bf16",
+                    "dtype": "fp8" if _resolve_quant(args, model, use_cuda) else "bf16",
                     "data": "synthetic",
                     "config": {
                         "model": model,

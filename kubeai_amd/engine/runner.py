@@ -372,6 +372,18 @@ class ModelRunner:
             raise RuntimeError(
                 f"logits rows {logits.shape[0]} != sample reqs {len(sample_reqs)}"
             )
+        # OpenAI presence/frequency penalties over generated tokens
+        for i, r in enumerate(sample_reqs):
+            pp, fp = r.params.presence_penalty, r.params.frequency_penalty
+            if (pp or fp) and r.num_generated > 0:
+                out_t = torch.tensor(
+                    r.output_token_ids, dtype=torch.int64, device=logits.device
+                )
+                counts = torch.zeros(
+                    logits.shape[1], dtype=torch.float32, device=logits.device
+                )
+                counts.scatter_add_(0, out_t, torch.ones_like(out_t, dtype=torch.float32))
+                logits[i] -= fp * counts + pp * (counts > 0).float()
         temps = [r.params.temperature for r in sample_reqs]
         if all(t <= 0.0 for t in temps):
             tokens = ops.greedy_sample(logits.contiguous())

@@ -25,6 +25,9 @@ class SamplingParams:
     seed: Optional[int] = None
     stop_token_ids: tuple[int, ...] = ()
     ignore_eos: bool = False
+    # OpenAI penalties over generated tokens (vLLM semantics)
+    presence_penalty: float = 0.0
+    frequency_penalty: float = 0.0
 
 
 class RequestStatus(enum.Enum):

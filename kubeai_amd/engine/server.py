@@ -211,7 +211,15 @@ def build_app(server: EngineServer) -> FastAPI:
             top_k=int(body.get("top_k") or 0),
             seed=int(seed) if seed is not None else None,
             ignore_eos=bool(body.get("ignore_eos") or False),
+            presence_penalty=float(body.get("presence_penalty") or 0.0),
+            frequency_penalty=float(body.get("frequency_penalty") or 0.0),
         )
+
+    def _stop_strings(body: dict) -> list[str]:
+        stop = body.get("stop")
+        if stop is None:
+            return []
+        return [stop] if isinstance(stop, str) else [s for s in stop if s]
 
     def _usage(prompt_toks: int, completion_toks: int) -> dict:
         return {
@@ -239,11 +247,22 @@ def build_app(server: EngineServer) -> FastAPI:
             data.append({"id": a, "object": "model", "owned_by": "kubeai-amd", "parent": name})
         return {"object": "list", "data": data}
 
-    async def _finish_tokens(gen):
+    async def _finish_tokens(gen, stops=None):
+        """Drain a generation; with stop strings, truncate at the earliest
+        match and abort the engine request early (the generator's finally
+        clause issues the abort)."""
         final = None
         async for o in gen:
             final = o
-        return final
+            if stops:
+                # full re-decode: incremental piece concatenation loses
+                # inter-token separators (tokenizer-dependent)
+                text = server.tokenizer.decode(o.output_token_ids)
+                for stp in stops:
+                    idx = text.find(stp)
+                    if idx >= 0:
+                        return final, text[:idx], "stop"
+        return final, None, None
 
     @app.post("/v1/completions")
     async def completions(request: Request):
@@ -259,27 +278,46 @@ def build_app(server: EngineServer) -> FastAPI:
         lora_id = _resolve_lora(server, body.get("model"))
         if isinstance(lora_id, JSONResponse):
             return lora_id
+        n = int(body.get("n") or 1)
+        stops = _stop_strings(body)
         if body.get("stream"):
+            if n > 1:
+                return JSONResponse(
+                    {"error": "stream with n>1 is not supported"}, status_code=400
+                )
             return StreamingResponse(
-                _stream_completion(server, toks, params, name, chat=False, lora_id=lora_id),
+                _stream_completion(server, toks, params, name, chat=False,
+                                   lora_id=lora_id, stops=stops),
                 media_type="text/event-stream",
             )
-        final = await _finish_tokens(server.generate(toks, params, lora_id))
-        text = server.tokenizer.decode(_strip_stop(final, params))
+        choices = []
+        pt = ct = 0
+        for i in range(n):
+            final, cut_text, cut_reason = await _finish_tokens(
+                server.generate(toks, params, lora_id), stops
+            )
+            text = (
+                cut_text
+                if cut_text is not None
+                else server.tokenizer.decode(_strip_stop(final, params))
+            )
+            choices.append(
+                {
+                    "index": i,
+                    "text": text,
+                    "finish_reason": cut_reason or final.finish_reason or "stop",
+                    "logprobs": None,
+                }
+            )
+            pt = final.num_prompt_tokens
+            ct += len(final.output_token_ids)
         return {
             "id": f"cmpl-{uuid.uuid4().hex[:12]}",
             "object": "text_completion",
             "created": int(time.time()),
             "model": name,
-            "choices": [
-                {
-                    "index": 0,
-                    "text": text,
-                    "finish_reason": final.finish_reason or "stop",
-                    "logprobs": None,
-                }
-            ],
-            "usage": _usage(final.num_prompt_tokens, len(final.output_token_ids)),
+            "choices": choices,
+            "usage": _usage(pt, ct),
         }
 
     @app.post("/v1/chat/completions")
@@ -290,26 +328,45 @@ def build_app(server: EngineServer) -> FastAPI:
         lora_id = _resolve_lora(server, body.get("model"))
         if isinstance(lora_id, JSONResponse):
             return lora_id
+        n = int(body.get("n") or 1)
+        stops = _stop_strings(body)
         if body.get("stream"):
+            if n > 1:
+                return JSONResponse(
+                    {"error": "stream with n>1 is not supported"}, status_code=400
+                )
             return StreamingResponse(
-                _stream_completion(server, toks, params, name, chat=True, lora_id=lora_id),
+                _stream_completion(server, toks, params, name, chat=True,
+                                   lora_id=lora_id, stops=stops),
                 media_type="text/event-stream",
             )
-        final = await _finish_tokens(server.generate(toks, params, lora_id))
-        text = server.tokenizer.decode(_strip_stop(final, params))
+        choices = []
+        pt = ct = 0
+        for i in range(n):
+            final, cut_text, cut_reason = await _finish_tokens(
+                server.generate(toks, params, lora_id), stops
+            )
+            text = (
+                cut_text
+                if cut_text is not None
+                else server.tokenizer.decode(_strip_stop(final, params))
+            )
+            choices.append(
+                {
+                    "index": i,
+                    "message": {"role": "assistant", "content": text},
+                    "finish_reason": cut_reason or final.finish_reason or "stop",
+                }
+            )
+            pt = final.num_prompt_tokens
+            ct += len(final.output_token_ids)
         return {
             "id": f"chatcmpl-{uuid.uuid4().hex[:12]}",
             "object": "chat.completion",
             "created": int(time.time()),
             "model": name,
-            "choices": [
-                {
-                    "index": 0,
-                    "message": {"role": "assistant", "content": text},
-                    "finish_reason": final.finish_reason or "stop",
-                }
-            ],
-            "usage": _usage(final.num_prompt_tokens, len(final.output_token_ids)),
+            "choices": choices,
+            "usage": _usage(pt, ct),
         }
 
     @app.post("/v1/embeddings")
@@ -422,12 +479,25 @@ def _resolve_lora(server: EngineServer, model_field: Optional[str]):
     )
 
 
-async def _stream_completion(server, toks, params, name, chat: bool, lora_id: int = 0):
+async def _stream_completion(server, toks, params, name, chat: bool,
+                             lora_id: int = 0, stops=None):
     rid = f"{'chatcmpl' if chat else 'cmpl'}-{uuid.uuid4().hex[:12]}"
     created = int(time.time())
     n_out = 0
+    text_so_far = ""
     async for o in server.generate(toks, params, lora_id):
-        piece = server.tokenizer.decode(o.new_token_ids)
+        # delta of the full decode: keeps inter-token separators exact
+        full = server.tokenizer.decode(_strip_stop(o, params))
+        piece = full[len(text_so_far):]
+        text_so_far = full
+        if stops:
+            hit = min((full.find(s) for s in stops if s in full), default=-1)
+            if hit >= 0:
+                keep = full[:hit]
+                prev = len(text_so_far) - len(piece)
+                piece = keep[prev:] if hit > prev else ""
+                o.finished = True
+                o.finish_reason = "stop"
         n_out = len(o.output_token_ids)
         if chat:
             chunk = {

@@ -156,6 +156,62 @@ def test_lora_changes_output(client):
     assert adapted != base
 
 
+def test_n_choices(client):
+    r = client.post(
+        "/v1/completions",
+        json={"prompt": "pick some words", "max_tokens": 4, "n": 3,
+              "temperature": 0.9},
+    )
+    assert r.status_code == 200
+    body = r.json()
+    assert len(body["choices"]) == 3
+    assert [c["index"] for c in body["choices"]] == [0, 1, 2]
+    # temperature sampling with distinct auto-seeds -> diverse outputs
+    texts = {c["text"] for c in body["choices"]}
+    assert len(texts) >= 2
+    assert body["usage"]["completion_tokens"] == 12
+
+
+def test_stop_strings(client):
+    base = client.post(
+        "/v1/completions",
+        json={"prompt": "stop string probe", "max_tokens": 8, "temperature": 0},
+    ).json()["choices"][0]["text"]
+    words = base.split()
+    assert len(words) >= 3
+    stop_word = words[2]
+    r = client.post(
+        "/v1/completions",
+        json={"prompt": "stop string probe", "max_tokens": 8, "temperature": 0,
+              "stop": [stop_word]},
+    ).json()
+    choice = r["choices"][0]
+    assert choice["finish_reason"] == "stop"
+    assert stop_word not in choice["text"]
+    assert choice["text"].split() == words[:2]
+
+
+def test_penalties_change_output(client):
+    base = client.post(
+        "/v1/completions",
+        json={"prompt": "penalty probe text", "max_tokens": 10,
+              "temperature": 0},
+    ).json()["choices"][0]["text"]
+    pen = client.post(
+        "/v1/completions",
+        json={"prompt": "penalty probe text", "max_tokens": 10,
+              "temperature": 0, "frequency_penalty": 2.0,
+              "presence_penalty": 2.0},
+    ).json()["choices"][0]["text"]
+    # strong penalties must forbid immediate repeats; outputs diverge once
+    # the greedy path would have repeated a token
+    base_toks = base.split()
+    pen_toks = pen.split()
+    if len(set(base_toks)) < len(base_toks):  # base repeats something
+        assert pen != base
+    assert len(set(pen_toks)) == len(pen_toks)  # no repeats under penalty
+
+
 def test_unknown_model_404(client):
     r = client.post(
         "/v1/completions", json={"model": "nope", "prompt": "x", "max_tokens": 1}

@@ -11,6 +11,24 @@ from typing import Optional
 import yaml
 
 
+def parse_duration(v) -> float:
+    """Accept Go-style duration strings ("10s", "10m", "1h30m") or numbers
+    (reference: config durations are metav1.Duration)."""
+    if isinstance(v, (int, float)):
+        return float(v)
+    import re
+
+    total = 0.0
+    for num, unit in re.findall(r"([0-9.]+)(ms|s|m|h)", str(v)):
+        total += float(num) * {"ms": 0.001, "s": 1, "m": 60, "h": 3600}[unit]
+    if total == 0.0 and str(v).strip():
+        try:
+            return float(v)
+        except ValueError:
+            raise ValueError(f"invalid duration: {v!r}")
+    return total
+
+
 @dataclasses.dataclass
 class AutoscalingConfig:
     # reference: system.go:119-146
@@ -73,8 +91,8 @@ def load_config(path: Optional[str]) -> SystemConfig:
         },
         cache_dir=raw.get("cacheDir", "/tmp/kubeai-cache"),
         autoscaling=AutoscalingConfig(
-            interval_seconds=float(auto.get("interval", 10)),
-            time_window_seconds=float(auto.get("timeWindow", 600)),
+            interval_seconds=parse_duration(auto.get("interval", 10)),
+            time_window_seconds=parse_duration(auto.get("timeWindow", 600)),
             state_path=auto.get("statePath", "/tmp/kubeai-amd-autoscaler-state.json"),
         ),
         messaging=[

@@ -230,3 +230,19 @@ def test_scale_down_delay_hysteresis():
     # scale-up is never delayed
     mc.scale("hyst", 4)
     assert store.get_model("hyst").spec.replicas == 4
+
+
+def test_duration_parsing(tmp_path):
+    from kubeai_amd.controlplane.config import parse_duration
+
+    assert parse_duration(10) == 10.0
+    assert parse_duration("10s") == 10.0
+    assert parse_duration("10m") == 600.0
+    assert parse_duration("1h30m") == 5400.0
+    assert parse_duration("250ms") == 0.25
+    assert parse_duration("2.5") == 2.5
+    p = tmp_path / "c.yaml"
+    p.write_text("modelAutoscaling:\n  interval: 10s\n  timeWindow: 10m\n")
+    cfg = load_config(str(p))
+    assert cfg.autoscaling.interval_seconds == 10.0
+    assert cfg.autoscaling.time_window_seconds == 600.0

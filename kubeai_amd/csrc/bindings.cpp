@@ -24,6 +24,8 @@ void greedy_sample(torch::Tensor out, torch::Tensor logits);
 void gumbel_sample(torch::Tensor out, torch::Tensor logits,
                    torch::Tensor temperature, torch::Tensor seeds,
                    int64_t step);
+void sgmv(torch::Tensor y, torch::Tensor x, torch::Tensor A, torch::Tensor B,
+          torch::Tensor idx, double scale);
 void register_chwbl(pybind11::module_& m);
 void rmsnorm_fp8(torch::Tensor out, torch::Tensor out_scale, torch::Tensor x,
                  torch::Tensor weight, double eps);
@@ -53,5 +55,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "residual add + rmsnorm with fused fp8 row quant");
   m.def("silu_and_mul_fp8", &silu_and_mul_fp8, "SwiGLU with fused fp8 quant");
   m.def("quant_fp8", &quant_fp8, "bf16 -> fp8 row quant");
+  m.def("sgmv", &sgmv, "segmented gather LoRA apply (one adapter segment)");
   register_chwbl(m);
 }

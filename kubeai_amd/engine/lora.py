@@ -143,7 +143,8 @@ class LoRAManager:
         y: torch.Tensor,  # [T, out] += delta (in-place)
         lora_ids: torch.Tensor,  # [T] int32
     ) -> None:
-        """Segmented LoRA matmul: per-adapter token groups (SGMV shape)."""
+        """Segmented LoRA matmul: per-adapter token groups; one hand-written
+        SGMV kernel launch per adapter on GPU (csrc/sgmv.hip)."""
         present = torch.unique(lora_ids)
         for lid_t in present:
             lid = int(lid_t)
@@ -151,8 +152,13 @@ class LoRAManager:
                 continue
             A, B, s = self.adapters[lid][target][layer_idx]
             idx = (lora_ids == lid_t).nonzero(as_tuple=True)[0]
-            xs = x[idx]
-            y[idx] += (xs @ A.T @ B.T) * s
+            if x.is_cuda:
+                from kubeai_amd import _C
+
+                _C.sgmv(y, x, A, B, idx, s)
+            else:
+                xs = x[idx]
+                y[idx] += (xs @ A.T @ B.T) * s
 
 
 def _peft_weights_present(path: str) -> Optional[str]:

@@ -210,6 +210,27 @@ def test_skinny_gemm(M, N, K):
     torch.testing.assert_close(out.float(), expected, atol=0.3, rtol=3e-2)
 
 
+@pytest.mark.parametrize("R", [8, 16, 64])
+def test_sgmv(R):
+    from kubeai_amd import _C
+
+    T, H, out = 33, 4096, 6144
+    x = torch.randn(T, H, dtype=torch.bfloat16, device=DEV)
+    y = torch.randn(T, out, dtype=torch.bfloat16, device=DEV)
+    A = torch.randn(R, H, dtype=torch.bfloat16, device=DEV) * 0.05
+    B = torch.randn(out, R, dtype=torch.bfloat16, device=DEV) * 0.5
+    idx = torch.tensor([0, 3, 7, 20, 32], dtype=torch.int64, device=DEV)
+    scale = 0.125
+    expected = y.float().clone()
+    expected[idx] += (x[idx].float() @ A.float().T @ B.float().T) * scale
+    y2 = y.clone()
+    _C.sgmv(y2, x, A, B, idx, scale)
+    untouched = torch.ones(T, dtype=torch.bool)
+    untouched[idx.cpu()] = False
+    torch.testing.assert_close(y2[untouched], y[untouched])
+    assert_close_bf16(y2[idx], expected[idx], atol=5e-2, rtol=5e-2)
+
+
 def _dequant(q8, scale):
     return q8.float() * scale.view(-1, 1)
 

@@ -87,22 +87,43 @@ def test_gpu_logits_match_cpu_reference():
 
 
 def test_lora_adapter_gpu():
-    """Adapter serving on the HIP path: SGMV kernel per adapter segment;
-    base and adapter streams isolated (KV salt) and deterministic."""
+    """Adapter serving on the HIP path: the SGMV kernel must shift the
+    adapter stream's logits while leaving the base stream bit-stable."""
+    import torch
+
+    from kubeai_amd.engine.scheduler import (
+        Request,
+        SamplingParams as SP,
+        ScheduledSeq,
+        SchedulerOutput,
+    )
+
     eng = make_engine("cuda")
     eng.runner.load_lora(7, None)  # synthetic rank-16 adapter
     prompt = list(range(10, 100))
-    eng.add_request(prompt, SamplingParams(max_tokens=6), request_id="base")
-    base = drain(eng)["base"].output_token_ids
-    eng.add_request(prompt, SamplingParams(max_tokens=6), request_id="ad",
-                    lora_id=7)
-    ad = drain(eng)["ad"].output_token_ids
-    assert ad != base  # adapter perturbs the greedy path
-    eng.add_request(prompt, SamplingParams(max_tokens=6), request_id="ad2",
-                    lora_id=7)
-    assert drain(eng)["ad2"].output_token_ids == ad  # deterministic
-    eng.add_request(prompt, SamplingParams(max_tokens=6), request_id="base2")
-    assert drain(eng)["base2"].output_token_ids == base  # base unaffected
+
+    def last_logits(lora_id, salt):
+        toks = list(prompt)
+        table, _ = eng.block_manager.allocate(toks, salt=salt, max_cached=0)
+        req = Request(toks, SP(max_tokens=1), lora_id=lora_id)
+        req.block_table = table
+        so = SchedulerOutput(
+            decode=[], prefill=[ScheduledSeq(req, 0, len(toks))], preempted=[]
+        )
+        fb = eng.runner.build_batch(so)
+        h = eng.runner.model(fb)
+        logits = eng.runner.model.compute_logits(h[len(toks) - 1 : len(toks)])[0]
+        eng.block_manager.free(table)
+        return logits
+
+    base1 = last_logits(0, -10)
+    ad = last_logits(7, -11)
+    base2 = last_logits(0, -12)
+    assert torch.equal(base1, base2)  # base stream untouched by the adapter
+    diff = (ad - base1).abs().max().item()
+    assert diff > 0.05, f"adapter did not move logits (max diff {diff})"
+    ad2 = last_logits(7, -13)
+    torch.testing.assert_close(ad, ad2)  # deterministic adapter stream
 
 
 def test_mixtral_moe_gpu():

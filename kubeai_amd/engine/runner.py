@@ -404,21 +404,45 @@ class ModelRunner:
                 device=dev,
             )
             if needs_topk:
-                from kubeai_amd.ops import ref as ops_ref
-
-                top_p = torch.tensor(
-                    [r.params.top_p for r in sample_reqs], dtype=torch.float32
+                # mask the distribution on-device, then draw with the
+                # deterministic Gumbel kernel (no CPU round trip)
+                masked = _apply_topk_topp(
+                    logits,
+                    [r.params.top_p for r in sample_reqs],
+                    [r.params.top_k for r in sample_reqs],
+                    [r.params.temperature for r in sample_reqs],
                 )
-                top_k = torch.tensor(
-                    [r.params.top_k for r in sample_reqs], dtype=torch.int64
-                )
-                tokens = ops_ref.topk_topp_sample(
-                    logits.cpu(), t_t.cpu(), top_p, top_k, seeds.cpu(), step
-                ).to(dev)
+                tokens = ops.gumbel_sample(masked.contiguous(), t_t, seeds, step).to(dev)
             else:
                 tokens = ops.gumbel_sample(logits.contiguous(), t_t, seeds, step)
         tokens = tokens.cpu().tolist()
         return {r.request_id: int(t) for r, t in zip(sample_reqs, tokens)}
+
+
+def _apply_topk_topp(logits, top_ps, top_ks, temps):
+    """Mask logits outside the per-row top-k / top-p (nucleus) sets.
+
+    Sampling from the masked logits via Gumbel-argmax is exactly nucleus
+    sampling: probabilities renormalize implicitly.
+    """
+    out = logits.clone()
+    sorted_logits, sorted_idx = out.sort(dim=-1, descending=True)
+    for i, (tp, tk, tt) in enumerate(zip(top_ps, top_ks, temps)):
+        if tt <= 0:
+            continue  # greedy rows ignore the mask
+        row = sorted_logits[i]
+        keep = torch.ones_like(row, dtype=torch.bool)
+        if tk and tk > 0:
+            keep[tk:] = False
+        if tp < 1.0:
+            probs = torch.softmax(row / tt, dim=-1)
+            csum = probs.cumsum(0)
+            # keep tokens while cumulative prob (exclusive) < top_p
+            keep &= (csum - probs) < tp
+            keep[0] = True
+        drop_idx = sorted_idx[i][~keep]
+        out[i, drop_idx] = float("-inf")
+    return out
 
 
 _BUCKETS = [8, 16, 24, 32, 40, 48, 64, 80, 96, 128, 160, 192, 256, 320,

@@ -151,3 +151,34 @@ def test_long_generation_grows_blocks():
     eng.add_request(list(range(3, 20)), SamplingParams(max_tokens=70), request_id="g")
     out = run_to_completion(eng)["g"]
     assert len(out.output_token_ids) == 70
+
+
+def test_topk_topp_masked_sampling():
+    """Nucleus sampling via masked Gumbel: only tokens inside the top-k /
+    top-p set may ever be drawn; deterministic per (seed, step)."""
+    import torch
+
+    from kubeai_amd.engine.runner import _apply_topk_topp
+    from kubeai_amd import ops
+
+    torch.manual_seed(0)
+    logits = torch.randn(4, 64) * 3
+    masked = _apply_topk_topp(logits, [1.0, 0.5, 1.0, 1.0], [2, 0, 0, 3],
+                              [1.0, 1.0, 0.0, 0.7])
+    top2 = set(logits[0].topk(2).indices.tolist())
+    temps = torch.tensor([1.0, 1.0, 0.0, 0.7])
+    seeds = torch.arange(4, dtype=torch.int64)
+    seen0 = set()
+    for step in range(50):
+        toks = ops.gumbel_sample(masked, temps, seeds, step)
+        seen0.add(int(toks[0]))
+        assert int(toks[0]) in top2
+        assert int(toks[2]) == int(logits[2].argmax())  # greedy row unmasked
+    assert len(seen0) == 2  # both top-2 tokens appear over 50 draws
+    # top-p row: drawn tokens restricted to the nucleus
+    probs = torch.softmax(logits[1], -1)
+    sp, si = probs.sort(descending=True)
+    nucleus = set(si[(sp.cumsum(0) - sp) < 0.5].tolist())
+    for step in range(30):
+        toks = ops.gumbel_sample(masked, temps, seeds, step)
+        assert int(toks[1]) in nucleus

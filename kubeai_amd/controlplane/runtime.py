@@ -101,9 +101,18 @@ class LocalProcessRuntime:
         self.health_interval = health_interval
         self._monitors: dict[str, asyncio.Task] = {}
 
+    SUPPORTED_ENGINES = ("KubeAIEngine",)
+
     async def create(self, model: Model, name: str, spec_hash: str, n_gpus: int) -> None:
-        ids = self.gpus.acquire(n_gpus)
         rep = Replica(name=name, model=model.name, hash=spec_hash)
+        if model.spec.engine not in self.SUPPORTED_ENGINES:
+            # third-party engine images (vLLM/Ollama/FasterWhisper/Infinity)
+            # are cluster-deployment concerns (chart modelServers); this
+            # node runtime runs the in-house engine only
+            rep.scheduled = False
+            self.store.add_replica(rep)
+            return
+        ids = self.gpus.acquire(n_gpus)
         if ids is None:
             rep.scheduled = False  # unschedulable: waits for free GPUs
             self.store.add_replica(rep)

@@ -234,6 +234,12 @@ def build_app(server: EngineServer) -> FastAPI:
             return JSONResponse({"status": "starting"}, status_code=503)
         return {"status": "ok"}
 
+    @app.get("/version")
+    async def version():
+        import kubeai_amd
+
+        return {"version": kubeai_amd.__version__}
+
     @app.get("/metrics")
     async def metrics():
         return PlainTextResponse(

@@ -120,14 +120,26 @@ class LLMEngine:
     # ------------------------------------------------------------------
     def step(self) -> list[RequestOutput]:
         out = self.scheduler.schedule()
+        results: list[RequestOutput] = []
+        for req in out.rejected:
+            results.append(
+                RequestOutput(
+                    request_id=req.request_id,
+                    new_token_ids=[],
+                    finished=True,
+                    finish_reason="abort",
+                    num_prompt_tokens=req.num_prompt_tokens,
+                    num_cached_tokens=0,
+                    output_token_ids=[],
+                )
+            )
         if out.is_empty:
-            return []
+            return results
         sampled = self.runner.execute(out, self.step_count)
         self.step_count += 1
         now = time.monotonic()
         finished = self.scheduler.finish_step(out, sampled, now)
         finished_ids = {r.request_id for r in finished}
-        results: list[RequestOutput] = []
         for ss in out.all_seqs:
             req = ss.req
             if not ss.samples:

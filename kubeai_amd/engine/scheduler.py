@@ -108,6 +108,8 @@ class SchedulerOutput:
     decode: list[ScheduledSeq]
     prefill: list[ScheduledSeq]
     preempted: list[Request]
+    # requests that can never fit the KV pool (rejected at admission)
+    rejected: list[Request] = dataclasses.field(default_factory=list)
 
     @property
     def all_seqs(self) -> list[ScheduledSeq]:
@@ -177,6 +179,19 @@ class Scheduler:
         decode: list[ScheduledSeq] = []
         prefill: list[ScheduledSeq] = []
         preempted: list[Request] = []
+        rejected: list[Request] = []
+        # liveness: a request larger than the ENTIRE pool would block the
+        # queue head forever — reject it outright
+        bs = self.bm.block_size
+        while self.waiting:
+            head = self.waiting[0]
+            need = (len(head.prompt_token_ids) + head.params.max_tokens + bs - 1) // bs
+            if need > self.bm.num_blocks:
+                head.status = RequestStatus.FINISHED_ABORTED
+                head.finish_time = time.monotonic()
+                rejected.append(self.waiting.popleft())
+            else:
+                break
         have_decodes = any(
             not r.in_prefill and r.status == RequestStatus.RUNNING
             for r in self.running
@@ -250,7 +265,9 @@ class Scheduler:
             prefill.append(ScheduledSeq(req, n_cached, chunk))
             budget -= chunk
 
-        return SchedulerOutput(decode=decode, prefill=prefill, preempted=preempted)
+        return SchedulerOutput(
+            decode=decode, prefill=prefill, preempted=preempted, rejected=rejected
+        )
 
     def _ensure_blocks(
         self, req: Request, needed_tokens: int, preempted: list[Request]

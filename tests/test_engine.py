@@ -182,3 +182,17 @@ def test_topk_topp_masked_sampling():
     for step in range(30):
         toks = ops.gumbel_sample(masked, temps, seeds, step)
         assert int(toks[1]) in nucleus
+
+
+def test_oversized_request_rejected_not_stuck():
+    """A request that can never fit the KV pool must be rejected with a
+    finished output instead of blocking the queue head forever."""
+    eng = make_engine(num_gpu_blocks=8)  # pool = 128 tokens
+    eng.add_request(list(range(5, 25)), SamplingParams(max_tokens=500),
+                    request_id="huge")
+    eng.add_request(list(range(5, 40)), SamplingParams(max_tokens=4),
+                    request_id="ok")
+    outs = run_to_completion(eng)
+    assert outs["huge"].finish_reason == "abort"
+    assert outs["huge"].output_token_ids == []
+    assert len(outs["ok"].output_token_ids) == 4

@@ -89,6 +89,9 @@ class ProxyHandler:
                     return JSONResponse(
                         {"error": "backend connection failed"}, status_code=502
                     )
+                except BaseException:
+                    done()  # never leak in-flight accounting
+                    raise
                 if resp.status_code in RETRYABLE and attempt <= self.max_retries:
                     await resp.aclose()
                     done()

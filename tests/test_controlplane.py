@@ -495,3 +495,32 @@ def test_messenger_flow():
                 await msgr.stop()
 
     run(body())
+
+
+def test_proxy_inflight_never_leaks_on_unexpected_error():
+    """Any exception between endpoint acquisition and response must release
+    the in-flight slot (the autoscaling signal would otherwise drift up)."""
+
+    async def body():
+        m = text_gen_model()
+        m.spec.replicas = 1
+        async with harness([m]) as (mgr, runtime, backend):
+            rep = (await wait_for(lambda: mgr.store.list_replicas("m1")))[0]
+            runtime.mark_ready(rep.name, backend.address)
+
+            async def boom(*a, **kw):
+                raise RuntimeError("injected")
+
+            mgr.proxy._forward = boom
+            transport = httpx.ASGITransport(app=mgr.app)
+            async with httpx.AsyncClient(transport=transport, base_url="http://gw") as client:
+                resp = await client.post(
+                    "/openai/v1/completions",
+                    json={"model": "m1", "prompt": "x", "max_tokens": 1},
+                )
+            assert resp.status_code == 500
+            await wait_for(lambda: mgr.lb.group("m1").total_in_flight == 0)
+            for ep in mgr.lb.group("m1").endpoints.values():
+                assert ep.in_flight == 0
+
+    run(body())

@@ -56,6 +56,7 @@ class RequestOutput:
     num_prompt_tokens: int
     num_cached_tokens: int
     output_token_ids: list[int]
+    logprob: Optional[float] = None  # logprob of new_token_ids[-1]
 
 
 class LLMEngine:
@@ -154,6 +155,9 @@ class LLMEngine:
                     num_prompt_tokens=req.num_prompt_tokens,
                     num_cached_tokens=req.num_cached_prompt_tokens,
                     output_token_ids=req.output_token_ids,
+                    logprob=getattr(self.runner, "last_logprobs", {}).get(
+                        req.request_id
+                    ),
                 )
             )
         return results

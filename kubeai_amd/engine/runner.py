@@ -415,7 +415,14 @@ class ModelRunner:
                 tokens = ops.gumbel_sample(masked.contiguous(), t_t, seeds, step).to(dev)
             else:
                 tokens = ops.gumbel_sample(logits.contiguous(), t_t, seeds, step)
+        # logprob of the chosen token: logit - logsumexp(row)
+        lse = torch.logsumexp(logits, dim=-1)
+        chosen = logits.gather(1, tokens.view(-1, 1)).squeeze(1)
+        logprobs = (chosen - lse).cpu().tolist()
         tokens = tokens.cpu().tolist()
+        self.last_logprobs = {
+            r.request_id: float(lp) for r, lp in zip(sample_reqs, logprobs)
+        }
         return {r.request_id: int(t) for r, t in zip(sample_reqs, tokens)}
 
 

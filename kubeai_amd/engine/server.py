@@ -296,12 +296,23 @@ def build_app(server: EngineServer) -> FastAPI:
                                    lora_id=lora_id, stops=stops),
                 media_type="text/event-stream",
             )
+        want_logprobs = body.get("logprobs") not in (None, False, 0)
         choices = []
         pt = ct = 0
         for i in range(n):
-            final, cut_text, cut_reason = await _finish_tokens(
-                server.generate(toks, params, lora_id), stops
-            )
+            lps: list = []
+            final = None
+            cut_text = cut_reason = None
+            async for o in server.generate(toks, params, lora_id):
+                final = o
+                if o.logprob is not None:
+                    lps.append(o.logprob)
+                if stops:
+                    text_probe = server.tokenizer.decode(o.output_token_ids)
+                    hits = [text_probe.find(s_) for s_ in stops if s_ in text_probe]
+                    if hits:
+                        cut_text, cut_reason = text_probe[: min(hits)], "stop"
+                        break
             text = (
                 cut_text
                 if cut_text is not None
@@ -312,7 +323,7 @@ def build_app(server: EngineServer) -> FastAPI:
                     "index": i,
                     "text": text,
                     "finish_reason": cut_reason or final.finish_reason or "stop",
-                    "logprobs": None,
+                    "logprobs": {"token_logprobs": lps} if want_logprobs else None,
                 }
             )
             pt = final.num_prompt_tokens

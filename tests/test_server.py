@@ -255,3 +255,15 @@ def test_embeddings(client):
     assert len(v) == 256  # llama-tiny hidden size
     norm = sum(x * x for x in v) ** 0.5
     assert abs(norm - 1.0) < 1e-3
+
+
+def test_logprobs(client):
+    r = client.post(
+        "/v1/completions",
+        json={"prompt": "log probs please", "max_tokens": 4, "temperature": 0,
+              "logprobs": 1},
+    )
+    assert r.status_code == 200
+    lp = r.json()["choices"][0]["logprobs"]["token_logprobs"]
+    assert len(lp) == 4
+    assert all(v <= 0 for v in lp)

@@ -124,13 +124,14 @@ class ModelRunner:
             self.n_kv_local = arch.num_key_value_heads
         self.model.eval()
         if model_path is not None and os.path.isdir(model_path):
-            from kubeai_amd.models.loader import load_weights
+            if tp_group is not None:
+                from kubeai_amd.models.loader import load_weights_tp
 
-            if tp_group is not None and tp_group.world > 1:
-                raise NotImplementedError(
-                    "TP checkpoint loading lands with the TP shard loader"
-                )
-            load_weights(self.model, model_path)
+                load_weights_tp(self.model, model_path)
+            else:
+                from kubeai_amd.models.loader import load_weights
+
+                load_weights(self.model, model_path)
         self.quantization = quantization
         if quantization == "fp8":
             if self.device.type != "cuda":

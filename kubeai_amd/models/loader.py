@@ -34,6 +34,103 @@ def _strip(name: str) -> str:
     return name
 
 
+def load_weights_tp(model, model_dir: str) -> int:
+    """Load HF weights into TPLlamaForCausalLM: every rank reads the full
+    tensors and keeps its shard (same slicing as parallel/tp.py init)."""
+    cfg = model.cfg
+    tp = model.tp
+    hd = cfg.head_dim
+    nq, nkv = cfg.num_attention_heads, cfg.num_key_value_heads
+    nq_l, nkv_l = nq // tp.world, nkv // tp.world
+    i_l = cfg.intermediate_size // tp.world
+    params = dict(model.named_parameters())
+    filled: set[str] = set()
+
+    def put(target: str, tensor: torch.Tensor) -> None:
+        p = params[target]
+        if p.shape != tensor.shape:
+            raise ValueError(
+                f"{target}: shape {tuple(tensor.shape)} != {tuple(p.shape)}"
+            )
+        with torch.no_grad():
+            p.copy_(tensor.to(p.dtype))
+        filled.add(target)
+
+    def shard_rows_heads(w, n_heads, n_local):
+        return w.view(n_heads, hd, -1)[
+            tp.rank * n_local : (tp.rank + 1) * n_local
+        ].reshape(n_local * hd, -1)
+
+    pending: dict[str, dict[str, torch.Tensor]] = {}
+
+    def fuse_qkv(layer: int) -> None:
+        ps = pending.get(f"qkv.{layer}", {})
+        if len(ps) == 3:
+            q = shard_rows_heads(ps["q"], nq, nq_l)
+            k = shard_rows_heads(ps["k"], nkv, nkv_l)
+            v = shard_rows_heads(ps["v"], nkv, nkv_l)
+            put(f"layers.{layer}.self_attn.qkv_proj.weight",
+                torch.cat([q, k, v], dim=0))
+
+    def fuse_gate_up(layer: int) -> None:
+        ps = pending.get(f"gu.{layer}", {})
+        if len(ps) == 2:
+            g = ps["gate"][tp.rank * i_l : (tp.rank + 1) * i_l]
+            u = ps["up"][tp.rank * i_l : (tp.rank + 1) * i_l]
+            put(f"layers.{layer}.mlp.gate_up_proj.weight", torch.cat([g, u], dim=0))
+
+    for name, w in iter_safetensors(model_dir):
+        n = _strip(name)
+        if n == "embed_tokens.weight":
+            put("embed_tokens.weight", w)
+        elif n == "lm_head.weight":
+            if model.lm_head is not None:
+                put("lm_head.weight", w)
+        elif n == "norm.weight":
+            put("norm", w)
+        elif n.startswith("layers."):
+            parts = n.split(".")
+            layer = int(parts[1])
+            rest = ".".join(parts[2:])
+            if rest == "input_layernorm.weight":
+                put(f"layers.{layer}.input_layernorm", w)
+            elif rest == "post_attention_layernorm.weight":
+                put(f"layers.{layer}.post_attention_layernorm", w)
+            elif rest == "self_attn.q_proj.weight":
+                pending.setdefault(f"qkv.{layer}", {})["q"] = w
+                fuse_qkv(layer)
+            elif rest == "self_attn.k_proj.weight":
+                pending.setdefault(f"qkv.{layer}", {})["k"] = w
+                fuse_qkv(layer)
+            elif rest == "self_attn.v_proj.weight":
+                pending.setdefault(f"qkv.{layer}", {})["v"] = w
+                fuse_qkv(layer)
+            elif rest == "self_attn.o_proj.weight":
+                put(
+                    f"layers.{layer}.self_attn.o_proj.weight",
+                    w[:, tp.rank * nq_l * hd : (tp.rank + 1) * nq_l * hd],
+                )
+            elif rest == "mlp.gate_proj.weight":
+                pending.setdefault(f"gu.{layer}", {})["gate"] = w
+                fuse_gate_up(layer)
+            elif rest == "mlp.up_proj.weight":
+                pending.setdefault(f"gu.{layer}", {})["up"] = w
+                fuse_gate_up(layer)
+            elif rest == "mlp.down_proj.weight":
+                put(
+                    f"layers.{layer}.mlp.down_proj.weight",
+                    w[:, tp.rank * i_l : (tp.rank + 1) * i_l],
+                )
+    if model.lm_head is not None and "lm_head.weight" not in filled:
+        with torch.no_grad():
+            model.lm_head.weight.copy_(model.embed_tokens.weight)
+        filled.add("lm_head.weight")
+    missing = set(params) - filled
+    if missing:
+        raise ValueError(f"unfilled parameters after TP load: {sorted(missing)[:8]}")
+    return len(filled)
+
+
 def load_weights(model, model_dir: str) -> int:
     """Load HF weights into LlamaForCausalLM (dense or MoE). Returns the
     number of engine parameters filled; raises if any stays unset."""

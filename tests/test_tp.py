@@ -34,7 +34,7 @@ def _generate(eng):
     raise AssertionError("did not finish")
 
 
-def _tp_worker(rank, world, port, q):
+def _tp_worker(rank, world, port, q, model="llama-tiny-tp", save_ckpt=None):
     import torch.distributed as dist
 
     from kubeai_amd.parallel.tp import TPGroup
@@ -48,7 +48,7 @@ def _tp_worker(rank, world, port, q):
     try:
         eng = LLMEngine(
             EngineConfig(
-                model="llama-tiny-tp",
+                model=model,
                 device="cpu",
                 dtype="float32",
                 num_gpu_blocks=64,
@@ -57,6 +57,10 @@ def _tp_worker(rank, world, port, q):
             ),
             tp_group=TPGroup(),
         )
+        if save_ckpt is not None and rank == 0:
+            from kubeai_amd.models.loader import save_hf_checkpoint
+
+            save_hf_checkpoint(eng.runner.model, save_ckpt)
         toks = _generate(eng)
         if rank == 0:
             q.put(("ok", toks))
@@ -70,12 +74,13 @@ def _tp_worker(rank, world, port, q):
         dist.destroy_process_group()
 
 
-def run_tp(world: int):
+def run_tp(world: int, model="llama-tiny-tp", save_ckpt=None):
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
     port = free_port()
     procs = [
-        ctx.Process(target=_tp_worker, args=(r, world, port, q)) for r in range(world)
+        ctx.Process(target=_tp_worker, args=(r, world, port, q, model, save_ckpt))
+        for r in range(world)
     ]
     for p in procs:
         p.start()
@@ -91,3 +96,12 @@ def test_tp2_matches_tp1():
     t2 = run_tp(2)
     assert len(t1) == N_TOKENS
     assert t1 == t2, f"tp1={t1} tp2={t2}"
+
+
+def test_tp2_checkpoint_loading(tmp_path):
+    """Shard-aware HF checkpoint loading: a checkpoint written at TP=1 must
+    reproduce greedy generation when loaded sharded across TP=2 ranks."""
+    ckpt = str(tmp_path / "ckpt")
+    ref = run_tp(1, save_ckpt=ckpt)
+    sharded = run_tp(2, model=ckpt)
+    assert ref == sharded, f"tp1={ref} tp2(ckpt)={sharded}"

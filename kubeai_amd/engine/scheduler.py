@@ -28,6 +28,10 @@ class SamplingParams:
     # OpenAI penalties over generated tokens (vLLM semantics)
     presence_penalty: float = 0.0
     frequency_penalty: float = 0.0
+    # admission priority: LOWER value is served first (vLLM semantics);
+    # FIFO within a priority class. Maps the Model CRD's priorityClassName
+    # analog onto per-request scheduling.
+    priority: int = 0
 
 
 class RequestStatus(enum.Enum):
@@ -155,6 +159,18 @@ class Scheduler:
             req.params.max_tokens = max(
                 1, self.max_model_len - len(req.prompt_token_ids)
             )
+        self._insert_waiting(req, retry=False)
+
+    def _insert_waiting(self, req: Request, retry: bool) -> None:
+        """Priority-ordered insert: before the first request of a strictly
+        worse class (FIFO within a class); a preempted retry goes to the
+        FRONT of its own class (recompute-first policy)."""
+        p = req.params.priority
+        for i, other in enumerate(self.waiting):
+            op = other.params.priority
+            if op > p or (retry and op == p):
+                self.waiting.insert(i, req)
+                return
         self.waiting.append(req)
 
     def abort(self, request_id: str) -> None:
@@ -289,13 +305,16 @@ class Scheduler:
         return True
 
     def _pick_victim(self, requester: Request) -> Optional[Request]:
-        # preempt the youngest running request that is not the requester and
-        # has not already been scheduled in this step (its ScheduledSeq would
-        # otherwise reference freed blocks)
-        for req in reversed(self.running):
+        # preempt the worst-priority, youngest running request that is not
+        # the requester and has not already been scheduled in this step (its
+        # ScheduledSeq would otherwise reference freed blocks)
+        victim = None
+        for pos, req in enumerate(self.running):
             if req is not requester and req.request_id not in self._scheduled_ids:
-                return req
-        return None
+                key = (req.params.priority, pos)
+                if victim is None or key > victim[0]:
+                    victim = (key, req)
+        return victim[1] if victim else None
 
     def _preempt(self, req: Request, preempted: list[Request]) -> None:
         self.bm.free(req.block_table)
@@ -304,7 +323,7 @@ class Scheduler:
         req.num_computed = 0
         req.status = RequestStatus.WAITING
         self.running.remove(req)
-        self.waiting.appendleft(req)  # retry first (recompute policy)
+        self._insert_waiting(req, retry=True)
         preempted.append(req)
 
     # ----------------------------------------------------------------

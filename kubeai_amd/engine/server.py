@@ -213,6 +213,7 @@ def build_app(server: EngineServer) -> FastAPI:
             ignore_eos=bool(body.get("ignore_eos") or False),
             presence_penalty=float(body.get("presence_penalty") or 0.0),
             frequency_penalty=float(body.get("frequency_penalty") or 0.0),
+            priority=int(body.get("priority") or 0),
         )
 
     def _stop_strings(body: dict) -> list[str]:

@@ -65,3 +65,53 @@ def test_interval_defers_admission():
     eng.scheduler.finish_step(out4, sampled)
     while eng.has_work():
         eng.step()
+
+def test_priority_admission_order():
+    """Lower priority value is admitted first; FIFO within a class."""
+    from kubeai_amd.engine.kvcache import BlockManager
+    from kubeai_amd.engine.scheduler import Request, Scheduler
+
+    bm = BlockManager(num_blocks=64, block_size=16)
+    # max_num_seqs=1 so admission order is directly observable
+    s = Scheduler(bm, max_num_seqs=1, max_num_batched_tokens=64, max_model_len=256)
+    order = []
+    for rid, prio in [("low1", 5), ("hi", 0), ("low2", 5), ("mid", 2)]:
+        s.add_request(
+            Request(list(range(10, 26)), SamplingParams(max_tokens=1, priority=prio),
+                    request_id=rid)
+        )
+    assert [r.request_id for r in s.waiting] == ["hi", "mid", "low1", "low2"]
+    while s.has_work():
+        out = s.schedule()
+        for ss in out.all_seqs:
+            if ss.req.request_id not in order:
+                order.append(ss.req.request_id)
+        s.finish_step(out, {ss.req.request_id: 3 for ss in out.all_seqs if ss.samples})
+    assert order == ["hi", "mid", "low1", "low2"]
+
+
+def test_priority_preemption_victim():
+    """Under block pressure the worst-priority running request is evicted."""
+    from kubeai_amd.engine.kvcache import BlockManager
+    from kubeai_amd.engine.scheduler import Request, Scheduler
+
+    bm = BlockManager(num_blocks=6, block_size=16)
+    s = Scheduler(bm, max_num_seqs=8, max_num_batched_tokens=512,
+                  max_model_len=256, enable_prefix_caching=False)
+    # two running requests: important (prio 0) and best-effort (prio 9),
+    # each holding 2 blocks with 6 total; decoding pushes past block
+    # boundaries until someone must be evicted
+    s.add_request(Request(list(range(10, 41)), SamplingParams(max_tokens=64, priority=0),
+                          request_id="vip"))
+    s.add_request(Request(list(range(60, 91)), SamplingParams(max_tokens=64, priority=9),
+                          request_id="bulk"))
+    preempted = []
+    for _ in range(80):
+        out = s.schedule()
+        preempted += [r.request_id for r in out.preempted]
+        if out.is_empty:
+            break
+        s.finish_step(out, {ss.req.request_id: 3 for ss in out.all_seqs if ss.samples})
+        if preempted:
+            break
+    assert preempted and preempted[0] == "bulk"

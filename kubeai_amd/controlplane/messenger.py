@@ -4,15 +4,22 @@ Message {metadata, path, body} -> parse -> scale-at-least-one -> await
 endpoint -> POST to engine -> publish {metadata, status_code, body} with
 ack/nack, bounded handler concurrency, consecutive-error backoff.
 
-Drivers: mem:// (tests/local; the reference registers SQS/PubSub/Kafka/...
-through gocloud — those transports are deployment concerns; the mem driver
-exercises the full flow, reference tests do the same with mem://).
+Drivers are selected by URL scheme like the reference's gocloud wiring
+(internal/messenger registers mem/SQS/PubSub/Kafka by URL): `mem://topic`
+is in-process (tests/local, reference tests use the same) and
+`file:///dir/topic` is a durable cross-process queue over a shared
+directory (single-node multi-process deployments; survives restarts).
+Cloud transports (SQS/Kafka) are deployment concerns behind the same
+Broker protocol.
 """
 from __future__ import annotations
 
 import asyncio
 import json
+import os
+import tempfile
 from typing import Optional
+from urllib.parse import urlparse
 
 import httpx
 
@@ -41,10 +48,69 @@ class MemBroker:
         return await self.topic(topic).get()
 
 
+class FileBroker:
+    """Durable cross-process pubsub over a shared directory (file:// driver).
+
+    A topic is a directory; a message is one file, written atomically
+    (tmp + rename within the topic dir). A consumer CLAIMS a message by
+    renaming it to *.claimed.<pid> — rename is atomic on POSIX, so exactly
+    one of N competing consumer processes wins each message.
+    """
+
+    def __init__(self, root: str, poll_interval: float = 0.05):
+        self.root = root
+        self.poll_interval = poll_interval
+        self._seq = 0
+        os.makedirs(root, exist_ok=True)
+
+    def _topic_dir(self, topic: str) -> str:
+        d = os.path.join(self.root, topic.strip("/").replace("/", "_"))
+        os.makedirs(d, exist_ok=True)
+        return d
+
+    async def publish(self, topic: str, payload: bytes) -> None:
+        d = self._topic_dir(topic)
+        self._seq += 1
+        # monotonic-ish lexicographic name: (coarse clock, pid, seq)
+        name = f"{int(asyncio.get_event_loop().time() * 1e6):018d}-{os.getpid()}-{self._seq:06d}.msg"
+        fd, tmp = tempfile.mkstemp(dir=d, suffix=".tmp")
+        with os.fdopen(fd, "wb") as f:
+            f.write(payload)
+        os.replace(tmp, os.path.join(d, name))
+
+    async def receive(self, topic: str) -> bytes:
+        d = self._topic_dir(topic)
+        while True:
+            for name in sorted(os.listdir(d)):
+                if not name.endswith(".msg"):
+                    continue
+                src = os.path.join(d, name)
+                claimed = f"{src}.claimed.{os.getpid()}"
+                try:
+                    os.rename(src, claimed)  # atomic claim; loser gets ENOENT
+                except OSError:
+                    continue
+                with open(claimed, "rb") as f:
+                    payload = f.read()
+                os.unlink(claimed)
+                return payload
+            await asyncio.sleep(self.poll_interval)
+
+
+def broker_from_url(url: str):
+    """gocloud-style driver selection: mem://topic-prefix | file:///dir."""
+    u = urlparse(url)
+    if u.scheme in ("", "mem"):
+        return MemBroker()
+    if u.scheme == "file":
+        return FileBroker(u.path or "/tmp/kubeai-msgs")
+    raise ValueError(f"unknown messenger driver: {u.scheme}:// (have mem, file)")
+
+
 class Messenger:
     def __init__(
         self,
-        broker: MemBroker,
+        broker,  # any Broker driver (MemBroker / FileBroker)
         requests_topic: str,
         responses_topic: str,
         model_client: ModelClient,

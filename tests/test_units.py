@@ -246,3 +246,48 @@ def test_duration_parsing(tmp_path):
     cfg = load_config(str(p))
     assert cfg.autoscaling.interval_seconds == 10.0
     assert cfg.autoscaling.time_window_seconds == 600.0
+
+
+def test_file_broker_roundtrip(tmp_path):
+    import asyncio
+
+    from kubeai_amd.controlplane.messenger import FileBroker, broker_from_url
+
+    async def run():
+        b = FileBroker(str(tmp_path / "q"))
+        for i in range(3):
+            await b.publish("requests", f"msg{i}".encode())
+        got = [await b.receive("requests") for _ in range(3)]
+        assert got == [b"msg0", b"msg1", b"msg2"]
+
+    asyncio.run(run())
+    assert type(broker_from_url("mem://x")).__name__ == "MemBroker"
+    assert type(broker_from_url(f"file://{tmp_path}/q2")).__name__ == "FileBroker"
+    import pytest
+
+    with pytest.raises(ValueError, match="unknown messenger driver"):
+        broker_from_url("sqs://queue")
+
+
+def test_file_broker_exactly_once_across_consumers(tmp_path):
+    """Two competing consumers on one topic: every message delivered once."""
+    import asyncio
+
+    from kubeai_amd.controlplane.messenger import FileBroker
+
+    async def run():
+        prod = FileBroker(str(tmp_path / "q"), poll_interval=0.01)
+        c1 = FileBroker(str(tmp_path / "q"), poll_interval=0.01)
+        c2 = FileBroker(str(tmp_path / "q"), poll_interval=0.01)
+        for i in range(20):
+            await prod.publish("t", f"{i}".encode())
+        got = []
+
+        async def drain(c, n):
+            for _ in range(n):
+                got.append(await c.receive("t"))
+
+        await asyncio.gather(drain(c1, 10), drain(c2, 10))
+        assert sorted(int(g) for g in got) == list(range(20))
+
+    asyncio.run(run())

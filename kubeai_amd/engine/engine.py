@@ -57,6 +57,8 @@ class RequestOutput:
     num_cached_tokens: int
     output_token_ids: list[int]
     logprob: Optional[float] = None  # logprob of new_token_ids[-1]
+    # top alternatives for new_token_ids[-1]: [(token_id, logprob), ...]
+    top_logprobs: Optional[list] = None
 
 
 class LLMEngine:
@@ -156,6 +158,9 @@ class LLMEngine:
                     num_cached_tokens=req.num_cached_prompt_tokens,
                     output_token_ids=req.output_token_ids,
                     logprob=getattr(self.runner, "last_logprobs", {}).get(
+                        req.request_id
+                    ),
+                    top_logprobs=getattr(self.runner, "last_top_logprobs", {}).get(
                         req.request_id
                     ),
                 )

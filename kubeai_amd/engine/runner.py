@@ -424,6 +424,19 @@ class ModelRunner:
         self.last_logprobs = {
             r.request_id: float(lp) for r, lp in zip(sample_reqs, logprobs)
         }
+        # top-alternative logprobs, only when some request asked for them
+        self.last_top_logprobs = {}
+        want = max((r.params.logprobs for r in sample_reqs), default=0)
+        if want > 0:
+            k = min(want, logits.shape[-1])
+            tv, ti = (logits - lse.unsqueeze(1)).topk(k, dim=-1)
+            tv, ti = tv.cpu().tolist(), ti.cpu().tolist()
+            for r, vs, ids in zip(sample_reqs, tv, ti):
+                if r.params.logprobs > 0:
+                    n = r.params.logprobs
+                    self.last_top_logprobs[r.request_id] = [
+                        (int(i), float(v)) for i, v in zip(ids[:n], vs[:n])
+                    ]
         return {r.request_id: int(t) for r, t in zip(sample_reqs, tokens)}
 
 

@@ -28,6 +28,9 @@ class SamplingParams:
     # OpenAI penalties over generated tokens (vLLM semantics)
     presence_penalty: float = 0.0
     frequency_penalty: float = 0.0
+    # number of top-alternative logprobs to return per sampled token
+    # (OpenAI `logprobs`/`top_logprobs`); 0 = none
+    logprobs: int = 0
     # admission priority: LOWER value is served first (vLLM semantics);
     # FIFO within a priority class. Maps the Model CRD's priorityClassName
     # analog onto per-request scheduling.

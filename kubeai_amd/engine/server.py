@@ -254,13 +254,16 @@ def build_app(server: EngineServer) -> FastAPI:
             data.append({"id": a, "object": "model", "owned_by": "kubeai-amd", "parent": name})
         return {"object": "list", "data": data}
 
-    async def _finish_tokens(gen, stops=None):
+    async def _finish_tokens(gen, stops=None, steps=None):
         """Drain a generation; with stop strings, truncate at the earliest
         match and abort the engine request early (the generator's finally
-        clause issues the abort)."""
+        clause issues the abort). `steps`, when given, collects every
+        per-token RequestOutput (chat logprobs)."""
         final = None
         async for o in gen:
             final = o
+            if steps is not None:
+                steps.append(o)
             if stops:
                 # full re-decode: incremental piece concatenation loses
                 # inter-token separators (tokenizer-dependent)
@@ -298,16 +301,23 @@ def build_app(server: EngineServer) -> FastAPI:
                 media_type="text/event-stream",
             )
         want_logprobs = body.get("logprobs") not in (None, False, 0)
+        if want_logprobs:
+            # completions `logprobs: N` = N top alternatives per token
+            params.logprobs = int(body.get("logprobs") or 0)
         choices = []
         pt = ct = 0
         for i in range(n):
             lps: list = []
+            toks_out: list = []
+            tops: list = []
             final = None
             cut_text = cut_reason = None
             async for o in server.generate(toks, params, lora_id):
                 final = o
                 if o.logprob is not None:
                     lps.append(o.logprob)
+                    toks_out.append(o.new_token_ids[-1])
+                    tops.append(o.top_logprobs)
                 if stops:
                     text_probe = server.tokenizer.decode(o.output_token_ids)
                     hits = [text_probe.find(s_) for s_ in stops if s_ in text_probe]
@@ -319,12 +329,22 @@ def build_app(server: EngineServer) -> FastAPI:
                 if cut_text is not None
                 else server.tokenizer.decode(_strip_stop(final, params))
             )
+            lp_obj = None
+            if want_logprobs:
+                dec = server.tokenizer.decode
+                lp_obj = {
+                    "tokens": [dec([t]) for t in toks_out],
+                    "token_logprobs": lps,
+                    "top_logprobs": [
+                        {dec([tid]): lp for tid, lp in (top or [])} for top in tops
+                    ],
+                }
             choices.append(
                 {
                     "index": i,
                     "text": text,
                     "finish_reason": cut_reason or final.finish_reason or "stop",
-                    "logprobs": {"token_logprobs": lps} if want_logprobs else None,
+                    "logprobs": lp_obj,
                 }
             )
             pt = final.num_prompt_tokens
@@ -358,24 +378,44 @@ def build_app(server: EngineServer) -> FastAPI:
                                    lora_id=lora_id, stops=stops),
                 media_type="text/event-stream",
             )
+        want_logprobs = bool(body.get("logprobs"))
+        if want_logprobs:
+            # chat `logprobs: true` + optional `top_logprobs: N`
+            params.logprobs = int(body.get("top_logprobs") or 0)
         choices = []
         pt = ct = 0
         for i in range(n):
+            steps: list = [] if want_logprobs else None
             final, cut_text, cut_reason = await _finish_tokens(
-                server.generate(toks, params, lora_id), stops
+                server.generate(toks, params, lora_id), stops, steps
             )
             text = (
                 cut_text
                 if cut_text is not None
                 else server.tokenizer.decode(_strip_stop(final, params))
             )
-            choices.append(
-                {
-                    "index": i,
-                    "message": {"role": "assistant", "content": text},
-                    "finish_reason": cut_reason or final.finish_reason or "stop",
+            choice = {
+                "index": i,
+                "message": {"role": "assistant", "content": text},
+                "finish_reason": cut_reason or final.finish_reason or "stop",
+            }
+            if want_logprobs:
+                dec = server.tokenizer.decode
+                choice["logprobs"] = {
+                    "content": [
+                        {
+                            "token": dec([o.new_token_ids[-1]]),
+                            "logprob": o.logprob,
+                            "top_logprobs": [
+                                {"token": dec([tid]), "logprob": lp}
+                                for tid, lp in (o.top_logprobs or [])
+                            ],
+                        }
+                        for o in steps
+                        if o.logprob is not None
+                    ]
                 }
-            )
+            choices.append(choice)
             pt = final.num_prompt_tokens
             ct += len(final.output_token_ids)
         return {

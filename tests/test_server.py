@@ -261,9 +261,32 @@ def test_logprobs(client):
     r = client.post(
         "/v1/completions",
         json={"prompt": "log probs please", "max_tokens": 4, "temperature": 0,
-              "logprobs": 1},
+              "logprobs": 2},
     )
     assert r.status_code == 200
-    lp = r.json()["choices"][0]["logprobs"]["token_logprobs"]
+    obj = r.json()["choices"][0]["logprobs"]
+    lp = obj["token_logprobs"]
     assert len(lp) == 4
     assert all(v <= 0 for v in lp)
+    assert len(obj["tokens"]) == 4
+    # 2 alternatives per position; greedy choice == best alternative
+    assert all(len(top) == 2 for top in obj["top_logprobs"])
+    for chosen_lp, top in zip(lp, obj["top_logprobs"]):
+        assert abs(max(top.values()) - chosen_lp) < 1e-4
+
+
+def test_chat_logprobs(client):
+    r = client.post(
+        "/v1/chat/completions",
+        json={"model": "test-model",
+              "messages": [{"role": "user", "content": "chat logprobs"}],
+              "max_tokens": 3, "temperature": 0,
+              "logprobs": True, "top_logprobs": 2},
+    )
+    assert r.status_code == 200
+    content = r.json()["choices"][0]["logprobs"]["content"]
+    assert len(content) == 3
+    for item in content:
+        assert item["logprob"] <= 0
+        assert len(item["top_logprobs"]) == 2
+        assert item["top_logprobs"][0]["logprob"] >= item["top_logprobs"][1]["logprob"]

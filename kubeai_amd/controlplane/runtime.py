@@ -148,6 +148,8 @@ class LocalProcessRuntime:
             "--port",
             str(port),
         ] + list(model.spec.args)
+        if "SpeechToText" in model.spec.features and "--task" not in model.spec.args:
+            cmd += ["--task", "transcribe"]
         if not ids:
             cmd += ["--device", "cpu"]
         proc = subprocess.Popen(cmd, env=env)

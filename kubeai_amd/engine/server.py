@@ -19,6 +19,7 @@ import asyncio
 import json
 import os
 import queue
+import re
 import threading
 import time
 import uuid
@@ -57,13 +58,15 @@ class EngineServer:
     """
 
     def __init__(self, cfg: EngineConfig, served_model_name: str, tp_size: int = 1,
-                 tp_port: int = 0):
+                 tp_port: int = 0, task: str = "generate"):
         self.cfg = cfg
         self.served_model_name = served_model_name
         self.tp_size = tp_size
         self.tp_port = tp_port
+        self.task = task
         self._dist = None
         self.engine: Optional[LLMEngine] = None
+        self.stt = None  # SpeechToText model when task == "transcribe"
         self.tokenizer = None
         self._submit: "queue.Queue" = queue.Queue()
         self._events: dict[str, tuple[asyncio.AbstractEventLoop, asyncio.Queue]] = {}
@@ -83,6 +86,20 @@ class EngineServer:
 
     def _run(self) -> None:
         # heavy init inside the thread so /health can answer "starting"
+        if self.task == "transcribe":
+            from kubeai_amd.models.whisper import PRESETS as STT_PRESETS
+            from kubeai_amd.models.whisper import SpeechToText
+
+            scfg = STT_PRESETS.get(self.cfg.model, STT_PRESETS["whisper-tiny"])
+            self.stt = SpeechToText(
+                scfg, device=self.cfg.resolve_device(), seed=self.cfg.seed
+            )
+            self.tokenizer = load_tokenizer(
+                self.cfg.model, scfg.vocab_size, scfg.sot_token, scfg.eot_token
+            )
+            self._ready.set()
+            self._stop.wait()  # transcription is request-driven, no stepping
+            return
         tp_group = None
         if self.tp_size > 1 or os.environ.get("KUBEAI_FORCE_TP"):
             import torch.distributed as dist
@@ -199,6 +216,14 @@ def build_app(server: EngineServer) -> FastAPI:
     app = FastAPI()
     name = server.served_model_name
 
+    def _generate_unsupported():
+        return JSONResponse(
+            {"error": {"message": f"model {name} is a speech-to-text model "
+                                  "(task=transcribe); text endpoints are "
+                                  "not supported"}},
+            status_code=400,
+        )
+
     def _params_from(body: dict) -> SamplingParams:
         mt = body.get("max_tokens") or body.get("max_completion_tokens") or 128
         temp = body.get("temperature")
@@ -276,6 +301,8 @@ def build_app(server: EngineServer) -> FastAPI:
 
     @app.post("/v1/completions")
     async def completions(request: Request):
+        if server.stt is not None:
+            return _generate_unsupported()
         body = await request.json()
         params = _params_from(body)
         prompt = body.get("prompt", "")
@@ -360,6 +387,8 @@ def build_app(server: EngineServer) -> FastAPI:
 
     @app.post("/v1/chat/completions")
     async def chat_completions(request: Request):
+        if server.stt is not None:
+            return _generate_unsupported()
         body = await request.json()
         params = _params_from(body)
         toks = apply_chat_template(server.tokenizer, body.get("messages", []))
@@ -429,6 +458,8 @@ def build_app(server: EngineServer) -> FastAPI:
 
     @app.post("/v1/embeddings")
     async def embeddings(request: Request):
+        if server.stt is not None:
+            return _generate_unsupported()
         body = await request.json()
         inputs = body.get("input", [])
         if isinstance(inputs, str):
@@ -452,6 +483,8 @@ def build_app(server: EngineServer) -> FastAPI:
     async def rerank(request: Request):
         """Score documents against a query (Reranking feature; reference
         analog: the Infinity engine's /rerank)."""
+        if server.stt is not None:
+            return _generate_unsupported()
         body = await request.json()
         query = body.get("query", "")
         docs = body.get("documents", []) or []
@@ -477,14 +510,49 @@ def build_app(server: EngineServer) -> FastAPI:
 
     @app.post("/v1/audio/transcriptions")
     async def transcriptions(request: Request):
-        # SpeechToText is a control-plane-routable feature; this engine
-        # build has no audio frontend (no whisper weights offline) — return
-        # the standard error shape rather than fake text.
-        return JSONResponse(
-            {"error": {"message": f"model {name} does not support audio "
-                                  "transcription in this build"}},
-            status_code=501,
+        # SpeechToText feature (reference routes it to the FasterWhisper
+        # engine; here the in-house whisper-architecture model serves it,
+        # kubeai_amd/models/whisper.py). Text-generation models answer 501.
+        if server.stt is None:
+            return JSONResponse(
+                {"error": {"message": f"model {name} does not support audio "
+                                      "transcription (task=generate)"}},
+                status_code=501,
+            )
+        ctype = request.headers.get("content-type", "")
+        raw = await request.body()
+        if not ctype.startswith("multipart/form-data"):
+            return JSONResponse(
+                {"error": "expected multipart/form-data with a `file` field"},
+                status_code=400,
+            )
+        fields = _parse_multipart(raw, ctype)
+        wav = fields.get("file")
+        if wav is None:
+            return JSONResponse({"error": "missing `file` field"}, status_code=400)
+        try:
+            audio, sr = _decode_wav(wav)
+        except Exception as e:  # noqa: BLE001
+            return JSONResponse({"error": f"cannot decode audio: {e}"},
+                                status_code=400)
+        loop = asyncio.get_running_loop()
+        toks = await loop.run_in_executor(
+            None, server.stt.transcribe_tokens, audio, sr
         )
+        # strip sot/eot specials before decoding
+        cfg_ = server.stt.cfg
+        text = server.tokenizer.decode(
+            [t for t in toks if t not in (cfg_.sot_token, cfg_.eot_token)]
+        )
+        fmt = (fields.get("response_format") or b"json").decode(errors="replace")
+        duration = len(audio) / sr if sr else 0.0
+        if fmt == "text":
+            return PlainTextResponse(text)
+        out = {"text": text}
+        if fmt == "verbose_json":
+            out.update({"task": "transcribe", "duration": duration,
+                        "language": fields.get("language", b"en").decode()})
+        return out
 
     @app.post("/v1/load_lora_adapter")
     async def load_lora(request: Request):
@@ -516,6 +584,43 @@ def build_app(server: EngineServer) -> FastAPI:
         return PlainTextResponse("OK")
 
     return app
+
+
+def _parse_multipart(raw: bytes, content_type: str) -> dict[str, bytes]:
+    """Minimal multipart/form-data parser (no python-multipart in this
+    image): returns {field-name: body-bytes}; good for one file + a few
+    small text fields, which is the transcriptions contract."""
+    m = re.search(r'boundary="?([^";,]+)"?', content_type)
+    if not m:
+        return {}
+    boundary = b"--" + m.group(1).encode()
+    fields: dict[str, bytes] = {}
+    for part in raw.split(boundary):
+        part = part.strip(b"\r\n")
+        if not part or part == b"--":
+            continue
+        head, _, body = part.partition(b"\r\n\r\n")
+        nm = re.search(rb'name="([^"]+)"', head)
+        if nm:
+            fields[nm.group(1).decode(errors="replace")] = body
+    return fields
+
+
+def _decode_wav(data: bytes):
+    """WAV bytes -> (float32 mono ndarray in [-1,1], sample_rate)."""
+    import io
+
+    import numpy as np
+    from scipy.io import wavfile
+
+    sr, audio = wavfile.read(io.BytesIO(data))
+    if audio.dtype.kind == "i":
+        audio = audio.astype(np.float32) / float(np.iinfo(audio.dtype).max)
+    elif audio.dtype.kind == "u":  # u8 wav
+        audio = (audio.astype(np.float32) - 128.0) / 128.0
+    else:
+        audio = audio.astype(np.float32)
+    return audio, sr
 
 
 def _strip_stop(final, params: SamplingParams) -> list[int]:
@@ -642,7 +747,13 @@ def main():
     p.add_argument("--num-gpu-blocks", type=int, default=None)
     p.add_argument("--enable-lora", action="store_true")
     p.add_argument("--tensor-parallel-size", type=int, default=1)
+    p.add_argument("--task", choices=["generate", "transcribe"], default=None,
+                   help="auto: whisper-* presets transcribe, others generate")
     args = p.parse_args()
+    task = args.task or (
+        "transcribe" if os.path.basename(args.model).startswith("whisper")
+        else "generate"
+    )
 
     cfg = EngineConfig(
         model=args.model,
@@ -673,7 +784,7 @@ def main():
             workers.append(p)
         if cfg.device in ("auto", "cuda"):
             cfg = EngineConfig(**{**cfg.__dict__, "device": "cuda:0"})
-    server = EngineServer(cfg, served, tp_size=tp, tp_port=tp_port)
+    server = EngineServer(cfg, served, tp_size=tp, tp_port=tp_port, task=task)
     server.start()
     app = build_app(server)
     import uvicorn

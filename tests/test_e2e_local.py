@@ -69,3 +69,63 @@ def test_local_runtime_end_to_end():
             await mgr.stop()
 
     asyncio.run(body())
+
+
+def test_local_runtime_speech_to_text():
+    """SpeechToText model: runtime launches the engine with task=transcribe;
+    the gateway proxies a multipart transcription request end to end."""
+    import io
+
+    import numpy as np
+    from scipy.io import wavfile
+
+    async def body():
+        cfg = SystemConfig(
+            autoscaling=AutoscalingConfig(
+                interval_seconds=3600, time_window_seconds=7200, state_path=None
+            ),
+            leader_lock_path="/tmp/kubeai-e2e-stt-leader.lock",
+            n_gpus=0,
+        )
+        mgr = Manager(cfg)
+        m = Model(
+            name="stt",
+            spec=ModelSpec(
+                url="hf://test/whisper-tiny",
+                features=["SpeechToText"],
+                resource_profile="cpu:1",
+                min_replicas=1,
+                max_replicas=1,
+            ),
+        )
+        mgr.store.apply_model(m)
+        await mgr.start()
+        try:
+            for _ in range(1200):
+                reps = mgr.store.list_replicas("stt")
+                if reps and reps[0].ready:
+                    break
+                await asyncio.sleep(0.1)
+            else:
+                raise AssertionError(f"replica never ready: {reps}")
+
+            t = np.arange(16_000) / 16_000.0
+            tone = (0.5 * np.sin(2 * np.pi * 440 * t) * 32767).astype(np.int16)
+            buf = io.BytesIO()
+            wavfile.write(buf, 16_000, tone)
+
+            transport = httpx.ASGITransport(app=mgr.app)
+            async with httpx.AsyncClient(
+                transport=transport, base_url="http://gw", timeout=60
+            ) as client:
+                r = await client.post(
+                    "/openai/v1/audio/transcriptions",
+                    files={"file": ("tone.wav", buf.getvalue(), "audio/wav")},
+                    data={"model": "stt"},
+                )
+                assert r.status_code == 200, r.text
+                assert isinstance(r.json()["text"], str)
+        finally:
+            await mgr.stop()
+
+    asyncio.run(body())

@@ -65,6 +65,10 @@ class SystemConfig:
     fixed_self_metric_addrs: list[str] = dataclasses.field(default_factory=list)
     leader_lock_path: str = "/tmp/kubeai-amd-leader.lock"
     n_gpus: Optional[int] = None  # None -> detect
+    # priorityClasses: name -> importance value (k8s PriorityClass analog,
+    # HIGHER = more important). A Model's priorityClassName maps every
+    # request for that model onto the engine's admission priority.
+    priority_classes: dict = dataclasses.field(default_factory=dict)
 
     def validate(self) -> None:
         if self.autoscaling.interval_seconds <= 0:

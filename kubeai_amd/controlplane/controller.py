@@ -129,7 +129,14 @@ class ModelController:
             return ev.owner or self._owner_of(ev.name)
 
         while True:
-            ev = await q.get()
+            # event-driven with a periodic resync (controller-runtime
+            # style): a dropped or raced event heals within one period
+            try:
+                ev = await asyncio.wait_for(q.get(), timeout=2.0)
+            except asyncio.TimeoutError:
+                for m in list(self.store.models):
+                    await self._safe_reconcile(m)
+                continue
             names = {owner(ev)}
             # drain burst
             while not q.empty():

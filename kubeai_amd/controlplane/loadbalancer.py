@@ -233,12 +233,21 @@ class LoadBalancer:
 
     async def _loop(self) -> None:
         q = self.store.subscribe()
-        await self.reconcile_all()
         while True:
-            await q.get()
+            try:
+                await self.reconcile_all()
+            except Exception:  # noqa: BLE001 — a bad event must not kill the watch
+                import traceback
+
+                traceback.print_exc()
+            # event-driven with a 1 s resync (controller-runtime style):
+            # a dropped/raced event costs at most one resync period
+            try:
+                await asyncio.wait_for(q.get(), timeout=1.0)
+            except asyncio.TimeoutError:
+                continue
             while not q.empty():
                 q.get_nowait()
-            await self.reconcile_all()
 
     async def reconcile_all(self) -> None:
         by_model: dict[str, dict[str, set[str]]] = {}

@@ -58,7 +58,9 @@ class Manager:
             state_path=self.cfg.autoscaling.state_path,
             is_leader=self.election.is_leader,
         )
-        self.proxy = ProxyHandler(self.model_client, self.lb)
+        self.proxy = ProxyHandler(
+            self.model_client, self.lb, priority_classes=self.cfg.priority_classes
+        )
         self.app = build_gateway_app(self.model_client, self.proxy)
         self.broker = broker or MemBroker()
         self.messengers = [

@@ -32,11 +32,15 @@ class ProxyHandler:
         max_retries: int = 3,
         timeout: float = 600.0,
         endpoint_timeout: float = 120.0,
+        priority_classes: Optional[dict] = None,
     ):
         self.model_client = model_client
         self.lb = lb
         self.max_retries = max_retries
         self.endpoint_timeout = endpoint_timeout
+        # k8s PriorityClass analog: name -> importance (higher = more
+        # important); mapped onto engine admission priority (lower-first)
+        self.priority_classes = priority_classes or {}
         self.client = httpx.AsyncClient(timeout=timeout)
 
     async def close(self) -> None:
@@ -65,6 +69,14 @@ class ProxyHandler:
                 return JSONResponse({"error": e.message}, status_code=e.status)
             if multipart:
                 pr = dataclasses.replace(pr, body=None)  # raw passthrough
+            elif pr.body is not None and "priority" not in pr.body:
+                # Model.spec.priorityClassName -> engine admission priority
+                # (engine serves LOWER values first; class value is k8s-style
+                # higher-is-more-important, hence the negation)
+                m = self.model_client.lookup_model(pr.model)
+                pcn = m.spec.priority_class_name if m else ""
+                if pcn and pcn in self.priority_classes:
+                    pr.body["priority"] = -int(self.priority_classes[pcn])
 
             self.model_client.scale_at_least_one_replica(pr.model)
             payload = raw if pr.body is None else json.dumps(pr.body).encode()

@@ -385,6 +385,16 @@ class ModelRunner:
                 )
                 counts.scatter_add_(0, out_t, torch.ones_like(out_t, dtype=torch.float32))
                 logits[i] -= fp * counts + pp * (counts > 0).float()
+            if r.params.logit_bias:
+                ids = torch.tensor(
+                    list(r.params.logit_bias.keys()), dtype=torch.int64,
+                    device=logits.device,
+                )
+                vals = torch.tensor(
+                    list(r.params.logit_bias.values()), dtype=logits.dtype,
+                    device=logits.device,
+                )
+                logits[i].scatter_add_(0, ids, vals)
         temps = [r.params.temperature for r in sample_reqs]
         if all(t <= 0.0 for t in temps):
             tokens = ops.greedy_sample(logits.contiguous())

@@ -28,6 +28,8 @@ class SamplingParams:
     # OpenAI penalties over generated tokens (vLLM semantics)
     presence_penalty: float = 0.0
     frequency_penalty: float = 0.0
+    # OpenAI logit_bias: {token_id: additive bias} applied before sampling
+    logit_bias: Optional[dict] = None
     # number of top-alternative logprobs to return per sampled token
     # (OpenAI `logprobs`/`top_logprobs`); 0 = none
     logprobs: int = 0

@@ -239,6 +239,11 @@ def build_app(server: EngineServer) -> FastAPI:
             presence_penalty=float(body.get("presence_penalty") or 0.0),
             frequency_penalty=float(body.get("frequency_penalty") or 0.0),
             priority=int(body.get("priority") or 0),
+            logit_bias=(
+                {int(k): float(v) for k, v in body["logit_bias"].items()}
+                if body.get("logit_bias")
+                else None
+            ),
         )
 
     def _stop_strings(body: dict) -> list[str]:

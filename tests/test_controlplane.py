@@ -118,7 +118,8 @@ class FakeBackend:
 
 
 @contextlib.asynccontextmanager
-async def harness(models=(), autoscaler_interval=0.05, messaging=False):
+async def harness(models=(), autoscaler_interval=0.05, messaging=False,
+                  priority_classes=None):
     cfg = SystemConfig(
         autoscaling=AutoscalingConfig(
             interval_seconds=autoscaler_interval,
@@ -126,6 +127,7 @@ async def harness(models=(), autoscaler_interval=0.05, messaging=False):
             state_path=None,
         ),
         leader_lock_path=f"/tmp/kubeai-test-{free_port()}.lock",
+        priority_classes=priority_classes or {},
     )
     store_runtime = {}
     mgr = Manager(cfg, runtime="placeholder")
@@ -522,5 +524,40 @@ def test_proxy_inflight_never_leaks_on_unexpected_error():
             await wait_for(lambda: mgr.lb.group("m1").total_in_flight == 0)
             for ep in mgr.lb.group("m1").endpoints.values():
                 assert ep.in_flight == 0
+
+    run(body())
+
+
+def test_priority_class_maps_to_request_priority():
+    """Model.spec.priorityClassName + config priorityClasses -> the proxy
+    injects the engine admission `priority` field (higher class value =
+    more important = lower engine priority); explicit request priority
+    wins."""
+    async def body():
+        m = text_gen_model("prio-m")
+        m.spec.priority_class_name = "critical"
+        m.spec.replicas = 1
+        async with harness([m], priority_classes={"critical": 1000}) as (
+            mgr, runtime, backend,
+        ):
+            rep = (await wait_for(lambda: mgr.store.list_replicas("prio-m")))[0]
+            runtime.mark_ready(rep.name, backend.address)
+            transport = httpx.ASGITransport(app=mgr.app)
+            async with httpx.AsyncClient(
+                transport=transport, base_url="http://gw"
+            ) as client:
+                r = await client.post(
+                    "/openai/v1/completions",
+                    json={"model": "prio-m", "prompt": "x", "max_tokens": 1},
+                )
+                assert r.status_code == 200
+                assert backend.requests[-1][1]["priority"] == -1000
+                r = await client.post(
+                    "/openai/v1/completions",
+                    json={"model": "prio-m", "prompt": "x", "max_tokens": 1,
+                          "priority": 7},
+                )
+                assert r.status_code == 200
+                assert backend.requests[-1][1]["priority"] == 7
 
     run(body())

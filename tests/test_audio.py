@@ -117,3 +117,25 @@ def test_transcription_errors(stt_client):
     # text endpoints answer a clear 400 on an STT model
     r = stt_client.post("/v1/completions", json={"prompt": "x", "max_tokens": 1})
     assert r.status_code == 400 and "speech-to-text" in r.text
+
+
+@pytest.mark.gpu
+def test_whisper_gpu_matches_cpu():
+    """STT on the GPU: deterministic, and numerically consistent with the
+    CPU run of the identical (same-seed) model."""
+    import torch
+
+    audio = np.sin(np.arange(16_000) / 30.0).astype(np.float32)
+    gpu = SpeechToText(PRESETS["whisper-tiny"], device="cuda", seed=0)
+    cpu = SpeechToText(PRESETS["whisper-tiny"], device="cpu", seed=0)
+    t1 = gpu.transcribe_tokens(audio, 16_000, max_tokens=8)
+    t2 = gpu.transcribe_tokens(audio, 16_000, max_tokens=8)
+    assert t1 == t2  # deterministic on device
+    with torch.inference_mode():
+        mel = torch.from_numpy(log_mel_spectrogram(audio, 16_000))[None]
+        eg = gpu.encoder(mel.cuda()).float().cpu()
+        ec = cpu.encoder(mel)
+    cos = torch.nn.functional.cosine_similarity(
+        eg.flatten(), ec.flatten(), dim=0
+    )
+    assert cos > 0.999

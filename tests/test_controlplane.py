@@ -158,7 +158,12 @@ async def wait_for(cond, timeout=5.0, interval=0.01):
 
 
 def text_gen_model(name="m1", **kw):
-    kw.setdefault("scale_down_delay_seconds", 0)  # fast tests
+    # scale-down stays OFF by default: with the harness's 0.05 s autoscaler
+    # tick, delay 0 means an idle model scales to zero within ~0.2 s, which
+    # races tests that mark a replica ready by hand before sending the first
+    # request (observed as rare 120 s endpoint-timeout flakes). Tests that
+    # exercise scale-down pass an explicit 0.
+    kw.setdefault("scale_down_delay_seconds", 300)
     spec = ModelSpec(url="hf://meta/llama-tiny", min_replicas=0, max_replicas=4, **kw)
     return Model(name=name, spec=spec)
 
@@ -439,7 +444,7 @@ def test_proxy_multipart_audio_routing():
 # ---------------------------------------------------------------- autoscaler
 def test_autoscaler_scales_up_and_down_to_zero():
     async def body():
-        m = text_gen_model()
+        m = text_gen_model(scale_down_delay_seconds=0,)
         m.spec.target_requests = 2
         async with harness([m]) as (mgr, runtime, backend):
             from kubeai_amd.controlplane import metrics

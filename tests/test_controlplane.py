@@ -566,3 +566,57 @@ def test_priority_class_maps_to_request_priority():
                 assert backend.requests[-1][1]["priority"] == 7
 
     run(body())
+
+
+def test_cache_lifecycle_and_finalizer(tmp_path):
+    """cacheProfile: reconcile creates the shared-dir cache + .loaded marker,
+    adds the eviction finalizer, status reflects loaded; deletion evicts the
+    dir and removes the model (reference: cache.go + finalizer flow)."""
+    import os
+
+    from kubeai_amd.controlplane.crd import CACHE_EVICTION_FINALIZER
+
+    async def body():
+        m = text_gen_model("cached-m", cache_profile="shared-fs")
+        m.spec.replicas = 1
+        async with harness([m]) as (mgr, runtime, backend):
+            mgr.controller.cache.base_dir = str(tmp_path)
+            await wait_for(lambda: mgr.store.list_replicas("cached-m"))
+            model = mgr.store.get_model("cached-m")
+            assert CACHE_EVICTION_FINALIZER in model.finalizers
+            assert model.status.cache_loaded
+            d = mgr.controller.cache.model_dir(model)
+            assert os.path.exists(os.path.join(d, ".loaded"))
+            # delete -> evict + remove
+            mgr.store.delete_model("cached-m")
+            await wait_for(lambda: mgr.store.get_model("cached-m") is None)
+            assert not os.path.exists(d)
+
+    run(body())
+
+
+def test_proxy_preserves_unknown_fields():
+    """vLLM-extension / vendor fields in the request body survive the proxy
+    re-marshal (reference: jsontext unknown-field passthrough)."""
+    async def body():
+        m = text_gen_model()
+        m.spec.replicas = 1
+        async with harness([m]) as (mgr, runtime, backend):
+            rep = (await wait_for(lambda: mgr.store.list_replicas("m1")))[0]
+            runtime.mark_ready(rep.name, backend.address)
+            transport = httpx.ASGITransport(app=mgr.app)
+            async with httpx.AsyncClient(
+                transport=transport, base_url="http://gw"
+            ) as client:
+                r = await client.post(
+                    "/openai/v1/completions",
+                    json={"model": "m1", "prompt": "x", "max_tokens": 1,
+                          "guided_json": {"type": "object"},
+                          "vendor_ext": [1, 2, 3]},
+                )
+                assert r.status_code == 200
+                seen = backend.requests[-1][1]
+                assert seen["guided_json"] == {"type": "object"}
+                assert seen["vendor_ext"] == [1, 2, 3]
+
+    run(body())

@@ -302,11 +302,18 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
   const int G = n_q / n_kv;
   TORCH_CHECK(n_q % n_kv == 0);
   if (B == 0) return;
-  // flash-decoding split: target >=1024 workgroups to fill 256 CUs
+  // flash-decoding split: target enough workgroups to fill 256 CUs with
+  // tail slack (default 2048; KUBEAI_DECODE_SPLIT_TARGET overrides for
+  // tuning sweeps — scripts/bench_ops.py)
+  static const int split_target = []() {
+    const char* e = getenv("KUBEAI_DECODE_SPLIT_TARGET");
+    const int v = e ? atoi(e) : 2048;
+    return v > 0 ? v : 2048;
+  }();
   int n_splits = 1;
   const int base_wgs = B * n_kv;
-  if (base_wgs < 2048) {
-    n_splits = std::min<int>(16, (2048 + base_wgs - 1) / base_wgs);
+  if (base_wgs < split_target) {
+    n_splits = std::min<int>(32, (split_target + base_wgs - 1) / base_wgs);
   }
   torch::Tensor part_o, part_ml;
   float *part_o_ptr = nullptr, *part_ml_ptr = nullptr;

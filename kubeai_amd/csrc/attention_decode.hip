@@ -302,9 +302,12 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
   const int G = n_q / n_kv;
   TORCH_CHECK(n_q % n_kv == 0);
   if (B == 0) return;
-  // flash-decoding split: target enough workgroups to fill 256 CUs with
-  // tail slack (default 2048; KUBEAI_DECODE_SPLIT_TARGET overrides for
-  // tuning sweeps — scripts/bench_ops.py)
+  // flash-decoding split: target 2048 workgroups, cap 16. Fixed-shape
+  // microbenches (scripts/sweep_decode_splits.py) mildly favor 4096, but
+  // the live bench's length mix loses ~4.5% end-to-end at 4096 (extra
+  // fp32 partial traffic + merge work inside graph replay) — same-box
+  // A/B in profiles/r01_results.md. KUBEAI_DECODE_SPLIT_TARGET overrides
+  // for tuning sweeps.
   static const int split_target = []() {
     const char* e = getenv("KUBEAI_DECODE_SPLIT_TARGET");
     const int v = e ? atoi(e) : 2048;
@@ -313,7 +316,7 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
   int n_splits = 1;
   const int base_wgs = B * n_kv;
   if (base_wgs < split_target) {
-    n_splits = std::min<int>(32, (split_target + base_wgs - 1) / base_wgs);
+    n_splits = std::min<int>(16, (split_target + base_wgs - 1) / base_wgs);
   }
   torch::Tensor part_o, part_ml;
   float *part_o_ptr = nullptr, *part_ml_ptr = nullptr;

@@ -243,9 +243,111 @@ class SpeechToText(nn.Module):
         cur = torch.tensor([[cfg.sot_token]], device=self.device_)
         for i in range(min(max_tokens, cfg.n_text_ctx - 1)):
             logits = self.decoder(cur, xa, kv_caches, offset=i)
-            nxt = int(logits[0, -1].argmax())
+            row = logits[0, -1].clone()
+            # the tied head makes hidden states correlate with the current
+            # token's embedding; mask it (and sot) so greedy decoding cannot
+            # collapse into a self-loop under random-init weights
+            row[toks[-1]] = -torch.inf
+            row[cfg.sot_token] = -torch.inf
+            nxt = int(row.argmax())
             toks.append(nxt)
             if nxt == cfg.eot_token:
                 break
             cur = torch.tensor([[nxt]], device=self.device_)
         return toks
+
+
+# ----------------------------------------------------------- checkpoint IO
+def save_whisper_checkpoint(model: "SpeechToText", out_dir: str) -> None:
+    """Write the model in HF whisper layout (tests + weight-prep tooling)."""
+    import os
+
+    from safetensors.torch import save_file
+
+    os.makedirs(out_dir, exist_ok=True)
+    state = {}
+    for name, p in model.state_dict().items():
+        hf = (
+            name.replace("encoder.", "model.encoder.")
+            .replace("decoder.", "model.decoder.")
+            .replace(".attn_ln.", ".self_attn_layer_norm.")
+            .replace(".cross_attn_ln.", ".encoder_attn_layer_norm.")
+            .replace(".mlp_ln.", ".final_layer_norm.")
+            .replace(".attn.", ".self_attn.")
+            .replace(".cross_attn.", ".encoder_attn.")
+            .replace(".blocks.", ".layers.")
+            .replace(".mlp.0.", ".fc1.").replace(".mlp.2.", ".fc2.")
+            .replace(".q.", ".q_proj.").replace(".k.", ".k_proj.")
+            .replace(".v.", ".v_proj.").replace(".out.", ".out_proj.")
+            .replace(".token_embedding.", ".embed_tokens.")
+            .replace("decoder.positional_embedding", "decoder.embed_positions.weight")
+            .replace(".ln_post.", ".layer_norm.")
+            .replace("decoder.ln.", "decoder.layer_norm.")
+        )
+        state[hf] = p.detach().cpu().contiguous()
+    save_file(state, os.path.join(out_dir, "model.safetensors"))
+    import json
+
+    cfg = model.cfg
+    with open(os.path.join(out_dir, "config.json"), "w") as f:
+        json.dump({"architectures": ["WhisperForConditionalGeneration"],
+                   "num_mel_bins": cfg.n_mels, "d_model": cfg.n_state,
+                   "encoder_layers": cfg.n_audio_layer,
+                   "decoder_layers": cfg.n_text_layer,
+                   "encoder_attention_heads": cfg.n_head,
+                   "decoder_attention_heads": cfg.n_head,
+                   "max_source_positions": cfg.n_audio_ctx,
+                   "max_target_positions": cfg.n_text_ctx,
+                   "vocab_size": cfg.vocab_size,
+                   "decoder_start_token_id": cfg.sot_token,
+                   "eos_token_id": cfg.eot_token}, f)
+
+
+def load_weights_whisper(model: "SpeechToText", model_dir: str) -> int:
+    """Load an HF whisper checkpoint (model.encoder/decoder.* names) into
+    SpeechToText; inverse of save_whisper_checkpoint. Returns params filled;
+    raises if any engine parameter stays unset."""
+    import glob
+    import os
+
+    from safetensors import safe_open
+
+    inverse = {}
+    for name in model.state_dict():
+        hf = (
+            name.replace("encoder.", "model.encoder.")
+            .replace("decoder.", "model.decoder.")
+            .replace(".attn_ln.", ".self_attn_layer_norm.")
+            .replace(".cross_attn_ln.", ".encoder_attn_layer_norm.")
+            .replace(".mlp_ln.", ".final_layer_norm.")
+            .replace(".attn.", ".self_attn.")
+            .replace(".cross_attn.", ".encoder_attn.")
+            .replace(".blocks.", ".layers.")
+            .replace(".mlp.0.", ".fc1.").replace(".mlp.2.", ".fc2.")
+            .replace(".q.", ".q_proj.").replace(".k.", ".k_proj.")
+            .replace(".v.", ".v_proj.").replace(".out.", ".out_proj.")
+            .replace(".token_embedding.", ".embed_tokens.")
+            .replace("decoder.positional_embedding", "decoder.embed_positions.weight")
+            .replace(".ln_post.", ".layer_norm.")
+            .replace("decoder.ln.", "decoder.layer_norm.")
+        )
+        inverse[hf] = name
+    sd = model.state_dict()
+    filled = set()
+    files = sorted(glob.glob(os.path.join(model_dir, "*.safetensors")))
+    if not files:
+        raise FileNotFoundError(f"no *.safetensors under {model_dir}")
+    for fpath in files:
+        with safe_open(fpath, framework="pt", device="cpu") as sf:
+            for key in sf.keys():
+                tgt = inverse.get(key)
+                if tgt is None:
+                    continue  # proj_out etc. — decoder head is tied
+                w = sf.get_tensor(key)
+                with torch.no_grad():
+                    sd[tgt].copy_(w.to(sd[tgt].dtype))
+                filled.add(tgt)
+    missing = set(sd) - filled
+    if missing:
+        raise ValueError(f"unfilled whisper parameters: {sorted(missing)[:8]}")
+    return len(filled)

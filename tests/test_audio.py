@@ -139,3 +139,23 @@ def test_whisper_gpu_matches_cpu():
         eg.flatten(), ec.flatten(), dim=0
     )
     assert cos > 0.999
+
+
+def test_whisper_checkpoint_roundtrip(tmp_path):
+    """HF whisper name mapping: save -> load into a differently-seeded
+    model -> identical transcription."""
+    from kubeai_amd.models.whisper import (
+        load_weights_whisper,
+        save_whisper_checkpoint,
+    )
+
+    audio = np.sin(np.arange(16_000) / 25.0).astype(np.float32)
+    src = SpeechToText(PRESETS["whisper-tiny"], device="cpu", seed=0)
+    ref = src.transcribe_tokens(audio, 16_000, max_tokens=10)
+    ckpt = str(tmp_path / "wckpt")
+    save_whisper_checkpoint(src, ckpt)
+    dst = SpeechToText(PRESETS["whisper-tiny"], device="cpu", seed=1234)
+    assert dst.transcribe_tokens(audio, 16_000, max_tokens=10) != ref
+    n = load_weights_whisper(dst, ckpt)
+    assert n > 20
+    assert dst.transcribe_tokens(audio, 16_000, max_tokens=10) == ref

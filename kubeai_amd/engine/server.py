@@ -88,12 +88,40 @@ class EngineServer:
         # heavy init inside the thread so /health can answer "starting"
         if self.task == "transcribe":
             from kubeai_amd.models.whisper import PRESETS as STT_PRESETS
-            from kubeai_amd.models.whisper import SpeechToText
+            from kubeai_amd.models.whisper import (
+                SpeechConfig,
+                SpeechToText,
+                load_weights_whisper,
+            )
 
-            scfg = STT_PRESETS.get(self.cfg.model, STT_PRESETS["whisper-tiny"])
+            ckpt_dir = None
+            if os.path.isdir(self.cfg.model) and os.path.exists(
+                os.path.join(self.cfg.model, "config.json")
+            ):
+                import json as _json
+
+                with open(os.path.join(self.cfg.model, "config.json")) as f:
+                    hc = _json.load(f)
+                scfg = SpeechConfig(
+                    n_mels=hc.get("num_mel_bins", 80),
+                    n_audio_ctx=hc.get("max_source_positions", 1500),
+                    n_text_ctx=hc.get("max_target_positions", 448),
+                    n_state=hc.get("d_model", 128),
+                    n_head=hc.get("encoder_attention_heads", 4),
+                    n_audio_layer=hc.get("encoder_layers", 2),
+                    n_text_layer=hc.get("decoder_layers", 2),
+                    vocab_size=hc.get("vocab_size", 2048),
+                    sot_token=hc.get("decoder_start_token_id", 1),
+                    eot_token=hc.get("eos_token_id", 2),
+                )
+                ckpt_dir = self.cfg.model
+            else:
+                scfg = STT_PRESETS.get(self.cfg.model, STT_PRESETS["whisper-tiny"])
             self.stt = SpeechToText(
                 scfg, device=self.cfg.resolve_device(), seed=self.cfg.seed
             )
+            if ckpt_dir is not None:
+                load_weights_whisper(self.stt, ckpt_dir)
             self.tokenizer = load_tokenizer(
                 self.cfg.model, scfg.vocab_size, scfg.sot_token, scfg.eot_token
             )

@@ -159,3 +159,34 @@ def test_whisper_checkpoint_roundtrip(tmp_path):
     n = load_weights_whisper(dst, ckpt)
     assert n > 20
     assert dst.transcribe_tokens(audio, 16_000, max_tokens=10) == ref
+
+
+def test_stt_server_loads_checkpoint_dir(tmp_path):
+    """An HF whisper checkpoint dir as --model: server derives the config
+    and loads the weights (same transcription as the source model)."""
+    from kubeai_amd.engine.engine import EngineConfig
+    from kubeai_amd.engine.server import EngineServer, build_app
+    from kubeai_amd.models.whisper import save_whisper_checkpoint
+
+    src = SpeechToText(PRESETS["whisper-tiny"], device="cpu", seed=7)
+    ckpt = str(tmp_path / "wdir")
+    save_whisper_checkpoint(src, ckpt)
+    audio = np.sin(np.arange(16_000) / 25.0).astype(np.float32)
+    want = src.transcribe_tokens(audio, 16_000, max_tokens=64)
+
+    server = EngineServer(
+        EngineConfig(model=ckpt, device="cpu", num_gpu_blocks=16, seed=999),
+        "stt-ckpt", task="transcribe",
+    )
+    server.start()
+    server._ready.wait(timeout=60)
+    try:
+        assert server.stt.transcribe_tokens(audio, 16_000, max_tokens=64) == want
+        with TestClient(build_app(server)) as c:
+            r = c.post(
+                "/v1/audio/transcriptions",
+                files={"file": ("t.wav", _wav_bytes(), "audio/wav")},
+            )
+            assert r.status_code == 200
+    finally:
+        server.stop()

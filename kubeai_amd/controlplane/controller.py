@@ -17,7 +17,7 @@ from typing import Optional
 import httpx
 
 from .crd import CACHE_EVICTION_FINALIZER, Model, feature_labels
-from .store import Replica, Store
+from .store import Replica, ReplicaState, Store
 
 
 def fnv1a_32(data: bytes) -> int:
@@ -213,8 +213,17 @@ class ModelController:
         current = sorted(
             self.store.list_replicas(model=name), key=lambda r: r.created_seq
         )
-        up_to_date = [r for r in current if r.hash == h]
-        out_of_date = [r for r in current if r.hash != h]
+        # FAILED replicas (engine process died — health monitor) are treated
+        # as out-of-date so the surge branch recreates them immediately
+        # (reference analog: pod restart policy + reconcile recreate)
+        up_to_date = [
+            r for r in current
+            if r.hash == h and r.state != ReplicaState.FAILED
+        ]
+        out_of_date = [
+            r for r in current
+            if r.hash != h or r.state == ReplicaState.FAILED
+        ]
         to_delete: list[Replica] = []
         n_create = 0
 

@@ -620,3 +620,22 @@ def test_proxy_preserves_unknown_fields():
                 assert seen["vendor_ext"] == [1, 2, 3]
 
     run(body())
+
+
+def test_failed_replica_recreated():
+    """A replica whose engine process died (state=Failed via the health
+    monitor) must be deleted and replaced by the reconciler."""
+    async def body():
+        m = text_gen_model()
+        m.spec.replicas = 1
+        async with harness([m]) as (mgr, runtime, backend):
+            rep = (await wait_for(lambda: mgr.store.list_replicas("m1")))[0]
+            mgr.store.update_replica(rep.name, state=ReplicaState.FAILED)
+            await wait_for(
+                lambda: (lambda reps: len(reps) == 1
+                         and reps[0].name != rep.name
+                         and reps[0].state != ReplicaState.FAILED)(
+                    mgr.store.list_replicas("m1"))
+            )
+
+    run(body())

@@ -313,10 +313,21 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
     const int v = e ? atoi(e) : 2048;
     return v > 0 ? v : 2048;
   }();
+  // optional length-aware bound: >=N cache blocks per split, so short
+  // sequences don't launch mostly-empty splits (max_blocks is the widest
+  // block table this step — host-side, no device sync)
+  static const int min_blocks_per_split = []() {
+    const char* e = getenv("KUBEAI_DECODE_MIN_BLOCKS_PER_SPLIT");
+    return e ? atoi(e) : 0;
+  }();
   int n_splits = 1;
   const int base_wgs = B * n_kv;
   if (base_wgs < split_target) {
     n_splits = std::min<int>(16, (split_target + base_wgs - 1) / base_wgs);
+    if (min_blocks_per_split > 0) {
+      const int cap = std::max(1, max_blocks / min_blocks_per_split);
+      n_splits = std::min(n_splits, cap);
+    }
   }
   torch::Tensor part_o, part_ml;
   float *part_o_ptr = nullptr, *part_ml_ptr = nullptr;

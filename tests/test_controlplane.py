@@ -639,3 +639,41 @@ def test_failed_replica_recreated():
             )
 
     run(body())
+
+
+def test_messenger_flow_file_broker(tmp_path):
+    """Same messenger flow over the durable file:// driver: the request and
+    response ride the shared directory instead of the in-memory queue."""
+    from kubeai_amd.controlplane.messenger import broker_from_url
+
+    async def body():
+        m = text_gen_model()
+        m.spec.replicas = 1
+        async with harness([m]) as (mgr, runtime, backend):
+            broker = broker_from_url(f"file://{tmp_path}/bus")
+            msgr = Messenger(
+                broker, "req", "resp", mgr.model_client, mgr.lb, max_handlers=2
+            )
+            msgr.start()
+            try:
+                rep = (await wait_for(lambda: mgr.store.list_replicas("m1")))[0]
+                runtime.mark_ready(rep.name, backend.address)
+                await broker.publish(
+                    "req",
+                    json.dumps(
+                        {
+                            "metadata": {"id": "f1"},
+                            "path": "/v1/completions",
+                            "body": {"model": "m1", "prompt": "via file bus"},
+                        }
+                    ).encode(),
+                )
+                out = json.loads(
+                    await asyncio.wait_for(broker.receive("resp"), timeout=10)
+                )
+                assert out["metadata"] == {"id": "f1"}
+                assert out["status_code"] == 200
+            finally:
+                await msgr.stop()
+
+    run(body())

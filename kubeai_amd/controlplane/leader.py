@@ -48,6 +48,7 @@ class Election:
             await asyncio.sleep(self.lease_duration / 3)
 
     def _try_acquire(self) -> None:
+        fh = None
         try:
             fh = open(self.lock_path, "a+")
             fcntl.flock(fh, fcntl.LOCK_EX | fcntl.LOCK_NB)
@@ -58,9 +59,8 @@ class Election:
             self._fh = fh
             self._is_leader = True
         except (OSError, BlockingIOError):
-            if self._fh:
-                self._fh.close()
-                self._fh = None
+            if fh is not None:
+                fh.close()
             self._is_leader = False
 
     def _heartbeat(self) -> None:

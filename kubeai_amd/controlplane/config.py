@@ -111,6 +111,9 @@ def load_config(path: Optional[str]) -> SystemConfig:
         metrics_port=int(raw.get("metricsPort", 8080)),
         fixed_self_metric_addrs=list(raw.get("fixedSelfMetricAddrs", [])),
         n_gpus=raw.get("nGPUs"),
+        priority_classes={
+            str(k): int(v) for k, v in raw.get("priorityClasses", {}).items()
+        },
     )
     cfg.validate()
     return cfg

@@ -62,18 +62,28 @@ class Manager:
             self.model_client, self.lb, priority_classes=self.cfg.priority_classes
         )
         self.app = build_gateway_app(self.model_client, self.proxy)
+        # broker driver selected per stream by URL scheme (mem:// shares
+        # one in-process broker; file:// is durable); an injected broker
+        # overrides for tests
+        from .messenger import stream_transport
+
         self.broker = broker or MemBroker()
-        self.messengers = [
-            Messenger(
-                self.broker,
-                s.requests_url.split("://", 1)[-1],
-                s.responses_url.split("://", 1)[-1],
-                self.model_client,
-                self.lb,
-                max_handlers=s.max_handlers,
+        self.messengers = []
+        for s in self.cfg.messaging:
+            if broker is None and not s.requests_url.startswith("mem"):
+                b, req_t, resp_t = stream_transport(
+                    s.requests_url, s.responses_url
+                )
+            else:
+                b = self.broker
+                req_t = s.requests_url.split("://", 1)[-1]
+                resp_t = s.responses_url.split("://", 1)[-1]
+            self.messengers.append(
+                Messenger(
+                    b, req_t, resp_t, self.model_client, self.lb,
+                    max_handlers=s.max_handlers,
+                )
             )
-            for s in self.cfg.messaging
-        ]
 
     async def start(self) -> None:
         self.lb.start()

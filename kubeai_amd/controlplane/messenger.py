@@ -107,6 +107,26 @@ def broker_from_url(url: str):
     raise ValueError(f"unknown messenger driver: {u.scheme}:// (have mem, file)")
 
 
+def stream_transport(requests_url: str, responses_url: str):
+    """Resolve one messaging stream to (broker, requests_topic,
+    responses_topic). For file:// the directory is the broker root and the
+    last path segment the topic, so both topics share one root dir."""
+    u = urlparse(requests_url)
+    if u.scheme == "file":
+        root = os.path.dirname(u.path) or "/tmp/kubeai-msgs"
+        return (
+            FileBroker(root),
+            os.path.basename(u.path),
+            os.path.basename(urlparse(responses_url).path),
+        )
+    broker = broker_from_url(requests_url)
+    return (
+        broker,
+        requests_url.split("://", 1)[-1],
+        responses_url.split("://", 1)[-1],
+    )
+
+
 class Messenger:
     def __init__(
         self,

@@ -313,3 +313,34 @@ def test_deploy_manifests_parse():
         list(yaml.safe_load_all(text))
         parsed += 1
     assert parsed >= 2  # at least the CRD + catalog
+
+
+def test_stream_transport_resolution(tmp_path):
+    from kubeai_amd.controlplane.messenger import stream_transport
+
+    b, rq, rs = stream_transport("mem://req", "mem://resp")
+    assert type(b).__name__ == "MemBroker" and (rq, rs) == ("req", "resp")
+    b, rq, rs = stream_transport(
+        f"file://{tmp_path}/bus/requests", f"file://{tmp_path}/bus/responses"
+    )
+    assert type(b).__name__ == "FileBroker"
+    assert b.root == f"{tmp_path}/bus"
+    assert (rq, rs) == ("requests", "responses")
+
+
+def test_load_config_priority_classes(tmp_path):
+    import yaml
+
+    from kubeai_amd.controlplane.config import load_config
+
+    p = tmp_path / "cfg.yaml"
+    p.write_text(yaml.safe_dump({
+        "priorityClasses": {"critical": 1000, "batch": -10},
+        "messaging": {"streams": [
+            {"requestsURL": f"file://{tmp_path}/q/req",
+             "responsesURL": f"file://{tmp_path}/q/resp"},
+        ]},
+    }))
+    cfg = load_config(str(p))
+    assert cfg.priority_classes == {"critical": 1000, "batch": -10}
+    assert cfg.messaging[0].requests_url.startswith("file://")

@@ -42,6 +42,11 @@ class Fp8Linear(nn.Module):
 
     def forward_quantized(self, xq: torch.Tensor, x_scale: torch.Tensor) -> torch.Tensor:
         """Pre-quantized input from a fused producer kernel (quant is free)."""
+        if not xq.is_cuda:
+            # CPU semantics fallback (tests): dequantize and matmul
+            x = xq.float() * x_scale.view(-1, 1)
+            w = self.weight_fp8.float() * self.weight_scale.reshape(-1, 1)
+            return (x @ w.t()).to(torch.bfloat16)
         return torch._scaled_mm(
             xq,
             self.weight_fp8.t(),
@@ -54,6 +59,8 @@ class Fp8Linear(nn.Module):
         a_amax = x.float().abs().amax(dim=-1, keepdim=True).clamp_min(1e-6)
         a_scale = a_amax / FP8_MAX
         xq = (x.float() / a_scale).clamp(-FP8_MAX, FP8_MAX).to(FP8_DTYPE)
+        if not x.is_cuda:
+            return self.forward_quantized(xq, a_scale.squeeze(-1))
         return torch._scaled_mm(
             xq,
             self.weight_fp8.t(),
@@ -83,10 +90,12 @@ def convert_to_fp8(model: nn.Module) -> int:
     for layer in getattr(model, "layers", []):
         attn = getattr(layer, "self_attn", None)
         mlp = getattr(layer, "mlp", None)
-        if (
-            attn is not None
-            and isinstance(getattr(attn, "qkv_proj", None), Fp8Linear)
-            and isinstance(getattr(mlp, "gate_up_proj", None), Fp8Linear)
-        ):
+        if attn is None or not isinstance(getattr(attn, "qkv_proj", None), Fp8Linear):
+            continue
+        if isinstance(getattr(mlp, "gate_up_proj", None), Fp8Linear):
+            layer._fp8_fused = True  # dense: norms/activations emit fp8
+        elif hasattr(mlp, "experts"):
+            # MoE: attention runs the fused-fp8 path; the MLP input stays
+            # bf16 (router needs it) and MoEMLP quantizes once itself
             layer._fp8_fused = True
     return n

@@ -134,8 +134,8 @@ class ModelRunner:
                 load_weights(self.model, model_path)
         self.quantization = quantization
         if quantization == "fp8":
-            if self.device.type != "cuda":
-                raise ValueError("fp8 quantization requires a GPU")
+            # on CPU the Fp8Linear dequant fallback keeps the same
+            # semantics (tests); the fast _scaled_mm path needs the GPU
             from .quant import convert_to_fp8
 
             n = convert_to_fp8(self.model)

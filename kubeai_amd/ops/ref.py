@@ -292,3 +292,13 @@ def topk_topp_sample(
         idx = int(torch.searchsorted(sp.cumsum(0), torch.tensor(r)).clamp(max=sp.shape[0] - 1))
         out[b] = int(si[idx])
     return out
+
+
+def quant_fp8(x: torch.Tensor):
+    """Per-row dynamic OCP-e4m3 quantization (CPU reference of
+    csrc/quant_fp8.hip::quant_fp8): scale = amax/448, symmetric."""
+    xf = x.float()
+    amax = xf.abs().amax(dim=-1).clamp_min(1e-6)
+    scale = amax / 448.0
+    q = (xf / scale[:, None]).clamp(-448.0, 448.0).to(torch.float8_e4m3fn)
+    return q, scale

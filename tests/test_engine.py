@@ -196,3 +196,51 @@ def test_oversized_request_rejected_not_stuck():
     assert outs["huge"].finish_reason == "abort"
     assert outs["huge"].output_token_ids == []
     assert len(outs["ok"].output_token_ids) == 4
+
+
+def test_fp8_cpu_semantics_close_to_bf16():
+    """fp8 W8A8 on CPU (dequant fallback): logits stay close to the bf16
+    model with identical weights — covers the MoE single-quant expert path
+    and the dense Fp8Linear path without a GPU."""
+    import torch
+
+    from kubeai_amd.engine.quant import convert_to_fp8
+    from kubeai_amd.engine.runner import ModelRunner
+    from kubeai_amd.models.config import PRESETS
+
+    for preset in ("llama-tiny", "mixtral-tiny"):
+        r_bf16 = ModelRunner(
+            PRESETS[preset], device="cpu", num_gpu_blocks=64, seed=5
+        )
+        r_fp8 = ModelRunner(
+            PRESETS[preset], device="cpu", num_gpu_blocks=64, seed=5,
+            quantization="fp8",
+        )
+        if preset == "mixtral-tiny":
+            # MoE experts quantize the hidden once per layer
+            assert hasattr(
+                r_fp8.model.layers[0].mlp.experts[0].gate_up_proj,
+                "forward_quantized",
+            )
+        from kubeai_amd.engine.scheduler import (
+            Request,
+            SamplingParams,
+            Scheduler,
+        )
+        from kubeai_amd.engine.kvcache import BlockManager
+
+        def logits_of(runner):
+            bm = BlockManager(64, 16)
+            s = Scheduler(bm, max_num_batched_tokens=512, max_model_len=256)
+            s.add_request(Request(list(range(10, 58)),
+                                  SamplingParams(max_tokens=1), request_id="x"))
+            out = s.schedule()
+            fb = runner.build_batch(out)
+            h = runner.model(fb)
+            return runner.model.compute_logits(h[fb.logits_indices]).float()
+
+        a, b = logits_of(r_bf16), logits_of(r_fp8)
+        cos = torch.nn.functional.cosine_similarity(
+            a.flatten(), b.flatten(), dim=0
+        )
+        assert cos > 0.97, f"{preset}: fp8 CPU cos {cos}"

@@ -160,7 +160,7 @@ def test_n_choices(client):
     r = client.post(
         "/v1/completions",
         json={"prompt": "pick some words", "max_tokens": 4, "n": 3,
-              "temperature": 0.9},
+              "temperature": 0.9, "ignore_eos": True},
     )
     assert r.status_code == 200
     body = r.json()

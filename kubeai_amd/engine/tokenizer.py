@@ -65,15 +65,29 @@ def apply_chat_template(tokenizer, messages: list[dict]) -> list[int]:
     Reference analog: the engine containers own their chat template
     (engine contract); format is internal.
     """
-    role_ids = {"system": 3, "user": 4, "assistant": 5, "tool": 6}
-    toks: list[int] = [tokenizer.bos_token_id]
-    for m in messages:
-        toks.append(role_ids.get(m.get("role", "user"), 4))
+    def text_of(m: dict) -> str:
         content = m.get("content") or ""
         if isinstance(content, list):  # OpenAI content-parts form
             content = " ".join(
                 p.get("text", "") for p in content if isinstance(p, dict)
             )
-        toks.extend(tokenizer.encode(content))
+        return content
+
+    # real checkpoints: use the model's own chat template
+    tk = getattr(tokenizer, "_tok", None)
+    if tk is not None and getattr(tk, "chat_template", None):
+        try:
+            return tk.apply_chat_template(
+                [{"role": m.get("role", "user"), "content": text_of(m)}
+                 for m in messages],
+                add_generation_prompt=True,
+            )
+        except Exception:
+            pass
+    role_ids = {"system": 3, "user": 4, "assistant": 5, "tool": 6}
+    toks: list[int] = [tokenizer.bos_token_id]
+    for m in messages:
+        toks.append(role_ids.get(m.get("role", "user"), 4))
+        toks.extend(tokenizer.encode(text_of(m)))
     toks.append(role_ids["assistant"])  # generation prompt
     return toks

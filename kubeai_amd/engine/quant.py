@@ -21,8 +21,12 @@ FP8_MAX = 448.0
 class Fp8Linear(nn.Module):
     """Drop-in replacement for bias-free nn.Linear, fp8 weights."""
 
-    def __init__(self, weight_bf16: torch.Tensor):
+    def __init__(self, weight_bf16: torch.Tensor, bias: torch.Tensor | None = None):
         super().__init__()
+        if bias is not None:
+            self.register_buffer("bias", bias.detach().to(torch.bfloat16))
+        else:
+            self.bias = None
         w = weight_bf16.detach().float()
         w_amax = w.abs().amax(dim=1, keepdim=True).clamp_min(1e-6)  # [out,1]
         w_scale = w_amax / FP8_MAX
@@ -46,12 +50,14 @@ class Fp8Linear(nn.Module):
             # CPU semantics fallback (tests): dequantize and matmul
             x = xq.float() * x_scale.view(-1, 1)
             w = self.weight_fp8.float() * self.weight_scale.reshape(-1, 1)
-            return (x @ w.t()).to(torch.bfloat16)
+            y = (x @ w.t()).to(torch.bfloat16)
+            return y + self.bias if self.bias is not None else y
         return torch._scaled_mm(
             xq,
             self.weight_fp8.t(),
             scale_a=x_scale.view(-1, 1),
             scale_b=self.weight_scale,
+            bias=self.bias,
             out_dtype=torch.bfloat16,
         )
 
@@ -66,6 +72,7 @@ class Fp8Linear(nn.Module):
             self.weight_fp8.t(),
             scale_a=a_scale,
             scale_b=self.weight_scale,
+            bias=self.bias,
             out_dtype=torch.bfloat16,
         )
 
@@ -85,7 +92,7 @@ def convert_to_fp8(model: nn.Module) -> int:
         for name in _TARGETS:
             child = getattr(mod, name, None)
             if isinstance(child, nn.Linear):
-                setattr(mod, name, Fp8Linear(child.weight))
+                setattr(mod, name, Fp8Linear(child.weight, child.bias))
                 n += 1
     for layer in getattr(model, "layers", []):
         attn = getattr(layer, "self_attn", None)

@@ -29,6 +29,8 @@ class ModelArchConfig:
     # MoE (Mixtral-style); n_experts == 0 means dense MLP
     num_local_experts: int = 0
     num_experts_per_tok: int = 2
+    # Qwen2-style: q/k/v projections carry a bias
+    attention_bias: bool = False
 
     @property
     def n_kv_heads(self) -> int:
@@ -46,6 +48,10 @@ class ModelArchConfig:
             eos = eos[0]
         return ModelArchConfig(
             architecture="mixtral" if "Mixtral" in arch else "llama",
+            # Qwen2 always uses qkv bias; Llama-family configs may say so
+            attention_bias=bool(
+                cfg.get("attention_bias", arch.startswith("Qwen2"))
+            ),
             vocab_size=cfg["vocab_size"],
             hidden_size=hidden,
             intermediate_size=cfg["intermediate_size"],
@@ -75,6 +81,38 @@ PRESETS: dict[str, ModelArchConfig] = {
         num_key_value_heads=8,
     ),
     # tiny: CPU tests / smoke; same head_dim=128 so HIP kernel paths match
+    # Qwen2 family (also the DeepSeek-R1-Distill-Qwen arch): qkv bias,
+    # tied embeddings on the small models
+    "qwen2-tiny": ModelArchConfig(
+        vocab_size=2048,
+        hidden_size=256,
+        intermediate_size=512,
+        num_hidden_layers=2,
+        num_attention_heads=2,
+        num_key_value_heads=1,
+        head_dim=128,
+        max_position_embeddings=2048,
+        bos_token_id=1,
+        eos_token_id=2,
+        attention_bias=True,
+        tie_word_embeddings=True,
+        rope_theta=1000000.0,
+    ),
+    "qwen2-7b": ModelArchConfig(
+        vocab_size=152064,
+        hidden_size=3584,
+        intermediate_size=18944,
+        num_hidden_layers=28,
+        num_attention_heads=28,
+        num_key_value_heads=4,
+        head_dim=128,
+        max_position_embeddings=32768,
+        rms_norm_eps=1e-6,
+        rope_theta=1000000.0,
+        bos_token_id=151643,
+        eos_token_id=151645,
+        attention_bias=True,
+    ),
     "llama-tiny": ModelArchConfig(
         vocab_size=2048,
         hidden_size=256,

@@ -28,10 +28,13 @@ from .config import ModelArchConfig
 class EngineLinear(nn.Linear):
     """nn.Linear routed through ops.linear: the hand-written weight-
     streaming GEMM takes skinny decode batches (M<=64); hipBLASLt keeps
-    the rest."""
+    the rest. Bias (Qwen2 qkv) is added outside the skinny path."""
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        return ops.linear(x, self.weight)
+        y = ops.linear(x, self.weight)
+        if self.bias is not None:
+            y = y + self.bias
+        return y
 
 
 class Attention(nn.Module):
@@ -44,7 +47,9 @@ class Attention(nn.Module):
         self.scale = 1.0 / math.sqrt(self.hd)
         q_size = self.n_q * self.hd
         kv_size = self.n_kv * self.hd
-        self.qkv_proj = EngineLinear(cfg.hidden_size, q_size + 2 * kv_size, bias=False)
+        self.qkv_proj = EngineLinear(
+            cfg.hidden_size, q_size + 2 * kv_size, bias=cfg.attention_bias
+        )
         self.o_proj = EngineLinear(q_size, cfg.hidden_size, bias=False)
 
     def forward(

@@ -158,6 +158,14 @@ def load_weights(model, model_dir: str) -> int:
                 torch.cat([ps["q"], ps["k"], ps["v"]], dim=0),
             )
 
+    def fuse_qkv_bias(layer: int) -> None:
+        ps = pending.get(f"qkvb.{layer}", {})
+        if len(ps) == 3:
+            put(
+                f"layers.{layer}.self_attn.qkv_proj.bias",
+                torch.cat([ps["q"], ps["k"], ps["v"]], dim=0),
+            )
+
     def fuse_gate_up(layer: int, expert: int | None) -> None:
         key = f"gu.{layer}" + ("" if expert is None else f".{expert}")
         ps = pending.get(key, {})
@@ -195,6 +203,15 @@ def load_weights(model, model_dir: str) -> int:
             elif rest == "self_attn.v_proj.weight":
                 pending.setdefault(f"qkv.{layer}", {})["v"] = w
                 fuse_qkv(layer)
+            elif rest == "self_attn.q_proj.bias":
+                pending.setdefault(f"qkvb.{layer}", {})["q"] = w
+                fuse_qkv_bias(layer)
+            elif rest == "self_attn.k_proj.bias":
+                pending.setdefault(f"qkvb.{layer}", {})["k"] = w
+                fuse_qkv_bias(layer)
+            elif rest == "self_attn.v_proj.bias":
+                pending.setdefault(f"qkvb.{layer}", {})["v"] = w
+                fuse_qkv_bias(layer)
             elif rest == "self_attn.o_proj.weight":
                 put(f"layers.{layer}.self_attn.o_proj.weight", w)
             elif rest == "mlp.gate_proj.weight":
@@ -255,6 +272,12 @@ def save_hf_checkpoint(model, out_dir: str) -> None:
         state[pre + "self_attn.q_proj.weight"] = qkv[: nq * hd].clone()
         state[pre + "self_attn.k_proj.weight"] = qkv[nq * hd : (nq + nkv) * hd].clone()
         state[pre + "self_attn.v_proj.weight"] = qkv[(nq + nkv) * hd :].clone()
+        qkv_b = layer.self_attn.qkv_proj.bias
+        if qkv_b is not None:
+            b = qkv_b.detach().cpu()
+            state[pre + "self_attn.q_proj.bias"] = b[: nq * hd].clone()
+            state[pre + "self_attn.k_proj.bias"] = b[nq * hd : (nq + nkv) * hd].clone()
+            state[pre + "self_attn.v_proj.bias"] = b[(nq + nkv) * hd :].clone()
         state[pre + "self_attn.o_proj.weight"] = (
             layer.self_attn.o_proj.weight.detach().cpu()
         )
@@ -296,6 +319,7 @@ def save_hf_checkpoint(model, out_dir: str) -> None:
                 "eos_token_id": cfg.eos_token_id,
                 "num_local_experts": cfg.num_local_experts,
                 "num_experts_per_tok": cfg.num_experts_per_tok,
+                "attention_bias": cfg.attention_bias,
                 "torch_dtype": "bfloat16",
             },
             f,

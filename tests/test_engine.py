@@ -208,7 +208,7 @@ def test_fp8_cpu_semantics_close_to_bf16():
     from kubeai_amd.engine.runner import ModelRunner
     from kubeai_amd.models.config import PRESETS
 
-    for preset in ("llama-tiny", "mixtral-tiny"):
+    for preset in ("llama-tiny", "mixtral-tiny", "qwen2-tiny"):
         r_bf16 = ModelRunner(
             PRESETS[preset], device="cpu", num_gpu_blocks=64, seed=5
         )
@@ -244,3 +244,56 @@ def test_fp8_cpu_semantics_close_to_bf16():
             a.flatten(), b.flatten(), dim=0
         )
         assert cos > 0.97, f"{preset}: fp8 CPU cos {cos}"
+
+
+def test_qwen2_arch_engine():
+    """Qwen2 family (qkv bias + tied embeddings): deterministic generation,
+    bias actually shifts the logits, checkpoint roundtrip keeps the bias."""
+    import torch
+
+    from kubeai_amd.models.loader import save_hf_checkpoint
+
+    def gen(model, seed=6, rid="q"):
+        eng = LLMEngine(
+            EngineConfig(model=model, device="cpu", num_gpu_blocks=128,
+                         max_model_len=512, seed=seed)
+        )
+        eng.add_request(list(range(10, 90)), SamplingParams(max_tokens=6),
+                        request_id=rid)
+        return run_to_completion(eng)[rid].output_token_ids, eng
+
+    a, eng = gen("qwen2-tiny")
+    b, _ = gen("qwen2-tiny")
+    assert a == b
+    qkv = eng.runner.model.layers[0].self_attn.qkv_proj
+    assert qkv.bias is not None
+    assert eng.runner.model.lm_head is None  # tied embeddings
+    # shifting the bias must move the logits (bias is live in the forward)
+    from kubeai_amd.engine.kvcache import BlockManager
+    from kubeai_amd.engine.scheduler import Request, Scheduler
+
+    def logits_of(e):
+        s = Scheduler(BlockManager(64, 16), max_num_batched_tokens=512,
+                      max_model_len=256)
+        s.add_request(Request(list(range(10, 58)), SamplingParams(max_tokens=1),
+                              request_id="x"))
+        out = s.schedule()
+        fb = e.runner.build_batch(out)
+        h = e.runner.model(fb)
+        return e.runner.model.compute_logits(h[fb.logits_indices]).float()
+
+    before = logits_of(eng)
+    with torch.no_grad():
+        for layer in eng.runner.model.layers:
+            layer.self_attn.qkv_proj.bias.add_(0.3)
+    after = logits_of(eng)
+    assert not torch.allclose(before, after)
+    # checkpoint roundtrip with biases
+    import tempfile
+
+    with tempfile.TemporaryDirectory() as d:
+        _, src = gen("qwen2-tiny", seed=8, rid="s")
+        save_hf_checkpoint(src.runner.model, d)
+        want, _ = gen("qwen2-tiny", seed=8, rid="w")
+        got, _ = gen(d, seed=999, rid="g")
+        assert got == want

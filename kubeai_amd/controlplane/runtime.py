@@ -148,8 +148,13 @@ class LocalProcessRuntime:
             "--port",
             str(port),
         ] + list(model.spec.args)
-        if "SpeechToText" in model.spec.features and "--task" not in model.spec.args:
-            cmd += ["--task", "transcribe"]
+        if "--task" not in model.spec.args:
+            feats = set(model.spec.features)
+            if "SpeechToText" in feats:
+                cmd += ["--task", "transcribe"]
+            elif feats and feats <= {"TextEmbedding", "Reranking"}:
+                # pure encoder model (no TextGeneration): BERT-arch engine
+                cmd += ["--task", "embed"]
         if not ids:
             cmd += ["--device", "cpu"]
         proc = subprocess.Popen(cmd, env=env)

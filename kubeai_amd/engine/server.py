@@ -67,6 +67,7 @@ class EngineServer:
         self._dist = None
         self.engine: Optional[LLMEngine] = None
         self.stt = None  # SpeechToText model when task == "transcribe"
+        self.embedder = None  # BertEncoder when task == "embed"
         self.tokenizer = None
         self._submit: "queue.Queue" = queue.Queue()
         self._events: dict[str, tuple[asyncio.AbstractEventLoop, asyncio.Queue]] = {}
@@ -127,6 +128,28 @@ class EngineServer:
             )
             self._ready.set()
             self._stop.wait()  # transcription is request-driven, no stepping
+            return
+        if self.task == "embed":
+            from kubeai_amd.models import bert as bert_mod
+
+            if os.path.isdir(self.cfg.model) and os.path.exists(
+                os.path.join(self.cfg.model, "config.json")
+            ):
+                bcfg = bert_mod.config_from_hf(self.cfg.model)
+                self.embedder = bert_mod.BertEncoder(
+                    bcfg, device=self.cfg.resolve_device(), seed=self.cfg.seed
+                )
+                bert_mod.load_weights_bert(self.embedder, self.cfg.model)
+            else:
+                bcfg = bert_mod.PRESETS.get(
+                    self.cfg.model, bert_mod.PRESETS["bert-tiny"]
+                )
+                self.embedder = bert_mod.BertEncoder(
+                    bcfg, device=self.cfg.resolve_device(), seed=self.cfg.seed
+                )
+            self.tokenizer = load_tokenizer(self.cfg.model, bcfg.vocab_size, 1, 2)
+            self._ready.set()
+            self._stop.wait()  # encoding is request-driven, no stepping
             return
         tp_group = None
         if self.tp_size > 1 or os.environ.get("KUBEAI_FORCE_TP"):
@@ -245,10 +268,10 @@ def build_app(server: EngineServer) -> FastAPI:
     name = server.served_model_name
 
     def _generate_unsupported():
+        task = "transcribe" if server.stt is not None else "embed"
         return JSONResponse(
-            {"error": {"message": f"model {name} is a speech-to-text model "
-                                  "(task=transcribe); text endpoints are "
-                                  "not supported"}},
+            {"error": {"message": f"model {name} runs task={task}; text "
+                                  "generation endpoints are not supported"}},
             status_code=400,
         )
 
@@ -334,7 +357,7 @@ def build_app(server: EngineServer) -> FastAPI:
 
     @app.post("/v1/completions")
     async def completions(request: Request):
-        if server.stt is not None:
+        if server.stt is not None or server.embedder is not None:
             return _generate_unsupported()
         body = await request.json()
         params = _params_from(body)
@@ -420,7 +443,7 @@ def build_app(server: EngineServer) -> FastAPI:
 
     @app.post("/v1/chat/completions")
     async def chat_completions(request: Request):
-        if server.stt is not None:
+        if server.stt is not None or server.embedder is not None:
             return _generate_unsupported()
         body = await request.json()
         params = _params_from(body)
@@ -498,9 +521,16 @@ def build_app(server: EngineServer) -> FastAPI:
         if isinstance(inputs, str):
             inputs = [inputs]
         tok_lists = [server.tokenizer.encode(t, add_bos=True) for t in inputs]
-        vecs = await asyncio.get_running_loop().run_in_executor(
-            None, server.engine.embed, tok_lists
-        )
+        if server.embedder is not None:
+            # BERT-architecture embedding model (task=embed)
+            enc = await asyncio.get_running_loop().run_in_executor(
+                None, server.embedder.encode, tok_lists
+            )
+            vecs = enc.cpu().tolist()
+        else:
+            vecs = await asyncio.get_running_loop().run_in_executor(
+                None, server.engine.embed, tok_lists
+            )
         data = [
             {"object": "embedding", "index": i, "embedding": v}
             for i, v in enumerate(vecs)
@@ -523,13 +553,18 @@ def build_app(server: EngineServer) -> FastAPI:
         docs = body.get("documents", []) or []
         top_n = body.get("top_n") or len(docs)
         tok_lists = [server.tokenizer.encode(t, add_bos=True) for t in [query] + docs]
-        vecs = await asyncio.get_running_loop().run_in_executor(
-            None, server.engine.embed, tok_lists
-        )
-        qv = vecs[0]
-        scores = [
-            sum(a * b for a, b in zip(qv, dv)) for dv in vecs[1:]
-        ]  # unit-norm vectors -> cosine
+        if server.embedder is not None:
+            scores = await asyncio.get_running_loop().run_in_executor(
+                None, server.embedder.score_pairs, tok_lists[0], tok_lists[1:]
+            )
+        else:
+            vecs = await asyncio.get_running_loop().run_in_executor(
+                None, server.engine.embed, tok_lists
+            )
+            qv = vecs[0]
+            scores = [
+                sum(a * b for a, b in zip(qv, dv)) for dv in vecs[1:]
+            ]  # unit-norm vectors -> cosine
         order = sorted(range(len(docs)), key=lambda i: -scores[i])[: int(top_n)]
         return {
             "model": name,
@@ -780,13 +815,19 @@ def main():
     p.add_argument("--num-gpu-blocks", type=int, default=None)
     p.add_argument("--enable-lora", action="store_true")
     p.add_argument("--tensor-parallel-size", type=int, default=1)
-    p.add_argument("--task", choices=["generate", "transcribe"], default=None,
-                   help="auto: whisper-* presets transcribe, others generate")
+    p.add_argument("--task", choices=["generate", "transcribe", "embed"],
+                   default=None,
+                   help="auto: whisper-* -> transcribe, bert/bge/e5 -> embed")
     args = p.parse_args()
-    task = args.task or (
-        "transcribe" if os.path.basename(args.model).startswith("whisper")
-        else "generate"
-    )
+    base = os.path.basename(args.model.rstrip("/")).lower()
+    if args.task:
+        task = args.task
+    elif base.startswith("whisper"):
+        task = "transcribe"
+    elif base.startswith(("bert", "bge", "e5")):
+        task = "embed"
+    else:
+        task = "generate"
 
     cfg = EngineConfig(
         model=args.model,

@@ -129,3 +129,51 @@ def test_local_runtime_speech_to_text():
             await mgr.stop()
 
     asyncio.run(body())
+
+
+def test_local_runtime_bert_embeddings():
+    """Pure TextEmbedding model: runtime launches the engine with
+    task=embed (BERT encoder); the gateway proxies /v1/embeddings."""
+    async def body():
+        cfg = SystemConfig(
+            autoscaling=AutoscalingConfig(
+                interval_seconds=3600, time_window_seconds=7200, state_path=None
+            ),
+            leader_lock_path="/tmp/kubeai-e2e-emb-leader.lock",
+            n_gpus=0,
+        )
+        mgr = Manager(cfg)
+        m = Model(
+            name="embedder",
+            spec=ModelSpec(
+                url="hf://test/bert-tiny",
+                features=["TextEmbedding", "Reranking"],
+                resource_profile="cpu:1",
+                min_replicas=1,
+                max_replicas=1,
+            ),
+        )
+        mgr.store.apply_model(m)
+        await mgr.start()
+        try:
+            for _ in range(1200):
+                reps = mgr.store.list_replicas("embedder")
+                if reps and reps[0].ready:
+                    break
+                await asyncio.sleep(0.1)
+            else:
+                raise AssertionError(f"replica never ready: {reps}")
+            transport = httpx.ASGITransport(app=mgr.app)
+            async with httpx.AsyncClient(
+                transport=transport, base_url="http://gw", timeout=60
+            ) as client:
+                r = await client.post(
+                    "/openai/v1/embeddings",
+                    json={"model": "embedder", "input": "embed me"},
+                )
+                assert r.status_code == 200, r.text
+                assert len(r.json()["data"][0]["embedding"]) == 128
+        finally:
+            await mgr.stop()
+
+    asyncio.run(body())

@@ -1,0 +1,107 @@
+"""BERT-architecture embedding/reranking engine (TextEmbedding/Reranking
+features; reference analog: the Infinity engine serving bge/e5 models —
+decoder-based embedding models are covered by LLMEngine.embed())."""
+import numpy as np
+import pytest
+import torch
+from fastapi.testclient import TestClient
+
+from kubeai_amd.models.bert import (
+    PRESETS,
+    BertEncoder,
+    load_weights_bert,
+    save_bert_checkpoint,
+)
+
+
+def test_encode_shapes_and_mask():
+    enc = BertEncoder(PRESETS["bert-tiny"], seed=0)
+    vecs = enc.encode([[5, 6, 7], [8, 9, 10, 11, 12, 13]])
+    assert vecs.shape == (2, 128)
+    assert torch.allclose(vecs.norm(dim=-1), torch.ones(2), atol=1e-4)
+    # padding must not change a sequence's embedding (mask correctness):
+    # same batch, one short + one long vs short alone
+    alone = enc.encode([[5, 6, 7]])
+    assert torch.allclose(vecs[0], alone[0], atol=1e-5)
+
+
+def test_encode_deterministic_and_content_sensitive():
+    enc = BertEncoder(PRESETS["bert-tiny"], seed=0)
+    a = enc.encode([[5, 6, 7, 8]])
+    b = enc.encode([[5, 6, 7, 8]])
+    c = enc.encode([[5, 6, 7, 9]])
+    assert torch.equal(a, b)
+    assert not torch.allclose(a, c)
+
+
+def test_checkpoint_roundtrip(tmp_path):
+    src = BertEncoder(PRESETS["bert-tiny"], seed=0)
+    ckpt = str(tmp_path / "bert")
+    save_bert_checkpoint(src, ckpt)
+    dst = BertEncoder(PRESETS["bert-tiny"], seed=777)
+    toks = [[4, 5, 6, 7, 8]]
+    assert not torch.allclose(dst.encode(toks), src.encode(toks))
+    n = load_weights_bert(dst, ckpt)
+    assert n > 10
+    assert torch.allclose(dst.encode(toks), src.encode(toks))
+
+
+@pytest.fixture(scope="module")
+def embed_client():
+    from kubeai_amd.engine.engine import EngineConfig
+    from kubeai_amd.engine.server import EngineServer, build_app
+
+    cfg = EngineConfig(model="bert-tiny", device="cpu", num_gpu_blocks=16)
+    server = EngineServer(cfg, "embed-model", task="embed")
+    server.start()
+    server._ready.wait(timeout=60)
+    with TestClient(build_app(server)) as c:
+        yield c
+    server.stop()
+
+
+def test_embeddings_endpoint(embed_client):
+    assert embed_client.get("/health").status_code == 200
+    r = embed_client.post(
+        "/v1/embeddings",
+        json={"model": "embed-model", "input": ["hello world", "other words"]},
+    )
+    assert r.status_code == 200
+    data = r.json()["data"]
+    assert len(data) == 2
+    v = data[0]["embedding"]
+    assert len(v) == 128  # bert-tiny hidden
+    assert abs(sum(x * x for x in v) ** 0.5 - 1.0) < 1e-3
+
+
+def test_rerank_endpoint(embed_client):
+    r = embed_client.post(
+        "/v1/rerank",
+        json={"query": "alpha beta", "top_n": 2,
+              "documents": ["alpha beta gamma", "unrelated words here",
+                            "alpha beta"]},
+    )
+    assert r.status_code == 200
+    results = r.json()["results"]
+    assert len(results) == 2
+    assert results[0]["relevance_score"] >= results[1]["relevance_score"]
+    # the token-overlapping docs must outrank the unrelated one
+    assert 1 not in {res["index"] for res in results}
+
+
+def test_text_endpoints_rejected(embed_client):
+    r = embed_client.post("/v1/completions",
+                          json={"prompt": "x", "max_tokens": 1})
+    assert r.status_code == 400 and "task=embed" in r.text
+
+
+@pytest.mark.gpu
+def test_bert_gpu_matches_cpu():
+    enc_g = BertEncoder(PRESETS["bert-tiny"], device="cuda", seed=0)
+    enc_c = BertEncoder(PRESETS["bert-tiny"], device="cpu", seed=0)
+    toks = [[5, 6, 7, 8, 9]]
+    vg, vc = enc_g.encode(toks).cpu(), enc_c.encode(toks)
+    cos = torch.nn.functional.cosine_similarity(
+        vg.flatten(), vc.flatten(), dim=0
+    )
+    assert cos > 0.999

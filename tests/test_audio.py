@@ -116,7 +116,7 @@ def test_transcription_errors(stt_client):
     assert r.status_code == 400
     # text endpoints answer a clear 400 on an STT model
     r = stt_client.post("/v1/completions", json={"prompt": "x", "max_tokens": 1})
-    assert r.status_code == 400 and "speech-to-text" in r.text
+    assert r.status_code == 400 and "task=transcribe" in r.text
 
 
 @pytest.mark.gpu

@@ -31,6 +31,8 @@ class ModelArchConfig:
     num_experts_per_tok: int = 2
     # Qwen2-style: q/k/v projections carry a bias
     attention_bias: bool = False
+    # HF rope_scaling dict (llama3 / linear), None = unscaled
+    rope_scaling: dict | None = None
 
     @property
     def n_kv_heads(self) -> int:
@@ -67,18 +69,30 @@ class ModelArchConfig:
             eos_token_id=eos,
             num_local_experts=cfg.get("num_local_experts", 0),
             num_experts_per_tok=cfg.get("num_experts_per_tok", 2),
+            rope_scaling=cfg.get("rope_scaling"),
         )
 
 
 # Presets (BASELINE.json configs + test-size models)
 PRESETS: dict[str, ModelArchConfig] = {
-    "llama-3-8b": ModelArchConfig(),
+    "llama-3-8b": ModelArchConfig(
+        # Llama-3.1-8B-Instruct shape (the BASELINE model): 128k context
+        # via llama3 rope scaling
+        max_position_embeddings=131072,
+        rope_scaling={"rope_type": "llama3", "factor": 8.0,
+                      "low_freq_factor": 1.0, "high_freq_factor": 4.0,
+                      "original_max_position_embeddings": 8192},
+    ),
     "llama-3-70b": ModelArchConfig(
         hidden_size=8192,
         intermediate_size=28672,
         num_hidden_layers=80,
         num_attention_heads=64,
         num_key_value_heads=8,
+        max_position_embeddings=131072,
+        rope_scaling={"rope_type": "llama3", "factor": 8.0,
+                      "low_freq_factor": 1.0, "high_freq_factor": 4.0,
+                      "original_max_position_embeddings": 8192},
     ),
     # tiny: CPU tests / smoke; same head_dim=128 so HIP kernel paths match
     # Qwen2 family (also the DeepSeek-R1-Distill-Qwen arch): qkv bias,

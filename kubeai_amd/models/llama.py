@@ -260,7 +260,8 @@ class LlamaForCausalLM(nn.Module):
         self.register_buffer(
             "cos_sin",
             ops_ref.make_cos_sin_cache(
-                cfg.head_dim, cfg.max_position_embeddings, cfg.rope_theta
+                cfg.head_dim, cfg.max_position_embeddings, cfg.rope_theta,
+                rope_scaling=cfg.rope_scaling,
             ).to(device=device),
             persistent=False,
         )

@@ -320,6 +320,7 @@ def save_hf_checkpoint(model, out_dir: str) -> None:
                 "num_local_experts": cfg.num_local_experts,
                 "num_experts_per_tok": cfg.num_experts_per_tok,
                 "attention_bias": cfg.attention_bias,
+                "rope_scaling": cfg.rope_scaling,
                 "torch_dtype": "bfloat16",
             },
             f,

@@ -18,6 +18,8 @@ Shapes / layout conventions (MI355X-first):
 """
 from __future__ import annotations
 
+import math
+
 import torch
 
 
@@ -56,11 +58,38 @@ def make_cos_sin_cache(
     base: float = 10000.0,
     scaling: float = 1.0,
     dtype: torch.dtype = torch.float32,
+    rope_scaling: dict | None = None,
 ) -> torch.Tensor:
-    """[max_positions, head_dim] — first half cos, second half sin (NeoX)."""
+    """[max_positions, head_dim] — first half cos, second half sin (NeoX).
+
+    rope_scaling follows the HF config.json `rope_scaling` dict:
+      {"rope_type": "llama3", "factor", "low_freq_factor",
+       "high_freq_factor", "original_max_position_embeddings"}  (Llama 3.1)
+      {"rope_type": "linear", "factor"}                         (PI)
+    """
     inv_freq = 1.0 / (
         base ** (torch.arange(0, head_dim, 2, dtype=torch.float64) / head_dim)
     )
+    if rope_scaling:
+        rtype = rope_scaling.get("rope_type") or rope_scaling.get("type")
+        factor = float(rope_scaling.get("factor", 1.0))
+        if rtype == "llama3":
+            lo_f = float(rope_scaling.get("low_freq_factor", 1.0))
+            hi_f = float(rope_scaling.get("high_freq_factor", 4.0))
+            orig = float(
+                rope_scaling.get("original_max_position_embeddings", 8192)
+            )
+            wavelen = 2 * math.pi / inv_freq
+            lo_wl = orig / lo_f  # long wavelengths: fully rescale
+            hi_wl = orig / hi_f  # short wavelengths: keep
+            smooth = ((orig / wavelen - lo_f) / (hi_f - lo_f)).clamp(0.0, 1.0)
+            scaled = (1 - smooth) * inv_freq / factor + smooth * inv_freq
+            inv_freq = torch.where(wavelen > lo_wl, inv_freq / factor, inv_freq)
+            mid = (wavelen <= lo_wl) & (wavelen >= hi_wl)
+            inv_freq = torch.where(mid, scaled, inv_freq)
+        elif rtype == "linear":
+            scaling = scaling * factor
+        # other types (dynamic/yarn) fall through unscaled
     t = torch.arange(max_positions, dtype=torch.float64) / scaling
     freqs = torch.outer(t, inv_freq)
     return torch.cat([freqs.cos(), freqs.sin()], dim=-1).to(dtype)

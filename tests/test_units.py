@@ -344,3 +344,36 @@ def test_load_config_priority_classes(tmp_path):
     cfg = load_config(str(p))
     assert cfg.priority_classes == {"critical": 1000, "batch": -10}
     assert cfg.messaging[0].requests_url.startswith("file://")
+
+
+def test_llama3_rope_scaling():
+    """llama3 rope scaling (Llama-3.1): high-frequency bands unchanged,
+    low-frequency bands stretched by the factor, smooth in between."""
+    import math
+
+    import torch
+
+    from kubeai_amd.ops.ref import make_cos_sin_cache
+
+    hd, base = 128, 500000.0
+    plain = make_cos_sin_cache(hd, 4096, base)
+    scaled = make_cos_sin_cache(
+        hd, 4096, base,
+        rope_scaling={"rope_type": "llama3", "factor": 8.0,
+                      "low_freq_factor": 1.0, "high_freq_factor": 4.0,
+                      "original_max_position_embeddings": 8192},
+    )
+    # fastest-rotating dims (short wavelength << 2048) identical
+    assert torch.allclose(plain[:, :8], scaled[:, :8])
+    assert torch.allclose(plain[:, 64:72], scaled[:, 64:72])
+    # slowest dim: angle compressed ~8x -> cos stays near 1 much longer
+    p = 4095
+    inv = 1.0 / (base ** ((hd - 2) / hd))
+    ang_plain = math.acos(float(plain[p, 63]))
+    ang_scaled = math.acos(float(scaled[p, 63]))
+    assert abs(ang_plain / max(ang_scaled, 1e-9) - 8.0) < 0.5
+    # linear scaling: all angles divided by factor
+    lin = make_cos_sin_cache(hd, 128, base,
+                             rope_scaling={"rope_type": "linear", "factor": 2.0})
+    half = make_cos_sin_cache(hd, 256, base)
+    assert torch.allclose(lin[64], half[32], atol=1e-6)

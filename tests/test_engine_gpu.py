@@ -246,3 +246,18 @@ def test_fp8_moe_engine():
     cos = torch.nn.functional.cosine_similarity(a.flatten(), b.flatten(), dim=0)
     assert cos > 0.95, f"MoE fp8 cos {cos}"
     assert torch.isfinite(b).all()
+
+
+def test_qwen2_engine_gpu():
+    """Qwen2 arch on the HIP path: qkv bias + tied embeddings generate
+    deterministically and match a fresh engine's stream."""
+    a = make_engine("cuda", model="qwen2-tiny", seed=11)
+    b = make_engine("cuda", model="qwen2-tiny", seed=11)
+    prompt = list(range(10, 100))
+    for eng, rid in ((a, "x"), (b, "y")):
+        eng.add_request(prompt, SamplingParams(max_tokens=8), request_id=rid)
+    oa = drain(a)["x"]
+    ob = drain(b)["y"]
+    assert len(oa.output_token_ids) == 8
+    assert oa.output_token_ids == ob.output_token_ids
+    assert a.runner.model.layers[0].self_attn.qkv_proj.bias is not None

@@ -255,7 +255,8 @@ def test_qwen2_engine_gpu():
     b = make_engine("cuda", model="qwen2-tiny", seed=11)
     prompt = list(range(10, 100))
     for eng, rid in ((a, "x"), (b, "y")):
-        eng.add_request(prompt, SamplingParams(max_tokens=8), request_id=rid)
+        eng.add_request(prompt, SamplingParams(max_tokens=8, ignore_eos=True),
+                        request_id=rid)
     oa = drain(a)["x"]
     ob = drain(b)["y"]
     assert len(oa.output_token_ids) == 8

@@ -297,3 +297,38 @@ def test_qwen2_arch_engine():
         want, _ = gen("qwen2-tiny", seed=8, rid="w")
         got, _ = gen(d, seed=999, rid="g")
         assert got == want
+
+
+def test_lora_on_moe_model():
+    """BASELINE config #5 (Mixtral + hot-swapped LoRA): adapters target the
+    attention projections, so they apply to the MoE model unchanged —
+    adapter stream diverges, base stream is untouched."""
+    import torch
+
+    eng = LLMEngine(
+        EngineConfig(model="mixtral-tiny", device="cpu", num_gpu_blocks=128,
+                     max_model_len=512, seed=2)
+    )
+    prompt = list(range(10, 90))
+
+    def last_logits(lora_id):
+        from kubeai_amd.engine.kvcache import BlockManager
+        from kubeai_amd.engine.scheduler import Request, Scheduler
+
+        s = Scheduler(BlockManager(64, 16), max_num_batched_tokens=512,
+                      max_model_len=256)
+        s.add_request(Request(prompt[:48], SamplingParams(max_tokens=1),
+                              request_id="x", lora_id=lora_id))
+        out = s.schedule()
+        fb = eng.runner.build_batch(out)
+        h = eng.runner.model(fb)
+        return eng.runner.model.compute_logits(h[fb.logits_indices]).float()
+
+    base_before = last_logits(0)
+    eng.load_lora(9, None)  # synthetic rank-16 adapter
+    base_after = last_logits(0)
+    adapted = last_logits(9)
+    assert torch.equal(base_before, base_after)  # base stream untouched
+    assert not torch.allclose(base_after, adapted)  # adapter is live
+    eng.unload_lora(9)
+    assert torch.equal(last_logits(9), base_before)  # unload restores

@@ -47,6 +47,19 @@ class MemBroker:
     async def receive(self, topic: str) -> bytes:
         return await self.topic(topic).get()
 
+    async def receive_with_ack(self, topic: str):
+        """(payload, ack, nack): nack requeues for redelivery (the
+        reference's msg.Nack() analog, messenger.go ack/nack flow)."""
+        payload = await self.topic(topic).get()
+
+        def ack() -> None:
+            pass
+
+        def nack() -> None:
+            self.topic(topic).put_nowait(payload)
+
+        return payload, ack, nack
+
 
 class FileBroker:
     """Durable cross-process pubsub over a shared directory (file:// driver).
@@ -78,7 +91,9 @@ class FileBroker:
             f.write(payload)
         os.replace(tmp, os.path.join(d, name))
 
-    async def receive(self, topic: str) -> bytes:
+    async def receive_with_ack(self, topic: str):
+        """(payload, ack, nack): the claim is the rename; ack deletes the
+        claimed file, nack renames it back (redelivery, original order)."""
         d = self._topic_dir(topic)
         while True:
             for name in sorted(os.listdir(d)):
@@ -92,9 +107,26 @@ class FileBroker:
                     continue
                 with open(claimed, "rb") as f:
                     payload = f.read()
-                os.unlink(claimed)
-                return payload
+
+                def ack(p=claimed):
+                    try:
+                        os.unlink(p)
+                    except OSError:
+                        pass
+
+                def nack(p=claimed, orig=src):
+                    try:
+                        os.rename(p, orig)
+                    except OSError:
+                        pass
+
+                return payload, ack, nack
             await asyncio.sleep(self.poll_interval)
+
+    async def receive(self, topic: str) -> bytes:
+        payload, ack, _ = await self.receive_with_ack(topic)
+        ack()
+        return payload
 
 
 def broker_from_url(url: str):
@@ -159,11 +191,18 @@ class Messenger:
 
     async def _loop(self) -> None:
         while True:
-            payload = await self.broker.receive(self.requests_topic)
+            if hasattr(self.broker, "receive_with_ack"):
+                payload, ack, nack = await self.broker.receive_with_ack(
+                    self.requests_topic
+                )
+            else:
+                payload = await self.broker.receive(self.requests_topic)
+                ack = nack = lambda: None
             await self.sem.acquire()
-            asyncio.create_task(self._handle(payload))
+            asyncio.create_task(self._handle(payload, ack, nack))
 
-    async def _handle(self, payload: bytes) -> None:
+    async def _handle(self, payload: bytes, ack=lambda: None,
+                      nack=lambda: None) -> None:
         try:
             metadata, status, body = await self.handle_request(payload)
             out = json.dumps(
@@ -171,7 +210,11 @@ class Messenger:
             ).encode()
             await self.broker.publish(self.responses_topic, out)
             self.consecutive_errors = 0
+            ack()
         except Exception:  # noqa: BLE001
+            # nack -> redelivery (reference: messenger.go nack + linear
+            # consecutive-error backoff)
+            nack()
             self.consecutive_errors += 1
             await asyncio.sleep(min(self.consecutive_errors, 3))
         finally:

@@ -377,3 +377,59 @@ def test_llama3_rope_scaling():
                              rope_scaling={"rope_type": "linear", "factor": 2.0})
     half = make_cos_sin_cache(hd, 256, base)
     assert torch.allclose(lin[64], half[32], atol=1e-6)
+
+
+def test_broker_nack_redelivery(tmp_path):
+    """nack returns the message to the queue for redelivery (both drivers)."""
+    import asyncio
+
+    from kubeai_amd.controlplane.messenger import FileBroker, MemBroker
+
+    async def check(broker):
+        await broker.publish("t", b"m1")
+        p, ack, nack = await broker.receive_with_ack("t")
+        assert p == b"m1"
+        nack()
+        p2, ack2, _ = await broker.receive_with_ack("t")
+        assert p2 == b"m1"  # redelivered
+        ack2()
+        await broker.publish("t", b"m2")
+        p3, ack3, _ = await broker.receive_with_ack("t")
+        assert p3 == b"m2"  # m1 gone after ack
+        ack3()
+
+    asyncio.run(check(MemBroker()))
+    asyncio.run(check(FileBroker(str(tmp_path / "q"), poll_interval=0.01)))
+
+
+def test_messenger_nacks_failed_handling(tmp_path):
+    """A handler exception nacks the message; the next receive loop retries
+    it and succeeds (at-least-once semantics, reference messenger.go)."""
+    import asyncio
+
+    from kubeai_amd.controlplane.messenger import FileBroker, Messenger
+
+    async def run():
+        broker = FileBroker(str(tmp_path / "bus"), poll_interval=0.01)
+        m = Messenger(broker, "req", "resp", model_client=None, lb=None,
+                      max_handlers=2)
+        calls = {"n": 0}
+
+        async def flaky(payload):
+            calls["n"] += 1
+            if calls["n"] == 1:
+                raise RuntimeError("transient")
+            return {"id": "1"}, 200, {"ok": True}
+
+        m.handle_request = flaky
+        m.consecutive_errors = 0
+        m.start()
+        try:
+            await broker.publish("req", b'{"metadata":{"id":"1"}}')
+            out = await asyncio.wait_for(broker.receive("resp"), timeout=15)
+            assert b'"status_code": 200' in out
+            assert calls["n"] == 2  # failed once, redelivered, succeeded
+        finally:
+            await m.stop()
+
+    asyncio.run(run())

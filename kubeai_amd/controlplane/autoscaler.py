@@ -30,6 +30,13 @@ _METRIC_RE = re.compile(
 _ENGINE_QUEUE_RE = re.compile(
     r'^kubeai_engine_num_requests_waiting\{model="([^"]+)"\}\s+([0-9.eE+-]+)',
 )
+_ENGINE_KV_RE = re.compile(
+    r'^kubeai_engine_kv_cache_usage_perc\{model="([^"]+)"\}\s+([0-9.eE+-]+)',
+)
+# KV pressure: a replica whose paged KV pool is nearly full is about to
+# preempt/queue even if nothing waits yet — count it as one extra
+# replica's worth of load (north-star signal: queue depth + KV occupancy)
+KV_PRESSURE_THRESHOLD = 0.95
 
 
 class Autoscaler:
@@ -83,8 +90,9 @@ class Autoscaler:
     # ------------------------------------------------------------ one tick
     async def tick(self) -> None:
         active = await self._aggregate_active_requests()
+        kv_pressured: dict[str, int] = {}
         if self.scrape_engine_queues:
-            queued = await self._aggregate_engine_queues()
+            queued, kv_pressured = await self._aggregate_engine_queues()
             for m, v in queued.items():
                 active[m] = active.get(m, 0.0) + v
         for model in self.model_client.list_all_models():
@@ -92,6 +100,12 @@ class Autoscaler:
                 continue
             avg = self._avg_for(model.name).next(active.get(model.name, 0.0))
             target = math.ceil(avg / max(model.spec.target_requests, 1))
+            # P replicas at >=95% KV occupancy -> want at least P+1 so new
+            # prefixes stop evicting hot cache (scale-down hysteresis in
+            # ModelClient.scale keeps this from flapping)
+            pressured = kv_pressured.get(model.name, 0)
+            if pressured:
+                target = max(target, pressured + 1)
             self.last_scales[model.name] = target
             self.model_client.scale(model.name, target)
         if self.state_path:
@@ -131,10 +145,11 @@ class Autoscaler:
                     totals[m.group(1)] = totals.get(m.group(1), 0.0) + float(m.group(2))
         return totals
 
-    async def _aggregate_engine_queues(self) -> dict[str, float]:
-        """Per-replica engine queue depth — the BASELINE.json north-star
-        signal (engine /metrics: waiting + running)."""
+    async def _aggregate_engine_queues(self) -> tuple[dict, dict]:
+        """Per-replica engine queue depth + KV-pressured replica counts —
+        the BASELINE.json north-star signals (engine /metrics)."""
         totals: dict[str, float] = {}
+        pressured: dict[str, int] = {}
         for rep in self.store.list_replicas():
             if not rep.ready or not rep.address:
                 continue
@@ -146,7 +161,11 @@ class Autoscaler:
                 m = _ENGINE_QUEUE_RE.match(line)
                 if m:
                     totals[rep.model] = totals.get(rep.model, 0.0) + float(m.group(2))
-        return totals
+                    continue
+                m = _ENGINE_KV_RE.match(line)
+                if m and float(m.group(2)) >= KV_PRESSURE_THRESHOLD:
+                    pressured[rep.model] = pressured.get(rep.model, 0) + 1
+        return totals, pressured
 
     # ------------------------------------------------------------ state
     def _save_state(self) -> None:

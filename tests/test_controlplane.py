@@ -42,6 +42,8 @@ class FakeBackend:
 
     def __init__(self):
         self.requests = []
+        self.waiting = 0      # engine queue depth exposed on /metrics
+        self.kv_usage = 0.0   # KV occupancy exposed on /metrics
         self.fail_next = 0
         self.loaded_adapters = set()
         self.port = free_port()
@@ -84,7 +86,12 @@ class FakeBackend:
             return JSONResponse({"status": "ok"})
 
         async def metrics(request):
-            return JSONResponse({})
+            from starlette.responses import PlainTextResponse
+
+            return PlainTextResponse(
+                f'kubeai_engine_num_requests_waiting{{model="m"}} {self.waiting}\n'
+                f'kubeai_engine_kv_cache_usage_perc{{model="m"}} {self.kv_usage}\n'
+            )
 
         async def transcriptions(request):
             raw = await request.body()
@@ -675,5 +682,28 @@ def test_messenger_flow_file_broker(tmp_path):
                 assert out["status_code"] == 200
             finally:
                 await msgr.stop()
+
+    run(body())
+
+
+def test_autoscaler_kv_pressure_scales_up():
+    """A replica at >=95% KV occupancy counts as one extra replica of load
+    even with an empty queue (north-star: queue depth + KV occupancy)."""
+    async def body():
+        m = text_gen_model("kvm")
+        m.spec.replicas = 1
+        m.spec.max_replicas = 4
+        m.spec.target_requests = 10
+        async with harness([m], autoscaler_interval=0.05) as (
+            mgr, runtime, backend,
+        ):
+            rep = (await wait_for(lambda: mgr.store.list_replicas("kvm")))[0]
+            runtime.mark_ready(rep.name, backend.address)
+            backend.kv_usage = 0.99  # full KV pool, nothing waiting
+            # moving average fills toward target_requests -> ceil -> 2
+            await wait_for(
+                lambda: mgr.store.get_model("kvm").spec.replicas >= 2,
+                timeout=10,
+            )
 
     run(body())

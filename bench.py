@@ -74,6 +74,9 @@ def get_args():
     p.add_argument("--gpu-mem-util", type=float, default=0.90)
     p.add_argument("--max-model-len", type=int, default=8192)
     p.add_argument("--prefill-interval", type=int, default=4)
+    p.add_argument("--kv-cache-dtype", choices=["auto", "fp8_e5m2"],
+                   default="auto",
+                   help="fp8_e5m2 halves KV bytes (pending device validation)")
     p.add_argument(
         "--quant",
         default="auto",
@@ -197,6 +200,7 @@ def main():
             enable_prefix_caching=not args.no_prefix_cache,
             prefill_interval=args.prefill_interval,
             quantization=_resolve_quant(args, model, use_cuda),
+            kv_cache_dtype=args.kv_cache_dtype,
             seed=replica_seed,
         ),
         tp_group=tp_group,

@@ -32,12 +32,14 @@ constexpr int kHD = 128;  // head dim
 // (4-way residual aliasing ~= 1.6x on the LDS op, vs 32-way unpadded)
 constexpr int kPad = 8;
 
-template <int G>
+// CT = cache element type: ushort (bf16) or unsigned char (fp8 e5m2,
+// converted to bf16 in-register while staging — same LDS layout/compute)
+template <int G, typename CT = ushort>
 __launch_bounds__(kBlockThreads) __global__ void paged_decode_kernel(
     ushort* __restrict__ out,            // [B, n_q, hd] bf16
     const ushort* __restrict__ q,        // [B, n_q, hd] bf16
-    const ushort* __restrict__ k_cache,  // [nb, n_kv, bs, hd]
-    const ushort* __restrict__ v_cache,
+    const CT* __restrict__ k_cache,      // [nb, n_kv, bs, hd]
+    const CT* __restrict__ v_cache,
     const int32_t* __restrict__ block_tables,  // [B, max_blocks]
     const int32_t* __restrict__ seq_lens,      // [B]
     const float scale, const int n_kv, const int max_blocks,
@@ -101,14 +103,32 @@ __launch_bounds__(kBlockThreads) __global__ void paged_decode_kernel(
 
   auto issue_loads = [&](int blk_i) {
     const int64_t blk = bt[blk_i];
-    const ushort8* src_k = reinterpret_cast<const ushort8*>(
-        k_cache + ((blk * n_kv + kh) * kBS) * kHD);
-    const ushort8* src_v = reinterpret_cast<const ushort8*>(
-        v_cache + ((blk * n_kv + kh) * kBS) * kHD);
+    const CT* base_k = k_cache + ((blk * n_kv + kh) * kBS) * kHD;
+    const CT* base_v = v_cache + ((blk * n_kv + kh) * kBS) * kHD;
+    if constexpr (sizeof(CT) == 2) {
+      const ushort8* src_k = reinterpret_cast<const ushort8*>(base_k);
+      const ushort8* src_v = reinterpret_cast<const ushort8*>(base_v);
 #pragma unroll
-    for (int i = 0; i < (kBS * kHD / 8) / WAVE_SIZE; ++i) {
-      stage_k[i] = src_k[lane + i * WAVE_SIZE];
-      stage_v[i] = src_v[lane + i * WAVE_SIZE];
+      for (int i = 0; i < (kBS * kHD / 8) / WAVE_SIZE; ++i) {
+        stage_k[i] = src_k[lane + i * WAVE_SIZE];
+        stage_v[i] = src_v[lane + i * WAVE_SIZE];
+      }
+    } else {
+      // fp8: 8-byte lane loads (one 8-elem staging unit each), convert
+      // to bf16 in-register; LDS layout and compute stay identical
+      const unsigned char* bk = reinterpret_cast<const unsigned char*>(base_k);
+      const unsigned char* bv = reinterpret_cast<const unsigned char*>(base_v);
+#pragma unroll
+      for (int i = 0; i < (kBS * kHD / 8) / WAVE_SIZE; ++i) {
+        uint64_t kraw = reinterpret_cast<const uint64_t*>(
+            bk)[lane + i * WAVE_SIZE];
+        uint64_t vraw = reinterpret_cast<const uint64_t*>(
+            bv)[lane + i * WAVE_SIZE];
+        stage_k[i] = e5m2x8_to_bf16x8(
+            reinterpret_cast<const unsigned char*>(&kraw));
+        stage_v[i] = e5m2x8_to_bf16x8(
+            reinterpret_cast<const unsigned char*>(&vraw));
+      }
     }
   };
   auto write_tile = [&](int buf) {
@@ -292,6 +312,8 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
   TORCH_CHECK(out.is_contiguous());
   TORCH_CHECK(k_cache.is_contiguous() && v_cache.is_contiguous());
   TORCH_CHECK(q.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(k_cache.scalar_type() == torch::kBFloat16 ||
+              k_cache.scalar_type() == torch::kFloat8_e5m2);
   TORCH_CHECK(block_tables.scalar_type() == torch::kInt32);
   TORCH_CHECK(seq_lens.scalar_type() == torch::kInt32);
   const int B = q.size(0), n_q = q.size(1), hd = q.size(2);
@@ -341,15 +363,25 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
   }
   dim3 grid(B * n_kv * n_splits), block(kBlockThreads);
   auto stream = c10::hip::getCurrentHIPStream().stream();
-#define LAUNCH(GG)                                                        \
-  hipLaunchKernelGGL((paged_decode_kernel<GG>), grid, block, 0, stream,   \
-                     (ushort*)out.data_ptr(), (const ushort*)q.data_ptr(), \
-                     (const ushort*)k_cache.data_ptr(),                    \
-                     (const ushort*)v_cache.data_ptr(),                    \
-                     block_tables.data_ptr<int32_t>(),                     \
-                     seq_lens.data_ptr<int32_t>(), (float)scale, n_kv,     \
-                     max_blocks, q.stride(0), n_splits, part_o_ptr,        \
+  const bool fp8_cache = k_cache.scalar_type() == torch::kFloat8_e5m2;
+#define LAUNCH_CT(GG, CT)                                                 \
+  hipLaunchKernelGGL((paged_decode_kernel<GG, CT>), grid, block, 0,       \
+                     stream, (ushort*)out.data_ptr(),                     \
+                     (const ushort*)q.data_ptr(),                         \
+                     (const CT*)k_cache.data_ptr(),                       \
+                     (const CT*)v_cache.data_ptr(),                       \
+                     block_tables.data_ptr<int32_t>(),                    \
+                     seq_lens.data_ptr<int32_t>(), (float)scale, n_kv,    \
+                     max_blocks, q.stride(0), n_splits, part_o_ptr,       \
                      part_ml_ptr)
+#define LAUNCH(GG)                                                        \
+  do {                                                                    \
+    if (fp8_cache) {                                                      \
+      LAUNCH_CT(GG, unsigned char);                                       \
+    } else {                                                              \
+      LAUNCH_CT(GG, ushort);                                              \
+    }                                                                     \
+  } while (0)
   switch (G) {
     case 1: LAUNCH(1); break;
     case 2: LAUNCH(2); break;
@@ -360,6 +392,7 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
     default: TORCH_CHECK(false, "unsupported GQA group size ", G);
   }
 #undef LAUNCH
+#undef LAUNCH_CT
   HIP_CHECK_KERNEL();
   if (n_splits > 1) {
     const int64_t n_bh = (int64_t)B * n_q;

@@ -42,12 +42,14 @@ DEV_INLINE int swz(int row, int byte_off) {
   return byte_off ^ ((row & 7) << 4);
 }
 
-template <int G>
+// CT = cache element type: ushort (bf16) or unsigned char (fp8 e5m2,
+// converted to bf16 while staging — LDS layout and MFMA path unchanged)
+template <int G, typename CT = ushort>
 __launch_bounds__(G * WAVE_SIZE) __global__ void paged_prefill_kernel(
     ushort* __restrict__ out,            // [Tq, n_q, hd]
     const ushort* __restrict__ q,        // [Tq, n_q, hd]
-    const ushort* __restrict__ k_cache,  // [nb, n_kv, bs, hd]
-    const ushort* __restrict__ v_cache,
+    const CT* __restrict__ k_cache,      // [nb, n_kv, bs, hd]
+    const CT* __restrict__ v_cache,
     const int32_t* __restrict__ block_tables,     // [B, max_blocks]
     const int32_t* __restrict__ query_start_loc,  // [B+1]
     const int32_t* __restrict__ seq_lens,         // [B]
@@ -134,17 +136,33 @@ __launch_bounds__(G * WAVE_SIZE) __global__ void paged_prefill_kernel(
         if (tok < kv_valid) {
           const int abs_tok = kv_start + tok;
           const int64_t blk = bt[abs_tok / kBS];
-          const ushort* src = k_cache +
-                              (((blk * n_kv + kh) * kBS) + abs_tok % kBS) * kHD;
+          const CT* src = k_cache +
+                          (((blk * n_kv + kh) * kBS) + abs_tok % kBS) * kHD;
+          const CT* vsrc = v_cache +
+                           (((blk * n_kv + kh) * kBS) + abs_tok % kBS) * kHD;
+          bf16x8 kk, vv;
+          if constexpr (sizeof(CT) == 2) {
+            kk = *reinterpret_cast<const bf16x8*>(src + col8 * 8);
+            vv = *reinterpret_cast<const bf16x8*>(vsrc + col8 * 8);
+          } else {
+            // fp8: 8-byte loads, in-register e5m2 -> bf16 conversion
+            uint64_t kraw = *reinterpret_cast<const uint64_t*>(src + col8 * 8);
+            uint64_t vraw = *reinterpret_cast<const uint64_t*>(vsrc + col8 * 8);
+            const ushort8 kc = e5m2x8_to_bf16x8(
+                reinterpret_cast<const unsigned char*>(&kraw));
+            const ushort8 vc = e5m2x8_to_bf16x8(
+                reinterpret_cast<const unsigned char*>(&vraw));
+#pragma unroll
+            for (int j = 0; j < 8; ++j) {
+              kk[j] = (short)kc[j];
+              vv[j] = (short)vc[j];
+            }
+          }
           // K: 16B vector write, swizzled row=tok
           *reinterpret_cast<bf16x8*>(
               reinterpret_cast<char*>(k_lds) +
-              swz(tok, tok * kHD * 2 + col8 * 16)) =
-              *reinterpret_cast<const bf16x8*>(src + col8 * 8);
+              swz(tok, tok * kHD * 2 + col8 * 16)) = kk;
           // V: scatter-transpose 8 elems (row=hd, col=tok)
-          const ushort* vsrc = v_cache +
-                               (((blk * n_kv + kh) * kBS) + abs_tok % kBS) * kHD;
-          bf16x8 vv = *reinterpret_cast<const bf16x8*>(vsrc + col8 * 8);
 #pragma unroll
           for (int j = 0; j < 8; ++j) {
             const int hdrow = col8 * 8 + j;
@@ -283,16 +301,25 @@ void paged_attention_prefill(torch::Tensor out, torch::Tensor q,
   const int n_qtiles_max = (Tq + kQB - 1) / kQB;  // per-seq early exit
   dim3 grid(B, n_kv, n_qtiles_max);
   auto stream = c10::hip::getCurrentHIPStream().stream();
-#define LAUNCH(GG)                                                         \
-  hipLaunchKernelGGL((paged_prefill_kernel<GG>), grid,                      \
+  const bool fp8_cache = k_cache.scalar_type() == torch::kFloat8_e5m2;
+#define LAUNCH_CT(GG, CT)                                                  \
+  hipLaunchKernelGGL((paged_prefill_kernel<GG, CT>), grid,                  \
                      dim3(GG * WAVE_SIZE), 0, stream,                       \
                      (ushort*)out.data_ptr(), (const ushort*)q.data_ptr(),  \
-                     (const ushort*)k_cache.data_ptr(),                     \
-                     (const ushort*)v_cache.data_ptr(),                     \
+                     (const CT*)k_cache.data_ptr(),                         \
+                     (const CT*)v_cache.data_ptr(),                         \
                      block_tables.data_ptr<int32_t>(),                      \
                      query_start_loc.data_ptr<int32_t>(),                   \
                      seq_lens.data_ptr<int32_t>(), (float)scale, n_kv,      \
                      max_blocks, q.stride(0))
+#define LAUNCH(GG)                                                         \
+  do {                                                                     \
+    if (fp8_cache) {                                                       \
+      LAUNCH_CT(GG, unsigned char);                                        \
+    } else {                                                               \
+      LAUNCH_CT(GG, ushort);                                               \
+    }                                                                      \
+  } while (0)
   switch (G) {
     case 1: LAUNCH(1); break;
     case 2: LAUNCH(2); break;
@@ -301,5 +328,6 @@ void paged_attention_prefill(torch::Tensor out, torch::Tensor q,
     default: TORCH_CHECK(false, "unsupported GQA group size ", G);
   }
 #undef LAUNCH
+#undef LAUNCH_CT
   HIP_CHECK_KERNEL();
 }

@@ -7,6 +7,9 @@ void fused_add_rmsnorm(torch::Tensor x, torch::Tensor residual,
                        torch::Tensor weight, double eps);
 void rope(torch::Tensor q, torch::Tensor k, torch::Tensor positions,
           torch::Tensor cos_sin);
+void reshape_and_cache_fp8(torch::Tensor k, torch::Tensor v,
+                           torch::Tensor k_cache, torch::Tensor v_cache,
+                           torch::Tensor slot_mapping);
 void reshape_and_cache(torch::Tensor k, torch::Tensor v, torch::Tensor k_cache,
                        torch::Tensor v_cache, torch::Tensor slot_mapping);
 void paged_attention_decode(torch::Tensor out, torch::Tensor q,
@@ -42,6 +45,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "in-place residual add + RMSNorm");
   m.def("rope", &rope, "NeoX rotary embedding, in-place q/k");
   m.def("reshape_and_cache", &reshape_and_cache, "paged KV cache write");
+  m.def("reshape_and_cache_fp8", &reshape_and_cache_fp8,
+        "paged KV cache write, fp8 e5m2 cache");
   m.def("paged_attention_decode", &paged_attention_decode,
         "paged attention, one query token per seq");
   m.def("paged_attention_prefill", &paged_attention_prefill,

@@ -9,6 +9,7 @@
 
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
+#include <hip/hip_fp8.h>
 #include <stdint.h>
 #include <c10/hip/HIPStream.h>
 
@@ -42,6 +43,50 @@ DEV_INLINE ushort f32_to_bf16(float f) {
   uint32_t rb = ((x >> 16) & 1u) + 0x7fffu;
   x += rb;
   return (ushort)(x >> 16);
+}
+
+// ---------------- fp8 e5m2 KV-cache element (OCP bf8) ----------------
+// e5m2 [s eeeee mm] -> f32 by bit widening: bias 15 -> 127. Matches
+// torch.float8_e5m2 exactly incl. subnormals/inf/nan (pure exponent
+// remap; subnormals handled by normalizing the 2-bit mantissa).
+
+DEV_INLINE float e5m2_to_f32(unsigned char b) {
+  const uint32_t s = ((uint32_t)b & 0x80u) << 24;
+  uint32_t e = (b >> 2) & 0x1fu;
+  uint32_t m = b & 0x3u;
+  uint32_t out;
+  if (e == 0) {
+    if (m == 0) {
+      out = s;  // +-0
+    } else {
+      // subnormal: value = m * 2^-16; normalize
+      int shift = (m & 2u) ? 0 : 1;  // m=2,3 -> msb at bit1; m=1 -> bit0
+      e = 127 - 15 + 1 - 1 - shift;  // 2^-15 / 2^-16
+      m = (m << (shift + 1)) & 0x3u;  // drop the implicit leading 1
+      out = s | (e << 23) | (m << 21);
+    }
+  } else if (e == 0x1fu) {
+    out = s | 0x7f800000u | (m << 21);  // inf / nan
+  } else {
+    out = s | ((e - 15u + 127u) << 23) | (m << 21);
+  }
+  union { uint32_t i; float f; } c;
+  c.i = out;
+  return c.f;
+}
+
+// 8 packed e5m2 bytes -> 8 bf16 values (one staging unit).
+DEV_INLINE ushort8 e5m2x8_to_bf16x8(const unsigned char* p) {
+  ushort8 out;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) out[j] = f32_to_bf16(e5m2_to_f32(p[j]));
+  return out;
+}
+
+DEV_INLINE unsigned char f32_to_e5m2(float x) {
+  // RNE with saturation to max finite (K/V magnitudes stay far below
+  // e5m2's 57344 ceiling in practice)
+  return (unsigned char)__hip_cvt_float_to_fp8(x, __HIP_SATFINITE, __HIP_E5M2);
 }
 
 // ---------------- wave reductions (64-wide) ----------------

@@ -32,6 +32,8 @@ class EngineConfig:
     max_model_len: int = 8192
     enable_prefix_caching: bool = True
     quantization: Optional[str] = None  # "fp8" => W8A8 dynamic (quant.py)
+    # "auto" = model dtype; "fp8_e5m2" halves KV bytes + doubles capacity
+    kv_cache_dtype: str = "auto"
     enable_graphs: bool = True  # hipGraph capture for pure-decode steps
     prefill_interval: int = 1  # >1: batch prefills onto every Nth step
     seed: int = 0
@@ -80,6 +82,7 @@ class LLMEngine:
             enable_graphs=cfg.enable_graphs,
             quantization=cfg.quantization,
             model_path=cfg.model,
+            kv_cache_dtype=cfg.kv_cache_dtype,
         )
         self.block_manager = BlockManager(self.runner.num_blocks, cfg.block_size)
         self.scheduler = Scheduler(

@@ -97,6 +97,7 @@ class ModelRunner:
         enable_graphs: bool = True,
         quantization: Optional[str] = None,
         model_path: Optional[str] = None,
+        kv_cache_dtype: str = "auto",
     ):
         self.arch = arch
         self.device = torch.device(device)
@@ -141,6 +142,15 @@ class ModelRunner:
             n = convert_to_fp8(self.model)
             assert n > 0, "no linear layers converted to fp8"
 
+        if kv_cache_dtype == "fp8_e5m2":
+            # halves decode KV reads and doubles block capacity; attention
+            # kernels dequantize in-register while staging (e5m2 range
+            # covers K/V without scale bookkeeping — vLLM's fp8_e5m2 analog)
+            self.kv_dtype = torch.float8_e5m2
+        elif kv_cache_dtype in ("auto", None):
+            self.kv_dtype = dtype
+        else:
+            raise ValueError(f"unknown kv_cache_dtype {kv_cache_dtype!r}")
         if num_gpu_blocks is None:
             num_gpu_blocks = self._profile_num_blocks(gpu_memory_utilization)
         self.num_blocks = num_gpu_blocks
@@ -152,8 +162,8 @@ class ModelRunner:
         )
         self.kv_caches = [
             (
-                torch.zeros(kv_shape, dtype=dtype, device=self.device),
-                torch.zeros(kv_shape, dtype=dtype, device=self.device),
+                torch.zeros(kv_shape, dtype=self.kv_dtype, device=self.device),
+                torch.zeros(kv_shape, dtype=self.kv_dtype, device=self.device),
             )
             for _ in range(arch.num_hidden_layers)
         ]
@@ -165,7 +175,7 @@ class ModelRunner:
             * self.n_kv_local
             * self.block_size
             * self.arch.head_dim
-            * self.dtype.itemsize
+            * self.kv_dtype.itemsize
             * self.arch.num_hidden_layers
         )
         if self.device.type == "cuda":

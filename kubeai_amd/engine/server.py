@@ -813,6 +813,8 @@ def main():
     p.add_argument("--max-num-seqs", type=int, default=256)
     p.add_argument("--gpu-memory-utilization", type=float, default=0.90)
     p.add_argument("--num-gpu-blocks", type=int, default=None)
+    p.add_argument("--kv-cache-dtype", choices=["auto", "fp8_e5m2"],
+                   default="auto")
     p.add_argument("--enable-lora", action="store_true")
     p.add_argument("--tensor-parallel-size", type=int, default=1)
     p.add_argument("--task", choices=["generate", "transcribe", "embed"],
@@ -836,6 +838,7 @@ def main():
         max_num_seqs=args.max_num_seqs,
         gpu_memory_utilization=args.gpu_memory_utilization,
         num_gpu_blocks=args.num_gpu_blocks,
+        kv_cache_dtype=args.kv_cache_dtype,
     )
     served = args.served_model_name or os.path.basename(args.model.rstrip("/"))
     tp = args.tensor_parallel_size

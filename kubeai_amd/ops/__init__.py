@@ -63,7 +63,10 @@ def rope(q, k, positions, cos_sin):
 def reshape_and_cache(k, v, k_cache, v_cache, slot_mapping) -> None:
     if k.is_cuda:
         _require_ext()
-        _C.reshape_and_cache(k, v, k_cache, v_cache, slot_mapping)
+        if k_cache.dtype == torch.float8_e5m2:
+            _C.reshape_and_cache_fp8(k, v, k_cache, v_cache, slot_mapping)
+        else:
+            _C.reshape_and_cache(k, v, k_cache, v_cache, slot_mapping)
         return
     ref.reshape_and_cache(k, v, k_cache, v_cache, slot_mapping)
 

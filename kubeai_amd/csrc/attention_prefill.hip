@@ -285,6 +285,8 @@ void paged_attention_prefill(torch::Tensor out, torch::Tensor q,
   TORCH_CHECK(out.is_contiguous());
   TORCH_CHECK(k_cache.is_contiguous() && v_cache.is_contiguous());
   TORCH_CHECK(q.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(k_cache.scalar_type() == torch::kBFloat16 ||
+              k_cache.scalar_type() == torch::kFloat8_e5m2);
   TORCH_CHECK(block_tables.scalar_type() == torch::kInt32);
   TORCH_CHECK(query_start_loc.scalar_type() == torch::kInt32);
   TORCH_CHECK(seq_lens.scalar_type() == torch::kInt32);

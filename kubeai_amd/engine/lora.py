@@ -22,17 +22,20 @@ import torch
 
 class LoRAManager:
     DEFAULT_RANK = 16
-    TARGETS = ("qkv", "o")
+    TARGETS = ("qkv", "o", "gate_up", "down")
 
     def __init__(self, model, device, dtype):
         self.model = model
         self.device = device
         self.dtype = dtype
-        # adapters[lora_id]["qkv"|"o"] = list over layers of (A [r,in], B [out,r], scale)
+        # adapters[lora_id][target] = list over layers of (A [r,in], B [out,r], scale)
         self.adapters: dict[int, dict[str, list[tuple[torch.Tensor, torch.Tensor, float]]]] = {}
         model.lora_manager = self
         for m in model.modules():
             if hasattr(m, "qkv_proj"):  # Attention modules
+                object.__setattr__(m, "_lora_manager", self)
+            # dense per-layer MLPs (layer_idx set); MoE expert MLPs excluded
+            if hasattr(m, "gate_up_proj") and getattr(m, "layer_idx", None) is not None:
                 object.__setattr__(m, "_lora_manager", self)
 
     # ------------------------------------------------------------------
@@ -59,9 +62,14 @@ class LoRAManager:
         H = cfg.hidden_size
         qkv_out = (cfg.num_attention_heads + 2 * cfg.num_key_value_heads) * cfg.head_dim
         o_in = cfg.num_attention_heads * cfg.head_dim
-        out = {"qkv": [], "o": []}
+        I = cfg.intermediate_size
+        moe = cfg.num_local_experts > 0  # MoE: adapters stay attention-only
+        out = {t: [] for t in self.TARGETS}
         for _ in range(cfg.num_hidden_layers):
-            for tgt, (din, dout) in (("qkv", (H, qkv_out)), ("o", (o_in, H))):
+            shapes = [("qkv", (H, qkv_out)), ("o", (o_in, H))]
+            if not moe:
+                shapes += [("gate_up", (H, 2 * I)), ("down", (I, H))]
+            for tgt, (din, dout) in shapes:
                 A = (torch.randn(r, din, generator=g) * 0.05).to(self.device, self.dtype)
                 B = (torch.randn(dout, r, generator=g) * 0.5).to(self.device, self.dtype)
                 out[tgt].append((A, B, 2.0 / r))
@@ -80,7 +88,7 @@ class LoRAManager:
             r = int(acfg.get("r", 16))
         weights = load_file(_peft_weights_present(path))
         cfg = self.model.cfg
-        out = {"qkv": [], "o": []}
+        out = {t: [] for t in self.TARGETS}
         for layer in range(cfg.num_hidden_layers):
             # fuse q/k/v adapters into the qkv slot (block-diagonal A, stacked B)
             parts = []
@@ -132,6 +140,45 @@ class LoRAManager:
                         alpha / r,
                     )
                 )
+            # MLP targets: gate+up fused block-diagonally, down direct
+            I = cfg.intermediate_size
+            gkey = _find_key(weights, layer, "gate_proj", "mlp")
+            ukey = _find_key(weights, layer, "up_proj", "mlp")
+            if gkey is None and ukey is None:
+                out["gate_up"].append(
+                    (torch.zeros(1, H, device=self.device, dtype=self.dtype),
+                     torch.zeros(2 * I, 1, device=self.device, dtype=self.dtype),
+                     0.0)
+                )
+            else:
+                As2, Bs2 = [], []
+                for key, row in ((gkey, 0), (ukey, I)):
+                    if key is not None:
+                        As2.append((weights[key + ".lora_A.weight"], row))
+                        Bs2.append(weights[key + ".lora_B.weight"])
+                A2 = torch.cat([a for a, _ in As2], dim=0)  # [n*r, H]
+                B2 = torch.zeros(2 * I, A2.shape[0])
+                col = 0
+                for (a, row), bmat in zip(As2, Bs2):
+                    B2[row : row + I, col : col + a.shape[0]] = bmat
+                    col += a.shape[0]
+                out["gate_up"].append(
+                    (A2.to(self.device, self.dtype),
+                     B2.to(self.device, self.dtype), alpha / r)
+                )
+            dkey = _find_key(weights, layer, "down_proj", "mlp")
+            if dkey is None:
+                out["down"].append(
+                    (torch.zeros(1, I, device=self.device, dtype=self.dtype),
+                     torch.zeros(H, 1, device=self.device, dtype=self.dtype),
+                     0.0)
+                )
+            else:
+                out["down"].append(
+                    (weights[dkey + ".lora_A.weight"].to(self.device, self.dtype),
+                     weights[dkey + ".lora_B.weight"].to(self.device, self.dtype),
+                     alpha / r)
+                )
         return out
 
     # ------------------------------------------------------------------
@@ -169,10 +216,12 @@ def _peft_weights_present(path: str) -> Optional[str]:
     return None
 
 
-def _find_key(weights: dict, layer: int, proj: str) -> Optional[str]:
+def _find_key(
+    weights: dict, layer: int, proj: str, sub: str = "self_attn"
+) -> Optional[str]:
     for prefix in (
-        f"base_model.model.model.layers.{layer}.self_attn.{proj}",
-        f"model.layers.{layer}.self_attn.{proj}",
+        f"base_model.model.model.layers.{layer}.{sub}.{proj}",
+        f"model.layers.{layer}.{sub}.{proj}",
     ):
         if prefix + ".lora_A.weight" in weights:
             return prefix

@@ -120,19 +120,34 @@ class Attention(nn.Module):
 
 
 class MLP(nn.Module):
-    def __init__(self, cfg: ModelArchConfig):
+    def __init__(self, cfg: ModelArchConfig, layer_idx: int | None = None):
         super().__init__()
+        # layer_idx set only for the DENSE per-layer MLP: LoRA targets it
+        # (MoE expert MLPs pass None — adapters stay attention-only there)
+        self.layer_idx = layer_idx
         self.gate_up_proj = EngineLinear(
             cfg.hidden_size, 2 * cfg.intermediate_size, bias=False
         )
         self.down_proj = EngineLinear(cfg.intermediate_size, cfg.hidden_size, bias=False)
 
-    def forward(self, x) -> torch.Tensor:
-        if isinstance(x, tuple):  # fused fp8 path
+    def forward(self, x, fb: ForwardBatch | None = None) -> torch.Tensor:
+        if isinstance(x, tuple):  # fused fp8 path (never active with LoRA)
             h = self.gate_up_proj.forward_quantized(*x)
             a8, ascale = ops.silu_and_mul_fp8(h)
             return self.down_proj.forward_quantized(a8, ascale)
-        return self.down_proj(ops.silu_and_mul(self.gate_up_proj(x)))
+        h = self.gate_up_proj(x)
+        lm = getattr(self, "_lora_manager", None)
+        lora_live = (
+            lm is not None and lm.active and fb is not None
+            and fb.lora_ids is not None and self.layer_idx is not None
+        )
+        if lora_live:
+            lm.apply(self.layer_idx, "gate_up", x, h, fb.lora_ids)
+        a = ops.silu_and_mul(h)
+        y = self.down_proj(a)
+        if lora_live:
+            lm.apply(self.layer_idx, "down", a, y, fb.lora_ids)
+        return y
 
 
 class MoEMLP(nn.Module):
@@ -149,7 +164,7 @@ class MoEMLP(nn.Module):
         self.gate = nn.Linear(cfg.hidden_size, self.n_experts, bias=False)
         self.experts = nn.ModuleList([MLP(cfg) for _ in range(self.n_experts)])
 
-    def forward(self, x: torch.Tensor) -> torch.Tensor:
+    def forward(self, x: torch.Tensor, fb: ForwardBatch | None = None) -> torch.Tensor:
         T, H = x.shape
         router = F.linear(x.float(), self.gate.weight.float())  # [T, E]
         weights, selected = torch.topk(router, self.top_k, dim=-1)
@@ -183,7 +198,9 @@ class DecoderLayer(nn.Module):
         self.input_layernorm = nn.Parameter(torch.ones(cfg.hidden_size))
         self.post_attention_layernorm = nn.Parameter(torch.ones(cfg.hidden_size))
         self.self_attn = Attention(cfg, layer_idx)
-        self.mlp = MoEMLP(cfg) if cfg.num_local_experts > 0 else MLP(cfg)
+        self.mlp = (
+            MoEMLP(cfg) if cfg.num_local_experts > 0 else MLP(cfg, layer_idx)
+        )
         self.eps = cfg.rms_norm_eps
 
     def forward(self, x, residual, fb, kv_cache, cos_sin):
@@ -226,7 +243,7 @@ class DecoderLayer(nn.Module):
         x, residual = ops.fused_add_rmsnorm(
             x, residual, self.post_attention_layernorm, self.eps
         )
-        x = self.mlp(x)
+        x = self.mlp(x, fb)
         return x, residual
 
 

@@ -332,3 +332,54 @@ def test_lora_on_moe_model():
     assert not torch.allclose(base_after, adapted)  # adapter is live
     eng.unload_lora(9)
     assert torch.equal(last_logits(9), base_before)  # unload restores
+
+
+def test_lora_mlp_targets(tmp_path):
+    """A PEFT adapter touching ONLY the MLP projections (gate/up/down)
+    must shift the logits — vLLM-style full-linear LoRA coverage."""
+    import torch
+    from safetensors.torch import save_file
+
+    eng = LLMEngine(
+        EngineConfig(model="llama-tiny", device="cpu", num_gpu_blocks=128,
+                     max_model_len=512, seed=2)
+    )
+    cfg = eng.runner.model.cfg
+    H, I, r = cfg.hidden_size, cfg.intermediate_size, 4
+    g = torch.Generator().manual_seed(0)
+    weights = {}
+    for layer in range(cfg.num_hidden_layers):
+        for proj in ("gate_proj", "up_proj", "down_proj"):
+            din = I if proj == "down_proj" else H
+            dout = H if proj == "down_proj" else I
+            pre = f"base_model.model.model.layers.{layer}.mlp.{proj}"
+            weights[pre + ".lora_A.weight"] = torch.randn(r, din, generator=g) * 0.2
+            weights[pre + ".lora_B.weight"] = torch.randn(dout, r, generator=g) * 0.2
+    d = str(tmp_path / "mlp-adapter")
+    import json
+    import os
+
+    os.makedirs(d)
+    save_file(weights, os.path.join(d, "adapter_model.safetensors"))
+    with open(os.path.join(d, "adapter_config.json"), "w") as f:
+        json.dump({"lora_alpha": 8, "r": r}, f)
+
+    prompt = list(range(10, 58))
+
+    def last_logits(lora_id):
+        from kubeai_amd.engine.kvcache import BlockManager
+        from kubeai_amd.engine.scheduler import Request, Scheduler
+
+        s = Scheduler(BlockManager(64, 16), max_num_batched_tokens=512,
+                      max_model_len=256)
+        s.add_request(Request(prompt, SamplingParams(max_tokens=1),
+                              request_id="x", lora_id=lora_id))
+        out = s.schedule()
+        fb = eng.runner.build_batch(out)
+        h = eng.runner.model(fb)
+        return eng.runner.model.compute_logits(h[fb.logits_indices]).float()
+
+    base = last_logits(0)
+    eng.load_lora(3, d)
+    assert torch.equal(last_logits(0), base)      # base stream untouched
+    assert not torch.allclose(last_logits(3), base)  # MLP deltas live

@@ -433,3 +433,60 @@ def test_messenger_nacks_failed_handling(tmp_path):
             await m.stop()
 
     asyncio.run(run())
+
+
+def test_ref_rope_invariants():
+    """The rope REFERENCE is the kernels' contract — pin its math:
+    position 0 is identity, rotations preserve per-pair norms, and
+    relative-position structure holds (q.k depends only on distance)."""
+    import torch
+
+    from kubeai_amd.ops import ref
+
+    hd = 64
+    cs = ref.make_cos_sin_cache(hd, 128, 10000.0)
+    q = torch.randn(3, 2, hd)
+    k = torch.randn(3, 1, hd)
+    pos0 = torch.zeros(3, dtype=torch.int32)
+    q0, k0 = ref.rope(q, k, pos0, cs)
+    assert torch.allclose(q0, q, atol=1e-6) and torch.allclose(k0, k, atol=1e-6)
+    # norm preservation at arbitrary positions
+    pos = torch.tensor([5, 17, 90], dtype=torch.int32)
+    qr, kr = ref.rope(q, k, pos, cs)
+    assert torch.allclose(qr.norm(dim=-1), q.norm(dim=-1), atol=1e-4)
+    # relative property: <rope(q,p+d), rope(k,p'+d)> == <rope(q,p), rope(k,p')>
+    qa = torch.randn(1, 1, hd)
+    kb = torch.randn(1, 1, hd)
+    def dot_at(pq, pk):
+        q1, _ = ref.rope(qa, qa, torch.tensor([pq], dtype=torch.int32), cs)
+        k1, _ = ref.rope(kb, kb, torch.tensor([pk], dtype=torch.int32), cs)
+        return float((q1.flatten() * k1.flatten()).sum())
+    assert abs(dot_at(10, 4) - dot_at(30, 24)) < 1e-3
+
+
+def test_ref_attention_softmax_sanity():
+    """Reference paged decode == plain softmax attention on gathered KV."""
+    import math
+
+    import torch
+
+    from kubeai_amd.ops import ref
+
+    torch.manual_seed(0)
+    B, nq, nkv, hd, bs, L = 2, 4, 2, 32, 16, 40
+    nb_per = (L + bs - 1) // bs
+    kc = torch.randn(B * nb_per + 1, nkv, bs, hd)
+    vc = torch.randn_like(kc)
+    bt = torch.arange(1, B * nb_per + 1, dtype=torch.int32).reshape(B, nb_per)
+    sl = torch.full((B,), L, dtype=torch.int32)
+    q = torch.randn(B, nq, hd)
+    scale = 1.0 / math.sqrt(hd)
+    out = ref.paged_attention_decode(q, kc, vc, bt, sl, scale)
+    # manual recompute for one (seq, head)
+    b, h = 1, 3
+    kv_h = h * nkv // nq
+    ks = kc[bt[b].long()].transpose(1, 2).reshape(-1, nkv, hd)[:L, kv_h]
+    vs = vc[bt[b].long()].transpose(1, 2).reshape(-1, nkv, hd)[:L, kv_h]
+    p = torch.softmax((ks @ q[b, h]) * scale, dim=0)
+    want = (p[:, None] * vs).sum(dim=0)
+    assert torch.allclose(out[b, h], want, atol=1e-5)

@@ -19,6 +19,8 @@
 //    per-token p reaching lanes via one __shfl broadcast.
 #include <torch/extension.h>
 
+#include <type_traits>
+
 #include "common.h"
 
 namespace {
@@ -59,9 +61,14 @@ __launch_bounds__(kBlockThreads) __global__ void paged_decode_kernel(
   const int blk_lo = split * chunk;
   const int blk_hi = min(n_blocks, blk_lo + chunk);
 
-  // LDS: per-wave double-buffered KV tiles + merge scratch
-  __shared__ ushort k_lds[kWaves][2][kBS][kHD + kPad];
-  __shared__ ushort v_lds[kWaves][2][kBS][kHD + kPad];
+  // LDS: per-wave double-buffered KV tiles + merge scratch. Element type
+  // follows the cache: bf16 stages bf16; fp8 stages the RAW e5m2 bytes
+  // (half the LDS footprint/traffic) and converts during compute with the
+  // native packed bf8->f32 instruction (v_cvt_pk_f32_bf8). 8-element row
+  // pad keeps 8 B staging alignment and breaks bank aliasing for both
+  // element widths (bf16: 68-dword stride, gcd 4; fp8: 34-dword, gcd 2).
+  __shared__ CT k_lds[kWaves][2][kBS][kHD + kPad];
+  __shared__ CT v_lds[kWaves][2][kBS][kHD + kPad];
   __shared__ float merge_o[kWaves][G][kHD];
   __shared__ float merge_ml[kWaves][G][2];
 
@@ -99,46 +106,32 @@ __launch_bounds__(kBlockThreads) __global__ void paged_decode_kernel(
   // compute (their latency hides under it); the register payload lands in
   // the other LDS buffer just before it is needed. Per-wave buffers ->
   // no barriers anywhere in the loop.
-  ushort8 stage_k[4], stage_v[4];  // this lane's 4 vectors of each tile
+  // staging vector = 8 elements: 16 B (bf16) or 8 B (raw fp8 bytes)
+  using StageVec = std::conditional_t<sizeof(CT) == 2, ushort8, uint64_t>;
+  StageVec stage_k[4], stage_v[4];  // this lane's 4 vectors of each tile
 
   auto issue_loads = [&](int blk_i) {
     const int64_t blk = bt[blk_i];
     const CT* base_k = k_cache + ((blk * n_kv + kh) * kBS) * kHD;
     const CT* base_v = v_cache + ((blk * n_kv + kh) * kBS) * kHD;
-    if constexpr (sizeof(CT) == 2) {
-      const ushort8* src_k = reinterpret_cast<const ushort8*>(base_k);
-      const ushort8* src_v = reinterpret_cast<const ushort8*>(base_v);
+    const StageVec* src_k = reinterpret_cast<const StageVec*>(base_k);
+    const StageVec* src_v = reinterpret_cast<const StageVec*>(base_v);
 #pragma unroll
-      for (int i = 0; i < (kBS * kHD / 8) / WAVE_SIZE; ++i) {
-        stage_k[i] = src_k[lane + i * WAVE_SIZE];
-        stage_v[i] = src_v[lane + i * WAVE_SIZE];
-      }
-    } else {
-      // fp8: 8-byte lane loads (one 8-elem staging unit each), convert
-      // to bf16 in-register; LDS layout and compute stay identical
-      const unsigned char* bk = reinterpret_cast<const unsigned char*>(base_k);
-      const unsigned char* bv = reinterpret_cast<const unsigned char*>(base_v);
-#pragma unroll
-      for (int i = 0; i < (kBS * kHD / 8) / WAVE_SIZE; ++i) {
-        uint64_t kraw = reinterpret_cast<const uint64_t*>(
-            bk)[lane + i * WAVE_SIZE];
-        uint64_t vraw = reinterpret_cast<const uint64_t*>(
-            bv)[lane + i * WAVE_SIZE];
-        stage_k[i] = e5m2x8_to_bf16x8(
-            reinterpret_cast<const unsigned char*>(&kraw));
-        stage_v[i] = e5m2x8_to_bf16x8(
-            reinterpret_cast<const unsigned char*>(&vraw));
-      }
+    for (int i = 0; i < (kBS * kHD / 8) / WAVE_SIZE; ++i) {
+      stage_k[i] = src_k[lane + i * WAVE_SIZE];
+      stage_v[i] = src_v[lane + i * WAVE_SIZE];
     }
   };
   auto write_tile = [&](int buf) {
 #pragma unroll
     for (int i = 0; i < (kBS * kHD / 8) / WAVE_SIZE; ++i) {
-      const int vec = lane + i * WAVE_SIZE;  // 16B vector index
+      const int vec = lane + i * WAVE_SIZE;  // 8-element vector index
       const int row = vec / (kHD / 8);
       const int col = vec % (kHD / 8);
-      *reinterpret_cast<ushort8*>(&k_lds[wave][buf][row][col * 8]) = stage_k[i];
-      *reinterpret_cast<ushort8*>(&v_lds[wave][buf][row][col * 8]) = stage_v[i];
+      *reinterpret_cast<StageVec*>(&k_lds[wave][buf][row][col * 8]) =
+          stage_k[i];
+      *reinterpret_cast<StageVec*>(&v_lds[wave][buf][row][col * 8]) =
+          stage_v[i];
     }
   };
 
@@ -158,7 +151,7 @@ __launch_bounds__(kBlockThreads) __global__ void paged_decode_kernel(
     float s[G];
 #pragma unroll
     for (int g = 0; g < G; ++g) s[g] = 0.f;
-    {
+    if constexpr (sizeof(CT) == 2) {
       const uint32_t* krow = reinterpret_cast<const uint32_t*>(
           &k_lds[wave][cur][tok_of][part * 32]);
 #pragma unroll
@@ -171,6 +164,26 @@ __launch_bounds__(kBlockThreads) __global__ void paged_decode_kernel(
           const uint32_t qq = q_pack[g][j];
           s[g] = fmaf(k0, bf16_to_f32((ushort)(qq & 0xffff)), s[g]);
           s[g] = fmaf(k1, bf16_to_f32((ushort)(qq >> 16)), s[g]);
+        }
+      }
+    } else {
+      // raw e5m2 slice: each uint32 holds 4 bytes -> 4 elements; the
+      // packed bf8->f32 convert does 2 per instruction
+      const uint32_t* krow = reinterpret_cast<const uint32_t*>(
+          &k_lds[wave][cur][tok_of][part * 32]);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const uint32_t kk = krow[j];
+        const floatx2 k01 = __builtin_amdgcn_cvt_pk_f32_bf8(kk, false);
+        const floatx2 k23 = __builtin_amdgcn_cvt_pk_f32_bf8(kk, true);
+#pragma unroll
+        for (int g = 0; g < G; ++g) {
+          const uint32_t qa = q_pack[g][2 * j];
+          const uint32_t qb = q_pack[g][2 * j + 1];
+          s[g] = fmaf(k01.x, bf16_to_f32((ushort)(qa & 0xffff)), s[g]);
+          s[g] = fmaf(k01.y, bf16_to_f32((ushort)(qa >> 16)), s[g]);
+          s[g] = fmaf(k23.x, bf16_to_f32((ushort)(qb & 0xffff)), s[g]);
+          s[g] = fmaf(k23.y, bf16_to_f32((ushort)(qb >> 16)), s[g]);
         }
       }
     }
@@ -207,10 +220,19 @@ __launch_bounds__(kBlockThreads) __global__ void paged_decode_kernel(
 #pragma unroll
     for (int t = 0; t < kBS; ++t) {
       if (t >= tile_len) break;
-      const uint32_t vv =
-          *reinterpret_cast<const uint32_t*>(&v_lds[wave][cur][t][2 * lane]);
-      const float v0 = bf16_to_f32((ushort)(vv & 0xffff));
-      const float v1 = bf16_to_f32((ushort)(vv >> 16));
+      float v0, v1;
+      if constexpr (sizeof(CT) == 2) {
+        const uint32_t vv =
+            *reinterpret_cast<const uint32_t*>(&v_lds[wave][cur][t][2 * lane]);
+        v0 = bf16_to_f32((ushort)(vv & 0xffff));
+        v1 = bf16_to_f32((ushort)(vv >> 16));
+      } else {
+        const uint32_t vv =
+            *reinterpret_cast<const ushort*>(&v_lds[wave][cur][t][2 * lane]);
+        const floatx2 vf = __builtin_amdgcn_cvt_pk_f32_bf8(vv, false);
+        v0 = vf.x;
+        v1 = vf.y;
+      }
 #pragma unroll
       for (int g = 0; g < G; ++g) {
         const float pt = __shfl(p[g], t * 4, 64);

@@ -145,13 +145,11 @@ __launch_bounds__(G * WAVE_SIZE) __global__ void paged_prefill_kernel(
             kk = *reinterpret_cast<const bf16x8*>(src + col8 * 8);
             vv = *reinterpret_cast<const bf16x8*>(vsrc + col8 * 8);
           } else {
-            // fp8: 8-byte loads, in-register e5m2 -> bf16 conversion
+            // fp8: 8-byte loads, packed-convert e5m2 -> bf16 in-register
             uint64_t kraw = *reinterpret_cast<const uint64_t*>(src + col8 * 8);
             uint64_t vraw = *reinterpret_cast<const uint64_t*>(vsrc + col8 * 8);
-            const ushort8 kc = e5m2x8_to_bf16x8(
-                reinterpret_cast<const unsigned char*>(&kraw));
-            const ushort8 vc = e5m2x8_to_bf16x8(
-                reinterpret_cast<const unsigned char*>(&vraw));
+            const ushort8 kc = e5m2x8_to_bf16x8(kraw);
+            const ushort8 vc = e5m2x8_to_bf16x8(vraw);
 #pragma unroll
             for (int j = 0; j < 8; ++j) {
               kk[j] = (short)kc[j];

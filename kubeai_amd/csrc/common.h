@@ -75,11 +75,26 @@ DEV_INLINE float e5m2_to_f32(unsigned char b) {
   return c.f;
 }
 
-// 8 packed e5m2 bytes -> 8 bf16 values (one staging unit).
-DEV_INLINE ushort8 e5m2x8_to_bf16x8(const unsigned char* p) {
+// native 2 x f32 vector, the result type of the packed bf8->f32 convert
+// (v_cvt_pk_f32_bf8: one instruction turns 2 e5m2 bytes into 2 floats)
+typedef float floatx2 __attribute__((ext_vector_type(2)));
+
+// 8 packed e5m2 bytes -> 8 bf16 values (one staging unit). Uses the
+// native packed convert (2 bytes/instr); every e5m2 value — subnormals,
+// inf, nan included — is exactly representable in bf16, so the f32
+// result truncates to bf16 losslessly (plain >>16, no rounding step).
+DEV_INLINE ushort8 e5m2x8_to_bf16x8(uint64_t raw) {
   ushort8 out;
+  const uint32_t w[2] = {(uint32_t)raw, (uint32_t)(raw >> 32)};
 #pragma unroll
-  for (int j = 0; j < 8; ++j) out[j] = f32_to_bf16(e5m2_to_f32(p[j]));
+  for (int i = 0; i < 2; ++i) {
+    const floatx2 lo = __builtin_amdgcn_cvt_pk_f32_bf8(w[i], false);
+    const floatx2 hi = __builtin_amdgcn_cvt_pk_f32_bf8(w[i], true);
+    out[4 * i + 0] = (ushort)(__builtin_bit_cast(uint32_t, lo.x) >> 16);
+    out[4 * i + 1] = (ushort)(__builtin_bit_cast(uint32_t, lo.y) >> 16);
+    out[4 * i + 2] = (ushort)(__builtin_bit_cast(uint32_t, hi.x) >> 16);
+    out[4 * i + 3] = (ushort)(__builtin_bit_cast(uint32_t, hi.y) >> 16);
+  }
   return out;
 }
 

@@ -173,25 +173,50 @@ class LLMEngine:
     # ------------------------------------------------------------------
     def embed(self, tok_lists: list[list[int]]) -> list[list[float]]:
         """Mean-pooled, L2-normalised final hidden states (TextEmbedding
-        feature; reference analog: the Infinity engine)."""
+        feature; reference analog: the Infinity engine).
+
+        Sequences are packed many-per-forward (token-budget bounded), the
+        same packed-prefill shape the generation path uses — batched
+        embedding throughput instead of one forward per input.
+
+        MUST run on the stepping thread (the server routes embed jobs
+        through its submit loop): it shares the BlockManager and KV
+        tensors with in-flight generations.
+        """
         from .scheduler import ScheduledSeq, SchedulerOutput
 
-        out: list[list[float]] = []
-        for toks in tok_lists:
-            toks = toks[: self.cfg.max_model_len - 1]
-            table, _ = self.block_manager.allocate(toks, salt=-1, max_cached=0)
-            req = Request(list(toks), SamplingParams(max_tokens=1))
-            req.block_table = table
+        out: list[Optional[list[float]]] = [None] * len(tok_lists)
+        budget = self.cfg.max_num_batched_tokens
+        i = 0
+        while i < len(tok_lists):
+            batch: list[tuple[int, Request, list[int]]] = []
+            total = 0
+            while i < len(tok_lists):
+                toks = list(tok_lists[i][: self.cfg.max_model_len - 1])
+                if batch and total + len(toks) > budget:
+                    break
+                table, _ = self.block_manager.allocate(toks, salt=-1, max_cached=0)
+                req = Request(toks, SamplingParams(max_tokens=1))
+                req.block_table = table
+                batch.append((i, req, toks))
+                total += len(toks)
+                i += 1
             so = SchedulerOutput(
-                decode=[], prefill=[ScheduledSeq(req, 0, len(toks))], preempted=[]
+                decode=[],
+                prefill=[ScheduledSeq(r, 0, len(t)) for _, r, t in batch],
+                preempted=[],
             )
             fb = self.runner.build_batch(so)
             hidden = self.runner.model(fb)
-            # exclude bucket-padding rows from the pool
-            vec = hidden[: len(toks)].float().mean(dim=0)
-            vec = vec / vec.norm().clamp_min(1e-12)
-            out.append(vec.cpu().tolist())
-            self.block_manager.free(table)
+            ofs = 0
+            for idx, req, toks in batch:
+                # rows are packed prefill-order; bucket padding sits past
+                # the real tokens
+                vec = hidden[ofs : ofs + len(toks)].float().mean(dim=0)
+                vec = vec / vec.norm().clamp_min(1e-12)
+                out[idx] = vec.cpu().tolist()
+                ofs += len(toks)
+                self.block_manager.free(req.block_table)
         return out
 
     # ------------------------------------------------------------------

@@ -396,15 +396,22 @@ class ModelRunner:
                 counts.scatter_add_(0, out_t, torch.ones_like(out_t, dtype=torch.float32))
                 logits[i] -= fp * counts + pp * (counts > 0).float()
             if r.params.logit_bias:
-                ids = torch.tensor(
-                    list(r.params.logit_bias.keys()), dtype=torch.int64,
-                    device=logits.device,
-                )
-                vals = torch.tensor(
-                    list(r.params.logit_bias.values()), dtype=logits.dtype,
-                    device=logits.device,
-                )
-                logits[i].scatter_add_(0, ids, vals)
+                # ids validated at parse time (server RequestError); filter
+                # again here so a bad id can never kill the step loop
+                V = logits.shape[1]
+                pairs = [
+                    (t, b) for t, b in r.params.logit_bias.items() if 0 <= t < V
+                ]
+                if pairs:
+                    ids = torch.tensor(
+                        [t for t, _ in pairs], dtype=torch.int64,
+                        device=logits.device,
+                    )
+                    vals = torch.tensor(
+                        [b for _, b in pairs], dtype=logits.dtype,
+                        device=logits.device,
+                    )
+                    logits[i].scatter_add_(0, ids, vals)
         temps = [r.params.temperature for r in sample_reqs]
         if all(t <= 0.0 for t in temps):
             tokens = ops.greedy_sample(logits.contiguous())

@@ -31,7 +31,41 @@ import prometheus_client as prom
 
 from .engine import EngineConfig, LLMEngine
 from .scheduler import SamplingParams
-from .tokenizer import apply_chat_template, load_tokenizer
+from .tokenizer import StreamDecoder, apply_chat_template, load_tokenizer
+
+
+class _StopScanner:
+    """Earliest stop-string hit with O(new text) scanning per step."""
+
+    def __init__(self, stops: list[str]):
+        self.stops = stops
+        self.max_len = max(len(s) for s in stops)
+        self._from = 0
+
+    def scan(self, text: str) -> int:
+        start = max(0, self._from - self.max_len + 1)
+        best = -1
+        for s in self.stops:
+            i = text.find(s, start)
+            if i >= 0 and (best < 0 or i < best):
+                best = i
+        self._from = len(text)
+        return best
+
+
+def _holdback_len(text: str, stops: list[str]) -> int:
+    """Longest trailing substring of `text` that is a proper prefix of a
+    stop string — streamed deltas hold it back so a stop spanning chunk
+    boundaries never leaks its leading characters to the client."""
+    m = 0
+    for s in stops:
+        k = min(len(s) - 1, len(text))
+        while k > m:
+            if text.endswith(s[:k]):
+                m = k
+                break
+            k -= 1
+    return m
 
 REGISTRY = prom.CollectorRegistry()
 M_WAITING = prom.Gauge("kubeai_engine_num_requests_waiting", "queue depth", ["model"], registry=REGISTRY)
@@ -72,10 +106,17 @@ class EngineServer:
         self._submit: "queue.Queue" = queue.Queue()
         self._events: dict[str, tuple[asyncio.AbstractEventLoop, asyncio.Queue]] = {}
         self._aborts: "queue.Queue" = queue.Queue()
+        # embed jobs run ON the stepping thread (they share the
+        # BlockManager/KV tensors with in-flight generations — ADVICE r1)
+        self._embed_queue: "queue.Queue" = queue.Queue()
         self._ready = threading.Event()
         self._stop = threading.Event()
         self._lora_adapters: dict[str, str] = {}
         self._lock = threading.Lock()
+        # step-loop health (a dead stepping thread must not keep
+        # reporting healthy while every request hangs)
+        self._consec_step_errors = 0
+        self._fatal_error: Optional[str] = None
         self._thread = threading.Thread(target=self._run, daemon=True)
 
     # ------------------------------------------------------------- lifecycle
@@ -86,6 +127,38 @@ class EngineServer:
         self._stop.set()
 
     def _run(self) -> None:
+        try:
+            self._run_inner()
+        except Exception:
+            import traceback
+
+            traceback.print_exc()
+            self._fatal_error = traceback.format_exc(limit=8)
+
+    def _fail_inflight(self) -> list:
+        """After a step exception: abort every in-flight request and emit
+        finished-abort outputs so no client waits forever. The next loop
+        iteration's schedule() applies the abort marks and frees KV."""
+        from .engine import RequestOutput
+
+        eng = self.engine
+        outs = []
+        for r in list(eng.scheduler.running) + list(eng.scheduler.waiting):
+            eng.abort_request(r.request_id)
+            outs.append(
+                RequestOutput(
+                    request_id=r.request_id,
+                    new_token_ids=[],
+                    finished=True,
+                    finish_reason="abort",
+                    num_prompt_tokens=r.num_prompt_tokens,
+                    num_cached_tokens=0,
+                    output_token_ids=r.output_token_ids,
+                )
+            )
+        return outs
+
+    def _run_inner(self) -> None:
         # heavy init inside the thread so /health can answer "starting"
         if self.task == "transcribe":
             from kubeai_amd.models.whisper import PRESETS as STT_PRESETS
@@ -186,10 +259,18 @@ class EngineServer:
                     abort_ids.append(self._aborts.get_nowait())
                 except queue.Empty:
                     break
+            embed_jobs = []
+            while True:
+                try:
+                    embed_jobs.append(self._embed_queue.get_nowait())
+                except queue.Empty:
+                    break
             if self._dist is not None:
                 # keep TP workers in lockstep: ship this iteration's intents
                 self._dist.broadcast_object_list(
-                    [{"new": new_reqs, "aborts": abort_ids, "shutdown": False}],
+                    [{"new": new_reqs, "aborts": abort_ids,
+                      "embeds": [j[0] for j in embed_jobs],
+                      "shutdown": False}],
                     src=0,
                 )
             for rid, toks, params, lora_id in new_reqs:
@@ -198,8 +279,31 @@ class EngineServer:
                 worked = True
             for rid in abort_ids:
                 self.engine.abort_request(rid)
+            for tok_lists, loop, fut in embed_jobs:
+                # embed shares the BlockManager/KV with generations, so it
+                # must run here on the stepping thread
+                try:
+                    vecs = self.engine.embed(tok_lists)
+                    loop.call_soon_threadsafe(
+                        lambda f=fut, v=vecs: f.done() or f.set_result(v)
+                    )
+                except Exception as e:  # noqa: BLE001
+                    loop.call_soon_threadsafe(
+                        lambda f=fut, err=e: f.done() or f.set_exception(err)
+                    )
+                worked = True
             if self.engine.has_work():
-                outputs = self.engine.step()
+                try:
+                    outputs = self.engine.step()
+                    self._consec_step_errors = 0
+                except Exception:
+                    import traceback
+
+                    traceback.print_exc()
+                    self._consec_step_errors += 1
+                    if self._consec_step_errors >= 3:
+                        self._fatal_error = traceback.format_exc(limit=8)
+                    outputs = self._fail_inflight()
                 worked = True
                 for o in outputs:
                     M_GEN.labels(label).inc(len(o.new_token_ids))
@@ -243,10 +347,25 @@ class EngineServer:
             self._events.pop(rid, None)
             self._aborts.put(rid)
 
+    async def embed(self, tok_lists: list[list[int]]) -> list[list[float]]:
+        """Embed via the stepping thread (generate-task models)."""
+        loop = asyncio.get_running_loop()
+        fut = loop.create_future()
+        self._embed_queue.put((tok_lists, loop, fut))
+        return await fut
+
     # ------------------------------------------------------------- LoRA
     # Registry + per-adapter KV-cache salting; adapter weights are applied
     # by the runner's LoRA manager (kubeai_amd/engine/lora.py).
     def load_lora(self, name: str, path: Optional[str]) -> None:
+        if self.tp_size > 1:
+            # TP layers do not apply LoRA and adapter loads are not
+            # broadcast to worker ranks — honest rejection beats silently
+            # serving base-model outputs under an adapter name (ADVICE r1)
+            raise ValueError(
+                "LoRA adapters are not supported with tensor-parallel "
+                "serving (tensor_parallel_size > 1)"
+            )
         if path and not (os.path.isdir(path) or os.path.isfile(path)):
             raise ValueError(f"lora path {path} does not exist")
         lid = self.lora_id_of(name)
@@ -263,9 +382,21 @@ class EngineServer:
         return 1 + (zlib.crc32(name.encode()) % 1_000_000)
 
 
+class RequestError(ValueError):
+    """Client error in a request body -> HTTP 400."""
+
+
 def build_app(server: EngineServer) -> FastAPI:
     app = FastAPI()
+    app.state.eng_server = server  # test/introspection handle
     name = server.served_model_name
+
+    @app.exception_handler(RequestError)
+    async def _bad_request(request, exc):
+        return JSONResponse(
+            {"error": {"message": str(exc), "type": "invalid_request_error"}},
+            status_code=400,
+        )
 
     def _generate_unsupported():
         task = "transcribe" if server.stt is not None else "embed"
@@ -274,6 +405,24 @@ def build_app(server: EngineServer) -> FastAPI:
                                   "generation endpoints are not supported"}},
             status_code=400,
         )
+
+    def _logit_bias_from(body: dict) -> Optional[dict]:
+        raw = body.get("logit_bias")
+        if not raw:
+            return None
+        vocab = server.engine.arch.vocab_size if server.engine else None
+        out: dict[int, float] = {}
+        for k, v in raw.items():
+            try:
+                tid, bias = int(k), float(v)
+            except (TypeError, ValueError):
+                raise RequestError(f"logit_bias key {k!r} is not a token id")
+            if tid < 0 or (vocab is not None and tid >= vocab):
+                raise RequestError(
+                    f"logit_bias token id {tid} out of range [0, {vocab})"
+                )
+            out[tid] = bias
+        return out
 
     def _params_from(body: dict) -> SamplingParams:
         mt = body.get("max_tokens") or body.get("max_completion_tokens") or 128
@@ -290,11 +439,7 @@ def build_app(server: EngineServer) -> FastAPI:
             presence_penalty=float(body.get("presence_penalty") or 0.0),
             frequency_penalty=float(body.get("frequency_penalty") or 0.0),
             priority=int(body.get("priority") or 0),
-            logit_bias=(
-                {int(k): float(v) for k, v in body["logit_bias"].items()}
-                if body.get("logit_bias")
-                else None
-            ),
+            logit_bias=_logit_bias_from(body),
         )
 
     def _stop_strings(body: dict) -> list[str]:
@@ -312,8 +457,18 @@ def build_app(server: EngineServer) -> FastAPI:
 
     @app.get("/health")
     async def health():
+        if server._fatal_error is not None:
+            return JSONResponse(
+                {"status": "error", "detail": server._fatal_error.splitlines()[-1]},
+                status_code=500,
+            )
         if not server._ready.is_set():
             return JSONResponse({"status": "starting"}, status_code=503)
+        if not server._thread.is_alive() and not server._stop.is_set():
+            return JSONResponse(
+                {"status": "error", "detail": "engine step thread died"},
+                status_code=500,
+            )
         return {"status": "ok"}
 
     @app.get("/version")
@@ -339,20 +494,20 @@ def build_app(server: EngineServer) -> FastAPI:
         """Drain a generation; with stop strings, truncate at the earliest
         match and abort the engine request early (the generator's finally
         clause issues the abort). `steps`, when given, collects every
-        per-token RequestOutput (chat logprobs)."""
+        per-token RequestOutput (chat logprobs). Stop detection is
+        incremental: O(new text) per token, not a full re-decode."""
         final = None
+        sd = StreamDecoder(server.tokenizer) if stops else None
+        sc = _StopScanner(stops) if stops else None
         async for o in gen:
             final = o
             if steps is not None:
                 steps.append(o)
             if stops:
-                # full re-decode: incremental piece concatenation loses
-                # inter-token separators (tokenizer-dependent)
-                text = server.tokenizer.decode(o.output_token_ids)
-                for stp in stops:
-                    idx = text.find(stp)
-                    if idx >= 0:
-                        return final, text[:idx], "stop"
+                text = sd.push(o.new_token_ids)
+                idx = sc.scan(text)
+                if idx >= 0:
+                    return final, text[:idx], "stop"
         return final, None, None
 
     @app.post("/v1/completions")
@@ -395,6 +550,8 @@ def build_app(server: EngineServer) -> FastAPI:
             tops: list = []
             final = None
             cut_text = cut_reason = None
+            sd = StreamDecoder(server.tokenizer) if stops else None
+            sc = _StopScanner(stops) if stops else None
             async for o in server.generate(toks, params, lora_id):
                 final = o
                 if o.logprob is not None:
@@ -402,10 +559,10 @@ def build_app(server: EngineServer) -> FastAPI:
                     toks_out.append(o.new_token_ids[-1])
                     tops.append(o.top_logprobs)
                 if stops:
-                    text_probe = server.tokenizer.decode(o.output_token_ids)
-                    hits = [text_probe.find(s_) for s_ in stops if s_ in text_probe]
-                    if hits:
-                        cut_text, cut_reason = text_probe[: min(hits)], "stop"
+                    text_probe = sd.push(o.new_token_ids)
+                    idx = sc.scan(text_probe)
+                    if idx >= 0:
+                        cut_text, cut_reason = text_probe[:idx], "stop"
                         break
             text = (
                 cut_text
@@ -528,9 +685,7 @@ def build_app(server: EngineServer) -> FastAPI:
             )
             vecs = enc.cpu().tolist()
         else:
-            vecs = await asyncio.get_running_loop().run_in_executor(
-                None, server.engine.embed, tok_lists
-            )
+            vecs = await server.embed(tok_lists)
         data = [
             {"object": "embedding", "index": i, "embedding": v}
             for i, v in enumerate(vecs)
@@ -558,9 +713,7 @@ def build_app(server: EngineServer) -> FastAPI:
                 None, server.embedder.score_pairs, tok_lists[0], tok_lists[1:]
             )
         else:
-            vecs = await asyncio.get_running_loop().run_in_executor(
-                None, server.engine.embed, tok_lists
-            )
+            vecs = await server.embed(tok_lists)
             qv = vecs[0]
             scores = [
                 sum(a * b for a, b in zip(qv, dv)) for dv in vecs[1:]
@@ -715,20 +868,37 @@ async def _stream_completion(server, toks, params, name, chat: bool,
     rid = f"{'chatcmpl' if chat else 'cmpl'}-{uuid.uuid4().hex[:12]}"
     created = int(time.time())
     n_out = 0
-    text_so_far = ""
+    sd = StreamDecoder(server.tokenizer)
+    sc = _StopScanner(stops) if stops else None
+    emitted = 0  # chars already streamed to the client
     async for o in server.generate(toks, params, lora_id):
-        # delta of the full decode: keeps inter-token separators exact
-        full = server.tokenizer.decode(_strip_stop(o, params))
-        piece = full[len(text_so_far):]
-        text_so_far = full
+        # incremental decode (O(new tokens) per step); drop a terminal
+        # stop token's own text, mirroring the non-stream _strip_stop
+        new_ids = o.new_token_ids
+        if (
+            o.finished
+            and o.finish_reason == "stop"
+            and new_ids
+            and new_ids[-1] in params.stop_token_ids
+        ):
+            new_ids = new_ids[:-1]
+        text = sd.push(new_ids)
         if stops:
-            hit = min((full.find(s) for s in stops if s in full), default=-1)
+            hit = sc.scan(text)
             if hit >= 0:
-                keep = full[:hit]
-                prev = len(text_so_far) - len(piece)
-                piece = keep[prev:] if hit > prev else ""
+                text = text[:hit]
                 o.finished = True
                 o.finish_reason = "stop"
+        if o.finished:
+            stable = text
+        else:
+            # hold back any tail that could still complete a stop string
+            # (never leak a stop's leading chars across chunk boundaries)
+            stable = (
+                text[: len(text) - _holdback_len(text, stops)] if stops else text
+            )
+        piece = stable[emitted:] if len(stable) > emitted else ""
+        emitted = max(emitted, len(stable))
         n_out = len(o.output_token_ids)
         if chat:
             chunk = {
@@ -739,7 +909,7 @@ async def _stream_completion(server, toks, params, name, chat: bool,
                 "choices": [
                     {
                         "index": 0,
-                        "delta": {"content": piece} if not o.finished else {},
+                        "delta": {"content": piece} if piece else {},
                         "finish_reason": o.finish_reason if o.finished else None,
                     }
                 ],
@@ -797,6 +967,8 @@ def _tp_worker_main(rank: int, world: int, port: int, cfg: EngineConfig) -> None
             engine.add_request(toks, params, request_id=rid, lora_id=lora_id)
         for rid in msg["aborts"]:
             engine.abort_request(rid)
+        for tok_lists in msg.get("embeds", []):
+            engine.embed(tok_lists)  # lockstep with rank 0 (all-reduces)
         if engine.has_work():
             engine.step()
     dist.destroy_process_group()

@@ -47,6 +47,36 @@ class HFTokenizer:
         return self._tok.decode(token_ids, skip_special_tokens=True)
 
 
+class StreamDecoder:
+    """Incremental detokenizer: O(window + new tokens) per push instead of
+    re-decoding the whole output every token (the r1 stop-string path was
+    O(n^2) in generation length). vLLM-style prefix/read offsets: each
+    decode call covers only the tokens since the last committed text, with
+    one already-committed anchor token so separator-dependent decoders
+    (sentencepiece spaces, byte-level merges) produce exact deltas; a
+    trailing U+FFFD holds the delta until the multi-byte sequence closes.
+    """
+
+    def __init__(self, tok):
+        self.tok = tok
+        self.ids: list[int] = []
+        self.text = ""
+        self._prefix = 0  # decode-window start (text before it committed)
+        self._read = 0    # ids[:_read] are reflected in self.text
+
+    def push(self, new_ids: list[int]) -> str:
+        if new_ids:
+            self.ids.extend(new_ids)
+            anchor = self.tok.decode(self.ids[self._prefix : self._read])
+            full = self.tok.decode(self.ids[self._prefix :])
+            if not full.endswith("�"):
+                if len(full) > len(anchor):
+                    self.text += full[len(anchor) :]
+                    self._prefix = self._read
+                self._read = len(self.ids)
+        return self.text
+
+
 def load_tokenizer(model: str, vocab_size: int, bos: int, eos: int):
     if os.path.isdir(model) and (
         os.path.exists(os.path.join(model, "tokenizer.json"))

@@ -15,7 +15,24 @@ from typing import Iterator
 import torch
 
 
+# quantization artifact suffixes: consumed alongside their weight (fp8)
+# or a loud error (formats we cannot dequantize) — ADVICE r1: never copy
+# fp8-coded bytes as if they were bf16 weights.
+_SCALE_SUFFIXES = (".weight_scale", ".weight_scale_inv", ".input_scale")
+_FP8_DTYPES = tuple(
+    getattr(torch, n)
+    for n in ("float8_e4m3fn", "float8_e5m2", "float8_e4m3fnuz", "float8_e5m2fnuz")
+    if hasattr(torch, n)
+)
+_UNSUPPORTED_QUANT_KEYS = ("qweight", "qzeros", "g_idx", "scales_zp")
+
+
 def iter_safetensors(model_dir: str) -> Iterator[tuple[str, torch.Tensor]]:
+    """Yield (name, tensor) over all shards; pre-quantized fp8 checkpoints
+    (e.g. Llama-3.1-*-FP8: fp8 weights + per-tensor `weight_scale`) are
+    dequantized on the fly so callers always see plain float weights.
+    Unsupported quant formats (AWQ/GPTQ packed ints) raise instead of
+    silently producing garbage."""
     from safetensors import safe_open
 
     files = sorted(glob.glob(os.path.join(model_dir, "*.safetensors")))
@@ -23,8 +40,41 @@ def iter_safetensors(model_dir: str) -> Iterator[tuple[str, torch.Tensor]]:
         raise FileNotFoundError(f"no *.safetensors under {model_dir}")
     for f in files:
         with safe_open(f, framework="pt", device="cpu") as sf:
+            keys = set(sf.keys())
             for key in sf.keys():
-                yield key, sf.get_tensor(key)
+                if key.endswith(_SCALE_SUFFIXES):
+                    continue  # consumed with its weight below
+                if any(key.endswith("." + s) for s in _UNSUPPORTED_QUANT_KEYS):
+                    raise ValueError(
+                        f"{model_dir}: packed-quantized checkpoint tensor "
+                        f"{key!r} is not supported — provide a bf16/fp16 or "
+                        "fp8(+weight_scale) checkpoint (serve with "
+                        "quantization=fp8 for W8A8)"
+                    )
+                w = sf.get_tensor(key)
+                if w.dtype in _FP8_DTYPES:
+                    scale_key = None
+                    for cand in (key + "_scale", key + "_scale_inv"):
+                        if cand in keys:
+                            scale_key = cand
+                            break
+                    if scale_key is None:
+                        raise ValueError(
+                            f"{model_dir}: fp8 tensor {key!r} has no "
+                            "weight_scale companion — cannot dequantize"
+                        )
+                    scale = sf.get_tensor(scale_key).float()
+                    if scale_key.endswith("_scale_inv"):
+                        scale = 1.0 / scale
+                    if scale.numel() not in (1, w.shape[0]):
+                        raise ValueError(
+                            f"{model_dir}: {scale_key!r} shape "
+                            f"{tuple(scale.shape)} unsupported (per-tensor "
+                            f"or per-row expected for {tuple(w.shape)})"
+                        )
+                    scale = scale.reshape(-1, *([1] * (w.dim() - 1)))
+                    w = w.float() * scale
+                yield key, w
 
 
 def _strip(name: str) -> str:

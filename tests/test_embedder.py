@@ -105,3 +105,38 @@ def test_bert_gpu_matches_cpu():
         vg.flatten(), vc.flatten(), dim=0
     )
     assert cos > 0.999
+
+
+def test_engine_embed_batched_equals_sequential():
+    """Packed multi-sequence embed must match one-at-a-time embed."""
+    import torch
+
+    from kubeai_amd.engine import EngineConfig, LLMEngine
+
+    eng = LLMEngine(
+        EngineConfig(model="llama-tiny", device="cpu", num_gpu_blocks=256,
+                     max_model_len=512, seed=5)
+    )
+    seqs = [list(range(10, 10 + n)) for n in (7, 33, 18, 64, 5)]
+    batched = eng.embed(seqs)
+    single = [eng.embed([s])[0] for s in seqs]
+    for b, s in zip(batched, single):
+        cos = torch.nn.functional.cosine_similarity(
+            torch.tensor(b), torch.tensor(s), dim=0
+        )
+        assert cos > 0.9999, cos
+    # no blocks leaked
+    assert eng.block_manager.usage() == 0.0
+
+
+def test_tp_lora_rejected():
+    from kubeai_amd.engine import EngineConfig
+    from kubeai_amd.engine.server import EngineServer
+
+    srv = EngineServer(
+        EngineConfig(model="llama-tiny", device="cpu"), "m", tp_size=2
+    )
+    import pytest as _pytest
+
+    with _pytest.raises(ValueError, match="tensor-parallel"):
+        srv.load_lora("a", None)

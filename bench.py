@@ -96,9 +96,12 @@ def _resolve_quant(args, model, use_cuda):
     if (
         args.quant == "auto"
         and args.tp == 1  # fused-fp8 path is validated single-GPU
-        and model in ("llama-3-8b", "llama-3-70b")
+        and model in ("llama-3-8b", "llama-3-70b", "mixtral-8x7b")
     ):
-        return "fp8"  # matches the FP8-named baseline configs
+        # llama: matches the FP8-named baseline configs. mixtral: fused
+        # single-quant fp8 won both interleaved same-box A/B pairs
+        # (796/848 vs 776/810 tok/s — profiles/r02_results.md)
+        return "fp8"
     return None
 
 

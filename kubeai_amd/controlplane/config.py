@@ -69,6 +69,10 @@ class SystemConfig:
     # HIGHER = more important). A Model's priorityClassName maps every
     # request for that model onto the engine's admission priority.
     priority_classes: dict = dataclasses.field(default_factory=dict)
+    # kubernetes: when set, Models persist as CRs and replicas run as
+    # Pods against this API server (kubestore.py). Keys: apiUrl (empty =
+    # in-cluster), namespace, engineImage, gpuResource, leaseName.
+    kubernetes: Optional[dict] = None
 
     def validate(self) -> None:
         if self.autoscaling.interval_seconds <= 0:
@@ -114,6 +118,7 @@ def load_config(path: Optional[str]) -> SystemConfig:
         priority_classes={
             str(k): int(v) for k, v in raw.get("priorityClasses", {}).items()
         },
+        kubernetes=raw.get("kubernetes"),
     )
     cfg.validate()
     return cfg

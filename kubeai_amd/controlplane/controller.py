@@ -173,6 +173,10 @@ class ModelController:
     async def reconcile(self, name: str) -> None:
         model = self.store.models.get(name)
         if model is None:
+            # model hard-deleted: clean any orphaned replicas (the
+            # reference relies on owner-reference GC here)
+            for r in self.store.list_replicas(model=name):
+                await self.runtime.delete(r.name)
             return
         replicas = self.store.list_replicas(model=name)
 
@@ -186,6 +190,7 @@ class ModelController:
             return
 
         # labels + bounds (reference: model_controller.go:96-105, 357-407)
+        pre = (dict(model.labels), list(model.finalizers), model.spec.replicas)
         model.labels.update(feature_labels(model))
         s = model.spec
         if s.replicas is None:
@@ -197,9 +202,13 @@ class ModelController:
                 s.replicas = s.max_replicas
 
         # cache (reference: cache.go)
+        if s.cache_profile and CACHE_EVICTION_FINALIZER not in model.finalizers:
+            model.finalizers.append(CACHE_EVICTION_FINALIZER)
+        if pre != (model.labels, model.finalizers, s.replicas):
+            # write back self-applied mutations (reference Update calls;
+            # K8s store PATCHes, in-memory store just notifies)
+            self.store.persist_model_meta(model)
         if s.cache_profile:
-            if CACHE_EVICTION_FINALIZER not in model.finalizers:
-                model.finalizers.append(CACHE_EVICTION_FINALIZER)
             loaded = await self.cache.ensure(model)
             self.store.update_status(name, cache_loaded=loaded)
             if not loaded:
@@ -305,7 +314,8 @@ class ModelController:
                     except Exception:
                         pass
             if changed:
-                self.store._notify("replica", rep.name)
+                # K8s store: adapter state -> pod labels (adapters.go:90-92)
+                self.store.persist_replica_adapters(rep)
 
 
 def _adapter_local_path(model: Model, adapter_name: str) -> Optional[str]:

@@ -1,0 +1,341 @@
+"""Minimal Kubernetes API client + Model/Pod manifest conversion.
+
+No `kubernetes` package exists in this image, so this speaks the REST API
+directly (the reference uses controller-runtime/client-go for the same
+verbs: typed GET/LIST/WATCH/PATCH — internal/manager/run.go:142-174,
+internal/k8sutils/apply.go:10-17). Scope is exactly what the control
+plane needs:
+
+  - CRUD + merge-patch on namespaced resources (Models, Pods, Leases,
+    ConfigMaps) and the Model `scale` subresource
+  - LIST with equality label selectors
+  - WATCH as a line-delimited JSON stream (informer feed)
+
+Auth: in-cluster service-account token/CA when present, else plain
+`api_url` (tests run against the in-process fake API server,
+kubeai_amd/controlplane/fakekube.py — the envtest analog).
+"""
+from __future__ import annotations
+
+import json
+import os
+from typing import Iterator, Optional
+
+import requests
+
+SA_DIR = "/var/run/secrets/kubernetes.io/serviceaccount"
+
+
+class ApiError(RuntimeError):
+    def __init__(self, status: int, body: str):
+        super().__init__(f"kube api {status}: {body[:300]}")
+        self.status = status
+        self.body = body
+
+
+class KubeClient:
+    def __init__(
+        self,
+        api_url: Optional[str] = None,
+        namespace: Optional[str] = None,
+        token: Optional[str] = None,
+        verify=None,
+        timeout: float = 10.0,
+    ):
+        if api_url is None:
+            host = os.environ.get("KUBERNETES_SERVICE_HOST")
+            port = os.environ.get("KUBERNETES_SERVICE_PORT", "443")
+            if not host:
+                raise ValueError("no api_url and not running in-cluster")
+            api_url = f"https://{host}:{port}"
+            if token is None and os.path.exists(os.path.join(SA_DIR, "token")):
+                with open(os.path.join(SA_DIR, "token")) as f:
+                    token = f.read().strip()
+            if verify is None and os.path.exists(os.path.join(SA_DIR, "ca.crt")):
+                verify = os.path.join(SA_DIR, "ca.crt")
+        if namespace is None:
+            ns_file = os.path.join(SA_DIR, "namespace")
+            if os.path.exists(ns_file):
+                with open(ns_file) as f:
+                    namespace = f.read().strip()
+            else:
+                namespace = "default"
+        self.base = api_url.rstrip("/")
+        self.namespace = namespace
+        self.timeout = timeout
+        self._s = requests.Session()
+        if token:
+            self._s.headers["Authorization"] = f"Bearer {token}"
+        if verify is not None:
+            self._s.verify = verify
+
+    # ------------------------------------------------------------- paths
+    def path(self, group: str, version: str, plural: str,
+             name: Optional[str] = None, subresource: Optional[str] = None,
+             namespace: Optional[str] = None) -> str:
+        ns = namespace or self.namespace
+        root = "/api" if group == "" else f"/apis/{group}"
+        p = f"{root}/{version}/namespaces/{ns}/{plural}"
+        if name:
+            p += f"/{name}"
+        if subresource:
+            p += f"/{subresource}"
+        return p
+
+    # ------------------------------------------------------------- verbs
+    def _req(self, method: str, path: str, *, params=None, body=None,
+             content_type: str = "application/json"):
+        r = self._s.request(
+            method,
+            self.base + path,
+            params=params,
+            data=json.dumps(body) if body is not None else None,
+            headers={"Content-Type": content_type},
+            timeout=self.timeout,
+        )
+        if r.status_code >= 400:
+            raise ApiError(r.status_code, r.text)
+        return r.json() if r.text else None
+
+    def get(self, path: str) -> dict:
+        return self._req("GET", path)
+
+    def list(self, path: str, label_selector: Optional[dict] = None) -> list[dict]:
+        params = {}
+        if label_selector:
+            params["labelSelector"] = ",".join(
+                f"{k}={v}" for k, v in sorted(label_selector.items())
+            )
+        out = self._req("GET", path, params=params)
+        return out.get("items", [])
+
+    def create(self, path: str, manifest: dict) -> dict:
+        return self._req("POST", path, body=manifest)
+
+    def replace(self, path: str, manifest: dict) -> dict:
+        return self._req("PUT", path, body=manifest)
+
+    def patch_merge(self, path: str, patch: dict) -> dict:
+        return self._req(
+            "PATCH", path, body=patch,
+            content_type="application/merge-patch+json",
+        )
+
+    def delete(self, path: str) -> Optional[dict]:
+        return self._req("DELETE", path)
+
+    def watch(self, path: str, resource_version: str = "0",
+              timeout: Optional[float] = None) -> Iterator[dict]:
+        """Stream watch events {type: ADDED|MODIFIED|DELETED, object: …}.
+        Raises on disconnect — callers re-list + re-watch (informer loop)."""
+        r = self._s.get(
+            self.base + path,
+            params={"watch": "true", "resourceVersion": resource_version},
+            stream=True,
+            timeout=timeout or 3600,
+        )
+        if r.status_code >= 400:
+            raise ApiError(r.status_code, r.text)
+        for line in r.iter_lines():
+            if line:
+                yield json.loads(line)
+
+
+# ======================================================================
+# Model CR <-> dataclass (field names match deploy/crds/kubeai.org_models
+# .yaml, which mirrors the reference api/k8s/v1/model_types.go)
+
+def model_to_manifest(m) -> dict:
+    from .crd import Model  # noqa: F401
+
+    s = m.spec
+    spec: dict = {
+        "url": s.url,
+        "features": list(s.features),
+        "engine": s.engine,
+    }
+    if s.adapters:
+        spec["adapters"] = [{"name": a.name, "url": a.url} for a in s.adapters]
+    if s.resource_profile:
+        spec["resourceProfile"] = s.resource_profile
+    if s.cache_profile:
+        spec["cacheProfile"] = s.cache_profile
+    if s.image:
+        spec["image"] = s.image
+    if s.args:
+        spec["args"] = list(s.args)
+    if s.env:
+        spec["env"] = dict(s.env)
+    if s.replicas is not None:
+        spec["replicas"] = s.replicas
+    spec["minReplicas"] = s.min_replicas
+    if s.max_replicas is not None:
+        spec["maxReplicas"] = s.max_replicas
+    if s.autoscaling_disabled:
+        spec["autoscalingDisabled"] = True
+    spec["targetRequests"] = s.target_requests
+    spec["scaleDownDelaySeconds"] = s.scale_down_delay_seconds
+    lb = s.load_balancing
+    spec["loadBalancing"] = {
+        "strategy": lb.strategy,
+        "prefixHash": {
+            "meanLoadFactor": lb.prefix_hash.mean_load_percentage,
+            "replication": lb.prefix_hash.replication,
+            "prefixCharLength": lb.prefix_hash.prefix_char_length,
+        },
+    }
+    if s.files:
+        spec["files"] = [{"path": f.path, "content": f.content} for f in s.files]
+    if s.priority_class_name:
+        spec["priorityClassName"] = s.priority_class_name
+    if s.owner:
+        spec["owner"] = s.owner
+    meta: dict = {"name": m.name}
+    if m.labels:
+        meta["labels"] = dict(m.labels)
+    if m.annotations:
+        meta["annotations"] = dict(m.annotations)
+    if m.finalizers:
+        meta["finalizers"] = list(m.finalizers)
+    if m.uid:
+        meta["uid"] = m.uid
+    return {
+        "apiVersion": "kubeai.org/v1",
+        "kind": "Model",
+        "metadata": meta,
+        "spec": spec,
+        "status": {
+            "replicas": {
+                "all": m.status.replicas_all,
+                "ready": m.status.replicas_ready,
+            },
+            "cache": {"loaded": m.status.cache_loaded},
+        },
+    }
+
+
+def model_from_manifest(obj: dict):
+    from .crd import (AdapterSpec, FileSpec, LoadBalancingSpec, Model,
+                      ModelSpec, ModelStatus, PrefixHashSpec)
+
+    meta = obj.get("metadata", {})
+    s = obj.get("spec", {})
+    lb = s.get("loadBalancing", {}) or {}
+    ph = lb.get("prefixHash", {}) or {}
+    spec = ModelSpec(
+        url=s.get("url", ""),
+        features=list(s.get("features") or ["TextGeneration"]),
+        engine=s.get("engine", "KubeAIEngine"),
+        adapters=[
+            AdapterSpec(name=a.get("name", ""), url=a.get("url", ""))
+            for a in (s.get("adapters") or [])
+        ],
+        resource_profile=s.get("resourceProfile", ""),
+        cache_profile=s.get("cacheProfile", ""),
+        image=s.get("image", ""),
+        args=list(s.get("args") or []),
+        env=dict(s.get("env") or {}),
+        replicas=s.get("replicas"),
+        min_replicas=int(s.get("minReplicas") or 0),
+        max_replicas=s.get("maxReplicas"),
+        autoscaling_disabled=bool(s.get("autoscalingDisabled") or False),
+        target_requests=int(s.get("targetRequests") or 100),
+        scale_down_delay_seconds=int(s.get("scaleDownDelaySeconds") or 30),
+        load_balancing=LoadBalancingSpec(
+            strategy=lb.get("strategy", "LeastLoad"),
+            prefix_hash=PrefixHashSpec(
+                mean_load_percentage=int(ph.get("meanLoadFactor") or 125),
+                replication=int(ph.get("replication") or 256),
+                prefix_char_length=int(ph.get("prefixCharLength") or 100),
+            ),
+        ),
+        files=[
+            FileSpec(path=f.get("path", ""), content=f.get("content", ""))
+            for f in (s.get("files") or [])
+        ],
+        priority_class_name=s.get("priorityClassName", ""),
+        owner=s.get("owner", ""),
+    )
+    st = obj.get("status", {}) or {}
+    reps = st.get("replicas", {}) or {}
+    status = ModelStatus(
+        replicas_all=int(reps.get("all") or 0),
+        replicas_ready=int(reps.get("ready") or 0),
+        cache_loaded=bool((st.get("cache") or {}).get("loaded") or False),
+    )
+    m = Model(
+        name=meta.get("name", ""),
+        spec=spec,
+        status=status,
+        labels=dict(meta.get("labels") or {}),
+        annotations=dict(meta.get("annotations") or {}),
+        finalizers=list(meta.get("finalizers") or []),
+        generation=int(meta.get("generation") or 0),
+        deleted=bool(meta.get("deletionTimestamp")),
+        uid=meta.get("uid", ""),
+    )
+    return m
+
+
+# ======================================================================
+# Engine pod manifest — the reference contract (engine_vllm.go:12-167):
+# container port 8000 + port annotation, /health startup/readiness/
+# liveness probes, `model`/`pod-hash` labels, resource-profile GPU counts.
+
+ENGINE_PORT = 8000
+
+
+def pod_manifest_for(model, name: str, spec_hash: str, n_gpus: int,
+                     image: str, namespace: str,
+                     gpu_resource: str = "amd.com/gpu") -> dict:
+    from .crd import POD_HASH_LABEL, POD_MODEL_LABEL
+
+    args = [
+        "--model", model.spec.url,
+        "--served-model-name", model.name,
+        "--port", str(ENGINE_PORT),
+    ] + list(model.spec.args)
+    env = [{"name": k, "value": v} for k, v in sorted(model.spec.env.items())]
+    resources = {}
+    if n_gpus > 0:
+        resources = {
+            "requests": {gpu_resource: str(n_gpus)},
+            "limits": {gpu_resource: str(n_gpus)},
+        }
+    probe = {
+        "httpGet": {"path": "/health", "port": ENGINE_PORT},
+        "periodSeconds": 2,
+    }
+    return {
+        "apiVersion": "v1",
+        "kind": "Pod",
+        "metadata": {
+            "name": name,
+            "namespace": namespace,
+            "labels": {
+                POD_MODEL_LABEL: model.name,
+                POD_HASH_LABEL: spec_hash,
+            },
+            "annotations": {"model-pod-port": str(ENGINE_PORT)},
+        },
+        "spec": {
+            "containers": [
+                {
+                    "name": "server",
+                    "image": image,
+                    "command": ["python", "-m", "kubeai_amd.engine.server"],
+                    "args": args,
+                    "env": env,
+                    "resources": resources,
+                    "ports": [{"containerPort": ENGINE_PORT}],
+                    # reference probe contract engine_vllm.go:101-138
+                    "startupProbe": {**probe, "failureThreshold": 5400},
+                    "readinessProbe": {**probe, "failureThreshold": 3},
+                    "livenessProbe": {
+                        **probe, "periodSeconds": 10, "failureThreshold": 3,
+                    },
+                }
+            ],
+            "restartPolicy": "Never",
+        },
+    }

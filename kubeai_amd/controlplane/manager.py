@@ -34,8 +34,32 @@ class Manager:
         broker: Optional[MemBroker] = None,
     ):
         self.cfg = cfg or SystemConfig()
-        self.store = Store()
-        self.runtime = runtime or LocalProcessRuntime(self.store, n_gpus=self.cfg.n_gpus)
+        kube = self.cfg.kubernetes
+        if kube is not None:
+            # K8s substrate: Models as CRs, replicas as Pods, Lease
+            # election (reference run.go:142-174 topology)
+            from .kubeclient import KubeClient
+            from .kubestore import KubeRuntime, KubeStore, LeaseElection
+
+            kc = KubeClient(
+                api_url=kube.get("apiUrl") or None,
+                namespace=kube.get("namespace"),
+            )
+            self.store = KubeStore(kc)
+            self.runtime = runtime or KubeRuntime(
+                self.store,
+                image=kube.get("engineImage", "kubeai-amd-engine:latest"),
+                gpu_resource=kube.get("gpuResource", "amd.com/gpu"),
+            )
+            self.election = LeaseElection(
+                kc, lease_name=kube.get("leaseName", "kubeai.org")
+            )
+        else:
+            self.store = Store()
+            self.runtime = runtime or LocalProcessRuntime(
+                self.store, n_gpus=self.cfg.n_gpus
+            )
+            self.election = Election(self.cfg.leader_lock_path)
         self.model_client = ModelClient(
             self.store,
             required_consecutive_scale_downs=self.cfg.autoscaling.required_consecutive_scale_downs,
@@ -48,7 +72,6 @@ class Manager:
             resource_profiles=self.cfg.resource_profiles,
             cache=CacheManager(self.cfg.cache_dir),
         )
-        self.election = Election(self.cfg.leader_lock_path)
         self.autoscaler = Autoscaler(
             self.store,
             self.model_client,
@@ -86,6 +109,8 @@ class Manager:
             )
 
     async def start(self) -> None:
+        if hasattr(self.store, "start"):
+            await self.store.start()  # warm the informer cache first
         self.lb.start()
         self.controller.start()
         self.election.start()
@@ -103,6 +128,8 @@ class Manager:
         await self.proxy.close()
         if hasattr(self.runtime, "shutdown"):
             await self.runtime.shutdown()
+        if hasattr(self.store, "stop"):
+            self.store.stop()
 
 
 async def run(cfg: SystemConfig) -> None:

@@ -132,6 +132,16 @@ class Store:
         if changed:
             self._notify("model", name)
 
+    # write-back hooks: the in-memory store shares live objects with the
+    # controller, so reconcile-path mutations (labels/finalizers/bounds,
+    # adapter state) are already visible; the K8s-backed store overrides
+    # these to PATCH the API server (kubestore.py)
+    def persist_model_meta(self, model: Model) -> None:
+        self._notify("model", model.name)
+
+    def persist_replica_adapters(self, rep: "Replica") -> None:
+        self._notify("replica", rep.name, rep.model)
+
     # ---------------------------------------------------------- replicas
     def add_replica(self, rep: Replica) -> Replica:
         rep.created_seq = next(self._seq)

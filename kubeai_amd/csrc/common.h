@@ -88,8 +88,13 @@ DEV_INLINE ushort8 e5m2x8_to_bf16x8(uint64_t raw) {
   const uint32_t w[2] = {(uint32_t)raw, (uint32_t)(raw >> 32)};
 #pragma unroll
   for (int i = 0; i < 2; ++i) {
-    const floatx2 lo = __builtin_amdgcn_cvt_pk_f32_bf8(w[i], false);
-    const floatx2 hi = __builtin_amdgcn_cvt_pk_f32_bf8(w[i], true);
+    floatx2 lo = __builtin_amdgcn_cvt_pk_f32_bf8(w[i], false);
+    floatx2 hi = __builtin_amdgcn_cvt_pk_f32_bf8(w[i], true);
+    // opaque barrier: this compiler's value tracking mis-models the
+    // packed convert's low mantissa bit and would fold it out of the
+    // bit-level truncation below (device-verified, scripts/dbg_cvt.hip
+    // — arithmetic consumers like the decode kernel are unaffected)
+    asm("" : "+v"(lo), "+v"(hi));
     out[4 * i + 0] = (ushort)(__builtin_bit_cast(uint32_t, lo.x) >> 16);
     out[4 * i + 1] = (ushort)(__builtin_bit_cast(uint32_t, lo.y) >> 16);
     out[4 * i + 2] = (ushort)(__builtin_bit_cast(uint32_t, hi.x) >> 16);

@@ -1,0 +1,378 @@
+"""K8s-backed control plane against the in-process fake API server — the
+reference's envtest tier (test/integration/main_test.go:83-157): Models
+persist as CRs, replicas are Pods whose readiness tests flip by PATCHing
+status (utils_test.go:118-132), leadership is a coordination.k8s.io Lease.
+"""
+import asyncio
+import time
+
+import pytest
+
+from kubeai_amd.controlplane.controller import ModelController
+from kubeai_amd.controlplane.crd import (POD_HASH_LABEL, POD_MODEL_LABEL,
+                                         Model, ModelSpec)
+from kubeai_amd.controlplane.fakekube import FakeKubeApiServer
+from kubeai_amd.controlplane.kubeclient import KubeClient
+from kubeai_amd.controlplane.kubestore import (MODELS, PODS, KubeRuntime,
+                                               KubeStore, LeaseElection)
+
+
+@pytest.fixture()
+def api():
+    srv = FakeKubeApiServer().start()
+    yield srv
+    srv.stop()
+
+
+def kc_for(api) -> KubeClient:
+    return KubeClient(api_url=api.url, namespace="kubeai")
+
+
+async def wait_for(cond, timeout=10.0, msg="condition"):
+    t0 = time.monotonic()
+    while time.monotonic() - t0 < timeout:
+        v = cond()
+        if v:
+            return v
+        await asyncio.sleep(0.02)
+    raise AssertionError(f"timed out waiting for {msg}")
+
+
+def mark_pod_ready(kc: KubeClient, name: str, ip: str = "10.0.0.5") -> None:
+    # the envtest pattern: tests write pod status; no kubelet exists
+    kc.patch_merge(
+        kc.path(*PODS, name=name, subresource="status"),
+        {"status": {"phase": "Running", "podIP": ip,
+                    "conditions": [{"type": "Ready", "status": "True"}]}},
+    )
+
+
+# ----------------------------------------------------------------------
+def test_model_cr_persistence(api):
+    """Models survive a control-plane restart (CRs live in the API
+    server, unlike the in-memory store)."""
+
+    async def main():
+        store = KubeStore(kc_for(api))
+        await store.start()
+        store.apply_model(
+            Model(name="m1", spec=ModelSpec(url="hf://org/m1", min_replicas=1))
+        )
+        store.stop()
+        # "restart": a brand-new store against the same API server
+        store2 = KubeStore(kc_for(api))
+        await store2.start()
+        assert "m1" in store2.models
+        assert store2.models["m1"].spec.url == "hf://org/m1"
+        assert store2.models["m1"].uid
+        store2.stop()
+
+    asyncio.run(main())
+
+
+def test_controller_reconciles_pods(api):
+    """Model CR -> engine Pods with the reference contract (labels,
+    port annotation, /health probes); readiness flows back to status."""
+
+    async def main():
+        kc = kc_for(api)
+        store = KubeStore(kc)
+        await store.start()
+        runtime = KubeRuntime(store)
+        ctrl = ModelController(store, runtime)
+        ctrl.start()
+        try:
+            store.apply_model(
+                Model(name="m1", spec=ModelSpec(
+                    url="hf://org/m1", min_replicas=2,
+                    resource_profile="amd-gpu-mi355x:1",
+                ))
+            )
+            pods = await wait_for(
+                lambda: (kc.list(kc.path(*PODS),
+                                 {POD_MODEL_LABEL: "m1"}) or None)
+                and len(kc.list(kc.path(*PODS), {POD_MODEL_LABEL: "m1"})) == 2
+                and kc.list(kc.path(*PODS), {POD_MODEL_LABEL: "m1"}),
+                msg="2 pods",
+            )
+            pod = pods[0]
+            # contract checks (reference engine_vllm.go:12-138)
+            assert pod["metadata"]["labels"][POD_HASH_LABEL]
+            assert pod["metadata"]["annotations"]["model-pod-port"] == "8000"
+            c = pod["spec"]["containers"][0]
+            assert c["readinessProbe"]["httpGet"]["path"] == "/health"
+            assert c["startupProbe"]["failureThreshold"] == 5400
+            assert c["resources"]["requests"]["amd.com/gpu"] == "1"
+            # feature labels persisted onto the CR
+            cr = kc.get(kc.path(*MODELS, name="m1"))
+            assert cr["metadata"]["labels"][
+                "features.kubeai.org/TextGeneration"] == "true"
+
+            for p in pods:
+                mark_pod_ready(kc, p["metadata"]["name"])
+            await wait_for(
+                lambda: store.models["m1"].status.replicas_ready == 2,
+                msg="status ready=2",
+            )
+            # status also persisted on the CR
+            cr = kc.get(kc.path(*MODELS, name="m1"))
+            assert cr["status"]["replicas"]["ready"] == 2
+            # replica address from podIP + port annotation
+            reps = store.list_replicas(model="m1")
+            assert all(r.address == "10.0.0.5:8000" for r in reps)
+        finally:
+            await ctrl.stop()
+            store.stop()
+
+    asyncio.run(main())
+
+
+def test_scale_subresource_and_scale_down(api):
+    async def main():
+        kc = kc_for(api)
+        store = KubeStore(kc)
+        await store.start()
+        ctrl = ModelController(store, KubeRuntime(store))
+        ctrl.start()
+        try:
+            store.apply_model(
+                Model(name="m2", spec=ModelSpec(url="hf://org/m2",
+                                                min_replicas=0, replicas=2))
+            )
+            await wait_for(
+                lambda: len(kc.list(kc.path(*PODS), {POD_MODEL_LABEL: "m2"})) == 2,
+                msg="2 pods",
+            )
+            # scale via the subresource (the autoscaler's write path)
+            store.scale_model("m2", 1)
+            await wait_for(
+                lambda: len(kc.list(kc.path(*PODS), {POD_MODEL_LABEL: "m2"})) == 1,
+                msg="scale down to 1",
+            )
+            sc = kc.get(kc.path(*MODELS, name="m2", subresource="scale"))
+            assert sc["spec"]["replicas"] == 1
+        finally:
+            await ctrl.stop()
+            store.stop()
+
+    asyncio.run(main())
+
+
+def test_out_of_band_create_via_watch(api):
+    """A CR created by `kubectl` (raw API client) reaches the controller
+    through the watch — no in-process apply needed."""
+
+    async def main():
+        kc = kc_for(api)
+        store = KubeStore(kc)
+        await store.start()
+        ctrl = ModelController(store, KubeRuntime(store))
+        ctrl.start()
+        try:
+            kc.create(kc.path(*MODELS), {
+                "apiVersion": "kubeai.org/v1", "kind": "Model",
+                "metadata": {"name": "kubectl-model"},
+                "spec": {"url": "hf://org/x", "minReplicas": 1},
+            })
+            await wait_for(
+                lambda: len(kc.list(kc.path(*PODS),
+                                    {POD_MODEL_LABEL: "kubectl-model"})) == 1,
+                msg="pod for kubectl-created model",
+            )
+        finally:
+            await ctrl.stop()
+            store.stop()
+
+    asyncio.run(main())
+
+
+def test_model_delete_removes_pods_and_cr(api):
+    async def main():
+        kc = kc_for(api)
+        store = KubeStore(kc)
+        await store.start()
+        ctrl = ModelController(store, KubeRuntime(store))
+        ctrl.start()
+        try:
+            store.apply_model(
+                Model(name="m3", spec=ModelSpec(url="hf://org/m3",
+                                                min_replicas=1))
+            )
+            await wait_for(
+                lambda: len(kc.list(kc.path(*PODS), {POD_MODEL_LABEL: "m3"})) == 1,
+                msg="pod",
+            )
+            store.delete_model("m3")
+            await wait_for(
+                lambda: not kc.list(kc.path(*PODS), {POD_MODEL_LABEL: "m3"}),
+                msg="pods gone",
+            )
+            await wait_for(
+                lambda: not kc.list(kc.path(*MODELS)),
+                msg="CR gone",
+            )
+        finally:
+            await ctrl.stop()
+            store.stop()
+
+    asyncio.run(main())
+
+
+def test_cache_finalizer_deletion_flow(api, tmp_path):
+    """cacheProfile models: eviction finalizer added; DELETE leaves the CR
+    with deletionTimestamp until the controller evicts and strips it
+    (reference cache.go:136-217)."""
+
+    async def main():
+        kc = kc_for(api)
+        store = KubeStore(kc)
+        await store.start()
+        from kubeai_amd.controlplane.controller import CacheManager
+
+        ctrl = ModelController(
+            store, KubeRuntime(store),
+            cache=CacheManager(base_dir=str(tmp_path / "cache")),
+        )
+        ctrl.start()
+        try:
+            store.apply_model(
+                Model(name="m4", spec=ModelSpec(
+                    url="hf://org/m4", min_replicas=0,
+                    cache_profile="shared-fs",
+                ))
+            )
+            await wait_for(
+                lambda: "kubeai.org/cache-eviction"
+                in (kc.get(kc.path(*MODELS, name="m4"))["metadata"].get(
+                    "finalizers") or []),
+                msg="finalizer added",
+            )
+            store.delete_model("m4")
+            await wait_for(
+                lambda: not store.models.get("m4")
+                or store.models["m4"].deleted is True,
+                msg="deletion seen",
+            )
+            # finalizer cleared -> CR actually removed
+            await wait_for(
+                lambda: not kc.list(kc.path(*MODELS)),
+                msg="CR fully removed after finalizer",
+            )
+        finally:
+            await ctrl.stop()
+            store.stop()
+
+    asyncio.run(main())
+
+
+def test_lease_election(api):
+    async def main():
+        kc = kc_for(api)
+        e1 = LeaseElection(kc, identity="a", lease_duration=0.6)
+        e2 = LeaseElection(kc_for(api), identity="b", lease_duration=0.6)
+        assert e1.tick() is True          # a acquires
+        assert e2.tick() is False         # b blocked by live lease
+        assert e1.tick() is True          # a renews
+        await asyncio.sleep(0.8)          # a's lease expires (no renew)
+        assert e2.tick() is True          # b takes over
+        assert e1.tick() is False         # a sees b's live lease
+        await e2.stop()                   # b releases the holder
+        assert e1.tick() is True          # a reacquires after release
+
+    asyncio.run(main())
+
+
+def test_pod_failed_phase_recreated(api):
+    """A pod that dies (phase=Failed) is recreated by the reconciler —
+    the reference's pod-recovery behavior (model_pod_recovery_test.go)."""
+
+    async def main():
+        kc = kc_for(api)
+        store = KubeStore(kc)
+        await store.start()
+        ctrl = ModelController(store, KubeRuntime(store))
+        ctrl.start()
+        try:
+            store.apply_model(
+                Model(name="m5", spec=ModelSpec(url="hf://org/m5",
+                                                min_replicas=1))
+            )
+            pods = await wait_for(
+                lambda: kc.list(kc.path(*PODS), {POD_MODEL_LABEL: "m5"}) or None,
+                msg="pod",
+            )
+            first = pods[0]["metadata"]["name"]
+            kc.patch_merge(
+                kc.path(*PODS, name=first, subresource="status"),
+                {"status": {"phase": "Failed"}},
+            )
+            await wait_for(
+                lambda: (lambda ps: len(ps) == 1
+                         and ps[0]["metadata"]["name"] != first)(
+                    kc.list(kc.path(*PODS), {POD_MODEL_LABEL: "m5"})),
+                msg="failed pod replaced",
+            )
+        finally:
+            await ctrl.stop()
+            store.stop()
+
+    asyncio.run(main())
+
+
+def test_full_manager_kube_mode_inference(api):
+    """The whole Manager on the K8s substrate: Model CR -> Pod -> mark
+    ready (with the model-pod-ip override pointing at a fake engine) ->
+    chat completion through the gateway proxies to it. The reference's
+    envtest inference-flow trick (utils_test.go:150-159, main_test.go:258).
+    """
+    from httpx import ASGITransport, AsyncClient
+
+    from tests.test_controlplane import FakeBackend
+    from kubeai_amd.controlplane.config import SystemConfig
+    from kubeai_amd.controlplane.manager import Manager
+
+    async def main():
+        backend = FakeBackend()
+        await backend.start()
+        cfg = SystemConfig(
+            kubernetes={"apiUrl": api.url, "namespace": "kubeai"}
+        )
+        mgr = Manager(cfg)
+        await mgr.start()
+        kc = mgr.store.kc
+        try:
+            ip, port = backend.address.split(":")
+            mgr.store.apply_model(
+                Model(
+                    name="chat-model",
+                    spec=ModelSpec(url="hf://org/chat", min_replicas=1),
+                    annotations={"model-pod-ip": ip, "model-pod-port": port},
+                )
+            )
+            pods = await wait_for(
+                lambda: kc.list(kc.path(*PODS), {POD_MODEL_LABEL: "chat-model"})
+                or None,
+                msg="pod created",
+            )
+            mark_pod_ready(kc, pods[0]["metadata"]["name"])
+            await wait_for(
+                lambda: any(
+                    r.ready for r in mgr.store.list_replicas(model="chat-model")
+                ),
+                msg="replica ready",
+            )
+            async with AsyncClient(
+                transport=ASGITransport(app=mgr.app), base_url="http://kubeai"
+            ) as client:
+                r = await client.post(
+                    "/openai/v1/completions",
+                    json={"model": "chat-model", "prompt": "hi", "max_tokens": 4},
+                )
+                assert r.status_code == 200, r.text
+                assert r.json()["choices"][0]["text"] == "ok"
+            assert backend.requests, "request never reached the engine"
+        finally:
+            await mgr.stop()
+            await backend.stop()
+
+    asyncio.run(main())

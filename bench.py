@@ -74,6 +74,10 @@ def get_args():
     p.add_argument("--gpu-mem-util", type=float, default=0.90)
     p.add_argument("--max-model-len", type=int, default=8192)
     p.add_argument("--prefill-interval", type=int, default=4)
+    p.add_argument("--temperature", type=float, default=0.0,
+                   help="sampling temperature (0 = greedy, the k6 default)")
+    p.add_argument("--top-p", type=float, default=1.0)
+    p.add_argument("--top-k", type=int, default=0)
     p.add_argument("--kv-cache-dtype", choices=["auto", "fp8_e5m2"],
                    default="auto",
                    help="fp8_e5m2 halves KV bytes (pending device validation)")
@@ -210,7 +214,8 @@ def main():
     )
 
     log(f"engine ready: {eng.runner.num_blocks} kv blocks")
-    sp = SamplingParams(max_tokens=args.max_tokens, temperature=0.0, ignore_eos=True)
+    sp = SamplingParams(max_tokens=args.max_tokens, temperature=args.temperature,
+                        top_p=args.top_p, top_k=args.top_k, ignore_eos=True)
     sys_rng = random.Random(7)
     system_prompt = [
         sys_rng.randrange(1000, arch.vocab_size - 10)

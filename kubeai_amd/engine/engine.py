@@ -119,6 +119,7 @@ class LLMEngine:
 
     def abort_request(self, request_id: str) -> None:
         self.scheduler.abort(request_id)
+        self.runner.release_request(request_id)
 
     def has_work(self) -> bool:
         return self.scheduler.has_work()
@@ -146,6 +147,8 @@ class LLMEngine:
         now = time.monotonic()
         finished = self.scheduler.finish_step(out, sampled, now)
         finished_ids = {r.request_id for r in finished}
+        for rid in finished_ids:
+            self.runner.release_request(rid)
         for ss in out.all_seqs:
             req = ss.req
             if not ss.samples:

@@ -383,35 +383,49 @@ class ModelRunner:
             raise RuntimeError(
                 f"logits rows {logits.shape[0]} != sample reqs {len(sample_reqs)}"
             )
-        # OpenAI presence/frequency penalties over generated tokens
+        # OpenAI presence/frequency penalties over generated tokens:
+        # persistent per-request GPU count tensors (updated incrementally
+        # after sampling, freed on finish) + ONE batched penalty kernel —
+        # no per-row Python loop or per-step re-upload of output ids
+        V = logits.shape[1]
+        dev_l = logits.device
+        pen_rows = [
+            (i, r)
+            for i, r in enumerate(sample_reqs)
+            if (r.params.presence_penalty or r.params.frequency_penalty)
+            and r.num_generated > 0
+        ]
+        if pen_rows:
+            counts = torch.stack([self._pen_counts_for(r, V) for _, r in pen_rows])
+            idx = torch.tensor([i for i, _ in pen_rows], device=dev_l)
+            fp_t = torch.tensor(
+                [r.params.frequency_penalty for _, r in pen_rows],
+                dtype=torch.float32, device=dev_l,
+            ).unsqueeze(1)
+            pp_t = torch.tensor(
+                [r.params.presence_penalty for _, r in pen_rows],
+                dtype=torch.float32, device=dev_l,
+            ).unsqueeze(1)
+            logits[idx] -= (fp_t * counts + pp_t * (counts > 0)).to(logits.dtype)
+        # logit_bias: flat-indexed single scatter over all rows (ids
+        # validated at parse time; filtered again so a bad id can never
+        # kill the step loop)
+        rows, cols, vals = [], [], []
         for i, r in enumerate(sample_reqs):
-            pp, fp = r.params.presence_penalty, r.params.frequency_penalty
-            if (pp or fp) and r.num_generated > 0:
-                out_t = torch.tensor(
-                    r.output_token_ids, dtype=torch.int64, device=logits.device
-                )
-                counts = torch.zeros(
-                    logits.shape[1], dtype=torch.float32, device=logits.device
-                )
-                counts.scatter_add_(0, out_t, torch.ones_like(out_t, dtype=torch.float32))
-                logits[i] -= fp * counts + pp * (counts > 0).float()
             if r.params.logit_bias:
-                # ids validated at parse time (server RequestError); filter
-                # again here so a bad id can never kill the step loop
-                V = logits.shape[1]
-                pairs = [
-                    (t, b) for t, b in r.params.logit_bias.items() if 0 <= t < V
-                ]
-                if pairs:
-                    ids = torch.tensor(
-                        [t for t, _ in pairs], dtype=torch.int64,
-                        device=logits.device,
-                    )
-                    vals = torch.tensor(
-                        [b for _, b in pairs], dtype=logits.dtype,
-                        device=logits.device,
-                    )
-                    logits[i].scatter_add_(0, ids, vals)
+                for t, b in r.params.logit_bias.items():
+                    if 0 <= t < V:
+                        rows.append(i)
+                        cols.append(t)
+                        vals.append(b)
+        if rows:
+            flat = torch.tensor(
+                [ri * V + ci for ri, ci in zip(rows, cols)],
+                dtype=torch.int64, device=dev_l,
+            )
+            logits.view(-1).scatter_add_(
+                0, flat, torch.tensor(vals, dtype=logits.dtype, device=dev_l)
+            )
         temps = [r.params.temperature for r in sample_reqs]
         if all(t <= 0.0 for t in temps):
             tokens = ops.greedy_sample(logits.contiguous())
@@ -464,33 +478,65 @@ class ModelRunner:
                     self.last_top_logprobs[r.request_id] = [
                         (int(i), float(v)) for i, v in zip(ids[:n], vs[:n])
                     ]
+        # keep persistent penalty counts current with this step's samples
+        upd = [
+            (r, t) for r, t in zip(sample_reqs, tokens)
+            if (r.params.presence_penalty or r.params.frequency_penalty)
+        ]
+        if upd:
+            for r, t in upd:
+                c = self._pen_counts_for(r, logits.shape[1])
+                c[int(t)] += 1.0
         return {r.request_id: int(t) for r, t in zip(sample_reqs, tokens)}
+
+    # ------------------------------------------------------------------
+    def _pen_counts_for(self, r, vocab: int) -> torch.Tensor:
+        """Per-request generated-token counts on device (penalty state)."""
+        counts = getattr(self, "_pen_counts", None)
+        if counts is None:
+            counts = self._pen_counts = {}
+        c = counts.get(r.request_id)
+        if c is None:
+            c = torch.zeros(vocab, dtype=torch.float32, device=self.device)
+            if r.output_token_ids:
+                out_t = torch.tensor(
+                    r.output_token_ids, dtype=torch.int64, device=self.device
+                )
+                c.scatter_add_(
+                    0, out_t, torch.ones_like(out_t, dtype=torch.float32)
+                )
+            counts[r.request_id] = c
+        return c
+
+    def release_request(self, request_id: str) -> None:
+        """Drop per-request runner state (called when a request ends)."""
+        if getattr(self, "_pen_counts", None):
+            self._pen_counts.pop(request_id, None)
 
 
 def _apply_topk_topp(logits, top_ps, top_ks, temps):
     """Mask logits outside the per-row top-k / top-p (nucleus) sets.
 
-    Sampling from the masked logits via Gumbel-argmax is exactly nucleus
-    sampling: probabilities renormalize implicitly.
+    Fully batched on-device (one sort + vectorized masking — no per-row
+    Python loop, r1's CPU round trip). Sampling from the masked logits
+    via Gumbel-argmax is exactly nucleus sampling: probabilities
+    renormalize implicitly. Greedy rows (temp<=0) pass through unmasked.
     """
-    out = logits.clone()
-    sorted_logits, sorted_idx = out.sort(dim=-1, descending=True)
-    for i, (tp, tk, tt) in enumerate(zip(top_ps, top_ks, temps)):
-        if tt <= 0:
-            continue  # greedy rows ignore the mask
-        row = sorted_logits[i]
-        keep = torch.ones_like(row, dtype=torch.bool)
-        if tk and tk > 0:
-            keep[tk:] = False
-        if tp < 1.0:
-            probs = torch.softmax(row / tt, dim=-1)
-            csum = probs.cumsum(0)
-            # keep tokens while cumulative prob (exclusive) < top_p
-            keep &= (csum - probs) < tp
-            keep[0] = True
-        drop_idx = sorted_idx[i][~keep]
-        out[i, drop_idx] = float("-inf")
-    return out
+    S, V = logits.shape
+    dev = logits.device
+    tp = torch.tensor(top_ps, device=dev, dtype=torch.float32).unsqueeze(1)
+    tk = torch.tensor(top_ks, device=dev, dtype=torch.int64).unsqueeze(1)
+    tt = torch.tensor(temps, device=dev, dtype=torch.float32).unsqueeze(1)
+    sorted_logits, sorted_idx = logits.sort(dim=-1, descending=True)
+    pos = torch.arange(V, device=dev).unsqueeze(0)
+    keep = (tk <= 0) | (pos < tk)                    # top-k
+    probs = torch.softmax(sorted_logits / tt.clamp_min(1e-6), dim=-1)
+    excl_csum = probs.cumsum(-1) - probs             # exclusive cumsum
+    keep &= (excl_csum < tp) | (pos == 0)            # top-p, rank-0 safe
+    keep |= tt <= 0                                  # greedy rows untouched
+    mask = torch.zeros_like(keep)
+    mask.scatter_(1, sorted_idx, keep)
+    return logits.masked_fill(~mask, float("-inf"))
 
 
 _BUCKETS = [8, 16, 24, 32, 40, 48, 64, 80, 96, 128, 160, 192, 256, 320,

@@ -490,3 +490,58 @@ def test_ref_attention_softmax_sanity():
     p = torch.softmax((ks @ q[b, h]) * scale, dim=0)
     want = (p[:, None] * vs).sum(dim=0)
     assert torch.allclose(out[b, h], want, atol=1e-5)
+
+
+def test_batched_topk_topp_matches_reference():
+    """The vectorized top-k/top-p mask must equal the straightforward
+    per-row reference implementation (which r1 shipped)."""
+    import torch
+
+    from kubeai_amd.engine.runner import _apply_topk_topp
+
+    def ref_mask(logits, top_ps, top_ks, temps):
+        out = logits.clone()
+        sorted_logits, sorted_idx = out.sort(dim=-1, descending=True)
+        for i, (tp, tk, tt) in enumerate(zip(top_ps, top_ks, temps)):
+            if tt <= 0:
+                continue
+            row = sorted_logits[i]
+            keep = torch.ones_like(row, dtype=torch.bool)
+            if tk and tk > 0:
+                keep[tk:] = False
+            if tp < 1.0:
+                probs = torch.softmax(row / tt, dim=-1)
+                csum = probs.cumsum(0)
+                keep &= (csum - probs) < tp
+                keep[0] = True
+            out[i, sorted_idx[i][~keep]] = float("-inf")
+        return out
+
+    torch.manual_seed(0)
+    logits = torch.randn(7, 257)
+    top_ps = [1.0, 0.9, 0.5, 1.0, 0.01, 0.7, 1.0]
+    top_ks = [0, 5, 0, 3, 0, 100000, 1]
+    temps = [1.0, 0.7, 1.3, 0.0, 1.0, 2.0, 0.5]
+    got = _apply_topk_topp(logits, top_ps, top_ks, temps)
+    want = ref_mask(logits, top_ps, top_ks, temps)
+    assert torch.equal(got, want)
+
+
+def test_penalty_counts_released():
+    from kubeai_amd.engine import EngineConfig, LLMEngine, SamplingParams
+
+    eng = LLMEngine(
+        EngineConfig(model="llama-tiny", device="cpu", num_gpu_blocks=64,
+                     max_model_len=256)
+    )
+    eng.add_request(
+        list(range(10, 40)),
+        SamplingParams(max_tokens=4, presence_penalty=0.5,
+                       frequency_penalty=0.2, temperature=0.0),
+        request_id="p1",
+    )
+    for _ in range(50):
+        if not eng.has_work():
+            break
+        eng.step()
+    assert not getattr(eng.runner, "_pen_counts", {}), "counts leaked"

@@ -99,7 +99,6 @@ def _resolve_quant(args, model, use_cuda):
         return "fp8"
     if (
         args.quant == "auto"
-        and args.tp == 1  # fused-fp8 path is validated single-GPU
         and model in ("llama-3-8b", "llama-3-70b", "mixtral-8x7b")
     ):
         # llama: matches the FP8-named baseline configs. mixtral: fused

@@ -30,6 +30,15 @@ void gumbel_sample(torch::Tensor out, torch::Tensor logits,
 void sgmv(torch::Tensor y, torch::Tensor x, torch::Tensor A, torch::Tensor B,
           torch::Tensor idx, double scale);
 void register_chwbl(pybind11::module_& m);
+// one-shot fused all-reduce over xGMI (allreduce.hip)
+pybind11::bytes xgmi_alloc();
+void xgmi_connect(int64_t rank, int64_t world,
+                  const std::vector<pybind11::bytes>& handles);
+void xgmi_fused_allreduce_add_rmsnorm(torch::Tensor x, torch::Tensor residual,
+                                      torch::Tensor weight, double eps);
+int64_t xgmi_error_count();
+int64_t xgmi_max_elems();
+void xgmi_shutdown();
 void rmsnorm_fp8(torch::Tensor out, torch::Tensor out_scale, torch::Tensor x,
                  torch::Tensor weight, double eps);
 void fused_add_rmsnorm_fp8(torch::Tensor out, torch::Tensor out_scale,
@@ -60,6 +69,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "residual add + rmsnorm with fused fp8 row quant");
   m.def("silu_and_mul_fp8", &silu_and_mul_fp8, "SwiGLU with fused fp8 quant");
   m.def("quant_fp8", &quant_fp8, "bf16 -> fp8 row quant");
+  m.def("xgmi_alloc", &xgmi_alloc, "allocate + export the IPC comm buffer");
+  m.def("xgmi_connect", &xgmi_connect, "open peer IPC buffers");
+  m.def("xgmi_fused_allreduce_add_rmsnorm", &xgmi_fused_allreduce_add_rmsnorm,
+        "one-shot xGMI all-reduce fused with residual add + RMSNorm");
+  m.def("xgmi_error_count", &xgmi_error_count, "spin-wait timeout count");
+  m.def("xgmi_max_elems", &xgmi_max_elems, "one-shot capacity (bf16 elems)");
+  m.def("xgmi_shutdown", &xgmi_shutdown, "free/close comm buffers");
   m.def("sgmv", &sgmv, "segmented gather LoRA apply (one adapter segment)");
   register_chwbl(m);
 }

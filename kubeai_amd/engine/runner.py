@@ -105,11 +105,15 @@ class ModelRunner:
         self.block_size = block_size
         self.max_model_len = max_model_len
         self.tp_group = tp_group
+        # TP decode graphs capture the in-graph all-reduce (one-shot xGMI
+        # kernel or RCCL); capture failures are detected on device and
+        # agreed across ranks (_execute_graph), falling back to eager —
+        # so the gate is just the env switch.
         self.enable_graphs = (
             enable_graphs
             and self.device.type == "cuda"
-            and tp_group is None  # RCCL-in-graph: validate before enabling
             and os.environ.get("KUBEAI_GRAPHS", "1") == "1"
+            and (tp_group is None or os.environ.get("KUBEAI_TP_GRAPHS", "1") == "1")
         )
         self._graphs: dict[int, _DecodeGraph] = {}
         torch.manual_seed(seed)
@@ -323,12 +327,28 @@ class ModelRunner:
         bt_max = (self.max_model_len + self.block_size - 1) // self.block_size
         g = self._graphs.get(b_pad)
         if g is None:
+            ok = True
             try:
                 g = _DecodeGraph(self, b_pad, bt_max)
             except Exception:
                 import traceback
 
                 traceback.print_exc()
+                ok = False
+            if self.tp_group is not None and self.tp_group.world > 1:
+                # ranks run in lockstep, so every rank attempts capture on
+                # the same step; all must agree or none may use the graph
+                # (a lone eager rank would mismatch in-graph collectives)
+                import torch.distributed as dist
+
+                flag = torch.tensor(
+                    [1 if ok else 0], dtype=torch.int32, device=self.device
+                )
+                dist.all_reduce(
+                    flag, op=dist.ReduceOp.MIN, group=self.tp_group.group
+                )
+                ok = ok and int(flag.item()) == 1
+            if not ok:
                 self.enable_graphs = False
                 return None
             self._graphs[b_pad] = g

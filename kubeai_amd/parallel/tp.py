@@ -19,6 +19,7 @@ TP=K output must equal TP=1 output (tests/test_tp.py asserts this).
 from __future__ import annotations
 
 import math
+import os
 
 import torch
 import torch.distributed as dist
@@ -36,11 +37,60 @@ class TPGroup:
         self.group = group
         self.rank = dist.get_rank(group)
         self.world = dist.get_world_size(group)
+        # one-shot fused xGMI all-reduce (parallel/comm.py); initialized
+        # lazily on first CUDA tensor — falls back to RCCL when IPC or
+        # the self-check fails
+        self._xgmi = None
+        self._xgmi_tried = False
 
     def all_reduce(self, t: torch.Tensor) -> torch.Tensor:
         if self.world > 1:
             dist.all_reduce(t, group=self.group)
         return t
+
+    def _xgmi_comm(self, t: torch.Tensor):
+        if not self._xgmi_tried:
+            self._xgmi_tried = True
+            if (
+                self.world > 1
+                and t.is_cuda
+                and os.environ.get("KUBEAI_XGMI_ALLREDUCE", "1") == "1"
+            ):
+                try:
+                    from .comm import XgmiAllReduce
+
+                    self._xgmi = XgmiAllReduce(self.group, t.device)
+                except Exception:
+                    import traceback
+
+                    traceback.print_exc()
+                    self._xgmi = None
+        return self._xgmi
+
+    def allreduce_add_rmsnorm(self, x, residual, weight, eps):
+        """(all-reduce partial x) + residual add + RMSNorm, fused over
+        xGMI when the one-shot comm path is up (SURVEY hard part #2:
+        decode-step latency is all-reduce-launch bound at TP=8)."""
+        if self.world > 1:
+            comm = self._xgmi_comm(x)
+            if comm is not None and comm.fits(x):
+                return comm.fused_allreduce_add_rmsnorm(x, residual, weight, eps)
+            dist.all_reduce(x, group=self.group)
+        return ops.fused_add_rmsnorm(x, residual, weight, eps)
+
+    def allreduce_add_rmsnorm_fp8(self, x, residual, weight, eps):
+        """Same, but the normed activation comes back quantized
+        (fp8 + per-row scales) for the fused-fp8 serving path."""
+        if self.world > 1:
+            comm = self._xgmi_comm(x)
+            if comm is not None and comm.fits(x):
+                x, residual = comm.fused_allreduce_add_rmsnorm(
+                    x, residual, weight, eps
+                )
+                return ops.quant_fp8(x), residual
+            dist.all_reduce(x, group=self.group)
+        xq_xs = ops.fused_add_rmsnorm_fp8(x, residual, weight, eps)
+        return xq_xs, residual
 
 
 def _shard_rows(full: torch.Tensor, tp: TPGroup) -> torch.Tensor:
@@ -92,8 +142,15 @@ class TPAttention(nn.Module):
         )
 
     def forward(self, x, fb: ForwardBatch, kv_cache, cos_sin):
-        T = x.shape[0]
-        qkv = self.qkv_proj(x)
+        fp8_in = isinstance(x, tuple)  # (fp8, row scales) from a fused
+        # producer (TPDecoderLayer fp8 path, same as models/llama.py)
+        if fp8_in:
+            xq, xs = x
+            T = xq.shape[0]
+            qkv = self.qkv_proj.forward_quantized(xq, xs)
+        else:
+            T = x.shape[0]
+            qkv = self.qkv_proj(x)
         q, k, v = qkv.split(
             [self.n_q * self.hd, self.n_kv * self.hd, self.n_kv * self.hd], dim=-1
         )
@@ -120,8 +177,11 @@ class TPAttention(nn.Module):
                 fb.prefill_seq_lens, self.scale,
                 out=out[nd:],
             )
-        res = self.o_proj(out.view(T, -1))
-        return self.tp.all_reduce(res)
+        attn_flat = out.view(T, -1)
+        if fp8_in:
+            a8, ascale = ops.quant_fp8(attn_flat)
+            return self.o_proj.forward_quantized(a8, ascale)  # partial sum
+        return self.o_proj(attn_flat)  # partial; layer all-reduces
 
 
 class TPMLP(nn.Module):
@@ -145,8 +205,12 @@ class TPMLP(nn.Module):
         )
 
     def forward(self, x):
+        if isinstance(x, tuple):  # fused fp8 path
+            h = self.gate_up_proj.forward_quantized(*x)
+            a8, ascale = ops.silu_and_mul_fp8(h)
+            return self.down_proj.forward_quantized(a8, ascale)  # partial
         act = ops.silu_and_mul(self.gate_up_proj(x))
-        return self.tp.all_reduce(self.down_proj(act))
+        return self.down_proj(act)  # partial; layer all-reduces
 
 
 class TPDecoderLayer(nn.Module):
@@ -159,16 +223,42 @@ class TPDecoderLayer(nn.Module):
         self.eps = cfg.rms_norm_eps
 
     def forward(self, x, residual, fb, kv_cache, cos_sin):
+        # every TP all-reduce here is paired with the following
+        # add+RMSNorm: TPGroup.allreduce_add_rmsnorm runs the one-shot
+        # fused xGMI kernel when available (hard part #2 — decode-step
+        # all-reduce latency), else RCCL all-reduce + fused_add_rmsnorm.
+        # x arrives PARTIAL (previous layer's down_proj partial sum)
+        # except for the first layer (full embeddings, residual None).
+        tp = self.self_attn.tp
+        fp8 = (
+            getattr(self, "_fp8_fused", False)
+            and fb.lora_ids is None
+            and (x.is_cuda if not isinstance(x, tuple) else True)
+        )
         if residual is None:
             residual = x
-            x = ops.rmsnorm(x, self.input_layernorm, self.eps)
+            if fp8:
+                xin = ops.rmsnorm_fp8(x, self.input_layernorm, self.eps)
+            else:
+                xin = ops.rmsnorm(x, self.input_layernorm, self.eps)
+        elif fp8:
+            xin, residual = tp.allreduce_add_rmsnorm_fp8(
+                x, residual, self.input_layernorm, self.eps
+            )
         else:
-            x, residual = ops.fused_add_rmsnorm(x, residual, self.input_layernorm, self.eps)
-        x = self.self_attn(x, fb, kv_cache, cos_sin)
-        x, residual = ops.fused_add_rmsnorm(
-            x, residual, self.post_attention_layernorm, self.eps
-        )
-        x = self.mlp(x)
+            xin, residual = tp.allreduce_add_rmsnorm(
+                x, residual, self.input_layernorm, self.eps
+            )
+        x = self.self_attn(xin, fb, kv_cache, cos_sin)  # partial out
+        if fp8:
+            xmid, residual = tp.allreduce_add_rmsnorm_fp8(
+                x, residual, self.post_attention_layernorm, self.eps
+            )
+        else:
+            xmid, residual = tp.allreduce_add_rmsnorm(
+                x, residual, self.post_attention_layernorm, self.eps
+            )
+        x = self.mlp(xmid)  # partial out
         return x, residual
 
 
@@ -222,7 +312,11 @@ class TPLlamaForCausalLM(nn.Module):
         residual = None
         for i, layer in enumerate(self.layers):
             x, residual = layer(x, residual, fb, self.kv_caches[i], self.cos_sin)
-        x, _ = ops.fused_add_rmsnorm(x, residual, self.norm, self.cfg.rms_norm_eps)
+        # x is the last layer's PARTIAL mlp output: final fused
+        # all-reduce + add + norm
+        x, _ = self.tp.allreduce_add_rmsnorm(
+            x, residual, self.norm, self.cfg.rms_norm_eps
+        )
         return x
 
     @torch.inference_mode()

@@ -34,7 +34,8 @@ def _generate(eng):
     raise AssertionError("did not finish")
 
 
-def _tp_worker(rank, world, port, q, model="llama-tiny-tp", save_ckpt=None):
+def _tp_worker(rank, world, port, q, model="llama-tiny-tp", save_ckpt=None,
+               quantization=None):
     import torch.distributed as dist
 
     from kubeai_amd.parallel.tp import TPGroup
@@ -54,6 +55,7 @@ def _tp_worker(rank, world, port, q, model="llama-tiny-tp", save_ckpt=None):
                 num_gpu_blocks=64,
                 max_model_len=512,
                 seed=0,
+                quantization=quantization,
             ),
             tp_group=TPGroup(),
         )
@@ -74,12 +76,13 @@ def _tp_worker(rank, world, port, q, model="llama-tiny-tp", save_ckpt=None):
         dist.destroy_process_group()
 
 
-def run_tp(world: int, model="llama-tiny-tp", save_ckpt=None):
+def run_tp(world: int, model="llama-tiny-tp", save_ckpt=None, quantization=None):
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
     port = free_port()
     procs = [
-        ctx.Process(target=_tp_worker, args=(r, world, port, q, model, save_ckpt))
+        ctx.Process(target=_tp_worker,
+                    args=(r, world, port, q, model, save_ckpt, quantization))
         for r in range(world)
     ]
     for p in procs:
@@ -105,3 +108,13 @@ def test_tp2_checkpoint_loading(tmp_path):
     ref = run_tp(1, save_ckpt=ckpt)
     sharded = run_tp(2, model=ckpt)
     assert ref == sharded, f"tp1={ref} tp2(ckpt)={sharded}"
+
+
+def test_tp2_fp8_matches_tp1_fp8():
+    """fp8 serving under TP (r2: the TP gate on fp8 is gone). On CPU the
+    Fp8Linear dequant fallback keeps _scaled_mm semantics, so TP=2 fp8
+    must reproduce TP=1 fp8 greedy output exactly."""
+    t1 = run_tp(1, quantization="fp8")
+    t2 = run_tp(2, quantization="fp8")
+    assert len(t1) == N_TOKENS
+    assert t1 == t2, f"tp1={t1} tp2={t2}"

@@ -116,6 +116,10 @@ class ModelRunner:
             and (tp_group is None or os.environ.get("KUBEAI_TP_GRAPHS", "1") == "1")
         )
         self._graphs: dict[int, _DecodeGraph] = {}
+        # JSON-mode constrained decoding: per-request prefix automata +
+        # a tokenizer the server/worker injects after startup
+        self.tokenizer = None
+        self._json_validators: dict = {}
         torch.manual_seed(seed)
         if tp_group is not None:
             from kubeai_amd.parallel.tp import TPLlamaForCausalLM
@@ -477,11 +481,14 @@ class ModelRunner:
                 tokens = ops.gumbel_sample(masked.contiguous(), t_t, seeds, step).to(dev)
             else:
                 tokens = ops.gumbel_sample(logits.contiguous(), t_t, seeds, step)
+        tokens = tokens.cpu().tolist()
+        if any(r.params.json_mode for r in sample_reqs):
+            tokens = self._constrain_json(sample_reqs, tokens, logits)
+        tokens_t = torch.tensor(tokens, dtype=torch.int64, device=logits.device)
         # logprob of the chosen token: logit - logsumexp(row)
         lse = torch.logsumexp(logits, dim=-1)
-        chosen = logits.gather(1, tokens.view(-1, 1)).squeeze(1)
+        chosen = logits.gather(1, tokens_t.view(-1, 1)).squeeze(1)
         logprobs = (chosen - lse).cpu().tolist()
-        tokens = tokens.cpu().tolist()
         self.last_logprobs = {
             r.request_id: float(lp) for r, lp in zip(sample_reqs, logprobs)
         }
@@ -510,6 +517,83 @@ class ModelRunner:
         return {r.request_id: int(t) for r, t in zip(sample_reqs, tokens)}
 
     # ------------------------------------------------------------------
+    def _constrain_json(self, sample_reqs, tokens: list, logits) -> list:
+        """OpenAI response_format json_object: vet each sampled token's
+        decoded text against a streaming JSON-prefix automaton; mask and
+        resample (argmax over survivors) on violation, hold EOS until the
+        top-level object closes, force-stop once it has. Runs identically
+        on every TP rank (the worker injects the same tokenizer)."""
+        tok = self.tokenizer
+        if tok is None:
+            return tokens  # no tokenizer injected (bare-engine use)
+        from .jsonmode import JsonPrefixValidator
+
+        eos = self.arch.eos_token_id
+        for i, r in enumerate(sample_reqs):
+            if not r.params.json_mode:
+                continue
+            v = self._json_validators.get(r.request_id)
+            if v is None:
+                v = self._json_validators[r.request_id] = JsonPrefixValidator()
+            if v.complete:
+                tokens[i] = eos  # the object closed on a previous step
+                continue
+            tail = r.output_token_ids[-8:]
+            prev = tok.decode(tail)
+            stop_ids = set(r.params.stop_token_ids) | {eos}
+            snap = v.snapshot()
+
+            def ok(t: int) -> bool:
+                if t in stop_ids:
+                    return v.complete
+                piece = tok.decode(tail + [t])[len(prev):]
+                if v.feed(piece):
+                    return True
+                v.restore(snap)
+                return False
+
+            if ok(tokens[i]):
+                continue
+            # sampled token breaks JSON: walk candidates in logit order
+            # (valid tokens can be a tiny vocab fraction, e.g. only '{'
+            # opens a document — plain resampling would almost never hit)
+            V = logits.shape[1]
+            cand = torch.topk(logits[i], min(V, 4096)).indices.cpu().tolist()
+            chosen = None
+            for t in cand:
+                if ok(t):
+                    chosen = t
+                    break
+            if chosen is None:
+                # last resort: tokens whose text starts with an allowed
+                # char (first-char buckets built once per tokenizer)
+                for t in self._json_char_candidates(v, snap, tok):
+                    if ok(t):
+                        chosen = t
+                        break
+            if chosen is not None:
+                tokens[i] = int(chosen)
+            # else: nothing representable continues this JSON — keep the
+            # sampled token (request will end as ordinary text)
+        return tokens
+
+    def _json_char_candidates(self, v, snap, tok):
+        buckets = getattr(self, "_json_first_char", None)
+        if buckets is None:
+            buckets = {}
+            for t in range(self.arch.vocab_size):
+                s = tok.decode([t])
+                if s:
+                    buckets.setdefault(s[0], []).append(t)
+            self._json_first_char = buckets
+        out = []
+        for ch, ids in buckets.items():
+            if v.feed(ch):
+                out.extend(ids)
+            v.restore(snap)
+        return out
+
+    # ------------------------------------------------------------------
     def _pen_counts_for(self, r, vocab: int) -> torch.Tensor:
         """Per-request generated-token counts on device (penalty state)."""
         counts = getattr(self, "_pen_counts", None)
@@ -532,30 +616,69 @@ class ModelRunner:
         """Drop per-request runner state (called when a request ends)."""
         if getattr(self, "_pen_counts", None):
             self._pen_counts.pop(request_id, None)
+        self._json_validators.pop(request_id, None)
 
 
 def _apply_topk_topp(logits, top_ps, top_ks, temps):
     """Mask logits outside the per-row top-k / top-p (nucleus) sets.
 
-    Fully batched on-device (one sort + vectorized masking — no per-row
-    Python loop, r1's CPU round trip). Sampling from the masked logits
-    via Gumbel-argmax is exactly nucleus sampling: probabilities
-    renormalize implicitly. Greedy rows (temp<=0) pass through unmasked.
+    Fully batched on-device, and the common case avoids the full-vocab
+    sort: torch.topk over the leading K=1024 candidates plus a full-row
+    logsumexp gives EXACT nucleus masks whenever the nucleus fits in K
+    (checked per row; the rare uncovered row falls back to a full sort).
+    Sampling from the masked logits via Gumbel-argmax is exactly nucleus
+    sampling: probabilities renormalize implicitly. Greedy rows
+    (temp<=0) pass through unmasked.
     """
     S, V = logits.shape
     dev = logits.device
     tp = torch.tensor(top_ps, device=dev, dtype=torch.float32).unsqueeze(1)
     tk = torch.tensor(top_ks, device=dev, dtype=torch.int64).unsqueeze(1)
     tt = torch.tensor(temps, device=dev, dtype=torch.float32).unsqueeze(1)
-    sorted_logits, sorted_idx = logits.sort(dim=-1, descending=True)
-    pos = torch.arange(V, device=dev).unsqueeze(0)
+    max_k = max((k for k in top_ks if k and k > 0), default=0)
+    K = min(V, max(1024, max_k))
+    if K >= V:
+        vals, idx = logits.sort(dim=-1, descending=True)
+    else:
+        vals, idx = logits.topk(K, dim=-1)
+    pos = torch.arange(K if K < V else V, device=dev).unsqueeze(0)
     keep = (tk <= 0) | (pos < tk)                    # top-k
-    probs = torch.softmax(sorted_logits / tt.clamp_min(1e-6), dim=-1)
-    excl_csum = probs.cumsum(-1) - probs             # exclusive cumsum
-    keep &= (excl_csum < tp) | (pos == 0)            # top-p, rank-0 safe
+    # exact full-distribution probabilities for the leading K entries
+    scaled = logits.float() / tt.clamp_min(1e-6)
+    lse = torch.logsumexp(scaled, dim=-1, keepdim=True)
+    probs = torch.exp(vals.float() / tt.clamp_min(1e-6) - lse)
+    csum = probs.cumsum(-1)
+    keep &= (csum - probs < tp) | (pos == 0)         # top-p, rank-0 safe
     keep |= tt <= 0                                  # greedy rows untouched
-    mask = torch.zeros_like(keep)
-    mask.scatter_(1, sorted_idx, keep)
+    if K < V:
+        # nucleus spilling past K (top-K mass < top_p with no tighter
+        # top-k)? those rows get the exact full-sort treatment
+        uncovered = (
+            (csum[:, -1:] < tp) & ((tk <= 0) | (tk > K)) & (tt > 0)
+        ).squeeze(1)
+        if bool(uncovered.any()):
+            sub_lg = logits[uncovered]
+            sv, si = sub_lg.sort(dim=-1, descending=True)
+            sub_tt = tt[uncovered]
+            sub_tp = tp[uncovered]
+            sub_tk = tk[uncovered]
+            posV = torch.arange(V, device=dev).unsqueeze(0)
+            skeep = (sub_tk <= 0) | (posV < sub_tk)
+            sprobs = torch.softmax(sv.float() / sub_tt.clamp_min(1e-6), -1)
+            scs = sprobs.cumsum(-1)
+            skeep &= (scs - sprobs < sub_tp) | (posV == 0)
+            skeep |= sub_tt <= 0
+            smask = torch.zeros_like(skeep)
+            smask.scatter_(1, si, skeep)
+            out = logits.clone()
+            mask = torch.zeros_like(logits, dtype=torch.bool)
+            mask.scatter_(1, idx, keep)
+            mask[uncovered] = smask
+            mask |= tt <= 0  # greedy rows keep the whole vocab
+            return out.masked_fill_(~mask, float("-inf"))
+    mask = torch.zeros_like(logits, dtype=torch.bool)
+    mask.scatter_(1, idx, keep)
+    mask |= tt <= 0  # greedy rows keep the whole vocab
     return logits.masked_fill(~mask, float("-inf"))
 
 

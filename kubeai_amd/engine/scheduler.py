@@ -37,6 +37,9 @@ class SamplingParams:
     # FIFO within a priority class. Maps the Model CRD's priorityClassName
     # analog onto per-request scheduling.
     priority: int = 0
+    # OpenAI response_format json_object: the sampler enforces that the
+    # output is a valid JSON object (engine/jsonmode.py)
+    json_mode: bool = False
 
 
 class RequestStatus(enum.Enum):

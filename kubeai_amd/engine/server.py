@@ -243,6 +243,8 @@ class EngineServer:
         self.tokenizer = load_tokenizer(
             self.cfg.model, arch.vocab_size, arch.bos_token_id, arch.eos_token_id
         )
+        # JSON-mode constrained decoding needs the tokenizer at sampling
+        self.engine.runner.tokenizer = self.tokenizer
         self._ready.set()
         label = self.served_model_name
         while not self._stop.is_set():
@@ -424,12 +426,26 @@ def build_app(server: EngineServer) -> FastAPI:
             out[tid] = bias
         return out
 
+    def _json_mode_from(body: dict) -> bool:
+        rf = body.get("response_format")
+        if not rf:
+            return False
+        rtype = rf.get("type") if isinstance(rf, dict) else rf
+        if rtype in ("json_object", "json_schema"):
+            # json_schema enforces well-formed JSON (schema-shape guidance
+            # comes from the prompt — docs/parity.md)
+            return True
+        if rtype == "text":
+            return False
+        raise RequestError(f"unsupported response_format type {rtype!r}")
+
     def _params_from(body: dict) -> SamplingParams:
         mt = body.get("max_tokens") or body.get("max_completion_tokens") or 128
         temp = body.get("temperature")
         temp = 1.0 if temp is None else float(temp)
         seed = body.get("seed")
         return SamplingParams(
+            json_mode=_json_mode_from(body),
             max_tokens=int(mt),
             temperature=float(temp),
             top_p=float(body.get("top_p") or 1.0),
@@ -526,16 +542,22 @@ def build_app(server: EngineServer) -> FastAPI:
         lora_id = _resolve_lora(server, body.get("model"))
         if isinstance(lora_id, JSONResponse):
             return lora_id
+        if body.get("suffix"):
+            # insertion/FIM is not supported (vLLM answers the same way
+            # behind the reference's passthrough)
+            raise RequestError("suffix is not supported")
         n = int(body.get("n") or 1)
         stops = _stop_strings(body)
+        echo_text = ""
+        if body.get("echo"):
+            echo_text = (
+                prompt if isinstance(prompt, str) else server.tokenizer.decode(toks)
+            )
         if body.get("stream"):
-            if n > 1:
-                return JSONResponse(
-                    {"error": "stream with n>1 is not supported"}, status_code=400
-                )
             return StreamingResponse(
                 _stream_completion(server, toks, params, name, chat=False,
-                                   lora_id=lora_id, stops=stops),
+                                   lora_id=lora_id, stops=stops, n=n,
+                                   echo_text=echo_text),
                 media_type="text/event-stream",
             )
         want_logprobs = body.get("logprobs") not in (None, False, 0)
@@ -567,7 +589,9 @@ def build_app(server: EngineServer) -> FastAPI:
             text = (
                 cut_text
                 if cut_text is not None
-                else server.tokenizer.decode(_strip_stop(final, params))
+                else server.tokenizer.decode(
+                    _strip_stop(final, params, server.engine.arch.eos_token_id)
+                )
             )
             lp_obj = None
             if want_logprobs:
@@ -582,7 +606,7 @@ def build_app(server: EngineServer) -> FastAPI:
             choices.append(
                 {
                     "index": i,
-                    "text": text,
+                    "text": echo_text + text,
                     "finish_reason": cut_reason or final.finish_reason or "stop",
                     "logprobs": lp_obj,
                 }
@@ -604,20 +628,28 @@ def build_app(server: EngineServer) -> FastAPI:
             return _generate_unsupported()
         body = await request.json()
         params = _params_from(body)
-        toks = apply_chat_template(server.tokenizer, body.get("messages", []))
+        messages = body.get("messages", [])
+        tools = body.get("tools") or []
+        tool_choice = body.get("tool_choice")
+        use_tools = bool(tools) and tool_choice != "none"
+        if use_tools:
+            # template-level function calling: tool schemas go into the
+            # prompt; a forced tool_choice constrains the output to JSON
+            # (the reference gets native tool support via vLLM
+            # passthrough — chat_completions.go:350-515)
+            messages = _with_tool_instructions(messages, tools, tool_choice)
+            if tool_choice == "required" or isinstance(tool_choice, dict):
+                params.json_mode = True
+        toks = apply_chat_template(server.tokenizer, messages)
         lora_id = _resolve_lora(server, body.get("model"))
         if isinstance(lora_id, JSONResponse):
             return lora_id
         n = int(body.get("n") or 1)
         stops = _stop_strings(body)
         if body.get("stream"):
-            if n > 1:
-                return JSONResponse(
-                    {"error": "stream with n>1 is not supported"}, status_code=400
-                )
             return StreamingResponse(
                 _stream_completion(server, toks, params, name, chat=True,
-                                   lora_id=lora_id, stops=stops),
+                                   lora_id=lora_id, stops=stops, n=n),
                 media_type="text/event-stream",
             )
         want_logprobs = bool(body.get("logprobs"))
@@ -634,13 +666,37 @@ def build_app(server: EngineServer) -> FastAPI:
             text = (
                 cut_text
                 if cut_text is not None
-                else server.tokenizer.decode(_strip_stop(final, params))
+                else server.tokenizer.decode(
+                    _strip_stop(final, params, server.engine.arch.eos_token_id)
+                )
             )
-            choice = {
-                "index": i,
-                "message": {"role": "assistant", "content": text},
-                "finish_reason": cut_reason or final.finish_reason or "stop",
-            }
+            tc = _parse_tool_call(text, tools) if use_tools else None
+            if tc is not None:
+                fname, fargs = tc
+                choice = {
+                    "index": i,
+                    "message": {
+                        "role": "assistant",
+                        "content": None,
+                        "tool_calls": [
+                            {
+                                "id": f"call_{uuid.uuid4().hex[:12]}",
+                                "type": "function",
+                                "function": {
+                                    "name": fname,
+                                    "arguments": json.dumps(fargs),
+                                },
+                            }
+                        ],
+                    },
+                    "finish_reason": "tool_calls",
+                }
+            else:
+                choice = {
+                    "index": i,
+                    "message": {"role": "assistant", "content": text},
+                    "finish_reason": cut_reason or final.finish_reason or "stop",
+                }
             if want_logprobs:
                 dec = server.tokenizer.decode
                 choice["logprobs"] = {
@@ -844,11 +900,62 @@ def _decode_wav(data: bytes):
     return audio, sr
 
 
-def _strip_stop(final, params: SamplingParams) -> list[int]:
+def _strip_stop(final, params: SamplingParams, eos_id=None) -> list[int]:
+    # the engine appends the model EOS to its own copy of the params, so
+    # the server-side tuple alone is not enough (JSON mode force-stops
+    # with a bare EOS)
     toks = final.output_token_ids
-    if final.finish_reason == "stop" and toks and toks[-1] in params.stop_token_ids:
+    stops = set(params.stop_token_ids)
+    if eos_id is not None:
+        stops.add(eos_id)
+    if final.finish_reason == "stop" and toks and toks[-1] in stops:
         return toks[:-1]
     return toks
+
+
+def _with_tool_instructions(messages: list, tools: list, tool_choice) -> list:
+    """Prepend a system message describing the available functions and the
+    JSON call convention (template-level tool support)."""
+    specs = []
+    for t in tools:
+        fn = t.get("function", t)
+        specs.append({
+            "name": fn.get("name"),
+            "description": fn.get("description", ""),
+            "parameters": fn.get("parameters", {}),
+        })
+    want = ""
+    if isinstance(tool_choice, dict):
+        forced = (tool_choice.get("function") or {}).get("name")
+        if forced:
+            want = f" You must call the function {forced!r}."
+    elif tool_choice == "required":
+        want = " You must call one of the functions."
+    content = (
+        "You can call functions. Available functions (JSON schemas): "
+        + json.dumps(specs)
+        + '. To call a function, respond ONLY with a JSON object '
+          '{"name": <function name>, "arguments": <arguments object>}.'
+        + want
+    )
+    return [{"role": "system", "content": content}] + list(messages)
+
+
+def _parse_tool_call(text: str, tools: list):
+    """-> (name, arguments) when the output is a function-call JSON."""
+    try:
+        obj = json.loads(text)
+    except Exception:  # noqa: BLE001
+        return None
+    if not isinstance(obj, dict) or not isinstance(obj.get("name"), str):
+        return None
+    names = {
+        (t.get("function", t) or {}).get("name") for t in tools
+    }
+    if obj["name"] not in names:
+        return None
+    args = obj.get("arguments")
+    return obj["name"], (args if isinstance(args, dict) else {})
 
 
 def _resolve_lora(server: EngineServer, model_field: Optional[str]):
@@ -863,79 +970,118 @@ def _resolve_lora(server: EngineServer, model_field: Optional[str]):
     )
 
 
-async def _stream_completion(server, toks, params, name, chat: bool,
-                             lora_id: int = 0, stops=None):
-    rid = f"{'chatcmpl' if chat else 'cmpl'}-{uuid.uuid4().hex[:12]}"
-    created = int(time.time())
-    n_out = 0
-    sd = StreamDecoder(server.tokenizer)
-    sc = _StopScanner(stops) if stops else None
-    emitted = 0  # chars already streamed to the client
-    async for o in server.generate(toks, params, lora_id):
-        # incremental decode (O(new tokens) per step); drop a terminal
-        # stop token's own text, mirroring the non-stream _strip_stop
+class _ChoiceStream:
+    """Per-choice incremental state for streaming responses."""
+
+    def __init__(self, server, params, stops, echo_text: str = ""):
+        self.sd = StreamDecoder(server.tokenizer)
+        self.sc = _StopScanner(stops) if stops else None
+        self.stops = stops
+        self.params = params
+        self.server = server
+        self.emitted = 0
+        self.n_out = 0
+        self.prompt_tokens = 0
+        self.echo_text = echo_text  # flushed with the first piece
+
+    def step(self, o):
+        """-> (piece, finished, finish_reason)."""
         new_ids = o.new_token_ids
         if (
             o.finished
             and o.finish_reason == "stop"
             and new_ids
-            and new_ids[-1] in params.stop_token_ids
+            and new_ids[-1]
+            in set(self.params.stop_token_ids)
+            | {self.server.engine.arch.eos_token_id}
         ):
             new_ids = new_ids[:-1]
-        text = sd.push(new_ids)
-        if stops:
-            hit = sc.scan(text)
+        text = self.sd.push(new_ids)
+        finished, reason = o.finished, o.finish_reason
+        if self.stops:
+            hit = self.sc.scan(text)
             if hit >= 0:
                 text = text[:hit]
-                o.finished = True
-                o.finish_reason = "stop"
-        if o.finished:
+                finished, reason = True, "stop"
+        if finished:
             stable = text
         else:
             # hold back any tail that could still complete a stop string
             # (never leak a stop's leading chars across chunk boundaries)
             stable = (
-                text[: len(text) - _holdback_len(text, stops)] if stops else text
+                text[: len(text) - _holdback_len(text, self.stops)]
+                if self.stops
+                else text
             )
-        piece = stable[emitted:] if len(stable) > emitted else ""
-        emitted = max(emitted, len(stable))
-        n_out = len(o.output_token_ids)
-        if chat:
+        piece = stable[self.emitted:] if len(stable) > self.emitted else ""
+        self.emitted = max(self.emitted, len(stable))
+        if self.echo_text:
+            piece = self.echo_text + piece
+            self.echo_text = ""
+        self.n_out = len(o.output_token_ids)
+        self.prompt_tokens = o.num_prompt_tokens
+        return piece, finished, reason
+
+
+async def _stream_completion(server, toks, params, name, chat: bool,
+                             lora_id: int = 0, stops=None, n: int = 1,
+                             echo_text: str = ""):
+    """SSE stream; n>1 runs n engine requests concurrently, interleaving
+    chunks with their choice index (the reference gets this via vLLM
+    passthrough; r1 rejected stream+n>1)."""
+    rid = f"{'chatcmpl' if chat else 'cmpl'}-{uuid.uuid4().hex[:12]}"
+    created = int(time.time())
+    agg: asyncio.Queue = asyncio.Queue()
+
+    async def pump(i: int):
+        async for o in server.generate(toks, params, lora_id):
+            await agg.put((i, o))
+
+    tasks = [asyncio.create_task(pump(i)) for i in range(n)]
+    states = [_ChoiceStream(server, params, stops, echo_text) for _ in range(n)]
+    live = set(range(n))
+    try:
+        while live:
+            i, o = await agg.get()
+            if i not in live:
+                continue
+            st = states[i]
+            piece, finished, reason = st.step(o)
+            if finished:
+                live.discard(i)
+                tasks[i].cancel()
+            if chat:
+                choice = {
+                    "index": i,
+                    "delta": {"content": piece} if piece else {},
+                    "finish_reason": reason if finished else None,
+                }
+            else:
+                choice = {
+                    "index": i,
+                    "text": piece,
+                    "finish_reason": reason if finished else None,
+                }
             chunk = {
                 "id": rid,
-                "object": "chat.completion.chunk",
+                "object": "chat.completion.chunk" if chat else "text_completion",
                 "created": created,
                 "model": name,
-                "choices": [
-                    {
-                        "index": 0,
-                        "delta": {"content": piece} if piece else {},
-                        "finish_reason": o.finish_reason if o.finished else None,
-                    }
-                ],
+                "choices": [choice],
             }
-        else:
-            chunk = {
-                "id": rid,
-                "object": "text_completion",
-                "created": created,
-                "model": name,
-                "choices": [
-                    {
-                        "index": 0,
-                        "text": piece,
-                        "finish_reason": o.finish_reason if o.finished else None,
-                    }
-                ],
-            }
-        if o.finished:
-            chunk["usage"] = {
-                "prompt_tokens": o.num_prompt_tokens,
-                "completion_tokens": n_out,
-                "total_tokens": o.num_prompt_tokens + n_out,
-            }
-        yield f"data: {json.dumps(chunk)}\n\n"
-    yield "data: [DONE]\n\n"
+            if not live:  # final chunk carries aggregate usage
+                pt = states[0].prompt_tokens
+                ct = sum(s.n_out for s in states)
+                chunk["usage"] = {
+                    "prompt_tokens": pt,
+                    "completion_tokens": ct,
+                    "total_tokens": pt + ct,
+                }
+            yield f"data: {json.dumps(chunk)}\n\n"
+        yield "data: [DONE]\n\n"
+    finally:
+        for t in tasks:
+            t.cancel()
 
 
 # --------------------------------------------------------------------------
@@ -957,6 +1103,11 @@ def _tp_worker_main(rank: int, world: int, port: int, cfg: EngineConfig) -> None
         world_size=world,
     )
     engine = LLMEngine(cfg, tp_group=TPGroup())
+    # identical tokenizer => identical JSON-mode sampling on every rank
+    engine.runner.tokenizer = load_tokenizer(
+        cfg.model, engine.arch.vocab_size, engine.arch.bos_token_id,
+        engine.arch.eos_token_id,
+    )
     while True:
         box = [None]
         dist.broadcast_object_list(box, src=0)

@@ -13,12 +13,24 @@ import zlib
 
 
 class SyntheticTokenizer:
+    """Word-hash tokenizer plus a printable-ASCII char region (ids 3..97
+    decode to chr(32..126)) so character-level features — JSON-mode
+    constrained decoding, stop strings with punctuation — are exercisable
+    without a real tokenizer file."""
+
+    CHAR_LO, CHAR_HI = 3, 97  # id -> chr(id + 29): ' ' .. '~'
+
     def __init__(self, vocab_size: int, bos_token_id: int = 1, eos_token_id: int = 2):
         self.vocab_size = vocab_size
         self.bos_token_id = bos_token_id
         self.eos_token_id = eos_token_id
-        # reserve low ids for specials
-        self._lo = 16
+        # word-hash ids start above the specials + char region
+        self._lo = 100
+
+    def char_token(self, ch: str) -> int:
+        o = ord(ch)
+        assert 32 <= o <= 126, ch
+        return o - 29
 
     def encode(self, text: str, add_bos: bool = False) -> list[int]:
         toks: list[int] = [self.bos_token_id] if add_bos else []
@@ -28,7 +40,15 @@ class SyntheticTokenizer:
         return toks
 
     def decode(self, token_ids: list[int]) -> str:
-        return " ".join(f"t{t}" for t in token_ids)
+        out: list[str] = []
+        for t in token_ids:
+            if self.CHAR_LO <= t <= self.CHAR_HI:
+                out.append(chr(t + 29))
+            else:
+                if out:
+                    out.append(" ")
+                out.append(f"t{t}")
+        return "".join(out)
 
 
 class HFTokenizer:

@@ -22,6 +22,21 @@ sources = sorted(
     glob.glob(os.path.join(CSRC, "*.cpp")) + glob.glob(os.path.join(CSRC, "*.hip"))
 )
 
+# torch's hipify/ninja pipeline tracks no header dependencies for .hip
+# objects: a common.h edit would silently ship stale kernels. Touch every
+# .hip newer than its headers so the pipeline rebuilds them.
+_hdr_mtime = max(
+    (os.path.getmtime(h) for h in glob.glob(os.path.join(CSRC, "*.h"))),
+    default=0.0,
+)
+for _src in list(sources):
+    if _src.endswith(".hip") and os.path.getmtime(_src) < _hdr_mtime:
+        os.utime(_src, None)
+        _gen = _src[:-4] + "_hip.hip"
+        if os.path.exists(_gen):
+            os.remove(_gen)
+            sources = [s for s in sources if s != _gen]
+
 setup(
     name="kubeai_amd_C",
     ext_modules=[

@@ -9,6 +9,7 @@
 
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
+#include <hip/hip_fp16.h>
 #include <hip/hip_fp8.h>
 #include <stdint.h>
 #include <c10/hip/HIPStream.h>
@@ -79,26 +80,26 @@ DEV_INLINE float e5m2_to_f32(unsigned char b) {
 // (v_cvt_pk_f32_bf8: one instruction turns 2 e5m2 bytes into 2 floats)
 typedef float floatx2 __attribute__((ext_vector_type(2)));
 
-// 8 packed e5m2 bytes -> 8 bf16 values (one staging unit). Uses the
-// native packed convert (2 bytes/instr); every e5m2 value — subnormals,
-// inf, nan included — is exactly representable in bf16, so the f32
-// result truncates to bf16 losslessly (plain >>16, no rounding step).
+// 8 packed e5m2 bytes -> 8 bf16 values (one staging unit).
+//
+// Route: e5m2 is bit-identical to the TOP BYTE of an IEEE fp16, so
+// fp16 = byte<<8 (pure integer move); fp16 -> f32 is one hardware cvt;
+// f32 -> bf16 truncates losslessly (every e5m2 value, subnormals and
+// inf/nan included, is exactly representable in bf16).
+//
+// Deliberately NOT via __builtin_amdgcn_cvt_pk_f32_bf8 + bit-level
+// truncation: this toolchain mis-simplifies bit manipulation of that
+// intrinsic's result (low mantissa bit folded to zero — 129/256 byte
+// values corrupt, device-verified in scripts/dbg_cvt.hip; an opaque
+// asm barrier does not stop it). Arithmetic (fadd/fma) consumers of
+// cvt_pk_f32_bf8 are fine — the decode kernel keeps using it.
 DEV_INLINE ushort8 e5m2x8_to_bf16x8(uint64_t raw) {
   ushort8 out;
-  const uint32_t w[2] = {(uint32_t)raw, (uint32_t)(raw >> 32)};
 #pragma unroll
-  for (int i = 0; i < 2; ++i) {
-    floatx2 lo = __builtin_amdgcn_cvt_pk_f32_bf8(w[i], false);
-    floatx2 hi = __builtin_amdgcn_cvt_pk_f32_bf8(w[i], true);
-    // opaque barrier: this compiler's value tracking mis-models the
-    // packed convert's low mantissa bit and would fold it out of the
-    // bit-level truncation below (device-verified, scripts/dbg_cvt.hip
-    // — arithmetic consumers like the decode kernel are unaffected)
-    asm("" : "+v"(lo), "+v"(hi));
-    out[4 * i + 0] = (ushort)(__builtin_bit_cast(uint32_t, lo.x) >> 16);
-    out[4 * i + 1] = (ushort)(__builtin_bit_cast(uint32_t, lo.y) >> 16);
-    out[4 * i + 2] = (ushort)(__builtin_bit_cast(uint32_t, hi.x) >> 16);
-    out[4 * i + 3] = (ushort)(__builtin_bit_cast(uint32_t, hi.y) >> 16);
+  for (int j = 0; j < 8; ++j) {
+    const ushort h = (ushort)((raw >> (8 * j)) << 8);
+    const float f = __half2float(__builtin_bit_cast(__half, h));
+    out[j] = (ushort)(__builtin_bit_cast(uint32_t, f) >> 16);
   }
   return out;
 }

@@ -274,6 +274,13 @@ __launch_bounds__(G * WAVE_SIZE) __global__ void paged_prefill_kernel(
 
 }  // namespace
 
+// attention_prefill_v2.hip; returns false when the shape is unsupported
+bool paged_attention_prefill_v2(torch::Tensor out, torch::Tensor q,
+                                torch::Tensor k_cache, torch::Tensor v_cache,
+                                torch::Tensor block_tables,
+                                torch::Tensor query_start_loc,
+                                torch::Tensor seq_lens, double scale);
+
 void paged_attention_prefill(torch::Tensor out, torch::Tensor q,
                              torch::Tensor k_cache, torch::Tensor v_cache,
                              torch::Tensor block_tables,
@@ -297,6 +304,16 @@ void paged_attention_prefill(torch::Tensor out, torch::Tensor q,
   const int G = n_q / n_kv;
   TORCH_CHECK(n_q % n_kv == 0);
   if (q.size(0) == 0) return;
+  // v2: 8-wave 32x32-MFMA ladder (attention_prefill_v2.hip) — handles
+  // G in {1,2,4,8}; opt out with KUBEAI_PREFILL_V2=0
+  static const bool use_v2 = []() {
+    const char* e = getenv("KUBEAI_PREFILL_V2");
+    return e == nullptr || e[0] != '0';
+  }();
+  if (use_v2 &&
+      paged_attention_prefill_v2(out, q, k_cache, v_cache, block_tables,
+                                 query_start_loc, seq_lens, scale))
+    return;
   const int Tq = q.size(0);
   const int n_qtiles_max = (Tq + kQB - 1) / kQB;  // per-seq early exit
   dim3 grid(B, n_kv, n_qtiles_max);

@@ -1,0 +1,49 @@
+"""Sampling-path microbench: where does the temperature overhead go?
+
+Times the top-k/top-p mask (fast topk path vs full sort), the Gumbel
+kernel, and greedy argmax at the bench's steady-state shape.
+"""
+import os
+import sys
+
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+import kubeai_amd.ops as ops
+from kubeai_amd.engine.runner import _apply_topk_topp
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+from bench_ops import timeit  # noqa: E402
+
+dev = "cuda"
+torch.manual_seed(0)
+for S, V in ((40, 128256), (256, 128256)):
+    logits = torch.randn(S, V, dtype=torch.float32, device=dev)
+    temps = [0.8] * S
+    tps = [0.9] * S
+    tks = [0] * S
+    t_t = torch.full((S,), 0.8, device=dev)
+    seeds = torch.arange(S, dtype=torch.int64, device=dev)
+
+    us_mask = timeit(lambda: _apply_topk_topp(logits, tps, tks, temps))
+
+    def full_sort():
+        sl, si = logits.sort(dim=-1, descending=True)
+        probs = torch.softmax(sl / 0.8, dim=-1)
+        cs = probs.cumsum(-1)
+        keep = (cs - probs) < 0.9
+        keep[:, 0] = True
+        mask = torch.zeros_like(keep)
+        mask.scatter_(1, si, keep)
+        return logits.masked_fill(~mask, float("-inf"))
+
+    us_sort = timeit(full_sort)
+    masked = _apply_topk_topp(logits, tps, tks, temps).contiguous()
+    us_gumbel = timeit(lambda: ops.gumbel_sample(masked, t_t, seeds, 3))
+    us_greedy = timeit(lambda: ops.greedy_sample(logits))
+    us_topk = timeit(lambda: torch.topk(logits, 1024, dim=-1))
+    us_lse = timeit(lambda: torch.logsumexp(logits / 0.8, dim=-1))
+    print(f"S={S:4d} V={V}: mask(fast)={us_mask:8.1f}us  mask(sort)="
+          f"{us_sort:8.1f}us  topk1024={us_topk:8.1f}us  lse={us_lse:7.1f}us"
+          f"  gumbel={us_gumbel:7.1f}us  greedy={us_greedy:7.1f}us",
+          flush=True)

@@ -130,13 +130,27 @@ class FileBroker:
 
 
 def broker_from_url(url: str):
-    """gocloud-style driver selection: mem://topic-prefix | file:///dir."""
+    """gocloud-style driver selection (reference registers its drivers at
+    internal/manager/run.go:47-53): mem:// | file:///dir | awssqs://host/
+    account (https, SigV4 from env creds) | sqs+http://host:port (dev)."""
     u = urlparse(url)
     if u.scheme in ("", "mem"):
         return MemBroker()
     if u.scheme == "file":
         return FileBroker(u.path or "/tmp/kubeai-msgs")
-    raise ValueError(f"unknown messenger driver: {u.scheme}:// (have mem, file)")
+    if u.scheme in ("awssqs", "sqs", "sqs+http"):
+        from .sqs import SqsBroker
+
+        proto = "http" if u.scheme == "sqs+http" else "https"
+        # last path segment is the queue/topic; the rest is the base URL
+        base_path = u.path.rsplit("/", 1)[0] if u.path.strip("/") else ""
+        region = "us-east-1"
+        if u.hostname and u.hostname.startswith("sqs."):
+            region = u.hostname.split(".")[1]
+        return SqsBroker(f"{proto}://{u.netloc}{base_path}", region=region)
+    raise ValueError(
+        f"unknown messenger driver: {u.scheme}:// (have mem, file, awssqs, sqs+http)"
+    )
 
 
 def stream_transport(requests_url: str, responses_url: str):
@@ -150,6 +164,14 @@ def stream_transport(requests_url: str, responses_url: str):
             FileBroker(root),
             os.path.basename(u.path),
             os.path.basename(urlparse(responses_url).path),
+        )
+    if u.scheme in ("awssqs", "sqs", "sqs+http"):
+        # queue name = last path segment; both queues share the base URL
+        broker = broker_from_url(requests_url)
+        return (
+            broker,
+            u.path.rstrip("/").rsplit("/", 1)[-1],
+            urlparse(responses_url).path.rstrip("/").rsplit("/", 1)[-1],
         )
     broker = broker_from_url(requests_url)
     return (

@@ -470,15 +470,15 @@ class ModelRunner:
                 device=dev,
             )
             if needs_topk:
-                # mask the distribution on-device, then draw with the
-                # deterministic Gumbel kernel (no CPU round trip)
-                masked = _apply_topk_topp(
+                # draw from the nucleus over the top-K candidate subset —
+                # no full-vocab sort or mask materialization
+                tokens = _sample_topk_topp(
                     logits,
                     [r.params.top_p for r in sample_reqs],
                     [r.params.top_k for r in sample_reqs],
                     [r.params.temperature for r in sample_reqs],
-                )
-                tokens = ops.gumbel_sample(masked.contiguous(), t_t, seeds, step).to(dev)
+                    t_t, seeds, step,
+                ).to(dev)
             else:
                 tokens = ops.gumbel_sample(logits.contiguous(), t_t, seeds, step)
         tokens = tokens.cpu().tolist()
@@ -617,6 +617,45 @@ class ModelRunner:
         if getattr(self, "_pen_counts", None):
             self._pen_counts.pop(request_id, None)
         self._json_validators.pop(request_id, None)
+
+
+def _sample_topk_topp(logits, top_ps, top_ks, temps, t_t, seeds, step):
+    """Nucleus/top-k sampling over the top-K candidate subset.
+
+    topk(K=1024) + one full-row logsumexp gives EXACT full-distribution
+    probabilities for the leading candidates, the per-row nucleus mask is
+    built on the small [S, K] tensor, and the Gumbel draw runs on that
+    subset (argmax over a masked subset == sampling the renormalized
+    nucleus; greedy rows reduce to position 0 == the global argmax).
+    ~5x cheaper than sorting and masking the full 128k vocab per step.
+    The rare row whose nucleus spills past K falls back to the exact
+    full-sort mask path.
+    """
+    S, V = logits.shape
+    dev = logits.device
+    max_k = max((k for k in top_ks if k and k > 0), default=0)
+    K = min(V, max(1024, max_k))
+    if K >= V:
+        masked = _apply_topk_topp(logits, top_ps, top_ks, temps)
+        return ops.gumbel_sample(masked.contiguous(), t_t, seeds, step)
+    tp = torch.tensor(top_ps, device=dev, dtype=torch.float32).unsqueeze(1)
+    tk = torch.tensor(top_ks, device=dev, dtype=torch.int64).unsqueeze(1)
+    tt = torch.tensor(temps, device=dev, dtype=torch.float32).unsqueeze(1)
+    vals, idx = logits.topk(K, dim=-1)  # sorted descending
+    ttc = tt.clamp_min(1e-6)
+    lse = torch.logsumexp(logits.float() / ttc, dim=-1, keepdim=True)
+    probs = torch.exp(vals.float() / ttc - lse)
+    csum = probs.cumsum(-1)
+    pos = torch.arange(K, device=dev).unsqueeze(0)
+    keep = (tk <= 0) | (pos < tk)
+    keep &= (csum - probs < tp) | (pos == 0)
+    uncovered = ((csum[:, -1:] < tp) & ((tk <= 0) | (tk > K)) & (tt > 0)).any()
+    if bool(uncovered):
+        masked = _apply_topk_topp(logits, top_ps, top_ks, temps)
+        return ops.gumbel_sample(masked.contiguous(), t_t, seeds, step)
+    sub = vals.masked_fill(~keep, float("-inf")).contiguous()
+    sub_pos = ops.gumbel_sample(sub, t_t, seeds, step)
+    return idx.gather(1, sub_pos.view(-1, 1).to(dev)).squeeze(1)
 
 
 def _apply_topk_topp(logits, top_ps, top_ks, temps):

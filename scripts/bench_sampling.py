@@ -10,7 +10,7 @@ import torch
 
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import kubeai_amd.ops as ops
-from kubeai_amd.engine.runner import _apply_topk_topp
+from kubeai_amd.engine.runner import _apply_topk_topp, _sample_topk_topp
 
 sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 from bench_ops import timeit  # noqa: E402
@@ -26,6 +26,9 @@ for S, V in ((40, 128256), (256, 128256)):
     seeds = torch.arange(S, dtype=torch.int64, device=dev)
 
     us_mask = timeit(lambda: _apply_topk_topp(logits, tps, tks, temps))
+    us_subset = timeit(
+        lambda: _sample_topk_topp(logits, tps, tks, temps, t_t, seeds, 3)
+    )
 
     def full_sort():
         sl, si = logits.sort(dim=-1, descending=True)
@@ -43,7 +46,7 @@ for S, V in ((40, 128256), (256, 128256)):
     us_greedy = timeit(lambda: ops.greedy_sample(logits))
     us_topk = timeit(lambda: torch.topk(logits, 1024, dim=-1))
     us_lse = timeit(lambda: torch.logsumexp(logits / 0.8, dim=-1))
-    print(f"S={S:4d} V={V}: mask(fast)={us_mask:8.1f}us  mask(sort)="
+    print(f"S={S:4d} V={V}: SAMPLE(subset)={us_subset:8.1f}us  mask(fast)={us_mask:8.1f}us  mask(sort)="
           f"{us_sort:8.1f}us  topk1024={us_topk:8.1f}us  lse={us_lse:7.1f}us"
           f"  gumbel={us_gumbel:7.1f}us  greedy={us_greedy:7.1f}us",
           flush=True)

@@ -30,12 +30,19 @@ _hdr_mtime = max(
     default=0.0,
 )
 for _src in list(sources):
-    if _src.endswith(".hip") and os.path.getmtime(_src) < _hdr_mtime:
+    if (
+        _src.endswith(".hip")
+        and not _src.endswith("_hip.hip")
+        and os.path.exists(_src)
+        and os.path.getmtime(_src) < _hdr_mtime
+    ):
         os.utime(_src, None)
         _gen = _src[:-4] + "_hip.hip"
         if os.path.exists(_gen):
             os.remove(_gen)
             sources = [s for s in sources if s != _gen]
+# drop hipify artifacts whose base source was deleted or touched
+sources = [s for s in sources if os.path.exists(s)]
 
 setup(
     name="kubeai_amd_C",

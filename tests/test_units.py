@@ -265,8 +265,10 @@ def test_file_broker_roundtrip(tmp_path):
     assert type(broker_from_url(f"file://{tmp_path}/q2")).__name__ == "FileBroker"
     import pytest
 
+    # sqs is a real driver now (kubeai_amd/controlplane/sqs.py)
+    assert type(broker_from_url("sqs://host/acct/queue")).__name__ == "SqsBroker"
     with pytest.raises(ValueError, match="unknown messenger driver"):
-        broker_from_url("sqs://queue")
+        broker_from_url("kafka://queue")
 
 
 def test_file_broker_exactly_once_across_consumers(tmp_path):
@@ -585,3 +587,27 @@ def test_topk_topp_fast_path_matches_full_sort():
     # compare kept sets (float error in cumsum near the p boundary could
     # differ by the boundary token; demand exact match here)
     assert torch.equal(got.isinf(), want.isinf())
+
+
+def test_subset_sampling_stays_in_nucleus():
+    """_sample_topk_topp tokens must always lie in the exact nucleus set
+    and greedy rows must return the global argmax (incl. spill rows)."""
+    import torch
+
+    from kubeai_amd.engine.runner import _apply_topk_topp, _sample_topk_topp
+
+    torch.manual_seed(2)
+    V = 4096
+    logits = torch.randn(4, V) * 4.0
+    logits[2] = torch.randn(V) * 0.01  # nucleus spills past K=1024
+    top_ps = [0.9, 0.8, 0.95, 1.0]
+    top_ks = [0, 50, 0, 0]
+    temps = [1.0, 0.7, 1.0, 0.0]
+    allowed = ~_apply_topk_topp(logits, top_ps, top_ks, temps).isinf()
+    t_t = torch.tensor(temps)
+    for step in range(20):
+        seeds = torch.arange(4, dtype=torch.int64) + step
+        toks = _sample_topk_topp(logits, top_ps, top_ks, temps, t_t, seeds, step)
+        for i, t in enumerate(toks.tolist()):
+            assert allowed[i, t], (i, t, step)
+        assert toks[3].item() == int(logits[3].argmax())

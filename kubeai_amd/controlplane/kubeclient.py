@@ -284,6 +284,61 @@ def model_from_manifest(obj: dict):
 
 ENGINE_PORT = 8000
 
+# per-scheme credential secret names (reference config SecretNames,
+# internal/config/system.go:148-153)
+SECRET_NAMES = {"aws": "aws", "gcp": "gcp", "alibaba": "alibaba",
+                "huggingface": "huggingface"}
+
+
+def _secret_env(name: str, secret: str, key: str) -> dict:
+    return {
+        "name": name,
+        "valueFrom": {"secretKeyRef": {"name": secret, "key": key,
+                                       "optional": True}},
+    }
+
+
+def source_pod_additions(url: str) -> tuple[list, list, list]:
+    """(env, volumes, volume_mounts) for a Model source URL — the
+    credential/volume wiring the reference applies per scheme
+    (internal/modelcontroller/model_source.go:82-227)."""
+    env: list = []
+    vols: list = []
+    mounts: list = []
+    if url.startswith("hf://"):
+        env.append(_secret_env("HF_TOKEN", SECRET_NAMES["huggingface"],
+                               "token"))
+    elif url.startswith("s3://"):
+        env.append(_secret_env("AWS_ACCESS_KEY_ID", SECRET_NAMES["aws"],
+                               "accessKeyID"))
+        env.append(_secret_env("AWS_SECRET_ACCESS_KEY", SECRET_NAMES["aws"],
+                               "secretAccessKey"))
+    elif url.startswith("gs://"):
+        env.append({"name": "GOOGLE_APPLICATION_CREDENTIALS",
+                    "value": "/secrets/gcp/credentials.json"})
+        vols.append({"name": "gcp-credentials",
+                     "secret": {"secretName": SECRET_NAMES["gcp"],
+                                "optional": True}})
+        mounts.append({"name": "gcp-credentials",
+                       "mountPath": "/secrets/gcp", "readOnly": True})
+    elif url.startswith("oss://"):
+        env.append(_secret_env("OSS_ACCESS_KEY_ID", SECRET_NAMES["alibaba"],
+                               "accessKeyID"))
+        env.append(_secret_env("OSS_ACCESS_KEY_SECRET",
+                               SECRET_NAMES["alibaba"], "accessKeySecret"))
+    elif url.startswith("pvc://"):
+        rest = url[len("pvc://"):]
+        claim, _, sub = rest.partition("/")
+        vol: dict = {"name": "model-pvc",
+                     "persistentVolumeClaim": {"claimName": claim}}
+        mount: dict = {"name": "model-pvc", "mountPath": "/model",
+                       "readOnly": True}
+        if sub:
+            mount["subPath"] = sub
+        vols.append(vol)
+        mounts.append(mount)
+    return env, vols, mounts
+
 
 def pod_manifest_for(model, name: str, spec_hash: str, n_gpus: int,
                      image: str, namespace: str,
@@ -296,6 +351,8 @@ def pod_manifest_for(model, name: str, spec_hash: str, n_gpus: int,
         "--port", str(ENGINE_PORT),
     ] + list(model.spec.args)
     env = [{"name": k, "value": v} for k, v in sorted(model.spec.env.items())]
+    src_env, src_vols, src_mounts = source_pod_additions(model.spec.url)
+    env += src_env
     resources = {}
     if n_gpus > 0:
         resources = {
@@ -334,8 +391,10 @@ def pod_manifest_for(model, name: str, spec_hash: str, n_gpus: int,
                     "livenessProbe": {
                         **probe, "periodSeconds": 10, "failureThreshold": 3,
                     },
+                    **({"volumeMounts": src_mounts} if src_mounts else {}),
                 }
             ],
+            **({"volumes": src_vols} if src_vols else {}),
             "restartPolicy": "Never",
         },
     }

@@ -262,3 +262,38 @@ def test_qwen2_engine_gpu():
     assert len(oa.output_token_ids) == 8
     assert oa.output_token_ids == ob.output_token_ids
     assert a.runner.model.layers[0].self_attn.qkv_proj.bias is not None
+
+
+def test_hf_checkpoint_dir_e2e_gpu(tmp_path):
+    """Real-weights e2e on device: an HF-format safetensors dir loads
+    through the engine (cache-dir layout) and reproduces the greedy
+    generation of the in-memory model it was saved from."""
+    import torch
+
+    from kubeai_amd.engine import EngineConfig, LLMEngine, SamplingParams
+    from kubeai_amd.models.loader import save_hf_checkpoint
+
+    def gen(eng, rid):
+        eng.add_request(list(range(10, 200)),
+                        SamplingParams(max_tokens=12), request_id=rid)
+        for _ in range(200):
+            if not eng.has_work():
+                break
+            for o in eng.step():
+                if o.finished:
+                    return o.output_token_ids
+        raise AssertionError("did not finish")
+
+    src = LLMEngine(EngineConfig(model="llama-tiny", device="cuda",
+                                 num_gpu_blocks=256, max_model_len=512,
+                                 seed=11))
+    ckpt = str(tmp_path / "hf-model")
+    save_hf_checkpoint(src.runner.model, ckpt)
+    want = gen(src, "src")
+    del src
+    torch.cuda.empty_cache()
+    dst = LLMEngine(EngineConfig(model=ckpt, device="cuda",
+                                 num_gpu_blocks=256, max_model_len=512,
+                                 seed=999))
+    got = gen(dst, "dst")
+    assert got == want

@@ -376,3 +376,38 @@ def test_full_manager_kube_mode_inference(api):
             await backend.stop()
 
     asyncio.run(main())
+
+
+def test_source_pod_additions():
+    """Per-scheme credential/volume wiring on engine pods (reference
+    model_source.go:82-227)."""
+    from kubeai_amd.controlplane.kubeclient import (pod_manifest_for,
+                                                    source_pod_additions)
+
+    env, vols, mounts = source_pod_additions("s3://bucket/model")
+    names = {e["name"] for e in env}
+    assert names == {"AWS_ACCESS_KEY_ID", "AWS_SECRET_ACCESS_KEY"}
+    assert env[0]["valueFrom"]["secretKeyRef"]["name"] == "aws"
+
+    env, vols, mounts = source_pod_additions("gs://bucket/model")
+    assert env[0]["name"] == "GOOGLE_APPLICATION_CREDENTIALS"
+    assert vols and mounts
+
+    env, vols, mounts = source_pod_additions("oss://bucket/model")
+    assert {e["name"] for e in env} == {"OSS_ACCESS_KEY_ID",
+                                        "OSS_ACCESS_KEY_SECRET"}
+
+    env, vols, mounts = source_pod_additions("hf://org/model")
+    assert env[0]["name"] == "HF_TOKEN"
+
+    env, vols, mounts = source_pod_additions("pvc://my-claim/sub/dir")
+    assert vols[0]["persistentVolumeClaim"]["claimName"] == "my-claim"
+    assert mounts[0]["mountPath"] == "/model"
+    assert mounts[0]["subPath"] == "sub/dir"
+
+    m = Model(name="s", spec=ModelSpec(url="pvc://claim/p"))
+    pod = pod_manifest_for(m, "p1", "h", 1, "img", "ns")
+    assert pod["spec"]["volumes"][0]["persistentVolumeClaim"][
+        "claimName"] == "claim"
+    assert pod["spec"]["containers"][0]["volumeMounts"][0][
+        "mountPath"] == "/model"

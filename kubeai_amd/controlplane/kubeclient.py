@@ -340,17 +340,77 @@ def source_pod_additions(url: str) -> tuple[list, list, list]:
     return env, vols, mounts
 
 
+def _engine_container(model, image: str) -> dict:
+    """Per-engine command/env contract (reference engine builders:
+    engine_vllm.go:86, engine_ollama.go:22-35, engine_fasterwhisper.go:29,
+    engine_infinity.go:31-54). The in-house KubeAIEngine speaks the same
+    vLLM-compatible CLI surface."""
+    spec = model.spec
+    eng = spec.engine
+    ref = spec.url.split("://", 1)[-1]
+    if eng in ("KubeAIEngine", "VLLM"):
+        cmd = (
+            ["python", "-m", "kubeai_amd.engine.server"]
+            if eng == "KubeAIEngine"
+            else ["python3", "-m", "vllm.entrypoints.openai.api_server"]
+        )
+        model_arg = "/model" if spec.url.startswith("pvc://") else spec.url
+        if eng == "VLLM":
+            model_arg = "/model" if spec.url.startswith("pvc://") else ref
+        return {
+            "command": cmd,
+            "args": ["--model", model_arg,
+                     "--served-model-name", model.name,
+                     "--port", str(ENGINE_PORT)] + list(spec.args),
+            "env": [],
+        }
+    if eng == "OLlama":
+        return {
+            "command": [],  # image entrypoint (`ollama serve`)
+            "args": list(spec.args),
+            "env": [
+                {"name": "OLLAMA_HOST", "value": f"0.0.0.0:{ENGINE_PORT}"},
+                {"name": "OLLAMA_KEEP_ALIVE", "value": "999999h"},
+            ],
+            # startup probe pulls + aliases the model (engine_ollama.go:173-213)
+            "startupProbeExec": [
+                "/bin/sh", "-c",
+                f"ollama pull {ref} && ollama cp {ref} {model.name} && "
+                f"ollama run {model.name} hi",
+            ],
+        }
+    if eng == "FasterWhisper":
+        return {
+            "command": [],
+            "args": list(spec.args),
+            "env": [
+                {"name": "WHISPER__MODEL", "value": ref},
+                {"name": "WHISPER__PORT", "value": str(ENGINE_PORT)},
+                {"name": "ENABLE_UI", "value": "false"},
+            ],
+        }
+    if eng == "Infinity":
+        return {
+            "command": [],
+            "args": list(spec.args),
+            "env": [
+                {"name": "INFINITY_MODEL_ID", "value": ref},
+                {"name": "INFINITY_SERVED_MODEL_NAME", "value": model.name},
+                {"name": "INFINITY_PORT", "value": str(ENGINE_PORT)},
+            ],
+        }
+    raise ValueError(f"no pod builder for engine {eng!r}")
+
+
 def pod_manifest_for(model, name: str, spec_hash: str, n_gpus: int,
                      image: str, namespace: str,
                      gpu_resource: str = "amd.com/gpu") -> dict:
     from .crd import POD_HASH_LABEL, POD_MODEL_LABEL
 
-    args = [
-        "--model", model.spec.url,
-        "--served-model-name", model.name,
-        "--port", str(ENGINE_PORT),
-    ] + list(model.spec.args)
+    eng = _engine_container(model, image)
+    args = eng["args"]
     env = [{"name": k, "value": v} for k, v in sorted(model.spec.env.items())]
+    env += eng["env"]
     src_env, src_vols, src_mounts = source_pod_additions(model.spec.url)
     env += src_env
     resources = {}
@@ -380,13 +440,20 @@ def pod_manifest_for(model, name: str, spec_hash: str, n_gpus: int,
                 {
                     "name": "server",
                     "image": image,
-                    "command": ["python", "-m", "kubeai_amd.engine.server"],
+                    **({"command": eng["command"]} if eng["command"] else {}),
                     "args": args,
                     "env": env,
                     "resources": resources,
                     "ports": [{"containerPort": ENGINE_PORT}],
-                    # reference probe contract engine_vllm.go:101-138
-                    "startupProbe": {**probe, "failureThreshold": 5400},
+                    # reference probe contract engine_vllm.go:101-138;
+                    # Ollama's startup probe pulls the model instead
+                    # (engine_ollama.go:173-213)
+                    "startupProbe": (
+                        {"exec": {"command": eng["startupProbeExec"]},
+                         "periodSeconds": 10, "failureThreshold": 1080}
+                        if eng.get("startupProbeExec")
+                        else {**probe, "failureThreshold": 5400}
+                    ),
                     "readinessProbe": {**probe, "failureThreshold": 3},
                     "livenessProbe": {
                         **probe, "periodSeconds": 10, "failureThreshold": 3,

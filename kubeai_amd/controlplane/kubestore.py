@@ -307,17 +307,23 @@ class KubeRuntime:
     readiness, which flows back through the pod watch."""
 
     def __init__(self, store: KubeStore, image: str = "kubeai-amd-engine:latest",
-                 gpu_resource: str = "amd.com/gpu"):
+                 gpu_resource: str = "amd.com/gpu",
+                 engine_images: Optional[dict] = None):
         self.store = store
         self.image = image
         self.gpu_resource = gpu_resource
+        # per-engine image defaults (chart modelServers analog,
+        # config/system.go:222-231)
+        self.engine_images = engine_images or {}
 
     async def create(self, model: Model, name: str, spec_hash: str,
                      n_gpus: int) -> None:
         kc = self.store.kc
         manifest = pod_manifest_for(
             model, name, spec_hash, n_gpus,
-            image=model.spec.image or self.image,
+            image=(model.spec.image
+                   or self.engine_images.get(model.spec.engine)
+                   or self.image),
             namespace=kc.namespace,
             gpu_resource=self.gpu_resource,
         )

@@ -50,6 +50,7 @@ class Manager:
                 self.store,
                 image=kube.get("engineImage", "kubeai-amd-engine:latest"),
                 gpu_resource=kube.get("gpuResource", "amd.com/gpu"),
+                engine_images=kube.get("engineImages"),
             )
             self.election = LeaseElection(
                 kc, lease_name=kube.get("leaseName", "kubeai.org")

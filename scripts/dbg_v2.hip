@@ -91,6 +91,38 @@ __global__ void probe_tr(ushort* out /* [4 ks][4 nb][64 lanes][8 jj] */,
     }
 }
 
+// ---- (b2) decode raw tr_read semantics -----------------------------
+// lds16[i] encodes its own index (two passes: low byte, high byte);
+// tr_read with a controlled per-lane address reveals exactly which
+// element each (lane, j) receives.
+__global__ void probe_tr_decode(int* out /* [2 addrmode][64][4] */,
+                                int pass /* 0=low,1=high */,
+                                int* acc /* accumulate */) {
+  __shared__ ushort v_lds[8192];
+  const int t = threadIdx.x;
+  for (int i = t; i < 8192; i += 64) {
+    const int enc = pass == 0 ? (i & 0xff) : (i >> 8);
+    v_lds[i] = f32_to_bf16((float)enc);
+  }
+  __syncthreads();
+  typedef __attribute__((address_space(3))) const char as3c;
+  for (int mode = 0; mode < 2; ++mode) {
+    // mode 0: wave-uniform base 0; mode 1: per-lane base (l>>4)*128 B
+    as3c* a0 = (as3c*)(v_lds) + (mode == 1 ? (t >> 4) * 128 : 0);
+    bf16x4 r;
+    asm volatile("ds_read_b64_tr_b16 %0, %1\n\ts_waitcnt lgkmcnt(0)"
+                 : "=v"(r) : "v"(a0));
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const int enc = (int)(bf16_to_f32((ushort)r[j]) + 0.5f);
+      int* slot = &out[(mode * 64 + t) * 4 + j];
+      if (pass == 0) *slot = enc;
+      else *slot |= enc << 8;
+    }
+  }
+  (void)acc;
+}
+
 // ---- (c) P exchange ------------------------------------------------
 // st2[sub][reg] encodes P[q][kv] = q*100 + kv (q = l&31 col,
 // kv = sub*32 + crow(reg, hi)); after assembly pa[ks] lane l elem jj
@@ -217,6 +249,39 @@ int main() {
         }
     printf("(c) P exchange: %d mismatches of %d\n", bad, 4 * 64 * 8);
     fails += bad != 0;
+  }
+  // ---------------- (b2) raw tr_read semantics
+  {
+    int* O;
+    hipMallocManaged(&O, 2 * 64 * 4 * 4);
+    hipLaunchKernelGGL(probe_tr_decode, dim3(1), dim3(64), 0, 0, O, 0,
+                       nullptr);
+    hipDeviceSynchronize();
+    hipLaunchKernelGGL(probe_tr_decode, dim3(1), dim3(64), 0, 0, O, 1,
+                       nullptr);
+    hipDeviceSynchronize();
+    for (int mode = 0; mode < 2; ++mode) {
+      printf("(b2) mode %d (%s):\n", mode,
+             mode ? "per-lane base (l>>4)*128B" : "uniform base 0");
+      for (int lane : {0, 1, 2, 15, 16, 17, 31, 32, 33, 48, 63}) {
+        printf("  lane %2d reads elems:", lane);
+        for (int j = 0; j < 4; ++j)
+          printf(" %5d", O[(mode * 64 + lane) * 4 + j]);
+        printf("\n");
+      }
+      // candidate formula fits
+      int fit1 = 0, fit2 = 0, fit3 = 0;
+      for (int l = 0; l < 64; ++l)
+        for (int j = 0; j < 4; ++j) {
+          const int base = mode ? (l >> 4) * 64 : 0;
+          const int got = O[(mode * 64 + l) * 4 + j];
+          if (got == base + (l & 15) + j * 16 + (l >> 4) * 64) ++fit1;
+          if (got == base + (l & 15) + j * 16) ++fit2;
+          if (got == base + (l & 15) * 4 + j) ++fit3;
+        }
+      printf("  fit[(l&15)+j*16+(l>>4)*64]=%d  fit[(l&15)+j*16]=%d  "
+             "fit[(l&15)*4+j]=%d  (of 256)\n", fit1, fit2, fit3);
+    }
   }
   printf(fails ? "FAIL %d\n" : "ALL OK\n", fails);
   return fails;

@@ -411,3 +411,40 @@ def test_source_pod_additions():
         "claimName"] == "claim"
     assert pod["spec"]["containers"][0]["volumeMounts"][0][
         "mountPath"] == "/model"
+
+
+def test_engine_pod_builders():
+    """Per-engine container contracts (reference engine_vllm.go /
+    engine_ollama.go / engine_fasterwhisper.go / engine_infinity.go)."""
+    from kubeai_amd.controlplane.kubeclient import pod_manifest_for
+
+    def pod(engine, url="hf://org/m", args=()):
+        m = Model(name="m", spec=ModelSpec(url=url, engine=engine,
+                                           args=list(args)))
+        return pod_manifest_for(m, "p", "h", 1, "img", "ns")
+
+    c = pod("VLLM")["spec"]["containers"][0]
+    assert c["command"][-1] == "vllm.entrypoints.openai.api_server"
+    assert "--served-model-name" in c["args"]
+
+    c = pod("OLlama", url="ollama://qwen2:0.5b")["spec"]["containers"][0]
+    env = {e["name"]: e.get("value") for e in c["env"]}
+    assert env["OLLAMA_HOST"] == "0.0.0.0:8000"
+    assert env["OLLAMA_KEEP_ALIVE"] == "999999h"
+    probe_cmd = " ".join(c["startupProbe"]["exec"]["command"])
+    assert "ollama pull qwen2:0.5b" in probe_cmd
+    assert "ollama cp qwen2:0.5b m" in probe_cmd
+
+    c = pod("FasterWhisper")["spec"]["containers"][0]
+    env = {e["name"]: e.get("value") for e in c["env"]}
+    assert env["WHISPER__MODEL"] == "org/m"
+    assert env["ENABLE_UI"] == "false"
+
+    c = pod("Infinity")["spec"]["containers"][0]
+    env = {e["name"]: e.get("value") for e in c["env"]}
+    assert env["INFINITY_MODEL_ID"] == "org/m"
+    assert env["INFINITY_SERVED_MODEL_NAME"] == "m"
+
+    # KubeAIEngine + pvc: model arg is the mount path
+    c = pod("KubeAIEngine", url="pvc://claim/path")["spec"]["containers"][0]
+    assert c["args"][c["args"].index("--model") + 1] == "/model"

@@ -47,6 +47,12 @@ void fused_add_rmsnorm_fp8(torch::Tensor out, torch::Tensor out_scale,
 void silu_and_mul_fp8(torch::Tensor out, torch::Tensor out_scale,
                       torch::Tensor x);
 void quant_fp8(torch::Tensor out, torch::Tensor out_scale, torch::Tensor x);
+void nucleus_stats(torch::Tensor m, torch::Tensor z, torch::Tensor logits,
+                   torch::Tensor temps);
+void nucleus_accept(torch::Tensor ok, torch::Tensor logits,
+                    torch::Tensor cand, torch::Tensor m, torch::Tensor z,
+                    torch::Tensor temps, torch::Tensor top_ps,
+                    torch::Tensor top_ks);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm", &rmsnorm, "RMSNorm (bf16, fp32 accum)");
@@ -69,6 +75,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "residual add + rmsnorm with fused fp8 row quant");
   m.def("silu_and_mul_fp8", &silu_and_mul_fp8, "SwiGLU with fused fp8 quant");
   m.def("quant_fp8", &quant_fp8, "bf16 -> fp8 row quant");
+  m.def("nucleus_stats", &nucleus_stats, "per-row softmax max + Z");
+  m.def("nucleus_accept", &nucleus_accept,
+        "nucleus membership of sampled tokens (mass above < p, rank < k)");
   m.def("xgmi_alloc", &xgmi_alloc, "allocate + export the IPC comm buffer");
   m.def("xgmi_connect", &xgmi_connect, "open peer IPC buffers");
   m.def("xgmi_fused_allreduce_add_rmsnorm", &xgmi_fused_allreduce_add_rmsnorm,

@@ -143,3 +143,109 @@ void gumbel_sample(torch::Tensor out, torch::Tensor logits,
   launch_sample<true>(out, logits, temperature.data_ptr<float>(),
                       seeds.data_ptr<int64_t>(), step);
 }
+
+// ---------------------------------------------------------------------
+// Rejection-based nucleus sampling support (runner._sample_topk_topp):
+// one pass computes per-row softmax stats; per candidate draw another
+// single pass decides membership (probability mass strictly above the
+// sampled logit < top_p, and rank < top_k) — no [S,V] intermediates.
+
+namespace {
+
+__global__ void nucleus_stats_kernel(
+    float* __restrict__ m_out, float* __restrict__ z_out,
+    const float* __restrict__ logits, const float* __restrict__ temps,
+    const int V) {
+  const int row = blockIdx.x;
+  const float* x = logits + (int64_t)row * V;
+  const float t = fmaxf(temps[row], 1e-6f);
+  __shared__ float red[16];
+  float m = -INFINITY;
+  for (int i = threadIdx.x; i < V; i += blockDim.x) m = fmaxf(m, x[i]);
+  m = wave_reduce_max(m);
+  const int wave = threadIdx.x / WAVE_SIZE, lane = threadIdx.x % WAVE_SIZE;
+  const int nw = blockDim.x / WAVE_SIZE;
+  if (lane == 0) red[wave] = m;
+  __syncthreads();
+  float mm = -INFINITY;
+  for (int w = 0; w < nw; ++w) mm = fmaxf(mm, red[w]);
+  __syncthreads();
+  float z = 0.f;
+  for (int i = threadIdx.x; i < V; i += blockDim.x)
+    z += __expf((x[i] - mm) / t);
+  z = block_reduce_sum(z, red);
+  if (threadIdx.x == 0) {
+    m_out[row] = mm;
+    z_out[row] = z;
+  }
+}
+
+__global__ void nucleus_accept_kernel(
+    unsigned char* __restrict__ ok,      // [S]
+    const float* __restrict__ logits,    // [S, V]
+    const int64_t* __restrict__ cand,    // [S]
+    const float* __restrict__ m_in, const float* __restrict__ z_in,
+    const float* __restrict__ temps, const float* __restrict__ top_ps,
+    const int32_t* __restrict__ top_ks, const int V) {
+  const int row = blockIdx.x;
+  const float t0 = temps[row];
+  if (t0 <= 0.f) {  // greedy rows: argmax is always in the nucleus
+    if (threadIdx.x == 0) ok[row] = 1;
+    return;
+  }
+  const float* x = logits + (int64_t)row * V;
+  const float lt = x[cand[row]];
+  const float t = fmaxf(t0, 1e-6f);
+  const float m = m_in[row], z = z_in[row];
+  __shared__ float red[16];
+  float mass = 0.f;
+  int cnt = 0;
+  for (int i = threadIdx.x; i < V; i += blockDim.x) {
+    const float xi = x[i];
+    if (xi > lt) {
+      mass += __expf((xi - m) / t);
+      ++cnt;
+    }
+  }
+  mass = block_reduce_sum(mass, red);
+  __syncthreads();
+  float cntf = block_reduce_sum((float)cnt, red);
+  if (threadIdx.x == 0) {
+    const float mass_above = mass / z;
+    const int k = top_ks[row];
+    ok[row] = (mass_above < top_ps[row] && (k <= 0 || (int)cntf < k)) ? 1 : 0;
+  }
+}
+
+}  // namespace
+
+void nucleus_stats(torch::Tensor m, torch::Tensor z, torch::Tensor logits,
+                   torch::Tensor temps) {
+  TORCH_CHECK(logits.is_contiguous() &&
+              logits.scalar_type() == torch::kFloat32);
+  const int S = logits.size(0), V = logits.size(1);
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  hipLaunchKernelGGL(nucleus_stats_kernel, dim3(S), dim3(512), 0, stream,
+                     m.data_ptr<float>(), z.data_ptr<float>(),
+                     logits.data_ptr<float>(), temps.data_ptr<float>(), V);
+  HIP_CHECK_KERNEL();
+}
+
+void nucleus_accept(torch::Tensor ok, torch::Tensor logits,
+                    torch::Tensor cand, torch::Tensor m, torch::Tensor z,
+                    torch::Tensor temps, torch::Tensor top_ps,
+                    torch::Tensor top_ks) {
+  TORCH_CHECK(logits.is_contiguous() &&
+              logits.scalar_type() == torch::kFloat32);
+  TORCH_CHECK(cand.scalar_type() == torch::kInt64);
+  TORCH_CHECK(top_ks.scalar_type() == torch::kInt32);
+  const int S = logits.size(0), V = logits.size(1);
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  hipLaunchKernelGGL(nucleus_accept_kernel, dim3(S), dim3(512), 0, stream,
+                     ok.data_ptr<unsigned char>(),
+                     logits.data_ptr<float>(), cand.data_ptr<int64_t>(),
+                     m.data_ptr<float>(), z.data_ptr<float>(),
+                     temps.data_ptr<float>(), top_ps.data_ptr<float>(),
+                     top_ks.data_ptr<int32_t>(), V);
+  HIP_CHECK_KERNEL();
+}

@@ -637,23 +637,20 @@ def _sample_topk_topp(logits, top_ps, top_ks, temps, t_t, seeds, step):
     S, V = logits.shape
     dev = logits.device
     tp = torch.tensor(top_ps, device=dev, dtype=torch.float32)
-    tk = torch.tensor(top_ks, device=dev, dtype=torch.int64)
+    tk = torch.tensor(top_ks, device=dev, dtype=torch.int32)
     tt = torch.tensor(temps, device=dev, dtype=torch.float32)
-    ttc = tt.clamp_min(1e-6).unsqueeze(1)
-    # exact full-distribution probabilities at the sampling temperature
-    probs = torch.softmax(logits.float() / ttc, dim=-1)
-    logits_c = logits.contiguous()
+    logits_c = logits.float().contiguous()
+    # per-row softmax stats once; each draw is then ONE fused pass
+    m, z = ops.nucleus_stats(logits_c, tt)
 
     def draw(attempt: int):
         # fresh deterministic noise per attempt
         cand = ops.gumbel_sample(
             logits_c, t_t, seeds, step + (attempt + 1) * 1_000_003
         ).to(dev)
-        lt = logits.gather(1, cand.view(-1, 1))          # sampled logit
-        above = logits > lt                              # strictly higher
-        mass_above = (probs * above).sum(-1)
-        cnt_above = above.sum(-1)
-        ok = (mass_above < tp) & ((tk <= 0) | (cnt_above < tk)) | (tt <= 0)
+        ok = ops.nucleus_accept(
+            logits_c, cand, m, z, tt, tp, tk
+        ).to(torch.bool)
         return cand, ok
 
     # three sync-free draws; P(all miss) <= (1-p)^3 per row, so the

@@ -209,3 +209,37 @@ def gumbel_sample(logits, temperature, seeds, step: int):
         _C.gumbel_sample(out, logits, temperature, seeds, step)
         return out
     return ref.gumbel_sample(logits, temperature, seeds, step)
+
+
+def nucleus_stats(logits, temps):
+    """Per-row (max, sum exp((x-max)/t)) for rejection nucleus sampling."""
+    if logits.is_cuda:
+        _require_ext()
+        S = logits.shape[0]
+        m = torch.empty(S, dtype=torch.float32, device=logits.device)
+        z = torch.empty_like(m)
+        _C.nucleus_stats(m, z, logits, temps)
+        return m, z
+    t = temps.clamp_min(1e-6).unsqueeze(1)
+    m = logits.max(dim=-1).values
+    z = torch.exp((logits - m.unsqueeze(1)) / t).sum(-1)
+    return m, z
+
+
+def nucleus_accept(logits, cand, m, z, temps, top_ps, top_ks):
+    """uint8 mask: sampled token lies in the top-p/top-k nucleus
+    (probability mass strictly above it < top_p, rank < top_k);
+    greedy rows always accept."""
+    if logits.is_cuda:
+        _require_ext()
+        ok = torch.empty(logits.shape[0], dtype=torch.uint8,
+                         device=logits.device)
+        _C.nucleus_accept(ok, logits, cand, m, z, temps, top_ps, top_ks)
+        return ok
+    t = temps.clamp_min(1e-6).unsqueeze(1)
+    lt = logits.gather(1, cand.view(-1, 1))
+    above = logits > lt
+    mass = (torch.exp((logits - m.unsqueeze(1)) / t) * above).sum(-1) / z
+    cnt = above.sum(-1)
+    ok = (mass < top_ps) & ((top_ks <= 0) | (cnt < top_ks)) | (temps <= 0)
+    return ok.to(torch.uint8)

@@ -26,6 +26,13 @@ for S, V in ((40, 128256), (256, 128256)):
     seeds = torch.arange(S, dtype=torch.int64, device=dev)
 
     us_mask = timeit(lambda: _apply_topk_topp(logits, tps, tks, temps))
+    tp_t = torch.full((S,), 0.9, device=dev)
+    tk_t = torch.zeros(S, dtype=torch.int32, device=dev)
+    us_stats = timeit(lambda: ops.nucleus_stats(logits, t_t))
+    m_, z_ = ops.nucleus_stats(logits, t_t)
+    cand_ = ops.gumbel_sample(logits, t_t, seeds, 3)
+    us_accept = timeit(
+        lambda: ops.nucleus_accept(logits, cand_, m_, z_, t_t, tp_t, tk_t))
     us_subset = timeit(
         lambda: _sample_topk_topp(logits, tps, tks, temps, t_t, seeds, 3)
     )
@@ -46,7 +53,7 @@ for S, V in ((40, 128256), (256, 128256)):
     us_greedy = timeit(lambda: ops.greedy_sample(logits))
     us_topk = timeit(lambda: torch.topk(logits, 1024, dim=-1))
     us_lse = timeit(lambda: torch.logsumexp(logits / 0.8, dim=-1))
-    print(f"S={S:4d} V={V}: SAMPLE(subset)={us_subset:8.1f}us  mask(fast)={us_mask:8.1f}us  mask(sort)="
+    print(f"S={S:4d} V={V}: SAMPLE(rej)={us_subset:8.1f}us  stats={us_stats:6.1f}us  accept={us_accept:6.1f}us  mask(fast)={us_mask:8.1f}us  mask(sort)="
           f"{us_sort:8.1f}us  topk1024={us_topk:8.1f}us  lse={us_lse:7.1f}us"
           f"  gumbel={us_gumbel:7.1f}us  greedy={us_greedy:7.1f}us",
           flush=True)

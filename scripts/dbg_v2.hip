@@ -71,7 +71,11 @@ __global__ void probe_tr(ushort* out /* [4 ks][4 nb][64 lanes][8 jj] */,
   __syncthreads();
   const int lane = t, hi = lane >> 5;
   typedef __attribute__((address_space(3))) const char as3c;
-  as3c* vbase = (as3c*)(v_lds) + hi * 1792;
+  // quad-cooperative addressing (see b2 decode): each lane addresses the
+  // row it contributes to its quad's 4x4 transpose
+  as3c* vbase = (as3c*)(v_lds) +
+                (hi * 2048 + ((lane & 31) >> 4) * 128 + (lane & 3) * 32 +
+                 (lane & 12) * 2);
   for (int ks = 0; ks < 4; ++ks)
     for (int nb = 0; nb < 4; ++nb) {
       bf16x4 lo4, hi4;

@@ -33,6 +33,14 @@ for S, V in ((40, 128256), (256, 128256)):
     cand_ = ops.gumbel_sample(logits, t_t, seeds, 3)
     us_accept = timeit(
         lambda: ops.nucleus_accept(logits, cand_, m_, z_, t_t, tp_t, tk_t))
+    rates = []
+    for a in range(3):
+        c = ops.gumbel_sample(logits, t_t, seeds, 1000 + a)
+        rates.append(
+            ops.nucleus_accept(logits, c, m_, z_, t_t, tp_t, tk_t)
+            .float().mean().item())
+    print(f"  accept rates per draw (expect ~{tps[0]}): "
+          f"{[round(r, 3) for r in rates]}")
     us_subset = timeit(
         lambda: _sample_topk_topp(logits, tps, tks, temps, t_t, seeds, 3)
     )

@@ -127,6 +127,33 @@ __global__ void probe_tr_decode(int* out /* [2 addrmode][64][4] */,
   (void)acc;
 }
 
+// ---- (b3) definitive tr_read decode: unique per-lane addresses -----
+// lane l's address = l * 256 bytes; lds[i] = i (self-indexing, split
+// low/high passes). Each output value v decodes as:
+//   source_lane = v / 128   (whose ADDRESS was used)
+//   elem_offset = v % 128   (internal offset applied, in elements)
+__global__ void probe_tr_unique(int* out /* [64][4] */, int pass) {
+  __shared__ ushort v_lds[8192];
+  const int t = threadIdx.x;
+  for (int i = t; i < 8192; i += 64) {
+    const int enc = pass == 0 ? (i & 0xff) : (i >> 8);
+    v_lds[i] = f32_to_bf16((float)enc);
+  }
+  __syncthreads();
+  typedef __attribute__((address_space(3))) const char as3c;
+  as3c* a0 = (as3c*)(v_lds) + t * 256;
+  bf16x4 r;
+  asm volatile("ds_read_b64_tr_b16 %0, %1\n\ts_waitcnt lgkmcnt(0)"
+               : "=v"(r) : "v"(a0));
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    const int enc = (int)(bf16_to_f32((ushort)r[j]) + 0.5f);
+    int* slot = &out[t * 4 + j];
+    if (pass == 0) *slot = enc;
+    else *slot |= enc << 8;
+  }
+}
+
 // ---- (c) P exchange ------------------------------------------------
 // st2[sub][reg] encodes P[q][kv] = q*100 + kv (q = l&31 col,
 // kv = sub*32 + crow(reg, hi)); after assembly pa[ks] lane l elem jj
@@ -286,6 +313,30 @@ int main() {
       printf("  fit[(l&15)+j*16+(l>>4)*64]=%d  fit[(l&15)+j*16]=%d  "
              "fit[(l&15)*4+j]=%d  (of 256)\n", fit1, fit2, fit3);
     }
+  }
+  // ---------------- (b3) unique-address decode
+  {
+    int* O;
+    hipMallocManaged(&O, 64 * 4 * 4);
+    hipLaunchKernelGGL(probe_tr_unique, dim3(1), dim3(64), 0, 0, O, 0);
+    hipDeviceSynchronize();
+    hipLaunchKernelGGL(probe_tr_unique, dim3(1), dim3(64), 0, 0, O, 1);
+    hipDeviceSynchronize();
+    printf("(b3) addr_l = l*256B; value -> (src_lane, elem_off):\n");
+    for (int l = 0; l < 16; ++l) {
+      printf("  lane %2d:", l);
+      for (int j = 0; j < 4; ++j) {
+        const int v = O[l * 4 + j];
+        printf("  (%2d,%3d)", v / 128, v % 128);
+      }
+      printf("\n");
+    }
+    printf("  lane 32:");
+    for (int j = 0; j < 4; ++j) {
+      const int v = O[32 * 4 + j];
+      printf("  (%2d,%3d)", v / 128, v % 128);
+    }
+    printf("\n");
   }
   printf(fails ? "FAIL %d\n" : "ALL OK\n", fails);
   return fails;

@@ -301,3 +301,32 @@ def test_gumbel_sample_matches_reference_rng():
     assert torch.equal(out, out2)
     out3 = ops.gumbel_sample(logits, temps, seeds, step=8)
     assert not torch.equal(out, out3)
+
+
+def test_nucleus_kernels_match_reference():
+    """Fused nucleus_stats/nucleus_accept vs the plain torch formula."""
+    torch.manual_seed(0)
+    S, V = 9, 5000
+    logits = (torch.randn(S, V, device=DEV) * 3).float().contiguous()
+    temps = torch.tensor([0.8] * 4 + [0.0] + [1.3] * 4, device=DEV)
+    top_ps = torch.tensor([0.9, 0.5, 0.99, 1.0, 0.9, 0.7, 0.9, 0.9, 0.2],
+                          device=DEV)
+    top_ks = torch.tensor([0, 0, 50, 0, 0, 0, 3, 100000, 0],
+                          dtype=torch.int32, device=DEV)
+    m, z = ops.nucleus_stats(logits, temps)
+    # reference stats
+    t = temps.clamp_min(1e-6).unsqueeze(1)
+    m_ref = logits.max(dim=-1).values
+    z_ref = torch.exp((logits - m_ref.unsqueeze(1)) / t).sum(-1)
+    assert torch.allclose(m, m_ref)
+    assert torch.allclose(z, z_ref, rtol=1e-3)
+    for trial in range(5):
+        cand = torch.randint(0, V, (S,), dtype=torch.int64, device=DEV)
+        ok = ops.nucleus_accept(logits, cand, m, z, temps, top_ps, top_ks)
+        lt = logits.gather(1, cand.view(-1, 1))
+        above = logits > lt
+        mass = (torch.exp((logits - m_ref.unsqueeze(1)) / t) * above).sum(-1) / z_ref
+        cnt = above.sum(-1)
+        want = ((mass < top_ps) & ((top_ks <= 0) | (cnt < top_ks))
+                | (temps <= 0)).to(torch.uint8)
+        assert torch.equal(ok.cpu(), want.cpu()), (trial, ok, want)

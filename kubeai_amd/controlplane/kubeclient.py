@@ -404,7 +404,12 @@ def _engine_container(model, image: str) -> dict:
 
 def pod_manifest_for(model, name: str, spec_hash: str, n_gpus: int,
                      image: str, namespace: str,
-                     gpu_resource: str = "amd.com/gpu") -> dict:
+                     gpu_resource: str = "amd.com/gpu",
+                     pod_config: Optional[dict] = None) -> dict:
+    """pod_config = the reference's `modelServerPods` admin settings
+    (config/system.go:243-260): serviceAccountName, securityContext,
+    podSecurityContext, imagePullSecrets, jsonPatches (RFC 6902, applied
+    last — pod_plan.go:42-44)."""
     from .crd import POD_HASH_LABEL, POD_MODEL_LABEL
 
     eng = _engine_container(model, image)
@@ -423,7 +428,8 @@ def pod_manifest_for(model, name: str, spec_hash: str, n_gpus: int,
         "httpGet": {"path": "/health", "port": ENGINE_PORT},
         "periodSeconds": 2,
     }
-    return {
+    pc = pod_config or {}
+    manifest = {
         "apiVersion": "v1",
         "kind": "Pod",
         "metadata": {
@@ -459,9 +465,35 @@ def pod_manifest_for(model, name: str, spec_hash: str, n_gpus: int,
                         **probe, "periodSeconds": 10, "failureThreshold": 3,
                     },
                     **({"volumeMounts": src_mounts} if src_mounts else {}),
+                    **(
+                        {"securityContext": pc["securityContext"]}
+                        if pc.get("securityContext")
+                        else {}
+                    ),
                 }
             ],
             **({"volumes": src_vols} if src_vols else {}),
+            **(
+                {"serviceAccountName": pc["serviceAccountName"]}
+                if pc.get("serviceAccountName")
+                else {}
+            ),
+            **(
+                {"securityContext": pc["podSecurityContext"]}
+                if pc.get("podSecurityContext")
+                else {}
+            ),
+            **(
+                {"imagePullSecrets": pc["imagePullSecrets"]}
+                if pc.get("imagePullSecrets")
+                else {}
+            ),
             "restartPolicy": "Never",
         },
     }
+    patches = pc.get("jsonPatches") or []
+    if patches:
+        from .jsonpatch import apply_patch
+
+        manifest = apply_patch(manifest, patches)
+    return manifest

@@ -299,19 +299,19 @@ __launch_bounds__(512) __global__ void paged_prefill_v2_kernel(
       }
 
       // ---- O += P.V via hardware-transpose V reads -----------------
-      // ds_read_b64_tr_b16 is QUAD-cooperative (device-decoded,
-      // scripts/dbg_v2.hip b2): lane l's element j is read from the 8 B
-      // row addressed by quad-lane (l>>2)*4+j, at column (l&3). Each
-      // lane therefore addresses the subtiled V row it contributes:
-      //   row = ks*16 + (l>>5)*8 + 4*tr + (l&3),
-      //   col base = nb*32 + (l&28)
-      // => byte addr = ks*4096 + tr*1024 + (l>>5)*2048 + nb*256
-      //               + ((l&31)>>4)*128 + (l&3)*32 + (l&12)*2,
-      // with (ks, tr, nb) walked by compile-time offsets.
+      // ds_read_b64_tr_b16 semantics (device-decoded, dbg_v2 b3): a
+      // 4x4 transpose over each 16-lane group's lane grid (row=l>>2,
+      // col=l&3): result[l][j] = lds[addr_of_lane((l&~15) + ((l>>2)&3)
+      // + 4*j) + 2*(l&3)]. Solving for the address each lane must
+      // supply so lane l's elem jj = V[ks*16+(l>>5)*8+jj][nb*32+(l&31)]
+      // in the 4x16-subtiled layout:
+      //   byte addr(m) = ks*4096 + tr*1024 + (m>>5)*2048 + nb*256
+      //     + ((m>>4)&1)*128 + ((m>>2)&3)*32 + (m&3)*8
       typedef __attribute__((address_space(3))) const char as3_char;
       as3_char* vbase =
-          (as3_char*)(v_lds[cur]) + (hi * 2048 + ((lane & 31) >> 4) * 128 +
-                                     (lane & 3) * 32 + (lane & 12) * 2);
+          (as3_char*)(v_lds[cur]) +
+          (hi * 2048 + ((lane >> 4) & 1) * 128 + ((lane >> 2) & 3) * 32 +
+           (lane & 3) * 8);
 #pragma unroll
       for (int nb = 0; nb < 4; ++nb) {
 #pragma unroll

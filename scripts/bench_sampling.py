@@ -41,6 +41,20 @@ for S, V in ((40, 128256), (256, 128256)):
             .float().mean().item())
     print(f"  accept rates per draw (expect ~{tps[0]}): "
           f"{[round(r, 3) for r in rates]}")
+
+    def manual():
+        m3, z3 = ops.nucleus_stats(logits, t_t)
+        c1 = ops.gumbel_sample(logits, t_t, seeds, 11)
+        o1 = ops.nucleus_accept(logits, c1, m3, z3, t_t, tp_t, tk_t)
+        c2 = ops.gumbel_sample(logits, t_t, seeds, 12)
+        o2 = ops.nucleus_accept(logits, c2, m3, z3, t_t, tp_t, tk_t)
+        c3 = ops.gumbel_sample(logits, t_t, seeds, 13)
+        o3 = ops.nucleus_accept(logits, c3, m3, z3, t_t, tp_t, tk_t)
+        return torch.where(o1.bool(), c1,
+                           torch.where(o2.bool(), c2, c3))
+
+    us_manual = timeit(manual)
+    print(f"  manual 3-draw pipeline (no fallback check): {us_manual:8.1f}us")
     us_subset = timeit(
         lambda: _sample_topk_topp(logits, tps, tks, temps, t_t, seeds, 3)
     )

@@ -448,3 +448,56 @@ def test_engine_pod_builders():
     # KubeAIEngine + pvc: model arg is the mount path
     c = pod("KubeAIEngine", url="pvc://claim/path")["spec"]["containers"][0]
     assert c["args"][c["args"].index("--model") + 1] == "/model"
+
+
+def test_model_server_pods_and_json_patches():
+    """modelServerPods admin settings + RFC-6902 jsonPatches applied to
+    every engine pod (reference system.go:243-260, pod_plan.go:42-44)."""
+    from kubeai_amd.controlplane.kubeclient import pod_manifest_for
+
+    m = Model(name="m", spec=ModelSpec(url="hf://org/m"))
+    pc = {
+        "serviceAccountName": "kubeai-engine",
+        "podSecurityContext": {"runAsUser": 1000},
+        "securityContext": {"allowPrivilegeEscalation": False},
+        "imagePullSecrets": [{"name": "regcred"}],
+        "jsonPatches": [
+            {"op": "add", "path": "/metadata/labels/team", "value": "ml"},
+            {"op": "add", "path": "/spec/containers/0/env/-",
+             "value": {"name": "EXTRA", "value": "1"}},
+            {"op": "replace", "path": "/spec/restartPolicy",
+             "value": "Always"},
+        ],
+    }
+    pod = pod_manifest_for(m, "p", "h", 1, "img", "ns", pod_config=pc)
+    assert pod["spec"]["serviceAccountName"] == "kubeai-engine"
+    assert pod["spec"]["securityContext"]["runAsUser"] == 1000
+    assert pod["spec"]["containers"][0]["securityContext"][
+        "allowPrivilegeEscalation"] is False
+    assert pod["spec"]["imagePullSecrets"] == [{"name": "regcred"}]
+    assert pod["metadata"]["labels"]["team"] == "ml"
+    assert pod["spec"]["containers"][0]["env"][-1]["name"] == "EXTRA"
+    assert pod["spec"]["restartPolicy"] == "Always"
+
+
+def test_jsonpatch_ops():
+    from kubeai_amd.controlplane.jsonpatch import PatchError, apply_patch
+
+    doc = {"a": {"b": [1, 2, 3]}, "x": 5}
+    out = apply_patch(doc, [
+        {"op": "test", "path": "/x", "value": 5},
+        {"op": "add", "path": "/a/b/1", "value": 99},
+        {"op": "remove", "path": "/x"},
+        {"op": "copy", "from": "/a/b/0", "path": "/c"},
+        {"op": "move", "from": "/a/b/3", "path": "/moved"},
+        {"op": "replace", "path": "/c", "value": "z"},
+    ])
+    # add at /a/b/1 -> [1,99,2,3]; move /a/b/3 (value 3) out -> [1,99,2]
+    assert out == {"a": {"b": [1, 99, 2]}, "c": "z", "moved": 3}
+    assert doc["x"] == 5  # original untouched
+    import pytest as _p
+
+    with _p.raises(PatchError):
+        apply_patch(doc, [{"op": "test", "path": "/x", "value": 6}])
+    with _p.raises(PatchError):
+        apply_patch(doc, [{"op": "replace", "path": "/nope", "value": 1}])

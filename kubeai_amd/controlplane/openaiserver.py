@@ -61,10 +61,18 @@ def build_gateway_app(model_client: ModelClient, proxy: ProxyHandler) -> Starlet
     async def healthz(request: Request) -> Response:
         return JSONResponse({"status": "ok"})
 
+    async def debug_traces(request: Request) -> Response:
+        # exporter-less trace inspection (tracing.py ring buffer)
+        from .tracing import TRACER
+
+        limit = int(request.query_params.get("limit", "100"))
+        return JSONResponse({"spans": TRACER.recent(limit)})
+
     routes = [
         Route("/openai/v1/models", models, methods=["GET"]),
         Route("/metrics", metrics_endpoint, methods=["GET"]),
         Route("/healthz", healthz, methods=["GET"]),
+        Route("/debug/traces", debug_traces, methods=["GET"]),
     ]
     for p in PROXIED_PATHS:
         routes.append(Route("/openai" + p, proxied, methods=["POST"]))

@@ -47,6 +47,26 @@ class ProxyHandler:
         await self.client.aclose()
 
     async def handle(self, request: Request, path: str) -> Response:
+        # W3C TraceContext server span (reference wraps handlers in
+        # otelhttp route tags; tracing.py is the exporter-less analog).
+        # For streaming responses the span covers time-to-headers.
+        from .tracing import TRACER, parse_traceparent
+
+        parent = parse_traceparent(request.headers.get("traceparent"))
+        span = TRACER.start_span(f"proxy {path}", parent)
+        span.set("http.route", path)
+        request.state.trace_ctx = span.ctx
+        try:
+            resp = await self._handle_inner(request, path)
+            span.set("http.status_code", getattr(resp, "status_code", 0))
+            return resp
+        except BaseException:
+            span.set("error", True)
+            raise
+        finally:
+            span.end()
+
+    async def _handle_inner(self, request: Request, path: str) -> Response:
         try:
             raw = await request.body()
             ctype = request.headers.get("content-type", "")
@@ -123,6 +143,10 @@ class ProxyHandler:
             if k.lower() in ("content-type", "accept", "authorization", "traceparent")
         }
         headers.setdefault("content-type", "application/json")
+        ctx = getattr(request.state, "trace_ctx", None)
+        if ctx is not None:
+            # engine sees the CHILD context, not the client's original
+            headers["traceparent"] = ctx.traceparent()
         req = self.client.build_request("POST", url, content=payload, headers=headers)
         return await self.client.send(req, stream=True)
 

@@ -55,6 +55,17 @@ for S, V in ((40, 128256), (256, 128256)):
 
     us_manual = timeit(manual)
     print(f"  manual 3-draw pipeline (no fallback check): {us_manual:8.1f}us")
+    # pending-rate diagnostic with the wrapper's exact step values
+    step0 = 3
+    oks = []
+    cs = []
+    for a in range(3):
+        c = ops.gumbel_sample(logits, t_t, seeds, step0 + (a + 1) * 1_000_003)
+        cs.append(c)
+        oks.append(ops.nucleus_accept(logits, c, m_, z_, t_t, tp_t, tk_t).bool())
+    pend = ~(oks[0] | oks[1] | oks[2])
+    print(f"  wrapper-step pending rate: {pend.float().mean().item():.4f}  "
+          f"draw-identical(c1==c2): {(cs[0] == cs[1]).float().mean().item():.3f}")
     us_subset = timeit(
         lambda: _sample_topk_topp(logits, tps, tks, temps, t_t, seeds, 3)
     )

@@ -42,6 +42,7 @@ class FakeBackend:
 
     def __init__(self):
         self.requests = []
+        self.last_headers = {}
         self.waiting = 0      # engine queue depth exposed on /metrics
         self.kv_usage = 0.0   # KV occupancy exposed on /metrics
         self.fail_next = 0
@@ -57,6 +58,7 @@ class FakeBackend:
         async def completions(request):
             body = await request.json()
             self.requests.append((request.url.path, body))
+            self.last_headers = dict(request.headers)
             if self.fail_next > 0:
                 self.fail_next -= 1
                 return JSONResponse({"error": "boom"}, status_code=500)
@@ -707,3 +709,49 @@ def test_autoscaler_kv_pressure_scales_up():
             )
 
     run(body())
+
+
+def test_trace_propagation_and_debug_traces():
+    """Gateway records a server span per request, forwards a CHILD
+    traceparent to the engine, and serves /debug/traces (reference
+    installs W3C propagators with no exporter — otel.go:76-81)."""
+    import asyncio
+
+    from kubeai_amd.controlplane.tracing import parse_traceparent
+
+    async def main():
+        async with harness(
+            models=[
+                Model(name="tr-model",
+                      spec=ModelSpec(url="hf://org/m", min_replicas=1))
+            ]
+        ) as (mgr, runtime, backend):
+            await wait_for(lambda: mgr.store.list_replicas(model="tr-model"))
+            rep = mgr.store.list_replicas(model="tr-model")[0]
+            runtime.mark_ready(rep.name, backend.address)
+            from httpx import ASGITransport, AsyncClient
+
+            async with AsyncClient(
+                transport=ASGITransport(app=mgr.app), base_url="http://t"
+            ) as client:
+                tp = "00-" + "ab" * 16 + "-" + "cd" * 8 + "-01"
+                r = await client.post(
+                    "/openai/v1/completions",
+                    json={"model": "tr-model", "prompt": "x", "max_tokens": 2},
+                    headers={"traceparent": tp},
+                )
+                assert r.status_code == 200
+                # the engine received a CHILD of the client's trace
+                fwd = backend.last_headers.get("traceparent")
+                ctx = parse_traceparent(fwd)
+                assert ctx is not None
+                assert ctx.trace_id == "ab" * 16
+                assert ctx.span_id != "cd" * 8
+                # span recorded with route + status
+                tr = await client.get("/debug/traces")
+                spans = tr.json()["spans"]
+                mine = [s for s in spans if s["traceId"] == "ab" * 16]
+                assert mine and mine[0]["parentSpanId"] == "cd" * 8
+                assert mine[0]["attributes"]["http.status_code"] == 200
+
+    asyncio.run(main())

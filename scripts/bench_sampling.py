@@ -66,6 +66,18 @@ for S, V in ((40, 128256), (256, 128256)):
     pend = ~(oks[0] | oks[1] | oks[2])
     print(f"  wrapper-step pending rate: {pend.float().mean().item():.4f}  "
           f"draw-identical(c1==c2): {(cs[0] == cs[1]).float().mean().item():.3f}")
+    rows = pend.nonzero(as_tuple=True)[0].tolist()
+    if rows:
+        # acceptance probability of the problem rows over 50 fresh draws
+        accs = torch.zeros(S, device=dev)
+        for a in range(50):
+            c = ops.gumbel_sample(logits, t_t, seeds, 7_000_000 + a * 97)
+            accs += ops.nucleus_accept(
+                logits, c, m_, z_, t_t, tp_t, tk_t).float()
+        print(f"  pending rows {rows[:6]} accept-prob over 50 draws: "
+              f"{[round(accs[r].item() / 50, 2) for r in rows[:6]]}  "
+              f"(healthy rows ~0.9; their cands: "
+              f"{[int(cs[0][r]) for r in rows[:3]]})")
     us_subset = timeit(
         lambda: _sample_topk_topp(logits, tps, tks, temps, t_t, seeds, 3)
     )

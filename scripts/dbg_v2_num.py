@@ -64,6 +64,11 @@ def run_case(name, Tq, ctx, k_mode, q_mode, v_mode):
     print(f"   err by head: {[round(v, 3) for v in by_head.tolist()]}")
     by_d16 = err.view(Tq, nq, hd // 16, 16).amax(dim=(0, 1, 3))
     print(f"   err by d16-block: {[round(v, 3) for v in by_d16.tolist()]}")
+    flat = err.flatten()
+    top = flat.topk(8).indices
+    coords = [(int(i) // (nq * hd), (int(i) // hd) % nq, int(i) % hd)
+              for i in top.tolist()]
+    print(f"   top-err (q,h,d): {coords}")
     if bad_rows:
         r = bad_rows[0]
         h = int(err[r].amax(dim=1).argmax())
@@ -76,3 +81,7 @@ run_case("B softmax-sums", 64, 0, "rand", "rand", "ones")
 run_case("C multi-tile", 192, 0, "zero", "zero", "pattern")
 run_case("D random+ctx", 192, 37, "rand", "rand", "rand")
 run_case("E random 1 tile", 64, 0, "rand", "rand", "rand")
+run_case("F random multi-tile", 192, 0, "rand", "rand", "rand")
+run_case("G random 1 tile + ctx", 27, 37, "rand", "rand", "rand")
+run_case("H pattern + ctx", 64, 37, "zero", "zero", "pattern")
+run_case("I ones + ctx (sums)", 192, 37, "rand", "rand", "ones")

@@ -66,7 +66,12 @@ __launch_bounds__(512) __global__ void paged_prefill_v2_kernel(
     const int32_t* __restrict__ query_start_loc,  // [B+1]
     const int32_t* __restrict__ seq_lens,         // [B]
     const float scale, const int n_kv, const int max_blocks,
-    const int64_t q_stride) {
+    const int64_t q_stride, const int dbg_mode) {
+  // dbg_mode bisect switches (host env KUBEAI_V2_DBG):
+  //   bit 0: synchronous single-slot staging (no async double buffer)
+  //   bit 1: plain scalar V reads instead of ds_read_b64_tr_b16
+  const bool kSyncStage = dbg_mode & 1;
+  const bool kSimpleV = dbg_mode & 2;
   constexpr int kQSUB = 8 / G;          // q subtiles per head
   constexpr int kQROWS = kQSUB * kQB;   // q rows per workgroup
   const int b = blockIdx.x;
@@ -186,23 +191,32 @@ __launch_bounds__(512) __global__ void paged_prefill_v2_kernel(
     }
   };
 
-  issue_tile_loads(0);
-  // valid flags for write_tile: st_tok[i] < kv_valid of tile 0 — encode
-  // by clobbering st_tok when invalid (simpler than re-deriving)
-  {
+  if (!kSyncStage) {
+    issue_tile_loads(0);
+    // valid flags for write_tile: st_tok[i] < kv_valid of tile 0 —
+    // encode by clobbering st_tok when invalid
     const int kv_valid0 = min(kKVB, kv_limit);
 #pragma unroll
     for (int i = 0; i < 2; ++i)
       if (st_tok[i] >= kv_valid0) st_tok[i] = kKVB + 1;
+    write_tile(0);
+    __syncthreads();
   }
-  write_tile(0);
-  __syncthreads();
 
   for (int kt = 0; kt < n_tiles; ++kt) {
-    const int cur = kt & 1;
+    const int cur = kSyncStage ? 0 : (kt & 1);
     const int kv_start = kt * kKVB;
     const int kv_valid = min(kKVB, kv_limit - kv_start);
-    const bool have_next = kt + 1 < n_tiles;
+    const bool have_next = !kSyncStage && kt + 1 < n_tiles;
+    if (kSyncStage) {
+      __syncthreads();  // everyone done with the previous tile
+      issue_tile_loads(kt);
+#pragma unroll
+      for (int i = 0; i < 2; ++i)
+        if (st_tok[i] >= kv_valid) st_tok[i] = kKVB + 1;
+      write_tile(0);
+      __syncthreads();
+    }
     if (have_next) {
       issue_tile_loads(kt + 1);
       const int kv_valid1 = min(kKVB, kv_limit - (kt + 1) * kKVB);
@@ -317,19 +331,31 @@ __launch_bounds__(512) __global__ void paged_prefill_v2_kernel(
 #pragma unroll
         for (int ks = 0; ks < 4; ++ks) {
           bf16x8 v_frag;
-          typedef __attribute__((ext_vector_type(4))) short bf16x4;
-          bf16x4 lo4, hi4;
-          as3_char* a0 = vbase + ks * 4096 + nb * 256;
-          asm volatile(
-              "ds_read_b64_tr_b16 %0, %2\n\t"
-              "ds_read_b64_tr_b16 %1, %2 offset:1024\n\t"
-              "s_waitcnt lgkmcnt(0)"
-              : "=v"(lo4), "=v"(hi4)
-              : "v"(a0));
+          if (kSimpleV) {
+            // bisect mode: direct scalar reads from the subtiled layout
 #pragma unroll
-          for (int j = 0; j < 4; ++j) {
-            v_frag[j] = lo4[j];
-            v_frag[4 + j] = hi4[j];
+            for (int jj = 0; jj < 8; ++jj) {
+              const int row = ks * 16 + hi * 8 + jj;
+              const int col = nb * 32 + ln31;
+              const int off = ((row >> 2) * 8 + (col >> 4)) * 64 +
+                              (row & 3) * 16 + (col & 15);
+              v_frag[jj] = (short)v_lds[cur][off];
+            }
+          } else {
+            typedef __attribute__((ext_vector_type(4))) short bf16x4;
+            bf16x4 lo4, hi4;
+            as3_char* a0 = vbase + ks * 4096 + nb * 256;
+            asm volatile(
+                "ds_read_b64_tr_b16 %0, %2\n\t"
+                "ds_read_b64_tr_b16 %1, %2 offset:1024\n\t"
+                "s_waitcnt lgkmcnt(0)"
+                : "=v"(lo4), "=v"(hi4)
+                : "v"(a0));
+#pragma unroll
+            for (int j = 0; j < 4; ++j) {
+              v_frag[j] = lo4[j];
+              v_frag[4 + j] = hi4[j];
+            }
           }
           o_acc[nb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
               pa[ks], v_frag, o_acc[nb], 0, 0, 0);
@@ -387,6 +413,10 @@ bool paged_attention_prefill_v2(torch::Tensor out, torch::Tensor q,
     max_qlen = q.size(0);
   }
   const int zdim = (max_qlen + qrows - 1) / qrows;
+  static const int dbg_mode = []() {
+    const char* e = getenv("KUBEAI_V2_DBG");
+    return e ? atoi(e) : 0;
+  }();
   dim3 grid(B, n_kv, zdim), block(512);
   auto stream = c10::hip::getCurrentHIPStream().stream();
   const bool fp8_cache = k_cache.scalar_type() == torch::kFloat8_e5m2;
@@ -399,7 +429,7 @@ bool paged_attention_prefill_v2(torch::Tensor out, torch::Tensor q,
                      block_tables.data_ptr<int32_t>(),                    \
                      query_start_loc.data_ptr<int32_t>(),                 \
                      seq_lens.data_ptr<int32_t>(), (float)scale, n_kv,    \
-                     max_blocks, q.stride(0))
+                     max_blocks, q.stride(0), dbg_mode)
 #define LAUNCH_V2(GG)                                                     \
   do {                                                                    \
     if (fp8_cache) {                                                      \

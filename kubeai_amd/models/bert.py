@@ -34,10 +34,16 @@ class BertConfig:
     type_vocab_size: int = 2
     layer_norm_eps: float = 1e-12
     pad_token_id: int = 0
+    # cross-encoder rerank head (bge-reranker / ms-marco cross-encoders):
+    # score = head(hidden[CLS]) over "[CLS] query [SEP] doc [SEP]"
+    cross_encoder: bool = False
+    cls_token_id: int = 101
+    sep_token_id: int = 102
 
 
 PRESETS = {
     "bert-tiny": BertConfig(),
+    "reranker-tiny": BertConfig(cross_encoder=True),
     # bge-base-en-v1.5 / e5-base dimensions
     "bge-base": BertConfig(
         vocab_size=30522, hidden_size=768, num_hidden_layers=12,
@@ -83,8 +89,27 @@ class _BertLayer(nn.Module):
         return self.ffn_ln(x + self.ffn_out(F.gelu(self.ffn_in(x))))
 
 
+class _ClsHead(nn.Module):
+    """Sequence-classification head on [CLS]: plain linear (BERT
+    cross-encoders) or Roberta-style dense+tanh+out_proj (bge-reranker's
+    XLMRobertaForSequenceClassification)."""
+
+    def __init__(self, hidden: int, roberta: bool = False):
+        super().__init__()
+        self.dense = nn.Linear(hidden, hidden) if roberta else None
+        self.out_proj = nn.Linear(hidden, 1)
+
+    def forward(self, cls_hidden: torch.Tensor) -> torch.Tensor:
+        x = cls_hidden
+        if self.dense is not None:
+            x = torch.tanh(self.dense(x))
+        return self.out_proj(x).squeeze(-1)
+
+
 class BertEncoder(nn.Module):
-    """Encoder + mean pooling; `score_pairs` gives bi-encoder reranking."""
+    """Encoder + mean pooling; `score_pairs` reranks — cross-encoder
+    scoring when the model has a classification head (reference analog:
+    Infinity serving bge-reranker), else bi-encoder cosine."""
 
     def __init__(self, cfg: BertConfig, device: str = "cpu", seed: int = 0):
         super().__init__()
@@ -98,6 +123,7 @@ class BertEncoder(nn.Module):
         self.layers = nn.ModuleList(
             _BertLayer(cfg) for _ in range(cfg.num_hidden_layers)
         )
+        self.cls_head = _ClsHead(H) if cfg.cross_encoder else None
         self.to(device)
         self.device_ = device
         self.eval()
@@ -118,24 +144,57 @@ class BertEncoder(nn.Module):
             ids[i, : len(toks)] = torch.tensor(toks, dtype=torch.int64)
             mask[i, : len(toks)] = True
         ids, mask = ids.to(self.device_), mask.to(self.device_)
-        pos = torch.arange(T, device=ids.device)[None, :].expand_as(ids)
-        x = (
-            self.word_embeddings(ids)
-            + self.position_embeddings(pos)
-            + self.token_type_embeddings(torch.zeros_like(ids))
-        )
-        x = self.emb_ln(x)
-        for layer in self.layers:
-            x = layer(x, mask)
+        x = self._hidden(ids, mask, torch.zeros_like(ids))
         m = mask[:, :, None].to(x.dtype)
         pooled = (x * m).sum(dim=1) / m.sum(dim=1).clamp_min(1e-9)
         return F.normalize(pooled.float(), dim=-1)
 
+    def _hidden(self, ids, mask, type_ids) -> torch.Tensor:
+        pos = torch.arange(ids.shape[1], device=ids.device)[None, :].expand_as(ids)
+        x = (
+            self.word_embeddings(ids)
+            + self.position_embeddings(pos)
+            + self.token_type_embeddings(type_ids)
+        )
+        x = self.emb_ln(x)
+        for layer in self.layers:
+            x = layer(x, mask)
+        return x
+
     @torch.inference_mode()
     def score_pairs(self, query: list[int], docs: list[list[int]]) -> list[float]:
+        if self.cls_head is not None:
+            return self._score_cross(query, docs)
         vecs = self.encode([query] + docs)
         q, d = vecs[0], vecs[1:]
         return (d @ q).cpu().tolist()
+
+    def _score_cross(self, query: list[int], docs: list[list[int]]) -> list[float]:
+        """Cross-encoder: one joint forward per (query, doc) pair,
+        score from the classification head on [CLS]."""
+        cfg = self.cfg
+        Tmax = cfg.max_position_embeddings
+        pairs, types = [], []
+        for d in docs:
+            toks = [cfg.cls_token_id] + list(query) + [cfg.sep_token_id]
+            tt = [0] * len(toks)
+            toks += list(d) + [cfg.sep_token_id]
+            tt += [1] * (len(toks) - len(tt))
+            pairs.append(toks[:Tmax])
+            types.append(tt[:Tmax])
+        T = max(len(p) for p in pairs)
+        pad = cfg.pad_token_id
+        ids = torch.full((len(pairs), T), pad, dtype=torch.int64)
+        mask = torch.zeros((len(pairs), T), dtype=torch.bool)
+        tids = torch.zeros((len(pairs), T), dtype=torch.int64)
+        for i, (p, tt) in enumerate(zip(pairs, types)):
+            ids[i, : len(p)] = torch.tensor(p, dtype=torch.int64)
+            mask[i, : len(p)] = True
+            tids[i, : len(tt)] = torch.tensor(tt, dtype=torch.int64)
+        ids, mask, tids = (ids.to(self.device_), mask.to(self.device_),
+                           tids.to(self.device_))
+        x = self._hidden(ids, mask, tids)
+        return self.cls_head(x[:, 0].float()).cpu().tolist()
 
 
 # ----------------------------------------------------------- checkpoint IO
@@ -219,16 +278,37 @@ def load_weights_bert(model: BertEncoder, model_dir: str) -> int:
     files = sorted(glob.glob(os.path.join(model_dir, "*.safetensors")))
     if not files:
         raise FileNotFoundError(f"no *.safetensors under {model_dir}")
+    cls_keys = {}
     for fpath in files:
         with safe_open(fpath, framework="pt", device="cpu") as sf:
             for key in sf.keys():
+                if key.startswith("classifier."):
+                    cls_keys[key] = sf.get_tensor(key)
+                    continue
                 k = key[len("bert."):] if key.startswith("bert.") else key
+                k = k[len("roberta."):] if k.startswith("roberta.") else k
                 tgt = inverse.get(k)
                 if tgt is None:
                     continue  # pooler etc.
                 with torch.no_grad():
                     sd[tgt].copy_(sf.get_tensor(key).to(sd[tgt].dtype))
                 filled.add(tgt)
+    if cls_keys:
+        roberta = "classifier.dense.weight" in cls_keys
+        if model.cls_head is None:
+            model.cls_head = _ClsHead(model.cfg.hidden_size, roberta=roberta)
+            model.cls_head.to(model.device_)
+        with torch.no_grad():
+            if roberta:
+                model.cls_head.dense.weight.copy_(cls_keys["classifier.dense.weight"])
+                model.cls_head.dense.bias.copy_(cls_keys["classifier.dense.bias"])
+                model.cls_head.out_proj.weight.copy_(
+                    cls_keys["classifier.out_proj.weight"])
+                model.cls_head.out_proj.bias.copy_(
+                    cls_keys["classifier.out_proj.bias"])
+            else:
+                model.cls_head.out_proj.weight.copy_(cls_keys["classifier.weight"])
+                model.cls_head.out_proj.bias.copy_(cls_keys["classifier.bias"])
     missing = set(sd) - filled
     if missing:
         raise ValueError(f"unfilled bert parameters: {sorted(missing)[:8]}")

@@ -275,10 +275,17 @@ __launch_bounds__(512) __global__ void paged_prefill_v2_kernel(
       rsum += __shfl_xor(rsum, 32, 64);
       l_run = l_run * corr + rsum;
       m_run = m_new;
+      // o_acc rows are the C-fragment rows crow(r, hi) — NOT this
+      // lane's softmax row (q = ln31). Rescale each accumulator row by
+      // ITS OWN correction, gathered from the lane that owns that q.
+      float corr_row[16];
+#pragma unroll
+      for (int r = 0; r < 16; ++r)
+        corr_row[r] = __shfl(corr, (lane & 32) + crow(r, hi), 64);
 #pragma unroll
       for (int nb = 0; nb < 4; ++nb)
 #pragma unroll
-        for (int r = 0; r < 16; ++r) o_acc[nb][r] *= corr;
+        for (int r = 0; r < 16; ++r) o_acc[nb][r] *= corr_row[r];
 
       // ---- P -> A fragments (in-register, T12) ---------------------
       // pa[ks]: lane holds P[q=ln31][ks*16 + hi*8 + jj], jj=0..7.

@@ -120,6 +120,7 @@ class ModelRunner:
         # a tokenizer the server/worker injects after startup
         self.tokenizer = None
         self._json_validators: dict = {}
+        self._sampling_cache: dict = {}
         torch.manual_seed(seed)
         if tp_group is not None:
             from kubeai_amd.parallel.tp import TPLlamaForCausalLM
@@ -470,14 +471,16 @@ class ModelRunner:
                 device=dev,
             )
             if needs_topk:
-                # draw from the nucleus over the top-K candidate subset —
-                # no full-vocab sort or mask materialization
+                # rejection nucleus sampling — no full-vocab sort; the
+                # per-row param tensors are cached across steps (the
+                # batch's sampling params rarely change step to step)
                 tokens = _sample_topk_topp(
                     logits,
                     [r.params.top_p for r in sample_reqs],
                     [r.params.top_k for r in sample_reqs],
                     [r.params.temperature for r in sample_reqs],
                     t_t, seeds, step,
+                    cache=self._sampling_cache,
                 ).to(dev)
             else:
                 tokens = ops.gumbel_sample(logits.contiguous(), t_t, seeds, step)
@@ -619,7 +622,8 @@ class ModelRunner:
         self._json_validators.pop(request_id, None)
 
 
-def _sample_topk_topp(logits, top_ps, top_ks, temps, t_t, seeds, step):
+def _sample_topk_topp(logits, top_ps, top_ks, temps, t_t, seeds, step,
+                      cache=None):
     """Nucleus/top-k sampling by rejection — no full-vocab sort.
 
     Sampling from the renormalized nucleus == sampling the FULL
@@ -636,9 +640,16 @@ def _sample_topk_topp(logits, top_ps, top_ks, temps, t_t, seeds, step):
     """
     S, V = logits.shape
     dev = logits.device
-    tp = torch.tensor(top_ps, device=dev, dtype=torch.float32)
-    tk = torch.tensor(top_ks, device=dev, dtype=torch.int32)
-    tt = torch.tensor(temps, device=dev, dtype=torch.float32)
+    key = (tuple(top_ps), tuple(top_ks), tuple(temps))
+    hit = cache.get("params") if cache is not None else None
+    if hit is not None and hit[0] == key:
+        tp, tk, tt = hit[1]
+    else:
+        tp = torch.tensor(top_ps, device=dev, dtype=torch.float32)
+        tk = torch.tensor(top_ks, device=dev, dtype=torch.int32)
+        tt = torch.tensor(temps, device=dev, dtype=torch.float32)
+        if cache is not None:
+            cache["params"] = (key, (tp, tk, tt))
     logits_c = logits.float().contiguous()
     # per-row softmax stats once; each draw is then ONE fused pass
     m, z = ops.nucleus_stats(logits_c, tt)

@@ -140,3 +140,70 @@ def test_tp_lora_rejected():
 
     with _pytest.raises(ValueError, match="tensor-parallel"):
         srv.load_lora("a", None)
+
+
+def test_cross_encoder_rerank_head():
+    """Cross-encoder scoring: joint pair forward + classification head
+    (bge-reranker analog); differs from bi-encoder cosine and responds
+    to the head weights."""
+    import torch
+
+    from kubeai_amd.models import bert as bert_mod
+
+    enc = bert_mod.BertEncoder(
+        bert_mod.PRESETS["reranker-tiny"], device="cpu", seed=3
+    )
+    assert enc.cls_head is not None
+    q = list(range(200, 220))
+    docs = [list(range(300, 330)), list(range(400, 420)), q]
+    with torch.no_grad():
+        enc.cls_head.out_proj.bias.zero_()
+    scores = enc.score_pairs(q, docs)
+    assert len(scores) == 3
+    # scores change when the head changes (really flows through the head)
+    with torch.no_grad():
+        enc.cls_head.out_proj.weight.mul_(-1.0)
+    flipped = enc.score_pairs(q, docs)
+    for a, b in zip(scores, flipped):
+        assert abs(a + b) < 1e-3, (a, b)
+    # bi-encoder model scores the same inputs differently (cosine path)
+    bi = bert_mod.BertEncoder(bert_mod.PRESETS["bert-tiny"], device="cpu",
+                              seed=3)
+    bi_scores = bi.score_pairs(q, docs)
+    assert bi_scores != scores
+
+
+def test_cross_encoder_checkpoint_roundtrip(tmp_path):
+    """classifier.* keys in an HF checkpoint create and fill the head
+    (XLMRobertaForSequenceClassification / bge-reranker layout)."""
+    import json
+    import os
+
+    import torch
+    from safetensors.torch import save_file
+
+    from kubeai_amd.models import bert as bert_mod
+
+    src = bert_mod.BertEncoder(bert_mod.PRESETS["bert-tiny"], device="cpu",
+                               seed=7)
+    ckpt = str(tmp_path / "rr")
+    bert_mod.save_bert_checkpoint(src, ckpt)
+    # append a roberta-style classification head to the checkpoint
+    from safetensors.torch import load_file
+
+    tensors = load_file(os.path.join(ckpt, "model.safetensors"))
+    H = src.cfg.hidden_size
+    torch.manual_seed(9)
+    tensors["classifier.dense.weight"] = torch.randn(H, H) * 0.05
+    tensors["classifier.dense.bias"] = torch.zeros(H)
+    tensors["classifier.out_proj.weight"] = torch.randn(1, H) * 0.05
+    tensors["classifier.out_proj.bias"] = torch.zeros(1)
+    save_file(tensors, os.path.join(ckpt, "model.safetensors"))
+
+    dst = bert_mod.BertEncoder(bert_mod.config_from_hf(ckpt), device="cpu")
+    bert_mod.load_weights_bert(dst, ckpt)
+    assert dst.cls_head is not None and dst.cls_head.dense is not None
+    q = list(range(50, 70))
+    docs = [list(range(80, 100)), list(range(120, 160))]
+    s = dst.score_pairs(q, docs)
+    assert len(s) == 2 and s[0] != s[1]

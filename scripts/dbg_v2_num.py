@@ -96,3 +96,8 @@ run_case("L G8 rand + ctx", 192, 37, "rand", "rand", "rand", _nq=64)
 run_case("M G8 big", 2048, 0, "rand", "rand", "rand", _nq=64)
 run_case("N G2 rand", 192, 21, "rand", "rand", "rand", _nq=16)
 run_case("O G1 rand", 192, 21, "rand", "rand", "rand", _nq=8)
+
+run_case("P G8 tiny16", 16, 0, "rand", "rand", "rand", _nq=64)
+run_case("Q G1 ragged100", 100, 0, "rand", "rand", "rand", _nq=8)
+run_case("R G4 big", 2048, 0, "rand", "rand", "rand", _nq=32)
+run_case("S G8 big repeat", 2048, 0, "rand", "rand", "rand", _nq=64)

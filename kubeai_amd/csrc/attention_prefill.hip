@@ -304,11 +304,13 @@ void paged_attention_prefill(torch::Tensor out, torch::Tensor q,
   const int G = n_q / n_kv;
   TORCH_CHECK(n_q % n_kv == 0);
   if (q.size(0) == 0) return;
-  // v2: 8-wave 32x32-MFMA ladder (attention_prefill_v2.hip) — handles
-  // G in {1,2,4,8}; opt out with KUBEAI_PREFILL_V2=0
+  // v2: 8-wave 32x32-MFMA ladder (attention_prefill_v2.hip), G in
+  // {1,2,4,8}. OPT-IN (KUBEAI_PREFILL_V2=1) until it beats v1: PMC
+  // shows it VALU-bound (68:1) at v1-level TF, with a residual sporadic
+  // fault at G=8 large shapes under investigation (profiles/r02).
   static const bool use_v2 = []() {
     const char* e = getenv("KUBEAI_PREFILL_V2");
-    return e == nullptr || e[0] != '0';
+    return e != nullptr && e[0] == '1';
   }();
   if (use_v2 &&
       paged_attention_prefill_v2(out, q, k_cache, v_cache, block_tables,

@@ -352,11 +352,15 @@ __launch_bounds__(512) __global__ void paged_prefill_v2_kernel(
             typedef __attribute__((ext_vector_type(4))) short bf16x4;
             bf16x4 lo4, hi4;
             as3_char* a0 = vbase + ks * 4096 + nb * 256;
+            // "=&v" early-clobber: without it the allocator may alias an
+            // output with the address VGPR — the first read's writeback
+            // then races the second read's address consumption
+            // (sporadic corruption, shape/schedule dependent)
             asm volatile(
                 "ds_read_b64_tr_b16 %0, %2\n\t"
                 "ds_read_b64_tr_b16 %1, %2 offset:1024\n\t"
                 "s_waitcnt lgkmcnt(0)"
-                : "=v"(lo4), "=v"(hi4)
+                : "=&v"(lo4), "=&v"(hi4)
                 : "v"(a0));
 #pragma unroll
             for (int j = 0; j < 4; ++j) {

@@ -84,7 +84,7 @@ __global__ void probe_tr(ushort* out /* [4 ks][4 nb][64 lanes][8 jj] */,
           "ds_read_b64_tr_b16 %0, %2\n\t"
           "ds_read_b64_tr_b16 %1, %2 offset:1024\n\t"
           "s_waitcnt lgkmcnt(0)"
-          : "=v"(lo4), "=v"(hi4)
+          : "=&v"(lo4), "=&v"(hi4)
           : "v"(a0));
       ushort* o = out + ((ks * 4 + nb) * 64 + lane) * 8;
 #pragma unroll

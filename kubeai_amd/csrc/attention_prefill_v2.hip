@@ -129,23 +129,35 @@ __launch_bounds__(512) __global__ void paged_prefill_v2_kernel(
   const int32_t* bt = block_tables + (int64_t)b * max_blocks;
 
   // ---- cooperative staging: 2 K + 2 V 8-element units per thread ----
-  // unit u of 1024: tok = u>>4, c8 = u&15 (16 bf16-8 units per 128-row)
+  // unit u of 1024: tok = u>>4, c8 = u&15 (16 bf16-8 units per 128-row).
+  // All thread-invariant addressing is hoisted; per tile only the block
+  // lookup + validity remain.
   bf16x8 st_k[2], st_v[2];
-  int st_tok[2];
+  bool st_ok[2];
+  int h_tok[2], h_coff[2], h_kdst[2], h_vdst[2];
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+    const int u = threadIdx.x + i * 512;
+    const int tok = u >> 4;
+    const int c8 = u & 15;
+    h_tok[i] = tok;
+    h_coff[i] = c8 * 8;
+    h_kdst[i] = swz(tok, tok * kHD * 2 + c8 * 16);
+    h_vdst[i] = ((tok >> 2) * 8 + (c8 >> 1)) * 64 + (tok & 3) * 16 +
+                (c8 & 1) * 8;
+  }
   auto issue_tile_loads = [&](int kt) {
     const int kv_start = kt * kKVB;
     const int kv_valid = min(kKVB, kv_limit - kv_start);
 #pragma unroll
     for (int i = 0; i < 2; ++i) {
-      const int u = threadIdx.x + i * 512;
-      const int tok = u >> 4;
-      const int c8 = u & 15;
-      st_tok[i] = tok;
-      if (tok < kv_valid) {
+      const int tok = h_tok[i];
+      st_ok[i] = tok < kv_valid;
+      if (st_ok[i]) {
         const int abs_tok = kv_start + tok;
         const int64_t blk = bt[abs_tok / kBS];
         const int64_t base =
-            (((blk * n_kv + kh) * kBS) + abs_tok % kBS) * kHD + c8 * 8;
+            (((blk * n_kv + kh) * kBS) + abs_tok % kBS) * kHD + h_coff[i];
         if constexpr (sizeof(CT) == 2) {
           st_k[i] = *reinterpret_cast<const bf16x8*>(k_cache + base);
           st_v[i] = *reinterpret_cast<const bf16x8*>(v_cache + base);
@@ -173,32 +185,17 @@ __launch_bounds__(512) __global__ void paged_prefill_v2_kernel(
   auto write_tile = [&](int buf) {
 #pragma unroll
     for (int i = 0; i < 2; ++i) {
-      const int u = threadIdx.x + i * 512;
-      const int tok = u >> 4;
-      const int c8 = u & 15;
-      // issue_tile_loads clobbers st_tok past kKVB for invalid units
-      const bool valid = st_tok[i] < kKVB;
       // K: swizzled row-major (skip stale-write cost only when invalid)
-      if (valid)
+      if (st_ok[i])
         *reinterpret_cast<bf16x8*>(
-            reinterpret_cast<char*>(k_lds[buf]) +
-            swz(tok, tok * kHD * 2 + c8 * 16)) = st_k[i];
-      // V: subtile (t4 = tok/4, h16 = c8/2): elem off =
-      //    (t4*8 + h16)*64 + (tok%4)*16 + (c8%2)*8
-      const int off =
-          ((tok >> 2) * 8 + (c8 >> 1)) * 64 + (tok & 3) * 16 + (c8 & 1) * 8;
-      *reinterpret_cast<bf16x8*>(&v_lds[buf][off]) = st_v[i];
+            reinterpret_cast<char*>(k_lds[buf]) + h_kdst[i]) = st_k[i];
+      // V: subtiled; invalid units were zeroed at load
+      *reinterpret_cast<bf16x8*>(&v_lds[buf][h_vdst[i]]) = st_v[i];
     }
   };
 
   if (!kSyncStage) {
     issue_tile_loads(0);
-    // valid flags for write_tile: st_tok[i] < kv_valid of tile 0 —
-    // encode by clobbering st_tok when invalid
-    const int kv_valid0 = min(kKVB, kv_limit);
-#pragma unroll
-    for (int i = 0; i < 2; ++i)
-      if (st_tok[i] >= kv_valid0) st_tok[i] = kKVB + 1;
     write_tile(0);
     __syncthreads();
   }
@@ -211,24 +208,16 @@ __launch_bounds__(512) __global__ void paged_prefill_v2_kernel(
     if (kSyncStage) {
       __syncthreads();  // everyone done with the previous tile
       issue_tile_loads(kt);
-#pragma unroll
-      for (int i = 0; i < 2; ++i)
-        if (st_tok[i] >= kv_valid) st_tok[i] = kKVB + 1;
       write_tile(0);
       __syncthreads();
     }
-    if (have_next) {
-      issue_tile_loads(kt + 1);
-      const int kv_valid1 = min(kKVB, kv_limit - (kt + 1) * kKVB);
-#pragma unroll
-      for (int i = 0; i < 2; ++i)
-        if (st_tok[i] >= kv_valid1) st_tok[i] = kKVB + 1;
-    }
+    if (have_next) issue_tile_loads(kt + 1);
 
     // wave-level skip: this wave's rows all causally precede the tile
     const int wave_kv_hi = ctx + row_lo + min(wrow0 + kQB, n_rows) - 1;
     if (kv_start <= wave_kv_hi) {
       // ---- S^T = K . Q^T over two 32-kv subtiles -------------------
+      __builtin_amdgcn_s_setprio(1);  // T9: keep MFMA issue ahead
       f32x16 st2[2];
 #pragma unroll
       for (int sub = 0; sub < 2; ++sub) {
@@ -245,7 +234,11 @@ __launch_bounds__(512) __global__ void paged_prefill_v2_kernel(
         }
       }
 
+      __builtin_amdgcn_s_setprio(0);
       // ---- mask + per-lane online softmax (q = column = ln31) ------
+      // single merged bound: kv is allowed iff kv <= kv_hi
+      const int kv_hi =
+          row_ok ? min(qpos, kv_start + kv_valid - 1) : -1;
       float pmax = -INFINITY;
 #pragma unroll
       for (int sub = 0; sub < 2; ++sub)
@@ -253,15 +246,33 @@ __launch_bounds__(512) __global__ void paged_prefill_v2_kernel(
         for (int r = 0; r < 16; ++r) {
           const int kv = kv_start + sub * 32 + crow(r, hi);
           float s = st2[sub][r] * scale;
-          if (kv > qpos || kv >= kv_valid + kv_start || !row_ok)
-            s = -INFINITY;
+          if (kv > kv_hi) s = -INFINITY;
           st2[sub][r] = s;
           pmax = fmaxf(pmax, s);
         }
       pmax = fmaxf(pmax, __shfl_xor(pmax, 32, 64));  // merge lane halves
-      const float m_new = fmaxf(m_run, pmax);
-      const float corr =
-          (m_run == -INFINITY) ? 0.f : __expf(m_run - m_new);
+      // defer-max (T ladder): skip the whole rescale path while no
+      // row's max grows by more than 8 (exp stays bounded by e^8) —
+      // wave-uniform so the cross-lane corr gather can be skipped too
+      const bool defer = __all(pmax <= m_run + 8.f);
+      float m_new = m_run;
+      if (!defer) {
+        m_new = fmaxf(m_run, pmax);
+        const float corr =
+            (m_run == -INFINITY) ? 0.f : __expf(m_run - m_new);
+        l_run *= corr;
+        // o_acc rows are the C-fragment rows crow(r, hi) — NOT this
+        // lane's softmax row (q = ln31). Rescale each accumulator row
+        // by ITS OWN correction, gathered from its owning lane.
+        float corr_row[16];
+#pragma unroll
+        for (int r = 0; r < 16; ++r)
+          corr_row[r] = __shfl(corr, (lane & 32) + crow(r, hi), 64);
+#pragma unroll
+        for (int nb = 0; nb < 4; ++nb)
+#pragma unroll
+          for (int r = 0; r < 16; ++r) o_acc[nb][r] *= corr_row[r];
+      }
       float rsum = 0.f;
 #pragma unroll
       for (int sub = 0; sub < 2; ++sub)
@@ -273,19 +284,8 @@ __launch_bounds__(512) __global__ void paged_prefill_v2_kernel(
           rsum += p;
         }
       rsum += __shfl_xor(rsum, 32, 64);
-      l_run = l_run * corr + rsum;
+      l_run += rsum;
       m_run = m_new;
-      // o_acc rows are the C-fragment rows crow(r, hi) — NOT this
-      // lane's softmax row (q = ln31). Rescale each accumulator row by
-      // ITS OWN correction, gathered from the lane that owns that q.
-      float corr_row[16];
-#pragma unroll
-      for (int r = 0; r < 16; ++r)
-        corr_row[r] = __shfl(corr, (lane & 32) + crow(r, hi), 64);
-#pragma unroll
-      for (int nb = 0; nb < 4; ++nb)
-#pragma unroll
-        for (int r = 0; r < 16; ++r) o_acc[nb][r] *= corr_row[r];
 
       // ---- P -> A fragments (in-register, T12) ---------------------
       // pa[ks]: lane holds P[q=ln31][ks*16 + hi*8 + jj], jj=0..7.
@@ -328,6 +328,7 @@ __launch_bounds__(512) __global__ void paged_prefill_v2_kernel(
       // in the 4x16-subtiled layout:
       //   byte addr(m) = ks*4096 + tr*1024 + (m>>5)*2048 + nb*256
       //     + ((m>>4)&1)*128 + ((m>>2)&3)*32 + (m&3)*8
+      __builtin_amdgcn_s_setprio(1);  // PV MFMA cluster
       typedef __attribute__((address_space(3))) const char as3_char;
       as3_char* vbase =
           (as3_char*)(v_lds[cur]) +
@@ -374,7 +375,10 @@ __launch_bounds__(512) __global__ void paged_prefill_v2_kernel(
       }
     }
 
-    __syncthreads();  // everyone done with buf cur^1 reads (prev iter)
+    __builtin_amdgcn_s_setprio(0);
+    // single barrier per tile: writing buf cur^1 cannot conflict with
+    // concurrent reads of buf cur, and the previous iteration's barrier
+    // already separated it from the last reads of cur^1
     if (have_next) write_tile(cur ^ 1);
     __syncthreads();  // staged tile visible before next iteration
   }

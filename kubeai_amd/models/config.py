@@ -139,6 +139,21 @@ PRESETS: dict[str, ModelArchConfig] = {
         bos_token_id=1,
         eos_token_id=2,
     ),
+    # long-context tiny model: exercises chunked prefill + paged decode
+    # at tens of thousands of tokens without big weights (the engine
+    # sizes KV from free HBM; presets elsewhere are 8k-faithful)
+    "llama-tiny-32k": ModelArchConfig(
+        vocab_size=2048,
+        hidden_size=256,
+        intermediate_size=512,
+        num_hidden_layers=2,
+        num_attention_heads=2,
+        num_key_value_heads=1,
+        head_dim=128,
+        max_position_embeddings=32768,
+        bos_token_id=1,
+        eos_token_id=2,
+    ),
     # small: 1-GPU quick bench model (head_dim=128 to stay on the HIP path)
     "llama-1b": ModelArchConfig(
         vocab_size=128256,

@@ -297,3 +297,38 @@ def test_hf_checkpoint_dir_e2e_gpu(tmp_path):
                                  seed=999))
     got = gen(dst, "dst")
     assert got == want
+
+
+def test_long_context_generation_gpu():
+    """Long-context serving path: a 20k-token prompt prefills in chunks
+    through the paged cache and decodes correctly (presets
+    were 8k-capped with no long run)."""
+    eng = LLMEngine(
+        EngineConfig(model="llama-tiny-32k", device="cuda",
+                     max_model_len=24576, max_num_batched_tokens=8192,
+                     gpu_memory_utilization=0.5)
+    )
+    prompt = [(7 + 13 * i) % 1900 + 100 for i in range(20000)]
+    eng.add_request(prompt, SamplingParams(max_tokens=8, ignore_eos=True),
+                    request_id="long")
+    out = None
+    for _ in range(200):
+        if not eng.has_work():
+            break
+        for o in eng.step():
+            if o.finished:
+                out = o
+    assert out is not None and len(out.output_token_ids) == 8
+    # prefix-cache equivalence at long length: resubmitting the same
+    # prompt hits the cache and reproduces the same greedy tokens
+    eng.add_request(prompt, SamplingParams(max_tokens=8, ignore_eos=True),
+                    request_id="long2")
+    out2 = None
+    for _ in range(200):
+        if not eng.has_work():
+            break
+        for o in eng.step():
+            if o.finished:
+                out2 = o
+    assert out2.num_cached_tokens > 15000
+    assert out2.output_token_ids == out.output_token_ids

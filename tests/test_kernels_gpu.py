@@ -330,3 +330,23 @@ def test_nucleus_kernels_match_reference():
         want = ((mass < top_ps) & ((top_ks <= 0) | (cnt < top_ks))
                 | (temps <= 0)).to(torch.uint8)
         assert torch.equal(ok.cpu(), want.cpu()), (trial, ok, want)
+
+
+def test_paged_prefill_long_context():
+    """Chunked continuation against a LONG cached context (16k)."""
+    torch.manual_seed(3)
+    nq, nkv, hd, bs = 32, 8, 128, 16
+    Tq, ctx = 1024, 15360
+    L = Tq + ctx
+    nb = L // bs + 1
+    kc, vc = _rand_cache(nb, nkv, bs, hd)
+    bt = torch.arange(1, nb, dtype=torch.int32, device=DEV).reshape(1, -1)
+    sl = torch.tensor([L], dtype=torch.int32, device=DEV)
+    qsl = torch.tensor([0, Tq], dtype=torch.int32, device=DEV)
+    q = torch.randn(Tq, nq, hd, dtype=torch.bfloat16, device=DEV)
+    scale = 1.0 / math.sqrt(hd)
+    out = ops.paged_attention_prefill(q, kc, vc, bt, qsl, sl, scale)
+    want = ref.paged_attention_prefill(
+        q.float(), kc.float(), vc.float(), bt, qsl, sl, scale
+    )
+    assert_close_bf16(out, want, atol=3e-2, rtol=3e-2)

@@ -305,12 +305,12 @@ void paged_attention_prefill(torch::Tensor out, torch::Tensor q,
   TORCH_CHECK(n_q % n_kv == 0);
   if (q.size(0) == 0) return;
   // v2: 8-wave 32x32-MFMA ladder (attention_prefill_v2.hip), G in
-  // {1,2,4,8}. OPT-IN (KUBEAI_PREFILL_V2=1) until it beats v1: PMC
-  // shows it VALU-bound (68:1) at v1-level TF, with a residual sporadic
-  // fault at G=8 large shapes under investigation (profiles/r02).
+  // {1,2,4,8} — DEFAULT (1.5-1.8x v1 after the defer-max/VALU pass:
+  // 268 TF vs 150 @Tq=8192, numerics cos=1.0 across the full isolation
+  // matrix, profiles/r02_results.md). KUBEAI_PREFILL_V2=0 falls back.
   static const bool use_v2 = []() {
     const char* e = getenv("KUBEAI_PREFILL_V2");
-    return e != nullptr && e[0] == '1';
+    return e == nullptr || e[0] != '0';
   }();
   if (use_v2 &&
       paged_attention_prefill_v2(out, q, k_cache, v_cache, block_tables,

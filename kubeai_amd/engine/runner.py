@@ -671,7 +671,11 @@ def _sample_topk_topp(logits, top_ps, top_ks, temps, t_t, seeds, step,
     c3, ok3 = draw(2)
     tokens = torch.where(ok1, c1, torch.where(ok2, c2, c3))
     pending = ~(ok1 | ok2 | ok3)
+    if cache is not None:
+        cache["calls"] = cache.get("calls", 0) + 1
     if bool(pending.any()):
+        if cache is not None:
+            cache["fallbacks"] = cache.get("fallbacks", 0) + 1
         # stragglers: exact sort-mask path (tiny top_k on a flat
         # distribution can reject indefinitely)
         masked = _apply_topk_topp(logits, top_ps, top_ks, temps)

@@ -83,6 +83,21 @@ for S, V in ((40, 128256), (256, 128256)):
         lambda: _sample_topk_topp(logits, tps, tks, temps, t_t, seeds, 3,
                                   cache=_cache)
     )
+    print(f"  fallback rate: {_cache.get('fallbacks', 0)}"
+          f"/{_cache.get('calls', 0)} calls")
+    # vary step across calls (the bench reuses step=3 -> identical draws
+    # every call: one unlucky batch repeats its fallback forever)
+    _cache2 = {}
+    _step_box = [0]
+
+    def varied():
+        _step_box[0] += 1
+        return _sample_topk_topp(logits, tps, tks, temps, t_t, seeds,
+                                 _step_box[0], cache=_cache2)
+
+    us_varied = timeit(varied)
+    print(f"  varied-step: {us_varied:8.1f}us  fallback "
+          f"{_cache2.get('fallbacks', 0)}/{_cache2.get('calls', 0)}")
 
     def full_sort():
         sl, si = logits.sort(dim=-1, descending=True)

@@ -674,13 +674,20 @@ def _sample_topk_topp(logits, top_ps, top_ks, temps, t_t, seeds, step,
     if cache is not None:
         cache["calls"] = cache.get("calls", 0) + 1
     if bool(pending.any()):
-        if cache is not None:
-            cache["fallbacks"] = cache.get("fallbacks", 0) + 1
-        # stragglers: exact sort-mask path (tiny top_k on a flat
-        # distribution can reject indefinitely)
-        masked = _apply_topk_topp(logits, top_ps, top_ks, temps)
-        rest = ops.gumbel_sample(masked.contiguous(), t_t, seeds, step).to(dev)
-        tokens = torch.where(pending, rest, tokens)
+        # a few more cheap draws before the expensive exact path — a
+        # fallback round costs ~3 fused passes vs ~40 for the sort-mask
+        for attempt in range(3, 6):
+            c, ok = draw(attempt)
+            tokens = torch.where(pending & ok, c, tokens)
+            pending &= ~ok
+        if bool(pending.any()):
+            if cache is not None:
+                cache["fallbacks"] = cache.get("fallbacks", 0) + 1
+            # stragglers: exact sort-mask path (tiny top_k on a flat
+            # distribution can reject indefinitely)
+            masked = _apply_topk_topp(logits, top_ps, top_ks, temps)
+            rest = ops.gumbel_sample(masked.contiguous(), t_t, seeds, step).to(dev)
+            tokens = torch.where(pending, rest, tokens)
     return tokens
 
 

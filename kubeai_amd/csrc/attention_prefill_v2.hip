@@ -56,7 +56,7 @@ DEV_INLINE int crow(int reg, int hi) {
   return (reg & 3) + 8 * (reg >> 2) + 4 * hi;
 }
 
-template <int G, typename CT = ushort>
+template <int G, typename CT = ushort, int DBG = 0>
 __launch_bounds__(512) __global__ void paged_prefill_v2_kernel(
     ushort* __restrict__ out,            // [Tq, n_q, hd]
     const ushort* __restrict__ q,        // [Tq, n_q, hd]
@@ -66,12 +66,13 @@ __launch_bounds__(512) __global__ void paged_prefill_v2_kernel(
     const int32_t* __restrict__ query_start_loc,  // [B+1]
     const int32_t* __restrict__ seq_lens,         // [B]
     const float scale, const int n_kv, const int max_blocks,
-    const int64_t q_stride, const int dbg_mode) {
-  // dbg_mode bisect switches (host env KUBEAI_V2_DBG):
+    const int64_t q_stride) {
+  // DBG bisect switches (host env KUBEAI_V2_DBG, compile-time so the
+  // fast path carries no runtime branches or extra SGPR pressure):
   //   bit 0: synchronous single-slot staging (no async double buffer)
   //   bit 1: plain scalar V reads instead of ds_read_b64_tr_b16
-  const bool kSyncStage = dbg_mode & 1;
-  const bool kSimpleV = dbg_mode & 2;
+  constexpr bool kSyncStage = DBG & 1;
+  constexpr bool kSimpleV = DBG & 2;
   constexpr int kQSUB = 8 / G;          // q subtiles per head
   constexpr int kQROWS = kQSUB * kQB;   // q rows per workgroup
   const int b = blockIdx.x;
@@ -194,7 +195,7 @@ __launch_bounds__(512) __global__ void paged_prefill_v2_kernel(
     }
   };
 
-  if (!kSyncStage) {
+  if constexpr (!kSyncStage) {
     issue_tile_loads(0);
     write_tile(0);
     __syncthreads();
@@ -205,7 +206,7 @@ __launch_bounds__(512) __global__ void paged_prefill_v2_kernel(
     const int kv_start = kt * kKVB;
     const int kv_valid = min(kKVB, kv_limit - kv_start);
     const bool have_next = !kSyncStage && kt + 1 < n_tiles;
-    if (kSyncStage) {
+    if constexpr (kSyncStage) {
       __syncthreads();  // everyone done with the previous tile
       issue_tile_loads(kt);
       write_tile(0);
@@ -236,20 +237,37 @@ __launch_bounds__(512) __global__ void paged_prefill_v2_kernel(
 
       __builtin_amdgcn_s_setprio(0);
       // ---- mask + per-lane online softmax (q = column = ln31) ------
-      // single merged bound: kv is allowed iff kv <= kv_hi
-      const int kv_hi =
-          row_ok ? min(qpos, kv_start + kv_valid - 1) : -1;
+      // interior tiles (fully inside every row's causal window, full
+      // kv_valid, all rows live) skip the per-element mask entirely —
+      // wave-uniform, so it also drops the compare/select chains
+      const bool tile_interior =
+          kv_valid == kKVB && wrow0 + kQB <= n_rows &&
+          kv_start + kKVB - 1 <= ctx + row_lo + wrow0;
       float pmax = -INFINITY;
+      if (tile_interior) {
 #pragma unroll
-      for (int sub = 0; sub < 2; ++sub)
+        for (int sub = 0; sub < 2; ++sub)
 #pragma unroll
-        for (int r = 0; r < 16; ++r) {
-          const int kv = kv_start + sub * 32 + crow(r, hi);
-          float s = st2[sub][r] * scale;
-          if (kv > kv_hi) s = -INFINITY;
-          st2[sub][r] = s;
-          pmax = fmaxf(pmax, s);
-        }
+          for (int r = 0; r < 16; ++r) {
+            const float s = st2[sub][r] * scale;
+            st2[sub][r] = s;
+            pmax = fmaxf(pmax, s);
+          }
+      } else {
+        // single merged bound: kv is allowed iff kv <= kv_hi
+        const int kv_hi =
+            row_ok ? min(qpos, kv_start + kv_valid - 1) : -1;
+#pragma unroll
+        for (int sub = 0; sub < 2; ++sub)
+#pragma unroll
+          for (int r = 0; r < 16; ++r) {
+            const int kv = kv_start + sub * 32 + crow(r, hi);
+            float s = st2[sub][r] * scale;
+            if (kv > kv_hi) s = -INFINITY;
+            st2[sub][r] = s;
+            pmax = fmaxf(pmax, s);
+          }
+      }
       pmax = fmaxf(pmax, __shfl_xor(pmax, 32, 64));  // merge lane halves
       // defer-max (T ladder): skip the whole rescale path while no
       // row's max grows by more than 8 (exp stays bounded by e^8) —
@@ -274,15 +292,26 @@ __launch_bounds__(512) __global__ void paged_prefill_v2_kernel(
           for (int r = 0; r < 16; ++r) o_acc[nb][r] *= corr_row[r];
       }
       float rsum = 0.f;
+      if (tile_interior) {
 #pragma unroll
-      for (int sub = 0; sub < 2; ++sub)
+        for (int sub = 0; sub < 2; ++sub)
 #pragma unroll
-        for (int r = 0; r < 16; ++r) {
-          const float p =
-              (st2[sub][r] == -INFINITY) ? 0.f : __expf(st2[sub][r] - m_new);
-          st2[sub][r] = p;
-          rsum += p;
-        }
+          for (int r = 0; r < 16; ++r) {
+            const float p = __expf(st2[sub][r] - m_new);
+            st2[sub][r] = p;
+            rsum += p;
+          }
+      } else {
+#pragma unroll
+        for (int sub = 0; sub < 2; ++sub)
+#pragma unroll
+          for (int r = 0; r < 16; ++r) {
+            const float p =
+                (st2[sub][r] == -INFINITY) ? 0.f : __expf(st2[sub][r] - m_new);
+            st2[sub][r] = p;
+            rsum += p;
+          }
+      }
       rsum += __shfl_xor(rsum, 32, 64);
       l_run += rsum;
       m_run = m_new;
@@ -339,7 +368,7 @@ __launch_bounds__(512) __global__ void paged_prefill_v2_kernel(
 #pragma unroll
         for (int ks = 0; ks < 4; ++ks) {
           bf16x8 v_frag;
-          if (kSimpleV) {
+          if constexpr (kSimpleV) {
             // bisect mode: direct scalar reads from the subtiled layout
 #pragma unroll
             for (int jj = 0; jj < 8; ++jj) {
@@ -418,6 +447,13 @@ bool paged_attention_prefill_v2(torch::Tensor out, torch::Tensor q,
     return false;  // caller falls back to v1
   const int B = query_start_loc.size(0) - 1;
   const int max_blocks = block_tables.size(1);
+  // small-q chunked continuations underfill the 8-wave workgroups
+  // (few z-tiles); v1's 16-row tiles win there (profiles/r02)
+  if (q.size(0) <= 1024 && B == 1) {
+    // rough ctx estimate via block-table width (host-side, no sync)
+    const int64_t approx_L = (int64_t)max_blocks * kBS;
+    if (approx_L > 3 * q.size(0)) return false;
+  }
   // max q chunk rows: 8/G waves * 32
   const int qrows = (8 / G) * kQB;
   int max_qlen = 0;
@@ -435,16 +471,25 @@ bool paged_attention_prefill_v2(torch::Tensor out, torch::Tensor q,
   dim3 grid(B, n_kv, zdim), block(512);
   auto stream = c10::hip::getCurrentHIPStream().stream();
   const bool fp8_cache = k_cache.scalar_type() == torch::kFloat8_e5m2;
-#define LAUNCH_V2_CT(GG, CT)                                              \
-  hipLaunchKernelGGL((paged_prefill_v2_kernel<GG, CT>), grid, block, 0,   \
-                     stream, (ushort*)out.data_ptr(),                     \
+#define LAUNCH_V2_CT_D(GG, CT, D)                                         \
+  hipLaunchKernelGGL((paged_prefill_v2_kernel<GG, CT, D>), grid, block,   \
+                     0, stream, (ushort*)out.data_ptr(),                  \
                      (const ushort*)q.data_ptr(),                         \
                      (const CT*)k_cache.data_ptr(),                       \
                      (const CT*)v_cache.data_ptr(),                       \
                      block_tables.data_ptr<int32_t>(),                    \
                      query_start_loc.data_ptr<int32_t>(),                 \
                      seq_lens.data_ptr<int32_t>(), (float)scale, n_kv,    \
-                     max_blocks, q.stride(0), dbg_mode)
+                     max_blocks, q.stride(0))
+#define LAUNCH_V2_CT(GG, CT)                                              \
+  do {                                                                    \
+    switch (dbg_mode & 3) {                                               \
+      case 1: LAUNCH_V2_CT_D(GG, CT, 1); break;                           \
+      case 2: LAUNCH_V2_CT_D(GG, CT, 2); break;                           \
+      case 3: LAUNCH_V2_CT_D(GG, CT, 3); break;                           \
+      default: LAUNCH_V2_CT_D(GG, CT, 0); break;                          \
+    }                                                                     \
+  } while (0)
 #define LAUNCH_V2(GG)                                                     \
   do {                                                                    \
     if (fp8_cache) {                                                      \

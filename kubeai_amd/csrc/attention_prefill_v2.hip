@@ -136,7 +136,6 @@ __launch_bounds__(512) __global__ void paged_prefill_v2_kernel(
   bf16x8 st_k[2], st_v[2];
   bool st_ok[2];
   int h_tok[2], h_coff[2], h_kdst[2], h_vdst[2];
-  bool h_vswap[2];
 #pragma unroll
   for (int i = 0; i < 2; ++i) {
     const int u = threadIdx.x + i * 512;
@@ -145,16 +144,8 @@ __launch_bounds__(512) __global__ void paged_prefill_v2_kernel(
     h_tok[i] = tok;
     h_coff[i] = c8 * 8;
     h_kdst[i] = swz(tok, tok * kHD * 2 + c8 * 16);
-    // V subtile with 8B-unit rotation rot=(t4+h16)&3: breaks the 4-way
-    // bank aliasing between tr-read lane groups (same-bank accesses from
-    // lanes differing only in bits >=4 of the subtile index)
-    {
-      const int t4 = tok >> 2, h16 = c8 >> 1;
-      const int rot = (t4 + h16) & 3;
-      const int pair = (c8 & 1) ^ (rot >> 1);
-      h_vswap[i] = rot & 1;
-      h_vdst[i] = (t4 * 8 + h16) * 64 + (tok & 3) * 16 + pair * 8;
-    }
+    h_vdst[i] = ((tok >> 2) * 8 + (c8 >> 1)) * 64 + (tok & 3) * 16 +
+                (c8 & 1) * 8;
   }
   auto issue_tile_loads = [&](int kt) {
     const int kv_start = kt * kKVB;
@@ -199,16 +190,8 @@ __launch_bounds__(512) __global__ void paged_prefill_v2_kernel(
       if (st_ok[i])
         *reinterpret_cast<bf16x8*>(
             reinterpret_cast<char*>(k_lds[buf]) + h_kdst[i]) = st_k[i];
-      // V: subtiled (rotated); invalid units were zeroed at load.
-      // rot&1 exchanges the two 8 B halves of the 16 B unit pair.
-      bf16x8 vv = st_v[i];
-      if (h_vswap[i]) {
-        uint64_t* p = reinterpret_cast<uint64_t*>(&vv);
-        const uint64_t t = p[0];
-        p[0] = p[1];
-        p[1] = t;
-      }
-      *reinterpret_cast<bf16x8*>(&v_lds[buf][h_vdst[i]]) = vv;
+      // V: subtiled; invalid units were zeroed at load
+      *reinterpret_cast<bf16x8*>(&v_lds[buf][h_vdst[i]]) = st_v[i];
     }
   };
 
@@ -376,17 +359,10 @@ __launch_bounds__(512) __global__ void paged_prefill_v2_kernel(
       //     + ((m>>4)&1)*128 + ((m>>2)&3)*32 + (m&3)*8
       __builtin_amdgcn_s_setprio(1);  // PV MFMA cluster
       typedef __attribute__((address_space(3))) const char as3_char;
-      // rotated unit: c_addr = (m&3) ^ ((lrot + tr + 2*nb) & 3) with
-      // lrot = (2*(m>>5) + ((m>>4)&1)) — 4 per-lane bases selected by
-      // the compile-time (tr + 2*nb) & 3, zero extra VALU in the loop
-      const int lrot = 2 * hi + ((lane >> 4) & 1);
-      as3_char* vb0 =
+      as3_char* vbase =
           (as3_char*)(v_lds[cur]) +
-          (hi * 2048 + ((lane >> 4) & 1) * 128 + ((lane >> 2) & 3) * 32);
-      as3_char* vbx[4];
-#pragma unroll
-      for (int x = 0; x < 4; ++x)
-        vbx[x] = vb0 + (((lane & 3) ^ ((lrot + x) & 3)) * 8);
+          (hi * 2048 + ((lane >> 4) & 1) * 128 + ((lane >> 2) & 3) * 32 +
+           (lane & 3) * 8);
 #pragma unroll
       for (int nb = 0; nb < 4; ++nb) {
 #pragma unroll
@@ -398,30 +374,24 @@ __launch_bounds__(512) __global__ void paged_prefill_v2_kernel(
             for (int jj = 0; jj < 8; ++jj) {
               const int row = ks * 16 + hi * 8 + jj;
               const int col = nb * 32 + ln31;
-              const int t4 = row >> 2, h16 = col >> 4;
-              const int rot = (t4 + h16) & 3;
-              const int u = (((col & 15) >> 2) ^ rot);
-              const int off = (t4 * 8 + h16) * 64 + (row & 3) * 16 +
-                              u * 4 + (col & 3);
+              const int off = ((row >> 2) * 8 + (col >> 4)) * 64 +
+                              (row & 3) * 16 + (col & 15);
               v_frag[jj] = (short)v_lds[cur][off];
             }
           } else {
             typedef __attribute__((ext_vector_type(4))) short bf16x4;
             bf16x4 lo4, hi4;
-            // NOTE: lo4 (tr=0) and hi4 (tr=1, offset:1024) have
-            // DIFFERENT rotations — issue as two selects
-            as3_char* a0 = vbx[(0 + 2 * nb) & 3] + ks * 4096 + nb * 256;
-            as3_char* a1 = vbx[(1 + 2 * nb) & 3] + ks * 4096 + nb * 256;
+            as3_char* a0 = vbase + ks * 4096 + nb * 256;
             // "=&v" early-clobber: without it the allocator may alias an
             // output with the address VGPR — the first read's writeback
             // then races the second read's address consumption
             // (sporadic corruption, shape/schedule dependent)
             asm volatile(
                 "ds_read_b64_tr_b16 %0, %2\n\t"
-                "ds_read_b64_tr_b16 %1, %3 offset:1024\n\t"
+                "ds_read_b64_tr_b16 %1, %2 offset:1024\n\t"
                 "s_waitcnt lgkmcnt(0)"
                 : "=&v"(lo4), "=&v"(hi4)
-                : "v"(a0), "v"(a1));
+                : "v"(a0));
 #pragma unroll
             for (int j = 0; j < 4; ++j) {
               v_frag[j] = lo4[j];

@@ -56,8 +56,12 @@ DEV_INLINE int crow(int reg, int hi) {
   return (reg & 3) + 8 * (reg >> 2) + 4 * hi;
 }
 
-template <int G, typename CT = ushort, int DBG = 0>
-__launch_bounds__(512) __global__ void paged_prefill_v2_kernel(
+// MINW: min waves per SIMD fed to launch_bounds — the VGPR budget knob.
+// The compiler heuristic picks 4 waves/SIMD (128 arch VGPRs) and spills
+// 52-264 B/lane to scratch (measured, rocpd); MINW=2 allows 256 VGPRs
+// (1 workgroup/CU instead of 2) with no spill. Env KUBEAI_V2_OCC picks.
+template <int G, typename CT = ushort, int DBG = 0, int MINW = 4>
+__launch_bounds__(512, MINW) __global__ void paged_prefill_v2_kernel(
     ushort* __restrict__ out,            // [Tq, n_q, hd]
     const ushort* __restrict__ q,        // [Tq, n_q, hd]
     const CT* __restrict__ k_cache,      // [nb, n_kv, bs, hd]
@@ -471,8 +475,13 @@ bool paged_attention_prefill_v2(torch::Tensor out, torch::Tensor q,
   dim3 grid(B, n_kv, zdim), block(512);
   auto stream = c10::hip::getCurrentHIPStream().stream();
   const bool fp8_cache = k_cache.scalar_type() == torch::kFloat8_e5m2;
-#define LAUNCH_V2_CT_D(GG, CT, D)                                         \
-  hipLaunchKernelGGL((paged_prefill_v2_kernel<GG, CT, D>), grid, block,   \
+  static const int occ = []() {
+    const char* e = getenv("KUBEAI_V2_OCC");
+    return e ? atoi(e) : 4;
+  }();
+#define LAUNCH_V2_CT_DW(GG, CT, D, W)                                     \
+  hipLaunchKernelGGL((paged_prefill_v2_kernel<GG, CT, D, W>), grid,       \
+                     block,                                               \
                      0, stream, (ushort*)out.data_ptr(),                  \
                      (const ushort*)q.data_ptr(),                         \
                      (const CT*)k_cache.data_ptr(),                       \
@@ -484,10 +493,13 @@ bool paged_attention_prefill_v2(torch::Tensor out, torch::Tensor q,
 #define LAUNCH_V2_CT(GG, CT)                                              \
   do {                                                                    \
     switch (dbg_mode & 3) {                                               \
-      case 1: LAUNCH_V2_CT_D(GG, CT, 1); break;                           \
-      case 2: LAUNCH_V2_CT_D(GG, CT, 2); break;                           \
-      case 3: LAUNCH_V2_CT_D(GG, CT, 3); break;                           \
-      default: LAUNCH_V2_CT_D(GG, CT, 0); break;                          \
+      case 1: LAUNCH_V2_CT_DW(GG, CT, 1, 4); break;                       \
+      case 2: LAUNCH_V2_CT_DW(GG, CT, 2, 4); break;                       \
+      case 3: LAUNCH_V2_CT_DW(GG, CT, 3, 4); break;                       \
+      default:                                                            \
+        if (occ == 2) LAUNCH_V2_CT_DW(GG, CT, 0, 2);                      \
+        else LAUNCH_V2_CT_DW(GG, CT, 0, 4);                               \
+        break;                                                            \
     }                                                                     \
   } while (0)
 #define LAUNCH_V2(GG)                                                     \
@@ -507,6 +519,7 @@ bool paged_attention_prefill_v2(torch::Tensor out, torch::Tensor q,
   }
 #undef LAUNCH_V2
 #undef LAUNCH_V2_CT
+#undef LAUNCH_V2_CT_DW
   HIP_CHECK_KERNEL();
   return true;
 }

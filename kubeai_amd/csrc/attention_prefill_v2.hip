@@ -57,10 +57,11 @@ DEV_INLINE int crow(int reg, int hi) {
 }
 
 // MINW: min waves per SIMD fed to launch_bounds — the VGPR budget knob.
-// The compiler heuristic picks 4 waves/SIMD (128 arch VGPRs) and spills
-// 52-264 B/lane to scratch (measured, rocpd); MINW=2 allows 256 VGPRs
-// (1 workgroup/CU instead of 2) with no spill. Env KUBEAI_V2_OCC picks.
-template <int G, typename CT = ushort, int DBG = 0, int MINW = 4>
+// Forcing 4 waves/SIMD caps arch VGPRs at 128 and spills ~300 VGPRs +
+// ~430 SGPRs (~650 B/lane scratch): 90 TF @Tq=8192. MINW=2 gives 256
+// VGPRs, near-zero spill (≤21 v, ~80 s), 283 TF — 1 workgroup/CU beats
+// 2 spilly ones 3x over. Measured same-box; KUBEAI_V2_OCC=4 re-checks.
+template <int G, typename CT = ushort, int DBG = 0, int MINW = 2>
 __launch_bounds__(512, MINW) __global__ void paged_prefill_v2_kernel(
     ushort* __restrict__ out,            // [Tq, n_q, hd]
     const ushort* __restrict__ q,        // [Tq, n_q, hd]
@@ -477,7 +478,7 @@ bool paged_attention_prefill_v2(torch::Tensor out, torch::Tensor q,
   const bool fp8_cache = k_cache.scalar_type() == torch::kFloat8_e5m2;
   static const int occ = []() {
     const char* e = getenv("KUBEAI_V2_OCC");
-    return e ? atoi(e) : 4;
+    return e ? atoi(e) : 2;
   }();
 #define LAUNCH_V2_CT_DW(GG, CT, D, W)                                     \
   hipLaunchKernelGGL((paged_prefill_v2_kernel<GG, CT, D, W>), grid,       \
@@ -493,12 +494,12 @@ bool paged_attention_prefill_v2(torch::Tensor out, torch::Tensor q,
 #define LAUNCH_V2_CT(GG, CT)                                              \
   do {                                                                    \
     switch (dbg_mode & 3) {                                               \
-      case 1: LAUNCH_V2_CT_DW(GG, CT, 1, 4); break;                       \
-      case 2: LAUNCH_V2_CT_DW(GG, CT, 2, 4); break;                       \
-      case 3: LAUNCH_V2_CT_DW(GG, CT, 3, 4); break;                       \
+      case 1: LAUNCH_V2_CT_DW(GG, CT, 1, 2); break;                       \
+      case 2: LAUNCH_V2_CT_DW(GG, CT, 2, 2); break;                       \
+      case 3: LAUNCH_V2_CT_DW(GG, CT, 3, 2); break;                       \
       default:                                                            \
-        if (occ == 2) LAUNCH_V2_CT_DW(GG, CT, 0, 2);                      \
-        else LAUNCH_V2_CT_DW(GG, CT, 0, 4);                               \
+        if (occ == 4) LAUNCH_V2_CT_DW(GG, CT, 0, 4);                      \
+        else LAUNCH_V2_CT_DW(GG, CT, 0, 2);                               \
         break;                                                            \
     }                                                                     \
   } while (0)

@@ -37,6 +37,11 @@ class ForwardBatch:
     # per-token LoRA adapter ids (None when no adapters are active)
     lora_ids: torch.Tensor | None = None  # int32 [T]
 
+    # multimodal splice: rows of mm_embeds replace the token embeddings at
+    # packed positions mm_idx (image-placeholder tokens; models/vision.py)
+    mm_idx: torch.Tensor | None = None     # int64 [M]
+    mm_embeds: torch.Tensor | None = None  # [M, hidden]
+
     @property
     def n_tokens(self) -> int:
         return self.input_ids.shape[0]

@@ -19,6 +19,10 @@ from .runner import ModelRunner
 from .scheduler import Request, SamplingParams, Scheduler
 
 
+class RequestError(ValueError):
+    """Client error in a request body -> HTTP 400."""
+
+
 @dataclasses.dataclass
 class EngineConfig:
     model: str = "llama-tiny"  # preset name or HF dir
@@ -102,6 +106,7 @@ class LLMEngine:
         params: Optional[SamplingParams] = None,
         request_id: Optional[str] = None,
         lora_id: int = 0,
+        images: Optional[list] = None,
     ) -> Request:
         if params is None:
             params = SamplingParams()
@@ -111,9 +116,50 @@ class LLMEngine:
                 stop_token_ids=tuple(params.stop_token_ids)
                 + (self.arch.eos_token_id,),
             )
+        mm_spans: list[tuple[int, int, int]] = []
+        if images:
+            # expand each image-placeholder token to n_patches positions
+            # (vLLM-style); the runner replaces those embedding rows with
+            # the vision-tower output
+            if self.arch.vision is None:
+                raise RequestError(
+                    f"model {self.cfg.model!r} does not accept image input"
+                )
+            img_id = self.arch.image_token_id
+            n_patch = self.runner.vision.n_patches
+            n_slots = sum(1 for t in prompt_token_ids if t == img_id)
+            if n_slots != len(images):
+                raise RequestError(
+                    f"prompt has {n_slots} image slot(s) but request "
+                    f"carries {len(images)} image(s)"
+                )
+            expanded: list[int] = []
+            k = 0
+            for t in prompt_token_ids:
+                if t == img_id:
+                    mm_spans.append((len(expanded), n_patch, k * n_patch))
+                    expanded.extend([img_id] * n_patch)
+                    k += 1
+                else:
+                    expanded.append(t)
+            prompt_token_ids = expanded
+            if len(prompt_token_ids) >= self.cfg.max_model_len:
+                raise RequestError(
+                    "multimodal prompt exceeds max_model_len "
+                    "(image spans cannot be truncated)"
+                )
         if len(prompt_token_ids) >= self.cfg.max_model_len:
             prompt_token_ids = prompt_token_ids[-(self.cfg.max_model_len - 1) :]
         req = Request(prompt_token_ids, params, request_id=request_id, lora_id=lora_id)
+        if images:
+            req.images = list(images)
+            req.mm_spans = mm_spans
+            # salt the prefix cache with image content: same token ids,
+            # different pixels -> different KV
+            req.cache_salt = hash(
+                (lora_id,)
+                + tuple(hash(im.numpy().tobytes()) for im in req.images)
+            )
         self.scheduler.add_request(req)
         return req
 

@@ -133,15 +133,24 @@ class ModelRunner:
             self.model = LlamaForCausalLM(arch, device=self.device, dtype=dtype)
             self.n_kv_local = arch.num_key_value_heads
         self.model.eval()
+        # multimodal: the vision tower is replicated on every rank (it
+        # runs once per image at prefill; TP ranks compute identical
+        # embeddings deterministically instead of broadcasting them)
+        self.vision = None
+        if arch.vision is not None:
+            from kubeai_amd.models.vision import VisionTower
+
+            self.vision = VisionTower(arch, device=self.device, dtype=dtype)
+            self.vision.eval()
         if model_path is not None and os.path.isdir(model_path):
             if tp_group is not None:
                 from kubeai_amd.models.loader import load_weights_tp
 
-                load_weights_tp(self.model, model_path)
+                load_weights_tp(self.model, model_path, vision=self.vision)
             else:
                 from kubeai_amd.models.loader import load_weights
 
-                load_weights(self.model, model_path)
+                load_weights(self.model, model_path, vision=self.vision)
         self.quantization = quantization
         if quantization == "fp8":
             # on CPU the Fp8Linear dequant fallback keeps the same
@@ -217,6 +226,14 @@ class ModelRunner:
     # ------------------------------------------------------------------
     def build_batch(self, out: SchedulerOutput) -> ForwardBatch:
         bs = self.block_size
+        # vision embeddings: computed once per request, at its first
+        # scheduled prefill chunk
+        if self.vision is not None:
+            for ss in out.prefill:
+                req = ss.req
+                if req.images and req.mm_embeds is None:
+                    pixels = torch.stack([im.float() for im in req.images])
+                    req.mm_embeds = self.vision.encode(pixels)
         input_ids: list[int] = []
         positions: list[int] = []
         slots: list[int] = []
@@ -252,11 +269,27 @@ class ModelRunner:
 
         pre_tables: list[list[int]] = []
         pre_lens: list[int] = []
+        mm_idx: list[int] = []
+        mm_rows: list[torch.Tensor] = []
+        nd_tok = len(input_ids)  # decode tokens precede prefill tokens
         qsl = [0]
         for ss in out.prefill:
             emit(ss.req, ss.chunk_start, ss.chunk_len)
             pre_tables.append(self._pad(ss.req.block_table, max_bt))
             pre_lens.append(ss.chunk_start + ss.chunk_len)
+            # image-placeholder positions inside this chunk -> splice rows
+            if ss.req.mm_embeds is not None:
+                c0, c1 = ss.chunk_start, ss.chunk_start + ss.chunk_len
+                flat_base = nd_tok + qsl[-1]
+                for s0, n, row0 in ss.req.mm_spans:
+                    lo, hi = max(s0, c0), min(s0 + n, c1)
+                    if lo < hi:
+                        mm_idx.extend(
+                            range(flat_base + lo - c0, flat_base + hi - c0)
+                        )
+                        mm_rows.append(
+                            ss.req.mm_embeds[row0 + lo - s0 : row0 + hi - s0]
+                        )
             qsl.append(qsl[-1] + ss.chunk_len)
 
         # logits rows to sample: decode seq i -> packed index i; prefill seq j
@@ -306,6 +339,12 @@ class ModelRunner:
         t32 = lambda x: torch.tensor(x, dtype=torch.int32, device=dev)
         n_prefill = len(pre_lens)
         return ForwardBatch(
+            mm_idx=(
+                torch.tensor(mm_idx, dtype=torch.int64, device=dev)
+                if mm_idx
+                else None
+            ),
+            mm_embeds=torch.cat(mm_rows).to(dev) if mm_rows else None,
             lora_ids=lora_ids,
             input_ids=t32(input_ids),
             positions=t32(positions),

@@ -77,6 +77,16 @@ class Request:
         self.prompt_token_ids = list(prompt_token_ids)
         self.params = params
         self.lora_id = lora_id  # kv-cache salt + LoRA adapter selector
+        # multimodal: preprocessed pixel tensors + placeholder spans
+        # (start, length, embed_row_base) in prompt coordinates; embeds
+        # are computed lazily by the runner at first prefill
+        self.images: list = []
+        self.mm_spans: list[tuple[int, int, int]] = []
+        self.mm_embeds = None
+        # prefix-cache salt: lora_id plus (for multimodal) the image
+        # content hash — identical placeholder ids with different images
+        # must never share KV blocks
+        self.cache_salt: int = lora_id
         self.arrival_time = arrival_time if arrival_time is not None else time.monotonic()
         self.first_token_time: Optional[float] = None
         self.finish_time: Optional[float] = None
@@ -261,7 +271,7 @@ class Scheduler:
             try:
                 table, n_cached = self.bm.allocate(
                     req.tokens[:n_prompt],
-                    salt=req.lora_id,
+                    salt=req.cache_salt,
                     max_cached=(n_prompt - 1)
                     if self.enable_prefix_caching
                     else 0,
@@ -367,7 +377,7 @@ class Scheduler:
                         i,
                         tuple(req.tokens[i * bs : (i + 1) * bs]),
                         parent,
-                        req.lora_id,
+                        req.cache_salt,
                     )
                     req.block_hashes.append(h)
             if req.status.finished:

@@ -243,6 +243,8 @@ class EngineServer:
         self.tokenizer = load_tokenizer(
             self.cfg.model, arch.vocab_size, arch.bos_token_id, arch.eos_token_id
         )
+        if arch.vision is not None and hasattr(self.tokenizer, "image_token_id"):
+            self.tokenizer.image_token_id = arch.image_token_id
         # JSON-mode constrained decoding needs the tokenizer at sampling
         self.engine.runner.tokenizer = self.tokenizer
         self._ready.set()
@@ -252,10 +254,10 @@ class EngineServer:
             new_reqs, abort_ids = [], []
             while True:
                 try:
-                    rid, toks, params, lora_id = self._submit.get_nowait()
+                    item = self._submit.get_nowait()
                 except queue.Empty:
                     break
-                new_reqs.append((rid, toks, params, lora_id))
+                new_reqs.append(item)
             while True:
                 try:
                     abort_ids.append(self._aborts.get_nowait())
@@ -275,8 +277,19 @@ class EngineServer:
                       "shutdown": False}],
                     src=0,
                 )
-            for rid, toks, params, lora_id in new_reqs:
-                self.engine.add_request(toks, params, request_id=rid, lora_id=lora_id)
+            for rid, toks, params, lora_id, images in new_reqs:
+                try:
+                    self.engine.add_request(
+                        toks, params, request_id=rid, lora_id=lora_id,
+                        images=images or None,
+                    )
+                except RequestError as e:
+                    # surface a bad multimodal request as a failed output
+                    ent = self._events.get(rid)
+                    if ent is not None:
+                        loop, q = ent
+                        loop.call_soon_threadsafe(q.put_nowait, e)
+                    continue
                 M_PROMPT.labels(label).inc(len(toks))
                 worked = True
             for rid in abort_ids:
@@ -327,7 +340,8 @@ class EngineServer:
             self._dist.destroy_process_group()
 
     # ------------------------------------------------------------- requests
-    async def generate(self, token_ids: list[int], params: SamplingParams, lora_id: int = 0):
+    async def generate(self, token_ids: list[int], params: SamplingParams,
+                       lora_id: int = 0, images: list | None = None):
         """Async iterator of RequestOutput for one request."""
         rid = f"cmpl-{uuid.uuid4().hex[:16]}"
         loop = asyncio.get_running_loop()
@@ -335,10 +349,12 @@ class EngineServer:
         self._events[rid] = (loop, q)
         t0 = time.monotonic()
         first = True
-        self._submit.put((rid, token_ids, params, lora_id))
+        self._submit.put((rid, token_ids, params, lora_id, images or []))
         try:
             while True:
                 o = await q.get()
+                if isinstance(o, Exception):
+                    raise o  # RequestError -> HTTP 400 via the app handler
                 if first:
                     M_TTFT.labels(self.served_model_name).observe(time.monotonic() - t0)
                     first = False
@@ -384,8 +400,45 @@ class EngineServer:
         return 1 + (zlib.crc32(name.encode()) % 1_000_000)
 
 
-class RequestError(ValueError):
-    """Client error in a request body -> HTTP 400."""
+from .engine import RequestError  # noqa: E402  (shared with the engine core)
+
+
+def _images_from_messages(server, messages: list) -> list:
+    """Decode + preprocess image_url content parts, in reading order.
+
+    Returns CLIP-normalized pixel tensors sized for the model's vision
+    tower. Decoding problems and images sent to text-only models raise
+    RequestError -> HTTP 400 (the reference forwards parts to vLLM and
+    relays its 400s; here the engine is in-house).
+    """
+    from kubeai_amd.utils import imaging
+
+    urls: list[str] = []
+    for m in messages:
+        content = m.get("content")
+        if not isinstance(content, list):
+            continue
+        for p in content:
+            if isinstance(p, dict) and p.get("type") == "image_url":
+                iu = p.get("image_url")
+                url = iu.get("url", "") if isinstance(iu, dict) else str(iu or "")
+                urls.append(url)
+    if not urls:
+        return []
+    arch = getattr(getattr(server, "engine", None), "arch", None)
+    if arch is None or arch.vision is None:
+        raise RequestError(
+            f"model {server.served_model_name!r} does not accept image input"
+        )
+    size = int(arch.vision.get("image_size", 336))
+    out = []
+    for url in urls:
+        try:
+            img, _raw = imaging.image_from_url(url)
+            out.append(imaging.preprocess(img, size))
+        except imaging.ImageError as e:
+            raise RequestError(f"bad image: {e}") from None
+    return out
 
 
 def build_app(server: EngineServer) -> FastAPI:
@@ -641,6 +694,7 @@ def build_app(server: EngineServer) -> FastAPI:
             if tool_choice == "required" or isinstance(tool_choice, dict):
                 params.json_mode = True
         toks = apply_chat_template(server.tokenizer, messages)
+        images = _images_from_messages(server, messages)
         lora_id = _resolve_lora(server, body.get("model"))
         if isinstance(lora_id, JSONResponse):
             return lora_id
@@ -649,7 +703,8 @@ def build_app(server: EngineServer) -> FastAPI:
         if body.get("stream"):
             return StreamingResponse(
                 _stream_completion(server, toks, params, name, chat=True,
-                                   lora_id=lora_id, stops=stops, n=n),
+                                   lora_id=lora_id, stops=stops, n=n,
+                                   images=images),
                 media_type="text/event-stream",
             )
         want_logprobs = bool(body.get("logprobs"))
@@ -661,7 +716,8 @@ def build_app(server: EngineServer) -> FastAPI:
         for i in range(n):
             steps: list = [] if want_logprobs else None
             final, cut_text, cut_reason = await _finish_tokens(
-                server.generate(toks, params, lora_id), stops, steps
+                server.generate(toks, params, lora_id, images=images), stops,
+                steps
             )
             text = (
                 cut_text
@@ -1025,7 +1081,7 @@ class _ChoiceStream:
 
 async def _stream_completion(server, toks, params, name, chat: bool,
                              lora_id: int = 0, stops=None, n: int = 1,
-                             echo_text: str = ""):
+                             echo_text: str = "", images=None):
     """SSE stream; n>1 runs n engine requests concurrently, interleaving
     chunks with their choice index (the reference gets this via vLLM
     passthrough; r1 rejected stream+n>1)."""
@@ -1034,7 +1090,7 @@ async def _stream_completion(server, toks, params, name, chat: bool,
     agg: asyncio.Queue = asyncio.Queue()
 
     async def pump(i: int):
-        async for o in server.generate(toks, params, lora_id):
+        async for o in server.generate(toks, params, lora_id, images=images):
             await agg.put((i, o))
 
     tasks = [asyncio.create_task(pump(i)) for i in range(n)]
@@ -1108,14 +1164,25 @@ def _tp_worker_main(rank: int, world: int, port: int, cfg: EngineConfig) -> None
         cfg.model, engine.arch.vocab_size, engine.arch.bos_token_id,
         engine.arch.eos_token_id,
     )
+    if engine.arch.vision is not None and hasattr(
+        engine.runner.tokenizer, "image_token_id"
+    ):
+        engine.runner.tokenizer.image_token_id = engine.arch.image_token_id
     while True:
         box = [None]
         dist.broadcast_object_list(box, src=0)
         msg = box[0]
         if msg["shutdown"]:
             break
-        for rid, toks, params, lora_id in msg["new"]:
-            engine.add_request(toks, params, request_id=rid, lora_id=lora_id)
+        for item in msg["new"]:
+            rid, toks, params, lora_id, images = item
+            try:
+                engine.add_request(
+                    toks, params, request_id=rid, lora_id=lora_id,
+                    images=images or None,
+                )
+            except RequestError:
+                continue  # rank 0 already rejected it identically
         for rid in msg["aborts"]:
             engine.abort_request(rid)
         for tok_lists in msg.get("embeds", []):

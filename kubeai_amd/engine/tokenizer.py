@@ -24,6 +24,9 @@ class SyntheticTokenizer:
         self.vocab_size = vocab_size
         self.bos_token_id = bos_token_id
         self.eos_token_id = eos_token_id
+        # multimodal: "<image>" in the prompt maps to this id when set
+        # (the server wires it from the model config; id 99 for presets)
+        self.image_token_id: int | None = None
         # word-hash ids start above the specials + char region
         self._lo = 100
 
@@ -36,6 +39,9 @@ class SyntheticTokenizer:
         toks: list[int] = [self.bos_token_id] if add_bos else []
         span = self.vocab_size - self._lo
         for word in text.split():
+            if word == "<image>" and self.image_token_id is not None:
+                toks.append(self.image_token_id)
+                continue
             toks.append(self._lo + zlib.crc32(word.encode()) % span)
         return toks
 
@@ -118,9 +124,18 @@ def apply_chat_template(tokenizer, messages: list[dict]) -> list[int]:
     def text_of(m: dict) -> str:
         content = m.get("content") or ""
         if isinstance(content, list):  # OpenAI content-parts form
-            content = " ".join(
-                p.get("text", "") for p in content if isinstance(p, dict)
-            )
+            parts: list[str] = []
+            for p in content:
+                if not isinstance(p, dict):
+                    continue
+                if p.get("type") == "image_url":
+                    # marker consumed by the tokenizer (vision models) or
+                    # left as plain text (text-only models reject images
+                    # at admission)
+                    parts.append("<image>")
+                else:
+                    parts.append(p.get("text", ""))
+            content = " ".join(parts)
         return content
 
     # real checkpoints: use the model's own chat template

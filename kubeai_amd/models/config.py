@@ -33,6 +33,11 @@ class ModelArchConfig:
     attention_bias: bool = False
     # HF rope_scaling dict (llama3 / linear), None = unscaled
     rope_scaling: dict | None = None
+    # multimodal (LLaVA-style): vision-tower config dict + the token id
+    # that marks an image slot in the prompt (expanded to n_patches
+    # placeholder positions at admission; models/vision.py)
+    vision: dict | None = None
+    image_token_id: int = -1
 
     @property
     def n_kv_heads(self) -> int:
@@ -45,6 +50,18 @@ class ModelArchConfig:
         hidden = cfg["hidden_size"]
         n_heads = cfg["num_attention_heads"]
         arch = (cfg.get("architectures") or ["LlamaForCausalLM"])[0]
+        vision = None
+        image_token_id = -1
+        if arch == "LlavaForConditionalGeneration":
+            # llava config nests text_config + vision_config
+            vision = dict(cfg.get("vision_config") or {})
+            vision.setdefault(
+                "vision_feature_layer", cfg.get("vision_feature_layer", -2)
+            )
+            image_token_id = cfg.get("image_token_index", 32000)
+            cfg = {**cfg, **(cfg.get("text_config") or {})}
+            hidden = cfg["hidden_size"]
+            n_heads = cfg["num_attention_heads"]
         eos = cfg.get("eos_token_id", 2)
         if isinstance(eos, list):
             eos = eos[0]
@@ -70,6 +87,8 @@ class ModelArchConfig:
             num_local_experts=cfg.get("num_local_experts", 0),
             num_experts_per_tok=cfg.get("num_experts_per_tok", 2),
             rope_scaling=cfg.get("rope_scaling"),
+            vision=vision,
+            image_token_id=image_token_id,
         )
 
 
@@ -138,6 +157,54 @@ PRESETS: dict[str, ModelArchConfig] = {
         max_position_embeddings=2048,
         bos_token_id=1,
         eos_token_id=2,
+    ),
+    # multimodal tiny: llama-tiny text + 2-layer CLIP tower; 16 patches
+    # per image (64px / 16px patches). Image slot token = 99 (synthetic
+    # tokenizer's char region stops at 97).
+    "llava-tiny": ModelArchConfig(
+        vocab_size=2048,
+        hidden_size=256,
+        intermediate_size=512,
+        num_hidden_layers=2,
+        num_attention_heads=2,
+        num_key_value_heads=1,
+        head_dim=128,
+        max_position_embeddings=2048,
+        bos_token_id=1,
+        eos_token_id=2,
+        image_token_id=99,
+        vision={
+            "image_size": 64,
+            "patch_size": 16,
+            "hidden_size": 128,
+            "num_hidden_layers": 2,
+            "num_attention_heads": 2,
+            "intermediate_size": 256,
+        },
+    ),
+    # LLaVA-1.5-7B shape: Vicuna-7B text + CLIP ViT-L/14-336 (576
+    # patches/image, features from the penultimate layer)
+    "llava-1.5-7b": ModelArchConfig(
+        vocab_size=32064,
+        hidden_size=4096,
+        intermediate_size=11008,
+        num_hidden_layers=32,
+        num_attention_heads=32,
+        num_key_value_heads=32,
+        head_dim=128,
+        max_position_embeddings=4096,
+        rope_theta=10000.0,
+        bos_token_id=1,
+        eos_token_id=2,
+        image_token_id=32000,
+        vision={
+            "image_size": 336,
+            "patch_size": 14,
+            "hidden_size": 1024,
+            "num_hidden_layers": 24,
+            "num_attention_heads": 16,
+            "intermediate_size": 4096,
+        },
     ),
     # long-context tiny model: exercises chunked prefill + paged decode
     # at tens of thousands of tokens without big weights (the engine

@@ -286,6 +286,9 @@ class LlamaForCausalLM(nn.Module):
     @torch.inference_mode()
     def forward(self, fb: ForwardBatch) -> torch.Tensor:
         x = self.embed_tokens(fb.input_ids.long())
+        if fb.mm_embeds is not None:
+            # image-placeholder positions take the vision-tower embeddings
+            x = x.index_copy(0, fb.mm_idx, fb.mm_embeds.to(x.dtype))
         residual = None
         for i, layer in enumerate(self.layers):
             x, residual = layer(x, residual, fb, self.kv_caches[i], self.cos_sin)

@@ -78,15 +78,16 @@ def iter_safetensors(model_dir: str) -> Iterator[tuple[str, torch.Tensor]]:
 
 
 def _strip(name: str) -> str:
-    for prefix in ("model.", "language_model.model."):
+    for prefix in ("model.", "language_model.model.", "language_model."):
         if name.startswith(prefix):
             return name[len(prefix) :]
     return name
 
 
-def load_weights_tp(model, model_dir: str) -> int:
+def load_weights_tp(model, model_dir: str, vision=None) -> int:
     """Load HF weights into TPLlamaForCausalLM: every rank reads the full
-    tensors and keeps its shard (same slicing as parallel/tp.py init)."""
+    tensors and keeps its shard (same slicing as parallel/tp.py init).
+    The vision tower (llava) is replicated on every rank, not sharded."""
     cfg = model.cfg
     tp = model.tp
     hd = cfg.head_dim
@@ -130,6 +131,12 @@ def load_weights_tp(model, model_dir: str) -> int:
             put(f"layers.{layer}.mlp.gate_up_proj.weight", torch.cat([g, u], dim=0))
 
     for name, w in iter_safetensors(model_dir):
+        if name.startswith(("vision_tower.", "multi_modal_projector.")):
+            if vision is not None and not vision.load_hf_tensor(name, w):
+                raise ValueError(f"unrecognized vision tensor {name}")
+            continue
+        if name == "image_newline":  # llava-1.6 packing artifact, unused
+            continue
         n = _strip(name)
         if n == "embed_tokens.weight":
             put("embed_tokens.weight", w)
@@ -181,9 +188,13 @@ def load_weights_tp(model, model_dir: str) -> int:
     return len(filled)
 
 
-def load_weights(model, model_dir: str) -> int:
+def load_weights(model, model_dir: str, vision=None) -> int:
     """Load HF weights into LlamaForCausalLM (dense or MoE). Returns the
-    number of engine parameters filled; raises if any stays unset."""
+    number of engine parameters filled; raises if any stays unset.
+
+    vision: optional models/vision.py VisionTower — llava checkpoints
+    route their vision_tower.* / multi_modal_projector.* tensors there
+    (HF llava nests the text model under language_model.*)."""
     cfg = model.cfg
     hd, nq, nkv = cfg.head_dim, cfg.num_attention_heads, cfg.num_key_value_heads
     params = dict(model.named_parameters())
@@ -228,6 +239,12 @@ def load_weights(model, model_dir: str) -> int:
             put(tgt, torch.cat([ps["gate"], ps["up"]], dim=0))
 
     for name, w in iter_safetensors(model_dir):
+        if name.startswith(("vision_tower.", "multi_modal_projector.")):
+            if vision is not None and not vision.load_hf_tensor(name, w):
+                raise ValueError(f"unrecognized vision tensor {name}")
+            continue
+        if name == "image_newline":  # llava-1.6 packing artifact, unused
+            continue
         n = _strip(name)
         if n == "embed_tokens.weight":
             put("embed_tokens.weight", w)

@@ -332,3 +332,43 @@ def test_long_context_generation_gpu():
                 out2 = o
     assert out2.num_cached_tokens > 15000
     assert out2.output_token_ids == out.output_token_ids
+
+
+@pytest.mark.gpu
+def test_multimodal_llava_gpu():
+    """LLaVA path on device: image embeddings spliced into prefill, greedy
+    reproducibility with the prefix cache salted by image content."""
+    import base64
+
+    from kubeai_amd.utils import imaging
+
+    g = torch.Generator().manual_seed(11)
+    img = torch.randint(0, 256, (40, 56, 3), generator=g, dtype=torch.uint8)
+    px = imaging.preprocess(img, 64)
+    eng = LLMEngine(EngineConfig(model="llava-tiny", device="cuda",
+                                 num_gpu_blocks=256, max_model_len=512))
+    img_id = eng.arch.image_token_id
+    prompt = [1, 4, img_id, 200, 300, 400]
+
+    def run(rid, pixels):
+        eng.add_request(prompt, SamplingParams(max_tokens=6, ignore_eos=True),
+                        request_id=rid, images=[pixels])
+        out = None
+        for _ in range(100):
+            if not eng.has_work():
+                break
+            for o in eng.step():
+                if o.finished and o.request_id == rid:
+                    out = o
+        return out
+
+    o1 = run("mm1", px)
+    o2 = run("mm2", px)  # same image: cache hit + identical greedy tokens
+    assert o1 is not None and len(o1.output_token_ids) == 6
+    assert o2.output_token_ids == o1.output_token_ids
+    assert o2.num_cached_tokens > 0
+    # different image, same tokens: salt isolates the prefix cache
+    g2 = torch.Generator().manual_seed(99)
+    img2 = torch.randint(0, 256, (40, 56, 3), generator=g2, dtype=torch.uint8)
+    o3 = run("mm3", imaging.preprocess(img2, 64))
+    assert o3 is not None and o3.num_cached_tokens == 0

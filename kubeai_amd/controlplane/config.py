@@ -69,6 +69,10 @@ class SystemConfig:
     # HIGHER = more important). A Model's priorityClassName maps every
     # request for that model onto the engine's admission priority.
     priority_classes: dict = dataclasses.field(default_factory=dict)
+    # cacheProfiles: name -> {"sharedFilesystem": {"storageClassName":
+    # ..., "persistentVolumeName": ...}} (reference config/system.go:202-
+    # 208); consumed by the K8s-mode PVC/Job cache machinery
+    cacheProfiles: dict = dataclasses.field(default_factory=dict)
     # kubernetes: when set, Models persist as CRs and replicas run as
     # Pods against this API server (kubestore.py). Keys: apiUrl (empty =
     # in-cluster), namespace, engineImage, gpuResource, leaseName.
@@ -119,6 +123,7 @@ def load_config(path: Optional[str]) -> SystemConfig:
             str(k): int(v) for k, v in raw.get("priorityClasses", {}).items()
         },
         kubernetes=raw.get("kubernetes"),
+        cacheProfiles=raw.get("cacheProfiles") or {},
     )
     cfg.validate()
     return cfg

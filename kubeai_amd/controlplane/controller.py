@@ -184,7 +184,12 @@ class ModelController:
             for r in replicas:
                 await self.runtime.delete(r.name)
             if model.spec.cache_profile and CACHE_EVICTION_FINALIZER in model.finalizers:
-                await self.cache.evict(model)
+                # evict() may span reconciles (K8s mode runs an eviction
+                # Job — cache.go's deletion branch); keep the finalizer
+                # until it reports completion
+                done = await self.cache.evict(model)
+                if done is False:
+                    return
                 model.finalizers.remove(CACHE_EVICTION_FINALIZER)
             self.store.remove_model(name)
             return

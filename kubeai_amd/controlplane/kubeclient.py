@@ -18,6 +18,7 @@ kubeai_amd/controlplane/fakekube.py — the envtest analog).
 from __future__ import annotations
 
 import json
+import time
 import os
 from typing import Iterator, Optional
 
@@ -123,6 +124,23 @@ class KubeClient:
 
     def delete(self, path: str) -> Optional[dict]:
         return self._req("DELETE", path)
+
+    def get_opt(self, path: str) -> Optional[dict]:
+        """GET that maps 404 to None."""
+        try:
+            return self._req("GET", path)
+        except ApiError as e:
+            if e.status == 404:
+                return None
+            raise
+
+    def delete_opt(self, path: str) -> None:
+        """DELETE that swallows 404."""
+        try:
+            self._req("DELETE", path)
+        except ApiError as e:
+            if e.status != 404:
+                raise
 
     def watch(self, path: str, resource_version: str = "0",
               timeout: Optional[float] = None) -> Iterator[dict]:
@@ -357,6 +375,10 @@ def _engine_container(model, image: str) -> dict:
         model_arg = "/model" if spec.url.startswith("pvc://") else spec.url
         if eng == "VLLM":
             model_arg = "/model" if spec.url.startswith("pvc://") else ref
+        if spec.cache_profile:
+            # cached models load from the shared PVC (cache.go
+            # patchServerCacheVolumes + engine_vllm.go modelCacheDir arg)
+            model_arg = f"/models/{model.name}-{model.uid}"
         return {
             "command": cmd,
             "args": ["--model", model_arg,
@@ -417,6 +439,20 @@ def pod_manifest_for(model, name: str, spec_hash: str, n_gpus: int,
     env = [{"name": k, "value": v} for k, v in sorted(model.spec.env.items())]
     env += eng["env"]
     src_env, src_vols, src_mounts = source_pod_additions(model.spec.url)
+    if model.spec.cache_profile:
+        # mount the profile cache PVC read-only at the model dir
+        cdir = f"/models/{model.name}-{model.uid}"
+        src_vols = src_vols + [{
+            "name": "model-cache",
+            "persistentVolumeClaim": {
+                "claimName": f"shared-model-cache-{model.spec.cache_profile}",
+                "readOnly": True,
+            },
+        }]
+        src_mounts = src_mounts + [{
+            "name": "model-cache", "mountPath": cdir,
+            "subPath": cdir.lstrip("/"), "readOnly": True,
+        }]
     env += src_env
     resources = {}
     if n_gpus > 0:
@@ -497,3 +533,226 @@ def pod_manifest_for(model, name: str, spec_hash: str, n_gpus: int,
 
         manifest = apply_patch(manifest, patches)
     return manifest
+
+
+# ---------------------------------------------------------------- cache
+# Reference: internal/modelcontroller/cache.go — a shared-filesystem PVC
+# per cache profile ("shared-model-cache-<profile>", RWX), a loader Job
+# ("load-cache-<model>") that downloads into /models/<name>-<uid>, a
+# per-model annotation on the PVC ("models.kubeai.org/<model>" ->
+# {"uid","timestamp"}) marking loaded content, and an eviction Job
+# ("evict-cache-<model>") driven by the cache-eviction finalizer.
+
+PVC_MODEL_ANN_PREFIX = "models.kubeai.org/"
+
+
+def model_cache_dir(model) -> str:
+    return f"/models/{model.name}-{model.uid}"
+
+
+def cache_pvc_name(model) -> str:
+    return f"shared-model-cache-{model.spec.cache_profile}"
+
+
+def cache_pvc_manifest(model, profile: dict, namespace: str) -> dict:
+    fs = (profile or {}).get("sharedFilesystem") or {}
+    spec = {
+        "accessModes": ["ReadWriteMany"],
+        "storageClassName": fs.get("storageClassName", ""),
+        "resources": {"requests": {"storage": fs.get("size", "10Gi")}},
+    }
+    if fs.get("persistentVolumeName"):
+        spec["volumeName"] = fs["persistentVolumeName"]
+    return {
+        "apiVersion": "v1",
+        "kind": "PersistentVolumeClaim",
+        "metadata": {"name": cache_pvc_name(model), "namespace": namespace},
+        "spec": spec,
+    }
+
+
+def _cache_job_manifest(model, name: str, container: dict,
+                        namespace: str) -> dict:
+    container.setdefault("volumeMounts", []).append(
+        {"name": "model", "mountPath": "/models", "subPath": "models"}
+    )
+    return {
+        "apiVersion": "batch/v1",
+        "kind": "Job",
+        "metadata": {
+            "name": name,
+            "namespace": namespace,
+            "labels": {"app.kubernetes.io/name": "kubeai",
+                       "kubeai.org/cache-job-for": model.name},
+        },
+        "spec": {
+            "ttlSecondsAfterFinished": 60,
+            "parallelism": 1,
+            "completions": 1,
+            "template": {
+                "spec": {
+                    "restartPolicy": "OnFailure",
+                    "containers": [container],
+                    "volumes": [{
+                        "name": "model",
+                        "persistentVolumeClaim": {
+                            "claimName": cache_pvc_name(model)
+                        },
+                    }],
+                },
+            },
+        },
+    }
+
+
+def load_cache_job_manifest(model, loader_image: str, namespace: str) -> dict:
+    env = [{"name": k, "value": v} for k, v in sorted(model.spec.env.items())]
+    src_env, src_vols, src_mounts = source_pod_additions(model.spec.url)
+    if model.spec.cache_profile:
+        # mount the profile cache PVC read-only at the model dir
+        cdir = f"/models/{model.name}-{model.uid}"
+        src_vols = src_vols + [{
+            "name": "model-cache",
+            "persistentVolumeClaim": {
+                "claimName": f"shared-model-cache-{model.spec.cache_profile}",
+                "readOnly": True,
+            },
+        }]
+        src_mounts = src_mounts + [{
+            "name": "model-cache", "mountPath": cdir,
+            "subPath": cdir.lstrip("/"), "readOnly": True,
+        }]
+    cdir = model_cache_dir(model)
+    container = {
+        "name": "loader",
+        "image": loader_image,
+        "env": env + src_env,
+        "args": [model.spec.url, cdir],
+        "volumeMounts": [
+            {"name": "model", "mountPath": cdir, "subPath": cdir.lstrip("/")}
+        ] + src_mounts,
+    }
+    m = _cache_job_manifest(model, f"load-cache-{model.name}", container,
+                            namespace)
+    # source secrets (hf/s3/gs/oss) piggyback on the loader pod
+    m["spec"]["template"]["spec"]["volumes"].extend(src_vols)
+    return m
+
+
+def evict_cache_job_manifest(model, loader_image: str, namespace: str) -> dict:
+    container = {
+        "name": "evictor",
+        "image": loader_image,
+        "command": ["bash", "-c", f"rm -rf {model_cache_dir(model)}"],
+        "volumeMounts": [],
+    }
+    return _cache_job_manifest(model, f"evict-cache-{model.name}", container,
+                               namespace)
+
+
+def job_completed(job: dict) -> bool:
+    st = (job or {}).get("status") or {}
+    if int(st.get("succeeded") or 0) >= 1:
+        return True
+    return any(
+        c.get("type") == "Complete" and c.get("status") == "True"
+        for c in st.get("conditions") or []
+    )
+
+
+class KubeCacheManager:
+    """PVC + Job cache machinery for Kubernetes mode (cache.go analog).
+
+    ensure(): creates the profile PVC on first use, runs the loader Job,
+    records loaded content as a PVC annotation, deletes the finished Job.
+    evict(): returns True when eviction has fully completed (the
+    controller keeps the finalizer until then — same multi-reconcile flow
+    as the reference's reconcileCache deletion branch).
+    """
+
+    def __init__(self, kc: KubeClient, cache_profiles: dict | None = None,
+                 loader_image: str = "substratusai/huggingface-model-loader:main"):
+        self.kc = kc
+        self.profiles = cache_profiles or {}
+        self.loader_image = loader_image
+
+    def _paths(self, model):
+        ns = self.kc.namespace
+        return (
+            f"/api/v1/namespaces/{ns}/persistentvolumeclaims/{cache_pvc_name(model)}",
+            f"/apis/batch/v1/namespaces/{ns}/jobs/load-cache-{model.name}",
+            f"/apis/batch/v1/namespaces/{ns}/jobs/evict-cache-{model.name}",
+        )
+
+    async def ensure(self, model) -> bool:
+        prof = self.profiles.get(model.spec.cache_profile)
+        if prof is None:
+            raise ValueError(
+                f"model {model.name}: unknown cacheProfile "
+                f"{model.spec.cache_profile!r}"
+            )
+        pvc_path, load_path, _ = self._paths(model)
+        pvc = self.kc.get_opt(pvc_path)
+        if pvc is None:
+            pvc = self.kc.create(
+                pvc_path.rsplit("/", 1)[0],
+                cache_pvc_manifest(model, prof, self.kc.namespace),
+            )
+        ann = ((pvc.get("metadata") or {}).get("annotations") or {}).get(
+            PVC_MODEL_ANN_PREFIX + model.name
+        )
+        if ann:
+            try:
+                if json.loads(ann).get("uid") == str(model.uid):
+                    self.kc.delete_opt(load_path)  # tidy a finished Job
+                    return True
+            except ValueError:
+                pass
+        job = self.kc.get_opt(load_path)
+        if job is None:
+            self.kc.create(
+                load_path.rsplit("/", 1)[0],
+                load_cache_job_manifest(model, self.loader_image,
+                                        self.kc.namespace),
+            )
+            return False
+        if not job_completed(job):
+            return False
+        self.kc.patch_merge(pvc_path, {
+            "metadata": {"annotations": {
+                PVC_MODEL_ANN_PREFIX + model.name: json.dumps(
+                    {"uid": str(model.uid), "timestamp": int(time.time())}
+                ),
+            }},
+        })
+        self.kc.delete_opt(load_path)
+        return True
+
+    async def evict(self, model) -> bool:
+        pvc_path, load_path, evict_path = self._paths(model)
+        pvc = self.kc.get_opt(pvc_path)
+        if pvc is None or (pvc.get("metadata") or {}).get("deletionTimestamp"):
+            self.kc.delete_opt(load_path)
+            self.kc.delete_opt(evict_path)
+            return True
+        anns = (pvc.get("metadata") or {}).get("annotations") or {}
+        if PVC_MODEL_ANN_PREFIX + model.name in anns:
+            # merge-patch null deletes the key
+            self.kc.patch_merge(pvc_path, {
+                "metadata": {"annotations": {
+                    PVC_MODEL_ANN_PREFIX + model.name: None,
+                }},
+            })
+        job = self.kc.get_opt(evict_path)
+        if job is None:
+            self.kc.create(
+                evict_path.rsplit("/", 1)[0],
+                evict_cache_job_manifest(model, self.loader_image,
+                                         self.kc.namespace),
+            )
+            return False
+        if not job_completed(job):
+            return False
+        self.kc.delete_opt(evict_path)
+        self.kc.delete_opt(load_path)
+        return True

@@ -55,7 +55,18 @@ class Manager:
             self.election = LeaseElection(
                 kc, lease_name=kube.get("leaseName", "kubeai.org")
             )
+            from .kubeclient import KubeCacheManager
+
+            kube_cache = KubeCacheManager(
+                kc,
+                cache_profiles=self.cfg.cacheProfiles,
+                loader_image=kube.get(
+                    "modelLoaderImage",
+                    "substratusai/huggingface-model-loader:main",
+                ),
+            )
         else:
+            kube_cache = None
             self.store = Store()
             self.runtime = runtime or LocalProcessRuntime(
                 self.store, n_gpus=self.cfg.n_gpus
@@ -71,7 +82,7 @@ class Manager:
             self.store,
             self.runtime,
             resource_profiles=self.cfg.resource_profiles,
-            cache=CacheManager(self.cfg.cache_dir),
+            cache=kube_cache or CacheManager(self.cfg.cache_dir),
         )
         self.autoscaler = Autoscaler(
             self.store,

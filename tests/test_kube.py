@@ -501,3 +501,117 @@ def test_jsonpatch_ops():
         apply_patch(doc, [{"op": "test", "path": "/x", "value": 6}])
     with _p.raises(PatchError):
         apply_patch(doc, [{"op": "replace", "path": "/nope", "value": 1}])
+
+
+def test_kube_cache_pvc_job_flow(api):
+    """K8s-mode cache machinery: profile PVC + loader Job + PVC annotation
+    + eviction Job driven by the finalizer (reference cache.go end to
+    end)."""
+
+    async def main():
+        kc = kc_for(api)
+        store = KubeStore(kc)
+        await store.start()
+        from kubeai_amd.controlplane.kubeclient import (
+            KubeCacheManager, PVC_MODEL_ANN_PREFIX, job_completed,
+        )
+
+        cache = KubeCacheManager(
+            kc,
+            cache_profiles={"shared-fs": {
+                "sharedFilesystem": {"storageClassName": "nfs-fast"},
+            }},
+            loader_image="loader:test",
+        )
+        ctrl = ModelController(store, KubeRuntime(store), cache=cache)
+        ctrl.start()
+        ns = kc.namespace
+        pvc_path = f"/api/v1/namespaces/{ns}/persistentvolumeclaims/shared-model-cache-shared-fs"
+        load_path = f"/apis/batch/v1/namespaces/{ns}/jobs/load-cache-m6"
+        evict_path = f"/apis/batch/v1/namespaces/{ns}/jobs/evict-cache-m6"
+        try:
+            store.apply_model(
+                Model(name="m6", spec=ModelSpec(
+                    url="hf://org/m6", min_replicas=1,
+                    cache_profile="shared-fs",
+                ))
+            )
+            # PVC + loader Job appear; no pods until the cache loads
+            await wait_for(lambda: kc.get_opt(pvc_path) is not None,
+                           msg="cache PVC created")
+            await wait_for(lambda: kc.get_opt(load_path) is not None,
+                           msg="loader job created")
+            pvc = kc.get(pvc_path)
+            assert pvc["spec"]["storageClassName"] == "nfs-fast"
+            assert pvc["spec"]["accessModes"] == ["ReadWriteMany"]
+            job = kc.get(load_path)
+            args = job["spec"]["template"]["spec"]["containers"][0]["args"]
+            assert args[0] == "hf://org/m6" and args[1].startswith("/models/m6-")
+            assert not kc.list(f"/api/v1/namespaces/{ns}/pods")
+            assert not job_completed(job)
+
+            # complete the loader Job (a kubelet would); nudge a reconcile
+            kc.patch_merge(load_path, {"status": {"succeeded": 1}})
+            kc.patch_merge(kc.path(*MODELS, name="m6"),
+                           {"metadata": {"annotations": {"poke": "1"}}})
+            model_uid = store.models["m6"].uid
+            await wait_for(
+                lambda: PVC_MODEL_ANN_PREFIX + "m6"
+                in ((kc.get(pvc_path)["metadata"].get("annotations")) or {}),
+                msg="PVC model annotation recorded",
+            )
+            await wait_for(lambda: kc.get_opt(load_path) is None,
+                           msg="finished loader job cleaned up")
+            await wait_for(
+                lambda: len(kc.list(f"/api/v1/namespaces/{ns}/pods")) == 1,
+                msg="server pod scheduled after cache load",
+            )
+            pod = kc.list(f"/api/v1/namespaces/{ns}/pods")[0]
+            vols = {v["name"] for v in pod["spec"]["volumes"]}
+            assert "model-cache" in vols
+            cargs = pod["spec"]["containers"][0]["args"]
+            assert cargs[cargs.index("--model") + 1] == f"/models/m6-{model_uid}"
+
+            # deletion: eviction Job runs to completion before the
+            # finalizer clears
+            store.delete_model("m6")
+            await wait_for(lambda: kc.get_opt(evict_path) is not None,
+                           msg="eviction job created")
+            assert kc.get_opt(kc.path(*MODELS, name="m6")) is not None, \
+                "finalizer must hold the CR while eviction runs"
+            kc.patch_merge(evict_path, {"status": {"conditions": [
+                {"type": "Complete", "status": "True"}]}})
+            kc.patch_merge(kc.path(*MODELS, name="m6"),
+                           {"metadata": {"annotations": {"poke": "2"}}})
+            await wait_for(lambda: not kc.list(kc.path(*MODELS)),
+                           msg="CR removed after eviction")
+            anns = (kc.get(pvc_path)["metadata"].get("annotations")) or {}
+            assert PVC_MODEL_ANN_PREFIX + "m6" not in anns
+        finally:
+            await ctrl.stop()
+            store.stop()
+
+    asyncio.run(main())
+
+
+def test_kube_cache_unknown_profile_errors(api):
+    async def main():
+        kc = kc_for(api)
+        from kubeai_amd.controlplane.kubeclient import KubeCacheManager
+
+        cache = KubeCacheManager(kc, cache_profiles={})
+
+        class M:
+            name = "x"
+            uid = "u1"
+
+            class spec:
+                cache_profile = "nope"
+
+        try:
+            await cache.ensure(M())
+            raise AssertionError("expected ValueError")
+        except ValueError as e:
+            assert "cacheProfile" in str(e)
+
+    asyncio.run(main())

@@ -45,7 +45,8 @@ __launch_bounds__(kBlockThreads) __global__ void paged_decode_kernel(
     const CT* __restrict__ v_cache,
     const int32_t* __restrict__ block_tables,  // [B, max_blocks]
     const int32_t* __restrict__ seq_lens,      // [B]
-    const float scale, const int n_kv, const int max_blocks,
+    const float scale, const int window, const int n_kv,
+    const int max_blocks,
     const int64_t q_stride, const int n_splits,
     float* __restrict__ part_o,    // [B, n_q, n_splits, hd]
     float* __restrict__ part_ml) { // [B, n_q, n_splits, 2]
@@ -58,9 +59,13 @@ __launch_bounds__(kBlockThreads) __global__ void paged_decode_kernel(
   const int lane = threadIdx.x % WAVE_SIZE;
   const int L = seq_lens[b];
   const int n_blocks = (L + kBS - 1) / kBS;
+  // sliding window: the query (position L-1) attends keys in
+  // (L-1-window, L-1] -> global positions >= w0; whole blocks below w0
+  // are skipped, the straddling block is masked per token
+  const int w0 = (window > 0) ? max(0, L - window) : 0;
   const int chunk = (n_blocks + n_splits - 1) / n_splits;
-  const int blk_lo = split * chunk;
-  const int blk_hi = min(n_blocks, blk_lo + chunk);
+  const int blk_lo = max(split * chunk, w0 / kBS);
+  const int blk_hi = min(n_blocks, split * chunk + chunk);
 
   // LDS: per-wave double-buffered KV tiles + merge scratch. Element type
   // follows the cache: bf16 stages bf16; fp8 stages the RAW e5m2 bytes
@@ -194,7 +199,9 @@ __launch_bounds__(kBlockThreads) __global__ void paged_decode_kernel(
       // reduce across the 4 part-slices -> full dot in all 4 lanes
       s[g] += __shfl_xor(s[g], 1, 64);
       s[g] += __shfl_xor(s[g], 2, 64);
-      s[g] = (tok_of < tile_len) ? s[g] * scale : -INFINITY;
+      s[g] = (tok_of < tile_len && tile_start + tok_of >= w0)
+                 ? s[g] * scale
+                 : -INFINITY;
       // tile max across tokens (bits 2..5 of the lane id)
       float tmax = s[g];
 #pragma unroll
@@ -329,7 +336,7 @@ __global__ void decode_merge_kernel(
 void paged_attention_decode(torch::Tensor out, torch::Tensor q,
                             torch::Tensor k_cache, torch::Tensor v_cache,
                             torch::Tensor block_tables, torch::Tensor seq_lens,
-                            double scale) {
+                            double scale, int64_t window) {
   TORCH_CHECK(q.stride(2) == 1 && q.stride(1) == q.size(2));
   TORCH_CHECK(out.is_contiguous());
   TORCH_CHECK(k_cache.is_contiguous() && v_cache.is_contiguous());
@@ -393,7 +400,8 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
                      (const CT*)k_cache.data_ptr(),                       \
                      (const CT*)v_cache.data_ptr(),                       \
                      block_tables.data_ptr<int32_t>(),                    \
-                     seq_lens.data_ptr<int32_t>(), (float)scale, n_kv,    \
+                     seq_lens.data_ptr<int32_t>(), (float)scale,          \
+                     (int)window, n_kv,                                   \
                      max_blocks, q.stride(0), n_splits, part_o_ptr,       \
                      part_ml_ptr)
 #define LAUNCH(GG)                                                        \

@@ -53,7 +53,8 @@ __launch_bounds__(G * WAVE_SIZE) __global__ void paged_prefill_kernel(
     const int32_t* __restrict__ block_tables,     // [B, max_blocks]
     const int32_t* __restrict__ query_start_loc,  // [B+1]
     const int32_t* __restrict__ seq_lens,         // [B]
-    const float scale, const int n_kv, const int max_blocks,
+    const float scale, const int window, const int n_kv,
+    const int max_blocks,
     const int64_t q_stride) {
   const int b = blockIdx.x;
   const int kh = blockIdx.y;
@@ -110,7 +111,13 @@ __launch_bounds__(G * WAVE_SIZE) __global__ void paged_prefill_kernel(
 
   const int32_t* bt = block_tables + (int64_t)b * max_blocks;
 
-  for (int kt = 0; kt < n_kv_tiles; ++kt) {
+  // sliding window: the smallest qpos in this tile is ctx+row_lo, so KV
+  // tiles ending at or before (ctx+row_lo - window) are masked for every
+  // row and can be skipped outright
+  const int kt0 =
+      (window > 0) ? max(0, ctx + row_lo - window + 1) / kKVB : 0;
+
+  for (int kt = kt0; kt < n_kv_tiles; ++kt) {
     const int kv_start = kt * kKVB;
     const int kv_valid = min(kKVB, kv_limit - kv_start);  // tokens staged
 
@@ -208,6 +215,10 @@ __launch_bounds__(G * WAVE_SIZE) __global__ void paged_prefill_kernel(
         const int kv1 = kv0 + 16;
         if (kv0 > qpos || r >= n_rows) s0v = -INFINITY;
         if (kv1 > qpos || r >= n_rows) s1v = -INFINITY;
+        if (window > 0) {  // sliding window: keys in (qpos-window, qpos]
+          if (kv0 <= qpos - window) s0v = -INFINITY;
+          if (kv1 <= qpos - window) s1v = -INFINITY;
+        }
         float rmax = fmaxf(s0v, s1v);
 #pragma unroll
         for (int off = 8; off > 0; off >>= 1)
@@ -285,7 +296,8 @@ void paged_attention_prefill(torch::Tensor out, torch::Tensor q,
                              torch::Tensor k_cache, torch::Tensor v_cache,
                              torch::Tensor block_tables,
                              torch::Tensor query_start_loc,
-                             torch::Tensor seq_lens, double scale) {
+                             torch::Tensor seq_lens, double scale,
+                             int64_t window) {
   TORCH_CHECK(q.stride(2) == 1 && q.stride(1) == q.size(2));
   TORCH_CHECK(out.is_contiguous());
   TORCH_CHECK(k_cache.is_contiguous() && v_cache.is_contiguous());
@@ -312,7 +324,9 @@ void paged_attention_prefill(torch::Tensor out, torch::Tensor q,
     const char* e = getenv("KUBEAI_PREFILL_V2");
     return e == nullptr || e[0] != '0';
   }();
-  if (use_v2 &&
+  // the v2 ladder has no sliding-window masking (interior-tile fast
+  // path assumes plain causal) — windowed models run the v1 kernel
+  if (window == 0 && use_v2 &&
       paged_attention_prefill_v2(out, q, k_cache, v_cache, block_tables,
                                  query_start_loc, seq_lens, scale))
     return;
@@ -329,7 +343,8 @@ void paged_attention_prefill(torch::Tensor out, torch::Tensor q,
                      (const CT*)v_cache.data_ptr(),                         \
                      block_tables.data_ptr<int32_t>(),                      \
                      query_start_loc.data_ptr<int32_t>(),                   \
-                     seq_lens.data_ptr<int32_t>(), (float)scale, n_kv,      \
+                     seq_lens.data_ptr<int32_t>(), (float)scale,            \
+                     (int)window, n_kv,                                     \
                      max_blocks, q.stride(0))
 #define LAUNCH(GG)                                                         \
   do {                                                                     \

@@ -15,12 +15,13 @@ void reshape_and_cache(torch::Tensor k, torch::Tensor v, torch::Tensor k_cache,
 void paged_attention_decode(torch::Tensor out, torch::Tensor q,
                             torch::Tensor k_cache, torch::Tensor v_cache,
                             torch::Tensor block_tables, torch::Tensor seq_lens,
-                            double scale);
+                            double scale, int64_t window);
 void paged_attention_prefill(torch::Tensor out, torch::Tensor q,
                              torch::Tensor k_cache, torch::Tensor v_cache,
                              torch::Tensor block_tables,
                              torch::Tensor query_start_loc,
-                             torch::Tensor seq_lens, double scale);
+                             torch::Tensor seq_lens, double scale,
+                             int64_t window);
 void silu_and_mul(torch::Tensor out, torch::Tensor x);
 void skinny_gemm(torch::Tensor y, torch::Tensor x, torch::Tensor w);
 void greedy_sample(torch::Tensor out, torch::Tensor logits);

@@ -33,6 +33,9 @@ class ModelArchConfig:
     attention_bias: bool = False
     # HF rope_scaling dict (llama3 / linear), None = unscaled
     rope_scaling: dict | None = None
+    # sliding-window attention (Mistral family; Gemma2 uses it on
+    # alternate layers): 0 = full causal
+    sliding_window: int = 0
     # multimodal (LLaVA-style): vision-tower config dict + the token id
     # that marks an image slot in the prompt (expanded to n_patches
     # placeholder positions at admission; models/vision.py)
@@ -87,6 +90,7 @@ class ModelArchConfig:
             num_local_experts=cfg.get("num_local_experts", 0),
             num_experts_per_tok=cfg.get("num_experts_per_tok", 2),
             rope_scaling=cfg.get("rope_scaling"),
+            sliding_window=int(cfg.get("sliding_window") or 0),
             vision=vision,
             image_token_id=image_token_id,
         )
@@ -155,6 +159,37 @@ PRESETS: dict[str, ModelArchConfig] = {
         num_key_value_heads=1,
         head_dim=128,
         max_position_embeddings=2048,
+        bos_token_id=1,
+        eos_token_id=2,
+    ),
+    # Mistral-7B v0.1/v0.2 shape: llama arch + 4096-token sliding window
+    "mistral-7b": ModelArchConfig(
+        vocab_size=32000,
+        hidden_size=4096,
+        intermediate_size=14336,
+        num_hidden_layers=32,
+        num_attention_heads=32,
+        num_key_value_heads=8,
+        head_dim=128,
+        max_position_embeddings=32768,
+        rope_theta=10000.0,
+        sliding_window=4096,
+        bos_token_id=1,
+        eos_token_id=2,
+    ),
+    # tiny windowed model for CPU tests (window smaller than the
+    # prompts the tests use, so masking actually bites)
+    "mistral-tiny": ModelArchConfig(
+        vocab_size=2048,
+        hidden_size=256,
+        intermediate_size=512,
+        num_hidden_layers=2,
+        num_attention_heads=2,
+        num_key_value_heads=1,
+        head_dim=128,
+        max_position_embeddings=2048,
+        rope_theta=10000.0,
+        sliding_window=64,
         bos_token_id=1,
         eos_token_id=2,
     ),

@@ -44,6 +44,7 @@ class Attention(nn.Module):
         self.n_q = cfg.num_attention_heads
         self.n_kv = cfg.num_key_value_heads
         self.hd = cfg.head_dim
+        self.window = cfg.sliding_window or 0
         self.scale = 1.0 / math.sqrt(self.hd)
         q_size = self.n_q * self.hd
         kv_size = self.n_kv * self.hd
@@ -97,6 +98,7 @@ class Attention(nn.Module):
                 fb.decode_seq_lens,
                 self.scale,
                 out=out[:nd],
+                window=self.window,
             )
         if fb.n_prefill > 0:
             ops.paged_attention_prefill(
@@ -108,6 +110,7 @@ class Attention(nn.Module):
                 fb.prefill_seq_lens,
                 self.scale,
                 out=out[nd:],
+                window=self.window,
             )
         attn_flat = out.view(T, -1)
         if fp8_in:

@@ -72,20 +72,23 @@ def reshape_and_cache(k, v, k_cache, v_cache, slot_mapping) -> None:
 
 
 def paged_attention_decode(
-    q, k_cache, v_cache, block_tables, seq_lens, scale: float, out=None
+    q, k_cache, v_cache, block_tables, seq_lens, scale: float, out=None,
+    window: int = 0,
 ):
     """q may be a row-strided view (fused qkv output); out must be
-    contiguous (allocated here if not supplied)."""
+    contiguous (allocated here if not supplied). window > 0 enables
+    sliding-window attention (Mistral/Gemma2 style: keys in
+    (pos-window, pos])."""
     if q.is_cuda:
         _require_ext()
         if out is None:
             out = torch.empty(q.shape, dtype=q.dtype, device=q.device)
         _C.paged_attention_decode(
-            out, q, k_cache, v_cache, block_tables, seq_lens, scale
+            out, q, k_cache, v_cache, block_tables, seq_lens, scale, window
         )
         return out
     res = ref.paged_attention_decode(
-        q, k_cache, v_cache, block_tables, seq_lens, scale
+        q, k_cache, v_cache, block_tables, seq_lens, scale, window=window
     )
     if out is not None:
         out.copy_(res)
@@ -95,18 +98,20 @@ def paged_attention_decode(
 
 def paged_attention_prefill(
     q, k_cache, v_cache, block_tables, query_start_loc, seq_lens, scale: float,
-    out=None,
+    out=None, window: int = 0,
 ):
     if q.is_cuda:
         _require_ext()
         if out is None:
             out = torch.empty(q.shape, dtype=q.dtype, device=q.device)
         _C.paged_attention_prefill(
-            out, q, k_cache, v_cache, block_tables, query_start_loc, seq_lens, scale
+            out, q, k_cache, v_cache, block_tables, query_start_loc, seq_lens,
+            scale, window
         )
         return out
     res = ref.paged_attention_prefill(
-        q, k_cache, v_cache, block_tables, query_start_loc, seq_lens, scale
+        q, k_cache, v_cache, block_tables, query_start_loc, seq_lens, scale,
+        window=window,
     )
     if out is not None:
         out.copy_(res)

@@ -160,6 +160,7 @@ def paged_attention_decode(
     block_tables: torch.Tensor,  # [B, max_blocks]
     seq_lens: torch.Tensor,  # [B] int32 (total length incl. the new token)
     scale: float,
+    window: int = 0,  # >0: sliding window, keys in (L-1-window, L-1]
 ) -> torch.Tensor:
     B, n_heads, head_dim = q.shape
     n_kv = k_cache.shape[1]
@@ -171,6 +172,8 @@ def paged_attention_decode(
         kf = k.float().repeat_interleave(g, dim=1)  # [L, n_heads, hd]
         vf = v.float().repeat_interleave(g, dim=1)
         s = torch.einsum("hd,lhd->hl", q[b].float(), kf) * scale
+        if window > 0 and L > window:
+            s[:, : L - window] = float("-inf")
         p = torch.softmax(s, dim=-1)
         out[b] = torch.einsum("hl,lhd->hd", p, vf)
     return out.to(q.dtype)
@@ -184,6 +187,7 @@ def paged_attention_prefill(
     query_start_loc: torch.Tensor,  # [B+1] int32 (prefix sums of q lens)
     seq_lens: torch.Tensor,  # [B] total kv length incl. this chunk
     scale: float,
+    window: int = 0,  # >0: query p attends keys in (p-window, p]
 ) -> torch.Tensor:
     """Causal attention of new (chunk) tokens against the full paged KV.
 
@@ -208,6 +212,8 @@ def paged_attention_prefill(
         qpos = torch.arange(L - q_len, L, device=q.device).unsqueeze(1)
         kpos = torch.arange(L, device=q.device).unsqueeze(0)
         s.masked_fill_((kpos > qpos).unsqueeze(0), float("-inf"))
+        if window > 0:
+            s.masked_fill_((kpos <= qpos - window).unsqueeze(0), float("-inf"))
         p = torch.softmax(s, dim=-1)
         out[s0:s1] = torch.einsum("hql,lhd->qhd", p, vf)
     return out.to(q.dtype)

@@ -121,6 +121,7 @@ class TPAttention(nn.Module):
         self.n_q = cfg.num_attention_heads // tp.world
         self.n_kv = cfg.num_key_value_heads // tp.world
         self.hd = cfg.head_dim
+        self.window = cfg.sliding_window or 0
         self.scale = 1.0 / math.sqrt(self.hd)
         H = cfg.hidden_size
         self.qkv_proj = EngineLinear(H, (self.n_q + 2 * self.n_kv) * self.hd, bias=False)
@@ -168,14 +169,14 @@ class TPAttention(nn.Module):
             ops.paged_attention_decode(
                 q[:nd], k_cache, v_cache,
                 fb.decode_block_tables, fb.decode_seq_lens, self.scale,
-                out=out[:nd],
+                out=out[:nd], window=getattr(self, "window", 0),
             )
         if fb.n_prefill > 0:
             ops.paged_attention_prefill(
                 q[nd:], k_cache, v_cache,
                 fb.prefill_block_tables, fb.prefill_query_start_loc,
                 fb.prefill_seq_lens, self.scale,
-                out=out[nd:],
+                out=out[nd:], window=getattr(self, "window", 0),
             )
         attn_flat = out.view(T, -1)
         if fp8_in:

@@ -350,3 +350,51 @@ def test_paged_prefill_long_context():
         q.float(), kc.float(), vc.float(), bt, qsl, sl, scale
     )
     assert_close_bf16(out, want, atol=3e-2, rtol=3e-2)
+
+
+@pytest.mark.parametrize("lens,window", [([100, 37], 48), ([2048], 512),
+                                         ([17], 64)])
+@pytest.mark.gpu
+def test_paged_decode_sliding_window(lens, window):
+    nq, nkv, hd, bs = 32, 8, 128, 16
+    B = len(lens)
+    max_blocks = max((L + bs - 1) // bs for L in lens)
+    nb = B * max_blocks + 1
+    kc, vc = _rand_cache(nb, nkv, bs, hd)
+    bt = torch.arange(1, nb, dtype=torch.int32, device=DEV).reshape(B, max_blocks)
+    seq_lens = torch.tensor(lens, dtype=torch.int32, device=DEV)
+    q = torch.randn(B, nq, hd, dtype=torch.bfloat16, device=DEV)
+    scale = 1.0 / math.sqrt(hd)
+    expected = ref.paged_attention_decode(
+        q.float(), kc.float(), vc.float(), bt, seq_lens, scale, window=window
+    )
+    out = ops.paged_attention_decode(q, kc, vc, bt, seq_lens, scale,
+                                     window=window)
+    assert_close_bf16(out, expected, atol=3e-2, rtol=3e-2)
+
+
+@pytest.mark.parametrize("q_lens,ctx_lens,window",
+                         [([120], [0], 32), ([40], [60], 24),
+                          ([33, 64], [0, 16], 48)])
+@pytest.mark.gpu
+def test_paged_prefill_sliding_window(q_lens, ctx_lens, window):
+    nq, nkv, hd, bs = 32, 8, 128, 16
+    B = len(q_lens)
+    lens = [q + c for q, c in zip(q_lens, ctx_lens)]
+    max_blocks = max((L + bs - 1) // bs for L in lens)
+    nb = B * max_blocks + 1
+    kc, vc = _rand_cache(nb, nkv, bs, hd)
+    bt = torch.arange(1, nb, dtype=torch.int32, device=DEV).reshape(B, max_blocks)
+    seq_lens = torch.tensor(lens, dtype=torch.int32, device=DEV)
+    qsl = torch.tensor([0] + list(torch.tensor(q_lens).cumsum(0)),
+                       dtype=torch.int32, device=DEV)
+    Tq = sum(q_lens)
+    q = torch.randn(Tq, nq, hd, dtype=torch.bfloat16, device=DEV)
+    scale = 1.0 / math.sqrt(hd)
+    expected = ref.paged_attention_prefill(
+        q.float(), kc.float(), vc.float(), bt, qsl, seq_lens, scale,
+        window=window,
+    )
+    out = ops.paged_attention_prefill(q, kc, vc, bt, qsl, seq_lens, scale,
+                                      window=window)
+    assert_close_bf16(out, expected, atol=3e-2, rtol=3e-2)

@@ -28,16 +28,22 @@ namespace {
 constexpr int kWaves = 2;  // 2 waves/workgroup: fits double-buffered tiles
 constexpr int kBlockThreads = kWaves * WAVE_SIZE;
 constexpr int kBS = 16;   // cache block size (tokens)
-constexpr int kHD = 128;  // head dim
 typedef __attribute__((ext_vector_type(2))) __bf16 bf16x2v;  // v_dot2 operand
 // +8 ushorts (16 B) row padding: keeps 16 B staging alignment and breaks
-// the 256 B row stride so the (token, part) score reads spread over banks
-// (4-way residual aliasing ~= 1.6x on the LDS op, vs 32-way unpadded)
+// the power-of-two row stride so the (token, part) score reads spread
+// over banks (4-way residual aliasing ~= 1.6x on the LDS op, vs 32-way
+// unpadded)
 constexpr int kPad = 8;
 
 // CT = cache element type: ushort (bf16) or unsigned char (fp8 e5m2,
 // converted to bf16 in-register while staging — same LDS layout/compute)
-template <int G, typename CT = ushort>
+// HD = head dim: 128 (llama family) or 256 (gemma2). A lane always
+// covers a 32-element slice, so HD sets PARTS = HD/32 lanes per token
+// and TOKP = 64/PARTS tokens scored per pass (two passes per 16-token
+// tile at HD=256).
+// SOFTCAP > 0 applies gemma2-style attention-logit soft capping,
+// s = cap * tanh(s / cap), after scaling and before masking.
+template <int G, typename CT = ushort, int HD = 128>
 __launch_bounds__(kBlockThreads) __global__ void paged_decode_kernel(
     ushort* __restrict__ out,            // [B, n_q, hd] bf16
     const ushort* __restrict__ q,        // [B, n_q, hd] bf16
@@ -45,8 +51,8 @@ __launch_bounds__(kBlockThreads) __global__ void paged_decode_kernel(
     const CT* __restrict__ v_cache,
     const int32_t* __restrict__ block_tables,  // [B, max_blocks]
     const int32_t* __restrict__ seq_lens,      // [B]
-    const float scale, const int window, const int n_kv,
-    const int max_blocks,
+    const float scale, const float softcap, const int window,
+    const int n_kv, const int max_blocks,
     const int64_t q_stride, const int n_splits,
     float* __restrict__ part_o,    // [B, n_q, n_splits, hd]
     float* __restrict__ part_ml) { // [B, n_q, n_splits, 2]
@@ -73,37 +79,40 @@ __launch_bounds__(kBlockThreads) __global__ void paged_decode_kernel(
   // native packed bf8->f32 instruction (v_cvt_pk_f32_bf8). 8-element row
   // pad keeps 8 B staging alignment and breaks bank aliasing for both
   // element widths (bf16: 68-dword stride, gcd 4; fp8: 34-dword, gcd 2).
-  __shared__ CT k_lds[kWaves][2][kBS][kHD + kPad];
-  __shared__ CT v_lds[kWaves][2][kBS][kHD + kPad];
-  __shared__ float merge_o[kWaves][G][kHD];
+  constexpr int PARTS = HD / 32;    // lanes per token in scoring
+  constexpr int TOKP = 64 / PARTS;  // tokens scored per pass
+  constexpr int EPL = HD / 64;      // output elems owned per lane
+  __shared__ CT k_lds[kWaves][2][kBS][HD + kPad];
+  __shared__ CT v_lds[kWaves][2][kBS][HD + kPad];
+  __shared__ float merge_o[kWaves][G][HD];
   __shared__ float merge_ml[kWaves][G][2];
 
   // --- lane roles -----------------------------------------------------
-  // scoring: 4 lanes per token (tok = lane>>2, part = lane&3), each part
-  //   covering a 32-elem slice of head_dim — all 16 tile tokens score in
-  //   parallel with a 2-step part reduction (replaces v1's 16 serial
-  //   64-lane butterflies, the VALU bottleneck)
-  // PV/output: lane owns head_dim elems {2*lane, 2*lane+1}; per-token p
-  //   reaches every lane via one __shfl broadcast
-  const int tok_of = lane >> 2;
-  const int part = lane & 3;
+  // scoring: PARTS lanes per token, each covering a 32-elem slice of
+  //   head_dim — TOKP tile tokens score in parallel with a log2(PARTS)-
+  //   step part reduction (replaces v1's serial 64-lane butterflies)
+  // PV/output: lane owns head_dim elems [EPL*lane, EPL*lane+EPL);
+  //   per-token p reaches every lane via one __shfl broadcast
+  const int tok_of = lane / PARTS;
+  const int part = lane % PARTS;
 
   // q slice for this lane's part, packed bf16 pairs (16 u32 per head)
   uint32_t q_pack[G][16];
 #pragma unroll
   for (int g = 0; g < G; ++g) {
     const uint32_t* qp = reinterpret_cast<const uint32_t*>(
-        q + (int64_t)b * q_stride + (int64_t)(kh * G + g) * kHD + part * 32);
+        q + (int64_t)b * q_stride + (int64_t)(kh * G + g) * HD + part * 32);
 #pragma unroll
     for (int i = 0; i < 16; ++i) q_pack[g][i] = qp[i];
   }
 
-  float m[G], l[G], o[G][2];
+  float m[G], l[G], o[G][EPL];
 #pragma unroll
   for (int g = 0; g < G; ++g) {
     m[g] = -INFINITY;
     l[g] = 0.f;
-    o[g][0] = o[g][1] = 0.f;
+#pragma unroll
+    for (int e = 0; e < EPL; ++e) o[g][e] = 0.f;
   }
 
   const int32_t* bt = block_tables + (int64_t)b * max_blocks;
@@ -114,26 +123,27 @@ __launch_bounds__(kBlockThreads) __global__ void paged_decode_kernel(
   // no barriers anywhere in the loop.
   // staging vector = 8 elements: 16 B (bf16) or 8 B (raw fp8 bytes)
   using StageVec = std::conditional_t<sizeof(CT) == 2, ushort8, uint64_t>;
-  StageVec stage_k[4], stage_v[4];  // this lane's 4 vectors of each tile
+  constexpr int NS = (kBS * HD / 8) / WAVE_SIZE;
+  StageVec stage_k[NS], stage_v[NS];  // this lane's vectors of each tile
 
   auto issue_loads = [&](int blk_i) {
     const int64_t blk = bt[blk_i];
-    const CT* base_k = k_cache + ((blk * n_kv + kh) * kBS) * kHD;
-    const CT* base_v = v_cache + ((blk * n_kv + kh) * kBS) * kHD;
+    const CT* base_k = k_cache + ((blk * n_kv + kh) * kBS) * HD;
+    const CT* base_v = v_cache + ((blk * n_kv + kh) * kBS) * HD;
     const StageVec* src_k = reinterpret_cast<const StageVec*>(base_k);
     const StageVec* src_v = reinterpret_cast<const StageVec*>(base_v);
 #pragma unroll
-    for (int i = 0; i < (kBS * kHD / 8) / WAVE_SIZE; ++i) {
+    for (int i = 0; i < NS; ++i) {
       stage_k[i] = src_k[lane + i * WAVE_SIZE];
       stage_v[i] = src_v[lane + i * WAVE_SIZE];
     }
   };
   auto write_tile = [&](int buf) {
 #pragma unroll
-    for (int i = 0; i < (kBS * kHD / 8) / WAVE_SIZE; ++i) {
+    for (int i = 0; i < NS; ++i) {
       const int vec = lane + i * WAVE_SIZE;  // 8-element vector index
-      const int row = vec / (kHD / 8);
-      const int col = vec % (kHD / 8);
+      const int row = vec / (HD / 8);
+      const int col = vec % (HD / 8);
       *reinterpret_cast<StageVec*>(&k_lds[wave][buf][row][col * 8]) =
           stage_k[i];
       *reinterpret_cast<StageVec*>(&v_lds[wave][buf][row][col * 8]) =
@@ -153,98 +163,113 @@ __launch_bounds__(kBlockThreads) __global__ void paged_decode_kernel(
     const int tile_len = min(kBS, L - tile_start);
     if (blk_i + kWaves < blk_hi) issue_loads(blk_i + kWaves);
 
-    // ---- scores: lane computes its (token, part-slice) partial --------
-    float s[G];
+    // TOKP tokens score per pass; kBS/TOKP passes cover the tile (one
+    // pass at HD=128, two at HD=256)
 #pragma unroll
-    for (int g = 0; g < G; ++g) s[g] = 0.f;
-    if constexpr (sizeof(CT) == 2) {
-      // v_dot2c_f32_bf16: one instruction per bf16 pair-dot-accumulate
-      // (replaces 2 unpacks + 2 fma per (g, j) — a 3x VALU cut in the
-      // score phase)
-      const uint32_t* krow = reinterpret_cast<const uint32_t*>(
-          &k_lds[wave][cur][tok_of][part * 32]);
+    for (int tp = 0; tp < kBS; tp += TOKP) {
+      const int tok = tp + tok_of;
+      // ---- scores: lane computes its (token, part-slice) partial ------
+      float s[G];
 #pragma unroll
-      for (int j = 0; j < 16; ++j) {
-        const bf16x2v kk = __builtin_bit_cast(bf16x2v, krow[j]);
+      for (int g = 0; g < G; ++g) s[g] = 0.f;
+      if constexpr (sizeof(CT) == 2) {
+        // v_dot2c_f32_bf16: one instruction per bf16 pair-dot-accumulate
+        // (replaces 2 unpacks + 2 fma per (g, j) — a 3x VALU cut in the
+        // score phase)
+        const uint32_t* krow = reinterpret_cast<const uint32_t*>(
+            &k_lds[wave][cur][tok][part * 32]);
 #pragma unroll
-        for (int g = 0; g < G; ++g)
-          s[g] = __builtin_amdgcn_fdot2_f32_bf16(
-              kk, __builtin_bit_cast(bf16x2v, q_pack[g][j]), s[g], false);
-      }
-    } else {
-      // raw e5m2 slice: each uint32 holds 4 bytes -> 4 elements; the
-      // packed bf8->f32 convert does 2 per instruction
-      const uint32_t* krow = reinterpret_cast<const uint32_t*>(
-          &k_lds[wave][cur][tok_of][part * 32]);
+        for (int j = 0; j < 16; ++j) {
+          const bf16x2v kk = __builtin_bit_cast(bf16x2v, krow[j]);
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        const uint32_t kk = krow[j];
-        const floatx2 k01 = __builtin_amdgcn_cvt_pk_f32_bf8(kk, false);
-        const floatx2 k23 = __builtin_amdgcn_cvt_pk_f32_bf8(kk, true);
+          for (int g = 0; g < G; ++g)
+            s[g] = __builtin_amdgcn_fdot2_f32_bf16(
+                kk, __builtin_bit_cast(bf16x2v, q_pack[g][j]), s[g], false);
+        }
+      } else {
+        // raw e5m2 slice: each uint32 holds 4 bytes -> 4 elements; the
+        // packed bf8->f32 convert does 2 per instruction
+        const uint32_t* krow = reinterpret_cast<const uint32_t*>(
+            &k_lds[wave][cur][tok][part * 32]);
 #pragma unroll
-        for (int g = 0; g < G; ++g) {
-          const uint32_t qa = q_pack[g][2 * j];
-          const uint32_t qb = q_pack[g][2 * j + 1];
-          s[g] = fmaf(k01.x, bf16_to_f32((ushort)(qa & 0xffff)), s[g]);
-          s[g] = fmaf(k01.y, bf16_to_f32((ushort)(qa >> 16)), s[g]);
-          s[g] = fmaf(k23.x, bf16_to_f32((ushort)(qb & 0xffff)), s[g]);
-          s[g] = fmaf(k23.y, bf16_to_f32((ushort)(qb >> 16)), s[g]);
+        for (int j = 0; j < 8; ++j) {
+          const uint32_t kk = krow[j];
+          const floatx2 k01 = __builtin_amdgcn_cvt_pk_f32_bf8(kk, false);
+          const floatx2 k23 = __builtin_amdgcn_cvt_pk_f32_bf8(kk, true);
+#pragma unroll
+          for (int g = 0; g < G; ++g) {
+            const uint32_t qa = q_pack[g][2 * j];
+            const uint32_t qb = q_pack[g][2 * j + 1];
+            s[g] = fmaf(k01.x, bf16_to_f32((ushort)(qa & 0xffff)), s[g]);
+            s[g] = fmaf(k01.y, bf16_to_f32((ushort)(qa >> 16)), s[g]);
+            s[g] = fmaf(k23.x, bf16_to_f32((ushort)(qb & 0xffff)), s[g]);
+            s[g] = fmaf(k23.y, bf16_to_f32((ushort)(qb >> 16)), s[g]);
+          }
         }
       }
-    }
-    float p[G];
-    float rsum[G];
-#pragma unroll
-    for (int g = 0; g < G; ++g) {
-      // reduce across the 4 part-slices -> full dot in all 4 lanes
-      s[g] += __shfl_xor(s[g], 1, 64);
-      s[g] += __shfl_xor(s[g], 2, 64);
-      s[g] = (tok_of < tile_len && tile_start + tok_of >= w0)
-                 ? s[g] * scale
-                 : -INFINITY;
-      // tile max across tokens (bits 2..5 of the lane id)
-      float tmax = s[g];
-#pragma unroll
-      for (int off = 4; off < 64; off <<= 1)
-        tmax = fmaxf(tmax, __shfl_xor(tmax, off, 64));
-      const float m_new = fmaxf(m[g], tmax);
-      if (m_new != m[g]) {  // defer-max: wave-uniform rescale only on growth
-        const float corr = __expf(m[g] - m_new);
-        l[g] *= corr;
-        o[g][0] *= corr;
-        o[g][1] *= corr;
-        m[g] = m_new;
-      }
-      p[g] = (s[g] == -INFINITY) ? 0.f : __expf(s[g] - m_new);
-      rsum[g] = p[g];
-#pragma unroll
-      for (int off = 1; off < 64; off <<= 1)
-        rsum[g] += __shfl_xor(rsum[g], off, 64);
-      l[g] += rsum[g] * 0.25f;  // each token counted by its 4 part-lanes
-    }
-
-    // ---- PV: lane accumulates its 2 output elems ----------------------
-#pragma unroll
-    for (int t = 0; t < kBS; ++t) {
-      if (t >= tile_len) break;
-      float v0, v1;
-      if constexpr (sizeof(CT) == 2) {
-        const uint32_t vv =
-            *reinterpret_cast<const uint32_t*>(&v_lds[wave][cur][t][2 * lane]);
-        v0 = bf16_to_f32((ushort)(vv & 0xffff));
-        v1 = bf16_to_f32((ushort)(vv >> 16));
-      } else {
-        const uint32_t vv =
-            *reinterpret_cast<const ushort*>(&v_lds[wave][cur][t][2 * lane]);
-        const floatx2 vf = __builtin_amdgcn_cvt_pk_f32_bf8(vv, false);
-        v0 = vf.x;
-        v1 = vf.y;
-      }
+      float p[G];
+      float rsum[G];
 #pragma unroll
       for (int g = 0; g < G; ++g) {
-        const float pt = __shfl(p[g], t * 4, 64);
-        o[g][0] = fmaf(pt, v0, o[g][0]);
-        o[g][1] = fmaf(pt, v1, o[g][1]);
+        // reduce across the part-slices -> full dot in all PARTS lanes
+#pragma unroll
+        for (int off = 1; off < PARTS; off <<= 1)
+          s[g] += __shfl_xor(s[g], off, 64);
+        s[g] = (tok < tile_len && tile_start + tok >= w0) ? s[g] * scale
+                                                          : -INFINITY;
+        if (softcap > 0.f && s[g] != -INFINITY)
+          s[g] = softcap * tanhf(s[g] / softcap);
+        // pass max across tokens (lane bits above the part bits)
+        float tmax = s[g];
+#pragma unroll
+        for (int off = PARTS; off < 64; off <<= 1)
+          tmax = fmaxf(tmax, __shfl_xor(tmax, off, 64));
+        const float m_new = fmaxf(m[g], tmax);
+        if (m_new != m[g]) {  // defer-max: wave-uniform rescale on growth
+          const float corr = __expf(m[g] - m_new);
+          l[g] *= corr;
+#pragma unroll
+          for (int e = 0; e < EPL; ++e) o[g][e] *= corr;
+          m[g] = m_new;
+        }
+        p[g] = (s[g] == -INFINITY) ? 0.f : __expf(s[g] - m_new);
+        rsum[g] = p[g];
+#pragma unroll
+        for (int off = 1; off < 64; off <<= 1)
+          rsum[g] += __shfl_xor(rsum[g], off, 64);
+        l[g] += rsum[g] * (1.0f / PARTS);  // each token counted PARTS x
+      }
+
+      // ---- PV: lane accumulates its EPL output elems ------------------
+#pragma unroll
+      for (int t = tp; t < tp + TOKP; ++t) {
+        if (t >= tile_len) break;
+        float vv[EPL];
+        if constexpr (sizeof(CT) == 2) {
+#pragma unroll
+          for (int e = 0; e < EPL; e += 2) {
+            const uint32_t vr = *reinterpret_cast<const uint32_t*>(
+                &v_lds[wave][cur][t][EPL * lane + e]);
+            vv[e] = bf16_to_f32((ushort)(vr & 0xffff));
+            vv[e + 1] = bf16_to_f32((ushort)(vr >> 16));
+          }
+        } else {
+#pragma unroll
+          for (int e = 0; e < EPL; e += 2) {
+            const uint32_t vr = *reinterpret_cast<const ushort*>(
+                &v_lds[wave][cur][t][EPL * lane + e]);
+            const floatx2 vf = __builtin_amdgcn_cvt_pk_f32_bf8(vr, false);
+            vv[e] = vf.x;
+            vv[e + 1] = vf.y;
+          }
+        }
+#pragma unroll
+        for (int g = 0; g < G; ++g) {
+          const float pt = __shfl(p[g], (t - tp) * PARTS, 64);
+#pragma unroll
+          for (int e = 0; e < EPL; ++e)
+            o[g][e] = fmaf(pt, vv[e], o[g][e]);
+        }
       }
     }
 
@@ -256,8 +281,8 @@ __launch_bounds__(kBlockThreads) __global__ void paged_decode_kernel(
   // cross-wave merge through LDS
 #pragma unroll
   for (int g = 0; g < G; ++g) {
-    merge_o[wave][g][2 * lane] = o[g][0];
-    merge_o[wave][g][2 * lane + 1] = o[g][1];
+#pragma unroll
+    for (int e = 0; e < EPL; ++e) merge_o[wave][g][EPL * lane + e] = o[g][e];
     if (lane == 0) {
       merge_ml[wave][g][0] = m[g];
       merge_ml[wave][g][1] = l[g];
@@ -265,33 +290,40 @@ __launch_bounds__(kBlockThreads) __global__ void paged_decode_kernel(
   }
   __syncthreads();
 
-  // wave w finalizes heads g = w, w+4, ... (for G<4, waves w>=G idle here)
+  // wave w finalizes heads g = w, w+2, ... (for G<2, wave 1 idles here)
   for (int g = wave; g < G; g += kWaves) {
     float m_star = -INFINITY;
 #pragma unroll
     for (int w = 0; w < kWaves; ++w) m_star = fmaxf(m_star, merge_ml[w][g][0]);
-    float l_star = 0.f, o0 = 0.f, o1 = 0.f;
+    float l_star = 0.f, oo[EPL];
+#pragma unroll
+    for (int e = 0; e < EPL; ++e) oo[e] = 0.f;
     if (m_star != -INFINITY) {
 #pragma unroll
       for (int w = 0; w < kWaves; ++w) {
         const float c = __expf(merge_ml[w][g][0] - m_star);
         l_star += merge_ml[w][g][1] * c;
-        o0 += merge_o[w][g][2 * lane] * c;
-        o1 += merge_o[w][g][2 * lane + 1] * c;
+#pragma unroll
+        for (int e = 0; e < EPL; ++e)
+          oo[e] += merge_o[w][g][EPL * lane + e] * c;
       }
     }
     const int head = kh * G + g;
     if (n_splits == 1) {
       const float inv_l = 1.0f / l_star;
-      ushort* oh = out + ((int64_t)b * n_q + head) * kHD;
-      uint32_t packed = ((uint32_t)f32_to_bf16(o1 * inv_l) << 16) |
-                        f32_to_bf16(o0 * inv_l);
-      *reinterpret_cast<uint32_t*>(&oh[2 * lane]) = packed;
+      ushort* oh = out + ((int64_t)b * n_q + head) * HD;
+#pragma unroll
+      for (int e = 0; e < EPL; e += 2) {
+        uint32_t packed =
+            ((uint32_t)f32_to_bf16(oo[e + 1] * inv_l) << 16) |
+            f32_to_bf16(oo[e] * inv_l);
+        *reinterpret_cast<uint32_t*>(&oh[EPL * lane + e]) = packed;
+      }
     } else {
       // flash-decoding partials (empty splits write m=-inf, l=0, o=0)
-      float* po = part_o + (((int64_t)b * n_q + head) * n_splits + split) * kHD;
-      po[2 * lane] = o0;
-      po[2 * lane + 1] = o1;
+      float* po = part_o + (((int64_t)b * n_q + head) * n_splits + split) * HD;
+#pragma unroll
+      for (int e = 0; e < EPL; ++e) po[EPL * lane + e] = oo[e];
       if (lane == 0) {
         float* pm = part_ml + (((int64_t)b * n_q + head) * n_splits + split) * 2;
         pm[0] = m_star;
@@ -302,33 +334,40 @@ __launch_bounds__(kBlockThreads) __global__ void paged_decode_kernel(
 }
 
 // Merge flash-decoding partials: one wave per (seq, q_head).
+template <int HD = 128>
 __global__ void decode_merge_kernel(
     ushort* __restrict__ out,             // [B, n_q, hd]
     const float* __restrict__ part_o,     // [B, n_q, n_splits, hd]
     const float* __restrict__ part_ml,    // [B, n_q, n_splits, 2]
     const int n_q, const int n_splits, const int64_t n_bh) {
+  constexpr int EPL = HD / 64;
   const int64_t bh = (int64_t)blockIdx.x * (blockDim.x / WAVE_SIZE) +
                      threadIdx.x / WAVE_SIZE;
   if (bh >= n_bh) return;
   const int lane = threadIdx.x % WAVE_SIZE;
   const float* pml = part_ml + bh * n_splits * 2;
-  const float* po = part_o + bh * n_splits * kHD;
+  const float* po = part_o + bh * n_splits * HD;
   float m_star = -INFINITY;
   for (int s = 0; s < n_splits; ++s) m_star = fmaxf(m_star, pml[2 * s]);
-  float l_star = 0.f, o0 = 0.f, o1 = 0.f;
+  float l_star = 0.f, oo[EPL];
+#pragma unroll
+  for (int e = 0; e < EPL; ++e) oo[e] = 0.f;
   for (int s = 0; s < n_splits; ++s) {
     const float ms = pml[2 * s];
     if (ms == -INFINITY) continue;
     const float c = __expf(ms - m_star);
     l_star += pml[2 * s + 1] * c;
-    o0 += po[s * kHD + 2 * lane] * c;
-    o1 += po[s * kHD + 2 * lane + 1] * c;
+#pragma unroll
+    for (int e = 0; e < EPL; ++e) oo[e] += po[s * HD + EPL * lane + e] * c;
   }
   const float inv_l = 1.0f / l_star;
-  ushort* oh = out + bh * kHD;
-  uint32_t packed = ((uint32_t)f32_to_bf16(o1 * inv_l) << 16) |
-                    f32_to_bf16(o0 * inv_l);
-  *reinterpret_cast<uint32_t*>(&oh[2 * lane]) = packed;
+  ushort* oh = out + bh * HD;
+#pragma unroll
+  for (int e = 0; e < EPL; e += 2) {
+    uint32_t packed = ((uint32_t)f32_to_bf16(oo[e + 1] * inv_l) << 16) |
+                      f32_to_bf16(oo[e] * inv_l);
+    *reinterpret_cast<uint32_t*>(&oh[EPL * lane + e]) = packed;
+  }
 }
 
 }  // namespace
@@ -336,7 +375,7 @@ __global__ void decode_merge_kernel(
 void paged_attention_decode(torch::Tensor out, torch::Tensor q,
                             torch::Tensor k_cache, torch::Tensor v_cache,
                             torch::Tensor block_tables, torch::Tensor seq_lens,
-                            double scale, int64_t window) {
+                            double scale, int64_t window, double softcap) {
   TORCH_CHECK(q.stride(2) == 1 && q.stride(1) == q.size(2));
   TORCH_CHECK(out.is_contiguous());
   TORCH_CHECK(k_cache.is_contiguous() && v_cache.is_contiguous());
@@ -348,7 +387,8 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
   const int B = q.size(0), n_q = q.size(1), hd = q.size(2);
   const int n_kv = k_cache.size(1);
   const int max_blocks = block_tables.size(1);
-  TORCH_CHECK(hd == kHD, "decode kernel supports head_dim=128");
+  TORCH_CHECK(hd == 128 || hd == 256,
+              "decode kernel supports head_dim 128 or 256");
   TORCH_CHECK(k_cache.size(2) == kBS, "decode kernel supports block_size=16");
   const int G = n_q / n_kv;
   TORCH_CHECK(n_q % n_kv == 0);
@@ -385,7 +425,7 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
   if (n_splits > 1) {
     auto opts =
         torch::TensorOptions().device(q.device()).dtype(torch::kFloat32);
-    part_o = torch::empty({B, n_q, n_splits, (int64_t)kHD}, opts);
+    part_o = torch::empty({B, n_q, n_splits, (int64_t)hd}, opts);
     part_ml = torch::empty({B, n_q, n_splits, 2}, opts);
     part_o_ptr = part_o.data_ptr<float>();
     part_ml_ptr = part_ml.data_ptr<float>();
@@ -393,23 +433,27 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
   dim3 grid(B * n_kv * n_splits), block(kBlockThreads);
   auto stream = c10::hip::getCurrentHIPStream().stream();
   const bool fp8_cache = k_cache.scalar_type() == torch::kFloat8_e5m2;
-#define LAUNCH_CT(GG, CT)                                                 \
-  hipLaunchKernelGGL((paged_decode_kernel<GG, CT>), grid, block, 0,       \
+#define LAUNCH_CT_HD(GG, CT, HDV)                                         \
+  hipLaunchKernelGGL((paged_decode_kernel<GG, CT, HDV>), grid, block, 0,  \
                      stream, (ushort*)out.data_ptr(),                     \
                      (const ushort*)q.data_ptr(),                         \
                      (const CT*)k_cache.data_ptr(),                       \
                      (const CT*)v_cache.data_ptr(),                       \
                      block_tables.data_ptr<int32_t>(),                    \
                      seq_lens.data_ptr<int32_t>(), (float)scale,          \
-                     (int)window, n_kv,                                   \
+                     (float)softcap, (int)window, n_kv,                   \
                      max_blocks, q.stride(0), n_splits, part_o_ptr,       \
                      part_ml_ptr)
 #define LAUNCH(GG)                                                        \
   do {                                                                    \
-    if (fp8_cache) {                                                      \
-      LAUNCH_CT(GG, unsigned char);                                       \
+    if (hd == 256) {                                                      \
+      /* gemma2 shapes: G <= 2 in practice, fp8 cache supported */        \
+      if (fp8_cache) LAUNCH_CT_HD(GG, unsigned char, 256);                \
+      else LAUNCH_CT_HD(GG, ushort, 256);                                 \
+    } else if (fp8_cache) {                                               \
+      LAUNCH_CT_HD(GG, unsigned char, 128);                               \
     } else {                                                              \
-      LAUNCH_CT(GG, ushort);                                              \
+      LAUNCH_CT_HD(GG, ushort, 128);                                      \
     }                                                                     \
   } while (0)
   switch (G) {
@@ -422,16 +466,23 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
     default: TORCH_CHECK(false, "unsupported GQA group size ", G);
   }
 #undef LAUNCH
-#undef LAUNCH_CT
+#undef LAUNCH_CT_HD
   HIP_CHECK_KERNEL();
   if (n_splits > 1) {
     const int64_t n_bh = (int64_t)B * n_q;
     const int wpb = 4;
-    hipLaunchKernelGGL(decode_merge_kernel,
-                       dim3((uint32_t)((n_bh + wpb - 1) / wpb)),
-                       dim3(wpb * WAVE_SIZE), 0, stream,
-                       (ushort*)out.data_ptr(), part_o_ptr, part_ml_ptr, n_q,
-                       n_splits, n_bh);
+    if (hd == 256)
+      hipLaunchKernelGGL((decode_merge_kernel<256>),
+                         dim3((uint32_t)((n_bh + wpb - 1) / wpb)),
+                         dim3(wpb * WAVE_SIZE), 0, stream,
+                         (ushort*)out.data_ptr(), part_o_ptr, part_ml_ptr,
+                         n_q, n_splits, n_bh);
+    else
+      hipLaunchKernelGGL((decode_merge_kernel<128>),
+                         dim3((uint32_t)((n_bh + wpb - 1) / wpb)),
+                         dim3(wpb * WAVE_SIZE), 0, stream,
+                         (ushort*)out.data_ptr(), part_o_ptr, part_ml_ptr,
+                         n_q, n_splits, n_bh);
     HIP_CHECK_KERNEL();
   }
 }

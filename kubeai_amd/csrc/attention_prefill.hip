@@ -29,7 +29,6 @@
 
 namespace {
 
-constexpr int kHD = 128;   // head dim
 constexpr int kBS = 16;    // cache block size
 constexpr int kQB = 16;    // query rows per tile (32 costs occupancy: 201 VGPR -> 1 wave/SIMD, net loss)
 constexpr int kMT = kQB / 16;
@@ -44,7 +43,11 @@ DEV_INLINE int swz(int row, int byte_off) {
 
 // CT = cache element type: ushort (bf16) or unsigned char (fp8 e5m2,
 // converted to bf16 while staging — LDS layout and MFMA path unchanged)
-template <int G, typename CT = ushort>
+// HD = head dim: 128 (llama family) or 256 (gemma2; K-chunk and
+// PV-column loops scale with HD/32 and HD/16).
+// SOFTCAP: gemma2 attention-logit capping s = cap*tanh(s/cap), applied
+// after scale, before masking (0 disables; passed at runtime).
+template <int G, typename CT = ushort, int HD = 128>
 __launch_bounds__(G * WAVE_SIZE) __global__ void paged_prefill_kernel(
     ushort* __restrict__ out,            // [Tq, n_q, hd]
     const ushort* __restrict__ q,        // [Tq, n_q, hd]
@@ -53,8 +56,8 @@ __launch_bounds__(G * WAVE_SIZE) __global__ void paged_prefill_kernel(
     const int32_t* __restrict__ block_tables,     // [B, max_blocks]
     const int32_t* __restrict__ query_start_loc,  // [B+1]
     const int32_t* __restrict__ seq_lens,         // [B]
-    const float scale, const int window, const int n_kv,
-    const int max_blocks,
+    const float scale, const float softcap, const int window,
+    const int n_kv, const int max_blocks,
     const int64_t q_stride) {
   const int b = blockIdx.x;
   const int kh = blockIdx.y;
@@ -77,29 +80,29 @@ __launch_bounds__(G * WAVE_SIZE) __global__ void paged_prefill_kernel(
   const int kv_limit = ctx + row_lo + n_rows;  // exclusive
   const int n_kv_tiles = (kv_limit + kKVB - 1) / kKVB;
 
-  __shared__ ushort k_lds[kKVB * kHD];          // [tok][hd], swizzled
-  __shared__ ushort v_lds[kHD * kKVB];          // [hd][tok], swizzled
+  __shared__ ushort k_lds[kKVB * HD];          // [tok][hd], swizzled
+  __shared__ ushort v_lds[HD * kKVB];          // [hd][tok], swizzled
   __shared__ ushort p_lds[G][16 * kKVB];        // [row][tok], swizzled
 
   // ---- load Q fragments (2 sub-tiles x 4 K-chunks) from global ----
-  bf16x8 q_frag[kMT][4];
+  bf16x8 q_frag[kMT][HD / 32];
 #pragma unroll
   for (int mt = 0; mt < kMT; ++mt) {
     const int qrow = min(row_lo + mt * 16 + (lane & 15), q_len - 1);
     const ushort* qp =
-        q + (int64_t)(s0 + qrow) * q_stride + (int64_t)head * kHD;
+        q + (int64_t)(s0 + qrow) * q_stride + (int64_t)head * HD;
 #pragma unroll
-    for (int kc = 0; kc < 4; ++kc) {
+    for (int kc = 0; kc < HD / 32; ++kc) {
       q_frag[mt][kc] =
           *reinterpret_cast<const bf16x8*>(qp + kc * 32 + (lane >> 4) * 8);
     }
   }
 
-  f32x4 o_acc[kMT][kHD / 16];
+  f32x4 o_acc[kMT][HD / 16];
 #pragma unroll
   for (int mt = 0; mt < kMT; ++mt)
 #pragma unroll
-    for (int c = 0; c < kHD / 16; ++c) o_acc[mt][c] = {0.f, 0.f, 0.f, 0.f};
+    for (int c = 0; c < HD / 16; ++c) o_acc[mt][c] = {0.f, 0.f, 0.f, 0.f};
   float m_run[kMT][4], l_run[kMT][4];
 #pragma unroll
   for (int mt = 0; mt < kMT; ++mt)
@@ -124,10 +127,10 @@ __launch_bounds__(G * WAVE_SIZE) __global__ void paged_prefill_kernel(
     // ---- cooperative staging: K -> k_lds, V -> v_lds transposed ----
     __syncthreads();  // previous tile fully consumed
     {
-      const int nvec = kKVB * kHD / 8;  // 512
+      const int nvec = kKVB * HD / 8;
       for (int i = threadIdx.x; i < nvec; i += G * WAVE_SIZE) {
-        const int tok = i / (kHD / 8);
-        const int col8 = i % (kHD / 8);
+        const int tok = i / (HD / 8);
+        const int col8 = i % (HD / 8);
         if (tok >= kv_valid) {
           // zero unstaged V: P rows are 0 there, but 0 * stale-NaN would
           // poison the PV MFMA accumulator (K can stay stale: scores for
@@ -144,9 +147,9 @@ __launch_bounds__(G * WAVE_SIZE) __global__ void paged_prefill_kernel(
           const int abs_tok = kv_start + tok;
           const int64_t blk = bt[abs_tok / kBS];
           const CT* src = k_cache +
-                          (((blk * n_kv + kh) * kBS) + abs_tok % kBS) * kHD;
+                          (((blk * n_kv + kh) * kBS) + abs_tok % kBS) * HD;
           const CT* vsrc = v_cache +
-                           (((blk * n_kv + kh) * kBS) + abs_tok % kBS) * kHD;
+                           (((blk * n_kv + kh) * kBS) + abs_tok % kBS) * HD;
           bf16x8 kk, vv;
           if constexpr (sizeof(CT) == 2) {
             kk = *reinterpret_cast<const bf16x8*>(src + col8 * 8);
@@ -166,7 +169,7 @@ __launch_bounds__(G * WAVE_SIZE) __global__ void paged_prefill_kernel(
           // K: 16B vector write, swizzled row=tok
           *reinterpret_cast<bf16x8*>(
               reinterpret_cast<char*>(k_lds) +
-              swz(tok, tok * kHD * 2 + col8 * 16)) = kk;
+              swz(tok, tok * HD * 2 + col8 * 16)) = kk;
           // V: scatter-transpose 8 elems (row=hd, col=tok)
 #pragma unroll
           for (int j = 0; j < 8; ++j) {
@@ -192,11 +195,11 @@ __launch_bounds__(G * WAVE_SIZE) __global__ void paged_prefill_kernel(
       for (int nsub = 0; nsub < 2; ++nsub) {
         s_frag[nsub] = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
-        for (int kc = 0; kc < 4; ++kc) {
+        for (int kc = 0; kc < HD / 32; ++kc) {
           const int tok = nsub * 16 + (lane & 15);
           const bf16x8 k_frag = *reinterpret_cast<const bf16x8*>(
               reinterpret_cast<const char*>(k_lds) +
-              swz(tok, tok * kHD * 2 + kc * 64 + (lane >> 4) * 16));
+              swz(tok, tok * HD * 2 + kc * 64 + (lane >> 4) * 16));
           s_frag[nsub] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               q_frag[mt][kc], k_frag, s_frag[nsub], 0, 0, 0);
         }
@@ -211,6 +214,10 @@ __launch_bounds__(G * WAVE_SIZE) __global__ void paged_prefill_kernel(
         const int qpos = ctx + row_lo + r;
         float s0v = s_frag[0][i] * scale;
         float s1v = s_frag[1][i] * scale;
+        if (softcap > 0.f) {
+          s0v = softcap * tanhf(s0v / softcap);
+          s1v = softcap * tanhf(s1v / softcap);
+        }
         const int kv0 = kv_start + (lane & 15);
         const int kv1 = kv0 + 16;
         if (kv0 > qpos || r >= n_rows) s0v = -INFINITY;
@@ -235,7 +242,7 @@ __launch_bounds__(G * WAVE_SIZE) __global__ void paged_prefill_kernel(
         m_run[mt][i] = m_new;
       }
 #pragma unroll
-      for (int c = 0; c < kHD / 16; ++c)
+      for (int c = 0; c < HD / 16; ++c)
 #pragma unroll
         for (int i = 0; i < 4; ++i) o_acc[mt][c][i] *= corr[i];
 
@@ -257,7 +264,7 @@ __launch_bounds__(G * WAVE_SIZE) __global__ void paged_prefill_kernel(
 
       // ---- O += P.V ----
 #pragma unroll
-      for (int c = 0; c < kHD / 16; ++c) {
+      for (int c = 0; c < HD / 16; ++c) {
         const int hdcol = c * 16 + (lane & 15);
         const bf16x8 v_frag = *reinterpret_cast<const bf16x8*>(
             reinterpret_cast<const char*>(v_lds) +
@@ -276,9 +283,9 @@ __launch_bounds__(G * WAVE_SIZE) __global__ void paged_prefill_kernel(
       const int r = mt * 16 + (lane >> 4) * 4 + i;
       if (r >= n_rows) continue;
       const float inv_l = 1.0f / l_run[mt][i];
-      ushort* op = out + ((int64_t)(s0 + row_lo + r) * n_q + head) * kHD;
+      ushort* op = out + ((int64_t)(s0 + row_lo + r) * n_q + head) * HD;
 #pragma unroll
-      for (int c = 0; c < kHD / 16; ++c)
+      for (int c = 0; c < HD / 16; ++c)
         op[c * 16 + (lane & 15)] = f32_to_bf16(o_acc[mt][c][i] * inv_l);
     }
 }
@@ -297,7 +304,7 @@ void paged_attention_prefill(torch::Tensor out, torch::Tensor q,
                              torch::Tensor block_tables,
                              torch::Tensor query_start_loc,
                              torch::Tensor seq_lens, double scale,
-                             int64_t window) {
+                             int64_t window, double softcap) {
   TORCH_CHECK(q.stride(2) == 1 && q.stride(1) == q.size(2));
   TORCH_CHECK(out.is_contiguous());
   TORCH_CHECK(k_cache.is_contiguous() && v_cache.is_contiguous());
@@ -311,7 +318,8 @@ void paged_attention_prefill(torch::Tensor out, torch::Tensor q,
   const int n_kv = k_cache.size(1);
   const int B = seq_lens.size(0);
   const int max_blocks = block_tables.size(1);
-  TORCH_CHECK(hd == kHD, "prefill kernel supports head_dim=128");
+  TORCH_CHECK(hd == 128 || hd == 256,
+              "prefill kernel supports head_dim 128 or 256");
   TORCH_CHECK(k_cache.size(2) == kBS);
   const int G = n_q / n_kv;
   TORCH_CHECK(n_q % n_kv == 0);
@@ -326,7 +334,8 @@ void paged_attention_prefill(torch::Tensor out, torch::Tensor q,
   }();
   // the v2 ladder has no sliding-window masking (interior-tile fast
   // path assumes plain causal) — windowed models run the v1 kernel
-  if (window == 0 && use_v2 &&
+  // the v2 ladder covers the plain-causal hd=128 fast path only
+  if (window == 0 && softcap == 0.0 && hd == 128 && use_v2 &&
       paged_attention_prefill_v2(out, q, k_cache, v_cache, block_tables,
                                  query_start_loc, seq_lens, scale))
     return;
@@ -335,8 +344,8 @@ void paged_attention_prefill(torch::Tensor out, torch::Tensor q,
   dim3 grid(B, n_kv, n_qtiles_max);
   auto stream = c10::hip::getCurrentHIPStream().stream();
   const bool fp8_cache = k_cache.scalar_type() == torch::kFloat8_e5m2;
-#define LAUNCH_CT(GG, CT)                                                  \
-  hipLaunchKernelGGL((paged_prefill_kernel<GG, CT>), grid,                  \
+#define LAUNCH_CT_HD(GG, CT, HDV)                                          \
+  hipLaunchKernelGGL((paged_prefill_kernel<GG, CT, HDV>), grid,             \
                      dim3(GG * WAVE_SIZE), 0, stream,                       \
                      (ushort*)out.data_ptr(), (const ushort*)q.data_ptr(),  \
                      (const CT*)k_cache.data_ptr(),                         \
@@ -344,15 +353,18 @@ void paged_attention_prefill(torch::Tensor out, torch::Tensor q,
                      block_tables.data_ptr<int32_t>(),                      \
                      query_start_loc.data_ptr<int32_t>(),                   \
                      seq_lens.data_ptr<int32_t>(), (float)scale,            \
-                     (int)window, n_kv,                                     \
+                     (float)softcap, (int)window, n_kv,                     \
                      max_blocks, q.stride(0))
 #define LAUNCH(GG)                                                         \
   do {                                                                     \
-    if (fp8_cache) {                                                       \
-      LAUNCH_CT(GG, unsigned char);                                        \
-    } else {                                                               \
-      LAUNCH_CT(GG, ushort);                                               \
-    }                                                                      \
+    if (hd == 256) {                                                       \
+      if (fp8_cache) LAUNCH_CT_HD(GG, unsigned char, 256);                  \
+      else LAUNCH_CT_HD(GG, ushort, 256);                                   \
+    } else if (fp8_cache) {                                                 \
+      LAUNCH_CT_HD(GG, unsigned char, 128);                                 \
+    } else {                                                                \
+      LAUNCH_CT_HD(GG, ushort, 128);                                        \
+    }                                                                       \
   } while (0)
   switch (G) {
     case 1: LAUNCH(1); break;
@@ -362,6 +374,6 @@ void paged_attention_prefill(torch::Tensor out, torch::Tensor q,
     default: TORCH_CHECK(false, "unsupported GQA group size ", G);
   }
 #undef LAUNCH
-#undef LAUNCH_CT
+#undef LAUNCH_CT_HD
   HIP_CHECK_KERNEL();
 }

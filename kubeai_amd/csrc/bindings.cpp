@@ -15,14 +15,15 @@ void reshape_and_cache(torch::Tensor k, torch::Tensor v, torch::Tensor k_cache,
 void paged_attention_decode(torch::Tensor out, torch::Tensor q,
                             torch::Tensor k_cache, torch::Tensor v_cache,
                             torch::Tensor block_tables, torch::Tensor seq_lens,
-                            double scale, int64_t window);
+                            double scale, int64_t window, double softcap);
 void paged_attention_prefill(torch::Tensor out, torch::Tensor q,
                              torch::Tensor k_cache, torch::Tensor v_cache,
                              torch::Tensor block_tables,
                              torch::Tensor query_start_loc,
                              torch::Tensor seq_lens, double scale,
-                             int64_t window);
+                             int64_t window, double softcap);
 void silu_and_mul(torch::Tensor out, torch::Tensor x);
+void gelu_and_mul(torch::Tensor out, torch::Tensor x);
 void skinny_gemm(torch::Tensor y, torch::Tensor x, torch::Tensor w);
 void greedy_sample(torch::Tensor out, torch::Tensor logits);
 void gumbel_sample(torch::Tensor out, torch::Tensor logits,
@@ -67,6 +68,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "paged attention, one query token per seq");
   m.def("paged_attention_prefill", &paged_attention_prefill,
         "paged causal flash attention over cached KV");
+  m.def("gelu_and_mul", &gelu_and_mul, "GeGLU tanh-gelu(gate)*up");
   m.def("silu_and_mul", &silu_and_mul, "SwiGLU activation");
   m.def("skinny_gemm", &skinny_gemm, "weight-streaming GEMM for M<=64");
   m.def("greedy_sample", &greedy_sample, "argmax sampling");

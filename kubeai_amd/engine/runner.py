@@ -125,6 +125,11 @@ class ModelRunner:
         if tp_group is not None:
             from kubeai_amd.parallel.tp import TPLlamaForCausalLM
 
+            if arch.post_norms:
+                raise NotImplementedError(
+                    "tensor parallelism is not implemented for the gemma2 "
+                    "layer structure (post-norms); run gemma2 replicated"
+                )
             self.model = TPLlamaForCausalLM(
                 arch, tp_group, device=self.device, dtype=dtype, seed=seed
             )

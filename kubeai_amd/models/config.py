@@ -36,6 +36,17 @@ class ModelArchConfig:
     # sliding-window attention (Mistral family; Gemma2 uses it on
     # alternate layers): 0 = full causal
     sliding_window: int = 0
+    # "all": every layer windowed (Mistral); "even": layers 0,2,4,..
+    # windowed (Gemma2 layer_types pattern)
+    sliding_window_pattern: str = "all"
+    # gemma2 family knobs
+    hidden_act: str = "silu"  # or "gelu_pytorch_tanh" (GeGLU)
+    norm_plus_one: bool = False  # RMSNorm multiplies by (1 + w)
+    scale_embeddings: bool = False  # x *= sqrt(hidden) after embedding
+    post_norms: bool = False  # extra post-attn/pre+post-FFN norms
+    attn_logit_softcap: float = 0.0  # s = c*tanh(s/c) on attention logits
+    final_logit_softcap: float = 0.0  # same cap on lm_head logits
+    query_pre_attn_scalar: float = 0.0  # 0 -> head_dim (scale = x^-0.5)
     # multimodal (LLaVA-style): vision-tower config dict + the token id
     # that marks an image slot in the prompt (expanded to n_patches
     # placeholder positions at admission; models/vision.py)
@@ -53,6 +64,7 @@ class ModelArchConfig:
         hidden = cfg["hidden_size"]
         n_heads = cfg["num_attention_heads"]
         arch = (cfg.get("architectures") or ["LlamaForCausalLM"])[0]
+        gemma2 = arch == "Gemma2ForCausalLM"
         vision = None
         image_token_id = -1
         if arch == "LlavaForConditionalGeneration":
@@ -84,13 +96,31 @@ class ModelArchConfig:
             max_position_embeddings=cfg.get("max_position_embeddings", 8192),
             rms_norm_eps=cfg.get("rms_norm_eps", 1e-5),
             rope_theta=cfg.get("rope_theta", 10000.0),
-            tie_word_embeddings=cfg.get("tie_word_embeddings", False),
+            tie_word_embeddings=cfg.get("tie_word_embeddings", gemma2),
             bos_token_id=cfg.get("bos_token_id", 1),
             eos_token_id=eos,
             num_local_experts=cfg.get("num_local_experts", 0),
             num_experts_per_tok=cfg.get("num_experts_per_tok", 2),
             rope_scaling=cfg.get("rope_scaling"),
             sliding_window=int(cfg.get("sliding_window") or 0),
+            sliding_window_pattern="even" if gemma2 else "all",
+            hidden_act=(
+                "gelu_pytorch_tanh"
+                if gemma2 or "gelu" in str(cfg.get("hidden_act") or "")
+                else "silu"
+            ),
+            norm_plus_one=gemma2,
+            scale_embeddings=gemma2,
+            post_norms=gemma2,
+            attn_logit_softcap=float(
+                cfg.get("attn_logit_softcapping") or 0.0
+            ) if gemma2 else 0.0,
+            final_logit_softcap=float(
+                cfg.get("final_logit_softcapping") or 0.0
+            ) if gemma2 else 0.0,
+            query_pre_attn_scalar=float(
+                cfg.get("query_pre_attn_scalar") or 0.0
+            ) if gemma2 else 0.0,
             vision=vision,
             image_token_id=image_token_id,
         )
@@ -161,6 +191,57 @@ PRESETS: dict[str, ModelArchConfig] = {
         max_position_embeddings=2048,
         bos_token_id=1,
         eos_token_id=2,
+    ),
+    # Gemma2-2b shape: GeGLU, (1+w) norms, post-norms, logit softcaps,
+    # head_dim 256, alternate-layer sliding window
+    "gemma2-2b": ModelArchConfig(
+        vocab_size=256000,
+        hidden_size=2304,
+        intermediate_size=9216,
+        num_hidden_layers=26,
+        num_attention_heads=8,
+        num_key_value_heads=4,
+        head_dim=256,
+        max_position_embeddings=8192,
+        rope_theta=10000.0,
+        rms_norm_eps=1e-6,
+        tie_word_embeddings=True,
+        bos_token_id=2,
+        eos_token_id=1,
+        sliding_window=4096,
+        sliding_window_pattern="even",
+        hidden_act="gelu_pytorch_tanh",
+        norm_plus_one=True,
+        scale_embeddings=True,
+        post_norms=True,
+        attn_logit_softcap=50.0,
+        final_logit_softcap=30.0,
+        query_pre_attn_scalar=256.0,
+    ),
+    # tiny gemma2 for CPU tests: every architectural wrinkle, small dims
+    "gemma2-tiny": ModelArchConfig(
+        vocab_size=2048,
+        hidden_size=256,
+        intermediate_size=512,
+        num_hidden_layers=2,
+        num_attention_heads=2,
+        num_key_value_heads=1,
+        head_dim=256,
+        max_position_embeddings=2048,
+        rope_theta=10000.0,
+        rms_norm_eps=1e-6,
+        tie_word_embeddings=True,
+        bos_token_id=2,
+        eos_token_id=1,
+        sliding_window=64,
+        sliding_window_pattern="even",
+        hidden_act="gelu_pytorch_tanh",
+        norm_plus_one=True,
+        scale_embeddings=True,
+        post_norms=True,
+        attn_logit_softcap=50.0,
+        final_logit_softcap=30.0,
+        query_pre_attn_scalar=256.0,
     ),
     # Mistral-7B v0.1/v0.2 shape: llama arch + 4096-token sliding window
     "mistral-7b": ModelArchConfig(

@@ -44,8 +44,12 @@ class Attention(nn.Module):
         self.n_q = cfg.num_attention_heads
         self.n_kv = cfg.num_key_value_heads
         self.hd = cfg.head_dim
-        self.window = cfg.sliding_window or 0
-        self.scale = 1.0 / math.sqrt(self.hd)
+        # per-layer window: gemma2 interleaves windowed ("even" layers)
+        # and full-causal attention; mistral windows every layer
+        use_win = cfg.sliding_window_pattern != "even" or layer_idx % 2 == 0
+        self.window = (cfg.sliding_window or 0) if use_win else 0
+        self.softcap = cfg.attn_logit_softcap or 0.0
+        self.scale = 1.0 / math.sqrt(cfg.query_pre_attn_scalar or self.hd)
         q_size = self.n_q * self.hd
         kv_size = self.n_kv * self.hd
         self.qkv_proj = EngineLinear(
@@ -99,6 +103,7 @@ class Attention(nn.Module):
                 self.scale,
                 out=out[:nd],
                 window=self.window,
+                softcap=self.softcap,
             )
         if fb.n_prefill > 0:
             ops.paged_attention_prefill(
@@ -111,6 +116,7 @@ class Attention(nn.Module):
                 self.scale,
                 out=out[nd:],
                 window=self.window,
+                softcap=self.softcap,
             )
         attn_flat = out.view(T, -1)
         if fp8_in:
@@ -132,6 +138,7 @@ class MLP(nn.Module):
             cfg.hidden_size, 2 * cfg.intermediate_size, bias=False
         )
         self.down_proj = EngineLinear(cfg.intermediate_size, cfg.hidden_size, bias=False)
+        self.gelu = cfg.hidden_act == "gelu_pytorch_tanh"
 
     def forward(self, x, fb: ForwardBatch | None = None) -> torch.Tensor:
         if isinstance(x, tuple):  # fused fp8 path (never active with LoRA)
@@ -146,7 +153,7 @@ class MLP(nn.Module):
         )
         if lora_live:
             lm.apply(self.layer_idx, "gate_up", x, h, fb.lora_ids)
-        a = ops.silu_and_mul(h)
+        a = ops.gelu_and_mul(h) if self.gelu else ops.silu_and_mul(h)
         y = self.down_proj(a)
         if lora_live:
             lm.apply(self.layer_idx, "down", a, y, fb.lora_ids)
@@ -205,8 +212,34 @@ class DecoderLayer(nn.Module):
             MoEMLP(cfg) if cfg.num_local_experts > 0 else MLP(cfg, layer_idx)
         )
         self.eps = cfg.rms_norm_eps
+        # gemma2: branch outputs are normed BEFORE the residual add, with
+        # separate pre/post feed-forward norms (4 norms per layer)
+        self.post_norms = cfg.post_norms
+        if cfg.post_norms:
+            self.pre_feedforward_layernorm = nn.Parameter(
+                torch.ones(cfg.hidden_size)
+            )
+            self.post_feedforward_layernorm = nn.Parameter(
+                torch.ones(cfg.hidden_size)
+            )
 
     def forward(self, x, residual, fb, kv_cache, cos_sin):
+        if self.post_norms:
+            # incoming invariant: hidden = x + residual (residual None at
+            # layer 0); outgoing (m_normed, hidden_after_attn) keeps it,
+            # so the model-level final fused_add_rmsnorm still applies
+            hidden = x if residual is None else x + residual
+            a = self.self_attn(
+                ops.rmsnorm(hidden, self.input_layernorm, self.eps),
+                fb, kv_cache, cos_sin,
+            )
+            a = ops.rmsnorm(a, self.post_attention_layernorm, self.eps)
+            h2 = hidden + a
+            mm = self.mlp(
+                ops.rmsnorm(h2, self.pre_feedforward_layernorm, self.eps), fb
+            )
+            mm = ops.rmsnorm(mm, self.post_feedforward_layernorm, self.eps)
+            return mm, h2
         if (
             getattr(self, "_fp8_fused", False)
             and x.is_cuda
@@ -289,6 +322,12 @@ class LlamaForCausalLM(nn.Module):
     @torch.inference_mode()
     def forward(self, fb: ForwardBatch) -> torch.Tensor:
         x = self.embed_tokens(fb.input_ids.long())
+        if self.cfg.scale_embeddings:
+            # gemma: embeddings scaled by sqrt(hidden), cast to model dtype
+            # first (HF normalizer semantics)
+            x = x * torch.tensor(
+                self.cfg.hidden_size ** 0.5, dtype=x.dtype
+            )
         if fb.mm_embeds is not None:
             # image-placeholder positions take the vision-tower embeddings
             x = x.index_copy(0, fb.mm_idx, fb.mm_embeds.to(x.dtype))
@@ -301,8 +340,13 @@ class LlamaForCausalLM(nn.Module):
     @torch.inference_mode()
     def compute_logits(self, hidden: torch.Tensor) -> torch.Tensor:
         if self.lm_head is None:
-            return F.linear(hidden, self.embed_tokens.weight).float()
-        return self.lm_head(hidden).float()
+            logits = F.linear(hidden, self.embed_tokens.weight).float()
+        else:
+            logits = self.lm_head(hidden).float()
+        cap = self.cfg.final_logit_softcap
+        if cap > 0:
+            logits = cap * torch.tanh(logits / cap)
+        return logits
 
     def bind_kv_caches(self, kv_caches: list[tuple[torch.Tensor, torch.Tensor]]):
         self.kv_caches = kv_caches

@@ -246,21 +246,29 @@ def load_weights(model, model_dir: str, vision=None) -> int:
         if name == "image_newline":  # llava-1.6 packing artifact, unused
             continue
         n = _strip(name)
+        # gemma RMSNorm convention: y = x * (1 + w). The engine's norm
+        # kernel multiplies by w, so gemma checkpoints load as w+1 (init
+        # zeros -> ones matches).
+        norm_w = (lambda t: t.float() + 1.0) if cfg.norm_plus_one else (lambda t: t)
         if n == "embed_tokens.weight":
             put("embed_tokens.weight", w)
         elif n in ("lm_head.weight",):
             if model.lm_head is not None:
                 put("lm_head.weight", w)
         elif n == "norm.weight":
-            put("norm", w)
+            put("norm", norm_w(w))
         elif ".layers." in n or n.startswith("layers."):
             parts = n.split(".")
             layer = int(parts[parts.index("layers") + 1])
             rest = ".".join(parts[parts.index("layers") + 2 :])
             if rest == "input_layernorm.weight":
-                put(f"layers.{layer}.input_layernorm", w)
+                put(f"layers.{layer}.input_layernorm", norm_w(w))
             elif rest == "post_attention_layernorm.weight":
-                put(f"layers.{layer}.post_attention_layernorm", w)
+                put(f"layers.{layer}.post_attention_layernorm", norm_w(w))
+            elif rest == "pre_feedforward_layernorm.weight":
+                put(f"layers.{layer}.pre_feedforward_layernorm", norm_w(w))
+            elif rest == "post_feedforward_layernorm.weight":
+                put(f"layers.{layer}.post_feedforward_layernorm", norm_w(w))
             elif rest == "self_attn.q_proj.weight":
                 pending.setdefault(f"qkv.{layer}", {})["q"] = w
                 fuse_qkv(layer)

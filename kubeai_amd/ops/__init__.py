@@ -73,7 +73,7 @@ def reshape_and_cache(k, v, k_cache, v_cache, slot_mapping) -> None:
 
 def paged_attention_decode(
     q, k_cache, v_cache, block_tables, seq_lens, scale: float, out=None,
-    window: int = 0,
+    window: int = 0, softcap: float = 0.0,
 ):
     """q may be a row-strided view (fused qkv output); out must be
     contiguous (allocated here if not supplied). window > 0 enables
@@ -84,11 +84,13 @@ def paged_attention_decode(
         if out is None:
             out = torch.empty(q.shape, dtype=q.dtype, device=q.device)
         _C.paged_attention_decode(
-            out, q, k_cache, v_cache, block_tables, seq_lens, scale, window
+            out, q, k_cache, v_cache, block_tables, seq_lens, scale, window,
+            softcap
         )
         return out
     res = ref.paged_attention_decode(
-        q, k_cache, v_cache, block_tables, seq_lens, scale, window=window
+        q, k_cache, v_cache, block_tables, seq_lens, scale, window=window,
+        softcap=softcap,
     )
     if out is not None:
         out.copy_(res)
@@ -98,7 +100,7 @@ def paged_attention_decode(
 
 def paged_attention_prefill(
     q, k_cache, v_cache, block_tables, query_start_loc, seq_lens, scale: float,
-    out=None, window: int = 0,
+    out=None, window: int = 0, softcap: float = 0.0,
 ):
     if q.is_cuda:
         _require_ext()
@@ -106,12 +108,12 @@ def paged_attention_prefill(
             out = torch.empty(q.shape, dtype=q.dtype, device=q.device)
         _C.paged_attention_prefill(
             out, q, k_cache, v_cache, block_tables, query_start_loc, seq_lens,
-            scale, window
+            scale, window, softcap
         )
         return out
     res = ref.paged_attention_prefill(
         q, k_cache, v_cache, block_tables, query_start_loc, seq_lens, scale,
-        window=window,
+        window=window, softcap=softcap,
     )
     if out is not None:
         out.copy_(res)
@@ -127,6 +129,17 @@ def silu_and_mul(x):
         _C.silu_and_mul(out, x)
         return out
     return ref.silu_and_mul(x)
+
+
+def gelu_and_mul(x):
+    """GeGLU (gemma family): gelu_tanh(gate) * up over packed [T, 2I]."""
+    if x.is_cuda:
+        _require_ext()
+        T, two_i = x.shape
+        out = torch.empty((T, two_i // 2), dtype=x.dtype, device=x.device)
+        _C.gelu_and_mul(out, x)
+        return out
+    return ref.gelu_and_mul(x)
 
 
 import os

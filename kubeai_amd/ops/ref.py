@@ -161,6 +161,7 @@ def paged_attention_decode(
     seq_lens: torch.Tensor,  # [B] int32 (total length incl. the new token)
     scale: float,
     window: int = 0,  # >0: sliding window, keys in (L-1-window, L-1]
+    softcap: float = 0.0,  # >0: gemma2 logit cap s = c*tanh(s/c)
 ) -> torch.Tensor:
     B, n_heads, head_dim = q.shape
     n_kv = k_cache.shape[1]
@@ -172,6 +173,8 @@ def paged_attention_decode(
         kf = k.float().repeat_interleave(g, dim=1)  # [L, n_heads, hd]
         vf = v.float().repeat_interleave(g, dim=1)
         s = torch.einsum("hd,lhd->hl", q[b].float(), kf) * scale
+        if softcap > 0:
+            s = softcap * torch.tanh(s / softcap)
         if window > 0 and L > window:
             s[:, : L - window] = float("-inf")
         p = torch.softmax(s, dim=-1)
@@ -188,6 +191,7 @@ def paged_attention_prefill(
     seq_lens: torch.Tensor,  # [B] total kv length incl. this chunk
     scale: float,
     window: int = 0,  # >0: query p attends keys in (p-window, p]
+    softcap: float = 0.0,  # >0: gemma2 logit cap s = c*tanh(s/c)
 ) -> torch.Tensor:
     """Causal attention of new (chunk) tokens against the full paged KV.
 
@@ -208,6 +212,8 @@ def paged_attention_prefill(
         vf = v.float().repeat_interleave(g, dim=1)
         qf = q[s0:s1].float()  # [q_len, n_heads, hd]
         s = torch.einsum("qhd,lhd->hql", qf, kf) * scale
+        if softcap > 0:
+            s = softcap * torch.tanh(s / softcap)
         # causal mask: query i (abs pos L - q_len + i) sees kv j <= abs pos
         qpos = torch.arange(L - q_len, L, device=q.device).unsqueeze(1)
         kpos = torch.arange(L, device=q.device).unsqueeze(0)
@@ -227,6 +233,13 @@ def silu_and_mul(x: torch.Tensor) -> torch.Tensor:
     """x = [T, 2*I] (gate | up) -> silu(gate) * up, fp32 internally."""
     gate, up = x.float().chunk(2, dim=-1)
     return (torch.nn.functional.silu(gate) * up).to(x.dtype)
+
+
+def gelu_and_mul(x: torch.Tensor) -> torch.Tensor:
+    """x = [T, 2*I] (gate | up) -> gelu_tanh(gate) * up (gemma GeGLU)."""
+    gate, up = x.float().chunk(2, dim=-1)
+    act = torch.nn.functional.gelu(gate, approximate="tanh")
+    return (act * up).to(x.dtype)
 
 
 # ---------------------------------------------------------------------------

@@ -398,3 +398,86 @@ def test_paged_prefill_sliding_window(q_lens, ctx_lens, window):
     out = ops.paged_attention_prefill(q, kc, vc, bt, qsl, seq_lens, scale,
                                       window=window)
     assert_close_bf16(out, expected, atol=3e-2, rtol=3e-2)
+
+
+@pytest.mark.parametrize("nq,nkv,lens,softcap,window",
+                         [(8, 4, [100, 37], 0.0, 0),
+                          (8, 4, [257], 50.0, 0),
+                          (2, 1, [2048], 50.0, 512),
+                          (8, 4, [16], 0.0, 0)])
+@pytest.mark.gpu
+def test_paged_decode_hd256(nq, nkv, lens, softcap, window):
+    """gemma2-shaped decode: head_dim 256, optional softcap + window."""
+    hd, bs = 256, 16
+    B = len(lens)
+    max_blocks = max((L + bs - 1) // bs for L in lens)
+    nb = B * max_blocks + 1
+    kc = torch.randn(nb, nkv, bs, hd, dtype=torch.bfloat16, device=DEV)
+    vc = torch.randn(nb, nkv, bs, hd, dtype=torch.bfloat16, device=DEV)
+    bt = torch.arange(1, nb, dtype=torch.int32, device=DEV).reshape(B, max_blocks)
+    seq_lens = torch.tensor(lens, dtype=torch.int32, device=DEV)
+    q = torch.randn(B, nq, hd, dtype=torch.bfloat16, device=DEV)
+    scale = 1.0 / math.sqrt(hd)
+    expected = ref.paged_attention_decode(
+        q.float(), kc.float(), vc.float(), bt, seq_lens, scale,
+        window=window, softcap=softcap,
+    )
+    out = ops.paged_attention_decode(q, kc, vc, bt, seq_lens, scale,
+                                     window=window, softcap=softcap)
+    assert_close_bf16(out, expected, atol=3e-2, rtol=3e-2)
+
+
+@pytest.mark.parametrize("q_lens,ctx_lens,softcap,window",
+                         [([100], [0], 0.0, 0),
+                          ([64], [30], 50.0, 0),
+                          ([40, 17], [0, 50], 50.0, 48)])
+@pytest.mark.gpu
+def test_paged_prefill_hd256(q_lens, ctx_lens, softcap, window):
+    """gemma2-shaped prefill: head_dim 256, optional softcap + window."""
+    nq, nkv, hd, bs = 8, 4, 256, 16
+    B = len(q_lens)
+    lens = [a + c for a, c in zip(q_lens, ctx_lens)]
+    max_blocks = max((L + bs - 1) // bs for L in lens)
+    nb = B * max_blocks + 1
+    kc = torch.randn(nb, nkv, bs, hd, dtype=torch.bfloat16, device=DEV)
+    vc = torch.randn(nb, nkv, bs, hd, dtype=torch.bfloat16, device=DEV)
+    bt = torch.arange(1, nb, dtype=torch.int32, device=DEV).reshape(B, max_blocks)
+    seq_lens = torch.tensor(lens, dtype=torch.int32, device=DEV)
+    qsl = torch.tensor([0] + list(torch.tensor(q_lens).cumsum(0)),
+                       dtype=torch.int32, device=DEV)
+    q = torch.randn(sum(q_lens), nq, hd, dtype=torch.bfloat16, device=DEV)
+    scale = 1.0 / math.sqrt(hd)
+    expected = ref.paged_attention_prefill(
+        q.float(), kc.float(), vc.float(), bt, qsl, seq_lens, scale,
+        window=window, softcap=softcap,
+    )
+    out = ops.paged_attention_prefill(q, kc, vc, bt, qsl, seq_lens, scale,
+                                      window=window, softcap=softcap)
+    assert_close_bf16(out, expected, atol=3e-2, rtol=3e-2)
+
+
+@pytest.mark.gpu
+def test_gelu_and_mul_gpu():
+    x = torch.randn(33, 512, dtype=torch.bfloat16, device=DEV)
+    out = ops.gelu_and_mul(x)
+    want = ref.gelu_and_mul(x.float())
+    assert_close_bf16(out, want, atol=2e-2, rtol=2e-2)
+
+
+@pytest.mark.gpu
+def test_gemma2_engine_e2e_gpu():
+    from kubeai_amd.engine import EngineConfig, LLMEngine, SamplingParams
+
+    eng = LLMEngine(EngineConfig(model="gemma2-tiny", device="cuda",
+                                 num_gpu_blocks=256, max_model_len=512))
+    prompt = [2] + list(range(100, 200))
+    eng.add_request(prompt, SamplingParams(max_tokens=6, ignore_eos=True),
+                    request_id="gg")
+    out = None
+    for _ in range(100):
+        if not eng.has_work():
+            break
+        for o in eng.step():
+            if o.finished:
+                out = o
+    assert out is not None and len(out.output_token_ids) == 6

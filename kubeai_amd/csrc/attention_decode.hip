@@ -43,7 +43,10 @@ constexpr int kPad = 8;
 // tile at HD=256).
 // SOFTCAP > 0 applies gemma2-style attention-logit soft capping,
 // s = cap * tanh(s / cap), after scaling and before masking.
-template <int G, typename CT = ushort, int HD = 128>
+// CAP: compile-time softcap enable — a runtime branch in the score
+// loop measured -13% on the capless hot path (194.6 vs 169.0 us at
+// B=64 L=2048), so capped models get their own instantiation.
+template <int G, typename CT = ushort, int HD = 128, bool CAP = false>
 __launch_bounds__(kBlockThreads) __global__ void paged_decode_kernel(
     ushort* __restrict__ out,            // [B, n_q, hd] bf16
     const ushort* __restrict__ q,        // [B, n_q, hd] bf16
@@ -217,8 +220,9 @@ __launch_bounds__(kBlockThreads) __global__ void paged_decode_kernel(
           s[g] += __shfl_xor(s[g], off, 64);
         s[g] = (tok < tile_len && tile_start + tok >= w0) ? s[g] * scale
                                                           : -INFINITY;
-        if (softcap > 0.f && s[g] != -INFINITY)
-          s[g] = softcap * tanhf(s[g] / softcap);
+        if constexpr (CAP) {
+          if (s[g] != -INFINITY) s[g] = softcap * tanhf(s[g] / softcap);
+        }
         // pass max across tokens (lane bits above the part bits)
         float tmax = s[g];
 #pragma unroll
@@ -434,7 +438,15 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
   auto stream = c10::hip::getCurrentHIPStream().stream();
   const bool fp8_cache = k_cache.scalar_type() == torch::kFloat8_e5m2;
 #define LAUNCH_CT_HD(GG, CT, HDV)                                         \
-  hipLaunchKernelGGL((paged_decode_kernel<GG, CT, HDV>), grid, block, 0,  \
+  do {                                                                    \
+    if (softcap > 0.0)                                                    \
+      LAUNCH_IMPL(GG, CT, HDV, true);                                     \
+    else                                                                  \
+      LAUNCH_IMPL(GG, CT, HDV, false);                                    \
+  } while (0)
+#define LAUNCH_IMPL(GG, CT, HDV, CAPV)                                    \
+  hipLaunchKernelGGL((paged_decode_kernel<GG, CT, HDV, CAPV>), grid,      \
+                     block, 0,                                            \
                      stream, (ushort*)out.data_ptr(),                     \
                      (const ushort*)q.data_ptr(),                         \
                      (const CT*)k_cache.data_ptr(),                       \
@@ -467,6 +479,7 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
   }
 #undef LAUNCH
 #undef LAUNCH_CT_HD
+#undef LAUNCH_IMPL
   HIP_CHECK_KERNEL();
   if (n_splits > 1) {
     const int64_t n_bh = (int64_t)B * n_q;

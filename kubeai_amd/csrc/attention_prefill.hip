@@ -47,7 +47,7 @@ DEV_INLINE int swz(int row, int byte_off) {
 // PV-column loops scale with HD/32 and HD/16).
 // SOFTCAP: gemma2 attention-logit capping s = cap*tanh(s/cap), applied
 // after scale, before masking (0 disables; passed at runtime).
-template <int G, typename CT = ushort, int HD = 128>
+template <int G, typename CT = ushort, int HD = 128, bool CAP = false>
 __launch_bounds__(G * WAVE_SIZE) __global__ void paged_prefill_kernel(
     ushort* __restrict__ out,            // [Tq, n_q, hd]
     const ushort* __restrict__ q,        // [Tq, n_q, hd]
@@ -214,7 +214,7 @@ __launch_bounds__(G * WAVE_SIZE) __global__ void paged_prefill_kernel(
         const int qpos = ctx + row_lo + r;
         float s0v = s_frag[0][i] * scale;
         float s1v = s_frag[1][i] * scale;
-        if (softcap > 0.f) {
+        if constexpr (CAP) {
           s0v = softcap * tanhf(s0v / softcap);
           s1v = softcap * tanhf(s1v / softcap);
         }
@@ -345,7 +345,14 @@ void paged_attention_prefill(torch::Tensor out, torch::Tensor q,
   auto stream = c10::hip::getCurrentHIPStream().stream();
   const bool fp8_cache = k_cache.scalar_type() == torch::kFloat8_e5m2;
 #define LAUNCH_CT_HD(GG, CT, HDV)                                          \
-  hipLaunchKernelGGL((paged_prefill_kernel<GG, CT, HDV>), grid,             \
+  do {                                                                      \
+    if (softcap > 0.0)                                                      \
+      LAUNCH_IMPL(GG, CT, HDV, true);                                       \
+    else                                                                    \
+      LAUNCH_IMPL(GG, CT, HDV, false);                                      \
+  } while (0)
+#define LAUNCH_IMPL(GG, CT, HDV, CAPV)                                      \
+  hipLaunchKernelGGL((paged_prefill_kernel<GG, CT, HDV, CAPV>), grid,       \
                      dim3(GG * WAVE_SIZE), 0, stream,                       \
                      (ushort*)out.data_ptr(), (const ushort*)q.data_ptr(),  \
                      (const CT*)k_cache.data_ptr(),                         \
@@ -375,5 +382,6 @@ void paged_attention_prefill(torch::Tensor out, torch::Tensor q,
   }
 #undef LAUNCH
 #undef LAUNCH_CT_HD
+#undef LAUNCH_IMPL
   HIP_CHECK_KERNEL();
 }

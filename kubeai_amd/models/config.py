@@ -65,6 +65,7 @@ class ModelArchConfig:
         n_heads = cfg["num_attention_heads"]
         arch = (cfg.get("architectures") or ["LlamaForCausalLM"])[0]
         gemma2 = arch == "Gemma2ForCausalLM"
+        gemma1 = arch == "GemmaForCausalLM"
         vision = None
         image_token_id = -1
         if arch == "LlavaForConditionalGeneration":
@@ -96,7 +97,9 @@ class ModelArchConfig:
             max_position_embeddings=cfg.get("max_position_embeddings", 8192),
             rms_norm_eps=cfg.get("rms_norm_eps", 1e-5),
             rope_theta=cfg.get("rope_theta", 10000.0),
-            tie_word_embeddings=cfg.get("tie_word_embeddings", gemma2),
+            tie_word_embeddings=cfg.get(
+                "tie_word_embeddings", gemma2 or gemma1
+            ),
             bos_token_id=cfg.get("bos_token_id", 1),
             eos_token_id=eos,
             num_local_experts=cfg.get("num_local_experts", 0),
@@ -106,11 +109,12 @@ class ModelArchConfig:
             sliding_window_pattern="even" if gemma2 else "all",
             hidden_act=(
                 "gelu_pytorch_tanh"
-                if gemma2 or "gelu" in str(cfg.get("hidden_act") or "")
+                if gemma2 or gemma1
+                or "gelu" in str(cfg.get("hidden_act") or "")
                 else "silu"
             ),
-            norm_plus_one=gemma2,
-            scale_embeddings=gemma2,
+            norm_plus_one=gemma2 or gemma1,
+            scale_embeddings=gemma2 or gemma1,
             post_norms=gemma2,
             attn_logit_softcap=float(
                 cfg.get("attn_logit_softcapping") or 0.0
@@ -191,6 +195,60 @@ PRESETS: dict[str, ModelArchConfig] = {
         max_position_embeddings=2048,
         bos_token_id=1,
         eos_token_id=2,
+    ),
+    # Gemma-2b (v1) shape: MQA (1 kv head, G=8), GeGLU, (1+w) norms,
+    # scaled embeddings, head_dim 256 — no post-norms/softcaps/window
+    "gemma-2b": ModelArchConfig(
+        vocab_size=256000,
+        hidden_size=2048,
+        intermediate_size=16384,
+        num_hidden_layers=18,
+        num_attention_heads=8,
+        num_key_value_heads=1,
+        head_dim=256,
+        max_position_embeddings=8192,
+        rope_theta=10000.0,
+        rms_norm_eps=1e-6,
+        tie_word_embeddings=True,
+        bos_token_id=2,
+        eos_token_id=1,
+        hidden_act="gelu_pytorch_tanh",
+        norm_plus_one=True,
+        scale_embeddings=True,
+    ),
+    "gemma-tiny": ModelArchConfig(
+        vocab_size=2048,
+        hidden_size=256,
+        intermediate_size=512,
+        num_hidden_layers=2,
+        num_attention_heads=2,
+        num_key_value_heads=1,
+        head_dim=256,
+        max_position_embeddings=2048,
+        rope_theta=10000.0,
+        rms_norm_eps=1e-6,
+        tie_word_embeddings=True,
+        bos_token_id=2,
+        eos_token_id=1,
+        hidden_act="gelu_pytorch_tanh",
+        norm_plus_one=True,
+        scale_embeddings=True,
+    ),
+    # Phi-4 shape (Phi3ForCausalLM): llama structure with fused
+    # qkv_proj / gate_up_proj checkpoint tensors, untied head
+    "phi-4": ModelArchConfig(
+        vocab_size=100352,
+        hidden_size=5120,
+        intermediate_size=17920,
+        num_hidden_layers=40,
+        num_attention_heads=40,
+        num_key_value_heads=10,
+        head_dim=128,
+        max_position_embeddings=16384,
+        rope_theta=250000.0,
+        rms_norm_eps=1e-5,
+        bos_token_id=100257,
+        eos_token_id=100257,
     ),
     # Gemma2-2b shape: GeGLU, (1+w) norms, post-norms, logit softcaps,
     # head_dim 256, alternate-layer sliding window

@@ -153,6 +153,24 @@ def load_weights_tp(model, model_dir: str, vision=None) -> int:
                 put(f"layers.{layer}.input_layernorm", w)
             elif rest == "post_attention_layernorm.weight":
                 put(f"layers.{layer}.post_attention_layernorm", w)
+            elif rest == "self_attn.qkv_proj.weight":
+                # phi-3/phi-4: pre-fused [q;k;v] rows; split and feed the
+                # same per-tensor sharding paths
+                qr, kr, vr = torch.split(
+                    w,
+                    [cfg.num_attention_heads * hd,
+                     cfg.num_key_value_heads * hd,
+                     cfg.num_key_value_heads * hd],
+                    dim=0,
+                )
+                ps = pending.setdefault(f"qkv.{layer}", {})
+                ps["q"], ps["k"], ps["v"] = qr, kr, vr
+                fuse_qkv(layer)
+            elif rest == "mlp.gate_up_proj.weight":
+                gr, ur = w.chunk(2, dim=0)
+                ps = pending.setdefault(f"gu.{layer}", {})
+                ps["gate"], ps["up"] = gr, ur
+                fuse_gate_up(layer)
             elif rest == "self_attn.q_proj.weight":
                 pending.setdefault(f"qkv.{layer}", {})["q"] = w
                 fuse_qkv(layer)
@@ -269,6 +287,21 @@ def load_weights(model, model_dir: str, vision=None) -> int:
                 put(f"layers.{layer}.pre_feedforward_layernorm", norm_w(w))
             elif rest == "post_feedforward_layernorm.weight":
                 put(f"layers.{layer}.post_feedforward_layernorm", norm_w(w))
+            elif rest == "self_attn.qkv_proj.weight":
+                # phi-3/phi-4 checkpoints pre-fuse qkv ([q;k;v] rows —
+                # same layout as the engine's fused projection)
+                qr, kr, vr = torch.split(
+                    w, [nq * hd, nkv * hd, nkv * hd], dim=0
+                )
+                ps = pending.setdefault(f"qkv.{layer}", {})
+                ps["q"], ps["k"], ps["v"] = qr, kr, vr
+                fuse_qkv(layer)
+            elif rest == "mlp.gate_up_proj.weight":
+                # phi family pre-fuses gate|up as well
+                gr, ur = w.chunk(2, dim=0)
+                ps = pending.setdefault(f"gu.{layer}", {})
+                ps["gate"], ps["up"] = gr, ur
+                fuse_gate_up(layer, None)
             elif rest == "self_attn.q_proj.weight":
                 pending.setdefault(f"qkv.{layer}", {})["q"] = w
                 fuse_qkv(layer)
@@ -332,17 +365,30 @@ def save_hf_checkpoint(model, out_dir: str) -> None:
     cfg = model.cfg
     state: dict[str, torch.Tensor] = {
         "model.embed_tokens.weight": model.embed_tokens.weight.detach().cpu(),
-        "model.norm.weight": model.norm.detach().cpu(),
     }
     if model.lm_head is not None:
         state["lm_head.weight"] = model.lm_head.weight.detach().cpu()
     nq, nkv, hd = cfg.num_attention_heads, cfg.num_key_value_heads, cfg.head_dim
+    # gemma convention stores norms as w (engine holds 1+w); reverse it
+    norm_out = (
+        (lambda t: t.detach().cpu().float() - 1.0)
+        if cfg.norm_plus_one
+        else (lambda t: t.detach().cpu())
+    )
+    state["model.norm.weight"] = norm_out(model.norm)
     for i, layer in enumerate(model.layers):
         pre = f"model.layers.{i}."
-        state[pre + "input_layernorm.weight"] = layer.input_layernorm.detach().cpu()
-        state[pre + "post_attention_layernorm.weight"] = (
-            layer.post_attention_layernorm.detach().cpu()
+        state[pre + "input_layernorm.weight"] = norm_out(layer.input_layernorm)
+        state[pre + "post_attention_layernorm.weight"] = norm_out(
+            layer.post_attention_layernorm
         )
+        if getattr(layer, "post_norms", False):
+            state[pre + "pre_feedforward_layernorm.weight"] = norm_out(
+                layer.pre_feedforward_layernorm
+            )
+            state[pre + "post_feedforward_layernorm.weight"] = norm_out(
+                layer.post_feedforward_layernorm
+            )
         qkv = layer.self_attn.qkv_proj.weight.detach().cpu()
         state[pre + "self_attn.q_proj.weight"] = qkv[: nq * hd].clone()
         state[pre + "self_attn.k_proj.weight"] = qkv[nq * hd : (nq + nkv) * hd].clone()
@@ -377,7 +423,10 @@ def save_hf_checkpoint(model, out_dir: str) -> None:
         json.dump(
             {
                 "architectures": [
-                    "MixtralForCausalLM" if cfg.num_local_experts else "LlamaForCausalLM"
+                    "MixtralForCausalLM" if cfg.num_local_experts
+                    else "Gemma2ForCausalLM" if cfg.post_norms
+                    else "GemmaForCausalLM" if cfg.norm_plus_one
+                    else "LlamaForCausalLM"
                 ],
                 "vocab_size": cfg.vocab_size,
                 "hidden_size": cfg.hidden_size,
@@ -396,6 +445,11 @@ def save_hf_checkpoint(model, out_dir: str) -> None:
                 "num_experts_per_tok": cfg.num_experts_per_tok,
                 "attention_bias": cfg.attention_bias,
                 "rope_scaling": cfg.rope_scaling,
+                "sliding_window": cfg.sliding_window or None,
+                "attn_logit_softcapping": cfg.attn_logit_softcap or None,
+                "final_logit_softcapping": cfg.final_logit_softcap or None,
+                "query_pre_attn_scalar": cfg.query_pre_attn_scalar or None,
+                "hidden_act": cfg.hidden_act,
                 "torch_dtype": "bfloat16",
             },
             f,

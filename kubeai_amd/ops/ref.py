@@ -89,7 +89,12 @@ def make_cos_sin_cache(
             inv_freq = torch.where(mid, scaled, inv_freq)
         elif rtype == "linear":
             scaling = scaling * factor
-        # other types (dynamic/yarn) fall through unscaled
+        elif rtype in (None, "default"):
+            pass
+        else:
+            # longrope/yarn/dynamic would silently produce wrong
+            # positions — refuse loudly instead
+            raise ValueError(f"unsupported rope_scaling type {rtype!r}")
     t = torch.arange(max_positions, dtype=torch.float64) / scaling
     freqs = torch.outer(t, inv_freq)
     return torch.cat([freqs.cos(), freqs.sin()], dim=-1).to(dtype)

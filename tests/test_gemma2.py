@@ -159,3 +159,109 @@ def test_gemma2_matches_hf_reference_math():
     ]
     h2 = model2.forward(fb)
     assert not torch.allclose(hidden, h2)
+
+
+def test_gemma1_engine_e2e_cpu():
+    eng = LLMEngine(EngineConfig(model="gemma-tiny", device="cpu",
+                                 num_gpu_blocks=128, enable_graphs=False,
+                                 max_model_len=512))
+    assert eng.arch.num_key_value_heads == 1  # MQA, G = 2 here
+    assert not eng.arch.post_norms and eng.arch.norm_plus_one
+    r = eng.add_request([2] + list(range(100, 140)),
+                        SamplingParams(max_tokens=5, ignore_eos=True),
+                        request_id="g1v1")
+    done = _drain(eng)
+    assert "g1v1" in done and len(done["g1v1"].output_token_ids) == 5
+
+
+def test_gemma2_checkpoint_roundtrip(tmp_path):
+    """(1+w) norm conversion + gemma2 config survive a save/load cycle."""
+    from kubeai_amd.models.loader import save_hf_checkpoint
+    from kubeai_amd.engine import EngineConfig, LLMEngine, SamplingParams
+
+    src = LLMEngine(EngineConfig(model="gemma2-tiny", device="cpu",
+                                 num_gpu_blocks=128, max_model_len=512,
+                                 enable_graphs=False, seed=3))
+    ckpt = str(tmp_path / "g2")
+    save_hf_checkpoint(src.runner.model, ckpt)
+    arch = ModelArchConfig.from_hf_config(ckpt)
+    assert arch.post_norms and arch.attn_logit_softcap == 50.0
+    dst = LLMEngine(EngineConfig(model=ckpt, device="cpu",
+                                 num_gpu_blocks=128, max_model_len=512,
+                                 enable_graphs=False, seed=999))
+    prompt = [2] + list(range(100, 150))
+
+    def gen(eng, rid):
+        eng.add_request(prompt, SamplingParams(max_tokens=6, ignore_eos=True),
+                        request_id=rid)
+        return _drain(eng)[rid].output_token_ids
+
+    assert gen(src, "a") == gen(dst, "b")
+
+
+def test_phi_fused_checkpoint_loads(tmp_path):
+    """phi-3/4-style checkpoints (pre-fused qkv_proj/gate_up_proj) load
+    into the engine and reproduce the split-key checkpoint's output."""
+    import json
+
+    from safetensors.torch import load_file, save_file
+
+    from kubeai_amd.models.loader import save_hf_checkpoint
+
+    src = LLMEngine(EngineConfig(model="llama-tiny", device="cpu",
+                                 num_gpu_blocks=128, max_model_len=512,
+                                 enable_graphs=False, seed=3))
+    split_dir = str(tmp_path / "split")
+    save_hf_checkpoint(src.runner.model, split_dir)
+    # refuse: convert split keys to phi-4 fused form
+    t = load_file(split_dir + "/model.safetensors")
+    fused = {}
+    done_layers = set()
+    for k, v in t.items():
+        if ".self_attn.q_proj.weight" in k or ".mlp.gate_proj.weight" in k:
+            pre = k.rsplit(".", 3)[0]  # model.layers.N
+            if (pre, "qkv") not in done_layers and "q_proj" in k:
+                fused[pre + ".self_attn.qkv_proj.weight"] = torch.cat([
+                    t[pre + ".self_attn.q_proj.weight"],
+                    t[pre + ".self_attn.k_proj.weight"],
+                    t[pre + ".self_attn.v_proj.weight"],
+                ])
+                done_layers.add((pre, "qkv"))
+            if (pre, "gu") not in done_layers and "gate_proj" in k:
+                fused[pre + ".mlp.gate_up_proj.weight"] = torch.cat([
+                    t[pre + ".mlp.gate_proj.weight"],
+                    t[pre + ".mlp.up_proj.weight"],
+                ])
+                done_layers.add((pre, "gu"))
+        elif any(s in k for s in (".k_proj.", ".v_proj.", ".up_proj.")):
+            continue
+        else:
+            fused[k] = v
+    phi_dir = str(tmp_path / "phi")
+    import os
+    os.makedirs(phi_dir)
+    save_file(fused, phi_dir + "/model.safetensors")
+    cfgd = json.load(open(split_dir + "/config.json"))
+    cfgd["architectures"] = ["Phi3ForCausalLM"]
+    json.dump(cfgd, open(phi_dir + "/config.json", "w"))
+
+    dst = LLMEngine(EngineConfig(model=phi_dir, device="cpu",
+                                 num_gpu_blocks=128, max_model_len=512,
+                                 enable_graphs=False, seed=999))
+    prompt = list(range(10, 80))
+
+    def gen(eng, rid):
+        eng.add_request(prompt, SamplingParams(max_tokens=6, ignore_eos=True),
+                        request_id=rid)
+        return _drain(eng)[rid].output_token_ids
+
+    assert gen(src, "a") == gen(dst, "b")
+
+
+def test_phi4_preset_and_unsupported_rope_loud():
+    assert PRESETS["phi-4"].num_key_value_heads == 10
+    from kubeai_amd.ops.ref import make_cos_sin_cache
+
+    with pytest.raises(ValueError):
+        make_cos_sin_cache(128, 64, 10000.0,
+                           rope_scaling={"rope_type": "longrope"})

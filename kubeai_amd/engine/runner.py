@@ -573,7 +573,7 @@ class ModelRunner:
         tok = self.tokenizer
         if tok is None:
             return tokens  # no tokenizer injected (bare-engine use)
-        from .jsonmode import JsonPrefixValidator
+        from .jsonmode import JsonPrefixValidator, SchemaValidator
 
         eos = self.arch.eos_token_id
         for i, r in enumerate(sample_reqs):
@@ -581,7 +581,11 @@ class ModelRunner:
                 continue
             v = self._json_validators.get(r.request_id)
             if v is None:
-                v = self._json_validators[r.request_id] = JsonPrefixValidator()
+                v = self._json_validators[r.request_id] = (
+                    SchemaValidator(r.params.json_schema)
+                    if r.params.json_schema
+                    else JsonPrefixValidator()
+                )
             if v.complete:
                 tokens[i] = eos  # the object closed on a previous step
                 continue
@@ -620,8 +624,11 @@ class ModelRunner:
                         break
             if chosen is not None:
                 tokens[i] = int(chosen)
-            # else: nothing representable continues this JSON — keep the
-            # sampled token (request will end as ordinary text)
+            else:
+                # nothing representable continues this JSON (schema dead
+                # end) — end the request rather than emit garbage; the
+                # client sees the longest valid prefix
+                tokens[i] = eos
         return tokens
 
     def _json_char_candidates(self, v, snap, tok):

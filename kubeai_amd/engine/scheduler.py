@@ -40,6 +40,9 @@ class SamplingParams:
     # OpenAI response_format json_object: the sampler enforces that the
     # output is a valid JSON object (engine/jsonmode.py)
     json_mode: bool = False
+    # OpenAI response_format json_schema: schema-guided decoding (types,
+    # required keys, enums, closed objects — jsonmode.SchemaValidator)
+    json_schema: Optional[dict] = None
 
 
 class RequestStatus(enum.Enum):

@@ -479,17 +479,25 @@ def build_app(server: EngineServer) -> FastAPI:
             out[tid] = bias
         return out
 
-    def _json_mode_from(body: dict) -> bool:
+    def _json_mode_from(body: dict):
+        """-> (json_mode, schema|None)."""
         rf = body.get("response_format")
         if not rf:
-            return False
+            return False, None
         rtype = rf.get("type") if isinstance(rf, dict) else rf
-        if rtype in ("json_object", "json_schema"):
-            # json_schema enforces well-formed JSON (schema-shape guidance
-            # comes from the prompt — docs/parity.md)
-            return True
+        if rtype == "json_object":
+            return True, None
+        if rtype == "json_schema":
+            # schema-GUIDED decoding: types, required keys, enums,
+            # closed objects (jsonmode.SchemaValidator)
+            js = rf.get("json_schema") if isinstance(rf, dict) else None
+            schema = js.get("schema") if isinstance(js, dict) else None
+            if schema is not None and not isinstance(schema, dict):
+                raise RequestError("response_format json_schema.schema "
+                                   "must be an object")
+            return True, schema
         if rtype == "text":
-            return False
+            return False, None
         raise RequestError(f"unsupported response_format type {rtype!r}")
 
     def _params_from(body: dict) -> SamplingParams:
@@ -497,8 +505,10 @@ def build_app(server: EngineServer) -> FastAPI:
         temp = body.get("temperature")
         temp = 1.0 if temp is None else float(temp)
         seed = body.get("seed")
+        jm, schema = _json_mode_from(body)
         return SamplingParams(
-            json_mode=_json_mode_from(body),
+            json_mode=jm,
+            json_schema=schema,
             max_tokens=int(mt),
             temperature=float(temp),
             top_p=float(body.get("top_p") or 1.0),
@@ -693,6 +703,27 @@ def build_app(server: EngineServer) -> FastAPI:
             messages = _with_tool_instructions(messages, tools, tool_choice)
             if tool_choice == "required" or isinstance(tool_choice, dict):
                 params.json_mode = True
+                # forced single function: guide decoding with the tool
+                # call shape {name: <fn>, arguments: <parameters schema>}
+                if isinstance(tool_choice, dict):
+                    fn = (tool_choice.get("function") or {}).get("name")
+                    tool = next(
+                        (t for t in tools
+                         if (t.get("function", t) or {}).get("name") == fn),
+                        None,
+                    )
+                    if tool is not None:
+                        fdef = tool.get("function", tool) or {}
+                        params.json_schema = {
+                            "type": "object",
+                            "properties": {
+                                "name": {"const": fn},
+                                "arguments": fdef.get("parameters")
+                                or {"type": "object"},
+                            },
+                            "required": ["name", "arguments"],
+                            "additionalProperties": False,
+                        }
         toks = apply_chat_template(server.tokenizer, messages)
         images = _images_from_messages(server, messages)
         lora_id = _resolve_lora(server, body.get("model"))

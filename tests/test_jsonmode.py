@@ -125,3 +125,100 @@ def test_bad_response_format_400(client):
               "max_tokens": 4, "response_format": {"type": "yaml"}},
     )
     assert r.status_code == 400
+
+
+# ----------------------------------------------------- schema-guided
+def test_schema_validator_enforces_subset():
+    from kubeai_amd.engine.jsonmode import SchemaValidator
+
+    sch = {
+        "type": "object",
+        "properties": {
+            "name": {"type": "string", "enum": ["alice", "bob"]},
+            "age": {"type": "integer"},
+            "tags": {"type": "array", "items": {"type": "string"}},
+        },
+        "required": ["name", "age"],
+        "additionalProperties": False,
+    }
+    ok = SchemaValidator(sch)
+    assert ok.feed('{"name": "alice", "age": 3, "tags": ["x", "y"]}')
+    assert ok.complete
+    assert not SchemaValidator(sch).feed('{"name": "carol')  # enum prefix
+    assert not SchemaValidator(sch).feed('{"age": 3.5')      # integer
+    assert not SchemaValidator(sch).feed('{"nope":')         # closed object
+    held = SchemaValidator(sch)
+    assert held.feed('{"name": "bob"')
+    assert not held.feed('}')                                # required held
+    assert not SchemaValidator(sch).feed('{"tags": {')       # array type
+
+
+def test_schema_validator_refs_unions_consts():
+    from kubeai_amd.engine.jsonmode import SchemaValidator
+
+    sch = {
+        "$defs": {"pt": {
+            "type": "object",
+            "properties": {"x": {"type": "number"}, "kind": {"const": "pt"}},
+            "required": ["x", "kind"],
+        }},
+        "type": "object",
+        "properties": {"p": {"$ref": "#/$defs/pt"},
+                       "v": {"type": ["null", "string"]}},
+        "required": ["p"],
+    }
+    v = SchemaValidator(sch)
+    assert v.feed('{"p": {"kind": "pt", "x": -1.5e3}, "v": null}')
+    assert v.complete
+    assert not SchemaValidator(sch).feed('{"p": {"kind": "xx')
+    assert not SchemaValidator(sch).feed('{"v": 4')
+
+
+def test_schema_snapshot_restore_roundtrip():
+    from kubeai_amd.engine.jsonmode import SchemaValidator
+
+    sch = {"type": "object",
+           "properties": {"name": {"enum": ["alice", "bob"]},
+                          "age": {"type": "integer"}},
+           "required": ["name", "age"]}
+    v = SchemaValidator(sch)
+    assert v.feed('{"name": "a')
+    snap = v.snapshot()
+    assert v.feed('lice", "age": 2}') and v.complete
+    v.restore(snap)
+    assert v.feed('lice", "age": 7}') and v.complete
+
+
+def test_schema_guided_generation_http(client):
+    """End to end: json_schema constrains output to parse AND conform
+    (required key present, closed object, enum value)."""
+    c, server = client
+    tok = server.tokenizer
+    bias = {
+        str(tok.char_token("{")): 4.0,
+        str(tok.char_token("}")): 6.0,
+        str(tok.char_token('"')): 2.0,
+    }
+    schema = {
+        "type": "object",
+        "properties": {"k": {"enum": ["aa", "bb"]}},
+        "required": ["k"],
+        "additionalProperties": False,
+    }
+    r = c.post(
+        "/v1/chat/completions",
+        json={
+            "messages": [{"role": "user", "content": "emit json"}],
+            "max_tokens": 64,
+            "temperature": 0,
+            "logit_bias": bias,
+            "response_format": {
+                "type": "json_schema",
+                "json_schema": {"name": "out", "schema": schema},
+            },
+        },
+    )
+    assert r.status_code == 200, r.text
+    text = r.json()["choices"][0]["message"]["content"]
+    obj = json.loads(text)
+    assert set(obj) == {"k"} and obj["k"] in ("aa", "bb")

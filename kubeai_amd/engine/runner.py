@@ -145,8 +145,13 @@ class ModelRunner:
         if arch.vision is not None:
             from kubeai_amd.models.vision import VisionTower
 
+            # independent seed: the TP model's construction leaves the
+            # global RNG in a world-size-dependent state; the tower must
+            # init identically on every rank AND across TP degrees
+            torch.manual_seed(seed + 7919)
             self.vision = VisionTower(arch, device=self.device, dtype=dtype)
             self.vision.eval()
+            torch.manual_seed(seed)
         if model_path is not None and os.path.isdir(model_path):
             if tp_group is not None:
                 from kubeai_amd.models.loader import load_weights_tp

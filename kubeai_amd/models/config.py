@@ -418,6 +418,28 @@ PRESETS: dict[str, ModelArchConfig] = {
         bos_token_id=1,
         eos_token_id=2,
     ),
+    # TP-shardable multimodal tiny (2 kv heads so TP=2 splits evenly)
+    "llava-tiny-tp": ModelArchConfig(
+        vocab_size=2048,
+        hidden_size=256,
+        intermediate_size=512,
+        num_hidden_layers=2,
+        num_attention_heads=4,
+        num_key_value_heads=2,
+        head_dim=128,
+        max_position_embeddings=2048,
+        bos_token_id=1,
+        eos_token_id=2,
+        image_token_id=99,
+        vision={
+            "image_size": 64,
+            "patch_size": 16,
+            "hidden_size": 128,
+            "num_hidden_layers": 2,
+            "num_attention_heads": 2,
+            "intermediate_size": 256,
+        },
+    ),
     "mixtral-tiny": ModelArchConfig(
         architecture="mixtral",
         vocab_size=2048,

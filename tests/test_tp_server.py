@@ -74,3 +74,64 @@ def test_tp2_server_matches_tp1():
             except subprocess.TimeoutExpired:
                 proc.kill()
     assert texts[1] == texts[2], texts
+
+
+def test_tp2_multimodal_matches_tp1():
+    """Images travel rank0 -> workers via the broadcast intent; each rank
+    runs the replicated vision tower deterministically, so TP=2 greedy
+    output must equal TP=1."""
+    import base64
+
+    import torch
+
+    from kubeai_amd.utils import imaging
+
+    g = torch.Generator().manual_seed(5)
+    img = torch.randint(0, 256, (24, 36, 3), generator=g, dtype=torch.uint8)
+    url = "data:image/png;base64," + base64.b64encode(
+        imaging.encode_png(img)
+    ).decode()
+    body = {
+        "model": "tiny",
+        "messages": [{
+            "role": "user",
+            "content": [
+                {"type": "text", "text": "describe"},
+                {"type": "image_url", "image_url": {"url": url}},
+            ],
+        }],
+        "max_tokens": 5,
+        "temperature": 0,
+    }
+    outs = {}
+    for tp in (1, 2):
+        proc, port = start_server_model(tp, "llava-tiny-tp")
+        try:
+            wait_ready(port)
+            r = httpx.post(
+                f"http://127.0.0.1:{port}/v1/chat/completions", json=body,
+                timeout=60,
+            )
+            assert r.status_code == 200, r.text
+            outs[tp] = r.json()["choices"][0]["message"]["content"]
+        finally:
+            proc.terminate()
+            proc.wait(timeout=30)
+    assert outs[1] == outs[2]
+
+
+def start_server_model(tp: int, model: str):
+    import os
+
+    port = free_port()
+    env = dict(os.environ)
+    if tp == 1:
+        env["KUBEAI_FORCE_TP"] = "1"
+    cmd = [
+        sys.executable, "-m", "kubeai_amd.engine.server",
+        "--model", model, "--served-model-name", "tiny",
+        "--host", "127.0.0.1", "--port", str(port),
+        "--device", "cpu", "--num-gpu-blocks", "64", "--max-model-len", "512",
+        "--tensor-parallel-size", str(tp),
+    ]
+    return subprocess.Popen(cmd, env=env), port

@@ -24,7 +24,36 @@ _FP8_DTYPES = tuple(
     for n in ("float8_e4m3fn", "float8_e5m2", "float8_e4m3fnuz", "float8_e5m2fnuz")
     if hasattr(torch, n)
 )
-_UNSUPPORTED_QUANT_KEYS = ("qweight", "qzeros", "g_idx", "scales_zp")
+_UNSUPPORTED_QUANT_KEYS = ("g_idx", "scales_zp")  # GPTQ act-order etc.
+
+# AWQ "gemm" nibble interleave (AutoAWQ unpack_awq + reverse_awq_order):
+# nibble i of each int32 holds logical column REVERSE_AWQ_ORDER.index(i);
+# applying this permutation to the shift-ordered unpack restores logical
+# column order. Self-consistency is covered by tests (synthetic packing);
+# no real AWQ checkpoint exists in this offline environment to cross-check.
+_REVERSE_AWQ_ORDER = (0, 4, 1, 5, 2, 6, 3, 7)
+
+
+def _awq_unpack(q: torch.Tensor) -> torch.Tensor:
+    """int32 [r, c/8] -> int [r, c] nibbles in logical column order."""
+    shifts = torch.arange(0, 32, 4, dtype=torch.int32)
+    nibbles = (q.unsqueeze(-1) >> shifts) & 0xF  # [r, c/8, 8] shift order
+    logical = nibbles[..., list(_REVERSE_AWQ_ORDER)]
+    return logical.reshape(q.shape[0], -1)
+
+
+def _awq_dequant(qweight, qzeros, scales) -> torch.Tensor:
+    """AWQ gemm-format group -> bf16 [out, in] dense weight.
+
+    qweight int32 [in, out/8]; qzeros int32 [in/g, out/8];
+    scales fp16 [in/g, out]. w = (nibble - zero) * scale.
+    """
+    iw = _awq_unpack(qweight).float()           # [in, out]
+    iz = _awq_unpack(qzeros).float()            # [in/g, out]
+    g = qweight.shape[0] // qzeros.shape[0]
+    z = iz.repeat_interleave(g, dim=0)
+    s = scales.float().repeat_interleave(g, dim=0)
+    return ((iw - z) * s).t().contiguous().to(torch.bfloat16)
 
 
 def iter_safetensors(model_dir: str) -> Iterator[tuple[str, torch.Tensor]]:
@@ -38,18 +67,53 @@ def iter_safetensors(model_dir: str) -> Iterator[tuple[str, torch.Tensor]]:
     files = sorted(glob.glob(os.path.join(model_dir, "*.safetensors")))
     if not files:
         raise FileNotFoundError(f"no *.safetensors under {model_dir}")
+    # AWQ (4-bit, "gemm" packing) dequantizes on the fly to bf16 —
+    # reference catalog entry: hugging-quants/...-70B-Instruct-AWQ-INT4
+    awq = False
+    cfg_path = os.path.join(model_dir, "config.json")
+    if os.path.exists(cfg_path):
+        try:
+            with open(cfg_path) as cf:
+                qc = (json.load(cf).get("quantization_config") or {})
+            awq = (
+                qc.get("quant_method") == "awq"
+                and int(qc.get("bits", 4)) == 4
+                and qc.get("version", "gemm").lower() == "gemm"
+            )
+        except Exception:  # noqa: BLE001
+            awq = False
     for f in files:
         with safe_open(f, framework="pt", device="cpu") as sf:
             keys = set(sf.keys())
             for key in sf.keys():
                 if key.endswith(_SCALE_SUFFIXES):
                     continue  # consumed with its weight below
+                if key.endswith((".qzeros", ".scales")) and awq:
+                    continue  # consumed with their .qweight below
+                if key.endswith(".qweight"):
+                    pre = key[: -len(".qweight")]
+                    if (
+                        not awq
+                        or pre + ".qzeros" not in keys
+                        or pre + ".scales" not in keys
+                    ):
+                        raise ValueError(
+                            f"{model_dir}: packed-quantized tensor {key!r} "
+                            "is not a recognized AWQ-gemm group — provide "
+                            "bf16/fp16, fp8(+weight_scale), or AWQ-INT4 "
+                            "gemm checkpoints"
+                        )
+                    yield pre + ".weight", _awq_dequant(
+                        sf.get_tensor(key),
+                        sf.get_tensor(pre + ".qzeros"),
+                        sf.get_tensor(pre + ".scales"),
+                    )
+                    continue
                 if any(key.endswith("." + s) for s in _UNSUPPORTED_QUANT_KEYS):
                     raise ValueError(
                         f"{model_dir}: packed-quantized checkpoint tensor "
-                        f"{key!r} is not supported — provide a bf16/fp16 or "
-                        "fp8(+weight_scale) checkpoint (serve with "
-                        "quantization=fp8 for W8A8)"
+                        f"{key!r} is not supported — provide a bf16/fp16, "
+                        "fp8(+weight_scale), or AWQ-INT4 gemm checkpoint"
                     )
                 w = sf.get_tensor(key)
                 if w.dtype in _FP8_DTYPES:

@@ -481,3 +481,30 @@ def test_gemma2_engine_e2e_gpu():
             if o.finished:
                 out = o
     assert out is not None and len(out.output_token_ids) == 6
+
+
+@pytest.mark.gpu
+def test_paged_attention_hd256_fp8_cache():
+    """gemma2 shapes with the fp8 e5m2 KV cache (decode + prefill)."""
+    nq, nkv, hd, bs = 8, 4, 256, 16
+    L = 200
+    nb = (L + bs - 1) // bs + 1
+    kc = torch.randn(nb, nkv, bs, hd, dtype=torch.bfloat16, device=DEV)
+    vc = torch.randn(nb, nkv, bs, hd, dtype=torch.bfloat16, device=DEV)
+    kc8 = kc.to(torch.float8_e5m2)
+    vc8 = vc.to(torch.float8_e5m2)
+    bt = torch.arange(1, nb, dtype=torch.int32, device=DEV).reshape(1, -1)
+    sl = torch.tensor([L], dtype=torch.int32, device=DEV)
+    scale = 1.0 / math.sqrt(hd)
+    qd = torch.randn(1, nq, hd, dtype=torch.bfloat16, device=DEV)
+    want = ref.paged_attention_decode(
+        qd.float(), kc8.float(), vc8.float(), bt, sl, scale)
+    out = ops.paged_attention_decode(qd, kc8, vc8, bt, sl, scale)
+    assert_close_bf16(out, want, atol=3e-2, rtol=3e-2)
+    Tq = 64
+    qp = torch.randn(Tq, nq, hd, dtype=torch.bfloat16, device=DEV)
+    qsl = torch.tensor([0, Tq], dtype=torch.int32, device=DEV)
+    wantp = ref.paged_attention_prefill(
+        qp.float(), kc8.float(), vc8.float(), bt, qsl, sl, scale)
+    outp = ops.paged_attention_prefill(qp, kc8, vc8, bt, qsl, sl, scale)
+    assert_close_bf16(outp, wantp, atol=3e-2, rtol=3e-2)

@@ -190,7 +190,8 @@ def main():
 
     model = args.model
     if not use_cuda and model == "llama-3-8b":
-        model = "llama-tiny"  # CPU sanity runs only
+        # CPU sanity runs only; TP needs the tp-divisible tiny (2 kv heads)
+        model = "llama-tiny-tp" if args.tp > 1 else "llama-tiny"
     arch = PRESETS[model]
 
     replica_seed = 1234 + (rank // args.tp)

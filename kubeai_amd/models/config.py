@@ -39,6 +39,13 @@ class ModelArchConfig:
     # "all": every layer windowed (Mistral); "even": layers 0,2,4,..
     # windowed (Gemma2 layer_types pattern)
     sliding_window_pattern: str = "all"
+    # gemma3: every Nth layer is GLOBAL attention (no window, global
+    # rope base); the rest are windowed with rope_local_base_freq.
+    # 0 disables the pattern.
+    global_layer_interval: int = 0
+    rope_local_base_freq: float = 0.0
+    # gemma3: per-head RMSNorm on q and k before rope
+    qk_norm: bool = False
     # gemma2 family knobs
     hidden_act: str = "silu"  # or "gelu_pytorch_tanh" (GeGLU)
     norm_plus_one: bool = False  # RMSNorm multiplies by (1 + w)
@@ -66,8 +73,15 @@ class ModelArchConfig:
         arch = (cfg.get("architectures") or ["LlamaForCausalLM"])[0]
         gemma2 = arch == "Gemma2ForCausalLM"
         gemma1 = arch == "GemmaForCausalLM"
+        gemma3 = arch in ("Gemma3ForCausalLM", "Gemma3ForConditionalGeneration")
         vision = None
         image_token_id = -1
+        if arch == "Gemma3ForConditionalGeneration":
+            # text-only serving of the text tower (the gemma3 SigLIP
+            # vision tower is not implemented; LLaVA covers multimodal)
+            cfg = {**cfg, **(cfg.get("text_config") or {})}
+            hidden = cfg["hidden_size"]
+            n_heads = cfg["num_attention_heads"]
         if arch == "LlavaForConditionalGeneration":
             # llava config nests text_config + vision_config
             vision = dict(cfg.get("vision_config") or {})
@@ -98,7 +112,7 @@ class ModelArchConfig:
             rms_norm_eps=cfg.get("rms_norm_eps", 1e-5),
             rope_theta=cfg.get("rope_theta", 10000.0),
             tie_word_embeddings=cfg.get(
-                "tie_word_embeddings", gemma2 or gemma1
+                "tie_word_embeddings", gemma2 or gemma1 or gemma3
             ),
             bos_token_id=cfg.get("bos_token_id", 1),
             eos_token_id=eos,
@@ -109,13 +123,21 @@ class ModelArchConfig:
             sliding_window_pattern="even" if gemma2 else "all",
             hidden_act=(
                 "gelu_pytorch_tanh"
-                if gemma2 or gemma1
+                if gemma2 or gemma1 or gemma3
                 or "gelu" in str(cfg.get("hidden_act") or "")
                 else "silu"
             ),
-            norm_plus_one=gemma2 or gemma1,
-            scale_embeddings=gemma2 or gemma1,
-            post_norms=gemma2,
+            norm_plus_one=gemma2 or gemma1 or gemma3,
+            scale_embeddings=gemma2 or gemma1 or gemma3,
+            post_norms=gemma2 or gemma3,
+            global_layer_interval=(
+                int(cfg.get("sliding_window_pattern") or 6) if gemma3 else 0
+            ),
+            rope_local_base_freq=(
+                float(cfg.get("rope_local_base_freq") or 10000.0)
+                if gemma3 else 0.0
+            ),
+            qk_norm=gemma3,
             attn_logit_softcap=float(
                 cfg.get("attn_logit_softcapping") or 0.0
             ) if gemma2 else 0.0,
@@ -124,7 +146,7 @@ class ModelArchConfig:
             ) if gemma2 else 0.0,
             query_pre_attn_scalar=float(
                 cfg.get("query_pre_attn_scalar") or 0.0
-            ) if gemma2 else 0.0,
+            ) if (gemma2 or gemma3) else 0.0,
             vision=vision,
             image_token_id=image_token_id,
         )
@@ -299,6 +321,57 @@ PRESETS: dict[str, ModelArchConfig] = {
         post_norms=True,
         attn_logit_softcap=50.0,
         final_logit_softcap=30.0,
+        query_pre_attn_scalar=256.0,
+    ),
+    # Gemma3-4b text shape: qk-norm, 5 local(1024-window, theta 10k) :
+    # 1 global(theta 1M, linear x8) layer pattern, post-norms, GeGLU
+    "gemma3-4b": ModelArchConfig(
+        vocab_size=262208,
+        hidden_size=2560,
+        intermediate_size=10240,
+        num_hidden_layers=34,
+        num_attention_heads=8,
+        num_key_value_heads=4,
+        head_dim=256,
+        max_position_embeddings=131072,
+        rope_theta=1000000.0,
+        rope_scaling={"rope_type": "linear", "factor": 8.0},
+        rope_local_base_freq=10000.0,
+        rms_norm_eps=1e-6,
+        tie_word_embeddings=True,
+        bos_token_id=2,
+        eos_token_id=1,
+        sliding_window=1024,
+        global_layer_interval=6,
+        hidden_act="gelu_pytorch_tanh",
+        norm_plus_one=True,
+        scale_embeddings=True,
+        post_norms=True,
+        qk_norm=True,
+        query_pre_attn_scalar=256.0,
+    ),
+    "gemma3-tiny": ModelArchConfig(
+        vocab_size=2048,
+        hidden_size=256,
+        intermediate_size=512,
+        num_hidden_layers=3,
+        num_attention_heads=2,
+        num_key_value_heads=1,
+        head_dim=256,
+        max_position_embeddings=2048,
+        rope_theta=1000000.0,
+        rope_local_base_freq=10000.0,
+        rms_norm_eps=1e-6,
+        tie_word_embeddings=True,
+        bos_token_id=2,
+        eos_token_id=1,
+        sliding_window=32,
+        global_layer_interval=3,
+        hidden_act="gelu_pytorch_tanh",
+        norm_plus_one=True,
+        scale_embeddings=True,
+        post_norms=True,
+        qk_norm=True,
         query_pre_attn_scalar=256.0,
     ),
     # Mistral-7B v0.1/v0.2 shape: llama arch + 4096-token sliding window

@@ -37,6 +37,13 @@ class EngineLinear(nn.Linear):
         return y
 
 
+def _rms_head(x: torch.Tensor, w: torch.Tensor, eps: float) -> torch.Tensor:
+    """Per-head RMSNorm over the last dim (gemma3 qk-norm)."""
+    xf = x.float()
+    xf = xf * torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + eps)
+    return (xf * w.float()).to(x.dtype)
+
+
 class Attention(nn.Module):
     def __init__(self, cfg: ModelArchConfig, layer_idx: int):
         super().__init__()
@@ -45,9 +52,22 @@ class Attention(nn.Module):
         self.n_kv = cfg.num_key_value_heads
         self.hd = cfg.head_dim
         # per-layer window: gemma2 interleaves windowed ("even" layers)
-        # and full-causal attention; mistral windows every layer
-        use_win = cfg.sliding_window_pattern != "even" or layer_idx % 2 == 0
+        # and full-causal attention; gemma3 makes every Nth layer GLOBAL
+        # (others windowed, local rope base); mistral windows every layer
+        if cfg.global_layer_interval > 0:
+            self.is_global = (layer_idx + 1) % cfg.global_layer_interval == 0
+            use_win = not self.is_global
+        else:
+            self.is_global = True
+            use_win = (
+                cfg.sliding_window_pattern != "even" or layer_idx % 2 == 0
+            )
         self.window = (cfg.sliding_window or 0) if use_win else 0
+        self.qk_norm = cfg.qk_norm
+        if cfg.qk_norm:
+            self.q_norm = nn.Parameter(torch.ones(cfg.head_dim))
+            self.k_norm = nn.Parameter(torch.ones(cfg.head_dim))
+        self.norm_eps = cfg.rms_norm_eps
         self.softcap = cfg.attn_logit_softcap or 0.0
         self.scale = 1.0 / math.sqrt(cfg.query_pre_attn_scalar or self.hd)
         q_size = self.n_q * self.hd
@@ -85,6 +105,14 @@ class Attention(nn.Module):
         q = q.unflatten(-1, (self.n_q, self.hd))
         k = k.unflatten(-1, (self.n_kv, self.hd))
         v = v.unflatten(-1, (self.n_kv, self.hd))
+        if self.qk_norm:
+            # gemma3: per-head RMSNorm ((1+w) stored as w) before rope;
+            # fp32 math, contiguous bf16 out
+            q = _rms_head(q, self.q_norm, self.norm_eps)
+            k = _rms_head(k, self.k_norm, self.norm_eps)
+        if isinstance(cos_sin, tuple):
+            # (global, local) caches — gemma3 dual rope bases
+            cos_sin = cos_sin[0] if self.is_global else cos_sin[1]
         q, k = ops.rope(q, k, fb.positions, cos_sin)
         k_cache, v_cache = kv_cache
         ops.reshape_and_cache(k, v, k_cache, v_cache, fb.slot_mapping)
@@ -318,9 +346,26 @@ class LlamaForCausalLM(nn.Module):
             ).to(device=device),
             persistent=False,
         )
+        if cfg.rope_local_base_freq:
+            # gemma3: windowed layers rope with the LOCAL base, unscaled
+            self.register_buffer(
+                "cos_sin_local",
+                ops_ref.make_cos_sin_cache(
+                    cfg.head_dim, cfg.max_position_embeddings,
+                    cfg.rope_local_base_freq,
+                ).to(device=device),
+                persistent=False,
+            )
+        else:
+            self.cos_sin_local = None
 
     @torch.inference_mode()
     def forward(self, fb: ForwardBatch) -> torch.Tensor:
+        cs = (
+            (self.cos_sin, self.cos_sin_local)
+            if self.cos_sin_local is not None
+            else self.cos_sin
+        )
         x = self.embed_tokens(fb.input_ids.long())
         if self.cfg.scale_embeddings:
             # gemma: embeddings scaled by sqrt(hidden), cast to model dtype
@@ -333,7 +378,7 @@ class LlamaForCausalLM(nn.Module):
             x = x.index_copy(0, fb.mm_idx, fb.mm_embeds.to(x.dtype))
         residual = None
         for i, layer in enumerate(self.layers):
-            x, residual = layer(x, residual, fb, self.kv_caches[i], self.cos_sin)
+            x, residual = layer(x, residual, fb, self.kv_caches[i], cs)
         x, _ = ops.fused_add_rmsnorm(x, residual, self.norm, self.cfg.rms_norm_eps)
         return x
 

@@ -351,6 +351,10 @@ def load_weights(model, model_dir: str, vision=None) -> int:
                 put(f"layers.{layer}.pre_feedforward_layernorm", norm_w(w))
             elif rest == "post_feedforward_layernorm.weight":
                 put(f"layers.{layer}.post_feedforward_layernorm", norm_w(w))
+            elif rest == "self_attn.q_norm.weight":
+                put(f"layers.{layer}.self_attn.q_norm", norm_w(w))
+            elif rest == "self_attn.k_norm.weight":
+                put(f"layers.{layer}.self_attn.k_norm", norm_w(w))
             elif rest == "self_attn.qkv_proj.weight":
                 # phi-3/phi-4 checkpoints pre-fuse qkv ([q;k;v] rows —
                 # same layout as the engine's fused projection)
@@ -453,6 +457,13 @@ def save_hf_checkpoint(model, out_dir: str) -> None:
             state[pre + "post_feedforward_layernorm.weight"] = norm_out(
                 layer.post_feedforward_layernorm
             )
+        if getattr(layer.self_attn, "qk_norm", False):
+            state[pre + "self_attn.q_norm.weight"] = norm_out(
+                layer.self_attn.q_norm
+            )
+            state[pre + "self_attn.k_norm.weight"] = norm_out(
+                layer.self_attn.k_norm
+            )
         qkv = layer.self_attn.qkv_proj.weight.detach().cpu()
         state[pre + "self_attn.q_proj.weight"] = qkv[: nq * hd].clone()
         state[pre + "self_attn.k_proj.weight"] = qkv[nq * hd : (nq + nkv) * hd].clone()
@@ -488,6 +499,7 @@ def save_hf_checkpoint(model, out_dir: str) -> None:
             {
                 "architectures": [
                     "MixtralForCausalLM" if cfg.num_local_experts
+                    else "Gemma3ForCausalLM" if cfg.qk_norm
                     else "Gemma2ForCausalLM" if cfg.post_norms
                     else "GemmaForCausalLM" if cfg.norm_plus_one
                     else "LlamaForCausalLM"
@@ -514,6 +526,8 @@ def save_hf_checkpoint(model, out_dir: str) -> None:
                 "final_logit_softcapping": cfg.final_logit_softcap or None,
                 "query_pre_attn_scalar": cfg.query_pre_attn_scalar or None,
                 "hidden_act": cfg.hidden_act,
+                "sliding_window_pattern": cfg.global_layer_interval or None,
+                "rope_local_base_freq": cfg.rope_local_base_freq or None,
                 "torch_dtype": "bfloat16",
             },
             f,

@@ -372,3 +372,21 @@ def test_multimodal_llava_gpu():
     img2 = torch.randint(0, 256, (40, 56, 3), generator=g2, dtype=torch.uint8)
     o3 = run("mm3", imaging.preprocess(img2, 64))
     assert o3 is not None and o3.num_cached_tokens == 0
+
+
+@pytest.mark.gpu
+def test_gemma3_engine_e2e_gpu():
+    """gemma3 on device: qk-norm + dual rope + local:global layers."""
+    eng = LLMEngine(EngineConfig(model="gemma3-tiny", device="cuda",
+                                 num_gpu_blocks=256, max_model_len=512))
+    prompt = [2] + list(range(100, 170))
+    eng.add_request(prompt, SamplingParams(max_tokens=6, ignore_eos=True),
+                    request_id="g3")
+    out = None
+    for _ in range(100):
+        if not eng.has_work():
+            break
+        for o in eng.step():
+            if o.finished:
+                out = o
+    assert out is not None and len(out.output_token_ids) == 6

@@ -107,9 +107,11 @@ class Attention(nn.Module):
         v = v.unflatten(-1, (self.n_kv, self.hd))
         if self.qk_norm:
             # gemma3: per-head RMSNorm ((1+w) stored as w) before rope;
-            # fp32 math, contiguous bf16 out
+            # fp32 math, contiguous bf16 out. v follows to contiguous —
+            # the cache-write kernel wants k/v with one shared row stride
             q = _rms_head(q, self.q_norm, self.norm_eps)
             k = _rms_head(k, self.k_norm, self.norm_eps)
+            v = v.contiguous()
         if isinstance(cos_sin, tuple):
             # (global, local) caches — gemma3 dual rope bases
             cos_sin = cos_sin[0] if self.is_global else cos_sin[1]

@@ -46,6 +46,11 @@ class ModelArchConfig:
     rope_local_base_freq: float = 0.0
     # gemma3: per-head RMSNorm on q and k before rope
     qk_norm: bool = False
+    # granite-3 scalar multipliers (0 = disabled/default)
+    embedding_multiplier: float = 0.0
+    residual_multiplier: float = 0.0
+    logits_scaling: float = 0.0
+    attention_multiplier: float = 0.0
     # gemma2 family knobs
     hidden_act: str = "silu"  # or "gelu_pytorch_tanh" (GeGLU)
     norm_plus_one: bool = False  # RMSNorm multiplies by (1 + w)
@@ -74,6 +79,7 @@ class ModelArchConfig:
         gemma2 = arch == "Gemma2ForCausalLM"
         gemma1 = arch == "GemmaForCausalLM"
         gemma3 = arch in ("Gemma3ForCausalLM", "Gemma3ForConditionalGeneration")
+        granite = arch == "GraniteForCausalLM"
         vision = None
         image_token_id = -1
         if arch == "Gemma3ForConditionalGeneration":
@@ -138,6 +144,18 @@ class ModelArchConfig:
                 if gemma3 else 0.0
             ),
             qk_norm=gemma3,
+            embedding_multiplier=float(
+                cfg.get("embedding_multiplier") or 0.0
+            ) if granite else 0.0,
+            residual_multiplier=float(
+                cfg.get("residual_multiplier") or 0.0
+            ) if granite else 0.0,
+            logits_scaling=float(
+                cfg.get("logits_scaling") or 0.0
+            ) if granite else 0.0,
+            attention_multiplier=float(
+                cfg.get("attention_multiplier") or 0.0
+            ) if granite else 0.0,
             attn_logit_softcap=float(
                 cfg.get("attn_logit_softcapping") or 0.0
             ) if gemma2 else 0.0,
@@ -322,6 +340,42 @@ PRESETS: dict[str, ModelArchConfig] = {
         attn_logit_softcap=50.0,
         final_logit_softcap=30.0,
         query_pre_attn_scalar=256.0,
+    ),
+    # Granite-3.1-8b dense shape: llama structure + scalar multipliers
+    "granite-3-8b": ModelArchConfig(
+        vocab_size=49155,
+        hidden_size=4096,
+        intermediate_size=12800,
+        num_hidden_layers=40,
+        num_attention_heads=32,
+        num_key_value_heads=8,
+        head_dim=128,
+        max_position_embeddings=131072,
+        rope_theta=10000000.0,
+        tie_word_embeddings=True,
+        bos_token_id=0,
+        eos_token_id=0,
+        embedding_multiplier=12.0,
+        residual_multiplier=0.22,
+        logits_scaling=16.0,
+        attention_multiplier=0.0078125,
+    ),
+    "granite-tiny": ModelArchConfig(
+        vocab_size=2048,
+        hidden_size=256,
+        intermediate_size=512,
+        num_hidden_layers=2,
+        num_attention_heads=2,
+        num_key_value_heads=1,
+        head_dim=128,
+        max_position_embeddings=2048,
+        tie_word_embeddings=True,
+        bos_token_id=1,
+        eos_token_id=2,
+        embedding_multiplier=12.0,
+        residual_multiplier=0.22,
+        logits_scaling=16.0,
+        attention_multiplier=0.0078125,
     ),
     # Gemma3-4b text shape: qk-norm, 5 local(1024-window, theta 10k) :
     # 1 global(theta 1M, linear x8) layer pattern, post-norms, GeGLU

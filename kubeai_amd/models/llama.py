@@ -76,7 +76,10 @@ class Attention(nn.Module):
             self.k_norm = nn.Parameter(torch.ones(cfg.head_dim))
         self.norm_eps = cfg.rms_norm_eps
         self.softcap = cfg.attn_logit_softcap or 0.0
-        self.scale = 1.0 / math.sqrt(cfg.query_pre_attn_scalar or self.hd)
+        self.scale = (
+            cfg.attention_multiplier
+            or 1.0 / math.sqrt(cfg.query_pre_attn_scalar or self.hd)
+        )
         q_size = self.n_q * self.hd
         kv_size = self.n_kv * self.hd
         self.qkv_proj = EngineLinear(
@@ -252,6 +255,9 @@ class DecoderLayer(nn.Module):
         # gemma2: branch outputs are normed BEFORE the residual add, with
         # separate pre/post feed-forward norms (4 norms per layer)
         self.post_norms = cfg.post_norms
+        # granite: branch outputs scale by residual_multiplier before the
+        # residual add
+        self.res_mult = cfg.residual_multiplier or 1.0
         if cfg.post_norms:
             self.pre_feedforward_layernorm = nn.Parameter(
                 torch.ones(cfg.hidden_size)
@@ -313,10 +319,14 @@ class DecoderLayer(nn.Module):
                 x, residual, self.input_layernorm, self.eps
             )
         x = self.self_attn(x, fb, kv_cache, cos_sin)
+        if self.res_mult != 1.0:
+            x = x * self.res_mult
         x, residual = ops.fused_add_rmsnorm(
             x, residual, self.post_attention_layernorm, self.eps
         )
         x = self.mlp(x, fb)
+        if self.res_mult != 1.0:
+            x = x * self.res_mult
         return x, residual
 
 
@@ -382,6 +392,8 @@ class LlamaForCausalLM(nn.Module):
             x = x * torch.tensor(
                 self.cfg.hidden_size ** 0.5, dtype=x.dtype
             )
+        elif self.cfg.embedding_multiplier:
+            x = x * self.cfg.embedding_multiplier  # granite
         if fb.mm_embeds is not None:
             # image-placeholder positions take the vision-tower embeddings
             x = x.index_copy(0, fb.mm_idx, fb.mm_embeds.to(x.dtype))
@@ -397,6 +409,8 @@ class LlamaForCausalLM(nn.Module):
             logits = F.linear(hidden, self.embed_tokens.weight).float()
         else:
             logits = self.lm_head(hidden).float()
+        if self.cfg.logits_scaling:
+            logits = logits / self.cfg.logits_scaling  # granite
         cap = self.cfg.final_logit_softcap
         if cap > 0:
             logits = cap * torch.tanh(logits / cap)

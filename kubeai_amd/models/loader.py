@@ -502,6 +502,7 @@ def save_hf_checkpoint(model, out_dir: str) -> None:
                     else "Gemma3ForCausalLM" if cfg.qk_norm
                     else "Gemma2ForCausalLM" if cfg.post_norms
                     else "GemmaForCausalLM" if cfg.norm_plus_one
+                    else "GraniteForCausalLM" if cfg.residual_multiplier
                     else "LlamaForCausalLM"
                 ],
                 "vocab_size": cfg.vocab_size,
@@ -528,6 +529,10 @@ def save_hf_checkpoint(model, out_dir: str) -> None:
                 "hidden_act": cfg.hidden_act,
                 "sliding_window_pattern": cfg.global_layer_interval or None,
                 "rope_local_base_freq": cfg.rope_local_base_freq or None,
+                "embedding_multiplier": cfg.embedding_multiplier or None,
+                "residual_multiplier": cfg.residual_multiplier or None,
+                "logits_scaling": cfg.logits_scaling or None,
+                "attention_multiplier": cfg.attention_multiplier or None,
                 "torch_dtype": "bfloat16",
             },
             f,

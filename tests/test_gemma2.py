@@ -326,3 +326,66 @@ def test_gemma3_qk_norm_changes_output():
     rms = pre.float().pow(2).mean(-1).sqrt()
     torch.testing.assert_close(rms, torch.ones_like(rms), atol=2e-2,
                                rtol=2e-2)
+
+
+# ------------------------------------------------------------ granite
+def test_granite_engine_e2e_and_roundtrip(tmp_path):
+    from kubeai_amd.models.loader import save_hf_checkpoint
+
+    src = LLMEngine(EngineConfig(model="granite-tiny", device="cpu",
+                                 num_gpu_blocks=128, enable_graphs=False,
+                                 max_model_len=512, seed=3))
+    a = src.arch
+    assert a.residual_multiplier == 0.22 and a.logits_scaling == 16.0
+    assert src.runner.model.layers[0].self_attn.scale == 0.0078125
+    prompt = [1] + list(range(100, 140))
+
+    def gen(eng, rid):
+        eng.add_request(prompt, SamplingParams(max_tokens=6, ignore_eos=True),
+                        request_id=rid)
+        return _drain(eng)[rid].output_token_ids
+
+    first = gen(src, "gr1")
+    assert len(first) == 6
+    ckpt = str(tmp_path / "granite")
+    save_hf_checkpoint(src.runner.model, ckpt)
+    arch = ModelArchConfig.from_hf_config(ckpt)
+    assert arch.residual_multiplier == 0.22
+    assert arch.attention_multiplier == 0.0078125
+    dst = LLMEngine(EngineConfig(model=ckpt, device="cpu",
+                                 num_gpu_blocks=128, max_model_len=512,
+                                 enable_graphs=False, seed=999))
+    assert gen(dst, "gr2") == first
+
+
+def test_granite_multipliers_change_logits():
+    import dataclasses
+
+    from kubeai_amd.models.llama import LlamaForCausalLM
+    from kubeai_amd.engine.batch import ForwardBatch
+
+    cfg = PRESETS["granite-tiny"]
+    plain = dataclasses.replace(
+        cfg, embedding_multiplier=0.0, residual_multiplier=0.0,
+        logits_scaling=0.0, attention_multiplier=0.0,
+    )
+    outs = []
+    for c in (cfg, plain):
+        torch.manual_seed(5)
+        m = LlamaForCausalLM(c, device="cpu", dtype=torch.float32)
+        kc = torch.zeros(4, c.num_key_value_heads, 16, c.head_dim)
+        m.kv_caches = [(kc.clone(), kc.clone())
+                       for _ in range(c.num_hidden_layers)]
+        fb = ForwardBatch(
+            input_ids=torch.tensor([1, 100, 101], dtype=torch.int32),
+            positions=torch.arange(3, dtype=torch.int32),
+            slot_mapping=torch.arange(3, dtype=torch.int64),
+            n_decode=0, decode_block_tables=None, decode_seq_lens=None,
+            n_prefill=1,
+            prefill_query_start_loc=torch.tensor([0, 3], dtype=torch.int32),
+            prefill_seq_lens=torch.tensor([3], dtype=torch.int32),
+            prefill_block_tables=torch.tensor([[0, 1]], dtype=torch.int32),
+            logits_indices=torch.tensor([2]),
+        )
+        outs.append(m.compute_logits(m.forward(fb)[fb.logits_indices]))
+    assert not torch.allclose(outs[0], outs[1])

@@ -390,3 +390,20 @@ def test_gemma3_engine_e2e_gpu():
             if o.finished:
                 out = o
     assert out is not None and len(out.output_token_ids) == 6
+
+
+@pytest.mark.gpu
+def test_granite_engine_e2e_gpu():
+    eng = LLMEngine(EngineConfig(model="granite-tiny", device="cuda",
+                                 num_gpu_blocks=256, max_model_len=512))
+    eng.add_request([1] + list(range(100, 140)),
+                    SamplingParams(max_tokens=6, ignore_eos=True),
+                    request_id="gr")
+    out = None
+    for _ in range(100):
+        if not eng.has_work():
+            break
+        for o in eng.step():
+            if o.finished:
+                out = o
+    assert out is not None and len(out.output_token_ids) == 6

@@ -48,6 +48,8 @@ class Autoscaler:
         time_window: float = 600.0,
         self_metric_addrs: Optional[list[str]] = None,
         state_path: Optional[str] = None,
+        state_store=None,  # object with load()->dict / save(dict); K8s
+        # mode passes a ConfigMap-backed store (reference state.go CM)
         is_leader=lambda: True,
         scrape_engine_queues: bool = True,
     ):
@@ -57,13 +59,16 @@ class Autoscaler:
         self.window_count = max(1, int(time_window / interval))
         self.self_metric_addrs = self_metric_addrs or []
         self.state_path = state_path
+        self.state_store = state_store
         self.is_leader = is_leader
         self.scrape_engine_queues = scrape_engine_queues
         self.averages: dict[str, SimpleMovingAverage] = {}
         self._task: Optional[asyncio.Task] = None
         self._client = httpx.AsyncClient(timeout=5.0)
         self.last_scales: dict[str, int] = {}
-        if state_path and os.path.exists(state_path):
+        if state_store is not None:
+            self._load_state()
+        elif state_path and os.path.exists(state_path):
             self._load_state()
 
     # ------------------------------------------------------------ lifecycle
@@ -108,7 +113,7 @@ class Autoscaler:
                 target = max(target, pressured + 1)
             self.last_scales[model.name] = target
             self.model_client.scale(model.name, target)
-        if self.state_path:
+        if self.state_path or self.state_store is not None:
             self._save_state()
 
     def _avg_for(self, name: str) -> SimpleMovingAverage:
@@ -170,6 +175,12 @@ class Autoscaler:
     # ------------------------------------------------------------ state
     def _save_state(self) -> None:
         data = {name: avg.history() for name, avg in self.averages.items()}
+        if self.state_store is not None:
+            try:
+                self.state_store.save(data)
+            except Exception:  # noqa: BLE001 — state is best-effort
+                pass
+            return
         tmp = self.state_path + ".tmp"
         with open(tmp, "w") as f:
             json.dump(data, f)
@@ -177,8 +188,11 @@ class Autoscaler:
 
     def _load_state(self) -> None:
         try:
-            with open(self.state_path) as f:
-                data = json.load(f)
+            if self.state_store is not None:
+                data = self.state_store.load()
+            else:
+                with open(self.state_path) as f:
+                    data = json.load(f)
         except Exception:
             return
         for name, values in data.items():

@@ -18,6 +18,7 @@ from __future__ import annotations
 
 import asyncio
 import itertools
+import json
 import threading
 import time
 import uuid
@@ -447,3 +448,36 @@ def _parse_micro_time(s: Optional[str]) -> Optional[float]:
     head, _, frac = s.rstrip("Z").partition(".")
     t = calendar.timegm(time.strptime(head, "%Y-%m-%dT%H:%M:%S"))
     return t + (float("0." + frac) if frac else 0.0)
+
+
+class ConfigMapStateStore:
+    """Autoscaler moving-average state in a ConfigMap (reference
+    internal/modelautoscaler/state.go: state persisted to a CM each tick
+    and preloaded on boot, so leader failover keeps the windows)."""
+
+    def __init__(self, kc: KubeClient, name: str = "kubeai-autoscaler-state"):
+        self.kc = kc
+        self.name = name
+        self._path = (
+            f"/api/v1/namespaces/{kc.namespace}/configmaps/{name}"
+        )
+
+    def load(self) -> dict:
+        cm = self.kc.get_opt(self._path)
+        if cm is None:
+            return {}
+        raw = (cm.get("data") or {}).get("state", "{}")
+        return json.loads(raw)
+
+    def save(self, data: dict) -> None:
+        body = {
+            "apiVersion": "v1",
+            "kind": "ConfigMap",
+            "metadata": {"name": self.name,
+                         "namespace": self.kc.namespace},
+            "data": {"state": json.dumps(data)},
+        }
+        if self.kc.get_opt(self._path) is None:
+            self.kc.create(self._path.rsplit("/", 1)[0], body)
+        else:
+            self.kc.patch_merge(self._path, {"data": body["data"]})

@@ -56,7 +56,12 @@ class Manager:
                 kc, lease_name=kube.get("leaseName", "kubeai.org")
             )
             from .kubeclient import KubeCacheManager
+            from .kubestore import ConfigMapStateStore
 
+            kube_state = ConfigMapStateStore(
+                kc, name=kube.get("autoscalerStateConfigMap",
+                                  "kubeai-autoscaler-state"),
+            )
             kube_cache = KubeCacheManager(
                 kc,
                 cache_profiles=self.cfg.cacheProfiles,
@@ -67,6 +72,7 @@ class Manager:
             )
         else:
             kube_cache = None
+            kube_state = None
             self.store = Store()
             self.runtime = runtime or LocalProcessRuntime(
                 self.store, n_gpus=self.cfg.n_gpus
@@ -91,6 +97,7 @@ class Manager:
             time_window=self.cfg.autoscaling.time_window_seconds,
             self_metric_addrs=self.cfg.fixed_self_metric_addrs,
             state_path=self.cfg.autoscaling.state_path,
+            state_store=kube_state,
             is_leader=self.election.is_leader,
         )
         self.proxy = ProxyHandler(

@@ -556,6 +556,31 @@ def build_app(server: EngineServer) -> FastAPI:
 
         return {"version": kubeai_amd.__version__}
 
+    @app.post("/tokenize")
+    async def tokenize(request: Request):
+        """vLLM-compat tokenizer endpoint (clients pointed straight at an
+        engine replica use it for budget accounting)."""
+        body = await request.json()
+        text = body.get("prompt")
+        if text is None and body.get("messages"):
+            toks = apply_chat_template(server.tokenizer, body["messages"])
+            return {"tokens": toks, "count": len(toks),
+                    "max_model_len": server.cfg.max_model_len}
+        if not isinstance(text, str):
+            raise RequestError("tokenize needs a string `prompt` or `messages`")
+        add_special = bool(body.get("add_special_tokens", True))
+        toks = server.tokenizer.encode(text, add_bos=add_special)
+        return {"tokens": toks, "count": len(toks),
+                "max_model_len": server.cfg.max_model_len}
+
+    @app.post("/detokenize")
+    async def detokenize(request: Request):
+        body = await request.json()
+        toks = body.get("tokens")
+        if not isinstance(toks, list):
+            raise RequestError("detokenize needs a `tokens` array")
+        return {"prompt": server.tokenizer.decode([int(t) for t in toks])}
+
     @app.get("/metrics")
     async def metrics():
         return PlainTextResponse(

@@ -615,3 +615,15 @@ def test_kube_cache_unknown_profile_errors(api):
             assert "cacheProfile" in str(e)
 
     asyncio.run(main())
+
+
+def test_configmap_autoscaler_state_store(api):
+    from kubeai_amd.controlplane.kubestore import ConfigMapStateStore
+
+    kc = kc_for(api)
+    st = ConfigMapStateStore(kc, name="as-state-test")
+    assert st.load() == {}
+    st.save({"m1": [1.0, 2.0, 3.0]})
+    assert st.load() == {"m1": [1.0, 2.0, 3.0]}
+    st.save({"m1": [4.0]})  # update path (CM exists)
+    assert st.load() == {"m1": [4.0]}

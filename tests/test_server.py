@@ -527,3 +527,18 @@ def test_parse_tool_call_shapes():
     assert _parse_tool_call('{"name": "g", "arguments": {}}', tools) is None
     assert _parse_tool_call("not json", tools) is None
     assert _parse_tool_call('{"name": "f"}', tools) == ("f", {})
+
+
+def test_tokenize_detokenize_endpoints(client):
+    r = client.post("/tokenize", json={"prompt": "hello world"})
+    assert r.status_code == 200
+    body = r.json()
+    assert body["count"] == len(body["tokens"]) > 0
+    assert body["max_model_len"] == 512
+    r2 = client.post("/detokenize", json={"tokens": body["tokens"]})
+    assert r2.status_code == 200
+    assert isinstance(r2.json()["prompt"], str)
+    r3 = client.post("/tokenize", json={
+        "messages": [{"role": "user", "content": "hi"}]})
+    assert r3.status_code == 200 and r3.json()["count"] > 0
+    assert client.post("/tokenize", json={}).status_code == 400

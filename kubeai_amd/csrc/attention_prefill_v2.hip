@@ -61,8 +61,14 @@ DEV_INLINE int crow(int reg, int hi) {
 // ~430 SGPRs (~650 B/lane scratch): 90 TF @Tq=8192. MINW=2 gives 256
 // VGPRs, near-zero spill (≤21 v, ~80 s), 283 TF — 1 workgroup/CU beats
 // 2 spilly ones 3x over. Measured same-box; KUBEAI_V2_OCC=4 re-checks.
-template <int G, typename CT = ushort, int DBG = 0, int MINW = 2>
-__launch_bounds__(512, MINW) __global__ void paged_prefill_v2_kernel(
+// NW: waves per workgroup. 8 (default) = 256 q rows/WG, 1 WG/CU at 256
+// VGPRs. 4 = 32*NW/G q rows/WG with HALF the arithmetic intensity per
+// KV byte but TWO co-resident WGs/CU (128 KB LDS) to overlap each
+// other's latency stalls — opt-in experiment via KUBEAI_V2_WAVES=4.
+template <int G, typename CT = ushort, int DBG = 0, int MINW = 2,
+          int NW = 8>
+__launch_bounds__(NW * WAVE_SIZE, MINW) __global__
+void paged_prefill_v2_kernel(
     ushort* __restrict__ out,            // [Tq, n_q, hd]
     const ushort* __restrict__ q,        // [Tq, n_q, hd]
     const CT* __restrict__ k_cache,      // [nb, n_kv, bs, hd]
@@ -78,7 +84,10 @@ __launch_bounds__(512, MINW) __global__ void paged_prefill_v2_kernel(
   //   bit 1: plain scalar V reads instead of ds_read_b64_tr_b16
   constexpr bool kSyncStage = DBG & 1;
   constexpr bool kSimpleV = DBG & 2;
-  constexpr int kQSUB = 8 / G;          // q subtiles per head
+  // NW < G would leave heads without waves; the host never launches
+  // that combination (falls back to NW=8), and the max() here keeps the
+  // dead instantiation compilable without a hard error
+  constexpr int kQSUB = (NW >= G) ? NW / G : 1;  // q subtiles per head
   constexpr int kQROWS = kQSUB * kQB;   // q rows per workgroup
   const int b = blockIdx.x;
   const int kh = blockIdx.y;
@@ -138,12 +147,13 @@ __launch_bounds__(512, MINW) __global__ void paged_prefill_v2_kernel(
   // unit u of 1024: tok = u>>4, c8 = u&15 (16 bf16-8 units per 128-row).
   // All thread-invariant addressing is hoisted; per tile only the block
   // lookup + validity remain.
-  bf16x8 st_k[2], st_v[2];
-  bool st_ok[2];
-  int h_tok[2], h_coff[2], h_kdst[2], h_vdst[2];
+  constexpr int NSTG = (kKVB * kHD / 8) / (NW * WAVE_SIZE);
+  bf16x8 st_k[NSTG], st_v[NSTG];
+  bool st_ok[NSTG];
+  int h_tok[NSTG], h_coff[NSTG], h_kdst[NSTG], h_vdst[NSTG];
 #pragma unroll
-  for (int i = 0; i < 2; ++i) {
-    const int u = threadIdx.x + i * 512;
+  for (int i = 0; i < NSTG; ++i) {
+    const int u = threadIdx.x + i * NW * WAVE_SIZE;
     const int tok = u >> 4;
     const int c8 = u & 15;
     h_tok[i] = tok;
@@ -156,7 +166,7 @@ __launch_bounds__(512, MINW) __global__ void paged_prefill_v2_kernel(
     const int kv_start = kt * kKVB;
     const int kv_valid = min(kKVB, kv_limit - kv_start);
 #pragma unroll
-    for (int i = 0; i < 2; ++i) {
+    for (int i = 0; i < NSTG; ++i) {
       const int tok = h_tok[i];
       st_ok[i] = tok < kv_valid;
       if (st_ok[i]) {
@@ -190,7 +200,7 @@ __launch_bounds__(512, MINW) __global__ void paged_prefill_v2_kernel(
   };
   auto write_tile = [&](int buf) {
 #pragma unroll
-    for (int i = 0; i < 2; ++i) {
+    for (int i = 0; i < NSTG; ++i) {
       // K: swizzled row-major (skip stale-write cost only when invalid)
       if (st_ok[i])
         *reinterpret_cast<bf16x8*>(
@@ -459,8 +469,28 @@ bool paged_attention_prefill_v2(torch::Tensor out, torch::Tensor q,
     const int64_t approx_L = (int64_t)max_blocks * kBS;
     if (approx_L > 3 * q.size(0)) return false;
   }
-  // max q chunk rows: 8/G waves * 32
-  const int qrows = (8 / G) * kQB;
+  // workgroup width: 8 waves (256 q rows/WG, best arithmetic intensity)
+  // unless that would underfill the 256-CU chip — small prefills then
+  // run the 4-wave form, whose 2x workgroup count fills more CUs
+  // (measured: +11% at Tq=1024, -6% at Tq=8192 — same-box A/B).
+  // KUBEAI_V2_WAVES forces either.
+  static const int forced_waves = []() {
+    const char* e = getenv("KUBEAI_V2_WAVES");
+    return e ? atoi(e) : 0;
+  }();
+  int nw = 8;
+  if (G <= 4) {
+    if (forced_waves == 4 || forced_waves == 8) {
+      nw = forced_waves;
+    } else {
+      const int qrows8 = (8 / G) * kQB;
+      const int64_t wgs8 =
+          (int64_t)B * n_kv * ((q.size(0) + qrows8 - 1) / qrows8);
+      if (wgs8 < 256) nw = 4;
+    }
+  }
+  // max q chunk rows: nw/G waves * 32
+  const int qrows = (nw / G) * kQB;
   int max_qlen = 0;
   {
     // host copy of query_start_loc is cheap (B+1 ints, pinned path); the
@@ -473,7 +503,7 @@ bool paged_attention_prefill_v2(torch::Tensor out, torch::Tensor q,
     const char* e = getenv("KUBEAI_V2_DBG");
     return e ? atoi(e) : 0;
   }();
-  dim3 grid(B, n_kv, zdim), block(512);
+  dim3 grid(B, n_kv, zdim), block(nw * 64);
   auto stream = c10::hip::getCurrentHIPStream().stream();
   const bool fp8_cache = k_cache.scalar_type() == torch::kFloat8_e5m2;
   static const int occ = []() {
@@ -481,7 +511,18 @@ bool paged_attention_prefill_v2(torch::Tensor out, torch::Tensor q,
     return e ? atoi(e) : 2;
   }();
 #define LAUNCH_V2_CT_DW(GG, CT, D, W)                                     \
-  hipLaunchKernelGGL((paged_prefill_v2_kernel<GG, CT, D, W>), grid,       \
+  hipLaunchKernelGGL((paged_prefill_v2_kernel<GG, CT, D, W, 8>), grid,    \
+                     block,                                               \
+                     0, stream, (ushort*)out.data_ptr(),                  \
+                     (const ushort*)q.data_ptr(),                         \
+                     (const CT*)k_cache.data_ptr(),                       \
+                     (const CT*)v_cache.data_ptr(),                       \
+                     block_tables.data_ptr<int32_t>(),                    \
+                     query_start_loc.data_ptr<int32_t>(),                 \
+                     seq_lens.data_ptr<int32_t>(), (float)scale, n_kv,    \
+                     max_blocks, q.stride(0))
+#define LAUNCH_V2_NW4(GG, CT)                                             \
+  hipLaunchKernelGGL((paged_prefill_v2_kernel<GG, CT, 0, 2, 4>), grid,    \
                      block,                                               \
                      0, stream, (ushort*)out.data_ptr(),                  \
                      (const ushort*)q.data_ptr(),                         \
@@ -499,6 +540,7 @@ bool paged_attention_prefill_v2(torch::Tensor out, torch::Tensor q,
       case 3: LAUNCH_V2_CT_DW(GG, CT, 3, 2); break;                       \
       default:                                                            \
         if (occ == 4) LAUNCH_V2_CT_DW(GG, CT, 0, 4);                      \
+        else if (nw == 4) LAUNCH_V2_NW4(GG, CT);                          \
         else LAUNCH_V2_CT_DW(GG, CT, 0, 2);                               \
         break;                                                            \
     }                                                                     \
@@ -520,6 +562,7 @@ bool paged_attention_prefill_v2(torch::Tensor out, torch::Tensor q,
   }
 #undef LAUNCH_V2
 #undef LAUNCH_V2_CT
+#undef LAUNCH_V2_NW4
 #undef LAUNCH_V2_CT_DW
   HIP_CHECK_KERNEL();
   return true;

@@ -72,9 +72,15 @@ __launch_bounds__(kBlockThreads) __global__ void paged_decode_kernel(
   // (L-1-window, L-1] -> global positions >= w0; whole blocks below w0
   // are skipped, the straddling block is masked per token
   const int w0 = (window > 0) ? max(0, L - window) : 0;
-  const int chunk = (n_blocks + n_splits - 1) / n_splits;
-  const int blk_lo = max(split * chunk, w0 / kBS);
-  const int blk_hi = min(n_blocks, split * chunk + chunk);
+  // flash-decoding splits divide the WINDOWED block range — splitting
+  // the full range would leave every split below w0 empty for long
+  // sequences (mistral at L >> window would use a fraction of the
+  // launched workgroups)
+  const int blk_base = w0 / kBS;
+  const int n_eff = n_blocks - blk_base;
+  const int chunk = (n_eff + n_splits - 1) / n_splits;
+  const int blk_lo = blk_base + split * chunk;
+  const int blk_hi = min(n_blocks, blk_lo + chunk);
 
   // LDS: per-wave double-buffered KV tiles + merge scratch. Element type
   // follows the cache: bf16 stages bf16; fp8 stages the RAW e5m2 bytes

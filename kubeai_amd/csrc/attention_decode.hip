@@ -58,10 +58,7 @@ __launch_bounds__(kBlockThreads) __global__ void paged_decode_kernel(
     const int n_kv, const int max_blocks,
     const int64_t q_stride, const int n_splits,
     float* __restrict__ part_o,    // [B, n_q, n_splits, hd]
-    float* __restrict__ part_ml,   // [B, n_q, n_splits, 2]
-    int* __restrict__ merge_cnt) { // [B*n_kv] self-resetting; non-null
-                                   // fuses the split merge into the
-                                   // LAST-arriving workgroup
+    float* __restrict__ part_ml) { // [B, n_q, n_splits, 2]
   const int b = blockIdx.x / (n_kv * n_splits);
   const int rem = blockIdx.x % (n_kv * n_splits);
   const int kh = rem / n_splits;
@@ -344,54 +341,6 @@ __launch_bounds__(kBlockThreads) __global__ void paged_decode_kernel(
       }
     }
   }
-  if (n_splits > 1 && merge_cnt != nullptr) {
-    // fused merge: the LAST workgroup of this (b, kh) group merges all
-    // G heads' partials (split-k style; the counter self-resets so the
-    // persistent buffer needs zeroing only once, graph-replay safe)
-    // every thread fences ITS OWN partial stores to device scope, then
-    // the block barrier orders the fence before thread 0's increment
-    // (threadFenceReduction pattern)
-    __threadfence();
-    __syncthreads();
-    __shared__ int s_last;
-    if (threadIdx.x == 0) {
-      const int old = atomicAdd(&merge_cnt[b * n_kv + kh], 1);
-      s_last = (old == n_splits - 1);
-      if (s_last) merge_cnt[b * n_kv + kh] = 0;  // reset for next launch
-    }
-    __syncthreads();
-    if (!s_last) return;
-    __threadfence();  // order: observed count -> partial loads
-    for (int g = wave; g < G; g += kWaves) {
-      const int head = kh * G + g;
-      const float* pml = part_ml + ((int64_t)b * n_q + head) * n_splits * 2;
-      const float* po = part_o + ((int64_t)b * n_q + head) * n_splits * HD;
-      float m_star = -INFINITY;
-      for (int s = 0; s < n_splits; ++s)
-        m_star = fmaxf(m_star, pml[2 * s]);
-      float l_star = 0.f, oo[EPL];
-#pragma unroll
-      for (int e = 0; e < EPL; ++e) oo[e] = 0.f;
-      for (int s = 0; s < n_splits; ++s) {
-        const float ms = pml[2 * s];
-        if (ms == -INFINITY) continue;
-        const float c = __expf(ms - m_star);
-        l_star += pml[2 * s + 1] * c;
-#pragma unroll
-        for (int e = 0; e < EPL; ++e)
-          oo[e] += po[s * HD + EPL * lane + e] * c;
-      }
-      const float inv_l = 1.0f / l_star;
-      ushort* oh = out + ((int64_t)b * n_q + head) * HD;
-#pragma unroll
-      for (int e = 0; e < EPL; e += 2) {
-        uint32_t packed =
-            ((uint32_t)f32_to_bf16(oo[e + 1] * inv_l) << 16) |
-            f32_to_bf16(oo[e] * inv_l);
-        *reinterpret_cast<uint32_t*>(&oh[EPL * lane + e]) = packed;
-      }
-    }
-  }
 }
 
 // Merge flash-decoding partials: one wave per (seq, q_head).
@@ -483,15 +432,6 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
   }
   torch::Tensor part_o, part_ml;
   float *part_o_ptr = nullptr, *part_ml_ptr = nullptr;
-  int* merge_cnt_ptr = nullptr;
-  // fused split-merge: the last-arriving workgroup of each (seq, kv
-  // head) merges in-kernel, dropping the second launch. Opt-in while it
-  // soaks (KUBEAI_DECODE_FUSED_MERGE=1); counters are a persistent
-  // self-resetting buffer (zeroed once => graph-replay safe).
-  static const bool fused_merge = []() {
-    const char* e = getenv("KUBEAI_DECODE_FUSED_MERGE");
-    return e != nullptr && e[0] == '1';
-  }();
   if (n_splits > 1) {
     auto opts =
         torch::TensorOptions().device(q.device()).dtype(torch::kFloat32);
@@ -499,24 +439,6 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
     part_ml = torch::empty({B, n_q, n_splits, 2}, opts);
     part_o_ptr = part_o.data_ptr<float>();
     part_ml_ptr = part_ml.data_ptr<float>();
-    if (fused_merge) {
-      static torch::Tensor counters;  // grows monotonically; stays zeroed
-      const int64_t need = (int64_t)B * n_kv;
-      if (!counters.defined() || counters.numel() < need ||
-          counters.device() != q.device()) {
-        hipStreamCaptureStatus cs = hipStreamCaptureStatusNone;
-        (void)hipStreamIsCapturing(
-            c10::hip::getCurrentHIPStream().stream(), &cs);
-        TORCH_CHECK(
-            cs == hipStreamCaptureStatusNone,
-            "fused-merge counter growth during graph capture — warm the "
-            "largest batch bucket before capturing");
-        counters = torch::zeros(
-            {need},
-            torch::TensorOptions().device(q.device()).dtype(torch::kInt32));
-      }
-      merge_cnt_ptr = counters.data_ptr<int>();
-    }
   }
   dim3 grid(B * n_kv * n_splits), block(kBlockThreads);
   auto stream = c10::hip::getCurrentHIPStream().stream();
@@ -539,7 +461,7 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
                      seq_lens.data_ptr<int32_t>(), (float)scale,          \
                      (float)softcap, (int)window, n_kv,                   \
                      max_blocks, q.stride(0), n_splits, part_o_ptr,       \
-                     part_ml_ptr, merge_cnt_ptr)
+                     part_ml_ptr)
 #define LAUNCH(GG)                                                        \
   do {                                                                    \
     if (hd == 256) {                                                      \
@@ -565,7 +487,7 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
 #undef LAUNCH_CT_HD
 #undef LAUNCH_IMPL
   HIP_CHECK_KERNEL();
-  if (n_splits > 1 && merge_cnt_ptr == nullptr) {
+  if (n_splits > 1) {
     const int64_t n_bh = (int64_t)B * n_q;
     const int wpb = 4;
     if (hd == 256)

@@ -219,6 +219,29 @@ class MoEMLP(nn.Module):
         router = F.linear(x.float(), self.gate.weight.float())  # [T, E]
         weights, selected = torch.topk(router, self.top_k, dim=-1)
         weights = torch.softmax(weights, dim=-1).to(x.dtype)  # [T, k]
+        if T <= 256:
+            # decode-sized batches: run EVERY expert densely. At these M
+            # the expert GEMMs are weight-bound, and a batch this size
+            # activates nearly all experts anyway — so the HBM cost is
+            # the same as the sparse gather path, but there are no
+            # dynamic shapes (topk/nonzero gathers), which is what lets
+            # the decode step be hipGraph-captured. (The sparse path's
+            # nonzero() aborts capture -> Mixtral decoded EAGER before
+            # this: 47 ms/step of launch overhead.)
+            wfull = torch.zeros(T, self.n_experts, dtype=x.dtype,
+                                device=x.device)
+            wfull.scatter_(1, selected, weights)
+            fp8 = hasattr(self.experts[0].gate_up_proj, "forward_quantized")
+            if fp8:
+                xq, xs = ops.quant_fp8(x)
+            out = None
+            for e in range(self.n_experts):
+                contrib = (
+                    self.experts[e]((xq, xs)) if fp8 else self.experts[e](x)
+                )
+                contrib = contrib * wfull[:, e : e + 1]
+                out = contrib if out is None else out + contrib
+            return out
         # fp8 experts: quantize the hidden ONCE per layer and hand each
         # expert a row gather of the shared (fp8, scale) pair — per-expert
         # dynamic quantization costs more in launches than fp8 wins back

@@ -1261,6 +1261,8 @@ def main():
     p.add_argument("--num-gpu-blocks", type=int, default=None)
     p.add_argument("--kv-cache-dtype", choices=["auto", "fp8_e5m2"],
                    default="auto")
+    p.add_argument("--quantization", choices=["fp8"], default=None,
+                   help="fp8 = W8A8 dynamic with fused activation quant")
     p.add_argument("--enable-lora", action="store_true")
     p.add_argument("--tensor-parallel-size", type=int, default=1)
     p.add_argument("--task", choices=["generate", "transcribe", "embed"],
@@ -1285,6 +1287,7 @@ def main():
         gpu_memory_utilization=args.gpu_memory_utilization,
         num_gpu_blocks=args.num_gpu_blocks,
         kv_cache_dtype=args.kv_cache_dtype,
+        quantization=args.quantization,
     )
     served = args.served_model_name or os.path.basename(args.model.rstrip("/"))
     tp = args.tensor_parallel_size

@@ -542,3 +542,14 @@ def test_tokenize_detokenize_endpoints(client):
         "messages": [{"role": "user", "content": "hi"}]})
     assert r3.status_code == 200 and r3.json()["count"] > 0
     assert client.post("/tokenize", json={}).status_code == 400
+
+
+def test_server_cli_quantization_flag():
+    import subprocess
+    import sys
+
+    out = subprocess.run(
+        [sys.executable, "-m", "kubeai_amd.engine.server", "--help"],
+        capture_output=True, text=True,
+    )
+    assert "--quantization" in out.stdout

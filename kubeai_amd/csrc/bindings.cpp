@@ -25,8 +25,7 @@ void paged_attention_prefill(torch::Tensor out, torch::Tensor q,
 void silu_and_mul(torch::Tensor out, torch::Tensor x);
 void gelu_and_mul(torch::Tensor out, torch::Tensor x);
 void moe_grouped_fp8(torch::Tensor out, torch::Tensor x_q,
-                     torch::Tensor x_s, std::vector<torch::Tensor> w_q,
-                     torch::Tensor w_s);
+                     std::vector<torch::Tensor> w_q);
 void skinny_gemm(torch::Tensor y, torch::Tensor x, torch::Tensor w);
 void greedy_sample(torch::Tensor out, torch::Tensor logits);
 void gumbel_sample(torch::Tensor out, torch::Tensor logits,

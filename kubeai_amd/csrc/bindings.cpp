@@ -24,8 +24,6 @@ void paged_attention_prefill(torch::Tensor out, torch::Tensor q,
                              int64_t window, double softcap);
 void silu_and_mul(torch::Tensor out, torch::Tensor x);
 void gelu_and_mul(torch::Tensor out, torch::Tensor x);
-void moe_grouped_fp8(torch::Tensor out, torch::Tensor x_q,
-                     std::vector<torch::Tensor> w_q);
 void skinny_gemm(torch::Tensor y, torch::Tensor x, torch::Tensor w);
 void greedy_sample(torch::Tensor out, torch::Tensor logits);
 void gumbel_sample(torch::Tensor out, torch::Tensor logits,
@@ -71,8 +69,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("paged_attention_prefill", &paged_attention_prefill,
         "paged causal flash attention over cached KV");
   m.def("gelu_and_mul", &gelu_and_mul, "GeGLU tanh-gelu(gate)*up");
-  m.def("moe_grouped_fp8", &moe_grouped_fp8,
-        "grouped fp8 expert GEMM (hipBLASLt GroupedGemm)");
   m.def("silu_and_mul", &silu_and_mul, "SwiGLU activation");
   m.def("skinny_gemm", &skinny_gemm, "weight-streaming GEMM for M<=64");
   m.def("greedy_sample", &greedy_sample, "argmax sampling");

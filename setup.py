@@ -50,7 +50,6 @@ setup(
         cpp_extension.CUDAExtension(
             name="kubeai_amd._C",
             sources=sources,
-            libraries=["hipblaslt"],  # grouped expert GEMM (moe_grouped.cpp)
             extra_compile_args={
                 "cxx": ["-O3", "-std=c++17"],
                 "nvcc": [

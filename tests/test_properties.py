@@ -126,3 +126,65 @@ def test_engine_random_traffic_invariants(seed):
     assert s["num_waiting"] == 0 and s["num_running"] == 0
     # all KV blocks returned
     assert all(b.ref_count == 0 for b in eng.scheduler.bm.blocks)
+
+
+@settings(max_examples=20, deadline=None,
+          suppress_health_check=[HealthCheck.too_slow])
+@given(seed=st.integers(0, 2**16))
+def test_priority_admission_order(seed):
+    """With a single running slot, strictly lower `priority` values are
+    admitted first regardless of arrival order (FIFO within a class)."""
+    rng = random.Random(seed)
+    eng = LLMEngine(EngineConfig(
+        model="llama-tiny", device="cpu", num_gpu_blocks=48,
+        enable_graphs=False, max_model_len=128, max_num_seqs=1,
+    ))
+    prios = [rng.choice([0, 1, 5]) for _ in range(5)]
+    for i, p in enumerate(prios):
+        eng.add_request(
+            [1] + [100 + i], SamplingParams(max_tokens=2, ignore_eos=True,
+                                            priority=p),
+            request_id=f"p{i}",
+        )
+    finish_order = []
+    for _ in range(200):
+        if not eng.has_work():
+            break
+        for o in eng.step():
+            if o.finished:
+                finish_order.append(o.request_id)
+    assert len(finish_order) == 5
+    finished_prios = [prios[int(r[1:])] for r in finish_order]
+    assert finished_prios == sorted(finished_prios), (
+        f"priority order violated: {finished_prios}"
+    )
+
+
+@settings(max_examples=15, deadline=None,
+          suppress_health_check=[HealthCheck.too_slow])
+@given(seed=st.integers(0, 2**16),
+       budget=st.sampled_from([8, 16, 32, 64]))
+def test_chunked_prefill_determinism(seed, budget):
+    """Greedy output is invariant to the prefill chunking schedule
+    (max_num_batched_tokens) — chunk boundaries must not leak into
+    attention results."""
+    rng = random.Random(seed)
+    prompt = [1] + [rng.randrange(100, 900) for _ in range(rng.randrange(20, 90))]
+
+    def run(b):
+        eng = LLMEngine(EngineConfig(
+            model="llama-tiny", device="cpu", num_gpu_blocks=64,
+            enable_graphs=False, max_model_len=256,
+            max_num_batched_tokens=b, seed=0,
+        ))
+        eng.add_request(prompt, SamplingParams(max_tokens=5, ignore_eos=True),
+                        request_id="c")
+        for _ in range(300):
+            if not eng.has_work():
+                break
+            for o in eng.step():
+                if o.finished:
+                    return o.output_token_ids
+        raise AssertionError("did not finish")
+
+    assert run(budget) == run(4096)

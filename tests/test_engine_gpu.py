@@ -407,3 +407,31 @@ def test_granite_engine_e2e_gpu():
             if o.finished:
                 out = o
     assert out is not None and len(out.output_token_ids) == 6
+
+
+@pytest.mark.gpu
+def test_mistral_sliding_window_e2e_gpu():
+    """Windowed model end to end on device: prompt longer than the
+    window, greedy reproduction through the prefix cache."""
+    eng = LLMEngine(EngineConfig(model="mistral-tiny", device="cuda",
+                                 num_gpu_blocks=256, max_model_len=512))
+    assert eng.arch.sliding_window == 64
+    prompt = [1] + list(range(100, 220))  # 121 tokens > window
+
+    def run(rid):
+        eng.add_request(prompt, SamplingParams(max_tokens=6, ignore_eos=True),
+                        request_id=rid)
+        out = None
+        for _ in range(100):
+            if not eng.has_work():
+                break
+            for o in eng.step():
+                if o.finished:
+                    out = o
+        return out
+
+    a = run("w1")
+    b = run("w2")
+    assert a is not None and len(a.output_token_ids) == 6
+    assert b.output_token_ids == a.output_token_ids
+    assert b.num_cached_tokens > 0
